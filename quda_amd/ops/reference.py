@@ -1,0 +1,197 @@
+"""CPU/complex-tensor oracle ops (analogue of the reference's
+tests/host_reference/, e.g. wilson_dslash_reference.cpp — reimplemented
+from the operator definitions, not the reference code).
+
+All functions work on the oracle layout:
+  spinor  psi : [V_cb, 4, 3] complex (single parity) or [2, V_cb, 4, 3]
+  gauge   u   : [4, 2, V_cb, 3, 3] complex
+Works on CPU or GPU tensors (pure torch ops) at complex64/complex128.
+
+Conventions (QUDA-compatible):
+  Dslash:  D psi(x) = sum_mu [ U_mu(x) P(-mu) psi(x+mu)
+                             + U_mu(x-mu)^dag P(+mu) psi(x-mu) ]
+  with P(+-mu) = (1 -+ gamma_mu)/2 for dagger=False  (sign flips for dagger).
+  Wilson operator (kappa normalization): M = 1 - kappa * D   (full lattice)
+  Even-odd preconditioned: M_pc = 1 - kappa^2 * D_eo D_oe    (on even sites)
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from ..fields.gamma import GAMMA, GAMMA5, projector, sigma_munu
+from ..fields.geometry import LatticeGeometry
+
+
+def _gamma_tensors(device, dtype):
+    key = (device, dtype)
+    cache = _gamma_tensors.__dict__.setdefault("cache", {})
+    if key not in cache:
+        P = np.stack([np.stack([projector(mu, -1), projector(mu, +1)])
+                      for mu in range(4)])  # [mu, 0:minus/1:plus, 4, 4]
+        cache[key] = torch.tensor(P, dtype=dtype, device=device)
+    return cache[key]
+
+
+def dslash_wilson_parity(u: torch.Tensor, psi: torch.Tensor,
+                         geo: LatticeGeometry, parity: int,
+                         dagger: bool = False) -> torch.Tensor:
+    """Apply the parity-hopping Wilson stencil: out_parity <- D psi_other.
+
+    psi is the SOURCE (at parity 1-parity), result is at `parity`.
+    """
+    dev, dt = psi.device, psi.dtype
+    P = _gamma_tensors(dev, dt)
+    other = 1 - parity
+    out = torch.zeros_like(psi)
+    sgn = 1 if not dagger else 0  # index into P: dagger swaps minus<->plus
+    for mu in range(4):
+        # forward: U_mu(x) P(-mu) psi(x+mu)
+        fwd_idx = geo.neighbor_cb(parity, mu, +1).to(dev)
+        proj = torch.einsum("st,vtc->vsc", P[mu, 1 - sgn], psi[fwd_idx])
+        out += torch.einsum("vij,vsj->vsi", u[mu, parity], proj)
+        # backward: U_mu(x-mu)^dag P(+mu) psi(x-mu)
+        bwd_idx = geo.neighbor_cb(parity, mu, -1).to(dev)
+        proj = torch.einsum("st,vtc->vsc", P[mu, sgn], psi[bwd_idx])
+        out += torch.einsum("vji,vsj->vsi", u[mu, other][bwd_idx].conj(), proj)
+    return out
+
+
+def dslash_wilson_full(u, psi_full, geo, dagger: bool = False):
+    """[2, V, 4, 3] -> [2, V, 4, 3]: out_p = D psi_{1-p} for both parities."""
+    out = torch.empty_like(psi_full)
+    for p in (0, 1):
+        out[p] = dslash_wilson_parity(u, psi_full[1 - p], geo, p, dagger)
+    return out
+
+
+def mat_wilson(u, psi_full, geo, kappa: float, dagger: bool = False):
+    """M psi = psi - kappa * D psi (full lattice, kappa normalization;
+    ref semantics: dirac_wilson.cpp DiracWilson::M)."""
+    return psi_full - kappa * dslash_wilson_full(u, psi_full, geo, dagger)
+
+
+def apply_gamma5(psi: torch.Tensor) -> torch.Tensor:
+    g5 = torch.tensor(np.diag(GAMMA5).real, dtype=psi.real.dtype,
+                      device=psi.device)
+    return psi * g5.view(*([1] * (psi.dim() - 2)), 4, 1)
+
+
+# ---------------------------------------------------------------------------
+# Clover term (ref: lib/clover_quda.cu CloverCompute + dslash_wilson_clover)
+# ---------------------------------------------------------------------------
+
+def field_strength(u: torch.Tensor, geo: LatticeGeometry):
+    """Clover-leaf F_munu (antihermitian traceless part).
+
+    Returns dict {(mu,nu): [2, V_cb, 3, 3] complex} for mu<nu.
+    F_munu = (1/8) sum_leaves (leaf - leaf^dag) (traceless antihermitian)
+    (ref: kernels/field_strength_tensor.cuh — rebuilt from the standard
+    4-leaf clover definition).
+    """
+    dev = u.device
+    # full-lattice link array indexed lexicographically for easy shifting
+    V = geo.volume
+    U = torch.empty((4, V, 3, 3), dtype=u.dtype, device=dev)
+    lo = geo.lex_of_cb.to(dev)
+    for mu in range(4):
+        U[mu, lo[0]] = u[mu, 0]
+        U[mu, lo[1]] = u[mu, 1]
+
+    def shift(f, mu, disp):
+        """f: [V,3,3] field; returns f(x + disp*mu_hat)."""
+        idx = geo.neighbor_lex(mu, disp).to(dev)
+        return f[idx]
+
+    out = {}
+    for mu in range(4):
+        for nu in range(mu + 1, 4):
+            Umu, Unu = U[mu], U[nu]
+            Umu_xnu = shift(Umu, nu, +1)   # U_mu(x+nu)
+            Unu_xmu = shift(Unu, mu, +1)   # U_nu(x+mu)
+            # leaf 1: U_mu(x) U_nu(x+mu) U_mu(x+nu)^d U_nu(x)^d
+            P1 = Umu @ Unu_xmu @ Umu_xnu.conj().mT @ Unu.conj().mT
+            # leaf 2: U_nu(x) U_mu(x-mu+nu)^d U_nu(x-mu)^d U_mu(x-mu)
+            Umu_mx = shift(Umu, mu, -1)
+            Unu_mx = shift(Unu, mu, -1)
+            Umu_mxnu = shift(shift(Umu, mu, -1), nu, +1)
+            P2 = Unu @ Umu_mxnu.conj().mT @ Unu_mx.conj().mT @ Umu_mx
+            # leaf 3: U_mu(x-mu)^d U_nu(x-mu-nu)^d U_mu(x-mu-nu) U_nu(x-nu)
+            Unu_mxmnu = shift(shift(Unu, mu, -1), nu, -1)
+            Umu_mxmnu = shift(shift(Umu, mu, -1), nu, -1)
+            Unu_mnu = shift(Unu, nu, -1)
+            P3 = Umu_mx.conj().mT @ Unu_mxmnu.conj().mT @ Umu_mxmnu @ Unu_mnu
+            # leaf 4: U_nu(x-nu)^d U_mu(x-nu) U_nu(x+mu-nu) U_mu(x)^d
+            Umu_mnu = shift(Umu, nu, -1)
+            Unu_xmu_mnu = shift(shift(Unu, nu, -1), mu, +1)
+            P4 = Unu_mnu.conj().mT @ Umu_mnu @ Unu_xmu_mnu @ Umu.conj().mT
+            Fsum = P1 + P2 + P3 + P4
+            F = (Fsum - Fsum.conj().mT) / 8.0
+            # remove trace
+            tr = torch.diagonal(F, dim1=-2, dim2=-1).sum(-1) / 3.0
+            F = F - tr[..., None, None] * torch.eye(3, dtype=u.dtype, device=dev)
+            # back to cb layout
+            Fcb = torch.stack([F[lo[0]], F[lo[1]]])
+            out[(mu, nu)] = Fcb
+    return out
+
+
+def clover_matrix(u: torch.Tensor, geo: LatticeGeometry, kappa: float,
+                  csw: float) -> torch.Tensor:
+    """Full clover site matrix A = 1 + (kappa*csw/2) * sigma_munu F_munu.
+
+    Returns [2, V_cb, 12, 12] complex (spin x color flattened, s*3+c),
+    hermitian. Block-diagonal in chirality in the DeGrand-Rossi basis.
+    """
+    F = field_strength(u, geo)
+    dev, dt = u.device, u.dtype
+    V = geo.volume_cb
+    A = torch.zeros((2, V, 4, 3, 4, 3), dtype=dt, device=dev)
+    eye = torch.eye(12, dtype=dt, device=dev).reshape(4, 3, 4, 3)
+    A += eye
+    coeff = kappa * csw / 2.0
+    for (mu, nu), Fmn in F.items():
+        sig = torch.tensor(sigma_munu(mu, nu), dtype=dt, device=dev)
+        # sigma acts on spin, i*F on color; sum over mu<nu twice (munu + numu)
+        # sigma_{nu mu} F_{nu mu} = sigma_{mu nu} F_{mu nu} so factor 2... but
+        # standard convention sums mu<nu with both orderings equal:
+        A += 2 * coeff * torch.einsum("st,pvij->pvsitj", sig, Fmn).reshape(2, V, 4, 3, 4, 3)
+    return A.reshape(2, V, 12, 12)
+
+
+def apply_clover(A: torch.Tensor, psi: torch.Tensor, parity=None) -> torch.Tensor:
+    """A psi, site-local 12x12. psi [(.,)V,4,3]; A [2,V,12,12] or [V,12,12]."""
+    if psi.dim() == 4:  # full field [2,V,4,3]
+        out = torch.einsum("pvij,pvj->pvi", A, psi.reshape(2, -1, 12))
+        return out.reshape(psi.shape)
+    Ap = A if A.dim() == 3 else A[parity]
+    out = torch.einsum("vij,vj->vi", Ap, psi.reshape(-1, 12))
+    return out.reshape(psi.shape)
+
+
+# ---------------------------------------------------------------------------
+# Gauge observables
+# ---------------------------------------------------------------------------
+
+def plaquette(u: torch.Tensor, geo: LatticeGeometry):
+    """Mean plaquette Re tr P / 3: returns (total, spatial, temporal)
+    (ref: lib/gauge_plaq.cu semantics)."""
+    dev = u.device
+    V = geo.volume
+    U = torch.empty((4, V, 3, 3), dtype=u.dtype, device=dev)
+    lo = geo.lex_of_cb.to(dev)
+    for mu in range(4):
+        U[mu, lo[0]] = u[mu, 0]
+        U[mu, lo[1]] = u[mu, 1]
+    sp, tp = [], []
+    for mu in range(4):
+        for nu in range(mu + 1, 4):
+            Unu_xmu = U[nu][geo.neighbor_lex(mu, +1).to(dev)]
+            Umu_xnu = U[mu][geo.neighbor_lex(nu, +1).to(dev)]
+            P = U[mu] @ Unu_xmu @ Umu_xnu.conj().mT @ U[nu].conj().mT
+            val = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).real.mean().item() / 3.0
+            (tp if nu == 3 else sp).append(val)
+    s = sum(sp) / len(sp)
+    t = sum(tp) / len(tp)
+    return (s + t) / 2, s, t

@@ -1,0 +1,81 @@
+"""Field layer tests: gamma algebra, layouts, geometry indexing."""
+import numpy as np
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.fields import gamma
+
+
+def test_gamma_algebra():
+    assert gamma.check_algebra()
+
+
+def test_gamma5_product():
+    g5 = gamma.GAMMA[0] @ gamma.GAMMA[1] @ gamma.GAMMA[2] @ gamma.GAMMA[3]
+    assert np.allclose(g5, gamma.GAMMA5)
+
+
+def test_geometry_parity_counts(rect_geo):
+    geo = rect_geo
+    assert geo.parity.to(torch.int64).sum().item() == geo.volume_cb
+    # cb <-> lex roundtrip
+    lo = geo.lex_of_cb
+    assert torch.unique(lo.flatten()).numel() == geo.volume
+
+
+def test_neighbor_tables_inverse(rect_geo):
+    geo = rect_geo
+    for mu in range(4):
+        fwd = geo.neighbor_lex(mu, +1)
+        bwd = geo.neighbor_lex(mu, -1)
+        idx = torch.arange(geo.volume)
+        assert torch.equal(bwd[fwd], idx)
+        assert torch.equal(fwd[bwd], idx)
+
+
+def test_neighbor_cb_parity_flip(small_geo):
+    geo = small_geo
+    for p in (0, 1):
+        for mu in range(4):
+            nbr = geo.neighbor_cb(p, mu, +1)
+            # displaced site must live on the opposite parity
+            nbr_lex = geo.lex_of_cb[1 - p][nbr]
+            expect = geo.neighbor_lex(mu, +1)[geo.lex_of_cb[p]]
+            assert torch.equal(nbr_lex, expect)
+
+
+@pytest.mark.parametrize("prec", ["double", "single", "half"])
+def test_spinor_roundtrip(small_geo, prec):
+    s = SpinorField(small_geo, "double").gaussian_(seed=3)
+    c = s.to_complex()
+    s2 = SpinorField(small_geo, prec).from_complex(c)
+    back = s2.to_complex()
+    tol = {"double": 1e-14, "single": 1e-6, "half": 1e-3}[prec]
+    assert (back - c).abs().max().item() <= tol * c.abs().max().item()
+
+
+@pytest.mark.parametrize("recon", ["none", "twelve"])
+@pytest.mark.parametrize("prec", ["double", "single"])
+def test_gauge_roundtrip(small_geo, prec, recon):
+    g = GaugeField(small_geo, "double").random_su3_(seed=4)
+    u = g.to_complex()
+    g2 = GaugeField(small_geo, prec, reconstruct=recon).from_complex(u)
+    back = g2.to_complex()
+    tol = {"double": 1e-12, "single": 1e-5}[prec]
+    assert (back - u).abs().max().item() <= tol
+
+
+def test_gauge_su3(small_geo):
+    g = GaugeField(small_geo, "double").random_su3_(seed=5)
+    u = g.to_complex()
+    eye = torch.eye(3, dtype=torch.complex128)
+    assert (u @ u.conj().mT - eye).abs().max().item() < 1e-12
+    assert (torch.linalg.det(u) - 1).abs().max().item() < 1e-12
+
+
+def test_parity_view_shares_storage(small_geo):
+    s = SpinorField(small_geo, "double").gaussian_(seed=6)
+    e = s.parity_view(0)
+    e.data.mul_(2.0)
+    assert torch.allclose(s.data[0], e.data[0])
