@@ -127,7 +127,8 @@ def field_strength(u: torch.Tensor, geo: LatticeGeometry):
             Unu_xmu_mnu = shift(shift(Unu, nu, -1), mu, +1)
             P4 = Unu_mnu.conj().mT @ Umu_mnu @ Unu_xmu_mnu @ Umu.conj().mT
             Fsum = P1 + P2 + P3 + P4
-            F = (Fsum - Fsum.conj().mT) / 8.0
+            # hermitian field strength: F = (Q - Q^dag) / (8i)
+            F = (Fsum - Fsum.conj().mT) / 8.0j
             # remove trace
             tr = torch.diagonal(F, dim1=-2, dim2=-1).sum(-1) / 3.0
             F = F - tr[..., None, None] * torch.eye(3, dtype=u.dtype, device=dev)
