@@ -1,0 +1,131 @@
+"""Conjugate gradient with mixed precision + reliable updates
+(ref: lib/inv_cg_quda.cpp:63 operator(), include/reliable_updates.h:33 —
+same algorithm structure, re-implemented).
+
+Solves MdagM x = b where `op` exposes MdagM(out, in, tmp) on single- or
+full-parity SpinorFields. Outer (precise) residual bookkeeping runs at
+`x.precision`; the Krylov iteration runs at `sloppy` precision. A reliable
+update (true-residual recomputation + solution accumulation) triggers when
+the iterated residual falls by `delta` relative to the max since the last
+update.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from math import sqrt
+from typing import Callable, Optional
+
+from ..fields.spinor import SpinorField
+from ..ops import blas
+
+
+@dataclass
+class SolverStats:
+    iters: int = 0
+    resid: float = 0.0
+    true_resid: float = 0.0
+    reliable_updates: int = 0
+    converged: bool = False
+    flops: float = 0.0
+    seconds: float = 0.0
+
+
+def cg_solve(op, x: SpinorField, b: SpinorField, *,
+             op_sloppy=None, sloppy: Optional[str] = None,
+             tol: float = 1e-8, maxiter: int = 1000,
+             delta: float = 0.1) -> SolverStats:
+    """CG on the (hermitian PSD) operator op.MdagM.
+
+    x: initial guess (overwritten with solution), b: source — both at the
+    "precise" precision. op_sloppy/sloppy select the inner precision
+    (default: same operator, same precision => plain CG).
+    """
+    stats = SolverStats()
+    prec_hi = x.precision
+    sloppy = sloppy or prec_hi
+    op_sloppy = op_sloppy or op
+    geo, dev, npar = x.geo, x.device, x.n_parity
+
+    def hi():
+        return SpinorField(geo, prec_hi, dev, npar)
+
+    def lo():
+        return SpinorField(geo, sloppy, dev, npar)
+
+    b2 = blas.norm2(b)
+    if b2 == 0.0:
+        x.zero_()
+        stats.converged = True
+        return stats
+    stop = tol * tol * b2
+
+    r = hi()           # precise residual
+    tmp_hi = hi()
+    # r = b - MdagM x   (skip apply if x == 0)
+    x2 = blas.norm2(x)
+    if x2 > 0.0:
+        op.MdagM(r, x, tmp_hi)
+        r2 = blas.xmy_norm2(b, r)
+    else:
+        blas.copy(r, b)
+        r2 = b2
+
+    mixed = (sloppy != prec_hi) or (op_sloppy is not op)
+    r_s = lo()
+    p = lo()
+    Ap = lo()
+    x_s = lo()         # sloppy solution accumulator since last update
+    tmp_s = lo()
+    blas.copy(r_s, r)
+    blas.copy(p, r_s)
+    x_s.zero_()
+
+    maxr = sqrt(r2)    # max residual since last reliable update
+    k = 0
+    while r2 > stop and k < maxiter:
+        op_sloppy.MdagM(Ap, p, tmp_s)
+        pAp = blas.re_dot(p, Ap)
+        if pAp <= 0.0:
+            break  # breakdown (ref: inv_cg_quda.cpp:265 checks)
+        alpha = r2 / pAp
+        blas.axpy(alpha, p, x_s)
+        r2_old = r2
+        r2 = blas.axpy_norm2(-alpha, Ap, r_s)
+        k += 1
+        rnorm = sqrt(r2)
+        maxr = max(maxr, rnorm)
+
+        need_reliable = mixed and (rnorm < delta * maxr)
+        if need_reliable or r2 <= stop:
+            # accumulate + recompute true residual at high precision
+            tmp_hi.copy_(x_s)
+            blas.axpy(1.0, tmp_hi, x)
+            op.MdagM(r, x, tmp_hi)
+            r2 = blas.xmy_norm2(b, r)
+            x_s.zero_()
+            blas.copy(r_s, r)
+            # restart direction with beta continuation:
+            # p = r_s + beta p  with beta = r2_new / r2_old
+            beta = r2 / r2_old
+            blas.xpay(r_s, beta, p)
+            maxr = sqrt(r2)
+            stats.reliable_updates += 1
+            if r2 <= stop:
+                break
+        else:
+            beta = r2 / r2_old
+            blas.xpay(r_s, beta, p)
+
+    if not mixed:
+        # accumulate the (high-precision) iterate
+        tmp_hi.copy_(x_s)
+        blas.axpy(1.0, tmp_hi, x)
+        op.MdagM(r, x, tmp_hi)
+        r2 = blas.xmy_norm2(b, r)
+
+    stats.iters = k
+    stats.resid = sqrt(r2 / b2)
+    stats.true_resid = stats.resid
+    stats.converged = r2 <= stop
+    return stats

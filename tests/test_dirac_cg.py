@@ -1,0 +1,163 @@
+"""Dirac operator + CG solver tests on the CPU oracle path."""
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.fields.clover import CloverField
+from quda_amd.models import (DiracClover, DiracCloverPC, DiracWilson,
+                             DiracWilsonPC)
+from quda_amd.ops import blas
+from quda_amd.ops import reference as ref
+from quda_amd.solvers import cg_solve
+
+KAPPA = 0.12
+CSW = 1.0
+
+
+@pytest.fixture
+def setup(small_geo):
+    g = GaugeField(small_geo, "double").random_su3_(seed=21)
+    b = SpinorField(small_geo, "double").gaussian_(seed=22)
+    return small_geo, g, b
+
+
+def _clover(geo, g):
+    A = ref.clover_matrix(g.to_complex(), geo, KAPPA, CSW)
+    return CloverField(geo, "double").from_matrices(A)
+
+
+def test_wilson_M_matches_oracle(setup):
+    geo, g, b = setup
+    d = DiracWilson(g, KAPPA)
+    out = SpinorField(geo, "double", n_parity=2)
+    d.M(out, b)
+    expect = ref.mat_wilson(g.to_complex(), b.to_complex(), geo, KAPPA)
+    assert (out.to_complex() - expect).abs().max().item() < 1e-12
+
+
+def test_wilson_M_dagger_is_g5Mg5(setup):
+    geo, g, b = setup
+    d = DiracWilson(g, KAPPA)
+    out = SpinorField(geo, "double", n_parity=2)
+    d.M(out, b, dagger=True)
+    u = g.to_complex()
+    g5Mg5 = ref.apply_gamma5(
+        ref.mat_wilson(u, ref.apply_gamma5(b.to_complex()), geo, KAPPA))
+    assert (out.to_complex() - g5Mg5).abs().max().item() < 1e-12
+
+
+def test_wilson_pc_consistency(setup):
+    """M_pc in_e == in_e - kappa^2 D_eo D_oe in_e via the oracle."""
+    geo, g, b = setup
+    d = DiracWilsonPC(g, KAPPA)
+    be = b.parity_view(0)
+    out = SpinorField(geo, "double", n_parity=1)
+    d.M(out, be)
+    u = g.to_complex()
+    in_e = be.to_complex()[0]
+    Doe = ref.dslash_wilson_parity(u, in_e, geo, 1)
+    DeoDoe = ref.dslash_wilson_parity(u, Doe, geo, 0)
+    expect = in_e - KAPPA ** 2 * DeoDoe
+    assert (out.to_complex()[0] - expect).abs().max().item() < 1e-12
+
+
+@pytest.mark.parametrize("pc", [False, True])
+def test_cg_wilson(setup, pc):
+    geo, g, b = setup
+    if pc:
+        d = DiracWilsonPC(g, KAPPA)
+        rhs = b.parity_view(0)
+        x = SpinorField(geo, "double", n_parity=1)
+    else:
+        d = DiracWilson(g, KAPPA)
+        rhs = b
+        x = SpinorField(geo, "double", n_parity=2)
+    stats = cg_solve(d, x, rhs, tol=1e-10, maxiter=500)
+    assert stats.converged, stats
+    # independent residual check: ||MdagM x - b|| / ||b||
+    out = x.clone_empty()
+    tmp = x.clone_empty()
+    d.MdagM(out, x, tmp)
+    r = (out.to_complex() - rhs.to_complex()).abs().max().item()
+    assert r < 1e-8
+
+
+def test_cg_mixed_precision(setup):
+    geo, g, b = setup
+    d = DiracWilsonPC(g, KAPPA)
+    rhs = b.parity_view(0)
+    x = SpinorField(geo, "double", n_parity=1)
+    stats = cg_solve(d, x, rhs, sloppy="single", tol=1e-10, maxiter=500)
+    assert stats.converged
+    assert stats.reliable_updates > 0
+    out = x.clone_empty()
+    tmp = x.clone_empty()
+    d.MdagM(out, x, tmp)
+    rel = ((out.to_complex() - rhs.to_complex()).abs().max().item()
+           / rhs.to_complex().abs().max().item())
+    assert rel < 1e-7
+
+
+def test_wilson_pc_prepare_reconstruct(setup):
+    """Full-lattice Wilson solve through the even-odd system."""
+    geo, g, b = setup
+    dpc = DiracWilsonPC(g, KAPPA)
+    be = dpc.prepare(b)
+    xe = SpinorField(geo, "double", n_parity=1)
+    stats = cg_solve(dpc, xe, be, tol=1e-12, maxiter=1000)
+    assert stats.converged
+    # CG solved MdagM xe = be => apply Mdag to get solution of M_pc y = be
+    y = xe.clone_empty()
+    dpc.M(y, xe, dagger=False)
+    # wait: MdagM x = b => x solves the normal equation; y = M x solves
+    # Mdag y = b. For the PC solve we need M_pc y = be, i.e. solve via
+    # CGNR: x = M^dag z. Instead verify the normal-equation residual and
+    # the full-lattice reconstruction with the NE solution:
+    xfull = SpinorField(geo, "double", n_parity=2)
+    # solution of M_pc xe' = be is xe' = (MdagM)^-1 Mdag be: redo with Mdag b
+    bed = be.clone_empty()
+    dpc.M(bed, be, dagger=True)
+    xe2 = SpinorField(geo, "double", n_parity=1)
+    stats2 = cg_solve(dpc, xe2, bed, tol=1e-12, maxiter=1000)
+    assert stats2.converged
+    dpc.reconstruct(xfull, xe2, b)
+    # check M_full x = b via oracle
+    u = g.to_complex()
+    Mx = ref.mat_wilson(u, xfull.to_complex(), geo, KAPPA)
+    rel = (Mx - b.to_complex()).abs().max().item() / b.to_complex().abs().max().item()
+    assert rel < 1e-8
+
+
+def test_clover_pc_cg(setup):
+    geo, g, b = setup
+    cl = _clover(geo, g)
+    dpc = DiracCloverPC(g, cl, KAPPA)
+    be = dpc.prepare(b)
+    bed = be.clone_empty()
+    dpc.M(bed, be, dagger=True)
+    xe = SpinorField(geo, "double", n_parity=1)
+    stats = cg_solve(dpc, xe, bed, tol=1e-12, maxiter=2000)
+    assert stats.converged
+    xfull = SpinorField(geo, "double", n_parity=2)
+    dpc.reconstruct(xfull, xe, b)
+    # verify against full-lattice clover operator M = A - kappa D
+    dfull = DiracClover(g, cl, KAPPA)
+    Mx = SpinorField(geo, "double", n_parity=2)
+    dfull.M(Mx, xfull)
+    rel = ((Mx.to_complex() - b.to_complex()).abs().max().item()
+           / b.to_complex().abs().max().item())
+    assert rel < 1e-8
+
+
+def test_clover_full_vs_blocks(setup):
+    """DiracClover.M == A psi - kappa D psi via oracle pieces."""
+    geo, g, b = setup
+    cl = _clover(geo, g)
+    d = DiracClover(g, cl, KAPPA)
+    out = SpinorField(geo, "double", n_parity=2)
+    d.M(out, b)
+    u = g.to_complex()
+    A = ref.clover_matrix(u, geo, KAPPA, CSW)
+    expect = (ref.apply_clover(A, b.to_complex())
+              - KAPPA * ref.dslash_wilson_full(u, b.to_complex(), geo))
+    assert (out.to_complex() - expect).abs().max().item() < 1e-12
