@@ -1,0 +1,119 @@
+// pybind11/torch bindings for the quda_amd HIP kernels (in-tree extension).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "launchers.h"
+
+namespace {
+
+int prec_of(const at::Tensor &t) {
+  switch (t.scalar_type()) {
+    case at::kDouble: return 0;
+    case at::kFloat: return 1;
+    case at::kHalf: return 2;
+    default: TORCH_CHECK(false, "unsupported dtype");
+  }
+}
+
+void *ptr_or_null(const at::Tensor &t) {
+  return t.numel() ? t.data_ptr() : nullptr;
+}
+
+BlasField field_of(const at::Tensor &data, const at::Tensor &norm, long Vcb) {
+  return BlasField{ptr_or_null(data), ptr_or_null(norm), Vcb};
+}
+
+hipStream_t stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+}  // namespace
+
+static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                          at::Tensor in_n, at::Tensor gauge, at::Tensor clover,
+                          at::Tensor x, at::Tensor x_n,
+                          std::vector<int64_t> dims, int64_t parity_offset,
+                          int64_t Vcb, int64_t parity, bool dagger,
+                          int64_t mode, bool xpay, double a, int64_t recon) {
+  TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
+  DslashCall c{};
+  c.out = field_of(out, out_n, Vcb);
+  c.in = field_of(in, in_n, Vcb);
+  c.x = field_of(x, x_n, Vcb);
+  c.gauge = gauge.data_ptr();
+  c.clover = ptr_or_null(clover);
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.dagger = dagger;
+  c.mode = (int)mode;
+  c.xpay = xpay;
+  c.a = a;
+  c.recon = (int)recon;
+  switch (prec_of(out)) {
+    case 0: launch_dslash_wilson_double(c, stream()); break;
+    case 1: launch_dslash_wilson_single(c, stream()); break;
+    case 2: launch_dslash_wilson_half(c, stream()); break;
+  }
+}
+
+static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
+                          at::Tensor x_n, at::Tensor y, at::Tensor y_n,
+                          int64_t Vcb, int64_t sites) {
+  BlasCall c{};
+  c.op = (int)op;
+  c.prec = prec_of(x);
+  c.a = a;
+  c.b = b;
+  c.x = field_of(x, x_n, Vcb);
+  c.y = field_of(y, y_n, Vcb);
+  c.sites = sites;
+  at::Tensor result;
+  bool reduction = (op == BLAS_AXPY_NORM2 || op == BLAS_XMY_NORM2 ||
+                    op == BLAS_NORM2 || op == BLAS_REDOT || op == BLAS_CDOT);
+  if (reduction) {
+    result = at::zeros({2}, x.options().dtype(at::kDouble));
+    c.result = result.data_ptr<double>();
+  }
+  launch_blas(c, stream());
+  return result;
+}
+
+static void convert(at::Tensor dst, at::Tensor dst_n, at::Tensor src,
+                    at::Tensor src_n, int64_t Vcb, int64_t sites) {
+  launch_convert(field_of(dst, dst_n, Vcb), prec_of(dst),
+                 field_of(src, src_n, Vcb), prec_of(src), sites, stream());
+}
+
+static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                         at::Tensor in_n, at::Tensor clover, int64_t parity,
+                         int64_t Vcb) {
+  CloverApplyCall c{};
+  c.out = field_of(out, out_n, Vcb);
+  c.in = field_of(in, in_n, Vcb);
+  c.clover = clover.data_ptr();
+  c.parity = (int)parity;
+  c.Vcb = Vcb;
+  c.sites = Vcb;
+  c.prec = prec_of(out);
+  launch_clover_apply(c, stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover) dslash");
+  m.def("blas_op", &blas_op, "fused blas/reduction");
+  m.def("convert", &convert, "precision conversion copy");
+  m.def("clover_apply", &clover_apply, "clover site-matrix apply");
+  m.attr("BLAS_AXPY") = (int)BLAS_AXPY;
+  m.attr("BLAS_AXPY_NORM2") = (int)BLAS_AXPY_NORM2;
+  m.attr("BLAS_XPAY") = (int)BLAS_XPAY;
+  m.attr("BLAS_AXPBY") = (int)BLAS_AXPBY;
+  m.attr("BLAS_CAXPY") = (int)BLAS_CAXPY;
+  m.attr("BLAS_XMY_NORM2") = (int)BLAS_XMY_NORM2;
+  m.attr("BLAS_SCAL") = (int)BLAS_SCAL;
+  m.attr("BLAS_NORM2") = (int)BLAS_NORM2;
+  m.attr("BLAS_REDOT") = (int)BLAS_REDOT;
+  m.attr("BLAS_CDOT") = (int)BLAS_CDOT;
+}

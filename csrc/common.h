@@ -1,0 +1,251 @@
+// quda_amd HIP device common: complex math, chunked-SoA field accessors,
+// checkerboard index helpers. MI355X (gfx950) only — wave64, 16-byte
+// vectorized global access (cdna_hip_programming.md G2/G13).
+//
+// Layouts match quda_amd/fields/layout.py:
+//   spinor: [n_chunk][V_cb][W] reals, comp = (s*3+c)*2+reim, W*sizeof = 16B
+//   gauge : [mu][parity][n_chunk][V_cb][2] complex-pair chunks
+//   clover: [parity][n_chunk][V_cb][W], 72 reals packed (fields/clover.py)
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+
+// ---------------------------------------------------------------------------
+// complex
+// ---------------------------------------------------------------------------
+template <typename T>
+struct cplx {
+  T re, im;
+  __device__ __forceinline__ cplx() {}
+  __device__ __forceinline__ cplx(T r, T i) : re(r), im(i) {}
+  __device__ __forceinline__ cplx operator+(const cplx &o) const { return {re + o.re, im + o.im}; }
+  __device__ __forceinline__ cplx operator-(const cplx &o) const { return {re - o.re, im - o.im}; }
+  __device__ __forceinline__ cplx operator*(const cplx &o) const {
+    return {re * o.re - im * o.im, re * o.im + im * o.re};
+  }
+  __device__ __forceinline__ cplx &operator+=(const cplx &o) { re += o.re; im += o.im; return *this; }
+};
+
+template <typename T>
+__device__ __forceinline__ cplx<T> operator*(T a, const cplx<T> &z) { return {a * z.re, a * z.im}; }
+template <typename T>
+__device__ __forceinline__ cplx<T> imul(const cplx<T> &z) { return {-z.im, z.re}; }   // i*z
+template <typename T>
+__device__ __forceinline__ cplx<T> mimul(const cplx<T> &z) { return {z.im, -z.re}; }  // -i*z
+template <typename T>
+__device__ __forceinline__ cplx<T> neg(const cplx<T> &z) { return {-z.re, -z.im}; }
+template <typename T>
+__device__ __forceinline__ cplx<T> conj(const cplx<T> &z) { return {z.re, -z.im}; }
+// fused a*b + c
+template <typename T>
+__device__ __forceinline__ cplx<T> cfma(const cplx<T> &a, const cplx<T> &b, const cplx<T> &c) {
+  return {fma(a.re, b.re, fma(-a.im, b.im, c.re)), fma(a.re, b.im, fma(a.im, b.re, c.im))};
+}
+// conj(a)*b + c
+template <typename T>
+__device__ __forceinline__ cplx<T> cfma_conj(const cplx<T> &a, const cplx<T> &b, const cplx<T> &c) {
+  return {fma(a.re, b.re, fma(a.im, b.im, c.re)), fma(a.re, b.im, fma(-a.im, b.re, c.im))};
+}
+
+// ---------------------------------------------------------------------------
+// storage traits: Store = memory type, Real = compute type
+// ---------------------------------------------------------------------------
+struct PrecDouble { using Store = double; using Real = double; static constexpr int W = 2;  static constexpr bool has_norm = false; };
+struct PrecSingle { using Store = float;  using Real = float;  static constexpr int W = 4;  static constexpr bool has_norm = false; };
+struct PrecHalf   { using Store = __half; using Real = float;  static constexpr int W = 8;  static constexpr bool has_norm = true;  };
+
+// 16-byte opaque chunk for vector loads
+struct alignas(16) chunk16 { unsigned int u[4]; };
+
+template <typename S, int W>
+__device__ __forceinline__ void load_chunk(const S *p, S out[W]) {
+  if constexpr (W * sizeof(S) == 16) {
+    *reinterpret_cast<chunk16 *>(out) = *reinterpret_cast<const chunk16 *>(p);
+  } else if constexpr (W * sizeof(S) == 8) {
+    *reinterpret_cast<unsigned long long *>(out) = *reinterpret_cast<const unsigned long long *>(p);
+  } else {
+    *reinterpret_cast<unsigned int *>(out) = *reinterpret_cast<const unsigned int *>(p);
+  }
+}
+
+template <typename S, int W>
+__device__ __forceinline__ void store_chunk(S *p, const S in[W]) {
+  if constexpr (W * sizeof(S) == 16) {
+    *reinterpret_cast<chunk16 *>(p) = *reinterpret_cast<const chunk16 *>(in);
+  } else if constexpr (W * sizeof(S) == 8) {
+    *reinterpret_cast<unsigned long long *>(p) = *reinterpret_cast<const unsigned long long *>(in);
+  } else {
+    *reinterpret_cast<unsigned int *>(p) = *reinterpret_cast<const unsigned int *>(in);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// spinor accessor: 24 reals/site in [24/W][V][W]
+// ---------------------------------------------------------------------------
+template <typename Prec>
+struct SpinorAcc {
+  using S = typename Prec::Store;
+  using R = typename Prec::Real;
+  static constexpr int W = Prec::W;
+  static constexpr int NCH = 24 / W;
+  S *data;
+  float *norm;  // only for half
+  long V;       // cb volume (chunk stride); fields span npar * V sites
+
+  // global site g in [0, npar*V): parity block p = g/V without a divide
+  __device__ __forceinline__ long chunk_base(long g) const {
+    long p = (g >= V) ? 1 : 0;
+    return (p * NCH * V + (g - p * V)) * W;
+  }
+
+  __device__ __forceinline__ void load_g(cplx<R> out[4][3], long g) const {
+    S tmp[24];
+    long base = chunk_base(g);
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      load_chunk<S, W>(data + base + (long)ch * V * W, tmp + ch * W);
+    R scale = (R)1;
+    if constexpr (Prec::has_norm) scale = norm[g];
+#pragma unroll
+    for (int k = 0; k < 12; ++k)
+      out[k / 3][k % 3] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+  }
+
+  __device__ __forceinline__ void store_g(const cplx<R> in[4][3], long g) const {
+    S tmp[24];
+    long base = chunk_base(g);
+    if constexpr (Prec::has_norm) {
+      R m = (R)0;
+#pragma unroll
+      for (int k = 0; k < 12; ++k)
+        m = fmax(m, fmax(fabs(in[k / 3][k % 3].re), fabs(in[k / 3][k % 3].im)));
+      norm[g] = m;
+      R inv = m > (R)0 ? (R)1 / m : (R)0;
+#pragma unroll
+      for (int k = 0; k < 12; ++k) {
+        tmp[2 * k] = (S)(in[k / 3][k % 3].re * inv);
+        tmp[2 * k + 1] = (S)(in[k / 3][k % 3].im * inv);
+      }
+    } else {
+#pragma unroll
+      for (int k = 0; k < 12; ++k) {
+        tmp[2 * k] = (S)in[k / 3][k % 3].re;
+        tmp[2 * k + 1] = (S)in[k / 3][k % 3].im;
+      }
+    }
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      store_chunk<S, W>(data + base + (long)ch * V * W, tmp + ch * W);
+  }
+
+  // single-parity (dslash) aliases: site index i in [0, V)
+  __device__ __forceinline__ void load(cplx<R> out[4][3], long i) const { load_g(out, i); }
+  __device__ __forceinline__ void store(const cplx<R> in[4][3], long i) const { store_g(in, i); }
+};
+
+// ---------------------------------------------------------------------------
+// gauge accessor: [mu][parity][nch][V][2] complex-pair chunks; recon 18|12
+// (recon-12: row2 = conj(row0 x row1), ref gauge_field_order.h:2369)
+// ---------------------------------------------------------------------------
+template <typename Prec, int RECON>
+struct GaugeAcc {
+  using S = typename Prec::Store;
+  using R = typename Prec::Real;
+  static constexpr int NCH = RECON / 2;  // complex chunks per link
+  const S *data;
+  long V;
+
+  __device__ __forceinline__ void load(cplx<R> u[3][3], int mu, int parity, long i) const {
+    const S *base = data + (((long)(mu * 2 + parity) * NCH) * V + i) * 2;
+    S tmp[RECON];
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      load_chunk<S, 2>(base + (long)ch * V * 2, tmp + ch * 2);
+#pragma unroll
+    for (int k = 0; k < RECON / 2; ++k)
+      u[k / 3][k % 3] = {(R)tmp[2 * k], (R)tmp[2 * k + 1]};
+    if constexpr (RECON == 12) {
+      // row2 = conj(row0 x row1)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) {
+        int c1 = (c + 1) % 3, c2 = (c + 2) % 3;
+        u[2][c] = conj(u[0][c1] * u[1][c2] - u[0][c2] * u[1][c1]);
+      }
+    }
+  }
+};
+
+// U * h (3x3 times 2x3 half-spinor, h indexed [spin][color])
+template <typename R>
+__device__ __forceinline__ void su3_mul_half(cplx<R> out[2][3], const cplx<R> u[3][3],
+                                             const cplx<R> h[2][3]) {
+#pragma unroll
+  for (int s = 0; s < 2; ++s)
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      cplx<R> acc = u[r][0] * h[s][0];
+      acc = cfma(u[r][1], h[s][1], acc);
+      acc = cfma(u[r][2], h[s][2], acc);
+      out[s][r] = acc;
+    }
+}
+
+// U^dag * h
+template <typename R>
+__device__ __forceinline__ void su3_dagmul_half(cplx<R> out[2][3], const cplx<R> u[3][3],
+                                                const cplx<R> h[2][3]) {
+#pragma unroll
+  for (int s = 0; s < 2; ++s)
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      cplx<R> acc = cfma_conj(u[0][r], h[s][0], cplx<R>((R)0, (R)0));
+      acc = cfma_conj(u[1][r], h[s][1], acc);
+      acc = cfma_conj(u[2][r], h[s][2], acc);
+      out[s][r] = acc;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// checkerboard coordinates (ref: include/index_helper.cuh coordsFromIndex)
+// ---------------------------------------------------------------------------
+struct LatDims {
+  int X[4];
+  int parity_offset;
+  long Vcb;
+};
+
+__device__ __forceinline__ void coords_from_cb(int x[4], long i, const LatDims &d, int parity) {
+  int X0h = d.X[0] >> 1;
+  x[0] = (int)(i % X0h);
+  long j = i / X0h;
+  x[1] = (int)(j % d.X[1]);
+  j /= d.X[1];
+  x[2] = (int)(j % d.X[2]);
+  x[3] = (int)(j / d.X[2]);
+  int odd = (x[1] + x[2] + x[3] + parity + d.parity_offset) & 1;
+  x[0] = 2 * x[0] + odd;
+}
+
+__device__ __forceinline__ long cb_from_coords(const int x[4], const LatDims &d) {
+  long lex = (((long)x[3] * d.X[2] + x[2]) * d.X[1] + x[1]) * d.X[0] + x[0];
+  return lex >> 1;
+}
+
+// neighbor cb index with periodic wrap in local lattice; dir = +1/-1
+__device__ __forceinline__ long neighbor_cb(const int x[4], int mu, int dir, const LatDims &d) {
+  int y[4] = {x[0], x[1], x[2], x[3]};
+  int v = y[mu] + dir;
+  if (v >= d.X[mu]) v -= d.X[mu];
+  if (v < 0) v += d.X[mu];
+  y[mu] = v;
+  return cb_from_coords(y, d);
+}
+
+#define HIP_CHECK(cmd)                                                         \
+  do {                                                                         \
+    hipError_t e = (cmd);                                                      \
+    if (e != hipSuccess) {                                                     \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(e), __FILE__, __LINE__); \
+    }                                                                          \
+  } while (0)

@@ -1,0 +1,179 @@
+// Wilson / Wilson-clover dslash stencil for MI355X (gfx950).
+// (role of reference include/kernels/dslash_wilson.cuh applyWilson +
+//  dslash_wilson_clover_preconditioned.cuh — redesigned: one thread per
+//  output site, grid-stride, 16B vector loads, fused clover/xpay epilogues)
+//
+// Kernel modes:
+//   PLAIN      : out = [x +] a * (D in)            (XPAY: x term present)
+//   CLOV_POST  : out = [x +] a * (A_clov (D in))   (A = packed site matrix,
+//                usually the inverse: even-odd preconditioned operator)
+//   CLOV_X     : out = A_clov x + a * (D in)       (full clover M, a = -kappa)
+//
+// so DiracWilsonPC::M = 2 launches, DiracCloverPC::M = 2 launches.
+#pragma once
+
+#include "common.h"
+#include "generated/proj.h"
+
+enum CloverMode { PLAIN = 0, CLOV_POST = 1, CLOV_X = 2 };
+
+// packed clover (fields/clover.py): per site 72 reals =
+// 2 chirality blocks x (6 diag + 15 lower-tri complex)
+template <typename Prec>
+struct CloverAcc {
+  using S = typename Prec::Store;
+  using R = typename Prec::Real;
+  static constexpr int W = Prec::W;
+  static constexpr int NCH = 72 / W;
+  const S *data;  // [parity][NCH][V][W]
+  long V;
+
+  __device__ __forceinline__ void load(R diag[2][6], cplx<R> tri[2][15],
+                                       int parity, long i) const {
+    S tmp[72];
+    const S *base = data + ((long)parity * NCH * V + i) * W;
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      load_chunk<S, W>(base + (long)ch * V * W, tmp + ch * W);
+#pragma unroll
+    for (int b = 0; b < 2; ++b) {
+#pragma unroll
+      for (int k = 0; k < 6; ++k) diag[b][k] = (R)tmp[36 * b + k];
+#pragma unroll
+      for (int k = 0; k < 15; ++k)
+        tri[b][k] = {(R)tmp[36 * b + 6 + 2 * k], (R)tmp[36 * b + 6 + 2 * k + 1]};
+    }
+  }
+};
+
+// apply packed hermitian 2x(6x6) clover to a spinor held as [4][3]
+// chirality block b covers spins {2b, 2b+1}; within-block index = s*3+c (s=0,1)
+template <typename R>
+__device__ __forceinline__ void clover_mul(cplx<R> out[4][3], const R diag[2][6],
+                                           const cplx<R> tri[2][15],
+                                           const cplx<R> in[4][3]) {
+  // lower-tri packing: k runs over (i,j), i>j, row-major: (1,0),(2,0),(2,1),...
+#pragma unroll
+  for (int b = 0; b < 2; ++b) {
+    cplx<R> v[6];
+#pragma unroll
+    for (int k = 0; k < 6; ++k) v[k] = in[2 * b + k / 3][k % 3];
+    cplx<R> r[6];
+#pragma unroll
+    for (int k = 0; k < 6; ++k) r[k] = diag[b][k] * v[k];
+    int k = 0;
+#pragma unroll
+    for (int i = 1; i < 6; ++i)
+#pragma unroll
+      for (int j = 0; j < i; ++j, ++k) {
+        r[i] = cfma(tri[b][k], v[j], r[i]);          // A[i][j] v[j]
+        r[j] = cfma_conj(tri[b][k], v[i], r[j]);     // conj(A[i][j]) v[i]
+      }
+#pragma unroll
+    for (int kk = 0; kk < 6; ++kk) out[2 * b + kk / 3][kk % 3] = r[kk];
+  }
+}
+
+template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
+__global__ __launch_bounds__(256) void k_dslash_wilson(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
+    CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
+    SpinorAcc<Prec> x) {
+  using R = typename Prec::Real;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= d.Vcb) return;
+  int xc[4];
+  coords_from_cb(xc, i, d, parity);
+
+  cplx<R> acc[4][3];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+#pragma unroll
+    for (int c = 0; c < 3; ++c) acc[s][c] = {(R)0, (R)0};
+
+  cplx<R> p[4][3], h[2][3], uh[2][3], U[3][3];
+  const R one = (R)1;
+
+#define QA_DIR(MU)                                                        \
+  {                                                                       \
+    long j = neighbor_cb(xc, MU, +1, d);                                  \
+    in.load(p, j);                                                        \
+    if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);    \
+    g.load(U, MU, parity, i);                                             \
+    su3_mul_half(uh, U, h);                                               \
+    if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
+    else recon_##MU##_1(acc, uh, one);                                    \
+    j = neighbor_cb(xc, MU, -1, d);                                       \
+    in.load(p, j);                                                        \
+    if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);    \
+    g.load(U, MU, parity ^ 1, j);                                         \
+    su3_dagmul_half(uh, U, h);                                            \
+    if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \
+    else recon_##MU##_0(acc, uh, one);                                    \
+  }
+
+  QA_DIR(0)
+  QA_DIR(1)
+  QA_DIR(2)
+  QA_DIR(3)
+#undef QA_DIR
+
+  if constexpr (MODE == CLOV_POST) {
+    R diag[2][6];
+    cplx<R> tri[2][15];
+    clov.load(diag, tri, parity, i);
+    cplx<R> tmp[4][3];
+    clover_mul(tmp, diag, tri, acc);
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) acc[s][c] = tmp[s][c];
+  }
+
+  if constexpr (XPAY || MODE == CLOV_X) {
+    cplx<R> xv[4][3];
+    x.load(xv, i);
+    if constexpr (MODE == CLOV_X) {
+      R diag[2][6];
+      cplx<R> tri[2][15];
+      clov.load(diag, tri, parity, i);
+      cplx<R> Ax[4][3];
+      clover_mul(Ax, diag, tri, xv);
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = Ax[s][c] + a * acc[s][c];
+    } else {
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = xv[s][c] + a * acc[s][c];
+    }
+  } else {
+    if (a != (R)1) {
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = a * acc[s][c];
+    }
+  }
+
+  out.store(acc, i);
+}
+
+// standalone clover apply: out = A(parity) in
+template <typename Prec>
+__global__ __launch_bounds__(256) void k_clover_apply(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, CloverAcc<Prec> clov,
+    int parity_slot, long Vcb) {
+  using R = typename Prec::Real;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= Vcb) return;
+  cplx<R> v[4][3], r[4][3];
+  R diag[2][6];
+  cplx<R> tri[2][15];
+  in.load(v, i);
+  clov.load(diag, tri, parity_slot, i);
+  clover_mul(r, diag, tri, v);
+  out.store(r, i);
+}

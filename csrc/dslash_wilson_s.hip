@@ -1,0 +1,7 @@
+// single-precision Wilson dslash TU (recon 18 + 12)
+#include "dslash_wilson_impl.h"
+
+void launch_dslash_wilson_single(const DslashCall &c, hipStream_t st) {
+  if (c.recon == 12) dslash_launch_all<PrecSingle, 12>(c, st);
+  else dslash_launch_all<PrecSingle, 18>(c, st);
+}

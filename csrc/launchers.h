@@ -1,0 +1,66 @@
+// C++ launcher API between the HIP kernel TUs and bindings.cpp.
+#pragma once
+#include <hip/hip_runtime.h>
+
+struct BlasField {
+  void *data;
+  void *norm;  // nullptr unless half
+  long Vcb;    // chunk stride (cb volume)
+};
+
+enum BlasOp {
+  BLAS_AXPY = 0,
+  BLAS_AXPY_NORM2,
+  BLAS_XPAY,
+  BLAS_AXPBY,
+  BLAS_CAXPY,
+  BLAS_XMY_NORM2,
+  BLAS_SCAL,
+  BLAS_NORM2,
+  BLAS_REDOT,
+  BLAS_CDOT,
+};
+
+struct BlasCall {
+  int op;        // BlasOp
+  int prec;      // 0 double, 1 single, 2 half
+  double a, b;   // scalars (caxpy: a=re, b=im)
+  BlasField x, y;
+  long sites;    // npar * Vcb
+  double *result;  // device ptr for reductions (>=2 doubles for cdot)
+};
+
+void launch_blas(const BlasCall &c, hipStream_t st);
+void launch_convert(const BlasField &dst, int pdst, const BlasField &src,
+                    int psrc, long sites, hipStream_t st);
+
+// ---------------------------------------------------------------------------
+struct DslashCall {
+  // single-parity field views (data points at the parity slice)
+  BlasField out, in, x;    // x used when xpay / CLOV_X
+  const void *gauge;       // [mu][par][nch][Vcb][2]
+  const void *clover;      // packed [par][nch][Vcb][W] (A or A^-1 slice)
+  int Xdim[4];
+  int parity_offset;
+  long Vcb;
+  int parity;
+  bool dagger;
+  int mode;   // 0 PLAIN, 1 CLOV_POST, 2 CLOV_X
+  bool xpay;
+  double a;
+  int recon;  // 18 or 12
+};
+
+void launch_dslash_wilson_double(const DslashCall &c, hipStream_t st);
+void launch_dslash_wilson_single(const DslashCall &c, hipStream_t st);
+void launch_dslash_wilson_half(const DslashCall &c, hipStream_t st);
+
+struct CloverApplyCall {
+  BlasField out, in;
+  const void *clover;  // full [2][nch][Vcb][W]
+  int parity;          // which parity slot of the clover field
+  long Vcb;
+  long sites;          // = Vcb (single parity)
+  int prec;
+};
+void launch_clover_apply(const CloverApplyCall &c, hipStream_t st);

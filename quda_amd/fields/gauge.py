@@ -43,6 +43,10 @@ class GaugeField:
     def device(self):
         return self.data.device
 
+    def to(self, device) -> "GaugeField":
+        return GaugeField(self.geo, self.precision, device, self.reconstruct,
+                          data=self.data.to(device))
+
     # ------------------------------------------------------------------
     def to_complex(self, dtype=torch.complex128) -> torch.Tensor:
         """-> [4, 2, V_cb, 3, 3] complex with row2 reconstructed if needed."""

@@ -112,6 +112,11 @@ class SpinorField:
         return self
 
     # -- views -----------------------------------------------------------
+    def to(self, device) -> "SpinorField":
+        return SpinorField(self.geo, self.precision, device, self.n_parity,
+                           data=self.data.to(device),
+                           norm=None if self.norm is None else self.norm.to(device))
+
     def parity_view(self, parity: int) -> "SpinorField":
         """Zero-copy single-parity view of a full field (ref Even()/Odd())."""
         assert self.n_parity == FULL

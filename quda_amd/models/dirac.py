@@ -1,50 +1,55 @@
 """Dirac operator hierarchy (ref: include/dirac_quda.h:156 + lib/dirac*.cpp
-— same class roles, MI355X-first implementation over ops.dispatch).
+— same class roles, MI355X-first implementation over the fused kernels in
+ops.dispatch: every operator application is the minimum number of launches).
 
 Conventions (kappa normalization, matching the reference semantics):
   full:        M psi = psi - kappa * D psi                  (Wilson)
                M psi = A psi - kappa * D psi                (clover)
   even-odd PC (symmetric, QUDA_MATPC_EVEN_EVEN):
-               M_pc chi_e = chi_e - kappa^2 A_ee^-1 D_eo A_oo^-1 D_oe chi_e
-  (A == 1 for plain Wilson).
-
-prepare()/reconstruct() map a full-lattice source/solution to/from the
-preconditioned system (ref: dirac_quda.h:358).
+               M_pc = 1 - kappa^2 A_ee^-1 D_eo A_oo^-1 D_oe
+  source prep: b' = A_ee^-1 (b_e + kappa D_eo A_oo^-1 b_o)
+  reconstruct: psi_o = A_oo^-1 (b_o + kappa D_oe psi_e)     (A == 1: Wilson)
 """
 
 from __future__ import annotations
 
 from typing import Optional
 
-import torch
-
 from ..fields.clover import CloverField
 from ..fields.gauge import GaugeField
 from ..fields.spinor import SpinorField
-from ..ops import blas, dispatch
+from ..ops import blas
+from ..ops.dispatch import CLOV_POST, CLOV_X, PLAIN, apply_clover, dslash_wilson
 
 
 class Dirac:
-    """Base: owns gauge (+clover) and scalar params; applies to parity or
-    full SpinorFields. Subclasses define M / MdagM on their solve space."""
+    """Base: owns gauge (+clover) and scalar params."""
 
     def __init__(self, gauge: GaugeField, kappa: float):
         self.gauge = gauge
         self.kappa = float(kappa)
         self.geo = gauge.geo
+        self._tmps = {}
 
-    # single-parity hop: out(par) = D in(1-par)
-    def dslash(self, out: SpinorField, inp: SpinorField, parity: int,
-               dagger: bool = False, xpay: Optional[tuple] = None):
-        return dispatch.dslash_wilson(out, inp, self.gauge, parity, dagger, xpay)
+    def dslash(self, out, inp, parity, dagger=False, mode=PLAIN, a=1.0,
+               x=None, clover=None, clover_inverse=False):
+        return dslash_wilson(out, inp, self.gauge, parity, dagger, mode, a, x,
+                             clover, clover_inverse)
+
+    def tmp(self, name: str, like: SpinorField, n_parity=1) -> SpinorField:
+        key = (name, like.precision, str(like.device), n_parity)
+        t = self._tmps.get(key)
+        if t is None:
+            t = SpinorField(self.geo, like.precision, like.device, n_parity)
+            self._tmps[key] = t
+        return t
 
     def new_spinor(self, precision=None, n_parity=1) -> SpinorField:
         return SpinorField(self.geo, precision or self.gauge.precision,
                            self.gauge.device, n_parity)
 
     def flops_per_site(self) -> int:
-        """Wilson dslash flop count (ref model: include/dslash.h:467 —
-        1320 flops/site for Nc=3, Ns=4)."""
+        """Wilson dslash flop model (ref: include/dslash.h:467): 1320/site."""
         return 1320
 
 
@@ -52,62 +57,45 @@ class DiracWilson(Dirac):
     """Full-lattice Wilson M = 1 - kappa D (ref: lib/dirac_wilson.cpp)."""
 
     def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
-        # out_e = in_e - kappa D_eo in_o ; out_o = in_o - kappa D_oe in_e
         for p in (0, 1):
             self.dslash(out.parity_view(p), inp.parity_view(1 - p), p, dagger,
-                        xpay=(-1.0 / self.kappa, inp.parity_view(p)))
-        blas.scal(-self.kappa, out)
+                        a=-self.kappa, x=inp.parity_view(p))
         return out
 
-    def MdagM(self, out: SpinorField, inp: SpinorField, tmp: SpinorField):
+    def MdagM(self, out, inp, tmp):
         self.M(tmp, inp, dagger=False)
         self.M(out, tmp, dagger=True)
         return out
 
 
 class DiracWilsonPC(Dirac):
-    """Even-odd preconditioned Wilson: M_pc = 1 - kappa^2 D_eo D_oe acting on
-    the even checkerboard (ref: lib/dirac_wilson.cpp DiracWilsonPC)."""
-
-    def __init__(self, gauge: GaugeField, kappa: float):
-        super().__init__(gauge, kappa)
-        self._tmp_o: Optional[SpinorField] = None
-
-    def _tmp(self, like: SpinorField) -> SpinorField:
-        if (self._tmp_o is None or self._tmp_o.precision != like.precision
-                or self._tmp_o.device != like.device):
-            self._tmp_o = SpinorField(self.geo, like.precision, like.device, 1)
-        return self._tmp_o
+    """Even-odd preconditioned Wilson: M_pc = 1 - kappa^2 D_eo D_oe on the
+    even checkerboard (ref: lib/dirac_wilson.cpp DiracWilsonPC)."""
 
     def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
-        t = self._tmp(inp)
-        # t_o = D_oe in_e ; out_e = in_e - kappa^2 D_eo t_o
+        t = self.tmp("pc_odd", inp)
         self.dslash(t, inp, 1, dagger)
-        self.dslash(out, t, 0, dagger, xpay=(-1.0 / self.kappa ** 2, inp))
-        blas.scal(-self.kappa ** 2, out)
+        self.dslash(out, t, 0, dagger, a=-self.kappa ** 2, x=inp)
         return out
 
-    def MdagM(self, out: SpinorField, inp: SpinorField, tmp: SpinorField):
+    def MdagM(self, out, inp, tmp):
         self.M(tmp, inp, dagger=False)
         self.M(out, tmp, dagger=True)
         return out
 
-    # -- source prep / solution reconstruction (ref dirac_quda.h:358) ------
     def prepare(self, b_full: SpinorField) -> SpinorField:
-        """b_e' = b_e + kappa D_eo b_o (kappa-normalized even source)."""
-        be = SpinorField(self.geo, b_full.precision, b_full.device, 1)
-        self.dslash(be, b_full.parity_view(1), 0,
-                    xpay=(1.0 / self.kappa, b_full.parity_view(0)))
-        blas.scal(self.kappa, be)
+        """b' = b_e + kappa D_eo b_o."""
+        be = self.new_spinor(b_full.precision)
+        self.dslash(be, b_full.parity_view(1), 0, a=self.kappa,
+                    x=b_full.parity_view(0))
         return be
 
     def reconstruct(self, x_full: SpinorField, x_e: SpinorField,
                     b_full: SpinorField):
-        """x_o = kappa (b_o_ + D_oe x_e) -> writes both parities of x_full."""
-        x_full.parity_view(0).copy_(x_e)
-        xo = x_full.parity_view(1)
-        self.dslash(xo, x_e, 1, xpay=(1.0 / self.kappa, b_full.parity_view(1)))
-        blas.scal(self.kappa, xo)
+        """x_o = b_o + kappa D_oe x_e."""
+        blas.copy(x_full.parity_view(0), x_e)
+        self.dslash(x_full.parity_view(1), x_e, 1, a=self.kappa,
+                    x=b_full.parity_view(1))
         return x_full
 
 
@@ -115,24 +103,22 @@ class _CloverMixin:
     clover: CloverField
 
     def apply_A(self, out, inp, parity, inverse=False):
-        return dispatch.apply_clover(out, inp, self.clover, parity, inverse)
+        return apply_clover(out, inp, self.clover, parity, inverse)
 
 
 class DiracClover(Dirac, _CloverMixin):
     """Full-lattice Wilson-clover M = A - kappa D
-    (ref: lib/dirac_clover.cpp)."""
+    (ref: lib/dirac_clover.cpp): one fused CLOV_X launch per parity."""
 
     def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float):
         super().__init__(gauge, kappa)
         self.clover = clover
 
     def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
-        t = SpinorField(self.geo, inp.precision, inp.device, 1)
         for p in (0, 1):
-            self.apply_A(t, inp.parity_view(p), p)
             self.dslash(out.parity_view(p), inp.parity_view(1 - p), p, dagger,
-                        xpay=(-1.0 / self.kappa, t))
-        blas.scal(-self.kappa, out)
+                        mode=CLOV_X, a=-self.kappa, x=inp.parity_view(p),
+                        clover=self.clover)
         return out
 
     def MdagM(self, out, inp, tmp):
@@ -141,44 +127,34 @@ class DiracClover(Dirac, _CloverMixin):
         return out
 
     def flops_per_site(self) -> int:
-        return 1320 + 504  # dslash + clover (ref dslash.h flop model)
+        return 1320 + 504
 
 
 class DiracCloverPC(Dirac, _CloverMixin):
-    """Symmetric even-odd preconditioned clover:
-    M_pc = 1 - kappa^2 A_ee^-1 D_eo A_oo^-1 D_oe
-    (ref: lib/dirac_clover.cpp DiracCloverPC, QUDA_MATPC_EVEN_EVEN)."""
+    """Symmetric even-odd preconditioned clover
+    (ref: lib/dirac_clover.cpp DiracCloverPC, QUDA_MATPC_EVEN_EVEN):
+    M_pc = 1 - kappa^2 A_ee^-1 D_eo A_oo^-1 D_oe  — two fused launches."""
 
     def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float):
         super().__init__(gauge, kappa)
         self.clover = clover
-        self._t1: Optional[SpinorField] = None
-        self._t2: Optional[SpinorField] = None
-
-    def _tmps(self, like: SpinorField):
-        if (self._t1 is None or self._t1.precision != like.precision
-                or self._t1.device != like.device):
-            self._t1 = SpinorField(self.geo, like.precision, like.device, 1)
-            self._t2 = SpinorField(self.geo, like.precision, like.device, 1)
-        return self._t1, self._t2
 
     def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
-        t1, t2 = self._tmps(inp)
+        k2 = -self.kappa ** 2
         if not dagger:
-            self.dslash(t1, inp, 1)            # t1_o = D_oe in_e
-            self.apply_A(t2, t1, 1, inverse=True)   # t2 = A_oo^-1 t1
-            self.dslash(t1, t2, 0)             # t1_e = D_eo t2
-            self.apply_A(t2, t1, 0, inverse=True)   # t2 = A_ee^-1 t1
+            t = self.tmp("pc_odd", inp)
+            self.dslash(t, inp, 1, mode=CLOV_POST, clover=self.clover,
+                        clover_inverse=True)
+            self.dslash(out, t, 0, mode=CLOV_POST, a=k2, x=inp,
+                        clover=self.clover, clover_inverse=True)
         else:
-            # (M_pc)^dag = 1 - kappa^2 D_oe^dag A_oo^-1 D_eo^dag A_ee^-1
-            self.apply_A(t2, inp, 0, inverse=True)
-            self.dslash(t1, t2, 1, dagger=True)
-            self.apply_A(t2, t1, 1, inverse=True)
-            self.dslash(t1, t2, 0, dagger=True)
-            t2, t1 = t1, t2
-        # out = in - kappa^2 t2
-        blas.copy(out, inp)
-        blas.axpy(-self.kappa ** 2, t2, out)
+            # M^dag = 1 - k^2 D_oe^dag A_oo^-1 D_eo^dag A_ee^-1
+            t0 = self.tmp("pc_even", inp)
+            t1 = self.tmp("pc_odd", inp)
+            self.apply_A(t0, inp, 0, inverse=True)
+            self.dslash(t1, t0, 1, dagger=True, mode=CLOV_POST,
+                        clover=self.clover, clover_inverse=True)
+            self.dslash(out, t1, 0, dagger=True, a=k2, x=inp)
         return out
 
     def MdagM(self, out, inp, tmp):
@@ -187,25 +163,22 @@ class DiracCloverPC(Dirac, _CloverMixin):
         return out
 
     def prepare(self, b_full: SpinorField) -> SpinorField:
-        """b_e' = kappa A_ee^-1 (b_e + kappa D_eo A_oo^-1 b_o)."""
-        t = SpinorField(self.geo, b_full.precision, b_full.device, 1)
-        be = SpinorField(self.geo, b_full.precision, b_full.device, 1)
+        """b' = A_ee^-1 (b_e + kappa D_eo A_oo^-1 b_o)."""
+        t = self.new_spinor(b_full.precision)
+        be = self.new_spinor(b_full.precision)
         self.apply_A(t, b_full.parity_view(1), 1, inverse=True)
-        self.dslash(be, t, 0, xpay=(1.0 / self.kappa, b_full.parity_view(0)))
+        self.dslash(be, t, 0, a=self.kappa, x=b_full.parity_view(0))
         self.apply_A(t, be, 0, inverse=True)
         blas.copy(be, t)
-        blas.scal(self.kappa, be)
         return be
 
     def reconstruct(self, x_full: SpinorField, x_e: SpinorField,
                     b_full: SpinorField):
-        """x_o = kappa A_oo^-1 (b_o + D_oe x_e)."""
-        x_full.parity_view(0).copy_(x_e)
-        t = SpinorField(self.geo, x_full.precision, x_full.device, 1)
-        self.dslash(t, x_e, 1, xpay=(1.0 / self.kappa, b_full.parity_view(1)))
-        xo = x_full.parity_view(1)
-        self.apply_A(xo, t, 1, inverse=True)
-        blas.scal(self.kappa, xo)
+        """x_o = A_oo^-1 (b_o + kappa D_oe x_e)."""
+        blas.copy(x_full.parity_view(0), x_e)
+        t = self.tmp("pc_odd", x_full)
+        self.dslash(t, x_e, 1, a=self.kappa, x=b_full.parity_view(1))
+        self.apply_A(x_full.parity_view(1), t, 1, inverse=True)
         return x_full
 
     def flops_per_site(self) -> int:

@@ -1,15 +1,10 @@
 """BLAS / reduction layer on SpinorFields (ref: lib/blas_quda.cu,
-lib/reduce_quda.cu — ~60 fused vector ops; here the core set the solvers
-need, with the same fusion boundaries so the HIP kernels drop in 1:1).
+lib/reduce_quda.cu — the fused op set the solvers need; HIP kernels in
+csrc/blas.hip are the GPU path for every precision, plain torch the CPU
+path; half on CPU goes through dequantized complex).
 
-Dispatch rule: double/single fields -> direct torch ops on the chunked
-storage (torch on ROCm is already a bandwidth-bound eager path; the fused
-HIP kernels in csrc/blas.hip replace the multi-kernel sequences on GPU).
-half fields -> dequantize/requantize via complex (CPU oracle only; on GPU
-the HIP kernels consume (data, norm) natively).
-
-All reductions return python floats/complex and, in multi-rank runs, are
-globally summed over the process grid (ref: reducer.h allreduce sites).
+All reductions return python floats/complex, globally summed over the
+process grid (ref: include/reducer.h allreduce call sites).
 """
 
 from __future__ import annotations
@@ -18,6 +13,7 @@ import torch
 
 from ..fields.spinor import SpinorField
 from ..parallel.comms import allreduce_sum
+from .dispatch import hip_ext, norm_or_empty, on_gpu
 
 
 def _is_half(x: SpinorField) -> bool:
@@ -25,12 +21,10 @@ def _is_half(x: SpinorField) -> bool:
 
 
 def _pairs(t: torch.Tensor) -> torch.Tensor:
-    """View chunked real storage as [..., n, 2] (re, im) pairs."""
     return t.reshape(*t.shape[:-1], t.shape[-1] // 2, 2)
 
 
 def _requant(x: SpinorField, vals: torch.Tensor):
-    """Write complex values [P,V,12] back into a (possibly half) field."""
     x.from_complex(vals.reshape(x.n_parity, x.volume_cb, 4, 3))
 
 
@@ -38,33 +32,47 @@ def _cvals(x: SpinorField) -> torch.Tensor:
     return x.to_complex(torch.complex128).reshape(x.n_parity, x.volume_cb, 12)
 
 
+def _gpu_blas(op: int, a, b, x: SpinorField, y: SpinorField):
+    ext = hip_ext()
+    return ext.blas_op(op, float(a), float(b), x.data, norm_or_empty(x),
+                       y.data, norm_or_empty(y), x.volume_cb,
+                       x.n_parity * x.volume_cb)
+
+
 # -- reductions -------------------------------------------------------------
 
 def norm2(x: SpinorField) -> float:
-    if _is_half(x):
+    if on_gpu(x):
+        ext = hip_ext()
+        r = _gpu_blas(ext.BLAS_NORM2, 0, 0, x, x)[0].item()
+    elif _is_half(x):
         v = _cvals(x)
         r = (v.real ** 2 + v.imag ** 2).sum().item()
     else:
-        d = x.data.to(torch.float64) if x.data.dtype != torch.float64 else x.data
+        d = x.data.to(torch.float64)
         r = (d * d).sum().item()
     return allreduce_sum(r)
 
 
 def re_dot(x: SpinorField, y: SpinorField) -> float:
-    """Re <x, y>."""
-    if _is_half(x) or _is_half(y):
-        vx, vy = _cvals(x), _cvals(y)
-        r = (vx.conj() * vy).real.sum().item()
+    if on_gpu(x, y):
+        ext = hip_ext()
+        r = _gpu_blas(ext.BLAS_REDOT, 0, 0, x, y)[0].item()
+    elif _is_half(x) or _is_half(y):
+        r = (_cvals(x).conj() * _cvals(y)).real.sum().item()
     else:
         r = (x.data.to(torch.float64) * y.data.to(torch.float64)).sum().item()
     return allreduce_sum(r)
 
 
 def c_dot(x: SpinorField, y: SpinorField) -> complex:
-    """<x, y> = sum conj(x) y."""
+    if on_gpu(x, y):
+        ext = hip_ext()
+        t = _gpu_blas(ext.BLAS_CDOT, 0, 0, x, y)
+        re, im = t[0].item(), t[1].item()
+        return complex(allreduce_sum(re), allreduce_sum(im))
     if _is_half(x) or _is_half(y):
-        vx, vy = _cvals(x), _cvals(y)
-        s = (vx.conj() * vy).sum().item()
+        s = (_cvals(x).conj() * _cvals(y)).sum().item()
         return complex(allreduce_sum(s.real), allreduce_sum(s.imag))
     px, py = _pairs(x.data.to(torch.float64)), _pairs(y.data.to(torch.float64))
     re = (px * py).sum().item()
@@ -73,13 +81,21 @@ def c_dot(x: SpinorField, y: SpinorField) -> complex:
 
 
 def axpy_norm2(a: float, x: SpinorField, y: SpinorField) -> float:
-    """y += a*x; returns ||y||^2 (fused in csrc/blas.hip; ref axpyNorm2)."""
+    """y += a*x; returns ||y||^2 (fused; ref axpyNorm2)."""
+    if on_gpu(x, y):
+        ext = hip_ext()
+        r = _gpu_blas(ext.BLAS_AXPY_NORM2, a, 0, x, y)[0].item()
+        return allreduce_sum(r)
     axpy(a, x, y)
     return norm2(y)
 
 
 def xmy_norm2(x: SpinorField, y: SpinorField) -> float:
     """y = x - y; returns ||y||^2."""
+    if on_gpu(x, y):
+        ext = hip_ext()
+        r = _gpu_blas(ext.BLAS_XMY_NORM2, 0, 0, x, y)[0].item()
+        return allreduce_sum(r)
     if _is_half(x) or _is_half(y):
         v = _cvals(x) - _cvals(y)
         _requant(y, v)
@@ -91,6 +107,13 @@ def xmy_norm2(x: SpinorField, y: SpinorField) -> float:
 # -- elementwise ------------------------------------------------------------
 
 def copy(dst: SpinorField, src: SpinorField) -> SpinorField:
+    if dst is src:
+        return dst
+    if on_gpu(dst, src) and dst.precision != src.precision:
+        ext = hip_ext()
+        ext.convert(dst.data, norm_or_empty(dst), src.data, norm_or_empty(src),
+                    dst.volume_cb, dst.n_parity * dst.volume_cb)
+        return dst
     dst.copy_(src)
     return dst
 
@@ -101,7 +124,10 @@ def zero(x: SpinorField) -> SpinorField:
 
 def axpy(a: float, x: SpinorField, y: SpinorField) -> SpinorField:
     """y = a*x + y."""
-    if _is_half(x) or _is_half(y):
+    if on_gpu(x, y):
+        ext = hip_ext()
+        _gpu_blas(ext.BLAS_AXPY, a, 0, x, y)
+    elif _is_half(x) or _is_half(y):
         _requant(y, _cvals(y) + a * _cvals(x))
     else:
         y.data.add_(x.data.to(y.data.dtype), alpha=float(a))
@@ -110,7 +136,10 @@ def axpy(a: float, x: SpinorField, y: SpinorField) -> SpinorField:
 
 def xpay(x: SpinorField, a: float, y: SpinorField) -> SpinorField:
     """y = x + a*y."""
-    if _is_half(x) or _is_half(y):
+    if on_gpu(x, y):
+        ext = hip_ext()
+        _gpu_blas(ext.BLAS_XPAY, a, 0, x, y)
+    elif _is_half(x) or _is_half(y):
         _requant(y, _cvals(x) + a * _cvals(y))
     else:
         y.data.mul_(float(a)).add_(x.data.to(y.data.dtype))
@@ -119,7 +148,10 @@ def xpay(x: SpinorField, a: float, y: SpinorField) -> SpinorField:
 
 def axpby(a: float, x: SpinorField, b: float, y: SpinorField) -> SpinorField:
     """y = a*x + b*y."""
-    if _is_half(x) or _is_half(y):
+    if on_gpu(x, y):
+        ext = hip_ext()
+        _gpu_blas(ext.BLAS_AXPBY, a, b, x, y)
+    elif _is_half(x) or _is_half(y):
         _requant(y, a * _cvals(x) + b * _cvals(y))
     else:
         y.data.mul_(float(b)).add_(x.data.to(y.data.dtype), alpha=float(a))
@@ -128,18 +160,27 @@ def axpby(a: float, x: SpinorField, b: float, y: SpinorField) -> SpinorField:
 
 def caxpy(a: complex, x: SpinorField, y: SpinorField) -> SpinorField:
     """y += a*x (complex a)."""
+    if on_gpu(x, y):
+        ext = hip_ext()
+        _gpu_blas(ext.BLAS_CAXPY, a.real, a.imag, x, y)
+        return y
     if _is_half(x) or _is_half(y):
         _requant(y, _cvals(y) + a * _cvals(x))
         return y
     px = _pairs(x.data.to(y.data.dtype))
     py = _pairs(y.data)
-    py[..., 0] += a.real * px[..., 0] - a.imag * px[..., 1]
-    py[..., 1] += a.real * px[..., 1] + a.imag * px[..., 0]
+    re = a.real * px[..., 0] - a.imag * px[..., 1]
+    im = a.real * px[..., 1] + a.imag * px[..., 0]
+    py[..., 0] += re
+    py[..., 1] += im
     return y
 
 
 def scal(a: float, x: SpinorField) -> SpinorField:
-    if _is_half(x):
+    if on_gpu(x):
+        ext = hip_ext()
+        _gpu_blas(ext.BLAS_SCAL, a, 0, x, x)
+    elif _is_half(x):
         x.norm.mul_(abs(float(a)))
         if a < 0:
             x.data.neg_()

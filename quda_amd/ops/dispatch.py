@@ -1,22 +1,27 @@
 """Op dispatch: HIP extension on GPU tensors, torch oracle on CPU.
 
-The HIP extension (quda_amd_hip, built from csrc/ by setup.py) is the ONLY
-compute path on a GPU box: if a field lives on a CUDA/HIP device and the
-extension is missing we raise — no silent eager fallback (per-project rule:
-GPU tests must exercise the native kernels).
+The HIP extension (quda_amd_hip.so, built in-tree from csrc/ by
+build_hip.py) is the ONLY compute path on a GPU box: if a field lives on a
+CUDA/HIP device and the extension is missing we raise — no silent eager
+fallback (GPU tests must exercise the native kernels).
+
+Fused dslash modes (csrc/dslash_wilson.h):
+  PLAIN      : out = [x +] a * (D in)
+  CLOV_POST  : out = [x +] a * (A (D in))     A = clover or its inverse
+  CLOV_X     : out = A x + a * (D in)
 """
 
 from __future__ import annotations
 
-import os
 from typing import Optional
 
 import torch
 
 from ..fields.gauge import GaugeField
-from ..fields.geometry import LatticeGeometry
 from ..fields.spinor import SpinorField
 from . import reference as ref
+
+PLAIN, CLOV_POST, CLOV_X = 0, 1, 2
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
@@ -27,87 +32,87 @@ def _load_ext():
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
     try:
-        import quda_amd_hip  # built in-tree: csrc/ -> quda_amd_hip.so
+        import quda_amd_hip
         _EXT = quda_amd_hip
     except ImportError as e:
         _EXT_ERR = str(e)
     return _EXT
 
 
-def hip_ext(required: bool):
+def hip_ext(required: bool = True):
     ext = _load_ext()
     if ext is None and required:
         raise RuntimeError(
             f"quda_amd_hip extension not available ({_EXT_ERR}); "
-            "build it with `python setup.py build_ext --inplace` — the GPU "
-            "path never falls back to eager torch.")
+            "build it with `python build_hip.py` — the GPU path never "
+            "falls back to eager torch.")
     return ext
 
 
-def _on_gpu(*fields) -> bool:
-    return any(f.device.type == "cuda" for f in fields if f is not None)
+def on_gpu(*fields) -> bool:
+    return any(f is not None and f.device.type == "cuda" for f in fields)
 
 
-# ---------------------------------------------------------------------------
-# Wilson dslash
-# ---------------------------------------------------------------------------
+def _empty(dev):
+    return torch.empty(0, dtype=torch.float32, device=dev)
+
+
+def norm_or_empty(f: Optional[SpinorField]):
+    if f is None or f.norm is None:
+        return _empty("cpu" if f is None else f.device)
+    return f.norm
+
+
+RECON_OF = {9: 18, 6: 12}  # gauge chunk count -> recon
+
 
 def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
-                  parity: int, dagger: bool = False,
-                  xpay: Optional[tuple] = None):
-    """out(parity) = D in(1-parity)  [+ a * x(parity) if xpay=(a, x)].
-
-    out/inp are single-parity fields (n_parity==1). The xpay fusion is the
-    reference's DslashXpay (dirac_quda.h:268).
-    """
+                  parity: int, dagger: bool = False, mode: int = PLAIN,
+                  a: float = 1.0, x: Optional[SpinorField] = None,
+                  clover=None, clover_inverse: bool = False):
+    """Apply the fused Wilson(-clover) stencil; out at `parity`, in at the
+    opposite parity; x (same parity as out) enables the xpay term."""
     geo = out.geo
-    if _on_gpu(out.data, inp.data):
-        ext = hip_ext(required=True)
-        _gpu_dslash_wilson(ext, out, inp, gauge, parity, dagger, xpay)
+    xpay = x is not None
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)
+        if mode != PLAIN:
+            cl_t = clover.inv_data if clover_inverse else clover.data
+        xf = x if x is not None else out
+        ext.dslash_wilson(
+            out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+            gauge.data, cl_t, xf.data, norm_or_empty(xf),
+            list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
+            bool(dagger), mode, xpay, float(a), RECON_OF[gauge.data.shape[2]])
         return out
-    # oracle path
+    # ---- oracle path ----
     u = gauge.to_complex()
     psi = inp.to_complex()[0]
     res = ref.dslash_wilson_parity(u, psi, geo, parity, dagger)
-    if xpay is not None:
-        a, x = xpay
-        res = a * x.to_complex()[0] + res
+    if mode == CLOV_POST:
+        A = clover.to_complex(inverse=clover_inverse)[parity]
+        res = ref.apply_clover(A, res)
+    if mode == CLOV_X:
+        A = clover.to_complex(inverse=clover_inverse)[parity]
+        res = ref.apply_clover(A, x.to_complex()[0]) + a * res
+    elif xpay:
+        res = x.to_complex()[0] + a * res
+    else:
+        res = a * res
     out.from_complex(res.unsqueeze(0))
     return out
 
 
-def _gpu_dslash_wilson(ext, out, inp, gauge, parity, dagger, xpay):
-    a = 0.0
-    xdata = out.data  # unused when a == 0
-    xnorm = _norm_or_empty(out)
-    if xpay is not None:
-        a, x = xpay
-        xdata, xnorm = x.data, _norm_or_empty(x)
-    ext.dslash_wilson(out.data, _norm_or_empty(out), inp.data,
-                      _norm_or_empty(inp), gauge.data,
-                      list(out.geo.dims), parity, bool(dagger),
-                      float(a), xdata, xnorm)
-
-
-def _norm_or_empty(f: SpinorField):
-    if f.norm is not None:
-        return f.norm
-    return torch.empty(0, dtype=torch.float32, device=f.device)
-
-
-# ---------------------------------------------------------------------------
-# Clover apply
-# ---------------------------------------------------------------------------
-
 def apply_clover(out: SpinorField, inp: SpinorField, clover, parity: int,
                  inverse: bool = False):
-    """out = A(parity) in  (site-local 12x12; clover is a CloverField)."""
-    if _on_gpu(out.data, inp.data):
-        ext = hip_ext(required=True)
-        ext.clover_apply(out.data, _norm_or_empty(out), inp.data,
-                         _norm_or_empty(inp),
+    """out = A(parity) in (standalone; used by prepare/reconstruct)."""
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        ext.clover_apply(out.data, norm_or_empty(out), inp.data,
+                         norm_or_empty(inp),
                          clover.inv_data if inverse else clover.data,
-                         parity)
+                         parity, out.geo.volume_cb)
         return out
     A = clover.to_complex(inverse=inverse)[parity]
     psi = inp.to_complex()[0]

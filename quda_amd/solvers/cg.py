@@ -99,7 +99,7 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
         need_reliable = mixed and (rnorm < delta * maxr)
         if need_reliable or r2 <= stop:
             # accumulate + recompute true residual at high precision
-            tmp_hi.copy_(x_s)
+            blas.copy(tmp_hi, x_s)
             blas.axpy(1.0, tmp_hi, x)
             op.MdagM(r, x, tmp_hi)
             r2 = blas.xmy_norm2(b, r)
@@ -119,7 +119,7 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
 
     if not mixed:
         # accumulate the (high-precision) iterate
-        tmp_hi.copy_(x_s)
+        blas.copy(tmp_hi, x_s)
         blas.axpy(1.0, tmp_hi, x)
         op.MdagM(r, x, tmp_hi)
         r2 = blas.xmy_norm2(b, r)

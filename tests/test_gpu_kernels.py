@@ -1,0 +1,210 @@
+"""GPU numerics tests: every HIP kernel vs the plain-PyTorch fp64 oracle.
+(analogue of the reference's dslash_ctest / blas_test GPU-vs-host checks)"""
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.fields.clover import CloverField
+from quda_amd.models import DiracCloverPC, DiracWilsonPC
+from quda_amd.ops import blas
+from quda_amd.ops import reference as ref
+from quda_amd.ops.dispatch import CLOV_POST, CLOV_X, PLAIN, dslash_wilson
+from quda_amd.solvers import cg_solve
+
+pytestmark = pytest.mark.gpu
+
+TOL = {"double": 1e-12, "single": 2e-5, "half": 2e-2}
+RECONS = {"double": [18], "single": [18, 12], "half": [12, 18]}
+KAPPA = 0.13
+CSW = 1.1
+
+
+@pytest.fixture(scope="module")
+def setup():
+    geo = LatticeGeometry((8, 8, 8, 8))
+    g = GaugeField(geo, "double").random_su3_(seed=31)
+    psi = SpinorField(geo, "double").gaussian_(seed=32)
+    chi = SpinorField(geo, "double").gaussian_(seed=33)
+    A = ref.clover_matrix(g.to_complex(), geo, KAPPA, CSW)
+    return geo, g, psi, chi, A
+
+
+def _gpu_fields(geo, g, psi, prec, recon):
+    gd = GaugeField(geo, prec, "cuda", reconstruct="none" if recon == 18 else "twelve")
+    gd.from_complex(g.to_complex().cuda())
+    sd = SpinorField(geo, prec, "cuda", n_parity=2).from_complex(
+        psi.to_complex().cuda())
+    return gd, sd
+
+
+@pytest.mark.parametrize("prec", ["double", "single", "half"])
+@pytest.mark.parametrize("dagger", [False, True])
+def test_dslash_vs_oracle(setup, prec, dagger):
+    geo, g, psi, chi, A = setup
+    u = g.to_complex()
+    for recon in RECONS[prec]:
+        gd, sd = _gpu_fields(geo, g, psi, prec, recon)
+        out = SpinorField(geo, prec, "cuda", n_parity=1)
+        for parity in (0, 1):
+            dslash_wilson(out, sd.parity_view(1 - parity), gd, parity, dagger)
+            got = out.to_complex().cpu()[0]
+            want = ref.dslash_wilson_parity(u, psi.to_complex()[1 - parity],
+                                            geo, parity, dagger)
+            err = (got - want).abs().max().item() / want.abs().max().item()
+            assert err < TOL[prec], (prec, recon, parity, dagger, err)
+
+
+@pytest.mark.parametrize("prec", ["double", "single"])
+def test_dslash_xpay(setup, prec):
+    geo, g, psi, chi, A = setup
+    u = g.to_complex()
+    gd, sd = _gpu_fields(geo, g, psi, prec, 18)
+    xd = SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+        chi.to_complex()[0:1].cuda())
+    out = SpinorField(geo, prec, "cuda", n_parity=1)
+    a = -0.37
+    dslash_wilson(out, sd.parity_view(1), gd, 0, a=a, x=xd)
+    got = out.to_complex().cpu()[0]
+    want = chi.to_complex()[0] + a * ref.dslash_wilson_parity(
+        u, psi.to_complex()[1], geo, 0)
+    err = (got - want).abs().max().item() / want.abs().max().item()
+    assert err < TOL[prec]
+
+
+@pytest.mark.parametrize("prec", ["double", "single", "half"])
+@pytest.mark.parametrize("mode", [CLOV_POST, CLOV_X])
+def test_dslash_clover_modes(setup, prec, mode):
+    geo, g, psi, chi, A = setup
+    u = g.to_complex()
+    recon = RECONS[prec][0]
+    gd, sd = _gpu_fields(geo, g, psi, prec, recon)
+    cl = CloverField(geo, prec, "cuda")
+    cl.data.copy_(cl._to_native(
+        __import__("quda_amd.fields.clover", fromlist=["pack_clover"])
+        .pack_clover(A.cuda())))
+    xd = SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+        chi.to_complex()[0:1].cuda())
+    out = SpinorField(geo, prec, "cuda", n_parity=1)
+    a = -0.29
+    D = ref.dslash_wilson_parity(u, psi.to_complex()[1], geo, 0)
+    if mode == CLOV_POST:
+        dslash_wilson(out, sd.parity_view(1), gd, 0, mode=CLOV_POST, a=a,
+                      x=xd, clover=cl)
+        want = chi.to_complex()[0] + a * ref.apply_clover(A[0], D)
+    else:
+        dslash_wilson(out, sd.parity_view(1), gd, 0, mode=CLOV_X, a=a, x=xd,
+                      clover=cl)
+        want = ref.apply_clover(A[0], chi.to_complex()[0]) + a * D
+    got = out.to_complex().cpu()[0]
+    err = (got - want).abs().max().item() / want.abs().max().item()
+    assert err < TOL[prec]
+
+
+@pytest.mark.parametrize("prec", ["double", "single", "half"])
+def test_blas_gpu(setup, prec):
+    geo, g, psi, chi, A = setup
+    tol = TOL[prec]
+    x = SpinorField(geo, prec, "cuda", n_parity=2).from_complex(psi.to_complex().cuda())
+    y = SpinorField(geo, prec, "cuda", n_parity=2).from_complex(chi.to_complex().cuda())
+    xc, yc = psi.to_complex(), chi.to_complex()
+    # reductions
+    assert abs(blas.norm2(x) - (xc.abs() ** 2).sum().item()) < tol * abs((xc.abs() ** 2).sum().item())
+    rd = blas.re_dot(x, y)
+    want_rd = (xc.conj() * yc).real.sum().item()
+    assert abs(rd - want_rd) < tol * (abs(want_rd) + 1)
+    cd = blas.c_dot(x, y)
+    want_cd = (xc.conj() * yc).sum().item()
+    assert abs(cd - want_cd) < tol * (abs(want_cd) + 1)
+    # axpy_norm2
+    r = blas.axpy_norm2(0.7, x, y)
+    yc = yc + 0.7 * xc
+    assert abs(r - (yc.abs() ** 2).sum().item()) < 10 * tol * r
+    # xpay
+    blas.xpay(x, -1.3, y)
+    yc = xc - 1.3 * yc
+    err = (y.to_complex().cpu() - yc).abs().max().item()
+    assert err < 10 * tol * yc.abs().max().item()
+    # caxpy
+    blas.caxpy(0.3 - 0.4j, x, y)
+    yc = yc + (0.3 - 0.4j) * xc
+    err = (y.to_complex().cpu() - yc).abs().max().item()
+    assert err < 20 * tol * yc.abs().max().item()
+    # xmy_norm2
+    r = blas.xmy_norm2(x, y)
+    yc = xc - yc
+    assert abs(r - (yc.abs() ** 2).sum().item()) < 20 * tol * (r + 1)
+    # scal
+    blas.scal(0.5, y)
+    yc = 0.5 * yc
+    err = (y.to_complex().cpu() - yc).abs().max().item()
+    assert err < 20 * tol * (yc.abs().max().item() + 1)
+
+
+def test_convert_gpu(setup):
+    geo, g, psi, chi, A = setup
+    d = SpinorField(geo, "double", "cuda", n_parity=2).from_complex(psi.to_complex().cuda())
+    s = SpinorField(geo, "single", "cuda", n_parity=2)
+    h = SpinorField(geo, "half", "cuda", n_parity=2)
+    blas.copy(s, d)
+    blas.copy(h, s)
+    d2 = SpinorField(geo, "double", "cuda", n_parity=2)
+    blas.copy(d2, h)
+    ref_c = psi.to_complex()
+    assert (s.to_complex().cpu() - ref_c).abs().max().item() < 1e-6
+    assert (d2.to_complex().cpu() - ref_c).abs().max().item() < 2e-3
+
+
+@pytest.mark.parametrize("sloppy", ["double", "single", "half"])
+def test_cg_wilson_gpu(setup, sloppy):
+    geo, g, psi, chi, A = setup
+    gd = GaugeField(geo, "double", "cuda").from_complex(g.to_complex().cuda())
+    gs_prec = "single" if sloppy == "single" else ("half" if sloppy == "half" else "double")
+    recon_s = "twelve" if sloppy != "double" else "none"
+    gs = GaugeField(geo, gs_prec, "cuda", reconstruct=recon_s).from_complex(
+        g.to_complex().cuda())
+    d = DiracWilsonPC(gd, KAPPA)
+    ds = DiracWilsonPC(gs, KAPPA)
+    b = SpinorField(geo, "double", "cuda", n_parity=1).from_complex(
+        psi.to_complex()[0:1].cuda())
+    x = SpinorField(geo, "double", "cuda", n_parity=1)
+    stats = cg_solve(d, x, b, op_sloppy=ds, sloppy=sloppy, tol=1e-8,
+                     maxiter=500)
+    assert stats.converged, (sloppy, stats)
+    # independent residual check on the oracle
+    u = g.to_complex()
+    xe = x.to_complex().cpu()[0]
+    t = ref.dslash_wilson_parity(u, xe, geo, 1)
+    Mx = xe - KAPPA ** 2 * ref.dslash_wilson_parity(u, t, geo, 0)
+    # Mdag
+    t = ref.dslash_wilson_parity(u, Mx, geo, 1, dagger=True)
+    MdMx = Mx - KAPPA ** 2 * ref.dslash_wilson_parity(u, t, geo, 0, dagger=True)
+    rel = (MdMx - psi.to_complex()[0]).abs().max().item()
+    assert rel < 1e-6
+
+
+def test_cg_clover_gpu(setup):
+    geo, g, psi, chi, A = setup
+    gd = GaugeField(geo, "double", "cuda").from_complex(g.to_complex().cuda())
+    gh = GaugeField(geo, "half", "cuda", reconstruct="twelve").from_complex(
+        g.to_complex().cuda())
+    cld = CloverField(geo, "double", "cuda")
+    clh = CloverField(geo, "half", "cuda")
+    Ac = A.cuda()
+    cld.from_matrices(Ac)
+    from quda_amd.fields.clover import pack_clover
+    clh.data.copy_(clh._to_native(pack_clover(Ac)))
+    Ainv = cld.to_complex(inverse=True)
+    clh.inv_data.copy_(clh._to_native(pack_clover(Ainv.to("cuda"))))
+    d = DiracCloverPC(gd, cld, KAPPA)
+    dh = DiracCloverPC(gh, clh, KAPPA)
+    b = SpinorField(geo, "double", "cuda", n_parity=1).from_complex(
+        psi.to_complex()[0:1].cuda())
+    x = SpinorField(geo, "double", "cuda", n_parity=1)
+    stats = cg_solve(d, x, b, op_sloppy=dh, sloppy="half", tol=1e-8,
+                     maxiter=1000)
+    assert stats.converged, stats
+    out = SpinorField(geo, "double", "cuda", n_parity=1)
+    tmp = SpinorField(geo, "double", "cuda", n_parity=1)
+    d.MdagM(out, x, tmp)
+    r = blas.xmy_norm2(b, out)
+    assert r < 1e-14 * blas.norm2(b) * 1e6  # rel residual < 1e-4 in norm2
