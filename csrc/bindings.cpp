@@ -28,6 +28,11 @@ hipStream_t stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
 
+void check_launch(const char *what) {
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, what, ": ", hipGetErrorString(e));
+}
+
 }  // namespace
 
 static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
@@ -57,6 +62,7 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
     case 1: launch_dslash_wilson_single(c, stream()); break;
     case 2: launch_dslash_wilson_half(c, stream()); break;
   }
+  check_launch("dslash_wilson");
 }
 
 static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
@@ -78,6 +84,7 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
     c.result = result.data_ptr<double>();
   }
   launch_blas(c, stream());
+  check_launch("blas_op");
   return result;
 }
 
@@ -85,6 +92,7 @@ static void convert(at::Tensor dst, at::Tensor dst_n, at::Tensor src,
                     at::Tensor src_n, int64_t Vcb, int64_t sites) {
   launch_convert(field_of(dst, dst_n, Vcb), prec_of(dst),
                  field_of(src, src_n, Vcb), prec_of(src), sites, stream());
+  check_launch("convert");
 }
 
 static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
@@ -99,6 +107,7 @@ static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
   c.sites = Vcb;
   c.prec = prec_of(out);
   launch_clover_apply(c, stream());
+  check_launch("clover_apply");
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
