@@ -92,7 +92,8 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
     for (int c = 0; c < 3; ++c) acc[s][c] = {(R)0, (R)0};
 
   cplx<R> p[4][3], h[2][3], uh[2][3], U[3][3];
-  const R one = (R)1;
+  // proj/recon tables encode 2P (generate_proj.py); the stencil needs P
+  const R one = (R)0.5;
 
 #define QA_DIR(MU)                                                        \
   {                                                                       \

@@ -1,8 +1,8 @@
 """Standalone GPU debug script (not a pytest test). Run via gpurun."""
-import torch
-import quda_amd_hip as ext
 import sys, os
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+import quda_amd_hip as ext
 from quda_amd import LatticeGeometry, SpinorField, GaugeField
 from quda_amd.ops import blas
 from quda_amd.ops.dispatch import dslash_wilson
