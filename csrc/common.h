@@ -145,32 +145,41 @@ struct SpinorAcc {
 };
 
 // ---------------------------------------------------------------------------
-// gauge accessor: [mu][parity][nch][V][2] complex-pair chunks; recon 18|12
-// (recon-12: row2 = conj(row0 x row1), ref gauge_field_order.h:2369)
+// gauge accessor — "stencil" layout (quda_amd/fields/gauge.py):
+// per parity, per site: 8 links (slots Q = 0-3 fwd U_mu(x), 4-7 bwd
+// U_mu(x-mu), pre-shifted) packed contiguously, 16-byte chunked.
+// recon-12: row2 = conj(row0 x row1) (ref gauge_field_order.h:2369).
+// Every load is site-local; the slot Q is compile-time so the chunk range
+// covering elements [Q*RECON, (Q+1)*RECON) folds to constants.
 // ---------------------------------------------------------------------------
 template <typename Prec, int RECON>
 struct GaugeAcc {
   using S = typename Prec::Store;
   using R = typename Prec::Real;
-  static constexpr int NCH = RECON / 2;  // complex chunks per link
-  const S *data;
+  static constexpr int W = Prec::W;
+  static constexpr int NCH = (8 * RECON) / W;  // chunks per site
+  const S *data;  // parity-adjusted base: [NCH][V][W]
   long V;
 
-  __device__ __forceinline__ void load(cplx<R> u[3][3], int mu, int parity, long i) const {
-    const S *base = data + (((long)(mu * 2 + parity) * NCH) * V + i) * 2;
-    S tmp[RECON];
+  template <int Q>
+  __device__ __forceinline__ void load(cplx<R> u[3][3], long i) const {
+    constexpr int e0 = Q * RECON;
+    constexpr int c0 = e0 / W;
+    constexpr int c1 = (e0 + RECON - 1) / W;
+    constexpr int NC = c1 - c0 + 1;
+    constexpr int off = e0 - c0 * W;
+    S tmp[NC * W];
 #pragma unroll
-    for (int ch = 0; ch < NCH; ++ch)
-      load_chunk<S, 2>(base + (long)ch * V * 2, tmp + ch * 2);
+    for (int c = 0; c < NC; ++c)
+      load_chunk<S, W>(data + ((long)(c0 + c) * V + i) * W, tmp + c * W);
 #pragma unroll
     for (int k = 0; k < RECON / 2; ++k)
-      u[k / 3][k % 3] = {(R)tmp[2 * k], (R)tmp[2 * k + 1]};
+      u[k / 3][k % 3] = {(R)tmp[off + 2 * k], (R)tmp[off + 2 * k + 1]};
     if constexpr (RECON == 12) {
-      // row2 = conj(row0 x row1)
 #pragma unroll
       for (int c = 0; c < 3; ++c) {
-        int c1 = (c + 1) % 3, c2 = (c + 2) % 3;
-        u[2][c] = conj(u[0][c1] * u[1][c2] - u[0][c2] * u[1][c1]);
+        int a = (c + 1) % 3, b = (c + 2) % 3;
+        u[2][c] = conj(u[0][a] * u[1][b] - u[0][b] * u[1][a]);
       }
     }
   }

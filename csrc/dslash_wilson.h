@@ -100,14 +100,14 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
     long j = neighbor_cb(xc, MU, +1, d);                                  \
     in.load(p, j);                                                        \
     if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);    \
-    g.load(U, MU, parity, i);                                             \
+    g.template load<MU>(U, i);                                            \
     su3_mul_half(uh, U, h);                                               \
     if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
     else recon_##MU##_1(acc, uh, one);                                    \
     j = neighbor_cb(xc, MU, -1, d);                                       \
     in.load(p, j);                                                        \
     if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);    \
-    g.load(U, MU, parity ^ 1, j);                                         \
+    g.template load<4 + MU>(U, i);                                        \
     su3_dagmul_half(uh, U, h);                                            \
     if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \
     else recon_##MU##_0(acc, uh, one);                                    \

@@ -1,15 +1,24 @@
-"""GaugeField equivalent (ref: lib/gauge_field.cpp).
+"""GaugeField (ref: lib/gauge_field.cpp, include/gauge_field_order.h) —
+MI355X-first "stencil" layout.
 
-Storage: [4 (mu), n_parity, n_chunk, V_cb, 2] in complex-pair chunks
-(double2 = 16 B, float2 = 8 B, half2 = 4 B per lane — all coalesced).
+Instead of the reference's per-direction FloatN arrays (which make the
+backward-hop link loads neighbor-indexed), each checkerboard site stores ALL
+8 links its dslash stencil needs, contiguously:
 
-reconstruct:
-  "none"   : 18 reals/link (full 3x3 complex matrix)
-  "twelve" : 12 reals/link (rows 0,1; row2 = conj(row0 x row1) in-kernel)
-(ref: gauge_field_order.h:2369 reconstruct mappers; 13/9/8 variants are
-future work)
+    site x, parity p:  [ U_0(x) .. U_3(x),  U_0(x-0) .. U_3(x-3) ]
 
-Oracle layout: to_complex() -> [4, n_parity, V_cb, 3, 3] complex, U_mu(x).
+i.e. the backward links are a pre-shifted copy. Every gauge load in the
+kernel is then site-local and 16-byte vectorized for every precision x
+reconstruct combination; only spinor loads remain neighbor-indexed. Costs
+2x gauge memory — 288 GB HBM3E per GPU makes that free (SURVEY.md 2.11
+rebuild note), and it removes 8 neighbor-index computations + all strided
+sub-16B gauge loads per site.
+
+Tensor shape: [2 (parity), 8*L/w, V_cb, w] with L = reals/link
+(18 = full 3x3, 12 = rows 0,1 with row2 = conj(row0 x row1) reconstructed
+in-kernel — ref gauge_field_order.h:2369), w = 16 bytes / itemsize.
+Component order within a site: link slot q in 0..7 (0-3 fwd mu, 4-7 bwd mu),
+comp = q*L + (row*3+col)*2 + reim.
 """
 
 from __future__ import annotations
@@ -19,7 +28,7 @@ from typing import Optional
 import torch
 
 from .geometry import LatticeGeometry
-from .layout import DTYPE_OF
+from .layout import DTYPE_OF, WIDTH_OF
 
 RECON_COMPS = {"none": 18, "twelve": 12}
 
@@ -31,8 +40,10 @@ class GaugeField:
         self.geo = geo
         self.precision = precision
         self.reconstruct = reconstruct
-        ncomp = RECON_COMPS[reconstruct]
-        shape = (4, 2, ncomp // 2, geo.volume_cb, 2)
+        L = RECON_COMPS[reconstruct]
+        w = WIDTH_OF[precision]
+        assert (8 * L) % w == 0
+        shape = (2, (8 * L) // w, geo.volume_cb, w)
         if data is not None:
             assert tuple(data.shape) == shape, (data.shape, shape)
             self.data = data
@@ -43,31 +54,55 @@ class GaugeField:
     def device(self):
         return self.data.device
 
+    @property
+    def L(self) -> int:
+        return RECON_COMPS[self.reconstruct]
+
     def to(self, device) -> "GaugeField":
         return GaugeField(self.geo, self.precision, device, self.reconstruct,
                           data=self.data.to(device))
 
     # ------------------------------------------------------------------
     def to_complex(self, dtype=torch.complex128) -> torch.Tensor:
-        """-> [4, 2, V_cb, 3, 3] complex with row2 reconstructed if needed."""
-        d = self.data.to(torch.float64 if dtype == torch.complex128 else torch.float32)
-        nch = d.shape[2]
+        """-> [4, 2, V_cb, 3, 3] complex U_mu(x) (from the fwd slots)."""
+        L = self.L
         V = self.geo.volume_cb
-        c = torch.view_as_complex(d.movedim(2, 3).contiguous())  # [4,2,V,nch]
-        if self.reconstruct == "none":
-            return c.reshape(4, 2, V, 3, 3)
-        rows01 = c.reshape(4, 2, V, 2, 3)
-        row2 = torch.cross(rows01[..., 0, :], rows01[..., 1, :], dim=-1).conj()
-        return torch.cat([rows01, row2.unsqueeze(-2)], dim=-2)
+        d = self.data.to(torch.float64 if dtype == torch.complex128 else torch.float32)
+        flat = d.movedim(1, 2).reshape(2, V, 8 * L)  # [p, V, comps]
+        out = torch.empty((4, 2, V, 3, 3), dtype=dtype, device=d.device)
+        for mu in range(4):
+            rows = torch.view_as_complex(
+                flat[:, :, mu * L:(mu + 1) * L].reshape(2, V, L // 2, 2).contiguous())
+            if self.reconstruct == "none":
+                out[mu] = rows.reshape(2, V, 3, 3)
+            else:
+                r01 = rows.reshape(2, V, 2, 3)
+                r2 = torch.cross(r01[..., 0, :], r01[..., 1, :], dim=-1).conj()
+                out[mu] = torch.cat([r01, r2.unsqueeze(-2)], dim=-2)
+        return out
 
     def from_complex(self, u: torch.Tensor) -> "GaugeField":
+        """u: [4, 2, V_cb, 3, 3] complex -> fill fwd + shifted bwd slots."""
+        L = self.L
         V = self.geo.volume_cb
         assert u.shape == (4, 2, V, 3, 3)
-        if self.reconstruct == "twelve":
-            u = u[..., 0:2, :]
-        nch = RECON_COMPS[self.reconstruct] // 2
-        flat = torch.view_as_real(u.reshape(4, 2, V, nch))  # [4,2,V,nch,2]
-        self.data.copy_(flat.movedim(3, 2).contiguous().to(self.data.dtype))
+        dev = u.device
+        flat = torch.empty((2, V, 8 * L),
+                           dtype=torch.float64 if u.dtype == torch.complex128 else torch.float32,
+                           device=dev)
+        nrows = L // 6  # 3 or 2 link rows stored
+        for p in (0, 1):
+            for mu in range(4):
+                fwd = u[mu, p][:, 0:nrows, :]                      # [V,nrows,3]
+                flat[p, :, mu * L:(mu + 1) * L] = torch.view_as_real(
+                    fwd).reshape(V, L)
+                bwd_idx = self.geo.neighbor_cb(p, mu, -1).to(dev)
+                bwd = u[mu, 1 - p][bwd_idx][:, 0:nrows, :]
+                flat[p, :, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
+                    bwd).reshape(V, L)
+        w = WIDTH_OF[self.precision]
+        native = flat.reshape(2, V, (8 * L) // w, w).movedim(2, 1).contiguous()
+        self.data.copy_(native.to(self.data.dtype))
         return self
 
     # -- fills ----------------------------------------------------------
@@ -78,18 +113,15 @@ class GaugeField:
         return self.from_complex(u)
 
     def random_su3_(self, seed: Optional[int] = None, sigma: float = 1.0) -> "GaugeField":
-        """Random SU(3) links: Gaussian complex matrix -> Gram-Schmidt rows ->
-        det-phase fix (same role as the reference tests'
+        """Random SU(3) links (same role as the reference tests'
         constructRandomGaugeField, tests/utils/host_utils.cpp:1022)."""
         g = torch.Generator(device="cpu")
         if seed is not None:
             g.manual_seed(seed)
         V = self.geo.volume_cb
         m = torch.randn((4, 2, V, 3, 3, 2), generator=g, dtype=torch.float64) * sigma
-        # bias toward identity for small sigma (keeps links near unit gauge)
-        u = torch.view_as_complex(m)
-        u = project_su3(u)
-        return self.from_complex(u.to(self.device))
+        u = project_su3(torch.view_as_complex(m).to(self.device))
+        return self.from_complex(u)
 
     def __repr__(self):
         return (f"GaugeField({self.geo.dims}, {self.precision}, "
@@ -104,6 +136,4 @@ def project_su3(u: torch.Tensor) -> torch.Tensor:
     r1 = r1 - (r0.conj() * r1).sum(-1, keepdim=True) * r0
     r1 = r1 / r1.norm(dim=-1, keepdim=True)
     r2 = torch.cross(r0, r1, dim=-1).conj()
-    out = torch.stack([r0, r1, r2], dim=-2)
-    # det is now exactly +1 by construction (r2 = conj(r0 x r1))
-    return out
+    return torch.stack([r0, r1, r2], dim=-2)

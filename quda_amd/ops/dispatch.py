@@ -63,7 +63,7 @@ def norm_or_empty(f: Optional[SpinorField]):
     return f.norm
 
 
-RECON_OF = {9: 18, 6: 12}  # gauge chunk count -> recon
+from ..fields.gauge import RECON_COMPS
 
 
 def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
@@ -84,7 +84,7 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
             out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
             gauge.data, cl_t, xf.data, norm_or_empty(xf),
             list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
-            bool(dagger), mode, xpay, float(a), RECON_OF[gauge.data.shape[2]])
+            bool(dagger), mode, xpay, float(a), RECON_COMPS[gauge.reconstruct])
         return out
     # ---- oracle path ----
     u = gauge.to_complex()
