@@ -40,9 +40,21 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
                           at::Tensor x, at::Tensor x_n,
                           std::vector<int64_t> dims, int64_t parity_offset,
                           int64_t Vcb, int64_t parity, bool dagger,
-                          int64_t mode, bool xpay, double a, int64_t recon) {
+                          int64_t mode, bool xpay, double a, int64_t recon,
+                          std::vector<at::Tensor> ghost,
+                          std::vector<at::Tensor> ghost_nrm,
+                          std::vector<int64_t> face_cb, int64_t comm_mask) {
   TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
   DslashCall c{};
+  c.comm_mask = (int)comm_mask;
+  if (comm_mask) {
+    TORCH_CHECK(ghost.size() == 8 && ghost_nrm.size() == 8 && face_cb.size() == 4);
+    for (int k = 0; k < 8; ++k) {
+      c.ghost[k] = ptr_or_null(ghost[k]);
+      c.ghost_nrm[k] = (const float *)ptr_or_null(ghost_nrm[k]);
+    }
+    for (int k = 0; k < 4; ++k) c.face_cb[k] = face_cb[k];
+  }
   c.out = field_of(out, out_n, Vcb);
   c.in = field_of(in, in_n, Vcb);
   c.x = field_of(x, x_n, Vcb);
@@ -63,6 +75,30 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
     case 2: launch_dslash_wilson_half(c, stream()); break;
   }
   check_launch("dslash_wilson");
+}
+
+static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
+                      at::Tensor in_n, std::vector<int64_t> dims,
+                      int64_t parity_offset, int64_t Vcb, int64_t parity,
+                      int64_t mu, int64_t s01, int64_t edge, int64_t Fcb) {
+  PackCall c{};
+  c.in = field_of(in, in_n, Vcb);
+  c.dst = dst.data_ptr();
+  c.dst_nrm = (float *)ptr_or_null(dst_nrm);
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.mu = (int)mu;
+  c.s01 = (int)s01;
+  c.edge = (int)edge;
+  c.Fcb = Fcb;
+  switch (prec_of(in)) {
+    case 0: launch_pack_face_double(c, stream()); break;
+    case 1: launch_pack_face_single(c, stream()); break;
+    case 2: launch_pack_face_half(c, stream()); break;
+  }
+  check_launch("pack_face");
 }
 
 static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
@@ -112,6 +148,7 @@ static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover) dslash");
+  m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
   m.def("blas_op", &blas_op, "fused blas/reduction");
   m.def("convert", &convert, "precision conversion copy");
   m.def("clover_apply", &clover_apply, "clover site-matrix apply");

@@ -14,6 +14,7 @@
 
 #include "common.h"
 #include "generated/proj.h"
+#include "halo.h"
 
 enum CloverMode { PLAIN = 0, CLOV_POST = 1, CLOV_X = 2 };
 
@@ -74,11 +75,11 @@ __device__ __forceinline__ void clover_mul(cplx<R> out[4][3], const R diag[2][6]
   }
 }
 
-template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
+template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY, bool COMMS = false>
 __global__ __launch_bounds__(256) void k_dslash_wilson(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
-    SpinorAcc<Prec> x) {
+    SpinorAcc<Prec> x, GhostAcc<Prec> gh) {
   using R = typename Prec::Real;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= d.Vcb) return;
@@ -97,16 +98,24 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
 
 #define QA_DIR(MU)                                                        \
   {                                                                       \
-    long j = neighbor_cb(xc, MU, +1, d);                                  \
-    in.load(p, j);                                                        \
-    if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);    \
+    if (COMMS && gh.active(MU) && xc[MU] == d.X[MU] - 1) {                \
+      gh.load(h, MU, 1, ghost_idx(xc, MU, d));                            \
+    } else {                                                              \
+      long j = neighbor_cb(xc, MU, +1, d);                                \
+      in.load(p, j);                                                      \
+      if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);  \
+    }                                                                     \
     g.template load<MU>(U, i);                                            \
     su3_mul_half(uh, U, h);                                               \
     if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
     else recon_##MU##_1(acc, uh, one);                                    \
-    j = neighbor_cb(xc, MU, -1, d);                                       \
-    in.load(p, j);                                                        \
-    if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);    \
+    if (COMMS && gh.active(MU) && xc[MU] == 0) {                          \
+      gh.load(h, MU, 0, ghost_idx(xc, MU, d));                            \
+    } else {                                                              \
+      long j = neighbor_cb(xc, MU, -1, d);                                \
+      in.load(p, j);                                                      \
+      if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);  \
+    }                                                                     \
     g.template load<4 + MU>(U, i);                                        \
     su3_dagmul_half(uh, U, h);                                            \
     if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \

@@ -15,27 +15,65 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   GaugeAcc<Prec, RECON> g{gbase, c.Vcb};
   CloverAcc<Prec> cl{(const typename Prec::Store *)c.clover, c.Vcb};
   LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
+  GhostAcc<Prec> gh{};
+  gh.mask = c.comm_mask;
+  for (int k = 0; k < 8; ++k) {
+    gh.buf[k] = (const typename Prec::Store *)c.ghost[k];
+    gh.nrm[k] = c.ghost_nrm[k];
+  }
+  for (int k = 0; k < 4; ++k) gh.Fcb[k] = c.face_cb[k];
   int blk = 256;
   int grid = (int)((c.Vcb + blk - 1) / blk);
   R a = (R)c.a;
 
-#define QA_LAUNCH(DAG, MODE, XPAY)                                            \
-  hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY>),         \
+#define QA_LAUNCH(DAG, MODE, XPAY, COMMS)                                     \
+  hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY, COMMS>),  \
                      dim3(grid), dim3(blk), 0, st, out, in, g, cl, d,         \
-                     c.parity, a, x)
+                     c.parity, a, x, gh)
 
-  if (!c.dagger) {
-    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(false, PLAIN, false);
-    else if (c.mode == PLAIN) QA_LAUNCH(false, PLAIN, true);
-    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(false, CLOV_POST, false);
-    else if (c.mode == CLOV_POST) QA_LAUNCH(false, CLOV_POST, true);
-    else QA_LAUNCH(false, CLOV_X, true);
-  } else {
-    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(true, PLAIN, false);
-    else if (c.mode == PLAIN) QA_LAUNCH(true, PLAIN, true);
-    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(true, CLOV_POST, false);
-    else if (c.mode == CLOV_POST) QA_LAUNCH(true, CLOV_POST, true);
-    else QA_LAUNCH(true, CLOV_X, true);
+#define QA_DISPATCH(COMMS)                                                     \
+  if (!c.dagger) {                                                             \
+    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(false, PLAIN, false, COMMS);     \
+    else if (c.mode == PLAIN) QA_LAUNCH(false, PLAIN, true, COMMS);            \
+    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(false, CLOV_POST, false, COMMS); \
+    else if (c.mode == CLOV_POST) QA_LAUNCH(false, CLOV_POST, true, COMMS);    \
+    else QA_LAUNCH(false, CLOV_X, true, COMMS);                                \
+  } else {                                                                     \
+    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(true, PLAIN, false, COMMS);      \
+    else if (c.mode == PLAIN) QA_LAUNCH(true, PLAIN, true, COMMS);             \
+    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(true, CLOV_POST, false, COMMS); \
+    else if (c.mode == CLOV_POST) QA_LAUNCH(true, CLOV_POST, true, COMMS);     \
+    else QA_LAUNCH(true, CLOV_X, true, COMMS);                                 \
   }
+
+  if (c.comm_mask) { QA_DISPATCH(true) } else { QA_DISPATCH(false) }
+#undef QA_DISPATCH
 #undef QA_LAUNCH
+}
+
+template <typename Prec>
+static void pack_launch(const PackCall &c, hipStream_t st) {
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.Vcb};
+  LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
+  int blk = 256;
+  int grid = (int)((c.Fcb + blk - 1) / blk);
+  auto *dst = (typename Prec::Store *)c.dst;
+
+#define QA_PACK(MU, S01, EDGE)                                                 \
+  hipLaunchKernelGGL((k_pack_face<Prec, MU, S01, EDGE>), dim3(grid), dim3(blk),\
+                     0, st, dst, c.dst_nrm, in, d, c.parity, c.Fcb)
+#define QA_PACK_MU(MU)                                                         \
+  case MU:                                                                     \
+    if (c.s01 == 0) { if (c.edge) QA_PACK(MU, 0, true); else QA_PACK(MU, 0, false); } \
+    else            { if (c.edge) QA_PACK(MU, 1, true); else QA_PACK(MU, 1, false); } \
+    break;
+
+  switch (c.mu) {
+    QA_PACK_MU(0)
+    QA_PACK_MU(1)
+    QA_PACK_MU(2)
+    QA_PACK_MU(3)
+  }
+#undef QA_PACK_MU
+#undef QA_PACK
 }

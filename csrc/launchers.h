@@ -49,11 +49,35 @@ struct DslashCall {
   bool xpay;
   double a;
   int recon;  // 18 or 12
+  // halo: ghost recv buffers per [2*mu+dir] (dir 1 = from +mu neighbor);
+  // null when mask bit mu unset. comm_mask==0 => pure-local kernel.
+  const void *ghost[8];
+  const float *ghost_nrm[8];
+  long face_cb[4];
+  int comm_mask;
 };
 
 void launch_dslash_wilson_double(const DslashCall &c, hipStream_t st);
 void launch_dslash_wilson_single(const DslashCall &c, hipStream_t st);
 void launch_dslash_wilson_half(const DslashCall &c, hipStream_t st);
+
+// pack one (mu, s01, edge) face of `in` into dst (see csrc/halo.h)
+struct PackCall {
+  BlasField in;       // dslash input spinor (single parity view)
+  void *dst;
+  float *dst_nrm;     // half only
+  int Xdim[4];
+  int parity_offset;
+  long Vcb;
+  int parity;         // parity of `in`
+  int mu;
+  int s01;            // projector sign index (dir XOR dagger)
+  int edge;           // 0: x_mu = 0 face, 1: x_mu = X-1 face
+  long Fcb;
+};
+void launch_pack_face_double(const PackCall &c, hipStream_t st);
+void launch_pack_face_single(const PackCall &c, hipStream_t st);
+void launch_pack_face_half(const PackCall &c, hipStream_t st);
 
 struct CloverApplyCall {
   BlasField out, in;

@@ -82,7 +82,14 @@ class GaugeField:
         return out
 
     def from_complex(self, u: torch.Tensor) -> "GaugeField":
-        """u: [4, 2, V_cb, 3, 3] complex -> fill fwd + shifted bwd slots."""
+        """u: [4, 2, V_cb, 3, 3] complex -> fill fwd + shifted bwd slots.
+
+        When the process grid partitions a dim (comms.comm_mask), the bwd
+        slots of the x_mu=0 face are U_mu links owned by the -mu neighbor:
+        exchanged here once at load time (analogue of the reference's gauge
+        exchangeGhost, lib/gauge_field.cpp:453 — our stencil layout folds
+        the gauge ghost INTO the field, so the dslash needs no gauge ghost
+        at apply time)."""
         L = self.L
         V = self.geo.volume_cb
         assert u.shape == (4, 2, V, 3, 3)
@@ -100,10 +107,56 @@ class GaugeField:
                 bwd = u[mu, 1 - p][bwd_idx][:, 0:nrows, :]
                 flat[p, :, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
                     bwd).reshape(V, L)
+        # fix up bwd slots on partitioned-dim boundary faces
+        from ..parallel import comms
+        from ..parallel.halo import active_dims, exchange_tensors
+        mask = comms.comm_mask()
+        for mu in active_dims(mask):
+            geo = self.geo
+            hi = geo.dims[mu] - 1
+            fcb = geo.face_volume_cb(mu)
+            # send my far-face U_mu rows (both parities) toward +mu
+            send = torch.empty((2, fcb, nrows, 3), dtype=u.dtype, device=dev)
+            for q in (0, 1):
+                send[q] = u[mu, q][geo.face_index_cb(q, mu, hi).to(dev)][:, 0:nrows, :]
+            recv = torch.empty_like(send)
+            exchange_tensors({(mu, 1): send}, {(mu, 0): recv})
+            for p in (0, 1):
+                fidx0 = geo.face_index_cb(p, mu, 0).to(dev)
+                flat[p, fidx0, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
+                    recv[1 - p]).reshape(fcb, L)
         w = WIDTH_OF[self.precision]
         native = flat.reshape(2, V, (8 * L) // w, w).movedim(2, 1).contiguous()
         self.data.copy_(native.to(self.data.dtype))
+        self._bwd_ghost_cache = {}
         return self
+
+    def bwd_ghost(self, mu: int, parity: int) -> torch.Tensor:
+        """[Fcb, 3, 3] complex U_mu(x-mu) for the x_mu=0 face sites of
+        `parity`, in ghost order — decoded from the stored bwd slots (used
+        by the CPU oracle on partitioned dims; exact same values the HIP
+        kernel reads)."""
+        cache = self.__dict__.setdefault("_bwd_ghost_cache", {})
+        key = (mu, parity)
+        if key not in cache:
+            L = self.L
+            geo = self.geo
+            V = geo.volume_cb
+            fidx0 = geo.face_index_cb(parity, mu, 0)
+            d = self.data[parity].to(torch.float64)          # [NCH, V, w]
+            flat = d.movedim(0, 1).reshape(V, 8 * L)
+            rows = torch.view_as_complex(
+                flat[fidx0.to(d.device), (4 + mu) * L:(5 + mu) * L]
+                .reshape(-1, L // 2, 2).contiguous())
+            nrows = L // 6
+            rows = rows.reshape(-1, nrows, 3)
+            if self.reconstruct == "none":
+                out = rows
+            else:
+                r2 = torch.cross(rows[:, 0, :], rows[:, 1, :], dim=-1).conj()
+                out = torch.cat([rows, r2.unsqueeze(1)], dim=1)
+            cache[key] = out
+        return cache[key]
 
     # -- fills ----------------------------------------------------------
     def unit_(self) -> "GaugeField":

@@ -146,6 +146,24 @@ class LatticeGeometry:
             cache[key] = self.cb_of_lex[nbr_lex[my_lex]].contiguous()
         return cache[key]
 
+    def face_index_cb(self, parity: int, dim: int, edge: int) -> torch.Tensor:
+        """[Fcb] int64 cb indices of the parity-`parity` sites on the face
+        coords[dim]==edge, ordered by ghost index (transverse coords
+        flattened lowest-dim-fastest, >>1) — the order csrc/halo.h
+        ghost_idx/face_coords use, shared by pack and unpack."""
+        key = ("face", parity, dim, edge)
+        cache = self.__dict__.setdefault("_nbr_cache", {})
+        if key not in cache:
+            c = self.coords_of_cb(parity).to(torch.int64)
+            sel = (c[:, dim] == edge).nonzero(as_tuple=True)[0]
+            rd = [i for i in range(4) if i != dim]
+            cc = c[sel]
+            flat3 = ((cc[:, rd[2]] * self.dims[rd[1]] + cc[:, rd[1]])
+                     * self.dims[rd[0]] + cc[:, rd[0]])
+            order = torch.argsort(flat3)
+            cache[key] = sel[order].contiguous()
+        return cache[key]
+
     def boundary_mask_cb(self, parity: int, dim: int, displacement: int) -> torch.Tensor:
         """[volume_cb] bool: True where the displaced neighbor wraps around the
         local lattice boundary in `dim` (i.e. lives in the halo when `dim` is

@@ -72,24 +72,43 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                   clover=None, clover_inverse: bool = False):
     """Apply the fused Wilson(-clover) stencil; out at `parity`, in at the
     opposite parity; x (same parity as out) enables the xpay term."""
+    from ..parallel import comms
     geo = out.geo
     xpay = x is not None
+    mask = comms.comm_mask()
     if on_gpu(out, inp):
         ext = hip_ext()
         cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)
         if mode != PLAIN:
             cl_t = clover.inv_data if clover_inverse else clover.data
         xf = x if x is not None else out
+        ghosts, nrms, face_cb = [], [], []
+        if mask:
+            from ..parallel.halo import get_spinor_halo
+            h = get_spinor_halo(geo, inp.precision, inp.device, mask)
+            h.pack(ext, inp, 1 - parity, bool(dagger))
+            h.exchange()
+            ghosts, nrms, face_cb = h.ghost_args()
         ext.dslash_wilson(
             out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
             gauge.data, cl_t, xf.data, norm_or_empty(xf),
             list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
-            bool(dagger), mode, xpay, float(a), RECON_COMPS[gauge.reconstruct])
+            bool(dagger), mode, xpay, float(a), RECON_COMPS[gauge.reconstruct],
+            ghosts, nrms, face_cb, mask)
         return out
     # ---- oracle path ----
     u = gauge.to_complex()
     psi = inp.to_complex()[0]
-    res = ref.dslash_wilson_parity(u, psi, geo, parity, dagger)
+    halo = None
+    if mask:
+        from ..parallel.halo import active_dims, exchange_psi_oracle
+        halo = {
+            "mask": mask,
+            "psi": exchange_psi_oracle(psi, geo, 1 - parity, mask),
+            "u_bwd": {mu: gauge.bwd_ghost(mu, parity)
+                      for mu in active_dims(mask)},
+        }
+    res = ref.dslash_wilson_parity(u, psi, geo, parity, dagger, halo=halo)
     if mode == CLOV_POST:
         A = clover.to_complex(inverse=clover_inverse)[parity]
         res = ref.apply_clover(A, res)

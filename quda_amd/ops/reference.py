@@ -36,25 +36,42 @@ def _gamma_tensors(device, dtype):
 
 def dslash_wilson_parity(u: torch.Tensor, psi: torch.Tensor,
                          geo: LatticeGeometry, parity: int,
-                         dagger: bool = False) -> torch.Tensor:
+                         dagger: bool = False, halo=None) -> torch.Tensor:
     """Apply the parity-hopping Wilson stencil: out_parity <- D psi_other.
 
     psi is the SOURCE (at parity 1-parity), result is at `parity`.
+    `halo` (multi-rank / forced-partition): dict with 'mask' (bit per
+    partitioned dim), 'psi' {(mu,dir): [Fcb,4,3]} ghost spinors in ghost
+    order (dir=1 from +mu), 'u_bwd' {mu: [Fcb,3,3]} U_mu(x-mu) ghosts for
+    the x_mu=0 face of this parity.
     """
     dev, dt = psi.device, psi.dtype
     P = _gamma_tensors(dev, dt)
     other = 1 - parity
     out = torch.zeros_like(psi)
     sgn = 1 if not dagger else 0  # index into P: dagger swaps minus<->plus
+    mask = halo["mask"] if halo else 0
     for mu in range(4):
+        part = (mask >> mu) & 1
         # forward: U_mu(x) P(-mu) psi(x+mu)
         fwd_idx = geo.neighbor_cb(parity, mu, +1).to(dev)
-        proj = torch.einsum("st,vtc->vsc", P[mu, 1 - sgn], psi[fwd_idx])
+        psi_f = psi[fwd_idx]
+        if part:
+            fidx = geo.face_index_cb(parity, mu, geo.dims[mu] - 1).to(dev)
+            psi_f[fidx] = halo["psi"][(mu, 1)].to(dt)
+        proj = torch.einsum("st,vtc->vsc", P[mu, 1 - sgn], psi_f)
         out += torch.einsum("vij,vsj->vsi", u[mu, parity], proj)
         # backward: U_mu(x-mu)^dag P(+mu) psi(x-mu)
         bwd_idx = geo.neighbor_cb(parity, mu, -1).to(dev)
-        proj = torch.einsum("st,vtc->vsc", P[mu, sgn], psi[bwd_idx])
-        out += torch.einsum("vji,vsj->vsi", u[mu, other][bwd_idx].conj(), proj)
+        psi_b = psi[bwd_idx]
+        u_b = u[mu, other][bwd_idx]
+        if part:
+            fidx0 = geo.face_index_cb(parity, mu, 0).to(dev)
+            psi_b[fidx0] = halo["psi"][(mu, 0)].to(dt)
+            u_b = u_b.clone()
+            u_b[fidx0] = halo["u_bwd"][mu].to(dt)
+        proj = torch.einsum("st,vtc->vsc", P[mu, sgn], psi_b)
+        out += torch.einsum("vji,vsj->vsi", u_b.conj(), proj)
     return out
 
 

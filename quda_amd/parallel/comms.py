@@ -18,7 +18,8 @@ from typing import Optional, Tuple
 import torch
 import torch.distributed as dist
 
-_STATE = {"grid": (1, 1, 1, 1), "coords": (0, 0, 0, 0), "initialized": False}
+_STATE = {"grid": (1, 1, 1, 1), "coords": (0, 0, 0, 0), "initialized": False,
+          "forced_mask": 0}
 
 
 def is_distributed() -> bool:
@@ -84,7 +85,24 @@ def neighbor_rank(dim: int, displacement: int) -> int:
 
 
 def is_partitioned(dim: int) -> bool:
-    return _STATE["grid"][dim] > 1
+    return _STATE["grid"][dim] > 1 or (_STATE["forced_mask"] >> dim) & 1
+
+
+def comm_mask() -> int:
+    """Bitmask of partitioned dims (halo-exchange dims). Includes forced
+    self-wraparound dims (the reference's --partition test mode,
+    tests/CMakeLists.txt partition matrix)."""
+    m = _STATE["forced_mask"]
+    for d in range(4):
+        if _STATE["grid"][d] > 1:
+            m |= 1 << d
+    return m
+
+
+def set_forced_partition(mask: int) -> None:
+    """Force halo-exchange code paths for dims in `mask` even at grid
+    extent 1 (self-wraparound; results must be identical)."""
+    _STATE["forced_mask"] = int(mask)
 
 
 def parity_offset_of_rank(local_dims) -> int:
@@ -100,7 +118,8 @@ def allreduce_sum(x):
     side). The latency-critical per-iteration collective."""
     if not is_distributed():
         return x
-    t = torch.tensor([x], dtype=torch.float64)
+    dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+    t = torch.tensor([x], dtype=torch.float64, device=dev)
     dist.all_reduce(t)
     return t.item()
 

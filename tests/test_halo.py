@@ -1,0 +1,210 @@
+"""Halo-exchange engine tests (analogue of the reference's partition-mask
+dslash_ctest matrix, tests/CMakeLists.txt:280 — self-wraparound forced
+partitions on one process, and real gloo multi-process runs compared
+against a single full-lattice oracle)."""
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.fields.clover import CloverField
+from quda_amd.models import DiracCloverPC, DiracWilson
+from quda_amd.ops import reference as ref
+from quda_amd.ops.dispatch import dslash_wilson
+from quda_amd.parallel import comms
+from quda_amd.solvers import cg_solve
+
+
+@pytest.fixture(autouse=True)
+def _reset_partition():
+    yield
+    comms.set_forced_partition(0)
+
+
+# ---------------------------------------------------------------------------
+# self-wraparound: forced partition on 1 process must not change results
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("mask", [0b1000, 0b0001, 0b1111])
+@pytest.mark.parametrize("dagger", [False, True])
+def test_self_wraparound_dslash_cpu(mask, dagger):
+    geo = LatticeGeometry((4, 6, 4, 8))
+    g = GaugeField(geo, "double").random_su3_(seed=3)
+    src = SpinorField(geo, "double", n_parity=1).gaussian_(seed=4)
+    out_ref = SpinorField(geo, "double", n_parity=1)
+    dslash_wilson(out_ref, src, g, 0, dagger=dagger)
+    comms.set_forced_partition(mask)
+    g2 = GaugeField(geo, "double").random_su3_(seed=3)  # rebuild w/ exchange
+    out = SpinorField(geo, "double", n_parity=1)
+    dslash_wilson(out, src, g2, 0, dagger=dagger)
+    err = (out.to_complex() - out_ref.to_complex()).abs().max().item()
+    assert err < 1e-12, f"mask={mask:04b} err={err}"
+
+
+def test_self_wraparound_cg_cpu():
+    geo = LatticeGeometry((4, 4, 4, 4))
+    kappa = 0.12
+    comms.set_forced_partition(0b1001)
+    g = GaugeField(geo, "double").random_su3_(seed=5)
+    u = g.to_complex()
+    A = ref.clover_matrix(u, geo, kappa, 1.0)
+    cl = CloverField(geo, "double").from_matrices(A)
+    d = DiracCloverPC(g, cl, kappa)
+    b = SpinorField(geo, "double", n_parity=1).gaussian_(seed=6)
+    x = SpinorField(geo, "double", n_parity=1)
+    stats = cg_solve(d, x, b, tol=1e-10, maxiter=200)
+    assert stats.converged
+
+
+# ---------------------------------------------------------------------------
+# real multi-process (gloo): local results must match the global lattice
+# ---------------------------------------------------------------------------
+
+GLOBAL_DIMS = (4, 4, 4, 8)
+
+
+def _global_fields(seed=11):
+    """Deterministic global gauge (lex layout) + source, same on all ranks."""
+    gg = LatticeGeometry(GLOBAL_DIMS)
+    gen = torch.Generator().manual_seed(seed)
+    from quda_amd.fields.gauge import project_su3
+    m = torch.randn((4, gg.volume, 3, 3, 2), generator=gen, dtype=torch.float64)
+    u_lex = project_su3(torch.view_as_complex(m))
+    s = torch.randn((gg.volume, 4, 3, 2), generator=gen, dtype=torch.float64)
+    src_lex = torch.view_as_complex(s)
+    return gg, u_lex, src_lex
+
+
+def _local_slice(gg, grid, coords, lex_field):
+    """Local-lex-ordered tensor for this rank's sub-lattice."""
+    ldims = tuple(GLOBAL_DIMS[i] // grid[i] for i in range(4))
+    lg = LatticeGeometry(ldims)
+    off = torch.tensor([coords[i] * ldims[i] for i in range(4)])
+    c = lg.coords.to(torch.int64) + off
+    X, Y, Z, _ = GLOBAL_DIMS
+    glex = ((c[:, 3] * Z + c[:, 2]) * Y + c[:, 1]) * X + c[:, 0]
+    return lg, lex_field[glex]
+
+
+def _worker(rank, world, grid, init_file, mode):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=grid)
+        gg, u_lex, src_lex = _global_fields()
+        from quda_amd.fields.geometry import checkerboard_split
+        lg, u_loc_lex = _local_slice(gg, grid, comms.grid_coords(),
+                                     u_lex.movedim(0, 1))
+        _, src_loc_lex = _local_slice(gg, grid, comms.grid_coords(), src_lex)
+        u_loc = checkerboard_split(u_loc_lex, lg)          # [2, Vcb, 4, 3, 3]
+        u_loc = u_loc.permute(2, 0, 1, 3, 4).contiguous()  # [4, 2, Vcb, 3, 3]
+        src_cb = checkerboard_split(src_loc_lex, lg)       # [2, Vcb, 4, 3]
+        g = GaugeField(lg, "double").from_complex(u_loc)
+        src = SpinorField(lg, "double")
+        src.from_complex(src_cb)
+
+        # global single-process truth
+        u_g = checkerboard_split(u_lex.movedim(0, 1), gg).permute(2, 0, 1, 3, 4).contiguous()
+        src_g = checkerboard_split(src_lex, gg)
+
+        if mode == "dslash":
+            out = SpinorField(lg, "double", n_parity=1)
+            dslash_wilson(out, src.parity_view(1), g, 0, dagger=False)
+            truth_g = ref.dslash_wilson_parity(u_g, src_g[1], gg, 0)
+            # compare local slice: local even-cb sites -> global cb index
+            from quda_amd.fields.geometry import checkerboard_join
+            truth_lex = torch.zeros((gg.volume, 4, 3), dtype=torch.complex128)
+            truth_lex[gg.lex_of_cb[0]] = truth_g
+            _, truth_loc_lex = _local_slice(gg, grid, comms.grid_coords(), truth_lex)
+            truth_loc = checkerboard_split(truth_loc_lex, lg)[0]
+            err = (out.to_complex()[0] - truth_loc).abs().max().item()
+            assert err < 1e-12, f"rank{rank} dslash err={err}"
+        else:  # cg on wilson full operator
+            kappa = 0.11
+            d = DiracWilson(g, kappa)
+            x = SpinorField(lg, "double")
+            stats = cg_solve(d, x, src, tol=1e-10, maxiter=400)
+            assert stats.converged, f"rank{rank}: {stats}"
+            # check true residual against global solve
+            dg = DiracWilson(GaugeField(gg, "double").from_complex(u_g), kappa)
+            xg = SpinorField(gg, "double")
+            sg = SpinorField(gg, "double")
+            sg.from_complex(src_g)
+            stats_g = cg_solve(dg, xg, sg, tol=1e-10, maxiter=400)
+            xg_lex = torch.zeros((gg.volume, 4, 3), dtype=torch.complex128)
+            from quda_amd.fields.geometry import checkerboard_join
+            xg_lex = checkerboard_join(xg.to_complex(), gg)
+            _, x_loc_lex = _local_slice(gg, grid, comms.grid_coords(), xg_lex)
+            x_loc = checkerboard_split(x_loc_lex, lg)
+            err = (x.to_complex() - x_loc).abs().max().item()
+            assert err < 1e-7, f"rank{rank} cg err={err}"
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("grid,world", [((1, 1, 1, 2), 2), ((1, 1, 2, 2), 4)])
+def test_multiproc_dslash_gloo(grid, world):
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        init_file = f.name
+    os.unlink(init_file)
+    mp.spawn(_worker, args=(world, grid, init_file, "dslash"), nprocs=world,
+             join=True)
+
+
+def test_multiproc_cg_gloo():
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        init_file = f.name
+    os.unlink(init_file)
+    mp.spawn(_worker, args=(2, (1, 1, 1, 2), init_file, "cg"), nprocs=2,
+             join=True)
+
+
+# ---------------------------------------------------------------------------
+# GPU: native pack kernels + ghost loads via self-wraparound
+# ---------------------------------------------------------------------------
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("prec,recon", [("double", "none"), ("single", "twelve"),
+                                        ("half", "twelve")])
+@pytest.mark.parametrize("mask", [0b1000, 0b1111])
+def test_self_wraparound_dslash_gpu(prec, recon, mask):
+    geo = LatticeGeometry((8, 8, 8, 8))
+    gen = torch.Generator().manual_seed(21)
+    from quda_amd.fields.gauge import project_su3
+    m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64)
+    u = project_su3(torch.view_as_complex(m)).cuda()
+    src = SpinorField(geo, prec, "cuda", n_parity=1).gaussian_(seed=22)
+    g = GaugeField(geo, prec, "cuda", reconstruct=recon).from_complex(u)
+    out_ref = SpinorField(geo, prec, "cuda", n_parity=1)
+    for dagger in (False, True):
+        dslash_wilson(out_ref, src, g, 0, dagger=dagger)
+        comms.set_forced_partition(mask)
+        g2 = GaugeField(geo, prec, "cuda", reconstruct=recon).from_complex(u)
+        out = SpinorField(geo, prec, "cuda", n_parity=1)
+        dslash_wilson(out, src, g2, 0, dagger=dagger)
+        comms.set_forced_partition(0)
+        err = (out.to_complex() - out_ref.to_complex()).abs().max().item()
+        tol = {"double": 1e-12, "single": 1e-5, "half": 5e-3}[prec]
+        assert err < tol, f"prec={prec} mask={mask:04b} dag={dagger} err={err}"
+
+
+@pytest.mark.gpu
+def test_self_wraparound_cg_gpu():
+    geo = LatticeGeometry((8, 8, 8, 8))
+    kappa, csw = 0.13, 1.0
+    comms.set_forced_partition(0b1000)
+    g_host = GaugeField(geo, "double").random_su3_(seed=31)
+    u = g_host.to_complex().cuda()
+    A = ref.clover_matrix(u, geo, kappa, csw)
+    g = GaugeField(geo, "double", "cuda").from_complex(u)
+    cl = CloverField(geo, "double", "cuda").from_matrices(A)
+    d = DiracCloverPC(g, cl, kappa)
+    b = SpinorField(geo, "double", "cuda", n_parity=1).gaussian_(seed=32)
+    x = SpinorField(geo, "double", "cuda", n_parity=1)
+    stats = cg_solve(d, x, b, tol=1e-8, maxiter=300)
+    assert stats.converged
