@@ -43,10 +43,12 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
                           int64_t mode, bool xpay, double a, int64_t recon,
                           std::vector<at::Tensor> ghost,
                           std::vector<at::Tensor> ghost_nrm,
-                          std::vector<int64_t> face_cb, int64_t comm_mask) {
+                          std::vector<int64_t> face_cb, int64_t comm_mask,
+                          int64_t kt) {
   TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
   DslashCall c{};
   c.comm_mask = (int)comm_mask;
+  c.kt = (int)kt;
   if (comm_mask) {
     TORCH_CHECK(ghost.size() == 8 && ghost_nrm.size() == 8 && face_cb.size() == 4);
     for (int k = 0; k < 8; ++k) {

@@ -18,6 +18,18 @@
 
 enum CloverMode { PLAIN = 0, CLOV_POST = 1, CLOV_X = 2 };
 
+// kernel roles for the comm-overlap split (role of the reference's
+// INTERIOR/EXTERIOR kernel_type, include/dslash_helper.cuh):
+//   LOCAL    : no partitioned dims — pure local stencil (zero overhead)
+//   FUSED    : ghost-aware single pass (blocking-comms policy)
+//   INTERIOR : skip hops crossing a partitioned boundary; boundary sites
+//              with MODE==CLOV_POST store the raw hop sum (epilogue
+//              deferred to EXTERIOR), other modes apply the (affine)
+//              epilogue to the partial sum
+// The EXTERIOR kernel (k_dslash_wilson_exterior) adds the ghost hops and
+// completes deferred epilogues; one owner thread per boundary site.
+enum DslashKT { KT_LOCAL = 0, KT_FUSED = 1, KT_INTERIOR = 2 };
+
 // packed clover (fields/clover.py): per site 72 reals =
 // 2 chirality blocks x (6 diag + 15 lower-tri complex)
 template <typename Prec>
@@ -75,7 +87,7 @@ __device__ __forceinline__ void clover_mul(cplx<R> out[4][3], const R diag[2][6]
   }
 }
 
-template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY, bool COMMS = false>
+template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY, int KT = KT_LOCAL>
 __global__ __launch_bounds__(256) void k_dslash_wilson(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
@@ -96,30 +108,43 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
   // proj/recon tables encode 2P (generate_proj.py); the stencil needs P
   const R one = (R)0.5;
 
+  bool bnd = false;  // interior: site has >=1 hop deferred to EXTERIOR
+  if constexpr (KT == KT_INTERIOR) {
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+      if (gh.active(m) && (xc[m] == 0 || xc[m] == d.X[m] - 1)) bnd = true;
+  }
+
 #define QA_DIR(MU)                                                        \
   {                                                                       \
-    if (COMMS && gh.active(MU) && xc[MU] == d.X[MU] - 1) {                \
-      gh.load(h, MU, 1, ghost_idx(xc, MU, d));                            \
-    } else {                                                              \
-      long j = neighbor_cb(xc, MU, +1, d);                                \
-      in.load(p, j);                                                      \
-      if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);  \
+    bool cross_p = KT != KT_LOCAL && gh.active(MU) && xc[MU] == d.X[MU] - 1; \
+    if (!(KT == KT_INTERIOR && cross_p)) {                                \
+      if (KT == KT_FUSED && cross_p) {                                    \
+        gh.load(h, MU, 1, ghost_idx(xc, MU, d));                          \
+      } else {                                                            \
+        long j = neighbor_cb(xc, MU, +1, d);                              \
+        in.load(p, j);                                                    \
+        if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);\
+      }                                                                   \
+      g.template load<MU>(U, i);                                          \
+      su3_mul_half(uh, U, h);                                             \
+      if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                   \
+      else recon_##MU##_1(acc, uh, one);                                  \
     }                                                                     \
-    g.template load<MU>(U, i);                                            \
-    su3_mul_half(uh, U, h);                                               \
-    if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
-    else recon_##MU##_1(acc, uh, one);                                    \
-    if (COMMS && gh.active(MU) && xc[MU] == 0) {                          \
-      gh.load(h, MU, 0, ghost_idx(xc, MU, d));                            \
-    } else {                                                              \
-      long j = neighbor_cb(xc, MU, -1, d);                                \
-      in.load(p, j);                                                      \
-      if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);  \
+    bool cross_m = KT != KT_LOCAL && gh.active(MU) && xc[MU] == 0;        \
+    if (!(KT == KT_INTERIOR && cross_m)) {                                \
+      if (KT == KT_FUSED && cross_m) {                                    \
+        gh.load(h, MU, 0, ghost_idx(xc, MU, d));                          \
+      } else {                                                            \
+        long j = neighbor_cb(xc, MU, -1, d);                              \
+        in.load(p, j);                                                    \
+        if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);\
+      }                                                                   \
+      g.template load<4 + MU>(U, i);                                      \
+      su3_dagmul_half(uh, U, h);                                          \
+      if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                   \
+      else recon_##MU##_0(acc, uh, one);                                  \
     }                                                                     \
-    g.template load<4 + MU>(U, i);                                        \
-    su3_dagmul_half(uh, U, h);                                            \
-    if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \
-    else recon_##MU##_0(acc, uh, one);                                    \
   }
 
   QA_DIR(0)
@@ -127,6 +152,14 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
   QA_DIR(2)
   QA_DIR(3)
 #undef QA_DIR
+
+  // boundary sites under CLOV_POST defer the whole epilogue: store raw sum
+  if constexpr (KT == KT_INTERIOR && MODE == CLOV_POST) {
+    if (bnd) {
+      out.store(acc, i);
+      return;
+    }
+  }
 
   if constexpr (MODE == CLOV_POST) {
     R diag[2][6];
@@ -168,6 +201,114 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
     }
   }
 
+  out.store(acc, i);
+}
+
+// EXTERIOR: one owner thread per boundary site adds all ghost-hop
+// contributions and completes deferred epilogues (see DslashKT above).
+// Thread space: faces in (mu asc, edge 0,1) order over active dims.
+template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
+__global__ __launch_bounds__(256) void k_dslash_wilson_exterior(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
+    CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
+    SpinorAcc<Prec> x, GhostAcc<Prec> gh, long n_threads) {
+  using R = typename Prec::Real;
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= n_threads) return;
+  int mu = -1, edge = 0;
+  long f = tid;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    if (mu < 0 && gh.active(m)) {
+      if (f < 2 * gh.Fcb[m]) {
+        mu = m;
+        edge = f >= gh.Fcb[m];
+        if (edge) f -= gh.Fcb[m];
+      } else {
+        f -= 2 * gh.Fcb[m];
+      }
+    }
+  }
+  if (mu < 0) return;
+  int xc[4];
+  face_coords(xc, f, mu, edge ? d.X[mu] - 1 : 0, d, parity);
+  // ownership: the first (m, edge) face containing this site handles ALL
+  // of its ghost hops (avoids read-modify-write races on corner sites)
+  int key = 2 * mu + edge;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    if (gh.active(m)) {
+      if (xc[m] == 0 && 2 * m < key) return;
+      if (xc[m] == d.X[m] - 1 && 2 * m + 1 < key) return;
+    }
+  }
+  long i = cb_from_coords(xc, d);
+
+  cplx<R> acc[4][3];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+#pragma unroll
+    for (int c = 0; c < 3; ++c) acc[s][c] = {(R)0, (R)0};
+  cplx<R> h[2][3], uh[2][3], U[3][3];
+  const R one = (R)0.5;
+
+#define QA_EXT_DIR(MU)                                                    \
+  if (gh.active(MU)) {                                                    \
+    if (xc[MU] == d.X[MU] - 1) {                                          \
+      gh.load(h, MU, 1, ghost_idx(xc, MU, d));                            \
+      g.template load<MU>(U, i);                                          \
+      su3_mul_half(uh, U, h);                                             \
+      if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                   \
+      else recon_##MU##_1(acc, uh, one);                                  \
+    }                                                                     \
+    if (xc[MU] == 0) {                                                    \
+      gh.load(h, MU, 0, ghost_idx(xc, MU, d));                            \
+      g.template load<4 + MU>(U, i);                                      \
+      su3_dagmul_half(uh, U, h);                                          \
+      if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                   \
+      else recon_##MU##_0(acc, uh, one);                                  \
+    }                                                                     \
+  }
+
+  QA_EXT_DIR(0)
+  QA_EXT_DIR(1)
+  QA_EXT_DIR(2)
+  QA_EXT_DIR(3)
+#undef QA_EXT_DIR
+
+  cplx<R> prev[4][3];
+  out.load(prev, i);
+  if constexpr (MODE == CLOV_POST) {
+    // prev = raw interior hop sum; complete sum, then epilogue
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) acc[s][c] += prev[s][c];
+    R diag[2][6];
+    cplx<R> tri[2][15];
+    clov.load(diag, tri, parity, i);
+    cplx<R> tmp[4][3];
+    clover_mul(tmp, diag, tri, acc);
+    if constexpr (XPAY) {
+      cplx<R> xv[4][3];
+      x.load(xv, i);
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = xv[s][c] + a * tmp[s][c];
+    } else {
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = a * tmp[s][c];
+    }
+  } else {
+    // prev already carries the (affine) epilogue on the partial sum
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) acc[s][c] = prev[s][c] + a * acc[s][c];
+  }
   out.store(acc, i);
 }
 

@@ -25,28 +25,42 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   int blk = 256;
   int grid = (int)((c.Vcb + blk - 1) / blk);
   R a = (R)c.a;
+  long n_ext = 0;
+  for (int m = 0; m < 4; ++m)
+    if ((c.comm_mask >> m) & 1) n_ext += 2 * c.face_cb[m];
+  int grid_ext = (int)((n_ext + blk - 1) / blk);
 
-#define QA_LAUNCH(DAG, MODE, XPAY, COMMS)                                     \
-  hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY, COMMS>),  \
-                     dim3(grid), dim3(blk), 0, st, out, in, g, cl, d,         \
-                     c.parity, a, x, gh)
+#define QA_LAUNCH(DAG, MODE, XPAY, KT)                                        \
+  if (c.kt == 3)                                                              \
+    hipLaunchKernelGGL((k_dslash_wilson_exterior<Prec, RECON, DAG, MODE, XPAY>), \
+                       dim3(grid_ext), dim3(blk), 0, st, out, in, g, cl, d,   \
+                       c.parity, a, x, gh, n_ext);                            \
+  else                                                                        \
+    hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY, KT>),   \
+                       dim3(grid), dim3(blk), 0, st, out, in, g, cl, d,       \
+                       c.parity, a, x, gh)
 
-#define QA_DISPATCH(COMMS)                                                     \
+#define QA_DISPATCH(KT)                                                        \
   if (!c.dagger) {                                                             \
-    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(false, PLAIN, false, COMMS);     \
-    else if (c.mode == PLAIN) QA_LAUNCH(false, PLAIN, true, COMMS);            \
-    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(false, CLOV_POST, false, COMMS); \
-    else if (c.mode == CLOV_POST) QA_LAUNCH(false, CLOV_POST, true, COMMS);    \
-    else QA_LAUNCH(false, CLOV_X, true, COMMS);                                \
+    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(false, PLAIN, false, KT);        \
+    else if (c.mode == PLAIN) QA_LAUNCH(false, PLAIN, true, KT);               \
+    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(false, CLOV_POST, false, KT); \
+    else if (c.mode == CLOV_POST) QA_LAUNCH(false, CLOV_POST, true, KT);       \
+    else QA_LAUNCH(false, CLOV_X, true, KT);                                   \
   } else {                                                                     \
-    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(true, PLAIN, false, COMMS);      \
-    else if (c.mode == PLAIN) QA_LAUNCH(true, PLAIN, true, COMMS);             \
-    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(true, CLOV_POST, false, COMMS); \
-    else if (c.mode == CLOV_POST) QA_LAUNCH(true, CLOV_POST, true, COMMS);     \
-    else QA_LAUNCH(true, CLOV_X, true, COMMS);                                 \
+    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(true, PLAIN, false, KT);         \
+    else if (c.mode == PLAIN) QA_LAUNCH(true, PLAIN, true, KT);                \
+    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(true, CLOV_POST, false, KT); \
+    else if (c.mode == CLOV_POST) QA_LAUNCH(true, CLOV_POST, true, KT);        \
+    else QA_LAUNCH(true, CLOV_X, true, KT);                                    \
   }
 
-  if (c.comm_mask) { QA_DISPATCH(true) } else { QA_DISPATCH(false) }
+  switch (c.kt) {
+    case 0: { QA_DISPATCH(KT_LOCAL) } break;
+    case 1: { QA_DISPATCH(KT_FUSED) } break;
+    case 2: { QA_DISPATCH(KT_INTERIOR) } break;
+    case 3: { QA_DISPATCH(KT_LOCAL) } break;  // KT arg unused for exterior
+  }
 #undef QA_DISPATCH
 #undef QA_LAUNCH
 }

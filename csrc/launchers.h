@@ -55,6 +55,7 @@ struct DslashCall {
   const float *ghost_nrm[8];
   long face_cb[4];
   int comm_mask;
+  int kt;  // 0 local, 1 fused, 2 interior, 3 exterior (csrc/dslash_wilson.h)
 };
 
 void launch_dslash_wilson_double(const DslashCall &c, hipStream_t st);

@@ -39,6 +39,26 @@ def _load_ext():
     return _EXT
 
 
+_POLICY = None
+
+
+def dslash_policy() -> str:
+    """'overlap' (interior/exterior split around async RCCL transfers,
+    default) or 'fused' (blocking exchange, single ghost-aware kernel).
+    Env QUDA_AMD_DSLASH_POLICY overrides."""
+    global _POLICY
+    if _POLICY is None:
+        import os
+        _POLICY = os.environ.get("QUDA_AMD_DSLASH_POLICY", "overlap")
+    return _POLICY
+
+
+def set_dslash_policy(p: str) -> None:
+    global _POLICY
+    assert p in ("overlap", "fused")
+    _POLICY = p
+
+
 def hip_ext(required: bool = True):
     ext = _load_ext()
     if ext is None and required:
@@ -82,19 +102,36 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
         if mode != PLAIN:
             cl_t = clover.inv_data if clover_inverse else clover.data
         xf = x if x is not None else out
-        ghosts, nrms, face_cb = [], [], []
-        if mask:
-            from ..parallel.halo import get_spinor_halo
-            h = get_spinor_halo(geo, inp.precision, inp.device, mask)
-            h.pack(ext, inp, 1 - parity, bool(dagger))
+
+        def launch(kt, ghosts=[], nrms=[], face_cb=[]):
+            ext.dslash_wilson(
+                out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+                gauge.data, cl_t, xf.data, norm_or_empty(xf),
+                list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
+                bool(dagger), mode, xpay, float(a),
+                RECON_COMPS[gauge.reconstruct], ghosts, nrms, face_cb,
+                mask if kt else 0, kt)
+
+        if not mask:
+            launch(0)
+            return out
+        # comm-overlap policy (role of ref lib/dslash_policy.hpp): pack ->
+        # start RCCL transfers (their kernels run on NCCL's streams) ->
+        # INTERIOR on the compute stream overlaps them -> stream-waits on
+        # the transfers -> EXTERIOR adds ghost hops. "fused" = blocking.
+        from ..parallel.halo import get_spinor_halo
+        h = get_spinor_halo(geo, inp.precision, inp.device, mask)
+        h.pack(ext, inp, 1 - parity, bool(dagger))
+        ghosts, nrms, face_cb = h.ghost_args()
+        if dslash_policy() == "fused":
             h.exchange()
-            ghosts, nrms, face_cb = h.ghost_args()
-        ext.dslash_wilson(
-            out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
-            gauge.data, cl_t, xf.data, norm_or_empty(xf),
-            list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
-            bool(dagger), mode, xpay, float(a), RECON_COMPS[gauge.reconstruct],
-            ghosts, nrms, face_cb, mask)
+            launch(1, ghosts, nrms, face_cb)
+        else:
+            reqs = h.exchange_start()
+            launch(2, ghosts, nrms, face_cb)   # interior
+            for r in reqs:
+                r.wait()
+            launch(3, ghosts, nrms, face_cb)   # exterior
         return out
     # ---- oracle path ----
     u = gauge.to_complex()

@@ -43,9 +43,20 @@ def _travel_key(mu: int, travel: int, extra: int = 0):
 
 def exchange_tensors(sends: Dict[Key, torch.Tensor],
                      recvs: Dict[Key, torch.Tensor]) -> None:
-    """Exchange face tensors: recv[(mu,1)] <- +mu neighbor's send[(mu,0)],
-    recv[(mu,0)] <- -mu neighbor's send[(mu,1)]. Complex tensors are viewed
-    as real. Blocks until all transfers complete (stream-ordered on NCCL)."""
+    """Blocking exchange (see exchange_tensors_start)."""
+    for r in exchange_tensors_start(sends, recvs):
+        r.wait()
+
+
+def exchange_tensors_start(sends: Dict[Key, torch.Tensor],
+                           recvs: Dict[Key, torch.Tensor]) -> list:
+    """Start exchanging face tensors: recv[(mu,1)] <- +mu neighbor's
+    send[(mu,0)], recv[(mu,0)] <- -mu neighbor's send[(mu,1)]. Complex
+    tensors are viewed as real. Returns the request list; self-wraparound
+    copies run immediately (stream-ordered). wait() on each request before
+    touching recv buffers (on NCCL that makes the CURRENT stream wait — the
+    host does not block, so an interior kernel launched in between overlaps
+    with the transfer)."""
     ops = []  # (order_key, is_send, tensor, peer)
     my_rank = comms.comm_rank()
     local_copies = []
@@ -68,7 +79,7 @@ def exchange_tensors(sends: Dict[Key, torch.Tensor],
     for key, s in local_copies:
         recvs[key].copy_(s)
     if not ops:
-        return
+        return []
     ops.sort(key=lambda o: (o[0], not o[1]))  # sends before recvs per key
     reqs = []
     for _, is_send, t, peer in ops:
@@ -78,8 +89,7 @@ def exchange_tensors(sends: Dict[Key, torch.Tensor],
         else:
             assert tt.is_contiguous()
             reqs.append(dist.irecv(tt, peer))
-    for r in reqs:
-        r.wait()
+    return reqs
 
 
 # ---------------------------------------------------------------------------
@@ -137,9 +147,14 @@ class SpinorHalo:
                               geo.volume_cb, parity, mu, s01, edge, fcb)
 
     def exchange(self) -> None:
-        exchange_tensors(self.send, self.recv)
+        for r in self.exchange_start():
+            r.wait()
+
+    def exchange_start(self) -> list:
+        reqs = exchange_tensors_start(self.send, self.recv)
         if self.precision == "half":
-            exchange_tensors(self.send_nrm, self.recv_nrm)
+            reqs += exchange_tensors_start(self.send_nrm, self.recv_nrm)
+        return reqs
 
     def ghost_args(self):
         """(ghost[8], ghost_nrm[8], face_cb[4]) lists for ext.dslash_*."""
