@@ -105,12 +105,15 @@ static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
 
 static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
                           at::Tensor x_n, at::Tensor y, at::Tensor y_n,
-                          int64_t Vcb, int64_t sites) {
+                          int64_t Vcb, int64_t sites, double c2 = 0.0,
+                          double d2 = 0.0) {
   BlasCall c{};
   c.op = (int)op;
   c.prec = prec_of(x);
   c.a = a;
   c.b = b;
+  c.c = c2;
+  c.d = d2;
   c.x = field_of(x, x_n, Vcb);
   c.y = field_of(y, y_n, Vcb);
   c.sites = sites;
@@ -151,7 +154,10 @@ static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover) dslash");
   m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
-  m.def("blas_op", &blas_op, "fused blas/reduction");
+  m.def("blas_op", &blas_op, "fused blas/reduction",
+        py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
+        py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),
+        py::arg("sites"), py::arg("c2") = 0.0, py::arg("d2") = 0.0);
   m.def("convert", &convert, "precision conversion copy");
   m.def("clover_apply", &clover_apply, "clover site-matrix apply");
   m.attr("BLAS_AXPY") = (int)BLAS_AXPY;
@@ -164,4 +170,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.attr("BLAS_NORM2") = (int)BLAS_NORM2;
   m.attr("BLAS_REDOT") = (int)BLAS_REDOT;
   m.attr("BLAS_CDOT") = (int)BLAS_CDOT;
+  m.attr("BLAS_CAXPBY") = (int)BLAS_CAXPBY;
 }

@@ -116,6 +116,26 @@ __global__ __launch_bounds__(256) void k_caxpy(
   }
 }
 
+// y = (ar+i ai)*x + (br+i bi)*y
+template <typename Prec>
+__global__ __launch_bounds__(256) void k_caxpby(
+    typename Prec::Real ar, typename Prec::Real ai, SpinorAcc<Prec> x,
+    typename Prec::Real br, typename Prec::Real bi, SpinorAcc<Prec> y,
+    long sites) {
+  using R = typename Prec::Real;
+  cplx<R> a{ar, ai}, b{br, bi};
+  GRID_STRIDE(g, sites) {
+    cplx<R> xv[4][3], yv[4][3];
+    x.load_g(xv, g);
+    y.load_g(yv, g);
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) yv[s][c] = cfma(a, xv[s][c], b * yv[s][c]);
+    y.store_g(yv, g);
+  }
+}
+
 // y = x - y ; returns norm2(y)
 template <typename Prec>
 __global__ __launch_bounds__(256) void k_xmy_norm2(
@@ -245,6 +265,10 @@ static void blas_dispatch(const BlasCall &c, hipStream_t st) {
     case BLAS_CAXPY:
       hipLaunchKernelGGL((k_caxpy<Prec>), dim3(gr), dim3(BLK), 0, st, (R)c.a,
                          (R)c.b, x, y, n);
+      break;
+    case BLAS_CAXPBY:
+      hipLaunchKernelGGL((k_caxpby<Prec>), dim3(gr), dim3(BLK), 0, st, (R)c.a,
+                         (R)c.b, x, (R)c.c, (R)c.d, y, n);
       break;
     case BLAS_XMY_NORM2:
       hipLaunchKernelGGL((k_xmy_norm2<Prec>), dim3(gr), dim3(BLK), 0, st, x, y, n,

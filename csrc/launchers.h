@@ -19,12 +19,14 @@ enum BlasOp {
   BLAS_NORM2,
   BLAS_REDOT,
   BLAS_CDOT,
+  BLAS_CAXPBY,
 };
 
 struct BlasCall {
   int op;        // BlasOp
   int prec;      // 0 double, 1 single, 2 half
   double a, b;   // scalars (caxpy: a=re, b=im)
+  double c, d;   // second complex scalar (caxpby: b=(c,d))
   BlasField x, y;
   long sites;    // npar * Vcb
   double *result;  // device ptr for reductions (>=2 doubles for cdot)

@@ -176,6 +176,24 @@ def caxpy(a: complex, x: SpinorField, y: SpinorField) -> SpinorField:
     return y
 
 
+def caxpby(a: complex, x: SpinorField, b: complex, y: SpinorField) -> SpinorField:
+    """y = a*x + b*y (complex a, b; ref blas_core.cuh caxpby)."""
+    a, b = complex(a), complex(b)
+    if on_gpu(x, y):
+        ext = hip_ext()
+        ext.blas_op(ext.BLAS_CAXPBY, a.real, a.imag, x.data, norm_or_empty(x),
+                    y.data, norm_or_empty(y), x.volume_cb,
+                    x.n_parity * x.volume_cb, b.real, b.imag)
+        return y
+    _requant_any(y, a * _cvals(x) + b * _cvals(y))
+    return y
+
+
+def _requant_any(y: SpinorField, vals):
+    """Write complex values back into y at its precision."""
+    y.from_complex(vals.reshape(y.n_parity, y.volume_cb, 4, 3))
+
+
 def scal(a: float, x: SpinorField) -> SpinorField:
     if on_gpu(x):
         ext = hip_ext()

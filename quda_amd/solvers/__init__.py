@@ -1,3 +1,42 @@
-from .cg import cg_solve, SolverStats
+"""Solver suite (ref: lib/solver.cpp Solver::create factory + the
+inv_*_quda.cpp implementations)."""
 
-__all__ = ["cg_solve", "SolverStats"]
+from .bicgstab import bicgstab_solve, bicgstabl_solve
+from .ca import ca_cg_solve, ca_gcr_solve
+from .cg import SolverStats, cg_solve
+from .gcr import gcr_solve, mr_solve
+from .mre import ChronoForecaster
+from .multishift import multishift_cg_solve
+from .variants import (cg3_solve, cgne_solve, cgnr_solve, pcg_solve,
+                       sd_solve)
+
+SOLVERS = {
+    "cg": cg_solve,
+    "cgne": cgne_solve,
+    "cgnr": cgnr_solve,
+    "cg3": cg3_solve,
+    "sd": sd_solve,
+    "pcg": pcg_solve,
+    "bicgstab": bicgstab_solve,
+    "bicgstab-l": bicgstabl_solve,
+    "gcr": gcr_solve,
+    "mr": mr_solve,
+    "ca-cg": ca_cg_solve,
+    "ca-gcr": ca_gcr_solve,
+    "multishift-cg": multishift_cg_solve,
+}
+
+
+def create_solver(name: str):
+    """Solver factory (ref: lib/solver.cpp:47 Solver::create)."""
+    try:
+        return SOLVERS[name]
+    except KeyError:
+        raise ValueError(f"unknown solver '{name}'; have {sorted(SOLVERS)}")
+
+
+__all__ = ["cg_solve", "SolverStats", "bicgstab_solve", "bicgstabl_solve",
+           "gcr_solve", "mr_solve", "ca_cg_solve", "ca_gcr_solve",
+           "multishift_cg_solve", "cgne_solve", "cgnr_solve", "cg3_solve",
+           "sd_solve", "pcg_solve", "ChronoForecaster", "create_solver",
+           "SOLVERS"]
