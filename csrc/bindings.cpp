@@ -44,7 +44,7 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
                           std::vector<at::Tensor> ghost,
                           std::vector<at::Tensor> ghost_nrm,
                           std::vector<int64_t> face_cb, int64_t comm_mask,
-                          int64_t kt) {
+                          int64_t kt, double b_re, double b_im) {
   TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
   DslashCall c{};
   c.comm_mask = (int)comm_mask;
@@ -70,6 +70,8 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
   c.mode = (int)mode;
   c.xpay = xpay;
   c.a = a;
+  c.b_re = b_re;
+  c.b_im = b_im;
   c.recon = (int)recon;
   switch (prec_of(out)) {
     case 0: launch_dslash_wilson_double(c, stream()); break;
@@ -151,6 +153,20 @@ static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
   check_launch("clover_apply");
 }
 
+static void twist_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                        at::Tensor in_n, double b_re, double b_im,
+                        int64_t Vcb, int64_t sites) {
+  TwistApplyCall c{};
+  c.out = field_of(out, out_n, Vcb);
+  c.in = field_of(in, in_n, Vcb);
+  c.b_re = b_re;
+  c.b_im = b_im;
+  c.sites = sites;
+  c.prec = prec_of(out);
+  launch_twist_apply(c, stream());
+  check_launch("twist_apply");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover) dslash");
   m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
@@ -160,6 +176,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("sites"), py::arg("c2") = 0.0, py::arg("d2") = 0.0);
   m.def("convert", &convert, "precision conversion copy");
   m.def("clover_apply", &clover_apply, "clover site-matrix apply");
+  m.def("twist_apply", &twist_apply, "twisted-mass T(b) apply");
   m.attr("BLAS_AXPY") = (int)BLAS_AXPY;
   m.attr("BLAS_AXPY_NORM2") = (int)BLAS_AXPY_NORM2;
   m.attr("BLAS_XPAY") = (int)BLAS_XPAY;

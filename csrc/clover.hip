@@ -20,3 +20,36 @@ void launch_clover_apply(const CloverApplyCall &c, hipStream_t st) {
     case 2: clover_apply_t<PrecHalf>(c, st); break;
   }
 }
+
+// standalone twist apply: out = T(b) in = b_re in + i b_im g5 in
+template <typename Prec>
+__global__ __launch_bounds__(256) void k_twist_apply(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, typename Prec::Real br,
+    typename Prec::Real bi, long sites) {
+  using R = typename Prec::Real;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= sites) return;
+  cplx<R> v[4][3];
+  in.load_g(v, g);
+  twist_mul(v, br, bi);
+  out.store_g(v, g);
+}
+
+template <typename Prec>
+static void twist_apply_t(const TwistApplyCall &c, hipStream_t st) {
+  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.out.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
+  int blk = 256;
+  int grid = (int)((c.sites + blk - 1) / blk);
+  using R = typename Prec::Real;
+  hipLaunchKernelGGL((k_twist_apply<Prec>), dim3(grid), dim3(blk), 0, st, out,
+                     in, (R)c.b_re, (R)c.b_im, c.sites);
+}
+
+void launch_twist_apply(const TwistApplyCall &c, hipStream_t st) {
+  switch (c.prec) {
+    case 0: twist_apply_t<PrecDouble>(c, st); break;
+    case 1: twist_apply_t<PrecSingle>(c, st); break;
+    case 2: twist_apply_t<PrecHalf>(c, st); break;
+  }
+}

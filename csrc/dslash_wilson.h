@@ -16,7 +16,29 @@
 #include "generated/proj.h"
 #include "halo.h"
 
-enum CloverMode { PLAIN = 0, CLOV_POST = 1, CLOV_X = 2 };
+// epilogue modes (b = (b_re, b_im) twist scalar, T(b) v = b_re v + i b_im g5 v):
+//   PLAIN      : out = [x +] a * (D in)
+//   CLOV_POST  : out = [x +] a * (A (D in))
+//   CLOV_X     : out = A x + a * (D in)
+//   TWIST_POST : out = [x +] a * (T(b) (D in))   (twisted-mass PC: T = A^-1)
+//   TWIST_X    : out = T(b) x + a * (D in)       (twisted-mass full op)
+//   CLOVTW_X   : out = (A + i b_im g5) x + a * (D in)  (twisted-clover full)
+enum CloverMode { PLAIN = 0, CLOV_POST = 1, CLOV_X = 2, TWIST_POST = 3,
+                  TWIST_X = 4, CLOVTW_X = 5 };
+
+// v <- b_re v + i b_im g5 v (DeGrand-Rossi g5 = diag(1,1,-1,-1))
+template <typename R>
+__device__ __forceinline__ void twist_mul(cplx<R> v[4][3], R br, R bi) {
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    R bs = (s < 2) ? bi : -bi;
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      cplx<R> t = v[s][c];
+      v[s][c] = {br * t.re - bs * t.im, br * t.im + bs * t.re};
+    }
+  }
+}
 
 // kernel roles for the comm-overlap split (role of the reference's
 // INTERIOR/EXTERIOR kernel_type, include/dslash_helper.cuh):
@@ -91,7 +113,8 @@ template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY, int KT = KT_L
 __global__ __launch_bounds__(256) void k_dslash_wilson(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
-    SpinorAcc<Prec> x, GhostAcc<Prec> gh) {
+    SpinorAcc<Prec> x, GhostAcc<Prec> gh, typename Prec::Real br,
+    typename Prec::Real bi) {
   using R = typename Prec::Real;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= d.Vcb) return;
@@ -172,20 +195,34 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
 #pragma unroll
       for (int c = 0; c < 3; ++c) acc[s][c] = tmp[s][c];
   }
+  if constexpr (MODE == TWIST_POST) twist_mul(acc, br, bi);
 
-  if constexpr (XPAY || MODE == CLOV_X) {
+  if constexpr (XPAY || MODE == CLOV_X || MODE == TWIST_X || MODE == CLOVTW_X) {
     cplx<R> xv[4][3];
     x.load(xv, i);
-    if constexpr (MODE == CLOV_X) {
+    if constexpr (MODE == CLOV_X || MODE == CLOVTW_X) {
       R diag[2][6];
       cplx<R> tri[2][15];
       clov.load(diag, tri, parity, i);
       cplx<R> Ax[4][3];
       clover_mul(Ax, diag, tri, xv);
+      if constexpr (MODE == CLOVTW_X) {
+        twist_mul(xv, (R)0, bi);  // i b_im g5 x
+#pragma unroll
+        for (int s = 0; s < 4; ++s)
+#pragma unroll
+          for (int c = 0; c < 3; ++c) Ax[s][c] += xv[s][c];
+      }
 #pragma unroll
       for (int s = 0; s < 4; ++s)
 #pragma unroll
         for (int c = 0; c < 3; ++c) acc[s][c] = Ax[s][c] + a * acc[s][c];
+    } else if constexpr (MODE == TWIST_X) {
+      twist_mul(xv, br, bi);
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = xv[s][c] + a * acc[s][c];
     } else {
 #pragma unroll
       for (int s = 0; s < 4; ++s)
@@ -211,7 +248,8 @@ template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
 __global__ __launch_bounds__(256) void k_dslash_wilson_exterior(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
-    SpinorAcc<Prec> x, GhostAcc<Prec> gh, long n_threads) {
+    SpinorAcc<Prec> x, GhostAcc<Prec> gh, long n_threads,
+    typename Prec::Real br, typename Prec::Real bi) {
   using R = typename Prec::Real;
   long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (tid >= n_threads) return;
@@ -303,7 +341,10 @@ __global__ __launch_bounds__(256) void k_dslash_wilson_exterior(
         for (int c = 0; c < 3; ++c) acc[s][c] = a * tmp[s][c];
     }
   } else {
-    // prev already carries the (affine) epilogue on the partial sum
+    // prev already carries the (affine) epilogue on the partial sum;
+    // T(b) is linear and site-diagonal, so TWIST_POST twists the ghost
+    // contribution alone
+    if constexpr (MODE == TWIST_POST) twist_mul(acc, br, bi);
 #pragma unroll
     for (int s = 0; s < 4; ++s)
 #pragma unroll

@@ -25,6 +25,7 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   int blk = 256;
   int grid = (int)((c.Vcb + blk - 1) / blk);
   R a = (R)c.a;
+  R br = (R)c.b_re, bi = (R)c.b_im;
   long n_ext = 0;
   for (int m = 0; m < 4; ++m)
     if ((c.comm_mask >> m) & 1) n_ext += 2 * c.face_cb[m];
@@ -34,26 +35,33 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   if (c.kt == 3)                                                              \
     hipLaunchKernelGGL((k_dslash_wilson_exterior<Prec, RECON, DAG, MODE, XPAY>), \
                        dim3(grid_ext), dim3(blk), 0, st, out, in, g, cl, d,   \
-                       c.parity, a, x, gh, n_ext);                            \
+                       c.parity, a, x, gh, n_ext, br, bi);                    \
   else                                                                        \
     hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY, KT>),   \
                        dim3(grid), dim3(blk), 0, st, out, in, g, cl, d,       \
-                       c.parity, a, x, gh)
+                       c.parity, a, x, gh, br, bi)
+
+#define QA_MODES(DAG, KT)                                                      \
+  switch (c.mode) {                                                            \
+    case PLAIN:                                                                \
+      if (c.xpay) QA_LAUNCH(DAG, PLAIN, true, KT);                             \
+      else QA_LAUNCH(DAG, PLAIN, false, KT);                                   \
+      break;                                                                   \
+    case CLOV_POST:                                                            \
+      if (c.xpay) QA_LAUNCH(DAG, CLOV_POST, true, KT);                         \
+      else QA_LAUNCH(DAG, CLOV_POST, false, KT);                               \
+      break;                                                                   \
+    case CLOV_X: QA_LAUNCH(DAG, CLOV_X, true, KT); break;                      \
+    case TWIST_POST:                                                           \
+      if (c.xpay) QA_LAUNCH(DAG, TWIST_POST, true, KT);                        \
+      else QA_LAUNCH(DAG, TWIST_POST, false, KT);                              \
+      break;                                                                   \
+    case TWIST_X: QA_LAUNCH(DAG, TWIST_X, true, KT); break;                    \
+    case CLOVTW_X: QA_LAUNCH(DAG, CLOVTW_X, true, KT); break;                  \
+  }
 
 #define QA_DISPATCH(KT)                                                        \
-  if (!c.dagger) {                                                             \
-    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(false, PLAIN, false, KT);        \
-    else if (c.mode == PLAIN) QA_LAUNCH(false, PLAIN, true, KT);               \
-    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(false, CLOV_POST, false, KT); \
-    else if (c.mode == CLOV_POST) QA_LAUNCH(false, CLOV_POST, true, KT);       \
-    else QA_LAUNCH(false, CLOV_X, true, KT);                                   \
-  } else {                                                                     \
-    if (c.mode == PLAIN && !c.xpay) QA_LAUNCH(true, PLAIN, false, KT);         \
-    else if (c.mode == PLAIN) QA_LAUNCH(true, PLAIN, true, KT);                \
-    else if (c.mode == CLOV_POST && !c.xpay) QA_LAUNCH(true, CLOV_POST, false, KT); \
-    else if (c.mode == CLOV_POST) QA_LAUNCH(true, CLOV_POST, true, KT);        \
-    else QA_LAUNCH(true, CLOV_X, true, KT);                                    \
-  }
+  if (!c.dagger) { QA_MODES(false, KT) } else { QA_MODES(true, KT) }
 
   switch (c.kt) {
     case 0: { QA_DISPATCH(KT_LOCAL) } break;
@@ -62,6 +70,7 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
     case 3: { QA_DISPATCH(KT_LOCAL) } break;  // KT arg unused for exterior
   }
 #undef QA_DISPATCH
+#undef QA_MODES
 #undef QA_LAUNCH
 }
 

@@ -50,6 +50,7 @@ struct DslashCall {
   int mode;   // 0 PLAIN, 1 CLOV_POST, 2 CLOV_X
   bool xpay;
   double a;
+  double b_re, b_im;  // twist scalar for TWIST_*/CLOVTW_* modes
   int recon;  // 18 or 12
   // halo: ghost recv buffers per [2*mu+dir] (dir 1 = from +mu neighbor);
   // null when mask bit mu unset. comm_mask==0 => pure-local kernel.
@@ -91,3 +92,11 @@ struct CloverApplyCall {
   int prec;
 };
 void launch_clover_apply(const CloverApplyCall &c, hipStream_t st);
+
+struct TwistApplyCall {
+  BlasField out, in;
+  double b_re, b_im;
+  long sites;
+  int prec;
+};
+void launch_twist_apply(const TwistApplyCall &c, hipStream_t st);

@@ -1,4 +1,7 @@
-from .dirac import Dirac, DiracWilson, DiracWilsonPC, DiracClover, DiracCloverPC
+from .dirac import (Dirac, DiracClover, DiracCloverPC, DiracTwistedClover,
+                    DiracTwistedMass, DiracTwistedMassPC, DiracWilson,
+                    DiracWilsonPC)
 
 __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
-           "DiracCloverPC"]
+           "DiracCloverPC", "DiracTwistedMass", "DiracTwistedMassPC",
+           "DiracTwistedClover"]

@@ -32,9 +32,9 @@ class Dirac:
         self._tmps = {}
 
     def dslash(self, out, inp, parity, dagger=False, mode=PLAIN, a=1.0,
-               x=None, clover=None, clover_inverse=False):
+               x=None, clover=None, clover_inverse=False, twist=(0.0, 0.0)):
         return dslash_wilson(out, inp, self.gauge, parity, dagger, mode, a, x,
-                             clover, clover_inverse)
+                             clover, clover_inverse, twist)
 
     def tmp(self, name: str, like: SpinorField, n_parity=1) -> SpinorField:
         key = (name, like.precision, str(like.device), n_parity)
@@ -183,3 +183,136 @@ class DiracCloverPC(Dirac, _CloverMixin):
 
     def flops_per_site(self) -> int:
         return 1320 + 504
+
+
+class DiracTwistedMass(Dirac):
+    """Degenerate twisted-mass Wilson, twisted basis
+    (ref: lib/dirac_twisted_mass.cpp): M = T(1, 2 kappa mu) - kappa D with
+    T(b) = b_re + i b_im g5. One fused TWIST_X launch per parity."""
+
+    def __init__(self, gauge: GaugeField, kappa: float, mu: float):
+        super().__init__(gauge, kappa)
+        self.mu = float(mu)
+
+    @property
+    def eps(self) -> float:
+        return 2.0 * self.kappa * self.mu
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        sgn = -1.0 if dagger else 1.0
+        from ..ops.dispatch import TWIST_X
+        for p in (0, 1):
+            self.dslash(out.parity_view(p), inp.parity_view(1 - p), p, dagger,
+                        mode=TWIST_X, a=-self.kappa, x=inp.parity_view(p),
+                        twist=(1.0, sgn * self.eps))
+        return out
+
+    def MdagM(self, out, inp, tmp):
+        self.M(tmp, inp, dagger=False)
+        self.M(out, tmp, dagger=True)
+        return out
+
+    def flops_per_site(self) -> int:
+        return 1320 + 48
+
+
+class DiracTwistedMassPC(Dirac):
+    """Symmetric even-odd preconditioned twisted mass
+    (ref: lib/dirac_twisted_mass.cpp DiracTwistedMassPC):
+    M_pc = 1 - kappa^2 T^-1 D_eo T^-1 D_oe, T^-1 = T(c, -eps c),
+    c = 1/(1+eps^2). Two fused TWIST_POST launches."""
+
+    def __init__(self, gauge: GaugeField, kappa: float, mu: float):
+        super().__init__(gauge, kappa)
+        self.mu = float(mu)
+
+    @property
+    def eps(self) -> float:
+        return 2.0 * self.kappa * self.mu
+
+    def _tinv(self, dagger: bool = False):
+        c = 1.0 / (1.0 + self.eps ** 2)
+        s = 1.0 if dagger else -1.0
+        return (c, s * self.eps * c)
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import TWIST_POST, apply_twist_field
+        k2 = -self.kappa ** 2
+        if not dagger:
+            t = self.tmp("pc_odd", inp)
+            self.dslash(t, inp, 1, mode=TWIST_POST, twist=self._tinv())
+            self.dslash(out, t, 0, mode=TWIST_POST, a=k2, x=inp,
+                        twist=self._tinv())
+        else:
+            # M^dag = 1 - k^2 D^dag T^-dag D^dag T^-dag
+            t0 = self.tmp("pc_even", inp)
+            t1 = self.tmp("pc_odd", inp)
+            apply_twist_field(t0, inp, *self._tinv(dagger=True))
+            self.dslash(t1, t0, 1, dagger=True, mode=TWIST_POST,
+                        twist=self._tinv(dagger=True))
+            self.dslash(out, t1, 0, dagger=True, a=k2, x=inp)
+        return out
+
+    def MdagM(self, out, inp, tmp):
+        self.M(tmp, inp, dagger=False)
+        self.M(out, tmp, dagger=True)
+        return out
+
+    def prepare(self, b_full: SpinorField) -> SpinorField:
+        """b' = T_ee^-1 (b_e + kappa D_eo T_oo^-1 b_o)."""
+        from ..ops.dispatch import apply_twist_field
+        t = self.new_spinor(b_full.precision)
+        be = self.new_spinor(b_full.precision)
+        apply_twist_field(t, b_full.parity_view(1), *self._tinv())
+        self.dslash(be, t, 0, a=self.kappa, x=b_full.parity_view(0))
+        apply_twist_field(t, be, *self._tinv())
+        from ..ops import blas
+        blas.copy(be, t)
+        return be
+
+    def reconstruct(self, x_full: SpinorField, x_e: SpinorField,
+                    b_full: SpinorField):
+        """x_o = T_oo^-1 (b_o + kappa D_oe x_e)."""
+        from ..ops import blas
+        from ..ops.dispatch import apply_twist_field
+        blas.copy(x_full.parity_view(0), x_e)
+        t = self.tmp("pc_odd", x_full)
+        self.dslash(t, x_e, 1, a=self.kappa, x=b_full.parity_view(1))
+        apply_twist_field(x_full.parity_view(1), t, *self._tinv())
+        return x_full
+
+    def flops_per_site(self) -> int:
+        return 1320 + 48
+
+
+class DiracTwistedClover(Dirac, _CloverMixin):
+    """Full-lattice twisted clover (ref: lib/dirac_twisted_clover.cpp):
+    M = (A + i 2 kappa mu g5) - kappa D — one fused CLOVTW_X launch per
+    parity."""
+
+    def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float,
+                 mu: float):
+        super().__init__(gauge, kappa)
+        self.clover = clover
+        self.mu = float(mu)
+
+    @property
+    def eps(self) -> float:
+        return 2.0 * self.kappa * self.mu
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import CLOVTW_X
+        sgn = -1.0 if dagger else 1.0
+        for p in (0, 1):
+            self.dslash(out.parity_view(p), inp.parity_view(1 - p), p, dagger,
+                        mode=CLOVTW_X, a=-self.kappa, x=inp.parity_view(p),
+                        clover=self.clover, twist=(1.0, sgn * self.eps))
+        return out
+
+    def MdagM(self, out, inp, tmp):
+        self.M(tmp, inp, dagger=False)
+        self.M(out, tmp, dagger=True)
+        return out
+
+    def flops_per_site(self) -> int:
+        return 1320 + 504 + 48

@@ -22,6 +22,7 @@ from ..fields.spinor import SpinorField
 from . import reference as ref
 
 PLAIN, CLOV_POST, CLOV_X = 0, 1, 2
+TWIST_POST, TWIST_X, CLOVTW_X = 3, 4, 5
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
@@ -89,9 +90,12 @@ from ..fields.gauge import RECON_COMPS
 def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                   parity: int, dagger: bool = False, mode: int = PLAIN,
                   a: float = 1.0, x: Optional[SpinorField] = None,
-                  clover=None, clover_inverse: bool = False):
-    """Apply the fused Wilson(-clover) stencil; out at `parity`, in at the
-    opposite parity; x (same parity as out) enables the xpay term."""
+                  clover=None, clover_inverse: bool = False,
+                  twist=(0.0, 0.0)):
+    """Apply the fused Wilson(-clover/-twisted) stencil; out at `parity`,
+    in at the opposite parity; x (same parity as out) enables the xpay
+    term; twist=(b_re,b_im) feeds the TWIST_*/CLOVTW_* epilogues
+    (T(b) v = b_re v + i b_im g5 v)."""
     from ..parallel import comms
     geo = out.geo
     xpay = x is not None
@@ -99,7 +103,7 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     if on_gpu(out, inp):
         ext = hip_ext()
         cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)
-        if mode != PLAIN:
+        if mode in (CLOV_POST, CLOV_X, CLOVTW_X):
             cl_t = clover.inv_data if clover_inverse else clover.data
         xf = x if x is not None else out
 
@@ -110,7 +114,7 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                 list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
                 bool(dagger), mode, xpay, float(a),
                 RECON_COMPS[gauge.reconstruct], ghosts, nrms, face_cb,
-                mask if kt else 0, kt)
+                mask if kt else 0, kt, float(twist[0]), float(twist[1]))
 
         if not mask:
             launch(0)
@@ -149,14 +153,35 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     if mode == CLOV_POST:
         A = clover.to_complex(inverse=clover_inverse)[parity]
         res = ref.apply_clover(A, res)
+    if mode == TWIST_POST:
+        res = ref.apply_twist(res, twist[0], twist[1])
     if mode == CLOV_X:
         A = clover.to_complex(inverse=clover_inverse)[parity]
         res = ref.apply_clover(A, x.to_complex()[0]) + a * res
+    elif mode == TWIST_X:
+        res = ref.apply_twist(x.to_complex()[0], twist[0], twist[1]) + a * res
+    elif mode == CLOVTW_X:
+        A = clover.to_complex(inverse=clover_inverse)[parity]
+        xv = x.to_complex()[0]
+        res = (ref.apply_clover(A, xv)
+               + ref.apply_twist(xv, 0.0, twist[1]) + a * res)
     elif xpay:
         res = x.to_complex()[0] + a * res
     else:
         res = a * res
     out.from_complex(res.unsqueeze(0))
+    return out
+
+
+def apply_twist_field(out: SpinorField, inp: SpinorField, br: float, bi: float):
+    """out = T(b) in = br*in + i*bi*g5*in (site-local)."""
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        ext.twist_apply(out.data, norm_or_empty(out), inp.data,
+                        norm_or_empty(inp), float(br), float(bi),
+                        out.geo.volume_cb, out.n_parity * out.geo.volume_cb)
+        return out
+    out.from_complex(ref.apply_twist(inp.to_complex(), br, bi))
     return out
 
 

@@ -89,6 +89,14 @@ def mat_wilson(u, psi_full, geo, kappa: float, dagger: bool = False):
     return psi_full - kappa * dslash_wilson_full(u, psi_full, geo, dagger)
 
 
+def apply_twist(psi: torch.Tensor, br: float, bi: float) -> torch.Tensor:
+    """T(b) psi = br psi + i bi g5 psi (g5 = diag(1,1,-1,-1))."""
+    g5 = torch.tensor([1.0, 1.0, -1.0, -1.0], dtype=psi.real.dtype,
+                      device=psi.device)
+    coef = br + 1j * bi * g5
+    return psi * coef.view(*([1] * (psi.dim() - 2)), 4, 1)
+
+
 def apply_gamma5(psi: torch.Tensor) -> torch.Tensor:
     g5 = torch.tensor(np.diag(GAMMA5).real, dtype=psi.real.dtype,
                       device=psi.device)
