@@ -97,7 +97,8 @@ static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
   c.s01 = (int)s01;
   c.edge = (int)edge;
   c.Fcb = Fcb;
-  switch (prec_of(in)) {
+  c.prec = prec_of(in);
+  switch (c.prec) {
     case 0: launch_pack_face_double(c, stream()); break;
     case 1: launch_pack_face_single(c, stream()); break;
     case 2: launch_pack_face_half(c, stream()); break;
@@ -105,10 +106,68 @@ static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
   check_launch("pack_face");
 }
 
+static void pack_face_stag(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
+                           at::Tensor in_n, std::vector<int64_t> dims,
+                           int64_t parity_offset, int64_t Vcb, int64_t parity,
+                           int64_t mu, int64_t edge, int64_t Fcb) {
+  PackCall c{};
+  c.in = field_of(in, in_n, Vcb);
+  c.dst = dst.data_ptr();
+  c.dst_nrm = (float *)ptr_or_null(dst_nrm);
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.mu = (int)mu;
+  c.edge = (int)edge;
+  c.Fcb = Fcb;
+  c.prec = prec_of(in);
+  launch_pack_face_stag(c, stream());
+  check_launch("pack_face_stag");
+}
+
+static void dslash_staggered(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                             at::Tensor in_n, at::Tensor gauge, at::Tensor x,
+                             at::Tensor x_n, std::vector<int64_t> dims,
+                             int64_t parity_offset, int64_t Vcb,
+                             int64_t parity, bool xpay, double a, double b,
+                             int64_t recon, std::vector<at::Tensor> ghost,
+                             std::vector<at::Tensor> ghost_nrm,
+                             std::vector<int64_t> face_cb, int64_t comm_mask,
+                             int64_t kt) {
+  TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
+  StagDslashCall c{};
+  c.out = field_of(out, out_n, Vcb);
+  c.in = field_of(in, in_n, Vcb);
+  c.x = field_of(x, x_n, Vcb);
+  c.gauge = gauge.data_ptr();
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.xpay = xpay;
+  c.a = a;
+  c.b = b;
+  c.recon = (int)recon;
+  c.comm_mask = (int)comm_mask;
+  c.kt = (int)kt;
+  c.prec = prec_of(out);
+  if (comm_mask) {
+    TORCH_CHECK(ghost.size() == 8 && ghost_nrm.size() == 8 && face_cb.size() == 4);
+    for (int k = 0; k < 8; ++k) {
+      c.ghost[k] = ptr_or_null(ghost[k]);
+      c.ghost_nrm[k] = (const float *)ptr_or_null(ghost_nrm[k]);
+    }
+    for (int k = 0; k < 4; ++k) c.face_cb[k] = face_cb[k];
+  }
+  launch_dslash_staggered(c, stream());
+  check_launch("dslash_staggered");
+}
+
 static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
                           at::Tensor x_n, at::Tensor y, at::Tensor y_n,
                           int64_t Vcb, int64_t sites, double c2 = 0.0,
-                          double d2 = 0.0) {
+                          double d2 = 0.0, int64_t ncomp = 24) {
   BlasCall c{};
   c.op = (int)op;
   c.prec = prec_of(x);
@@ -116,6 +175,7 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
   c.b = b;
   c.c = c2;
   c.d = d2;
+  c.ncomp = (int)ncomp;
   c.x = field_of(x, x_n, Vcb);
   c.y = field_of(y, y_n, Vcb);
   c.sites = sites;
@@ -132,9 +192,11 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
 }
 
 static void convert(at::Tensor dst, at::Tensor dst_n, at::Tensor src,
-                    at::Tensor src_n, int64_t Vcb, int64_t sites) {
+                    at::Tensor src_n, int64_t Vcb, int64_t sites,
+                    int64_t ncomp = 24) {
   launch_convert(field_of(dst, dst_n, Vcb), prec_of(dst),
-                 field_of(src, src_n, Vcb), prec_of(src), sites, stream());
+                 field_of(src, src_n, Vcb), prec_of(src), sites, (int)ncomp,
+                 stream());
   check_launch("convert");
 }
 
@@ -170,11 +232,16 @@ static void twist_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover) dslash");
   m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
+  m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
+  m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
   m.def("blas_op", &blas_op, "fused blas/reduction",
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
         py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),
-        py::arg("sites"), py::arg("c2") = 0.0, py::arg("d2") = 0.0);
-  m.def("convert", &convert, "precision conversion copy");
+        py::arg("sites"), py::arg("c2") = 0.0, py::arg("d2") = 0.0,
+        py::arg("ncomp") = 24);
+  m.def("convert", &convert, "precision conversion copy", py::arg("dst"),
+        py::arg("dst_n"), py::arg("src"), py::arg("src_n"), py::arg("Vcb"),
+        py::arg("sites"), py::arg("ncomp") = 24);
   m.def("clover_apply", &clover_apply, "clover site-matrix apply");
   m.def("twist_apply", &twist_apply, "twisted-mass T(b) apply");
   m.attr("BLAS_AXPY") = (int)BLAS_AXPY;

@@ -1,7 +1,9 @@
 // Fused BLAS + reduction kernels (role of reference lib/blas_quda.cu /
 // lib/reduce_quda.cu functors, kernels/blas_core.cuh + reduce_core.cuh).
-// Site-structured so HALF (per-site norm) precision composes with every op;
-// arithmetic always in Real (float for half), reductions accumulate double.
+// Generic over the site layout via the accessor (Wilson 24-real sites and
+// staggered 6-real sites share every functor); HALF (per-site norm)
+// precision composes with every op; arithmetic always in Real (float for
+// half), reductions accumulate double.
 // Block 256 (4 waves), grid-stride; block-reduce -> one f64 atomic per block.
 #include "common.h"
 
@@ -28,195 +30,167 @@ __device__ __forceinline__ void block_atomic_add(double v, double *out) {
   }
 }
 
-// ---------------------------------------------------------------------------
-// site-wise blas: each functor consumes/produces [4][3] complex site values
-// ---------------------------------------------------------------------------
-template <typename Prec>
-using Site = cplx<typename Prec::Real>[4][3];
-
 #define GRID_STRIDE(g, n)                                                     \
   for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < (n);         \
        g += (long)gridDim.x * blockDim.x)
 
 // y = a*x + y ; optional norm2(y) accumulation
-template <typename Prec, bool NORM2>
+template <typename A, bool NORM2>
 __global__ __launch_bounds__(256) void k_axpy(
-    typename Prec::Real a, SpinorAcc<Prec> x, SpinorAcc<Prec> y, long sites,
-    double *result) {
-  using R = typename Prec::Real;
+    typename A::R a, A x, A y, long sites, double *result) {
+  using R = typename A::R;
   double acc = 0.0;
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) {
-        yv[s][c] = yv[s][c] + a * xv[s][c];
-        if constexpr (NORM2)
-          acc += (double)yv[s][c].re * yv[s][c].re + (double)yv[s][c].im * yv[s][c].im;
-      }
-    y.store_g(yv, g);
+    for (int k = 0; k < A::NCPLX; ++k) {
+      yv[k] = yv[k] + a * xv[k];
+      if constexpr (NORM2)
+        acc += (double)yv[k].re * yv[k].re + (double)yv[k].im * yv[k].im;
+    }
+    y.store_v(yv, g);
   }
   if constexpr (NORM2) block_atomic_add(acc, result);
 }
 
 // y = x + a*y
-template <typename Prec>
+template <typename A>
 __global__ __launch_bounds__(256) void k_xpay(
-    SpinorAcc<Prec> x, typename Prec::Real a, SpinorAcc<Prec> y, long sites) {
-  using R = typename Prec::Real;
+    A x, typename A::R a, A y, long sites) {
+  using R = typename A::R;
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) yv[s][c] = xv[s][c] + a * yv[s][c];
-    y.store_g(yv, g);
+    for (int k = 0; k < A::NCPLX; ++k) yv[k] = xv[k] + a * yv[k];
+    y.store_v(yv, g);
   }
 }
 
 // y = a*x + b*y
-template <typename Prec>
+template <typename A>
 __global__ __launch_bounds__(256) void k_axpby(
-    typename Prec::Real a, SpinorAcc<Prec> x, typename Prec::Real b,
-    SpinorAcc<Prec> y, long sites) {
-  using R = typename Prec::Real;
+    typename A::R a, A x, typename A::R b, A y, long sites) {
+  using R = typename A::R;
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) yv[s][c] = a * xv[s][c] + b * yv[s][c];
-    y.store_g(yv, g);
+    for (int k = 0; k < A::NCPLX; ++k) yv[k] = a * xv[k] + b * yv[k];
+    y.store_v(yv, g);
   }
 }
 
 // y += (ar + i ai) * x
-template <typename Prec>
+template <typename A>
 __global__ __launch_bounds__(256) void k_caxpy(
-    typename Prec::Real ar, typename Prec::Real ai, SpinorAcc<Prec> x,
-    SpinorAcc<Prec> y, long sites) {
-  using R = typename Prec::Real;
+    typename A::R ar, typename A::R ai, A x, A y, long sites) {
+  using R = typename A::R;
   cplx<R> a{ar, ai};
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) yv[s][c] = cfma(a, xv[s][c], yv[s][c]);
-    y.store_g(yv, g);
+    for (int k = 0; k < A::NCPLX; ++k) yv[k] = cfma(a, xv[k], yv[k]);
+    y.store_v(yv, g);
   }
 }
 
 // y = (ar+i ai)*x + (br+i bi)*y
-template <typename Prec>
+template <typename A>
 __global__ __launch_bounds__(256) void k_caxpby(
-    typename Prec::Real ar, typename Prec::Real ai, SpinorAcc<Prec> x,
-    typename Prec::Real br, typename Prec::Real bi, SpinorAcc<Prec> y,
-    long sites) {
-  using R = typename Prec::Real;
+    typename A::R ar, typename A::R ai, A x, typename A::R br,
+    typename A::R bi, A y, long sites) {
+  using R = typename A::R;
   cplx<R> a{ar, ai}, b{br, bi};
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) yv[s][c] = cfma(a, xv[s][c], b * yv[s][c]);
-    y.store_g(yv, g);
+    for (int k = 0; k < A::NCPLX; ++k) yv[k] = cfma(a, xv[k], b * yv[k]);
+    y.store_v(yv, g);
   }
 }
 
 // y = x - y ; returns norm2(y)
-template <typename Prec>
+template <typename A>
 __global__ __launch_bounds__(256) void k_xmy_norm2(
-    SpinorAcc<Prec> x, SpinorAcc<Prec> y, long sites, double *result) {
-  using R = typename Prec::Real;
+    A x, A y, long sites, double *result) {
+  using R = typename A::R;
   double acc = 0.0;
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) {
-        yv[s][c] = xv[s][c] - yv[s][c];
-        acc += (double)yv[s][c].re * yv[s][c].re + (double)yv[s][c].im * yv[s][c].im;
-      }
-    y.store_g(yv, g);
+    for (int k = 0; k < A::NCPLX; ++k) {
+      yv[k] = xv[k] - yv[k];
+      acc += (double)yv[k].re * yv[k].re + (double)yv[k].im * yv[k].im;
+    }
+    y.store_v(yv, g);
   }
   block_atomic_add(acc, result);
 }
 
 // x *= a
-template <typename Prec>
+template <typename A>
 __global__ __launch_bounds__(256) void k_scal(
-    typename Prec::Real a, SpinorAcc<Prec> x, long sites) {
-  using R = typename Prec::Real;
+    typename A::R a, A x, long sites) {
+  using R = typename A::R;
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3];
-    x.load_g(xv, g);
+    cplx<R> xv[A::NCPLX];
+    x.load_v(xv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) xv[s][c] = a * xv[s][c];
-    x.store_g(xv, g);
+    for (int k = 0; k < A::NCPLX; ++k) xv[k] = a * xv[k];
+    x.store_v(xv, g);
   }
 }
 
 // reductions: norm2, re<x,y>, <x,y> (re+im)
-template <typename Prec, int KIND>  // 0 norm2, 1 redot, 2 cdot
+template <typename A, int KIND>  // 0 norm2, 1 redot, 2 cdot
 __global__ __launch_bounds__(256) void k_reduce(
-    SpinorAcc<Prec> x, SpinorAcc<Prec> y, long sites, double *result) {
-  using R = typename Prec::Real;
+    A x, A y, long sites, double *result) {
+  using R = typename A::R;
   double acc = 0.0, acc2 = 0.0;
   GRID_STRIDE(g, sites) {
-    cplx<R> xv[4][3], yv[4][3];
-    x.load_g(xv, g);
-    if constexpr (KIND != 0) y.load_g(yv, g);
+    cplx<R> xv[A::NCPLX], yv[A::NCPLX];
+    x.load_v(xv, g);
+    if constexpr (KIND != 0) y.load_v(yv, g);
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) {
-        if constexpr (KIND == 0) {
-          acc += (double)xv[s][c].re * xv[s][c].re + (double)xv[s][c].im * xv[s][c].im;
-        } else if constexpr (KIND == 1) {
-          acc += (double)xv[s][c].re * yv[s][c].re + (double)xv[s][c].im * yv[s][c].im;
-        } else {
-          acc += (double)xv[s][c].re * yv[s][c].re + (double)xv[s][c].im * yv[s][c].im;
-          acc2 += (double)xv[s][c].re * yv[s][c].im - (double)xv[s][c].im * yv[s][c].re;
-        }
+    for (int k = 0; k < A::NCPLX; ++k) {
+      if constexpr (KIND == 0) {
+        acc += (double)xv[k].re * xv[k].re + (double)xv[k].im * xv[k].im;
+      } else if constexpr (KIND == 1) {
+        acc += (double)xv[k].re * yv[k].re + (double)xv[k].im * yv[k].im;
+      } else {
+        acc += (double)xv[k].re * yv[k].re + (double)xv[k].im * yv[k].im;
+        acc2 += (double)xv[k].re * yv[k].im - (double)xv[k].im * yv[k].re;
       }
+    }
   }
   block_atomic_add(acc, result);
   if constexpr (KIND == 2) block_atomic_add(acc2, result + 1);
 }
 
 // precision conversion copy (ref: lib/copy_color_spinor_*.cu)
-template <typename PrecDst, typename PrecSrc>
-__global__ __launch_bounds__(256) void k_convert(
-    SpinorAcc<PrecDst> dst, SpinorAcc<PrecSrc> src, long sites) {
-  using RS = typename PrecSrc::Real;
-  using RD = typename PrecDst::Real;
+template <typename AD, typename AS>
+__global__ __launch_bounds__(256) void k_convert(AD dst, AS src, long sites) {
+  using RS = typename AS::R;
+  using RD = typename AD::R;
+  static_assert(AD::NCPLX == AS::NCPLX);
   GRID_STRIDE(g, sites) {
-    cplx<RS> v[4][3];
-    src.load_g(v, g);
-    cplx<RD> o[4][3];
+    cplx<RS> v[AS::NCPLX];
+    src.load_v(v, g);
+    cplx<RD> o[AD::NCPLX];
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-#pragma unroll
-      for (int c = 0; c < 3; ++c) o[s][c] = {(RD)v[s][c].re, (RD)v[s][c].im};
-    dst.store_g(o, g);
+    for (int k = 0; k < AS::NCPLX; ++k) o[k] = {(RD)v[k].re, (RD)v[k].im};
+    dst.store_v(o, g);
   }
 }
 
@@ -232,93 +206,109 @@ int grid_for(long sites) {
   // cap + grid-stride (cdna_hip_programming.md Guideline 11)
   return (int)(g < 2048 ? g : 2048);
 }
-
-template <typename Prec>
-SpinorAcc<Prec> acc_of(const BlasField &f) {
-  return SpinorAcc<Prec>{(typename Prec::Store *)f.data, (float *)f.norm, f.Vcb};
-}
 }  // namespace
 
-template <typename Prec>
+template <typename A>
 static void blas_dispatch(const BlasCall &c, hipStream_t st) {
-  auto x = acc_of<Prec>(c.x);
-  auto y = acc_of<Prec>(c.y);
+  using S = typename A::S;
+  A x{(S *)c.x.data, (float *)c.x.norm, c.x.Vcb};
+  A y{(S *)c.y.data, (float *)c.y.norm, c.y.Vcb};
   long n = c.sites;
   int gr = grid_for(n);
-  using R = typename Prec::Real;
+  using R = typename A::R;
   switch (c.op) {
     case BLAS_AXPY:
-      hipLaunchKernelGGL((k_axpy<Prec, false>), dim3(gr), dim3(BLK), 0, st,
+      hipLaunchKernelGGL((k_axpy<A, false>), dim3(gr), dim3(BLK), 0, st,
                          (R)c.a, x, y, n, nullptr);
       break;
     case BLAS_AXPY_NORM2:
-      hipLaunchKernelGGL((k_axpy<Prec, true>), dim3(gr), dim3(BLK), 0, st,
+      hipLaunchKernelGGL((k_axpy<A, true>), dim3(gr), dim3(BLK), 0, st,
                          (R)c.a, x, y, n, c.result);
       break;
     case BLAS_XPAY:
-      hipLaunchKernelGGL((k_xpay<Prec>), dim3(gr), dim3(BLK), 0, st, x, (R)c.a, y, n);
+      hipLaunchKernelGGL((k_xpay<A>), dim3(gr), dim3(BLK), 0, st, x, (R)c.a, y, n);
       break;
     case BLAS_AXPBY:
-      hipLaunchKernelGGL((k_axpby<Prec>), dim3(gr), dim3(BLK), 0, st, (R)c.a, x,
+      hipLaunchKernelGGL((k_axpby<A>), dim3(gr), dim3(BLK), 0, st, (R)c.a, x,
                          (R)c.b, y, n);
       break;
     case BLAS_CAXPY:
-      hipLaunchKernelGGL((k_caxpy<Prec>), dim3(gr), dim3(BLK), 0, st, (R)c.a,
+      hipLaunchKernelGGL((k_caxpy<A>), dim3(gr), dim3(BLK), 0, st, (R)c.a,
                          (R)c.b, x, y, n);
       break;
     case BLAS_CAXPBY:
-      hipLaunchKernelGGL((k_caxpby<Prec>), dim3(gr), dim3(BLK), 0, st, (R)c.a,
+      hipLaunchKernelGGL((k_caxpby<A>), dim3(gr), dim3(BLK), 0, st, (R)c.a,
                          (R)c.b, x, (R)c.c, (R)c.d, y, n);
       break;
     case BLAS_XMY_NORM2:
-      hipLaunchKernelGGL((k_xmy_norm2<Prec>), dim3(gr), dim3(BLK), 0, st, x, y, n,
+      hipLaunchKernelGGL((k_xmy_norm2<A>), dim3(gr), dim3(BLK), 0, st, x, y, n,
                          c.result);
       break;
     case BLAS_SCAL:
-      hipLaunchKernelGGL((k_scal<Prec>), dim3(gr), dim3(BLK), 0, st, (R)c.a, x, n);
+      hipLaunchKernelGGL((k_scal<A>), dim3(gr), dim3(BLK), 0, st, (R)c.a, x, n);
       break;
     case BLAS_NORM2:
-      hipLaunchKernelGGL((k_reduce<Prec, 0>), dim3(gr), dim3(BLK), 0, st, x, y, n,
+      hipLaunchKernelGGL((k_reduce<A, 0>), dim3(gr), dim3(BLK), 0, st, x, y, n,
                          c.result);
       break;
     case BLAS_REDOT:
-      hipLaunchKernelGGL((k_reduce<Prec, 1>), dim3(gr), dim3(BLK), 0, st, x, y, n,
+      hipLaunchKernelGGL((k_reduce<A, 1>), dim3(gr), dim3(BLK), 0, st, x, y, n,
                          c.result);
       break;
     case BLAS_CDOT:
-      hipLaunchKernelGGL((k_reduce<Prec, 2>), dim3(gr), dim3(BLK), 0, st, x, y, n,
+      hipLaunchKernelGGL((k_reduce<A, 2>), dim3(gr), dim3(BLK), 0, st, x, y, n,
                          c.result);
       break;
   }
 }
 
 void launch_blas(const BlasCall &c, hipStream_t st) {
-  switch (c.prec) {
-    case 0: blas_dispatch<PrecDouble>(c, st); break;
-    case 1: blas_dispatch<PrecSingle>(c, st); break;
-    case 2: blas_dispatch<PrecHalf>(c, st); break;
+  if (c.ncomp == 6) {
+    switch (c.prec) {
+      case 0: blas_dispatch<StagAcc<PrecDouble>>(c, st); break;
+      case 1: blas_dispatch<StagAcc<PrecSingle>>(c, st); break;
+      case 2: blas_dispatch<StagAcc<PrecHalf>>(c, st); break;
+    }
+  } else {
+    switch (c.prec) {
+      case 0: blas_dispatch<SpinorAcc<PrecDouble>>(c, st); break;
+      case 1: blas_dispatch<SpinorAcc<PrecSingle>>(c, st); break;
+      case 2: blas_dispatch<SpinorAcc<PrecHalf>>(c, st); break;
+    }
   }
 }
 
-template <typename PD, typename PS>
+template <template <typename> class AT, typename PD, typename PS>
 static void conv(const BlasField &d, const BlasField &s, long sites, hipStream_t st) {
-  auto da = SpinorAcc<PD>{(typename PD::Store *)d.data, (float *)d.norm, d.Vcb};
-  auto sa = SpinorAcc<PS>{(typename PS::Store *)s.data, (float *)s.norm, s.Vcb};
-  hipLaunchKernelGGL((k_convert<PD, PS>), dim3(grid_for(sites)), dim3(BLK), 0, st,
-                     da, sa, sites);
+  auto da = AT<PD>{(typename PD::Store *)d.data, (float *)d.norm, d.Vcb};
+  auto sa = AT<PS>{(typename PS::Store *)s.data, (float *)s.norm, s.Vcb};
+  hipLaunchKernelGGL((k_convert<AT<PD>, AT<PS>>), dim3(grid_for(sites)),
+                     dim3(BLK), 0, st, da, sa, sites);
 }
+
+template <template <typename> class AT>
+static void conv_dispatch(const BlasField &dst, int pdst, const BlasField &src,
+                          int psrc, long sites, hipStream_t st) {
+  switch (pdst * 3 + psrc) {
+    case 0 * 3 + 1: conv<AT, PrecDouble, PrecSingle>(dst, src, sites, st); break;
+    case 0 * 3 + 2: conv<AT, PrecDouble, PrecHalf>(dst, src, sites, st); break;
+    case 1 * 3 + 0: conv<AT, PrecSingle, PrecDouble>(dst, src, sites, st); break;
+    case 1 * 3 + 2: conv<AT, PrecSingle, PrecHalf>(dst, src, sites, st); break;
+    case 2 * 3 + 0: conv<AT, PrecHalf, PrecDouble>(dst, src, sites, st); break;
+    case 2 * 3 + 1: conv<AT, PrecHalf, PrecSingle>(dst, src, sites, st); break;
+    case 0 * 3 + 0: conv<AT, PrecDouble, PrecDouble>(dst, src, sites, st); break;
+    case 1 * 3 + 1: conv<AT, PrecSingle, PrecSingle>(dst, src, sites, st); break;
+    case 2 * 3 + 2: conv<AT, PrecHalf, PrecHalf>(dst, src, sites, st); break;
+  }
+}
+
+template <typename P>
+using WilsonAccT = SpinorAcc<P, 24>;
 
 void launch_convert(const BlasField &dst, int pdst, const BlasField &src,
-                    int psrc, long sites, hipStream_t st) {
-  switch (pdst * 3 + psrc) {
-    case 0 * 3 + 1: conv<PrecDouble, PrecSingle>(dst, src, sites, st); break;
-    case 0 * 3 + 2: conv<PrecDouble, PrecHalf>(dst, src, sites, st); break;
-    case 1 * 3 + 0: conv<PrecSingle, PrecDouble>(dst, src, sites, st); break;
-    case 1 * 3 + 2: conv<PrecSingle, PrecHalf>(dst, src, sites, st); break;
-    case 2 * 3 + 0: conv<PrecHalf, PrecDouble>(dst, src, sites, st); break;
-    case 2 * 3 + 1: conv<PrecHalf, PrecSingle>(dst, src, sites, st); break;
-    case 0 * 3 + 0: conv<PrecDouble, PrecDouble>(dst, src, sites, st); break;
-    case 1 * 3 + 1: conv<PrecSingle, PrecSingle>(dst, src, sites, st); break;
-    case 2 * 3 + 2: conv<PrecHalf, PrecHalf>(dst, src, sites, st); break;
-  }
+                    int psrc, long sites, int ncomp, hipStream_t st) {
+  if (ncomp == 6)
+    conv_dispatch<StagAcc>(dst, pdst, src, psrc, sites, st);
+  else
+    conv_dispatch<WilsonAccT>(dst, pdst, src, psrc, sites, st);
 }

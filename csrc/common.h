@@ -81,14 +81,26 @@ __device__ __forceinline__ void store_chunk(S *p, const S in[W]) {
 }
 
 // ---------------------------------------------------------------------------
-// spinor accessor: 24 reals/site in [24/W][V][W]
+// spinor accessor: NCOMP reals/site in [NCOMP/CW][V][CW]; CW is the widest
+// chunk (<= Prec::W) dividing NCOMP (24 -> 16B chunks everywhere; staggered
+// 6 -> one complex per chunk, the reference's staggered FloatN choice)
 // ---------------------------------------------------------------------------
-template <typename Prec>
+template <int NC, int W>
+struct chunk_w {
+  static constexpr int value = (NC % W == 0) ? W : chunk_w<NC, W / 2>::value;
+};
+template <int NC>
+struct chunk_w<NC, 1> {
+  static constexpr int value = 1;
+};
+
+template <typename Prec, int NCOMP = 24>
 struct SpinorAcc {
   using S = typename Prec::Store;
   using R = typename Prec::Real;
-  static constexpr int W = Prec::W;
-  static constexpr int NCH = 24 / W;
+  static constexpr int W = chunk_w<NCOMP, Prec::W>::value;
+  static constexpr int NCH = NCOMP / W;
+  static constexpr int NCPLX = NCOMP / 2;
   S *data;
   float *norm;  // only for half
   long V;       // cb volume (chunk stride); fields span npar * V sites
@@ -99,8 +111,8 @@ struct SpinorAcc {
     return (p * NCH * V + (g - p * V)) * W;
   }
 
-  __device__ __forceinline__ void load_g(cplx<R> out[4][3], long g) const {
-    S tmp[24];
+  __device__ __forceinline__ void load_v(cplx<R> out[NCPLX], long g) const {
+    S tmp[NCOMP];
     long base = chunk_base(g);
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch)
@@ -108,30 +120,30 @@ struct SpinorAcc {
     R scale = (R)1;
     if constexpr (Prec::has_norm) scale = norm[g];
 #pragma unroll
-    for (int k = 0; k < 12; ++k)
-      out[k / 3][k % 3] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+    for (int k = 0; k < NCPLX; ++k)
+      out[k] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
   }
 
-  __device__ __forceinline__ void store_g(const cplx<R> in[4][3], long g) const {
-    S tmp[24];
+  __device__ __forceinline__ void store_v(const cplx<R> in[NCPLX], long g) const {
+    S tmp[NCOMP];
     long base = chunk_base(g);
     if constexpr (Prec::has_norm) {
       R m = (R)0;
 #pragma unroll
-      for (int k = 0; k < 12; ++k)
-        m = fmax(m, fmax(fabs(in[k / 3][k % 3].re), fabs(in[k / 3][k % 3].im)));
+      for (int k = 0; k < NCPLX; ++k)
+        m = fmax(m, fmax(fabs(in[k].re), fabs(in[k].im)));
       norm[g] = m;
       R inv = m > (R)0 ? (R)1 / m : (R)0;
 #pragma unroll
-      for (int k = 0; k < 12; ++k) {
-        tmp[2 * k] = (S)(in[k / 3][k % 3].re * inv);
-        tmp[2 * k + 1] = (S)(in[k / 3][k % 3].im * inv);
+      for (int k = 0; k < NCPLX; ++k) {
+        tmp[2 * k] = (S)(in[k].re * inv);
+        tmp[2 * k + 1] = (S)(in[k].im * inv);
       }
     } else {
 #pragma unroll
-      for (int k = 0; k < 12; ++k) {
-        tmp[2 * k] = (S)in[k / 3][k % 3].re;
-        tmp[2 * k + 1] = (S)in[k / 3][k % 3].im;
+      for (int k = 0; k < NCPLX; ++k) {
+        tmp[2 * k] = (S)in[k].re;
+        tmp[2 * k + 1] = (S)in[k].im;
       }
     }
 #pragma unroll
@@ -139,10 +151,24 @@ struct SpinorAcc {
       store_chunk<S, W>(data + base + (long)ch * V * W, tmp + ch * W);
   }
 
+  // [4][3]-shaped views for the Wilson kernels (NCOMP == 24 only)
+  __device__ __forceinline__ void load_g(cplx<R> (&out)[4][3], long g) const {
+    static_assert(NCOMP == 24);
+    load_v(reinterpret_cast<cplx<R> *>(out), g);
+  }
+  __device__ __forceinline__ void store_g(const cplx<R> (&in)[4][3], long g) const {
+    static_assert(NCOMP == 24);
+    store_v(reinterpret_cast<const cplx<R> *>(in), g);
+  }
+
   // single-parity (dslash) aliases: site index i in [0, V)
-  __device__ __forceinline__ void load(cplx<R> out[4][3], long i) const { load_g(out, i); }
-  __device__ __forceinline__ void store(const cplx<R> in[4][3], long i) const { store_g(in, i); }
+  __device__ __forceinline__ void load(cplx<R> (&out)[4][3], long i) const { load_g(out, i); }
+  __device__ __forceinline__ void store(const cplx<R> (&in)[4][3], long i) const { store_g(in, i); }
 };
+
+// staggered (nSpin=1) spinor: 6 reals/site, one complex color per chunk
+template <typename Prec>
+using StagAcc = SpinorAcc<Prec, 6>;
 
 // ---------------------------------------------------------------------------
 // gauge accessor — "stencil" layout (quda_amd/fields/gauge.py):
@@ -251,7 +277,17 @@ __device__ __forceinline__ long neighbor_cb(const int x[4], int mu, int dir, con
   return cb_from_coords(y, d);
 }
 
-#define HIP_CHECK(cmd)                                                         \
+// kernel roles for the comm-overlap split (role of the reference's
+// INTERIOR/EXTERIOR kernel_type, include/dslash_helper.cuh):
+//   LOCAL    : no partitioned dims — pure local stencil (zero overhead)
+//   FUSED    : ghost-aware single pass (blocking-comms policy)
+//   INTERIOR : skip hops crossing a partitioned boundary; non-affine
+//              epilogues on boundary sites are deferred to EXTERIOR
+// The *_exterior kernels add the ghost hops and complete deferred
+// epilogues; one owner thread per boundary site.
+enum DslashKT { KT_LOCAL = 0, KT_FUSED = 1, KT_INTERIOR = 2 };
+
+#define HIP_CHECK(cmd)                                                       \
   do {                                                                         \
     hipError_t e = (cmd);                                                      \
     if (e != hipSuccess) {                                                     \

@@ -1,0 +1,195 @@
+// Naive staggered (Kogut-Susskind) stencil for MI355X (gfx950)
+// (role of reference include/kernels/dslash_staggered.cuh — redesigned for
+//  the site-local pre-shifted stencil gauge layout; staggered phases
+//  eta_mu(x) computed on the fly from local coords, valid because local
+//  extents are even so rank offsets never flip a phase).
+//
+//   D psi(x) = sum_mu eta_mu(x) [ U_mu(x) psi(x+mu) - U_mu(x-mu)^dag psi(x-mu) ]
+//   eta_mu(x) = (-1)^(x_0 + ... + x_{mu-1}),  eta_0 = 1
+//
+// Kernel computes out = [a*x +] b*(D psi). D^dag = -D, so the dagger is
+// b -> -b at the call site (no DAG template).
+#pragma once
+
+#include "common.h"
+#include "halo.h"
+
+template <typename R>
+__device__ __forceinline__ void su3_mul_vec(cplx<R> out[3], const cplx<R> u[3][3],
+                                            const cplx<R> v[3]) {
+#pragma unroll
+  for (int r = 0; r < 3; ++r) {
+    cplx<R> acc = u[r][0] * v[0];
+    acc = cfma(u[r][1], v[1], acc);
+    acc = cfma(u[r][2], v[2], acc);
+    out[r] = acc;
+  }
+}
+
+template <typename R>
+__device__ __forceinline__ void su3_dagmul_vec(cplx<R> out[3], const cplx<R> u[3][3],
+                                               const cplx<R> v[3]) {
+#pragma unroll
+  for (int r = 0; r < 3; ++r) {
+    cplx<R> acc = cfma_conj(u[0][r], v[0], cplx<R>((R)0, (R)0));
+    acc = cfma_conj(u[1][r], v[1], acc);
+    acc = cfma_conj(u[2][r], v[2], acc);
+    out[r] = acc;
+  }
+}
+
+template <typename Prec, int RECON, bool XPAY, int KT = KT_LOCAL>
+__global__ __launch_bounds__(256) void k_dslash_staggered(
+    StagAcc<Prec> out, StagAcc<Prec> in, GaugeAcc<Prec, RECON> g, LatDims d,
+    int parity, typename Prec::Real a, typename Prec::Real b,
+    StagAcc<Prec> x, GhostAcc<Prec, 6> gh) {
+  using R = typename Prec::Real;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= d.Vcb) return;
+  int xc[4];
+  coords_from_cb(xc, i, d, parity);
+
+  cplx<R> acc[3];
+#pragma unroll
+  for (int c = 0; c < 3; ++c) acc[c] = {(R)0, (R)0};
+  cplx<R> p[3], up[3], U[3][3];
+  int esum = 0;  // running x_0+..+x_{mu-1}
+
+#define QA_SDIR(MU)                                                       \
+  {                                                                       \
+    R eta = (esum & 1) ? (R)-1 : (R)1;                                    \
+    bool cross_p = KT != KT_LOCAL && gh.active(MU) && xc[MU] == d.X[MU] - 1; \
+    if (!(KT == KT_INTERIOR && cross_p)) {                                \
+      if (KT == KT_FUSED && cross_p) {                                    \
+        gh.load_v(p, MU, 1, ghost_idx(xc, MU, d));                        \
+      } else {                                                            \
+        in.load_v(p, neighbor_cb(xc, MU, +1, d));                         \
+      }                                                                   \
+      g.template load<MU>(U, i);                                          \
+      su3_mul_vec(up, U, p);                                              \
+      for (int c = 0; c < 3; ++c) acc[c] += eta * up[c];                  \
+    }                                                                     \
+    bool cross_m = KT != KT_LOCAL && gh.active(MU) && xc[MU] == 0;        \
+    if (!(KT == KT_INTERIOR && cross_m)) {                                \
+      if (KT == KT_FUSED && cross_m) {                                    \
+        gh.load_v(p, MU, 0, ghost_idx(xc, MU, d));                        \
+      } else {                                                            \
+        in.load_v(p, neighbor_cb(xc, MU, -1, d));                         \
+      }                                                                   \
+      g.template load<4 + MU>(U, i);                                      \
+      su3_dagmul_vec(up, U, p);                                           \
+      for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \
+    }                                                                     \
+    esum += xc[MU];                                                       \
+  }
+
+  QA_SDIR(0)
+  QA_SDIR(1)
+  QA_SDIR(2)
+  QA_SDIR(3)
+#undef QA_SDIR
+
+  cplx<R> res[3];
+  if constexpr (XPAY) {
+    cplx<R> xv[3];
+    x.load_v(xv, i);
+#pragma unroll
+    for (int c = 0; c < 3; ++c) res[c] = a * xv[c] + b * acc[c];
+  } else {
+#pragma unroll
+    for (int c = 0; c < 3; ++c) res[c] = b * acc[c];
+  }
+  out.store_v(res, i);
+}
+
+// EXTERIOR companion (same ownership pattern as the Wilson exterior):
+// adds b * eta * (ghost hops) to the interior partial result. All
+// epilogues here are affine so nothing is deferred.
+template <typename Prec, int RECON>
+__global__ __launch_bounds__(256) void k_dslash_staggered_exterior(
+    StagAcc<Prec> out, StagAcc<Prec> in, GaugeAcc<Prec, RECON> g, LatDims d,
+    int parity, typename Prec::Real b, GhostAcc<Prec, 6> gh, long n_threads) {
+  using R = typename Prec::Real;
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= n_threads) return;
+  int mu = -1, edge = 0;
+  long f = tid;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    if (mu < 0 && gh.active(m)) {
+      if (f < 2 * gh.Fcb[m]) {
+        mu = m;
+        edge = f >= gh.Fcb[m];
+        if (edge) f -= gh.Fcb[m];
+      } else {
+        f -= 2 * gh.Fcb[m];
+      }
+    }
+  }
+  if (mu < 0) return;
+  int xc[4];
+  face_coords(xc, f, mu, edge ? d.X[mu] - 1 : 0, d, parity);
+  int key = 2 * mu + edge;
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    if (gh.active(m)) {
+      if (xc[m] == 0 && 2 * m < key) return;
+      if (xc[m] == d.X[m] - 1 && 2 * m + 1 < key) return;
+    }
+  }
+  long i = cb_from_coords(xc, d);
+
+  cplx<R> acc[3];
+#pragma unroll
+  for (int c = 0; c < 3; ++c) acc[c] = {(R)0, (R)0};
+  cplx<R> p[3], up[3], U[3][3];
+  int esum = 0;
+
+#define QA_SEXT(MU)                                                       \
+  {                                                                       \
+    R eta = (esum & 1) ? (R)-1 : (R)1;                                    \
+    if (gh.active(MU)) {                                                  \
+      if (xc[MU] == d.X[MU] - 1) {                                        \
+        gh.load_v(p, MU, 1, ghost_idx(xc, MU, d));                        \
+        g.template load<MU>(U, i);                                        \
+        su3_mul_vec(up, U, p);                                            \
+        for (int c = 0; c < 3; ++c) acc[c] += eta * up[c];                \
+      }                                                                   \
+      if (xc[MU] == 0) {                                                  \
+        gh.load_v(p, MU, 0, ghost_idx(xc, MU, d));                        \
+        g.template load<4 + MU>(U, i);                                    \
+        su3_dagmul_vec(up, U, p);                                         \
+        for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];             \
+      }                                                                   \
+    }                                                                     \
+    esum += xc[MU];                                                       \
+  }
+
+  QA_SEXT(0)
+  QA_SEXT(1)
+  QA_SEXT(2)
+  QA_SEXT(3)
+#undef QA_SEXT
+
+  cplx<R> prev[3];
+  out.load_v(prev, i);
+#pragma unroll
+  for (int c = 0; c < 3; ++c) prev[c] = prev[c] + b * acc[c];
+  out.store_v(prev, i);
+}
+
+// staggered face pack: full site (6 reals), no projection
+template <typename Prec, bool EDGE>
+__global__ __launch_bounds__(256) void k_pack_face_stag(
+    typename Prec::Store *dst, float *dst_nrm, StagAcc<Prec> in, LatDims d,
+    int parity, int mu, long Fcb) {
+  using R = typename Prec::Real;
+  long f = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (f >= Fcb) return;
+  int xc[4];
+  face_coords(xc, f, mu, EDGE ? d.X[mu] - 1 : 0, d, parity);
+  long i = cb_from_coords(xc, d);
+  cplx<R> v[3];
+  in.load_v(v, i);
+  ghost_store_v<Prec, 6>(dst, dst_nrm, Fcb, f, v);
+}

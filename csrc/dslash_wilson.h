@@ -40,17 +40,6 @@ __device__ __forceinline__ void twist_mul(cplx<R> v[4][3], R br, R bi) {
   }
 }
 
-// kernel roles for the comm-overlap split (role of the reference's
-// INTERIOR/EXTERIOR kernel_type, include/dslash_helper.cuh):
-//   LOCAL    : no partitioned dims — pure local stencil (zero overhead)
-//   FUSED    : ghost-aware single pass (blocking-comms policy)
-//   INTERIOR : skip hops crossing a partitioned boundary; boundary sites
-//              with MODE==CLOV_POST store the raw hop sum (epilogue
-//              deferred to EXTERIOR), other modes apply the (affine)
-//              epilogue to the partial sum
-// The EXTERIOR kernel (k_dslash_wilson_exterior) adds the ghost hops and
-// completes deferred epilogues; one owner thread per boundary site.
-enum DslashKT { KT_LOCAL = 0, KT_FUSED = 1, KT_INTERIOR = 2 };
 
 // packed clover (fields/clover.py): per site 72 reals =
 // 2 chirality blocks x (6 diag + 15 lower-tri complex)

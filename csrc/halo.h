@@ -45,12 +45,14 @@ __device__ __forceinline__ void face_coords(int x[4], long f, int mu, int fix,
   x[rd0] = 2 * x0h + odd;
 }
 
-template <typename Prec>
+template <typename Prec, int NCOMP = 12>
 struct GhostAcc {
   using S = typename Prec::Store;
   using R = typename Prec::Real;
-  static constexpr int GW = (Prec::W == 8) ? 4 : Prec::W;
-  static constexpr int NCH = 12 / GW;
+  static constexpr int GW =
+      chunk_w<NCOMP, (Prec::W == 8) ? 4 : Prec::W>::value;
+  static constexpr int NCH = NCOMP / GW;
+  static constexpr int NCPLX = NCOMP / 2;
   const S *buf[8];      // [2*mu+dir]; null when dim not partitioned
   const float *nrm[8];  // half only
   long Fcb[4];
@@ -58,52 +60,69 @@ struct GhostAcc {
 
   __device__ __forceinline__ bool active(int mu) const { return (mask >> mu) & 1; }
 
-  __device__ __forceinline__ void load(cplx<R> h[2][3], int mu, int dir, long f) const {
+  __device__ __forceinline__ void load_v(cplx<R> out[NCPLX], int mu, int dir,
+                                         long f) const {
     const S *b = buf[2 * mu + dir];
-    S tmp[12];
+    S tmp[NCOMP];
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch)
       load_chunk<S, GW>(b + ((long)ch * Fcb[mu] + f) * GW, tmp + ch * GW);
     R scale = (R)1;
     if constexpr (Prec::has_norm) scale = nrm[2 * mu + dir][f];
 #pragma unroll
-    for (int k = 0; k < 6; ++k)
-      h[k / 3][k % 3] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+    for (int k = 0; k < NCPLX; ++k)
+      out[k] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+  }
+
+  // Wilson half-spinor view (NCOMP == 12)
+  __device__ __forceinline__ void load(cplx<R> (&h)[2][3], int mu, int dir,
+                                       long f) const {
+    static_assert(NCOMP == 12);
+    load_v(reinterpret_cast<cplx<R> *>(h), mu, dir, f);
   }
 };
 
-// write a projected half-spinor into a send buffer slot
-template <typename Prec>
-__device__ __forceinline__ void ghost_store(typename Prec::Store *buf, float *nrm,
-                                            long Fcb, long f,
-                                            const cplx<typename Prec::Real> h[2][3]) {
+// write NCOMP/2 complex values into a send buffer slot
+template <typename Prec, int NCOMP>
+__device__ __forceinline__ void ghost_store_v(
+    typename Prec::Store *buf, float *nrm, long Fcb, long f,
+    const cplx<typename Prec::Real> *h) {
   using S = typename Prec::Store;
   using R = typename Prec::Real;
-  constexpr int GW = GhostAcc<Prec>::GW;
-  constexpr int NCH = GhostAcc<Prec>::NCH;
-  S tmp[12];
+  constexpr int GW = GhostAcc<Prec, NCOMP>::GW;
+  constexpr int NCH = GhostAcc<Prec, NCOMP>::NCH;
+  constexpr int NCPLX = NCOMP / 2;
+  S tmp[NCOMP];
   if constexpr (Prec::has_norm) {
     R m = (R)0;
 #pragma unroll
-    for (int k = 0; k < 6; ++k)
-      m = fmax(m, fmax(fabs(h[k / 3][k % 3].re), fabs(h[k / 3][k % 3].im)));
+    for (int k = 0; k < NCPLX; ++k)
+      m = fmax(m, fmax(fabs(h[k].re), fabs(h[k].im)));
     nrm[f] = m;
     R inv = m > (R)0 ? (R)1 / m : (R)0;
 #pragma unroll
-    for (int k = 0; k < 6; ++k) {
-      tmp[2 * k] = (S)(h[k / 3][k % 3].re * inv);
-      tmp[2 * k + 1] = (S)(h[k / 3][k % 3].im * inv);
+    for (int k = 0; k < NCPLX; ++k) {
+      tmp[2 * k] = (S)(h[k].re * inv);
+      tmp[2 * k + 1] = (S)(h[k].im * inv);
     }
   } else {
 #pragma unroll
-    for (int k = 0; k < 6; ++k) {
-      tmp[2 * k] = (S)h[k / 3][k % 3].re;
-      tmp[2 * k + 1] = (S)h[k / 3][k % 3].im;
+    for (int k = 0; k < NCPLX; ++k) {
+      tmp[2 * k] = (S)h[k].re;
+      tmp[2 * k + 1] = (S)h[k].im;
     }
   }
 #pragma unroll
   for (int ch = 0; ch < NCH; ++ch)
     store_chunk<S, GW>(buf + ((long)ch * Fcb + f) * GW, tmp + ch * GW);
+}
+
+template <typename Prec>
+__device__ __forceinline__ void ghost_store(typename Prec::Store *buf, float *nrm,
+                                            long Fcb, long f,
+                                            const cplx<typename Prec::Real> h[2][3]) {
+  ghost_store_v<Prec, 12>(buf, nrm, Fcb, f,
+                          reinterpret_cast<const cplx<typename Prec::Real> *>(h));
 }
 
 // Pack one face of `in` (the dslash input spinor, at parity `parity`) into a

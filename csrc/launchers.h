@@ -29,12 +29,13 @@ struct BlasCall {
   double c, d;   // second complex scalar (caxpby: b=(c,d))
   BlasField x, y;
   long sites;    // npar * Vcb
+  int ncomp;     // reals per site: 24 (Wilson) or 6 (staggered)
   double *result;  // device ptr for reductions (>=2 doubles for cdot)
 };
 
 void launch_blas(const BlasCall &c, hipStream_t st);
 void launch_convert(const BlasField &dst, int pdst, const BlasField &src,
-                    int psrc, long sites, hipStream_t st);
+                    int psrc, long sites, int ncomp, hipStream_t st);
 
 // ---------------------------------------------------------------------------
 struct DslashCall {
@@ -78,6 +79,7 @@ struct PackCall {
   int s01;            // projector sign index (dir XOR dagger)
   int edge;           // 0: x_mu = 0 face, 1: x_mu = X-1 face
   long Fcb;
+  int prec;
 };
 void launch_pack_face_double(const PackCall &c, hipStream_t st);
 void launch_pack_face_single(const PackCall &c, hipStream_t st);
@@ -100,3 +102,24 @@ struct TwistApplyCall {
   int prec;
 };
 void launch_twist_apply(const TwistApplyCall &c, hipStream_t st);
+
+// ---------------------------------------------------------------------------
+struct StagDslashCall {
+  BlasField out, in, x;  // single-parity views
+  const void *gauge;     // stencil layout, like DslashCall
+  int Xdim[4];
+  int parity_offset;
+  long Vcb;
+  int parity;
+  bool xpay;
+  double a, b;  // out = [a*x +] b*(D in); dagger folds into b
+  int recon;
+  const void *ghost[8];
+  const float *ghost_nrm[8];
+  long face_cb[4];
+  int comm_mask;
+  int kt;  // 0 local, 1 fused, 2 interior, 3 exterior
+  int prec;
+};
+void launch_dslash_staggered(const StagDslashCall &c, hipStream_t st);
+void launch_pack_face_stag(const PackCall &c, hipStream_t st);

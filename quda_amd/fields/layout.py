@@ -37,11 +37,18 @@ WIDTH_OF = {  # reals per 16-byte chunk
 }
 
 
-def n_chunks(ncomp: int, prec: str) -> int:
+def chunk_width(ncomp: int, prec: str) -> int:
+    """Widest chunk (in reals) <= the precision's 16B width that divides
+    ncomp (mirrors csrc/common.h chunk_w<>: 24 -> 16B chunks, 6 -> one
+    complex per chunk)."""
     w = WIDTH_OF[prec]
-    if ncomp % w != 0:
-        raise ValueError(f"{ncomp} components not divisible by chunk width {w} ({prec})")
-    return ncomp // w
+    while ncomp % w:
+        w //= 2
+    return w
+
+
+def n_chunks(ncomp: int, prec: str) -> int:
+    return ncomp // chunk_width(ncomp, prec)
 
 
 def complex_to_chunked(site_comp: torch.Tensor, prec: str) -> torch.Tensor:
@@ -49,7 +56,7 @@ def complex_to_chunked(site_comp: torch.Tensor, prec: str) -> torch.Tensor:
 
     Leading dims (parity etc.) are preserved.
     """
-    w = WIDTH_OF[prec]
+    w = chunk_width(site_comp.shape[-1] * 2, prec)
     real = torch.view_as_real(site_comp)           # [..., V, ncomp/2, 2]
     flat = real.reshape(*site_comp.shape[:-1], -1)  # [..., V, ncomp]
     ncomp = flat.shape[-1]

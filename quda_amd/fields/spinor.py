@@ -24,18 +24,20 @@ PARITY = 1  # single-parity field (even-odd solves)
 
 
 class SpinorField:
-    nspin = 4
+    """nspin=4 Wilson-type or nspin=1 staggered-type fermion field."""
     ncolor = 3
 
     def __init__(self, geo: LatticeGeometry, precision: str = "double",
                  device="cpu", n_parity: int = FULL,
                  data: Optional[torch.Tensor] = None,
-                 norm: Optional[torch.Tensor] = None):
+                 norm: Optional[torch.Tensor] = None, nspin: int = 4):
         self.geo = geo
         self.precision = precision
         self.n_parity = n_parity
+        self.nspin = nspin
         self.ncomp = self.nspin * self.ncolor * 2
-        w = WIDTH_OF[precision]
+        from .layout import chunk_width
+        w = chunk_width(self.ncomp, precision)
         nch = n_chunks(self.ncomp, precision)
         shape = (n_parity, nch, geo.volume_cb, w)
         if data is not None:
@@ -62,9 +64,13 @@ class SpinorField:
     def volume_cb(self) -> int:
         return self.geo.volume_cb
 
+    @property
+    def site_shape(self):
+        return (self.nspin, self.ncolor) if self.nspin > 1 else (self.ncolor,)
+
     def clone_empty(self, precision: Optional[str] = None) -> "SpinorField":
         return SpinorField(self.geo, precision or self.precision,
-                           self.device, self.n_parity)
+                           self.device, self.n_parity, nspin=self.nspin)
 
     def copy_(self, src: "SpinorField") -> "SpinorField":
         """Any-precision copy (ref: lib/copy_color_spinor_*.cu)."""
@@ -79,14 +85,15 @@ class SpinorField:
     # -- oracle layout conversions -------------------------------------
     def to_complex(self, dtype=torch.complex128) -> torch.Tensor:
         """-> [n_parity, V_cb, 4, 3] complex (denormalized for half)."""
-        c = chunked_to_complex(self.data, dtype)  # [P, V, 12]
+        c = chunked_to_complex(self.data, dtype)  # [P, V, ncomp/2]
         if self.norm is not None:
             c = c * self.norm.to(c.real.dtype).unsqueeze(-1)
-        return c.reshape(self.n_parity, self.volume_cb, self.nspin, self.ncolor)
+        return c.reshape(self.n_parity, self.volume_cb, *self.site_shape)
 
     def from_complex(self, c: torch.Tensor) -> "SpinorField":
-        assert c.shape == (self.n_parity, self.volume_cb, self.nspin, self.ncolor)
-        flat = c.reshape(self.n_parity, self.volume_cb, 12)
+        assert c.shape == (self.n_parity, self.volume_cb, *self.site_shape), \
+            (c.shape, self.site_shape)
+        flat = c.reshape(self.n_parity, self.volume_cb, self.ncomp // 2)
         if self.precision == "half":
             mags = torch.view_as_real(flat).abs().amax(dim=(-1, -2))  # [P,V]
             self.norm.copy_(mags.to(torch.float32))
@@ -106,7 +113,7 @@ class SpinorField:
         g = torch.Generator(device="cpu")
         if seed is not None:
             g.manual_seed(seed)
-        c = torch.randn((self.n_parity, self.volume_cb, self.nspin, self.ncolor, 2),
+        c = torch.randn((self.n_parity, self.volume_cb, *self.site_shape, 2),
                         generator=g, dtype=torch.float64)
         self.from_complex(torch.view_as_complex(c).to(self.device))
         return self
@@ -115,14 +122,16 @@ class SpinorField:
     def to(self, device) -> "SpinorField":
         return SpinorField(self.geo, self.precision, device, self.n_parity,
                            data=self.data.to(device),
-                           norm=None if self.norm is None else self.norm.to(device))
+                           norm=None if self.norm is None else self.norm.to(device),
+                           nspin=self.nspin)
 
     def parity_view(self, parity: int) -> "SpinorField":
         """Zero-copy single-parity view of a full field (ref Even()/Odd())."""
         assert self.n_parity == FULL
         return SpinorField(self.geo, self.precision, self.device, PARITY,
                            data=self.data[parity:parity + 1],
-                           norm=None if self.norm is None else self.norm[parity:parity + 1])
+                           norm=None if self.norm is None else self.norm[parity:parity + 1],
+                           nspin=self.nspin)
 
     def __repr__(self):
         return (f"SpinorField({self.geo.dims}, {self.precision}, "

@@ -1,7 +1,8 @@
+from .staggered import DiracStaggered, DiracStaggeredPC
 from .dirac import (Dirac, DiracClover, DiracCloverPC, DiracTwistedClover,
                     DiracTwistedMass, DiracTwistedMassPC, DiracWilson,
                     DiracWilsonPC)
 
 __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracCloverPC", "DiracTwistedMass", "DiracTwistedMassPC",
-           "DiracTwistedClover"]
+           "DiracTwistedClover", "DiracStaggered", "DiracStaggeredPC"]

@@ -25,18 +25,19 @@ def _pairs(t: torch.Tensor) -> torch.Tensor:
 
 
 def _requant(x: SpinorField, vals: torch.Tensor):
-    x.from_complex(vals.reshape(x.n_parity, x.volume_cb, 4, 3))
+    x.from_complex(vals.reshape(x.n_parity, x.volume_cb, *x.site_shape))
 
 
 def _cvals(x: SpinorField) -> torch.Tensor:
-    return x.to_complex(torch.complex128).reshape(x.n_parity, x.volume_cb, 12)
+    return x.to_complex(torch.complex128).reshape(x.n_parity, x.volume_cb,
+                                                  x.ncomp // 2)
 
 
 def _gpu_blas(op: int, a, b, x: SpinorField, y: SpinorField):
     ext = hip_ext()
     return ext.blas_op(op, float(a), float(b), x.data, norm_or_empty(x),
                        y.data, norm_or_empty(y), x.volume_cb,
-                       x.n_parity * x.volume_cb)
+                       x.n_parity * x.volume_cb, ncomp=x.ncomp)
 
 
 # -- reductions -------------------------------------------------------------
@@ -112,7 +113,8 @@ def copy(dst: SpinorField, src: SpinorField) -> SpinorField:
     if on_gpu(dst, src) and dst.precision != src.precision:
         ext = hip_ext()
         ext.convert(dst.data, norm_or_empty(dst), src.data, norm_or_empty(src),
-                    dst.volume_cb, dst.n_parity * dst.volume_cb)
+                    dst.volume_cb, dst.n_parity * dst.volume_cb,
+                    ncomp=dst.ncomp)
         return dst
     dst.copy_(src)
     return dst
@@ -183,7 +185,7 @@ def caxpby(a: complex, x: SpinorField, b: complex, y: SpinorField) -> SpinorFiel
         ext = hip_ext()
         ext.blas_op(ext.BLAS_CAXPBY, a.real, a.imag, x.data, norm_or_empty(x),
                     y.data, norm_or_empty(y), x.volume_cb,
-                    x.n_parity * x.volume_cb, b.real, b.imag)
+                    x.n_parity * x.volume_cb, b.real, b.imag, ncomp=x.ncomp)
         return y
     _requant_any(y, a * _cvals(x) + b * _cvals(y))
     return y
@@ -191,7 +193,7 @@ def caxpby(a: complex, x: SpinorField, b: complex, y: SpinorField) -> SpinorFiel
 
 def _requant_any(y: SpinorField, vals):
     """Write complex values back into y at its precision."""
-    y.from_complex(vals.reshape(y.n_parity, y.volume_cb, 4, 3))
+    y.from_complex(vals.reshape(y.n_parity, y.volume_cb, *y.site_shape))
 
 
 def scal(a: float, x: SpinorField) -> SpinorField:

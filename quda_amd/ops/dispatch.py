@@ -199,3 +199,61 @@ def apply_clover(out: SpinorField, inp: SpinorField, clover, parity: int,
     psi = inp.to_complex()[0]
     out.from_complex(ref.apply_clover(A, psi).unsqueeze(0))
     return out
+
+
+def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
+                     parity: int, a: float = 0.0,
+                     b: float = 1.0, x: Optional[SpinorField] = None):
+    """Staggered stencil: out = [a*x +] b*(D in); out at `parity`, in at
+    the opposite parity (nspin=1 fields). D^dag = -D: pass b=-b for the
+    dagger."""
+    from ..parallel import comms
+    geo = out.geo
+    xpay = x is not None
+    mask = comms.comm_mask()
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        xf = x if x is not None else out
+
+        def launch(kt, ghosts=[], nrms=[], face_cb=[]):
+            ext.dslash_staggered(
+                out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+                gauge.data, xf.data, norm_or_empty(xf), list(geo.dims),
+                geo.parity_offset, geo.volume_cb, parity, xpay, float(a),
+                float(b), RECON_COMPS[gauge.reconstruct], ghosts, nrms,
+                face_cb, mask if kt else 0, kt)
+
+        if not mask:
+            launch(0)
+            return out
+        from ..parallel.halo import get_spinor_halo
+        h = get_spinor_halo(geo, inp.precision, inp.device, mask, ncomp=6)
+        h.pack(ext, inp, 1 - parity, False)
+        ghosts, nrms, face_cb = h.ghost_args()
+        if dslash_policy() == "fused":
+            h.exchange()
+            launch(1, ghosts, nrms, face_cb)
+        else:
+            reqs = h.exchange_start()
+            launch(2, ghosts, nrms, face_cb)
+            for r in reqs:
+                r.wait()
+            launch(3, ghosts, nrms, face_cb)
+        return out
+    # ---- oracle path ----
+    u = gauge.to_complex()
+    psi = inp.to_complex()[0]
+    halo = None
+    if mask:
+        from ..parallel.halo import active_dims, exchange_psi_oracle
+        halo = {
+            "mask": mask,
+            "psi": exchange_psi_oracle(psi, geo, 1 - parity, mask),
+            "u_bwd": {mu: gauge.bwd_ghost(mu, parity)
+                      for mu in active_dims(mask)},
+        }
+    res = b * ref.dslash_staggered_parity(u, psi, geo, parity, halo=halo)
+    if xpay:
+        res = a * x.to_complex()[0] + res
+    out.from_complex(res.unsqueeze(0))
+    return out

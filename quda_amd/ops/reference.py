@@ -221,3 +221,64 @@ def plaquette(u: torch.Tensor, geo: LatticeGeometry):
     s = sum(sp) / len(sp)
     t = sum(tp) / len(tp)
     return (s + t) / 2, s, t
+
+
+# ---------------------------------------------------------------------------
+# Staggered (Kogut-Susskind) oracle (ref: tests/host_reference/
+# staggered_dslash_reference.cpp — re-derived from the operator definition)
+# ---------------------------------------------------------------------------
+
+def staggered_phases(geo: LatticeGeometry, parity: int) -> torch.Tensor:
+    """[V_cb, 4] +-1 eta_mu(x) = (-1)^(x_0+..+x_{mu-1}) for parity sites."""
+    c = geo.coords_of_cb(parity).to(torch.int64)
+    pref = torch.zeros_like(c)
+    pref[:, 1] = c[:, 0]
+    pref[:, 2] = c[:, 0] + c[:, 1]
+    pref[:, 3] = c[:, 0] + c[:, 1] + c[:, 2]
+    return torch.where(pref % 2 == 0, 1.0, -1.0).to(torch.float64)
+
+
+def dslash_staggered_parity(u: torch.Tensor, psi: torch.Tensor,
+                            geo: LatticeGeometry, parity: int,
+                            halo=None) -> torch.Tensor:
+    """out(parity) = sum_mu eta_mu(x)[U_mu(x) psi(x+mu)
+                                      - U_mu(x-mu)^dag psi(x-mu)].
+
+    psi: [V_cb, 3] complex at parity 1-parity. halo as in
+    dslash_wilson_parity but with [Fcb, 3] spinor ghosts."""
+    dev, dt = psi.device, psi.dtype
+    other = 1 - parity
+    out = torch.zeros_like(psi)
+    eta = staggered_phases(geo, parity).to(dev)
+    mask = halo["mask"] if halo else 0
+    for mu in range(4):
+        part = (mask >> mu) & 1
+        fwd_idx = geo.neighbor_cb(parity, mu, +1).to(dev)
+        psi_f = psi[fwd_idx]
+        if part:
+            fidx = geo.face_index_cb(parity, mu, geo.dims[mu] - 1).to(dev)
+            psi_f[fidx] = halo["psi"][(mu, 1)].to(dt)
+        e = eta[:, mu].to(dt).unsqueeze(-1)
+        out += e * torch.einsum("vij,vj->vi", u[mu, parity], psi_f)
+        bwd_idx = geo.neighbor_cb(parity, mu, -1).to(dev)
+        psi_b = psi[bwd_idx]
+        u_b = u[mu, other][bwd_idx]
+        if part:
+            fidx0 = geo.face_index_cb(parity, mu, 0).to(dev)
+            psi_b[fidx0] = halo["psi"][(mu, 0)].to(dt)
+            u_b = u_b.clone()
+            u_b[fidx0] = halo["u_bwd"][mu].to(dt)
+        out -= e * torch.einsum("vji,vj->vi", u_b.conj(), psi_b)
+    return out
+
+
+def dslash_staggered_full(u, psi_full, geo):
+    out = torch.empty_like(psi_full)
+    for p in (0, 1):
+        out[p] = dslash_staggered_parity(u, psi_full[1 - p], geo, p)
+    return out
+
+
+def mat_staggered(u, psi_full, geo, mass: float):
+    """M psi = 2m psi + D psi (mass normalization; D antihermitian)."""
+    return 2.0 * mass * psi_full + dslash_staggered_full(u, psi_full, geo)

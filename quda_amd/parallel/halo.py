@@ -96,31 +96,41 @@ def exchange_tensors_start(sends: Dict[Key, torch.Tensor],
 # native (device-layout) spinor halo: persistent buffers + HIP pack kernels
 # ---------------------------------------------------------------------------
 
-GHOST_W = {"double": 2, "single": 4, "half": 4}  # reals per ghost chunk
+GHOST_W0 = {"double": 2, "single": 4, "half": 4}  # max reals per ghost chunk
+
+
+def ghost_width(ncomp: int, precision: str) -> int:
+    """Mirrors csrc/halo.h GhostAcc<> chunk width."""
+    w = GHOST_W0[precision]
+    while ncomp % w:
+        w //= 2
+    return w
 
 
 class SpinorHalo:
     """Persistent send/recv ghost buffers for one (geometry, precision,
-    device, mask) signature (role of the reference's static ghost arenas,
-    lattice_field.h:250)."""
+    device, mask, ncomp) signature (role of the reference's static ghost
+    arenas, lattice_field.h:250). ncomp = 12 (spin-projected Wilson) or
+    6 (staggered full site)."""
 
     def __init__(self, geo: LatticeGeometry, precision: str, device,
-                 mask: int):
+                 mask: int, ncomp: int = 12):
         from ..fields.layout import DTYPE_OF
         self.geo = geo
         self.precision = precision
         self.mask = mask
+        self.ncomp = ncomp
         self.device = torch.device(device)
         self.send: Dict[Key, torch.Tensor] = {}
         self.recv: Dict[Key, torch.Tensor] = {}
         self.send_nrm: Dict[Key, torch.Tensor] = {}
         self.recv_nrm: Dict[Key, torch.Tensor] = {}
-        gw = GHOST_W[precision]
+        gw = ghost_width(ncomp, precision)
         dt = DTYPE_OF[precision]
         for mu in active_dims(mask):
             fcb = geo.face_volume_cb(mu)
             for d in (0, 1):
-                shape = (12 // gw, fcb, gw)
+                shape = (ncomp // gw, fcb, gw)
                 self.send[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
                 self.recv[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
                 if precision == "half":
@@ -130,21 +140,29 @@ class SpinorHalo:
                                                          device=device)
 
     def pack(self, ext, inp, parity: int, dagger: bool) -> None:
-        """Pack all active faces of `inp` (the dslash input, at `parity`)
-        with the projector the consuming hop applies."""
+        """Pack all active faces of `inp` (the dslash input, at `parity`):
+        Wilson ghosts carry the projector the consuming hop applies,
+        staggered ghosts the full site."""
         geo = self.geo
         empty = torch.empty(0, dtype=torch.float32, device=self.device)
         from ..ops.dispatch import norm_or_empty
         for mu in active_dims(self.mask):
             fcb = geo.face_volume_cb(mu)
             for d in (0, 1):
-                s01 = d ^ (1 if dagger else 0)
                 edge = 0 if d == 0 else 1
-                ext.pack_face(self.send[(mu, d)],
-                              self.send_nrm.get((mu, d), empty),
-                              inp.data, norm_or_empty(inp),
-                              list(geo.dims), geo.parity_offset,
-                              geo.volume_cb, parity, mu, s01, edge, fcb)
+                if self.ncomp == 6:
+                    ext.pack_face_stag(self.send[(mu, d)],
+                                       self.send_nrm.get((mu, d), empty),
+                                       inp.data, norm_or_empty(inp),
+                                       list(geo.dims), geo.parity_offset,
+                                       geo.volume_cb, parity, mu, edge, fcb)
+                else:
+                    s01 = d ^ (1 if dagger else 0)
+                    ext.pack_face(self.send[(mu, d)],
+                                  self.send_nrm.get((mu, d), empty),
+                                  inp.data, norm_or_empty(inp),
+                                  list(geo.dims), geo.parity_offset,
+                                  geo.volume_cb, parity, mu, s01, edge, fcb)
 
     def exchange(self) -> None:
         for r in self.exchange_start():
@@ -176,11 +194,11 @@ _HALO_CACHE: Dict[tuple, SpinorHalo] = {}
 
 
 def get_spinor_halo(geo: LatticeGeometry, precision: str, device,
-                    mask: int) -> SpinorHalo:
-    key = (geo.dims, geo.parity_offset, precision, str(device), mask)
+                    mask: int, ncomp: int = 12) -> SpinorHalo:
+    key = (geo.dims, geo.parity_offset, precision, str(device), mask, ncomp)
     h = _HALO_CACHE.get(key)
     if h is None:
-        h = SpinorHalo(geo, precision, device, mask)
+        h = SpinorHalo(geo, precision, device, mask, ncomp)
         _HALO_CACHE[key] = h
     return h
 
@@ -191,8 +209,9 @@ def get_spinor_halo(geo: LatticeGeometry, precision: str, device,
 
 def exchange_psi_oracle(psi: torch.Tensor, geo: LatticeGeometry,
                         parity_in: int, mask: int) -> Dict[Key, torch.Tensor]:
-    """Exchange full (unprojected) spinor faces of `psi` ([V_cb,4,3] complex
-    at parity_in). Returns {(mu,dir): [Fcb,4,3]} ghosts in ghost-index order."""
+    """Exchange full (unprojected) spinor faces of `psi` ([V_cb,4,3] or
+    [V_cb,3] complex at parity_in). Returns {(mu,dir): faces} in
+    ghost-index order."""
     sends, recvs = {}, {}
     for mu in active_dims(mask):
         hi = geo.dims[mu] - 1

@@ -45,13 +45,13 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
     prec_hi = x.precision
     sloppy = sloppy or prec_hi
     op_sloppy = op_sloppy or op
-    geo, dev, npar = x.geo, x.device, x.n_parity
+    geo, dev, npar, nsp = x.geo, x.device, x.n_parity, x.nspin
 
     def hi():
-        return SpinorField(geo, prec_hi, dev, npar)
+        return SpinorField(geo, prec_hi, dev, npar, nspin=nsp)
 
     def lo():
-        return SpinorField(geo, sloppy, dev, npar)
+        return SpinorField(geo, sloppy, dev, npar, nspin=nsp)
 
     b2 = blas.norm2(b)
     if b2 == 0.0:
