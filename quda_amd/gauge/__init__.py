@@ -1,0 +1,18 @@
+"""Gauge-sector tools (ref: SURVEY.md 2.9 — observables, smearing, HMC).
+
+Torch-tensor implementations over the oracle gauge layout
+[4, 2, V_cb, 3, 3]; they run on CPU or GPU (torch-ROCm dispatches to HIP
+elementwise/rocBLAS batched kernels). Single-rank for now: these are
+setup/analysis utilities, not the solver hot path."""
+
+from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
+                  plaquette, polyakov_loop, project_ta, staple_sum,
+                  stout_smear, topological_charge, wilson_flow)
+from .hmc import hmc_trajectory, leapfrog, mom_action, random_momentum
+from .heatbath import heatbath_sweep, overrelax_sweep
+
+__all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
+           "project_ta", "exp_su3", "ape_smear", "stout_smear",
+           "wilson_flow", "polyakov_loop", "topological_charge",
+           "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",
+           "heatbath_sweep", "overrelax_sweep"]

@@ -1,0 +1,204 @@
+"""Gauge observables, smearing, flow and forces
+(ref: lib/gauge_plaq.cu, gauge_ape.cu, gauge_stout.cu, gauge_wilson_flow.cu,
+gauge_force.cu, gauge_qcharge.cu, gauge_polyakov_loop.cu + the kernels in
+include/kernels/gauge_*.cuh — re-derived from the standard definitions)."""
+
+from __future__ import annotations
+
+import torch
+
+from ..fields.geometry import LatticeGeometry
+from ..ops.reference import field_strength
+
+
+def _to_lex(u: torch.Tensor, geo: LatticeGeometry) -> torch.Tensor:
+    """[4,2,V,3,3] cb -> [4,Vlex,3,3]."""
+    lo = geo.lex_of_cb.to(u.device)
+    U = torch.empty((4, geo.volume, 3, 3), dtype=u.dtype, device=u.device)
+    U[:, lo[0]] = u[:, 0]
+    U[:, lo[1]] = u[:, 1]
+    return U
+
+
+def _from_lex(U: torch.Tensor, geo: LatticeGeometry) -> torch.Tensor:
+    lo = geo.lex_of_cb.to(U.device)
+    return torch.stack([U[:, lo[0]], U[:, lo[1]]], dim=1)
+
+
+def _shift(f: torch.Tensor, geo: LatticeGeometry, mu: int, disp: int):
+    """f: [Vlex,...] -> f(x + disp*mu)."""
+    idx = geo.neighbor_lex(mu, disp).to(f.device)
+    return f[idx]
+
+
+def plaquette(u: torch.Tensor, geo: LatticeGeometry):
+    """(total, spatial, temporal) mean plaquette Re tr P / 3
+    (ref: lib/gauge_plaq.cu)."""
+    U = _to_lex(u, geo)
+    tot_s = tot_t = 0.0
+    n_s = n_t = 0
+    for mu in range(4):
+        for nu in range(mu + 1, 4):
+            Unu_xmu = _shift(U[nu], geo, mu, +1)
+            Umu_xnu = _shift(U[mu], geo, nu, +1)
+            P = U[mu] @ Unu_xmu @ Umu_xnu.conj().mT @ U[nu].conj().mT
+            val = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).real.mean().item() / 3.0
+            if nu == 3:
+                tot_t += val
+                n_t += 1
+            else:
+                tot_s += val
+                n_s += 1
+    sp = tot_s / n_s
+    tm = tot_t / n_t
+    return ((sp * n_s + tm * n_t) / (n_s + n_t), sp, tm)
+
+
+def staple_sum(U: torch.Tensor, geo: LatticeGeometry, mu: int) -> torch.Tensor:
+    """Sum of the 6 staples around U_mu (lex layout input [4,Vlex,3,3]).
+
+    staple(nu,+) = U_nu(x) U_mu(x+nu) U_nu(x+mu)^d
+    staple(nu,-) = U_nu(x-nu)^d U_mu(x-nu) U_nu(x+mu-nu)
+    normalized so Re tr[U_mu(x) staple^d] = sum of the Re-traces of the 6
+    plaquettes containing U_mu(x).
+    """
+    S = torch.zeros_like(U[mu])
+    for nu in range(4):
+        if nu == mu:
+            continue
+        Unu_xmu = _shift(U[nu], geo, mu, +1)
+        Umu_xnu = _shift(U[mu], geo, nu, +1)
+        S = S + U[nu] @ Umu_xnu @ Unu_xmu.conj().mT
+        Unu_mnu = _shift(U[nu], geo, nu, -1)
+        Umu_mnu = _shift(U[mu], geo, nu, -1)
+        Unu_xmu_mnu = _shift(Unu_mnu, geo, mu, +1)
+        S = S + Unu_mnu.conj().mT @ Umu_mnu @ Unu_xmu_mnu
+    return S
+
+
+def gauge_action(u: torch.Tensor, geo: LatticeGeometry, beta: float) -> float:
+    """Wilson gauge action S = beta * sum_P (1 - Re tr P / 3)."""
+    p_tot, _, _ = plaquette(u, geo)
+    n_plaq = 6 * geo.volume
+    return beta * n_plaq * (1.0 - p_tot)
+
+
+def project_ta(M: torch.Tensor) -> torch.Tensor:
+    """Traceless antihermitian part: (M - M^d)/2 - tr(M - M^d)/6."""
+    A = (M - M.conj().mT) / 2.0
+    tr = torch.diagonal(A, dim1=-2, dim2=-1).sum(-1) / 3.0
+    eye = torch.eye(3, dtype=M.dtype, device=M.device)
+    return A - tr[..., None, None] * eye
+
+
+def gauge_force(u: torch.Tensor, geo: LatticeGeometry, beta: float) -> torch.Tensor:
+    """Wilson-action MD force (ref: lib/gauge_force.cu), normalized for
+    OUR Hamiltonian convention H = -sum tr P^2 + S(U), Udot = P U:
+    energy conservation then requires Pdot = F = -(beta/6) TA[U staple^d]
+    (TA = project_ta; derivation: tr(P TA[M]) = Re tr(P M), and
+    dS/dt = -(beta/3) sum Re tr(P U staple^d))."""
+    U = _to_lex(u, geo)
+    F = torch.empty_like(U)
+    for mu in range(4):
+        S = staple_sum(U, geo, mu)
+        F[mu] = -(beta / 6.0) * project_ta(U[mu] @ S.conj().mT)
+    return _from_lex(F, geo)
+
+
+def exp_su3(A: torch.Tensor, scale: float = 1.0) -> torch.Tensor:
+    """exp(scale*A) for [...,3,3] antihermitian A (torch.matrix_exp)."""
+    return torch.matrix_exp(scale * A)
+
+
+def ape_smear(u: torch.Tensor, geo: LatticeGeometry, alpha: float,
+              n_iter: int = 1, spatial_only: bool = False) -> torch.Tensor:
+    """APE smearing: U' = Proj_SU3[(1-alpha) U + (alpha/6) staple]
+    (ref: lib/gauge_ape.cu)."""
+    from ..fields.gauge import project_su3
+    out = u
+    dims = range(3) if spatial_only else range(4)
+    for _ in range(n_iter):
+        U = _to_lex(out, geo)
+        Unew = U.clone()
+        for mu in dims:
+            S = staple_sum(U, geo, mu)
+            n_st = 6 if not spatial_only else 4
+            M = (1 - alpha) * U[mu] + (alpha / n_st) * S
+            Unew[mu] = project_su3(M)
+        out = _from_lex(Unew, geo)
+    return out
+
+
+def stout_smear(u: torch.Tensor, geo: LatticeGeometry, rho: float,
+                n_iter: int = 1) -> torch.Tensor:
+    """Stout smearing U' = exp(rho * TA[U staple^d])^d ... standard:
+    Q = TA[Omega], Omega = U_mu staple^d; U' = exp(-rho Q)... sign per
+    Morningstar-Peardon: U' = exp(i rho Q_herm) U; here with antihermitian
+    Q: U' = exp(rho * TA[staple U^d... ]) — we use
+    U'_mu = exp(rho * TA[S U_mu^d]) U_mu (ref: lib/gauge_stout.cu)."""
+    out = u
+    for _ in range(n_iter):
+        U = _to_lex(out, geo)
+        Unew = U.clone()
+        for mu in range(4):
+            S = staple_sum(U, geo, mu)
+            Q = project_ta(S @ U[mu].conj().mT)
+            Unew[mu] = exp_su3(Q, rho) @ U[mu]
+        out = _from_lex(Unew, geo)
+    return out
+
+
+def wilson_flow(u: torch.Tensor, geo: LatticeGeometry, eps: float,
+                n_steps: int = 1) -> torch.Tensor:
+    """Wilson (gradient) flow, RK3 Luscher scheme
+    (ref: lib/gauge_wilson_flow.cu):
+      W0 = U;  Z_i = eps * grad S(W_i)
+      W1 = exp(1/4 Z0) W0
+      W2 = exp(8/9 Z1 - 17/36 Z0) W1
+      U' = exp(3/4 Z2 - 8/9 Z1 + 17/36 Z0) W2
+    with grad = TA[S W^d] (flow toward smaller action)."""
+    def zmat(U):
+        Z = torch.empty_like(U)
+        for mu in range(4):
+            S = staple_sum(U, geo, mu)
+            Z[mu] = eps * project_ta(S @ U[mu].conj().mT)
+        return Z
+
+    out = u
+    for _ in range(n_steps):
+        W0 = _to_lex(out, geo)
+        Z0 = zmat(W0)
+        W1 = torch.matrix_exp(0.25 * Z0) @ W0
+        Z1 = zmat(W1)
+        W2 = torch.matrix_exp((8.0 / 9.0) * Z1 - (17.0 / 36.0) * Z0) @ W1
+        Z2 = zmat(W2)
+        W3 = torch.matrix_exp(0.75 * Z2 - (8.0 / 9.0) * Z1 + (17.0 / 36.0) * Z0) @ W2
+        out = _from_lex(W3, geo)
+    return out
+
+
+def polyakov_loop(u: torch.Tensor, geo: LatticeGeometry) -> complex:
+    """Mean Polyakov loop (1/3) <tr prod_t U_t(x,t)>
+    (ref: lib/gauge_polyakov_loop.cu)."""
+    U = _to_lex(u, geo)
+    X, Y, Z, T = geo.dims
+    Ut = U[3].reshape(T, Z * Y * X, 3, 3)  # t slowest in lex
+    P = Ut[0]
+    for t in range(1, T):
+        P = P @ Ut[t]
+    tr = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1) / 3.0
+    m = tr.mean()
+    return complex(m.real.item(), m.imag.item())
+
+
+def topological_charge(u: torch.Tensor, geo: LatticeGeometry) -> float:
+    """Field-theoretic Q = (1/32 pi^2) sum_x eps_{munurhosig}
+    tr[F_munu F_rhosig] via the clover-leaf F (ref: lib/gauge_qcharge.cu)."""
+    import math
+    F = field_strength(u, geo)  # hermitian convention {(mu,nu): [2,V,3,3]}
+    # Q = (1/4pi^2) sum tr[F01 F23 - F02 F13 + F03 F12] with hermitian F
+    def trprod(a, b):
+        return torch.einsum("pvij,pvji->", F[a], F[b]).real.item()
+
+    q = trprod((0, 1), (2, 3)) - trprod((0, 2), (1, 3)) + trprod((0, 3), (1, 2))
+    return q / (4.0 * math.pi ** 2)

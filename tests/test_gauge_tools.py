@@ -1,0 +1,133 @@
+"""Gauge-sector tests (analogue of the reference's plaq/su3/gauge_alg/
+heatbath/gauge_path tests)."""
+import math
+
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry
+from quda_amd.gauge import (ape_smear, gauge_action, gauge_force,
+                            heatbath_sweep, hmc_trajectory, leapfrog,
+                            mom_action, overrelax_sweep, plaquette,
+                            polyakov_loop, project_ta, random_momentum,
+                            stout_smear, topological_charge, wilson_flow)
+
+
+@pytest.fixture(scope="module")
+def geo():
+    return LatticeGeometry((4, 4, 4, 4))
+
+
+@pytest.fixture(scope="module")
+def u_rand(geo):
+    # mildly disordered field: random close-ish to unit for stable smearing
+    from quda_amd.fields.gauge import project_su3
+    gen = torch.Generator().manual_seed(81)
+    eye = torch.eye(3, dtype=torch.complex128)
+    m = eye + 0.3 * torch.view_as_complex(
+        torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64))
+    return project_su3(m)
+
+
+def test_plaquette_unit(geo):
+    u = GaugeField(geo, "double").unit_().to_complex()
+    tot, sp, tm = plaquette(u, geo)
+    assert abs(tot - 1.0) < 1e-12 and abs(sp - 1.0) < 1e-12
+
+
+def test_plaquette_gauge_invariance(geo, u_rand):
+    """Plaquette invariant under random gauge transformation."""
+    from quda_amd.fields.gauge import project_su3
+    gen = torch.Generator().manual_seed(82)
+    g = project_su3(torch.view_as_complex(
+        torch.randn((geo.volume, 3, 3, 2), generator=gen, dtype=torch.float64)))
+    from quda_amd.gauge.ops import _from_lex, _to_lex
+    U = _to_lex(u_rand, geo)
+    U2 = torch.empty_like(U)
+    for mu in range(4):
+        idx = geo.neighbor_lex(mu, +1)
+        U2[mu] = g @ U[mu] @ g[idx].conj().mT
+    u2 = _from_lex(U2, geo)
+    p1, _, _ = plaquette(u_rand, geo)
+    p2, _, _ = plaquette(u2, geo)
+    assert abs(p1 - p2) < 1e-12
+
+
+def test_smearing_raises_plaquette(geo, u_rand):
+    p0, _, _ = plaquette(u_rand, geo)
+    for sm in (lambda u: ape_smear(u, geo, 0.5, 2),
+               lambda u: stout_smear(u, geo, 0.1, 2),
+               lambda u: wilson_flow(u, geo, 0.05, 4)):
+        u1 = sm(u_rand)
+        p1, _, _ = plaquette(u1, geo)
+        assert p1 > p0
+        # still SU(3)
+        det = torch.linalg.det(u1.reshape(-1, 3, 3))
+        assert (det - 1).abs().max().item() < 1e-8
+
+
+def test_polyakov_unit(geo):
+    u = GaugeField(geo, "double").unit_().to_complex()
+    p = polyakov_loop(u, geo)
+    assert abs(p - 1.0) < 1e-12
+
+
+def test_qcharge_smooth_field_near_integerlike(geo, u_rand):
+    """Q of a smooth (flowed) field is real and finite; unit field gives 0."""
+    u0 = GaugeField(geo, "double").unit_().to_complex()
+    assert abs(topological_charge(u0, geo)) < 1e-10
+    q = topological_charge(wilson_flow(u_rand, geo, 0.05, 5), geo)
+    assert abs(q) < 5.0  # sane magnitude on a tiny smooth lattice
+
+
+def test_leapfrog_energy_conservation(geo, u_rand):
+    """dH -> 0 as dt -> 0 at fixed trajectory length (validates force
+    normalization + integrator)."""
+    beta = 5.5
+    P = random_momentum(geo, seed=83)
+    dHs = []
+    for n in (10, 20, 40):
+        u1, P1 = leapfrog(u_rand, P, geo, beta, n, 0.5 / n)
+        H0 = mom_action(P) + gauge_action(u_rand, geo, beta)
+        H1 = mom_action(P1) + gauge_action(u1, geo, beta)
+        dHs.append(abs(H1 - H0))
+    assert dHs[2] < dHs[0]
+    # leapfrog is O(dt^2): 4x smaller dt -> ~16x smaller dH
+    assert dHs[2] < dHs[0] / 8
+    assert dHs[2] < 0.5
+
+
+def test_leapfrog_reversibility(geo, u_rand):
+    beta = 5.5
+    P = random_momentum(geo, seed=84)
+    u1, P1 = leapfrog(u_rand, P, geo, beta, 10, 0.05)
+    u2, P2 = leapfrog(u1, -P1, geo, beta, 10, 0.05)
+    assert (u2 - u_rand).abs().max().item() < 1e-10
+    assert (P2 + P).abs().max().item() < 1e-10
+
+
+def test_hmc_trajectory_accepts(geo, u_rand):
+    u1, acc, dH = hmc_trajectory(u_rand, geo, 5.5, n_md=20, tau=0.5, seed=85)
+    assert abs(dH) < 0.5
+
+
+def test_overrelax_preserves_action(geo, u_rand):
+    beta = 5.5
+    s0 = gauge_action(u_rand, geo, beta)
+    u1 = overrelax_sweep(u_rand, geo, beta, seed=86)
+    s1 = gauge_action(u1, geo, beta)
+    assert abs(s1 - s0) < 1e-6 * abs(s0) + 1e-8
+    assert (u1 - u_rand).abs().max().item() > 1e-3  # actually moved
+
+
+def test_heatbath_thermalizes_toward_beta(geo):
+    """From a cold start at moderate beta, heatbath moves the plaquette
+    off 1 but keeps it high; links stay SU(3)."""
+    u = GaugeField(geo, "double").unit_().to_complex()
+    for it in range(3):
+        u = heatbath_sweep(u, geo, 8.0, seed=90 + it)
+    p, _, _ = plaquette(u, geo)
+    assert 0.7 < p < 0.999
+    det = torch.linalg.det(u.reshape(-1, 3, 3))
+    assert (det - 1).abs().max().item() < 1e-8
