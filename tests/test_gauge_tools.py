@@ -108,7 +108,7 @@ def test_leapfrog_reversibility(geo, u_rand):
 
 
 def test_hmc_trajectory_accepts(geo, u_rand):
-    u1, acc, dH = hmc_trajectory(u_rand, geo, 5.5, n_md=20, tau=0.5, seed=85)
+    u1, acc, dH = hmc_trajectory(u_rand, geo, 5.5, n_md=40, tau=0.5, seed=85)
     assert abs(dH) < 0.5
 
 
