@@ -118,7 +118,8 @@ def test_iram_smallest_residuals(small):
 
 
 def test_deflated_cg_fewer_iters(small):
-    geo, g, d = small
+    geo, g, _ = small
+    d = DiracWilsonPC(g, 0.135)  # near-critical: ill-conditioned MdagM
     b = SpinorField(geo, "double", n_parity=1).gaussian_(seed=92)
     x_plain = SpinorField(geo, "double", n_parity=1)
     st0 = cg_solve(d, x_plain, b, tol=1e-10, maxiter=1000)
