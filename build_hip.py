@@ -29,6 +29,7 @@ KERNEL_SOURCES = [
     "dslash_wilson_s.hip",
     "dslash_wilson_h.hip",
     "dslash_staggered.hip",
+    "dslash_dwf.hip",
 ]
 BINDING_SOURCES = ["bindings.cpp"]
 

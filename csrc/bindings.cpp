@@ -24,6 +24,25 @@ BlasField field_of(const at::Tensor &data, const at::Tensor &norm, long Vcb) {
   return BlasField{ptr_or_null(data), ptr_or_null(norm), Vcb};
 }
 
+int chunk_w_of(const at::Tensor &t) {
+  switch (t.scalar_type()) {  // 16B chunks for 24-real sites
+    case at::kDouble: return 2;
+    case at::kFloat: return 4;
+    default: return 8;
+  }
+}
+
+// field view with a site offset into the chunk-stride dimension
+BlasField field_of_off(const at::Tensor &data, const at::Tensor &norm,
+                       long v_stride, long s_offset) {
+  BlasField f{ptr_or_null(data), ptr_or_null(norm), v_stride};
+  if (s_offset && f.data) {
+    f.data = (char *)f.data + (long)s_offset * chunk_w_of(data) * data.element_size();
+    if (f.norm) f.norm = (float *)f.norm + s_offset;
+  }
+  return f;
+}
+
 hipStream_t stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
@@ -44,8 +63,12 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
                           std::vector<at::Tensor> ghost,
                           std::vector<at::Tensor> ghost_nrm,
                           std::vector<int64_t> face_cb, int64_t comm_mask,
-                          int64_t kt, double b_re, double b_im) {
+                          int64_t kt, double b_re, double b_im,
+                          int64_t v_stride, int64_t s_offset) {
+  // v_stride: chunk stride of the spinor fields (Ls*Vcb for 5-d fields);
+  // s_offset: site offset of the 4-d slice being operated on (s*Vcb)
   TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
+  if (v_stride == 0) v_stride = Vcb;
   DslashCall c{};
   c.comm_mask = (int)comm_mask;
   c.kt = (int)kt;
@@ -57,9 +80,9 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
     }
     for (int k = 0; k < 4; ++k) c.face_cb[k] = face_cb[k];
   }
-  c.out = field_of(out, out_n, Vcb);
-  c.in = field_of(in, in_n, Vcb);
-  c.x = field_of(x, x_n, Vcb);
+  c.out = field_of_off(out, out_n, v_stride, s_offset);
+  c.in = field_of_off(in, in_n, v_stride, s_offset);
+  c.x = field_of_off(x, x_n, v_stride, s_offset);
   c.gauge = gauge.data_ptr();
   c.clover = ptr_or_null(clover);
   for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
@@ -229,11 +252,43 @@ static void twist_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
   check_launch("twist_apply");
 }
 
+static void dwf5(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                 at::Tensor in_n, at::Tensor x, at::Tensor x_n,
+                 int64_t Vcb4, int64_t Ls, bool xpay, bool dagger, double a,
+                 double alpha, double beta, double mf, int64_t kind) {
+  Dwf5Call c{};
+  long stride = Vcb4 * Ls;
+  c.out = field_of(out, out_n, stride);
+  c.in = field_of(in, in_n, stride);
+  c.x = field_of(x, x_n, stride);
+  c.Vcb4 = Vcb4;
+  c.Ls = (int)Ls;
+  c.xpay = xpay;
+  c.dagger = dagger;
+  c.a = a;
+  c.alpha = alpha;
+  c.beta = beta;
+  c.mf = mf;
+  c.prec = prec_of(out);
+  c.kind = (int)kind;
+  launch_dwf5(c, stream());
+  check_launch("dwf5");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover) dslash");
+  m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover/-twisted) dslash",
+        py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
+        py::arg("gauge"), py::arg("clover"), py::arg("x"), py::arg("x_n"),
+        py::arg("dims"), py::arg("parity_offset"), py::arg("Vcb"),
+        py::arg("parity"), py::arg("dagger"), py::arg("mode"),
+        py::arg("xpay"), py::arg("a"), py::arg("recon"), py::arg("ghost"),
+        py::arg("ghost_nrm"), py::arg("face_cb"), py::arg("comm_mask"),
+        py::arg("kt"), py::arg("b_re") = 0.0, py::arg("b_im") = 0.0,
+        py::arg("v_stride") = 0, py::arg("s_offset") = 0);
   m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
+  m.def("dwf5", &dwf5, "DWF/Moebius 5th-dim ops (Ds apply / M5 inverse)");
   m.def("blas_op", &blas_op, "fused blas/reduction",
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
         py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),

@@ -1,0 +1,37 @@
+// DWF/Moebius 5th-dimension TU (all precisions)
+#include "dslash_dwf.h"
+#include "launchers.h"
+
+template <typename Prec>
+static void dwf5_t(const Dwf5Call &c, hipStream_t st) {
+  using R = typename Prec::Real;
+  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.out.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
+  SpinorAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.x.Vcb};
+  int blk = 256;
+  R a = (R)c.a, al = (R)c.alpha, be = (R)c.beta, mf = (R)c.mf;
+
+#define QA_D5(XPAY, DAG)                                                      \
+  if (c.kind == 0) {                                                          \
+    long n = c.Vcb4 * c.Ls;                                                   \
+    hipLaunchKernelGGL((k_dslash5<Prec, XPAY, DAG>),                          \
+                       dim3((int)((n + blk - 1) / blk)), dim3(blk), 0, st,    \
+                       out, in, x, c.Vcb4, c.Ls, a, al, be, mf);              \
+  } else {                                                                    \
+    hipLaunchKernelGGL((k_m5inv<Prec, XPAY, DAG>),                            \
+                       dim3((int)((c.Vcb4 + blk - 1) / blk)), dim3(blk), 0,   \
+                       st, out, in, x, c.Vcb4, c.Ls, a, al, be, mf);          \
+  }
+
+  if (c.xpay) { if (c.dagger) QA_D5(true, true) else QA_D5(true, false) }
+  else        { if (c.dagger) QA_D5(false, true) else QA_D5(false, false) }
+#undef QA_D5
+}
+
+void launch_dwf5(const Dwf5Call &c, hipStream_t st) {
+  switch (c.prec) {
+    case 0: dwf5_t<PrecDouble>(c, st); break;
+    case 1: dwf5_t<PrecSingle>(c, st); break;
+    case 2: dwf5_t<PrecHalf>(c, st); break;
+  }
+}

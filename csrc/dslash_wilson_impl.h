@@ -6,9 +6,11 @@
 template <typename Prec, int RECON>
 static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   using R = typename Prec::Real;
-  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.Vcb};
-  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.Vcb};
-  SpinorAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.Vcb};
+  // BlasField.Vcb carries the CHUNK STRIDE (= Ls*Vcb for a 5-d field whose
+  // s-slice is addressed via a pointer offset); c.Vcb is the 4-d cb volume
+  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.out.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
+  SpinorAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.x.Vcb};
   // parity-offset the stencil gauge base: [2][NCH][V][W]
   const auto *gbase = (const typename Prec::Store *)c.gauge +
                       (long)c.parity * GaugeAcc<Prec, RECON>::NCH * c.Vcb * Prec::W;
@@ -76,7 +78,7 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
 
 template <typename Prec>
 static void pack_launch(const PackCall &c, hipStream_t st) {
-  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
   LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
   int blk = 256;
   int grid = (int)((c.Fcb + blk - 1) / blk);

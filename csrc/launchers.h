@@ -123,3 +123,16 @@ struct StagDslashCall {
 };
 void launch_dslash_staggered(const StagDslashCall &c, hipStream_t st);
 void launch_pack_face_stag(const PackCall &c, hipStream_t st);
+
+// ---------------------------------------------------------------------------
+struct Dwf5Call {
+  BlasField out, in, x;  // 5-d single-parity fields (Vcb = chunk stride Ls*Vcb4)
+  long Vcb4;
+  int Ls;
+  bool xpay;
+  bool dagger;
+  double a, alpha, beta, mf;
+  int prec;
+  int kind;  // 0 = dslash5 (Ds apply), 1 = m5inv
+};
+void launch_dwf5(const Dwf5Call &c, hipStream_t st);

@@ -30,23 +30,25 @@ class SpinorField:
     def __init__(self, geo: LatticeGeometry, precision: str = "double",
                  device="cpu", n_parity: int = FULL,
                  data: Optional[torch.Tensor] = None,
-                 norm: Optional[torch.Tensor] = None, nspin: int = 4):
+                 norm: Optional[torch.Tensor] = None, nspin: int = 4,
+                 ls: int = 1):
         self.geo = geo
         self.precision = precision
         self.n_parity = n_parity
         self.nspin = nspin
+        self.ls = ls  # 5th (domain-wall) extent; site index = s*Vcb + x
         self.ncomp = self.nspin * self.ncolor * 2
         from .layout import chunk_width
         w = chunk_width(self.ncomp, precision)
         nch = n_chunks(self.ncomp, precision)
-        shape = (n_parity, nch, geo.volume_cb, w)
+        shape = (n_parity, nch, ls * geo.volume_cb, w)
         if data is not None:
             assert tuple(data.shape) == shape, (data.shape, shape)
             self.data = data
         else:
             self.data = torch.zeros(shape, dtype=DTYPE_OF[precision], device=device)
         if precision == "half":
-            nshape = (n_parity, geo.volume_cb)
+            nshape = (n_parity, ls * geo.volume_cb)
             if norm is not None:
                 assert tuple(norm.shape) == nshape
                 self.norm = norm
@@ -62,7 +64,7 @@ class SpinorField:
 
     @property
     def volume_cb(self) -> int:
-        return self.geo.volume_cb
+        return self.ls * self.geo.volume_cb
 
     @property
     def site_shape(self):
@@ -70,7 +72,8 @@ class SpinorField:
 
     def clone_empty(self, precision: Optional[str] = None) -> "SpinorField":
         return SpinorField(self.geo, precision or self.precision,
-                           self.device, self.n_parity, nspin=self.nspin)
+                           self.device, self.n_parity, nspin=self.nspin,
+                           ls=self.ls)
 
     def copy_(self, src: "SpinorField") -> "SpinorField":
         """Any-precision copy (ref: lib/copy_color_spinor_*.cu)."""
@@ -123,7 +126,7 @@ class SpinorField:
         return SpinorField(self.geo, self.precision, device, self.n_parity,
                            data=self.data.to(device),
                            norm=None if self.norm is None else self.norm.to(device),
-                           nspin=self.nspin)
+                           nspin=self.nspin, ls=self.ls)
 
     def parity_view(self, parity: int) -> "SpinorField":
         """Zero-copy single-parity view of a full field (ref Even()/Odd())."""
@@ -131,7 +134,7 @@ class SpinorField:
         return SpinorField(self.geo, self.precision, self.device, PARITY,
                            data=self.data[parity:parity + 1],
                            norm=None if self.norm is None else self.norm[parity:parity + 1],
-                           nspin=self.nspin)
+                           nspin=self.nspin, ls=self.ls)
 
     def __repr__(self):
         return (f"SpinorField({self.geo.dims}, {self.precision}, "

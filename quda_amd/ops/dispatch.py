@@ -257,3 +257,70 @@ def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
         res = a * x.to_complex()[0] + res
     out.from_complex(res.unsqueeze(0))
     return out
+
+
+def dwf5_op(out: SpinorField, inp: SpinorField, alpha: float, beta: float,
+            mf: float, kind: int, dagger: bool = False, a: float = 1.0,
+            x: Optional[SpinorField] = None):
+    """5th-dimension ops on single-parity 5-d fields (csrc/dslash_dwf.h):
+    kind=0: out = [a*x +] alpha*in + beta*(Ds in)
+    kind=1: out = [x +] a * (alpha + beta*Ds)^{-1} in"""
+    Ls = inp.ls
+    xpay = x is not None
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        xf = x if x is not None else out
+        ext.dwf5(out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+                 xf.data, norm_or_empty(xf), inp.geo.volume_cb, Ls,
+                 xpay, bool(dagger), float(a), float(alpha), float(beta),
+                 float(mf), kind)
+        return out
+    psi = inp.to_complex()[0]
+    if kind == 0:
+        res = ref.dslash5(psi, Ls, alpha, beta, mf, dagger)
+        if xpay:
+            res = a * x.to_complex()[0] + res
+    else:
+        res = a * ref.m5inv(psi, Ls, alpha, beta, mf, dagger)
+        if xpay:
+            res = x.to_complex()[0] + res
+    out.from_complex(res.unsqueeze(0))
+    return out
+
+
+def dslash_wilson_slice(out: SpinorField, inp: SpinorField, gauge: GaugeField,
+                        parity: int, s: int, dagger: bool = False,
+                        a: float = 1.0, x: Optional[SpinorField] = None):
+    """4-d Wilson hop on the s-th slice of 5-d fields (out/in/x all 5-d
+    single-parity): out[s] = [x[s] +] a * (D in[s])."""
+    from ..parallel import comms
+    geo = out.geo
+    mask = comms.comm_mask()
+    if mask:
+        raise NotImplementedError(
+            "5-d halo exchange not wired yet (per-slice pack offsets)")
+    xpay = x is not None
+    Vcb = geo.volume_cb
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        xf = x if x is not None else out
+        cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)
+        ext.dslash_wilson(
+            out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+            gauge.data, cl_t, xf.data, norm_or_empty(xf),
+            list(geo.dims), geo.parity_offset, Vcb, parity, bool(dagger),
+            PLAIN, xpay, float(a), RECON_COMPS[gauge.reconstruct], [], [],
+            [], 0, 0, 0.0, 0.0, inp.volume_cb, s * Vcb)
+        return out
+    u = gauge.to_complex()
+    sl = slice(s * Vcb, (s + 1) * Vcb)
+    psi = inp.to_complex()[0][sl]
+    res = ref.dslash_wilson_parity(u, psi, geo, parity, dagger)
+    if xpay:
+        res = x.to_complex()[0][sl] + a * res
+    else:
+        res = a * res
+    oc = out.to_complex()
+    oc[0][sl] = res
+    out.from_complex(oc)
+    return out

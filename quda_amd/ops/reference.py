@@ -282,3 +282,63 @@ def dslash_staggered_full(u, psi_full, geo):
 def mat_staggered(u, psi_full, geo, mass: float):
     """M psi = 2m psi + D psi (mass normalization; D antihermitian)."""
     return 2.0 * mass * psi_full + dslash_staggered_full(u, psi_full, geo)
+
+
+# ---------------------------------------------------------------------------
+# Domain-wall / Moebius 5th-dimension oracle (ref: tests/host_reference/
+# domain_wall_dslash_reference.cpp — re-derived; P+ = spins 0,1 and
+# P- = spins 2,3 in DeGrand-Rossi, so s-hops are spin-diagonal)
+# ---------------------------------------------------------------------------
+
+def dslash5(psi5: torch.Tensor, Ls: int, alpha: float, beta: float,
+            mf: float, dagger: bool = False) -> torch.Tensor:
+    """[Ls*V, 4, 3] -> alpha psi + beta (Ds psi); Ds hops upper spins from
+    s-1 and lower from s+1 (swapped for dagger), with -mf boundary wraps."""
+    V = psi5.shape[0] // Ls
+    v = psi5.reshape(Ls, V, 4, 3)
+    if not dagger:
+        up_from = torch.roll(v, shifts=1, dims=0).clone()   # s-1 -> s
+        up_from[0] = -mf * v[Ls - 1]
+        dn_from = torch.roll(v, shifts=-1, dims=0).clone()  # s+1 -> s
+        dn_from[Ls - 1] = -mf * v[0]
+    else:
+        up_from = torch.roll(v, shifts=-1, dims=0).clone()
+        up_from[Ls - 1] = -mf * v[0]
+        dn_from = torch.roll(v, shifts=1, dims=0).clone()
+        dn_from[0] = -mf * v[Ls - 1]
+    out = alpha * v.clone()
+    out[:, :, 0:2, :] += beta * up_from[:, :, 0:2, :]
+    out[:, :, 2:4, :] += beta * dn_from[:, :, 2:4, :]
+    return out.reshape(Ls * V, 4, 3)
+
+
+def _m5_matrix(Ls, alpha, beta, mf, upper: bool, dagger: bool):
+    """Dense [Ls, Ls] chirality-block matrix of alpha + beta Ds."""
+    import numpy as np
+    A = alpha * np.eye(Ls)
+    for s in range(Ls):
+        src = s - 1 if (upper != dagger) else s + 1
+        w = 1.0
+        if src < 0:
+            src += Ls
+            w = -mf
+        if src >= Ls:
+            src -= Ls
+            w = -mf
+        A[s, src] += beta * w
+    return A
+
+
+def m5inv(psi5: torch.Tensor, Ls: int, alpha: float, beta: float, mf: float,
+          dagger: bool = False) -> torch.Tensor:
+    """(alpha + beta Ds)^{-1} psi via dense per-chirality [Ls,Ls] solves."""
+    import numpy as np
+    V = psi5.shape[0] // Ls
+    v = psi5.reshape(Ls, V, 4, 3)
+    out = torch.empty_like(v)
+    for upper, sl in ((True, slice(0, 2)), (False, slice(2, 4))):
+        A = _m5_matrix(Ls, alpha, beta, mf, upper, dagger)
+        Ainv = torch.tensor(np.linalg.inv(A), dtype=psi5.dtype,
+                            device=psi5.device)
+        out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", Ainv, v[:, :, sl, :])
+    return out.reshape(Ls * V, 4, 3)

@@ -25,7 +25,7 @@ def bicgstab_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, r0, p, v, t = new(), new(), new(), new(), new()
     # r = b - M x
@@ -90,7 +90,7 @@ def bicgstabl_solve(op, x: SpinorField, b: SpinorField, *, L: int = 2,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r = [new() for _ in range(L + 1)]
     u = [new() for _ in range(L + 1)]

@@ -42,7 +42,7 @@ def ca_cg_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     s = basis_size
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, tmp = new(), new()
     V = [new() for _ in range(s + 1)]  # V[s] = A V[s-1] completes AV
@@ -97,7 +97,7 @@ def ca_gcr_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     s = basis_size
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r = new()
     V = [new() for _ in range(s + 1)]

@@ -45,13 +45,13 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
     prec_hi = x.precision
     sloppy = sloppy or prec_hi
     op_sloppy = op_sloppy or op
-    geo, dev, npar, nsp = x.geo, x.device, x.n_parity, x.nspin
+    geo, dev, npar, nsp, nls = x.geo, x.device, x.n_parity, x.nspin, x.ls
 
     def hi():
-        return SpinorField(geo, prec_hi, dev, npar, nspin=nsp)
+        return SpinorField(geo, prec_hi, dev, npar, nspin=nsp, ls=nls)
 
     def lo():
-        return SpinorField(geo, sloppy, dev, npar, nspin=nsp)
+        return SpinorField(geo, sloppy, dev, npar, nspin=nsp, ls=nls)
 
     b2 = blas.norm2(b)
     if b2 == 0.0:

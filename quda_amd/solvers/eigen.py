@@ -27,7 +27,7 @@ class EigResult:
 
 
 def _new_like(x: SpinorField) -> SpinorField:
-    return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+    return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
 
 class ChebyshevOp:

@@ -27,7 +27,7 @@ def gcr_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r = new()
     x2 = blas.norm2(x)
@@ -96,7 +96,7 @@ def mr_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, Ar = new(), new()
     if zero_init:

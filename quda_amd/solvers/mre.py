@@ -25,7 +25,7 @@ class ChronoForecaster:
         self.basis.clear()
 
     def append(self, x: SpinorField):
-        keep = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        keep = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
         blas.copy(keep, x)
         self.basis.append(keep)
         if len(self.basis) > self.max_dim:
@@ -39,9 +39,9 @@ class ChronoForecaster:
             x.zero_()
             return x
         Aps = []
-        tmp = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        tmp = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
         for p in self.basis:
-            Ap = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+            Ap = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
             if use_mdagm:
                 op.MdagM(Ap, p, tmp)
             else:

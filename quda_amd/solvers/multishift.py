@@ -37,7 +37,7 @@ def multishift_cg_solve(op, xs: List[SpinorField], b: SpinorField,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(b.geo, b.precision, b.device, b.n_parity, nspin=b.nspin)
+        return SpinorField(b.geo, b.precision, b.device, b.n_parity, nspin=b.nspin, ls=b.ls)
 
     r, Ap, tmp = new(), new(), new()
     ps = [new() for _ in range(n)]

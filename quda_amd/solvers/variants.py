@@ -31,14 +31,14 @@ class _NormalOp:
 
 def cgnr_solve(op, x: SpinorField, b: SpinorField, **kw) -> SolverStats:
     """CG on the normal equations MdagM x = Mdag b (ref: inv_cgnr.cpp)."""
-    bp = SpinorField(b.geo, b.precision, b.device, b.n_parity, nspin=b.nspin)
+    bp = SpinorField(b.geo, b.precision, b.device, b.n_parity, nspin=b.nspin, ls=b.ls)
     op.M(bp, b, dagger=True)
     return cg_solve(_NormalOp(op), x, bp, **kw)
 
 
 def cgne_solve(op, x: SpinorField, b: SpinorField, **kw) -> SolverStats:
     """CG on M Mdag y = b, x = Mdag y (ref: inv_cgne.cpp)."""
-    y = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+    y = SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
     stats = cg_solve(_NormalOp(op, mmdag=True), y, b, **kw)
     op.M(x, y, dagger=True)
     return stats
@@ -56,7 +56,7 @@ def sd_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, Ar, tmp = new(), new(), new()
     x2 = blas.norm2(x)
@@ -96,7 +96,7 @@ def pcg_solve(op, x: SpinorField, b: SpinorField, *,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, z, p, Ap, tmp = new(), new(), new(), new(), new()
     x2 = blas.norm2(x)
@@ -144,7 +144,7 @@ def cg3_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     stop = tol * tol * b2
 
     def new():
-        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin)
+        return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, Ar, tmp = new(), new(), new()
     x_prev, r_prev = new(), new()

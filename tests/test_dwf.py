@@ -1,0 +1,159 @@
+"""Domain-wall / Moebius tests (analogue of the reference's DWF coverage
+in dslash_ctest + invert_test for dslash_type domain_wall/mobius)."""
+import math
+
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.models.dwf import (DiracDomainWall, DiracDomainWallPC,
+                                 DiracMobius, DiracMobiusPC)
+from quda_amd.ops import blas
+from quda_amd.ops import reference as ref
+from quda_amd.ops.dispatch import dwf5_op
+from quda_amd.solvers import cgnr_solve, cg_solve
+
+MF, M5, LS = 0.04, 1.8, 6
+
+
+@pytest.fixture(scope="module")
+def setup():
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=101)
+    return geo, g
+
+
+def spin5(geo, seed, n_parity=2):
+    return SpinorField(geo, "double", n_parity=n_parity, ls=LS).gaussian_(seed=seed)
+
+
+def test_m5inv_oracle_inverts(setup):
+    geo, g = setup
+    psi = spin5(geo, 102, n_parity=1).to_complex()[0]
+    for dagger in (False, True):
+        y = ref.m5inv(psi, LS, 1.3, -0.4, MF, dagger)
+        back = ref.dslash5(y, LS, 1.3, -0.4, MF, dagger)
+        assert (back - psi).abs().max().item() < 1e-10
+
+
+def test_dslash5_dagger_adjoint(setup):
+    geo, g = setup
+    a = spin5(geo, 103, n_parity=1).to_complex()[0]
+    b = spin5(geo, 104, n_parity=1).to_complex()[0]
+    Da = ref.dslash5(a, LS, 0.7, -0.3, MF, dagger=False)
+    Ddag_b = ref.dslash5(b, LS, 0.7, -0.3, MF, dagger=True)
+    lhs = (b.conj() * Da).sum()
+    rhs = (Ddag_b.conj() * a).sum()
+    assert abs(lhs - rhs) < 1e-10 * abs(lhs)
+
+
+def test_mobius_M_dagger_adjoint(setup):
+    geo, g = setup
+    d = DiracMobius(g, MF, M5, LS)
+    a = spin5(geo, 105)
+    b = spin5(geo, 106)
+    Ma = SpinorField(geo, "double", ls=LS)
+    Mdb = SpinorField(geo, "double", ls=LS)
+    d.M(Ma, a)
+    d.M(Mdb, b, dagger=True)
+    lhs = (b.to_complex().conj() * Ma.to_complex()).sum()
+    rhs = (Mdb.to_complex().conj() * a.to_complex()).sum()
+    assert abs(lhs - rhs) < 1e-10 * abs(lhs)
+
+
+def test_dwf_pc_vs_full_solve(setup):
+    geo, g = setup
+    full = DiracDomainWall(g, MF, M5, LS)
+    pc = DiracDomainWallPC(g, MF, M5, LS)
+    b = spin5(geo, 107)
+    x_full = SpinorField(geo, "double", ls=LS)
+    st = cgnr_solve(full, x_full, b, tol=1e-10, maxiter=3000)
+    assert st.converged
+    be = pc.prepare(b)
+    xe = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st2 = cgnr_solve(pc, xe, be, tol=1e-11, maxiter=3000)
+    assert st2.converged
+    x_rec = SpinorField(geo, "double", ls=LS)
+    pc.reconstruct(x_rec, xe, b)
+    err = (x_rec.to_complex() - x_full.to_complex()).abs().max().item()
+    assert err < 1e-6, err
+
+
+def test_mobius_pc_vs_full_solve(setup):
+    geo, g = setup
+    full = DiracMobius(g, MF, M5, LS)
+    pc = DiracMobiusPC(g, MF, M5, LS)
+    b = spin5(geo, 108)
+    x_full = SpinorField(geo, "double", ls=LS)
+    st = cgnr_solve(full, x_full, b, tol=1e-10, maxiter=4000)
+    assert st.converged
+    be = pc.prepare(b)
+    xe = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st2 = cgnr_solve(pc, xe, be, tol=1e-11, maxiter=4000)
+    assert st2.converged
+    x_rec = SpinorField(geo, "double", ls=LS)
+    pc.reconstruct(x_rec, xe, b)
+    err = (x_rec.to_complex() - x_full.to_complex()).abs().max().item()
+    assert err < 1e-6, err
+
+
+# ---------------------------------------------------------------------------
+# GPU numerics
+# ---------------------------------------------------------------------------
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("prec", ["double", "single"])
+@pytest.mark.parametrize("kind,dagger", [(0, False), (0, True), (1, False),
+                                         (1, True)])
+def test_dwf5_gpu_vs_oracle(setup, prec, kind, dagger):
+    geo, _ = setup
+    inp = SpinorField(geo, prec, "cuda", n_parity=1, ls=LS).gaussian_(seed=111)
+    out = SpinorField(geo, prec, "cuda", n_parity=1, ls=LS)
+    dwf5_op(out, inp, 1.9, -0.55, MF, kind=kind, dagger=dagger)
+    psi = inp.to_complex()[0]
+    if kind == 0:
+        expect = ref.dslash5(psi, LS, 1.9, -0.55, MF, dagger)
+    else:
+        expect = ref.m5inv(psi, LS, 1.9, -0.55, MF, dagger)
+    err = (out.to_complex()[0] - expect).abs().max().item()
+    tol = {"double": 1e-11, "single": 1e-4}[prec]
+    assert err < tol, f"{prec} kind={kind} dag={dagger}: {err}"
+
+
+@pytest.mark.gpu
+def test_mobius_M_gpu_vs_cpu(setup):
+    geo, _ = setup
+    gen = torch.Generator().manual_seed(112)
+    from quda_amd.fields.gauge import project_su3
+    m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64)
+    u = project_su3(torch.view_as_complex(m))
+    g_cpu = GaugeField(geo, "double").from_complex(u)
+    g_gpu = GaugeField(geo, "double", "cuda").from_complex(u.cuda())
+    psi = SpinorField(geo, "double", ls=LS).gaussian_(seed=113)
+    psi_g = SpinorField(geo, "double", "cuda", ls=LS)
+    psi_g.from_complex(psi.to_complex().cuda())
+    for dagger in (False, True):
+        out_c = SpinorField(geo, "double", ls=LS)
+        DiracMobius(g_cpu, MF, M5, LS).M(out_c, psi, dagger=dagger)
+        out_g = SpinorField(geo, "double", "cuda", ls=LS)
+        DiracMobius(g_gpu, MF, M5, LS).M(out_g, psi_g, dagger=dagger)
+        err = (out_g.to_complex().cpu() - out_c.to_complex()).abs().max().item()
+        assert err < 1e-11, f"dag={dagger}: {err}"
+
+
+@pytest.mark.gpu
+def test_mobius_pc_cg_gpu(setup):
+    geo, _ = setup
+    geo = LatticeGeometry((8, 8, 8, 8))
+    gen = torch.Generator().manual_seed(114)
+    from quda_amd.fields.gauge import project_su3
+    m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64)
+    u = project_su3(torch.view_as_complex(m)).cuda()
+    g = GaugeField(geo, "double", "cuda").from_complex(u)
+    pc = DiracMobiusPC(g, MF, M5, 8)
+    b = SpinorField(geo, "double", "cuda", n_parity=1, ls=8).gaussian_(seed=115)
+    x = SpinorField(geo, "double", "cuda", n_parity=1, ls=8)
+    st = cgnr_solve(pc, x, b, tol=1e-8, maxiter=2000)
+    assert st.converged
