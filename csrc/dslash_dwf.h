@@ -114,8 +114,8 @@ __global__ __launch_bounds__(256) void k_m5inv(
 #pragma unroll
     for (int k = 6; k < 12; ++k) {
       cplx<R> yl = (si == 0) ? inv_alpha * vin[k]
-                             : inv_alpha * (vin[k] - beta * prev_l[k]);
-      prev_l[k] = yl;
+                             : inv_alpha * (vin[k] - beta * prev_l[k - 6]);
+      prev_l[k - 6] = yl;
       vin[k] = yl;
     }
     out.store_v(vin, (long)s * Vcb4 + g);
