@@ -131,7 +131,7 @@ __global__ __launch_bounds__(256) void k_m5inv(
     int s = DAG ? (Ls - 1 - si) : si;
     out.load_v(vin, (long)s * Vcb4 + g);
 #pragma unroll
-    for (int k = 6; k < 12; ++k) vin[k] = vin[k] + (cu * kp) * yl_last[k];
+    for (int k = 6; k < 12; ++k) vin[k] = vin[k] + (cu * kp) * yl_last[k - 6];
     if constexpr (XPAY) {
       cplx<R> xv[12];
       x.load_v(xv, (long)s * Vcb4 + g);
