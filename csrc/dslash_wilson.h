@@ -116,7 +116,7 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
 #pragma unroll
     for (int c = 0; c < 3; ++c) acc[s][c] = {(R)0, (R)0};
 
-  cplx<R> p[4][3], h[2][3], uh[2][3], U[3][3];
+  cplx<R> p[4][3], h[8][2][3], uh[2][3], U[3][3];
   // proj/recon tables encode 2P (generate_proj.py); the stencil needs P
   const R one = (R)0.5;
 
@@ -126,44 +126,69 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
     for (int m = 0; m < 4; ++m)
       if (gh.active(m) && (xc[m] == 0 || xc[m] == d.X[m] - 1)) bnd = true;
   }
+  bool skip[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) skip[k] = false;
 
-#define QA_DIR(MU)                                                        \
+  // Phase 1: gather + spin-project all 8 neighbor half-spinors first — the
+  // loads are independent, so the wave has 8 spinor fetches in flight
+  // instead of a serialized load->project->multiply chain per direction
+  // (the dslash is latency-bound at ~30% of HBM peak otherwise; measured
+  // FETCH_SIZE is already near-minimal).
+#define QA_GATHER(MU)                                                     \
   {                                                                       \
     bool cross_p = KT != KT_LOCAL && gh.active(MU) && xc[MU] == d.X[MU] - 1; \
-    if (!(KT == KT_INTERIOR && cross_p)) {                                \
-      if (KT == KT_FUSED && cross_p) {                                    \
-        gh.load(h, MU, 1, ghost_idx(xc, MU, d));                          \
-      } else {                                                            \
-        long j = neighbor_cb(xc, MU, +1, d);                              \
-        in.load(p, j);                                                    \
-        if constexpr (!DAG) proj_##MU##_0(h, p); else proj_##MU##_1(h, p);\
-      }                                                                   \
+    if (KT == KT_INTERIOR && cross_p) {                                   \
+      skip[2 * MU] = true;                                                \
+    } else if (KT == KT_FUSED && cross_p) {                               \
+      gh.load(h[2 * MU], MU, 1, ghost_idx(xc, MU, d));                    \
+    } else {                                                              \
+      long j = neighbor_cb(xc, MU, +1, d);                                \
+      in.load(p, j);                                                      \
+      if constexpr (!DAG) proj_##MU##_0(h[2 * MU], p);                    \
+      else proj_##MU##_1(h[2 * MU], p);                                   \
+    }                                                                     \
+    bool cross_m = KT != KT_LOCAL && gh.active(MU) && xc[MU] == 0;        \
+    if (KT == KT_INTERIOR && cross_m) {                                   \
+      skip[2 * MU + 1] = true;                                            \
+    } else if (KT == KT_FUSED && cross_m) {                               \
+      gh.load(h[2 * MU + 1], MU, 0, ghost_idx(xc, MU, d));                \
+    } else {                                                              \
+      long j = neighbor_cb(xc, MU, -1, d);                                \
+      in.load(p, j);                                                      \
+      if constexpr (!DAG) proj_##MU##_1(h[2 * MU + 1], p);                \
+      else proj_##MU##_0(h[2 * MU + 1], p);                               \
+    }                                                                     \
+  }
+
+  QA_GATHER(0)
+  QA_GATHER(1)
+  QA_GATHER(2)
+  QA_GATHER(3)
+#undef QA_GATHER
+
+  // Phase 2: gauge multiplies + spin reconstruction (site-local gauge)
+#define QA_MUL(MU)                                                        \
+  {                                                                       \
+    if (!skip[2 * MU]) {                                                  \
       g.template load<MU>(U, i);                                          \
-      su3_mul_half(uh, U, h);                                             \
+      su3_mul_half(uh, U, h[2 * MU]);                                     \
       if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                   \
       else recon_##MU##_1(acc, uh, one);                                  \
     }                                                                     \
-    bool cross_m = KT != KT_LOCAL && gh.active(MU) && xc[MU] == 0;        \
-    if (!(KT == KT_INTERIOR && cross_m)) {                                \
-      if (KT == KT_FUSED && cross_m) {                                    \
-        gh.load(h, MU, 0, ghost_idx(xc, MU, d));                          \
-      } else {                                                            \
-        long j = neighbor_cb(xc, MU, -1, d);                              \
-        in.load(p, j);                                                    \
-        if constexpr (!DAG) proj_##MU##_1(h, p); else proj_##MU##_0(h, p);\
-      }                                                                   \
+    if (!skip[2 * MU + 1]) {                                              \
       g.template load<4 + MU>(U, i);                                      \
-      su3_dagmul_half(uh, U, h);                                          \
+      su3_dagmul_half(uh, U, h[2 * MU + 1]);                              \
       if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                   \
       else recon_##MU##_0(acc, uh, one);                                  \
     }                                                                     \
   }
 
-  QA_DIR(0)
-  QA_DIR(1)
-  QA_DIR(2)
-  QA_DIR(3)
-#undef QA_DIR
+  QA_MUL(0)
+  QA_MUL(1)
+  QA_MUL(2)
+  QA_MUL(3)
+#undef QA_MUL
 
   // boundary sites under CLOV_POST defer the whole epilogue: store raw sum
   if constexpr (KT == KT_INTERIOR && MODE == CLOV_POST) {
