@@ -1,7 +1,17 @@
+#include <cstdlib>
 // Shared runtime->template dispatch for the Wilson dslash TUs.
 #pragma once
 #include "dslash_wilson.h"
 #include "launchers.h"
+
+static inline int qa_dslash_block() {
+  static int blk = [] {
+    const char *e = getenv("QUDA_AMD_DSLASH_BLOCK");
+    int v = e ? atoi(e) : 64;  // 1 wave/WG measures best (see profiles/)
+    return (v == 64 || v == 128 || v == 256) ? v : 256;
+  }();
+  return blk;
+}
 
 template <typename Prec, int RECON>
 static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
@@ -24,7 +34,7 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
     gh.nrm[k] = c.ghost_nrm[k];
   }
   for (int k = 0; k < 4; ++k) gh.Fcb[k] = c.face_cb[k];
-  int blk = 256;
+  int blk = qa_dslash_block();
   int grid = (int)((c.Vcb + blk - 1) / blk);
   R a = (R)c.a;
   R br = (R)c.b_re, bi = (R)c.b_im;
