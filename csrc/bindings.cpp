@@ -275,6 +275,10 @@ static void dwf5(at::Tensor out, at::Tensor out_n, at::Tensor in,
   check_launch("dwf5");
 }
 
+static void set_dslash_block(int64_t b) {
+  if (b == 64 || b == 128 || b == 256) qa_dslash_block_ref() = (int)b;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover/-twisted) dslash",
         py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
@@ -288,6 +292,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
+  m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("dwf5", &dwf5, "DWF/Moebius 5th-dim ops (Ds apply / M5 inverse)");
   m.def("blas_op", &blas_op, "fused blas/reduction",
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),

@@ -4,15 +4,6 @@
 #include "dslash_wilson.h"
 #include "launchers.h"
 
-static inline int qa_dslash_block() {
-  static int blk = [] {
-    const char *e = getenv("QUDA_AMD_DSLASH_BLOCK");
-    int v = e ? atoi(e) : 64;  // 1 wave/WG measures best (see profiles/)
-    return (v == 64 || v == 128 || v == 256) ? v : 256;
-  }();
-  return blk;
-}
-
 template <typename Prec, int RECON>
 static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   using R = typename Prec::Real;

@@ -2,6 +2,20 @@
 #pragma once
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
+// runtime-tunable workgroup size (set by the Python autotuner via the
+// set_dslash_block binding; env QUDA_AMD_DSLASH_BLOCK seeds the default)
+inline int &qa_dslash_block_ref() {
+  static int blk = [] {
+    const char *e = getenv("QUDA_AMD_DSLASH_BLOCK");
+    int v = e ? atoi(e) : 64;  // 1 wave/WG measures best (see profiles/)
+    return (v == 64 || v == 128 || v == 256) ? v : 64;
+  }();
+  return blk;
+}
+inline int qa_dslash_block() { return qa_dslash_block_ref(); }
+
 struct BlasField {
   void *data;
   void *norm;  // nullptr unless half

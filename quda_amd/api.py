@@ -1,0 +1,398 @@
+"""quda.h-parity entry points (ref: include/quda.h + lib/interface_quda.cpp
+— the same param-struct + entry-point surface, expressed as Python
+dataclasses/functions over the MI355X-native engine; field layouts accepted
+at the boundary are the oracle complex layouts).
+
+Covers the reference's resident-field model: load_gauge_quda/
+load_clover_quda populate a module-level cache (gaugePrecise + sloppy /
+precondition precision copies, interface_quda.cpp:537), invert_quda builds
+the Dirac operator quadruple and runs the selected solver with mixed
+precision, dslash_quda/mat_quda apply operators, plus observables,
+smearing, eigensolver and multigrid hooks.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from enum import Enum
+from typing import List, Optional
+
+import torch
+
+from .fields.clover import CloverField, pack_clover
+from .fields.gauge import GaugeField
+from .fields.geometry import LatticeGeometry
+from .fields.spinor import SpinorField
+from .models import (DiracClover, DiracCloverPC, DiracStaggered,
+                     DiracStaggeredPC, DiracTwistedClover, DiracTwistedMass,
+                     DiracTwistedMassPC, DiracWilson, DiracWilsonPC)
+from .models.dwf import (DiracDomainWall, DiracDomainWallPC, DiracMobius,
+                         DiracMobiusPC)
+from .ops import blas
+from .ops import reference as ref
+from .solvers import (SolverStats, bicgstab_solve, bicgstabl_solve,
+                      ca_cg_solve, ca_gcr_solve, cg_solve, cgne_solve,
+                      cgnr_solve, gcr_solve, mr_solve, multishift_cg_solve)
+
+
+class DslashType(Enum):
+    WILSON = "wilson"
+    CLOVER = "clover"
+    TWISTED_MASS = "twisted-mass"
+    TWISTED_CLOVER = "twisted-clover"
+    STAGGERED = "staggered"
+    DOMAIN_WALL = "domain-wall"
+    MOBIUS = "mobius"
+
+
+class InverterType(Enum):
+    CG = "cg"
+    CGNE = "cgne"
+    CGNR = "cgnr"
+    BICGSTAB = "bicgstab"
+    BICGSTABL = "bicgstab-l"
+    GCR = "gcr"
+    MR = "mr"
+    CA_CG = "ca-cg"
+    CA_GCR = "ca-gcr"
+
+
+class SolutionType(Enum):
+    MAT = "mat"          # solve M x = b on the full lattice
+    MATPC = "matpc"      # solve the even-odd system directly
+
+
+@dataclass
+class GaugeParam:
+    """ref: QudaGaugeParam quda.h:31 (the fields the engine consumes)."""
+    X: tuple = (8, 8, 8, 8)
+    cuda_prec: str = "double"
+    cuda_prec_sloppy: str = "half"
+    reconstruct: str = "none"
+    reconstruct_sloppy: str = "twelve"
+    anisotropy: float = 1.0
+    device: str = "cuda" if torch.cuda.is_available() else "cpu"
+
+
+@dataclass
+class InvertParam:
+    """ref: QudaInvertParam quda.h:100 (consumed subset)."""
+    dslash_type: DslashType = DslashType.WILSON
+    inv_type: InverterType = InverterType.CG
+    solution_type: SolutionType = SolutionType.MAT
+    kappa: float = 0.135
+    mass: float = 0.05          # staggered
+    mu: float = 0.0             # twisted
+    m5: float = 1.8             # DWF height
+    Ls: int = 8
+    b5: float = 1.5
+    c5: float = 0.5
+    clover_csw: float = 0.0
+    tol: float = 1e-8
+    maxiter: int = 1000
+    reliable_delta: float = 0.1
+    cuda_prec: str = "double"
+    cuda_prec_sloppy: str = "half"
+    preconditioner: Optional[object] = None  # e.g. MG.precond
+    # outputs (ref: out-fields of QudaInvertParam)
+    iter: int = 0
+    true_res: float = 0.0
+    secs: float = 0.0
+    gflops: float = 0.0
+
+
+@dataclass
+class EigParam:
+    """ref: QudaEigParam quda.h:471 (consumed subset)."""
+    n_ev: int = 8
+    n_kr: int = 32
+    tol: float = 1e-8
+    max_restarts: int = 100
+    use_poly_acc: bool = False
+    poly_deg: int = 8
+    a_min: float = 0.1
+    a_max: float = 10.0
+    spectrum: str = "smallest"   # or "largest"
+    use_norm_op: bool = True     # TRLM on MdagM vs IRAM on M
+
+
+# -- resident state (ref: interface_quda.cpp:537 static gauge cache) --------
+
+class _Resident:
+    geo: Optional[LatticeGeometry] = None
+    gauge: Optional[GaugeField] = None
+    gauge_sloppy: Optional[GaugeField] = None
+    clover: Optional[CloverField] = None
+    clover_sloppy: Optional[CloverField] = None
+    gauge_param: Optional[GaugeParam] = None
+    u_complex: Optional[torch.Tensor] = None
+
+
+_R = _Resident()
+
+
+def init_quda(device: int = 0) -> None:
+    """ref: initQuda (device init + comms; comms come from torchrun env)."""
+    if torch.cuda.is_available():
+        torch.cuda.set_device(device)
+    from .parallel import comms
+    comms.init_comms()
+
+
+def end_quda() -> None:
+    free_gauge_quda()
+
+
+def free_gauge_quda() -> None:
+    _R.geo = _R.gauge = _R.gauge_sloppy = None
+    _R.clover = _R.clover_sloppy = None
+    _R.u_complex = None
+
+
+def load_gauge_quda(u: torch.Tensor, param: GaugeParam) -> None:
+    """u: [4, 2, V_cb, 3, 3] complex links (oracle layout) or a GaugeField.
+
+    Creates the precise + sloppy resident copies (ref: loadGaugeQuda
+    interface_quda.cpp:571)."""
+    geo = LatticeGeometry(param.X)
+    if isinstance(u, GaugeField):
+        u = u.to_complex()
+    u = u.to(param.device)
+    _R.geo = geo
+    _R.gauge_param = param
+    _R.u_complex = u
+    _R.gauge = GaugeField(geo, param.cuda_prec, param.device,
+                          reconstruct=param.reconstruct).from_complex(u)
+    _R.gauge_sloppy = GaugeField(
+        geo, param.cuda_prec_sloppy, param.device,
+        reconstruct=param.reconstruct_sloppy).from_complex(u)
+
+
+def load_clover_quda(inv_param: InvertParam,
+                     clover: Optional[torch.Tensor] = None) -> None:
+    """Build (or accept) the clover term for the resident gauge
+    (ref: loadCloverQuda — with clover=None computes it from the resident
+    field like QUDA's compute_clover path)."""
+    assert _R.geo is not None, "load_gauge_quda first"
+    A = clover if clover is not None else ref.clover_matrix(
+        _R.u_complex, _R.geo, inv_param.kappa, inv_param.clover_csw)
+    _R.clover = CloverField(_R.geo, _R.gauge.precision,
+                            _R.gauge.device).from_matrices(A)
+    sp = _R.gauge_sloppy.precision
+    cls = CloverField(_R.geo, sp, _R.gauge.device)
+    cls.data.copy_(cls._to_native(pack_clover(A)))
+    inv = _R.clover.to_complex(inverse=True).to(_R.gauge.device)
+    cls.inv_data.copy_(cls._to_native(pack_clover(inv)))
+    _R.clover_sloppy = cls
+
+
+def _make_dirac(p: InvertParam, sloppy: bool = False):
+    g = _R.gauge_sloppy if sloppy else _R.gauge
+    cl = _R.clover_sloppy if sloppy else _R.clover
+    pc = p.solution_type == SolutionType.MATPC
+    t = p.dslash_type
+    if t == DslashType.WILSON:
+        return DiracWilsonPC(g, p.kappa) if pc else DiracWilson(g, p.kappa)
+    if t == DslashType.CLOVER:
+        assert cl is not None, "load_clover_quda first"
+        return (DiracCloverPC(g, cl, p.kappa) if pc
+                else DiracClover(g, cl, p.kappa))
+    if t == DslashType.TWISTED_MASS:
+        return (DiracTwistedMassPC(g, p.kappa, p.mu) if pc
+                else DiracTwistedMass(g, p.kappa, p.mu))
+    if t == DslashType.TWISTED_CLOVER:
+        assert cl is not None and not pc
+        return DiracTwistedClover(g, cl, p.kappa, p.mu)
+    if t == DslashType.STAGGERED:
+        return (DiracStaggeredPC(g, p.mass) if pc
+                else DiracStaggered(g, p.mass))
+    if t == DslashType.DOMAIN_WALL:
+        return (DiracDomainWallPC(g, p.mass, p.m5, p.Ls) if pc
+                else DiracDomainWall(g, p.mass, p.m5, p.Ls))
+    if t == DslashType.MOBIUS:
+        return (DiracMobiusPC(g, p.mass, p.m5, p.Ls, p.b5, p.c5) if pc
+                else DiracMobius(g, p.mass, p.m5, p.Ls, p.b5, p.c5))
+    raise ValueError(t)
+
+
+def _wrap(v: torch.Tensor, p: InvertParam, n_parity: int) -> SpinorField:
+    geo = _R.geo
+    nspin = 1 if p.dslash_type == DslashType.STAGGERED else 4
+    ls = p.Ls if p.dslash_type in (DslashType.DOMAIN_WALL,
+                                   DslashType.MOBIUS) else 1
+    f = SpinorField(geo, p.cuda_prec, _R.gauge.device, n_parity,
+                    nspin=nspin, ls=ls)
+    f.from_complex(v.to(_R.gauge.device))
+    return f
+
+
+def dslash_quda(inp: torch.Tensor, p: InvertParam, parity: int) -> torch.Tensor:
+    """Apply the parity-hopping dslash (ref: dslashQuda
+    interface_quda.cpp:1709). inp at parity 1-parity."""
+    d = _make_dirac(p)
+    src = _wrap(inp.unsqueeze(0) if inp.dim() == 3 else inp, p, 1)
+    out = src.clone_empty()
+    d.dslash(out, src, parity)
+    return out.to_complex().cpu()[0]
+
+
+def mat_quda(inp: torch.Tensor, p: InvertParam,
+             dagger: bool = False) -> torch.Tensor:
+    """Apply the full operator M (ref: MatQuda)."""
+    d = _make_dirac(p)
+    src = _wrap(inp, p, 2)
+    out = src.clone_empty()
+    d.M(out, src, dagger=dagger)
+    return out.to_complex().cpu()
+
+
+def invert_quda(b: torch.Tensor, p: InvertParam) -> torch.Tensor:
+    """Solve M x = b (ref: invertQuda interface_quda.cpp:2986): builds the
+    precise/sloppy operator pair, runs the selected solver (PC solves go
+    through prepare/reconstruct), fills p.iter/true_res/secs/gflops."""
+    t0 = time.perf_counter()
+    d = _make_dirac(p)
+    b_f = _wrap(b, p, 2)
+    pc_capable = hasattr(d, "prepare")
+
+    if p.solution_type == SolutionType.MATPC and pc_capable:
+        src = d.prepare(b_f)
+        x_e = src.clone_empty()
+        stats = _run_solver(d, x_e, src, p, sloppy_pair=True)
+        x_f = b_f.clone_empty()
+        d.reconstruct(x_f, x_e, b_f)
+    else:
+        x_f = b_f.clone_empty()
+        stats = _run_solver(d, x_f, b_f, p, sloppy_pair=False)
+
+    # true residual on the requested system
+    r = b_f.clone_empty()
+    full = _make_dirac(InvertParam(**{**p.__dict__,
+                                      "solution_type": SolutionType.MAT})) \
+        if p.solution_type == SolutionType.MATPC else d
+    full_b = b_f
+    full.M(r, x_f) if p.solution_type == SolutionType.MATPC else d.M(r, x_f)
+    import math
+    p.true_res = math.sqrt(blas.xmy_norm2(full_b, r) / blas.norm2(full_b))
+    p.iter = stats.iters
+    p.secs = time.perf_counter() - t0
+    vol_factor = _R.geo.volume * (getattr(d, "Ls", 1))
+    p.gflops = (stats.iters * 2 * d.flops_per_site() * vol_factor / 1e9
+                / max(p.secs, 1e-12)) if hasattr(d, "flops_per_site") else 0.0
+    return x_f.to_complex().cpu()
+
+
+def _run_solver(d, x, b, p: InvertParam, sloppy_pair: bool):
+    inv = p.inv_type
+    if inv == InverterType.CG:
+        # CG delivers M x = b via the normal equations (ref solve-type
+        # matrix, lib/solve.cpp: NORMOP solves): rhs <- Mdag b, except for
+        # the staggered PC operator which is hermitian PD itself (its
+        # MdagM hook applies M_pc once).
+        rhs = b
+        if not isinstance(d, DiracStaggeredPC):
+            rhs = b.clone_empty()
+            d.M(rhs, b, dagger=True)
+        if sloppy_pair and p.cuda_prec_sloppy != p.cuda_prec:
+            ds = _make_dirac(p, sloppy=True)
+            return cg_solve(d, x, rhs, op_sloppy=ds,
+                            sloppy=p.cuda_prec_sloppy, tol=p.tol,
+                            maxiter=p.maxiter, delta=p.reliable_delta)
+        return cg_solve(d, x, rhs, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.CGNR:
+        return cgnr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.CGNE:
+        return cgne_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.BICGSTAB:
+        return bicgstab_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.BICGSTABL:
+        return bicgstabl_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.GCR:
+        return gcr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter,
+                         precond=p.preconditioner)
+    if inv == InverterType.MR:
+        return mr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.CA_CG:
+        return ca_cg_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.CA_GCR:
+        return ca_gcr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    raise ValueError(inv)
+
+
+def invert_multishift_quda(b: torch.Tensor, p: InvertParam,
+                           shifts: List[float]) -> List[torch.Tensor]:
+    """ref: invertMultiShiftQuda interface_quda.cpp:3405 — (MdagM+s)x=b on
+    the even-odd system."""
+    assert p.solution_type == SolutionType.MATPC
+    d = _make_dirac(p)
+    b_f = _wrap(b.unsqueeze(0) if b.dim() == 3 else b, p, 1)
+    xs = [b_f.clone_empty() for _ in shifts]
+    st = multishift_cg_solve(d, xs, b_f, shifts, tol=p.tol,
+                             maxiter=p.maxiter)
+    p.iter = st.iters
+    return [x.to_complex().cpu() for x in xs]
+
+
+def eigensolve_quda(p: InvertParam, e: EigParam):
+    """ref: eigensolveQuda interface_quda.cpp:2524."""
+    from .solvers.eigen import ChebyshevOp, iram_solve, trlm_solve
+    d = _make_dirac(p)
+    npar = 1 if p.solution_type == SolutionType.MATPC else 2
+    nspin = 1 if p.dslash_type == DslashType.STAGGERED else 4
+    x0 = SpinorField(_R.geo, p.cuda_prec, _R.gauge.device, npar, nspin=nspin)
+    if e.use_norm_op:
+        poly = (ChebyshevOp(d, e.a_min, e.a_max, e.poly_deg)
+                if e.use_poly_acc else None)
+        res = trlm_solve(d, e.n_ev, e.n_kr, x0, tol=e.tol,
+                         max_restarts=e.max_restarts, poly=poly,
+                         which=e.spectrum)
+    else:
+        res = iram_solve(d, e.n_ev, e.n_kr, x0, tol=e.tol,
+                         max_restarts=e.max_restarts,
+                         which=("smallest_abs" if e.spectrum == "smallest"
+                                else "largest_abs"))
+    return res.evals, [v.to_complex().cpu() for v in res.evecs]
+
+
+def new_multigrid_quda(p: InvertParam, block=(2, 2, 2, 2), n_vec: int = 4,
+                       **kw):
+    """ref: newMultigridQuda interface_quda.cpp:2772. Returns an MG object
+    whose .precond plugs into InvertParam.preconditioner."""
+    from .mg import MG, MGParam
+    d = _make_dirac(InvertParam(**{**p.__dict__,
+                                   "solution_type": SolutionType.MAT}))
+    return MG(d, MGParam(block=block, n_vec=n_vec, **kw))
+
+
+# -- observables / smearing (ref: gauge_observable.cpp entry points) --------
+
+def plaq_quda() -> tuple:
+    from .gauge import plaquette
+    return plaquette(_R.u_complex, _R.geo)
+
+
+def gauge_observables_quda() -> dict:
+    from .gauge import plaquette, polyakov_loop, topological_charge
+    tot, sp, tm = plaquette(_R.u_complex, _R.geo)
+    return {
+        "plaquette": (tot, sp, tm),
+        "polyakov_loop": polyakov_loop(_R.u_complex, _R.geo),
+        "qcharge": topological_charge(_R.u_complex, _R.geo),
+    }
+
+
+def perform_gauge_smear_quda(kind: str, n_steps: int, coeff: float) -> None:
+    """ref: performGaugeSmearQuda / performWFlowQuda — smears the resident
+    field in place."""
+    from .gauge import ape_smear, stout_smear, wilson_flow
+    u = _R.u_complex
+    if kind == "ape":
+        u = ape_smear(u, _R.geo, coeff, n_steps)
+    elif kind == "stout":
+        u = stout_smear(u, _R.geo, coeff, n_steps)
+    elif kind == "wilson_flow":
+        u = wilson_flow(u, _R.geo, coeff, n_steps)
+    else:
+        raise ValueError(kind)
+    load_gauge_quda(u, _R.gauge_param)
