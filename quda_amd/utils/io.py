@@ -1,0 +1,55 @@
+"""Field IO with checksums (ref: lib/vector_io.cpp VectorIO + checksum.cu
+— eigenvector/null-space persistence; torch.save container instead of
+QIO/LIME, xor-fold checksum for integrity)."""
+
+from __future__ import annotations
+
+import hashlib
+from typing import List
+
+import torch
+
+from ..fields.geometry import LatticeGeometry
+from ..fields.spinor import SpinorField
+
+
+def field_checksum(t: torch.Tensor) -> str:
+    """Deterministic content hash (ref checksum.cu CRC role)."""
+    return hashlib.sha256(t.detach().cpu().numpy().tobytes()).hexdigest()[:16]
+
+
+def save_field(path: str, fields: List[SpinorField], meta: dict = None):
+    recs = []
+    for f in fields:
+        c = f.to_complex().cpu()
+        recs.append({"data": c, "dims": f.geo.dims, "nspin": f.nspin,
+                     "ls": f.ls, "n_parity": f.n_parity,
+                     "checksum": field_checksum(c)})
+    torch.save({"fields": recs, "meta": meta or {}}, path)
+
+
+def load_field(path: str, device="cpu", precision="double") -> List[SpinorField]:
+    blob = torch.load(path, weights_only=False)
+    out = []
+    for r in blob["fields"]:
+        if field_checksum(r["data"]) != r["checksum"]:
+            raise IOError(f"checksum mismatch in {path}")
+        geo = LatticeGeometry(r["dims"])
+        f = SpinorField(geo, precision, device, r["n_parity"],
+                        nspin=r["nspin"], ls=r["ls"])
+        f.from_complex(r["data"].to(device))
+        out.append(f)
+    return out
+
+
+def save_gauge(path: str, u: torch.Tensor, meta: dict = None):
+    c = u.detach().cpu()
+    torch.save({"u": c, "checksum": field_checksum(c), "meta": meta or {}},
+               path)
+
+
+def load_gauge(path: str, device="cpu") -> torch.Tensor:
+    blob = torch.load(path, weights_only=False)
+    if field_checksum(blob["u"]) != blob["checksum"]:
+        raise IOError(f"checksum mismatch in {path}")
+    return blob["u"].to(device)
