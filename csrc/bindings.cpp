@@ -150,7 +150,8 @@ static void pack_face_stag(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
 }
 
 static void dslash_staggered(at::Tensor out, at::Tensor out_n, at::Tensor in,
-                             at::Tensor in_n, at::Tensor gauge, at::Tensor x,
+                             at::Tensor in_n, at::Tensor gauge,
+                             at::Tensor long_gauge, at::Tensor x,
                              at::Tensor x_n, std::vector<int64_t> dims,
                              int64_t parity_offset, int64_t Vcb,
                              int64_t parity, bool xpay, double a, double b,
@@ -164,6 +165,7 @@ static void dslash_staggered(at::Tensor out, at::Tensor out_n, at::Tensor in,
   c.in = field_of(in, in_n, Vcb);
   c.x = field_of(x, x_n, Vcb);
   c.gauge = gauge.data_ptr();
+  c.long_gauge = ptr_or_null(long_gauge);
   for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
   c.parity_offset = (int)parity_offset;
   c.Vcb = Vcb;

@@ -38,9 +38,21 @@ __device__ __forceinline__ void su3_dagmul_vec(cplx<R> out[3], const cplx<R> u[3
   }
 }
 
-template <typename Prec, int RECON, bool XPAY, int KT = KT_LOCAL>
+// 3-hop neighbor (Naik long links; local periodic wrap)
+__device__ __forceinline__ long neighbor_cb3(const int x[4], int mu, int dir,
+                                             const LatDims &d) {
+  int y[4] = {x[0], x[1], x[2], x[3]};
+  int v = y[mu] + 3 * dir;
+  while (v >= d.X[mu]) v -= d.X[mu];
+  while (v < 0) v += d.X[mu];
+  y[mu] = v;
+  return cb_from_coords(y, d);
+}
+
+template <typename Prec, int RECON, bool XPAY, int KT = KT_LOCAL, bool IMP = false>
 __global__ __launch_bounds__(256) void k_dslash_staggered(
-    StagAcc<Prec> out, StagAcc<Prec> in, GaugeAcc<Prec, RECON> g, LatDims d,
+    StagAcc<Prec> out, StagAcc<Prec> in, GaugeAcc<Prec, RECON> g,
+    GaugeAcc<Prec, 18> lng, LatDims d,
     int parity, typename Prec::Real a, typename Prec::Real b,
     StagAcc<Prec> x, GhostAcc<Prec, 6> gh) {
   using R = typename Prec::Real;
@@ -88,6 +100,30 @@ __global__ __launch_bounds__(256) void k_dslash_staggered(
   QA_SDIR(2)
   QA_SDIR(3)
 #undef QA_SDIR
+
+  // Naik 3-hop long-link term (improved staggered; single-rank local wrap,
+  // nFace=3 halos land with the multi-rank HISQ support)
+  if constexpr (IMP) {
+    int esum2 = 0;
+#define QA_LDIR(MU)                                                       \
+    {                                                                     \
+      R eta = (esum2 & 1) ? (R)-1 : (R)1;                                 \
+      in.load_v(p, neighbor_cb3(xc, MU, +1, d));                          \
+      lng.template load<MU>(U, i);                                        \
+      su3_mul_vec(up, U, p);                                              \
+      for (int c = 0; c < 3; ++c) acc[c] += eta * up[c];                  \
+      in.load_v(p, neighbor_cb3(xc, MU, -1, d));                          \
+      lng.template load<4 + MU>(U, i);                                    \
+      su3_dagmul_vec(up, U, p);                                           \
+      for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \
+      esum2 += xc[MU];                                                    \
+    }
+    QA_LDIR(0)
+    QA_LDIR(1)
+    QA_LDIR(2)
+    QA_LDIR(3)
+#undef QA_LDIR
+  }
 
   cplx<R> res[3];
   if constexpr (XPAY) {

@@ -11,6 +11,11 @@ static void stag_launch(const StagDslashCall &c, hipStream_t st) {
   const auto *gbase = (const typename Prec::Store *)c.gauge +
                       (long)c.parity * GaugeAcc<Prec, RECON>::NCH * c.Vcb * Prec::W;
   GaugeAcc<Prec, RECON> g{gbase, c.Vcb};
+  const typename Prec::Store *lbase = nullptr;
+  if (c.long_gauge)
+    lbase = (const typename Prec::Store *)c.long_gauge +
+            (long)c.parity * GaugeAcc<Prec, 18>::NCH * c.Vcb * Prec::W;
+  GaugeAcc<Prec, 18> lng{lbase, c.Vcb};
   LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
   GhostAcc<Prec, 6> gh{};
   gh.mask = c.comm_mask;
@@ -28,8 +33,14 @@ static void stag_launch(const StagDslashCall &c, hipStream_t st) {
   int grid_ext = (int)((n_ext + blk - 1) / blk);
 
 #define QA_SLAUNCH(XPAY, KT)                                                   \
-  hipLaunchKernelGGL((k_dslash_staggered<Prec, RECON, XPAY, KT>), dim3(grid),  \
-                     dim3(blk), 0, st, out, in, g, d, c.parity, a, b, x, gh)
+  if (c.long_gauge)                                                            \
+    hipLaunchKernelGGL((k_dslash_staggered<Prec, RECON, XPAY, KT, true>),      \
+                       dim3(grid), dim3(blk), 0, st, out, in, g, lng, d,       \
+                       c.parity, a, b, x, gh);                                 \
+  else                                                                         \
+    hipLaunchKernelGGL((k_dslash_staggered<Prec, RECON, XPAY, KT, false>),     \
+                       dim3(grid), dim3(blk), 0, st, out, in, g, lng, d,       \
+                       c.parity, a, b, x, gh)
   if (c.kt == 3) {
     hipLaunchKernelGGL((k_dslash_staggered_exterior<Prec, RECON>),
                        dim3(grid_ext), dim3(blk), 0, st, out, in, g, d,

@@ -121,6 +121,7 @@ void launch_twist_apply(const TwistApplyCall &c, hipStream_t st);
 struct StagDslashCall {
   BlasField out, in, x;  // single-parity views
   const void *gauge;     // stencil layout, like DslashCall
+  const void *long_gauge;  // Naik links (stencil layout, shift=3, recon 18); null = naive
   int Xdim[4];
   int parity_offset;
   long Vcb;

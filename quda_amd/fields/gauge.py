@@ -36,10 +36,11 @@ RECON_COMPS = {"none": 18, "twelve": 12}
 class GaugeField:
     def __init__(self, geo: LatticeGeometry, precision: str = "double",
                  device="cpu", reconstruct: str = "none",
-                 data: Optional[torch.Tensor] = None):
+                 data: Optional[torch.Tensor] = None, shift: int = 1):
         self.geo = geo
         self.precision = precision
         self.reconstruct = reconstruct
+        self.shift = shift  # hop distance of the stencil (3 = long links)
         L = RECON_COMPS[reconstruct]
         w = WIDTH_OF[precision]
         assert (8 * L) % w == 0
@@ -58,9 +59,21 @@ class GaugeField:
     def L(self) -> int:
         return RECON_COMPS[self.reconstruct]
 
+    def _nbr_shift(self, p: int, mu: int):
+        """cb index of x - shift*mu (local periodic wrap)."""
+        geo = self.geo
+        if self.shift == 1:
+            return geo.neighbor_cb(p, mu, -1)
+        c = geo.coords_of_cb(p).to(torch.int64).clone()
+        d = geo.dims[mu]
+        c[:, mu] = (c[:, mu] - self.shift) % d
+        X, Y, Z, _ = geo.dims
+        lex = ((c[:, 3] * Z + c[:, 2]) * Y + c[:, 1]) * X + c[:, 0]
+        return geo.cb_of_lex[lex]
+
     def to(self, device) -> "GaugeField":
         return GaugeField(self.geo, self.precision, device, self.reconstruct,
-                          data=self.data.to(device))
+                          data=self.data.to(device), shift=self.shift)
 
     # ------------------------------------------------------------------
     def to_complex(self, dtype=torch.complex128) -> torch.Tensor:
@@ -103,8 +116,8 @@ class GaugeField:
                 fwd = u[mu, p][:, 0:nrows, :]                      # [V,nrows,3]
                 flat[p, :, mu * L:(mu + 1) * L] = torch.view_as_real(
                     fwd).reshape(V, L)
-                bwd_idx = self.geo.neighbor_cb(p, mu, -1).to(dev)
-                bwd = u[mu, 1 - p][bwd_idx][:, 0:nrows, :]
+                bwd_idx = self._nbr_shift(p, mu).to(dev)
+                bwd = u[mu, (1 - p) if self.shift % 2 else p][bwd_idx][:, 0:nrows, :]
                 flat[p, :, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
                     bwd).reshape(V, L)
         # fix up bwd slots on partitioned-dim boundary faces

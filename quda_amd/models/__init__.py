@@ -1,8 +1,10 @@
-from .staggered import DiracStaggered, DiracStaggeredPC
+from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
+                        DiracStaggered, DiracStaggeredPC)
 from .dirac import (Dirac, DiracClover, DiracCloverPC, DiracTwistedClover,
                     DiracTwistedMass, DiracTwistedMassPC, DiracWilson,
                     DiracWilsonPC)
 
 __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracCloverPC", "DiracTwistedMass", "DiracTwistedMassPC",
-           "DiracTwistedClover", "DiracStaggered", "DiracStaggeredPC"]
+           "DiracTwistedClover", "DiracStaggered", "DiracStaggeredPC",
+           "DiracImprovedStaggered", "DiracImprovedStaggeredPC"]

@@ -203,22 +203,28 @@ def apply_clover(out: SpinorField, inp: SpinorField, clover, parity: int,
 
 def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                      parity: int, a: float = 0.0,
-                     b: float = 1.0, x: Optional[SpinorField] = None):
+                     b: float = 1.0, x: Optional[SpinorField] = None,
+                     long_gauge: Optional[GaugeField] = None):
     """Staggered stencil: out = [a*x +] b*(D in); out at `parity`, in at
     the opposite parity (nspin=1 fields). D^dag = -D: pass b=-b for the
-    dagger."""
+    dagger. `long_gauge` (shift=3 stencil field) adds the Naik 3-hop term
+    (improved staggered; single-rank only until nFace=3 halos land)."""
     from ..parallel import comms
     geo = out.geo
     xpay = x is not None
     mask = comms.comm_mask()
+    if long_gauge is not None and mask:
+        raise NotImplementedError("improved staggered needs nFace=3 halos")
     if on_gpu(out, inp):
         ext = hip_ext()
         xf = x if x is not None else out
+        lng = (long_gauge.data if long_gauge is not None
+               else torch.empty(0, dtype=out.data.dtype, device=out.device))
 
         def launch(kt, ghosts=[], nrms=[], face_cb=[]):
             ext.dslash_staggered(
                 out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
-                gauge.data, xf.data, norm_or_empty(xf), list(geo.dims),
+                gauge.data, lng, xf.data, norm_or_empty(xf), list(geo.dims),
                 geo.parity_offset, geo.volume_cb, parity, xpay, float(a),
                 float(b), RECON_COMPS[gauge.reconstruct], ghosts, nrms,
                 face_cb, mask if kt else 0, kt)
@@ -252,7 +258,11 @@ def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
             "u_bwd": {mu: gauge.bwd_ghost(mu, parity)
                       for mu in active_dims(mask)},
         }
-    res = b * ref.dslash_staggered_parity(u, psi, geo, parity, halo=halo)
+    res = ref.dslash_staggered_parity(u, psi, geo, parity, halo=halo)
+    if long_gauge is not None:
+        res = res + ref.dslash_staggered_naik_parity(
+            long_gauge.to_complex(), psi, geo, parity)
+    res = b * res
     if xpay:
         res = a * x.to_complex()[0] + res
     out.from_complex(res.unsqueeze(0))

@@ -342,3 +342,28 @@ def m5inv(psi5: torch.Tensor, Ls: int, alpha: float, beta: float, mf: float,
                             device=psi5.device)
         out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", Ainv, v[:, :, sl, :])
     return out.reshape(Ls * V, 4, 3)
+
+
+def dslash_staggered_naik_parity(n: torch.Tensor, psi: torch.Tensor,
+                                 geo: LatticeGeometry, parity: int) -> torch.Tensor:
+    """Naik 3-hop term: sum_mu eta_mu(x)[N_mu(x) psi(x+3mu)
+    - N_mu(x-3mu)^dag psi(x-3mu)]. n: long links [4,2,V,3,3]."""
+    dev, dt = psi.device, psi.dtype
+    other = 1 - parity
+    out = torch.zeros_like(psi)
+    eta = staggered_phases(geo, parity).to(dev)
+
+    def nbr3(p, mu, disp):
+        c = geo.coords_of_cb(p).to(torch.int64).clone()
+        c[:, mu] = (c[:, mu] + disp) % geo.dims[mu]
+        X, Y, Z, _ = geo.dims
+        lex = ((c[:, 3] * Z + c[:, 2]) * Y + c[:, 1]) * X + c[:, 0]
+        return geo.cb_of_lex[lex].to(dev)
+
+    for mu in range(4):
+        e = eta[:, mu].to(dt).unsqueeze(-1)
+        fwd = nbr3(parity, mu, +3)
+        out += e * torch.einsum("vij,vj->vi", n[mu, parity], psi[fwd])
+        bwd = nbr3(parity, mu, -3)
+        out -= e * torch.einsum("vji,vj->vi", n[mu, other][bwd].conj(), psi[bwd])
+    return out
