@@ -10,9 +10,13 @@ from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
                   stout_smear, topological_charge, wilson_flow)
 from .hmc import hmc_trajectory, leapfrog, mom_action, random_momentum
 from .heatbath import heatbath_sweep, overrelax_sweep
+from .fermion_force import (fermion_action_and_force, hmc_trajectory_2f,
+                            pseudofermion_refresh, wilson_fermion_force)
 
 __all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
            "project_ta", "exp_su3", "ape_smear", "stout_smear",
            "wilson_flow", "polyakov_loop", "topological_charge",
            "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",
-           "heatbath_sweep", "overrelax_sweep"]
+           "heatbath_sweep", "overrelax_sweep", "wilson_fermion_force",
+           "fermion_action_and_force", "hmc_trajectory_2f",
+           "pseudofermion_refresh"]

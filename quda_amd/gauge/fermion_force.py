@@ -1,0 +1,113 @@
+"""Two-flavor Wilson pseudofermion force + fermionic HMC
+(ref: lib/clover_force.cpp / wilson outer-product pieces,
+kernels/clover_outer_product.cuh — re-derived for OUR conventions:
+H = -sum tr P^2 + S_g + S_f,  Udot = P U,
+S_f = phi^dag (M M^dag)^{-1} phi,  M = 1 - kappa D.
+
+With X = (M M^dag)^{-1} phi and Y = M^dag X:
+  dS_f/dt = -2 Re[X^dag dM/dt Y] = 2 kappa Re sum_{x,mu} tr[ Pdot_mu(x)
+            ( U_mu(x) G1_mu(x) - G2_mu(x) U_mu(x)^dag ) ]
+  G1(x,mu)_{c,a} = sum_s [P(-mu) Y](x+mu)_{s,c} conj(X(x))_{s,a}
+  G2(x,mu)_{c,b} = sum_s [P(+mu) Y](x)_{s,c} conj(X(x+mu))_{s,b}
+so energy conservation fixes  F_f = kappa TA[U G1 - G2 U^dag]
+(TA = project_ta; validated by the finite-difference test)."""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from ..fields.geometry import LatticeGeometry, checkerboard_join, checkerboard_split
+from ..fields.gauge import GaugeField
+from ..fields.spinor import SpinorField
+from ..models import DiracWilson
+from ..ops import blas
+from ..ops.reference import _gamma_tensors
+from ..solvers.cg import cg_solve
+from ..solvers.variants import _NormalOp
+from .ops import _from_lex, _to_lex, exp_su3, gauge_action, gauge_force, project_ta
+from .hmc import mom_action, random_momentum
+
+
+def wilson_fermion_force(u: torch.Tensor, geo: LatticeGeometry, kappa: float,
+                         X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
+    """F_mu(x) = kappa TA[U_mu(x) G1 - G2 U_mu(x)^dag]; X/Y are full-parity
+    oracle fields [2,Vcb,4,3]."""
+    dev, dt = u.device, u.dtype
+    P = _gamma_tensors(dev, dt)  # [mu, 0:minus/1:plus, 4, 4]
+    U = _to_lex(u, geo)
+    Xl = checkerboard_join(X, geo)
+    Yl = checkerboard_join(Y, geo)
+    F = torch.empty_like(U)
+    for mu in range(4):
+        idx = geo.neighbor_lex(mu, +1).to(dev)
+        PmY_xmu = torch.einsum("st,vtc->vsc", P[mu, 0], Yl[idx])
+        PpY = torch.einsum("st,vtc->vsc", P[mu, 1], Yl)
+        G1 = torch.einsum("vsc,vsa->vca", PmY_xmu, Xl.conj())
+        G2 = torch.einsum("vsc,vsb->vcb", PpY, Xl[idx].conj())
+        F[mu] = kappa * project_ta(U[mu] @ G1 - G2 @ U[mu].conj().mT)
+    return _from_lex(F, geo)
+
+
+def pseudofermion_refresh(d: DiracWilson, seed: int) -> SpinorField:
+    """phi = M eta with Gaussian eta (exact heatbath: S_f = |eta|^2)."""
+    eta = d.new_spinor(n_parity=2)
+    eta.gaussian_(seed=seed)
+    phi = d.new_spinor(n_parity=2)
+    d.M(phi, eta)
+    return phi
+
+
+def fermion_action_and_force(u: torch.Tensor, geo: LatticeGeometry,
+                             kappa: float, phi: SpinorField, *,
+                             cg_tol: float = 1e-10, cg_maxiter: int = 2000
+                             ) -> Tuple[float, torch.Tensor]:
+    """S_f = phi^dag (M Mdag)^{-1} phi and its MD force."""
+    g = GaugeField(geo, "double", phi.device).from_complex(u)
+    d = DiracWilson(g, kappa)
+    X = d.new_spinor(n_parity=2)
+    st = cg_solve(_NormalOp(d, mmdag=True), X, phi, tol=cg_tol,
+                  maxiter=cg_maxiter)
+    assert st.converged, "fermion force CG failed"
+    Y = d.new_spinor(n_parity=2)
+    d.M(Y, X, dagger=True)
+    S_f = blas.re_dot(phi, X)
+    F = wilson_fermion_force(u, geo, kappa, X.to_complex(), Y.to_complex())
+    return S_f, F
+
+
+def hmc_trajectory_2f(u: torch.Tensor, geo: LatticeGeometry, beta: float,
+                      kappa: float, *, n_md: int = 20, tau: float = 0.5,
+                      seed: int = 0, cg_tol: float = 1e-10):
+    """One two-flavor Wilson HMC trajectory (leapfrog, Metropolis).
+    Returns (u', accepted, dH)."""
+    P = random_momentum(geo, u.device, seed, u.dtype)
+    gph = GaugeField(geo, "double", u.device).from_complex(u)
+    phi = pseudofermion_refresh(DiracWilson(gph, kappa), seed + 7)
+
+    def total_force(uc):
+        Sf, Ff = fermion_action_and_force(uc, geo, kappa, phi, cg_tol=cg_tol)
+        return gauge_force(uc, geo, beta) + Ff, Sf
+
+    def hamiltonian(uc, Pc):
+        Sf, _ = fermion_action_and_force(uc, geo, kappa, phi, cg_tol=cg_tol)
+        return mom_action(Pc) + gauge_action(uc, geo, beta) + Sf
+
+    dt = tau / n_md
+    H0 = hamiltonian(u, P)
+    uc = u.clone()
+    F, _ = total_force(uc)
+    P = P + 0.5 * dt * F
+    for k in range(n_md):
+        U = _to_lex(uc, geo)
+        U = exp_su3(_to_lex(P, geo), dt) @ U
+        uc = _from_lex(U, geo)
+        F, _ = total_force(uc)
+        P = P + (0.5 if k == n_md - 1 else 1.0) * dt * F
+    H1 = hamiltonian(uc, P)
+    dH = H1 - H0
+    g = torch.Generator().manual_seed(seed + 13)
+    accept = torch.rand(1, generator=g).item() < min(
+        1.0, float(torch.exp(torch.tensor(-dH))))
+    return (uc if accept else u), accept, dH

@@ -1,0 +1,61 @@
+"""Pseudofermion force tests: finite-difference consistency + fermionic
+HMC energy conservation (role of the reference's clover_force_test)."""
+import math
+
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.gauge import gauge_action, mom_action, random_momentum
+from quda_amd.gauge.fermion_force import (fermion_action_and_force,
+                                          hmc_trajectory_2f,
+                                          pseudofermion_refresh)
+from quda_amd.gauge.ops import _from_lex, _to_lex
+from quda_amd.models import DiracWilson
+
+KAPPA, BETA = 0.115, 5.5
+
+
+@pytest.fixture(scope="module")
+def setup():
+    geo = LatticeGeometry((4, 4, 4, 4))
+    from quda_amd.fields.gauge import project_su3
+    gen = torch.Generator().manual_seed(161)
+    eye = torch.eye(3, dtype=torch.complex128)
+    m = eye + 0.3 * torch.view_as_complex(
+        torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64))
+    u = project_su3(m)
+    g = GaugeField(geo, "double").from_complex(u)
+    phi = pseudofermion_refresh(DiracWilson(g, KAPPA), seed=162)
+    return geo, u, phi
+
+
+def test_force_matches_finite_difference(setup):
+    """dS_f/dt along Udot = P U must equal -2 tr(P F_f)."""
+    geo, u, phi = setup
+    P = random_momentum(geo, seed=163)
+    eps = 1e-6
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    Sp, _ = fermion_action_and_force(up, geo, KAPPA, phi, cg_tol=1e-12)
+    Sm, _ = fermion_action_and_force(um, geo, KAPPA, phi, cg_tol=1e-12)
+    dSdt = (Sp - Sm) / (2 * eps)
+    _, F = fermion_action_and_force(u, geo, KAPPA, phi, cg_tol=1e-12)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    # conservation: d/dt(-tr P^2) = -2 tr(P F) must cancel dS/dt
+    assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
+
+
+def test_hmc_2f_energy_conservation(setup):
+    geo, u, phi = setup
+    dHs = []
+    for n in (8, 16):
+        _, _, dH = hmc_trajectory_2f(u, geo, BETA, KAPPA, n_md=n, tau=0.25,
+                                     seed=164, cg_tol=1e-11)
+        dHs.append(abs(dH))
+    assert dHs[1] < dHs[0]  # second-order integrator
+    assert dHs[1] < 1.0
