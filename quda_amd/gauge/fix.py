@@ -14,20 +14,23 @@ from ..fields.geometry import LatticeGeometry
 from .ops import _from_lex, _to_lex, _shift
 
 
-def _polar_unitary(w: torch.Tensor) -> torch.Tensor:
-    """argmax_g Re tr(g w) over U(3), projected back to SU(3) by removing
-    the determinant phase (cube root)."""
-    U, S, Vh = torch.linalg.svd(w)
-    g = (U @ Vh).conj().mT
-    ph = torch.angle(torch.linalg.det(g)) / 3.0
-    return g * torch.exp(-1j * ph)[..., None, None]
-
-
-def _unitary_power(g: torch.Tensor, omega: float) -> torch.Tensor:
-    """g^omega for unitary g via eigendecomposition."""
-    w, V = torch.linalg.eig(g)
-    phase = torch.exp(1j * omega * torch.angle(w))
-    return (V * phase.unsqueeze(-2)) @ torch.linalg.inv(V)
+def _su3_maximize(w: torch.Tensor, hits: int = 3) -> torch.Tensor:
+    """g in SU(3) (locally) maximizing Re tr(g w) via Cabibbo-Marinari
+    SU(2)-subgroup hits (each hit takes the conjugate quaternion of the
+    subgroup projection, the per-subgroup argmax; same machinery as the
+    heatbath, ref kernels/gauge_fix_ovr.cuh)."""
+    from .heatbath import _SUBGROUPS, _su2_embed_mul, _su2_extract
+    g = torch.zeros_like(w)
+    g[...] = torch.eye(3, dtype=w.dtype, device=w.device)
+    W = w.clone()
+    for _ in range(hits):
+        for (i, j) in _SUBGROUPS:
+            a0, a1, a2, a3 = _su2_extract(W, i, j)
+            k = torch.sqrt(a0 ** 2 + a1 ** 2 + a2 ** 2 + a3 ** 2).clamp_min(1e-30)
+            r0, r1, r2, r3 = a0 / k, -a1 / k, -a2 / k, -a3 / k
+            W = _su2_embed_mul(W, r0, r1, r2, r3, i, j)
+            g = _su2_embed_mul(g, r0, r1, r2, r3, i, j)
+    return g
 
 
 def gauge_fix_quality(u: torch.Tensor, geo: LatticeGeometry,
@@ -62,9 +65,7 @@ def gauge_fix_ovr(u: torch.Tensor, geo: LatticeGeometry, *,
             w = torch.zeros((geo.volume, 3, 3), dtype=u.dtype, device=u.device)
             for mu in range(dirs):
                 w = w + U[mu] + _shift(U[mu], geo, mu, -1).conj().mT
-            g = _polar_unitary(w)
-            if omega != 1.0:
-                g = _unitary_power(g, omega)
+            g = _su3_maximize(w)
             eye = torch.eye(3, dtype=u.dtype, device=u.device)
             sel = (parity_lex == p)
             g = torch.where(sel[:, None, None], g, eye)
