@@ -86,14 +86,19 @@ def main():
     ap.add_argument("--sloppy", type=str, default="half",
                     choices=["double", "single", "half"])
     ap.add_argument("--seed", type=int, default=777)
+    ap.add_argument("--device", default=None,
+                    help="override (cpu for harness tests; default cuda:LOCAL_RANK)")
     args = ap.parse_args()
 
     dims = tuple(int(d) for d in args.lattice.split(","))
     rank, world = comms.init_comms()
     assert world == args.gpus or "WORLD_SIZE" not in os.environ, \
         f"world {world} != --gpus {args.gpus}"
-    device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
-    torch.cuda.set_device(device)
+    if args.device:
+        device = args.device
+    else:
+        device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
+        torch.cuda.set_device(device)
 
     geo = LatticeGeometry(dims, parity_offset=comms.parity_offset_of_rank(dims))
     g, cl, gs, cls = setup_fields(geo, device, args.sloppy, args.seed + rank)
@@ -114,19 +119,22 @@ def main():
     for _ in range(args.warmup):
         step()
     comms.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
     comms.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     t1 = time.perf_counter()
 
     elapsed = t1 - t0
     # max over ranks
     if comms.is_distributed():
-        t = torch.tensor([elapsed], dtype=torch.float64)
         import torch.distributed as dist
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
@@ -137,7 +145,7 @@ def main():
     if rank == 0:
         out = {
             "metric": "wilson_clover_cg_gflops",
-            "value": round(gflops, 1),
+            "value": round(gflops, 3),
             "unit": "GFLOPS",
             "n_gpus": world,
             "steps": args.steps,

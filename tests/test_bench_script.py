@@ -1,0 +1,43 @@
+"""The driver contract: bench.py must run multi-rank (weak scaling along T
+with real halo exchange). Exercised here as a 2-process gloo job on a tiny
+CPU lattice — the same code path the round-end 8-GPU RCCL run takes."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_two_ranks_gloo():
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = "29511"
+    procs = []
+    for r in range(2):
+        e = dict(env, RANK=str(r), LOCAL_RANK=str(r), WORLD_SIZE="2")
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(ROOT, "bench.py"), "--gpus", "2",
+             "--steps", "1", "--warmup", "0", "--iters", "2",
+             "--lattice", "4,4,4,8", "--sloppy", "double", "--device", "cpu"],
+            env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    outs = [p.communicate(timeout=300) for p in procs]
+    for p in procs:
+        assert p.returncode == 0, outs
+    line = [l for l in outs[0][0].splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["ms_per_step"] > 0 and rec["value"] > 0
+
+
+def test_bench_single_rank_cpu():
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "1",
+         "--warmup", "0", "--iters", "2", "--lattice", "4,4,4,8",
+         "--sloppy", "double", "--device", "cpu"],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    rec = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][-1])
+    assert rec["n_gpus"] == 1
