@@ -107,9 +107,10 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
 static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
                       at::Tensor in_n, std::vector<int64_t> dims,
                       int64_t parity_offset, int64_t Vcb, int64_t parity,
-                      int64_t mu, int64_t s01, int64_t edge, int64_t Fcb) {
+                      int64_t mu, int64_t s01, int64_t edge, int64_t Fcb,
+                      int64_t v_stride = 0, int64_t s_offset = 0) {
   PackCall c{};
-  c.in = field_of(in, in_n, Vcb);
+  c.in = field_of_off(in, in_n, v_stride ? v_stride : Vcb, s_offset);
   c.dst = dst.data_ptr();
   c.dst_nrm = (float *)ptr_or_null(dst_nrm);
   for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
@@ -291,7 +292,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ghost_nrm"), py::arg("face_cb"), py::arg("comm_mask"),
         py::arg("kt"), py::arg("b_re") = 0.0, py::arg("b_im") = 0.0,
         py::arg("v_stride") = 0, py::arg("s_offset") = 0);
-  m.def("pack_face", &pack_face, "halo face pack (spin-projected)");
+  m.def("pack_face", &pack_face, "halo face pack (spin-projected)",
+        py::arg("dst"), py::arg("dst_nrm"), py::arg("in"), py::arg("in_n"),
+        py::arg("dims"), py::arg("parity_offset"), py::arg("Vcb"),
+        py::arg("parity"), py::arg("mu"), py::arg("s01"), py::arg("edge"),
+        py::arg("Fcb"), py::arg("v_stride") = 0, py::arg("s_offset") = 0);
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");

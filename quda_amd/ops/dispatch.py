@@ -298,34 +298,71 @@ def dwf5_op(out: SpinorField, inp: SpinorField, alpha: float, beta: float,
     return out
 
 
+def dwf_halo_exchange(inp: SpinorField, parity_in: int, dagger: bool):
+    """Pack + exchange all s-slice faces of a 5-d input (blocking; returns
+    the halo object to hand to dslash_wilson_slice, or None when no dim is
+    partitioned)."""
+    from ..parallel import comms
+    mask = comms.comm_mask()
+    if not mask:
+        return None
+    if on_gpu(inp):
+        from ..parallel.halo import get_dwf_halo
+        h = get_dwf_halo(inp.geo, inp.precision, inp.device, mask, inp.ls)
+        h.pack_exchange(hip_ext(), inp, parity_in, bool(dagger))
+        return ("native", mask, h)
+    from ..parallel.halo import exchange_psi5_oracle
+    ghosts = exchange_psi5_oracle(inp.to_complex()[0], inp.geo, parity_in,
+                                  mask, inp.ls)
+    return ("oracle", mask, ghosts)
+
+
 def dslash_wilson_slice(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                         parity: int, s: int, dagger: bool = False,
-                        a: float = 1.0, x: Optional[SpinorField] = None):
+                        a: float = 1.0, x: Optional[SpinorField] = None,
+                        halo=None):
     """4-d Wilson hop on the s-th slice of 5-d fields (out/in/x all 5-d
-    single-parity): out[s] = [x[s] +] a * (D in[s])."""
+    single-parity): out[s] = [x[s] +] a * (D in[s]). `halo` from
+    dwf_halo_exchange(inp, 1-parity, dagger) when dims are partitioned."""
     from ..parallel import comms
     geo = out.geo
     mask = comms.comm_mask()
     if mask:
-        raise NotImplementedError(
-            "5-d halo exchange not wired yet (per-slice pack offsets)")
+        assert halo is not None, "call dwf_halo_exchange first"
     xpay = x is not None
     Vcb = geo.volume_cb
     if on_gpu(out, inp):
         ext = hip_ext()
         xf = x if x is not None else out
         cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)
+        ghosts, nrms, face_cb, kt = [], [], [], 0
+        if mask:
+            kind, hmask, h = halo
+            assert kind == "native" and hmask == mask
+            ghosts, nrms, face_cb = h.ghost_args(s)
+            kt = 1  # fused ghost-aware kernel (exchange already done)
         ext.dslash_wilson(
             out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
             gauge.data, cl_t, xf.data, norm_or_empty(xf),
             list(geo.dims), geo.parity_offset, Vcb, parity, bool(dagger),
-            PLAIN, xpay, float(a), RECON_COMPS[gauge.reconstruct], [], [],
-            [], 0, 0, 0.0, 0.0, inp.volume_cb, s * Vcb)
+            PLAIN, xpay, float(a), RECON_COMPS[gauge.reconstruct], ghosts,
+            nrms, face_cb, mask, kt, 0.0, 0.0, inp.volume_cb, s * Vcb)
         return out
     u = gauge.to_complex()
     sl = slice(s * Vcb, (s + 1) * Vcb)
     psi = inp.to_complex()[0][sl]
-    res = ref.dslash_wilson_parity(u, psi, geo, parity, dagger)
+    oh = None
+    if mask:
+        kind, hmask, ghosts5 = halo
+        assert kind == "oracle" and hmask == mask
+        from ..parallel.halo import active_dims
+        oh = {
+            "mask": mask,
+            "psi": {k: v[s] for k, v in ghosts5.items()},
+            "u_bwd": {mu: gauge.bwd_ghost(mu, parity)
+                      for mu in active_dims(mask)},
+        }
+    res = ref.dslash_wilson_parity(u, psi, geo, parity, dagger, halo=oh)
     if xpay:
         res = x.to_complex()[0][sl] + a * res
     else:

@@ -223,3 +223,114 @@ def exchange_psi_oracle(psi: torch.Tensor, geo: LatticeGeometry,
         recvs[(mu, 1)] = torch.empty_like(sends[(mu, 0)])
     exchange_tensors(sends, recvs)
     return recvs
+
+
+# ---------------------------------------------------------------------------
+# 5-d (domain-wall) halo: per-slice packs into one slab buffer, ONE exchange
+# ---------------------------------------------------------------------------
+
+class DwfHalo:
+    """Ghost buffers for 5-d fields: one slab per (mu,dir) shaped
+    [Ls, 12/gw, Fcb, gw]; each s-slice packs into its slab and the whole
+    buffer ships in a single message per (mu,dir)."""
+
+    def __init__(self, geo: LatticeGeometry, precision: str, device,
+                 mask: int, ls: int):
+        from ..fields.layout import DTYPE_OF
+        self.geo = geo
+        self.precision = precision
+        self.mask = mask
+        self.ls = ls
+        self.device = torch.device(device)
+        gw = ghost_width(12, precision)
+        dt = DTYPE_OF[precision]
+        self.send, self.recv = {}, {}
+        self.send_nrm, self.recv_nrm = {}, {}
+        for mu in active_dims(mask):
+            fcb = geo.face_volume_cb(mu)
+            for d in (0, 1):
+                shape = (ls, 12 // gw, fcb, gw)
+                self.send[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
+                self.recv[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
+                if precision == "half":
+                    self.send_nrm[(mu, d)] = torch.empty((ls, fcb),
+                                                         dtype=torch.float32,
+                                                         device=device)
+                    self.recv_nrm[(mu, d)] = torch.empty((ls, fcb),
+                                                         dtype=torch.float32,
+                                                         device=device)
+
+    def pack_exchange(self, ext, inp, parity: int, dagger: bool) -> None:
+        geo = self.geo
+        Vcb = geo.volume_cb
+        empty = torch.empty(0, dtype=torch.float32, device=self.device)
+        for mu in active_dims(self.mask):
+            fcb = geo.face_volume_cb(mu)
+            for d in (0, 1):
+                s01 = d ^ (1 if dagger else 0)
+                edge = 0 if d == 0 else 1
+                for s in range(self.ls):
+                    nrm = (self.send_nrm[(mu, d)][s]
+                           if self.precision == "half" else empty)
+                    ext.pack_face(self.send[(mu, d)][s], nrm,
+                                  inp.data, _norm_or_empty(inp),
+                                  list(geo.dims), geo.parity_offset, Vcb,
+                                  parity, mu, s01, edge, fcb,
+                                  v_stride=self.ls * Vcb, s_offset=s * Vcb)
+        exchange_tensors(self.send, self.recv)
+        if self.precision == "half":
+            exchange_tensors(self.send_nrm, self.recv_nrm)
+
+    def ghost_args(self, s: int):
+        empty_n = torch.empty(0, dtype=torch.float32, device=self.device)
+        ghosts, nrms = [], []
+        for mu in range(4):
+            for d in (0, 1):
+                g = self.recv.get((mu, d))
+                if g is None:
+                    ghosts.append(torch.empty(0, dtype=next(iter(self.recv.values())).dtype,
+                                              device=self.device))
+                    nrms.append(empty_n)
+                else:
+                    ghosts.append(g[s])
+                    nrms.append(self.recv_nrm[(mu, d)][s]
+                                if self.precision == "half" else empty_n)
+        face_cb = [self.geo.face_volume_cb(mu) for mu in range(4)]
+        return ghosts, nrms, face_cb
+
+
+def _norm_or_empty(f):
+    from ..ops.dispatch import norm_or_empty
+    return norm_or_empty(f)
+
+
+_DWF_HALO_CACHE: Dict[tuple, DwfHalo] = {}
+
+
+def get_dwf_halo(geo: LatticeGeometry, precision: str, device, mask: int,
+                 ls: int) -> DwfHalo:
+    key = (geo.dims, precision, str(device), mask, ls)
+    h = _DWF_HALO_CACHE.get(key)
+    if h is None:
+        h = DwfHalo(geo, precision, device, mask, ls)
+        _DWF_HALO_CACHE[key] = h
+    return h
+
+
+def exchange_psi5_oracle(psi5: torch.Tensor, geo: LatticeGeometry,
+                         parity_in: int, mask: int, ls: int):
+    """Oracle 5-d spinor face exchange: psi5 [Ls*V,4,3] -> ghosts
+    {(mu,dir): [Ls, Fcb, 4, 3]}."""
+    V = geo.volume_cb
+    v = psi5.reshape(ls, V, 4, 3)
+    sends, recvs = {}, {}
+    for mu in active_dims(mask):
+        hi = geo.dims[mu] - 1
+        idx0 = geo.face_index_cb(parity_in, mu, 0).to(psi5.device)
+        idx1 = geo.face_index_cb(parity_in, mu, hi).to(psi5.device)
+        sends[(mu, 0)] = v[:, idx0].contiguous()
+        sends[(mu, 1)] = v[:, idx1].contiguous()
+        recvs[(mu, 0)] = torch.empty_like(sends[(mu, 1)])
+        recvs[(mu, 1)] = torch.empty_like(sends[(mu, 0)])
+    exchange_tensors(sends, recvs)
+    return recvs

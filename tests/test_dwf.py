@@ -157,3 +157,54 @@ def test_mobius_pc_cg_gpu(setup):
     x = SpinorField(geo, "double", "cuda", n_parity=1, ls=8)
     st = cgnr_solve(pc, x, b, tol=1e-8, maxiter=2000)
     assert st.converged
+
+
+# ---------------------------------------------------------------------------
+# 5-d halo exchange
+# ---------------------------------------------------------------------------
+
+def test_mobius_self_wraparound_cpu(setup):
+    from quda_amd.parallel import comms
+    geo, g = setup
+    d = DiracMobius(g, MF, M5, LS)
+    psi = spin5(geo, 181)
+    out_ref = SpinorField(geo, "double", ls=LS)
+    d.M(out_ref, psi)
+    try:
+        comms.set_forced_partition(0b1010)
+        g2 = GaugeField(geo, "double").from_complex(g.to_complex())
+        d2 = DiracMobius(g2, MF, M5, LS)
+        out = SpinorField(geo, "double", ls=LS)
+        d2.M(out, psi)
+    finally:
+        comms.set_forced_partition(0)
+    err = (out.to_complex() - out_ref.to_complex()).abs().max().item()
+    assert err < 1e-12, err
+
+
+@pytest.mark.gpu
+def test_mobius_self_wraparound_gpu(setup):
+    from quda_amd.parallel import comms
+    geo, _ = setup
+    gen = torch.Generator().manual_seed(182)
+    from quda_amd.fields.gauge import project_su3
+    m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64)
+    u = project_su3(torch.view_as_complex(m)).cuda()
+    g = GaugeField(geo, "double", "cuda").from_complex(u)
+    d = DiracMobius(g, MF, M5, LS)
+    psi = SpinorField(geo, "double", "cuda", ls=LS).gaussian_(seed=183)
+    out_ref = SpinorField(geo, "double", "cuda", ls=LS)
+    d.M(out_ref, psi)
+    for dagger in (False, True):
+        d.M(out_ref, psi, dagger=dagger)
+        try:
+            comms.set_forced_partition(0b1111)
+            g2 = GaugeField(geo, "double", "cuda").from_complex(u)
+            d2 = DiracMobius(g2, MF, M5, LS)
+            out = SpinorField(geo, "double", "cuda", ls=LS)
+            d2.M(out, psi, dagger=dagger)
+        finally:
+            comms.set_forced_partition(0)
+        err = (out.to_complex() - out_ref.to_complex()).abs().max().item()
+        assert err < 1e-12, (dagger, err)
