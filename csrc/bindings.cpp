@@ -193,7 +193,8 @@ static void dslash_staggered(at::Tensor out, at::Tensor out_n, at::Tensor in,
 static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
                           at::Tensor x_n, at::Tensor y, at::Tensor y_n,
                           int64_t Vcb, int64_t sites, double c2 = 0.0,
-                          double d2 = 0.0, int64_t ncomp = 24) {
+                          double d2 = 0.0, int64_t ncomp = 24,
+                          bool deterministic = false) {
   BlasCall c{};
   c.op = (int)op;
   c.prec = prec_of(x);
@@ -202,6 +203,7 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
   c.c = c2;
   c.d = d2;
   c.ncomp = (int)ncomp;
+  c.det = deterministic;
   c.x = field_of(x, x_n, Vcb);
   c.y = field_of(y, y_n, Vcb);
   c.sites = sites;
@@ -209,11 +211,18 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
   bool reduction = (op == BLAS_AXPY_NORM2 || op == BLAS_XMY_NORM2 ||
                     op == BLAS_NORM2 || op == BLAS_REDOT || op == BLAS_CDOT);
   if (reduction) {
-    result = at::zeros({2}, x.options().dtype(at::kDouble));
+    if (deterministic) {
+      long g = (sites + 255) / 256;
+      if (g > 2048) g = 2048;
+      result = at::zeros({g, 2}, x.options().dtype(at::kDouble));
+    } else {
+      result = at::zeros({2}, x.options().dtype(at::kDouble));
+    }
     c.result = result.data_ptr<double>();
   }
   launch_blas(c, stream());
   check_launch("blas_op");
+  if (reduction && deterministic) result = result.sum(0);
   return result;
 }
 
@@ -305,7 +314,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
         py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),
         py::arg("sites"), py::arg("c2") = 0.0, py::arg("d2") = 0.0,
-        py::arg("ncomp") = 24);
+        py::arg("ncomp") = 24, py::arg("deterministic") = false);
   m.def("convert", &convert, "precision conversion copy", py::arg("dst"),
         py::arg("dst_n"), py::arg("src"), py::arg("src_n"), py::arg("Vcb"),
         py::arg("sites"), py::arg("ncomp") = 24);

@@ -16,7 +16,13 @@ __device__ __forceinline__ double wave_reduce(double v) {
   return v;
 }
 
-__device__ __forceinline__ void block_atomic_add(double v, double *out) {
+// finish mode: per_block=false -> one f64 atomic per block (fast path);
+// per_block=true -> write partial[blockIdx] for a deterministic ordered
+// host/torch sum (ref: QUDA_DETERMINISTIC_REDUCE, comm_quda.h:204)
+__device__ __forceinline__ void block_atomic_add(double v, double *out,
+                                                 bool per_block = false,
+                                                 int slot_stride = 1,
+                                                 int slot = 0) {
   __shared__ double partial[4];
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
@@ -26,7 +32,12 @@ __device__ __forceinline__ void block_atomic_add(double v, double *out) {
   if (wave == 0) {
     double s = (lane < (int)(blockDim.x >> 6)) ? partial[lane] : 0.0;
     s = wave_reduce(s);
-    if (lane == 0) atomicAdd(out, s);
+    if (lane == 0) {
+      if (per_block)
+        out[(long)blockIdx.x * slot_stride + slot] = s;
+      else
+        atomicAdd(out + slot, s);
+    }
   }
 }
 
@@ -37,7 +48,7 @@ __device__ __forceinline__ void block_atomic_add(double v, double *out) {
 // y = a*x + y ; optional norm2(y) accumulation
 template <typename A, bool NORM2>
 __global__ __launch_bounds__(256) void k_axpy(
-    typename A::R a, A x, A y, long sites, double *result) {
+    typename A::R a, A x, A y, long sites, double *result, bool det) {
   using R = typename A::R;
   double acc = 0.0;
   GRID_STRIDE(g, sites) {
@@ -52,7 +63,7 @@ __global__ __launch_bounds__(256) void k_axpy(
     }
     y.store_v(yv, g);
   }
-  if constexpr (NORM2) block_atomic_add(acc, result);
+  if constexpr (NORM2) block_atomic_add(acc, result, det, 2, 0);
 }
 
 // y = x + a*y
@@ -121,7 +132,7 @@ __global__ __launch_bounds__(256) void k_caxpby(
 // y = x - y ; returns norm2(y)
 template <typename A>
 __global__ __launch_bounds__(256) void k_xmy_norm2(
-    A x, A y, long sites, double *result) {
+    A x, A y, long sites, double *result, bool det) {
   using R = typename A::R;
   double acc = 0.0;
   GRID_STRIDE(g, sites) {
@@ -135,7 +146,7 @@ __global__ __launch_bounds__(256) void k_xmy_norm2(
     }
     y.store_v(yv, g);
   }
-  block_atomic_add(acc, result);
+  block_atomic_add(acc, result, det, 2, 0);
 }
 
 // x *= a
@@ -155,7 +166,7 @@ __global__ __launch_bounds__(256) void k_scal(
 // reductions: norm2, re<x,y>, <x,y> (re+im)
 template <typename A, int KIND>  // 0 norm2, 1 redot, 2 cdot
 __global__ __launch_bounds__(256) void k_reduce(
-    A x, A y, long sites, double *result) {
+    A x, A y, long sites, double *result, bool det) {
   using R = typename A::R;
   double acc = 0.0, acc2 = 0.0;
   GRID_STRIDE(g, sites) {
@@ -174,8 +185,8 @@ __global__ __launch_bounds__(256) void k_reduce(
       }
     }
   }
-  block_atomic_add(acc, result);
-  if constexpr (KIND == 2) block_atomic_add(acc2, result + 1);
+  block_atomic_add(acc, result, det, 2, 0);
+  if constexpr (KIND == 2) block_atomic_add(acc2, result, det, 2, 1);
 }
 
 // precision conversion copy (ref: lib/copy_color_spinor_*.cu)
@@ -219,11 +230,11 @@ static void blas_dispatch(const BlasCall &c, hipStream_t st) {
   switch (c.op) {
     case BLAS_AXPY:
       hipLaunchKernelGGL((k_axpy<A, false>), dim3(gr), dim3(BLK), 0, st,
-                         (R)c.a, x, y, n, nullptr);
+                         (R)c.a, x, y, n, nullptr, false);
       break;
     case BLAS_AXPY_NORM2:
       hipLaunchKernelGGL((k_axpy<A, true>), dim3(gr), dim3(BLK), 0, st,
-                         (R)c.a, x, y, n, c.result);
+                         (R)c.a, x, y, n, c.result, c.det);
       break;
     case BLAS_XPAY:
       hipLaunchKernelGGL((k_xpay<A>), dim3(gr), dim3(BLK), 0, st, x, (R)c.a, y, n);
@@ -242,22 +253,22 @@ static void blas_dispatch(const BlasCall &c, hipStream_t st) {
       break;
     case BLAS_XMY_NORM2:
       hipLaunchKernelGGL((k_xmy_norm2<A>), dim3(gr), dim3(BLK), 0, st, x, y, n,
-                         c.result);
+                         c.result, c.det);
       break;
     case BLAS_SCAL:
       hipLaunchKernelGGL((k_scal<A>), dim3(gr), dim3(BLK), 0, st, (R)c.a, x, n);
       break;
     case BLAS_NORM2:
       hipLaunchKernelGGL((k_reduce<A, 0>), dim3(gr), dim3(BLK), 0, st, x, y, n,
-                         c.result);
+                         c.result, c.det);
       break;
     case BLAS_REDOT:
       hipLaunchKernelGGL((k_reduce<A, 1>), dim3(gr), dim3(BLK), 0, st, x, y, n,
-                         c.result);
+                         c.result, c.det);
       break;
     case BLAS_CDOT:
       hipLaunchKernelGGL((k_reduce<A, 2>), dim3(gr), dim3(BLK), 0, st, x, y, n,
-                         c.result);
+                         c.result, c.det);
       break;
   }
 }

@@ -44,7 +44,8 @@ struct BlasCall {
   BlasField x, y;
   long sites;    // npar * Vcb
   int ncomp;     // reals per site: 24 (Wilson) or 6 (staggered)
-  double *result;  // device ptr for reductions (>=2 doubles for cdot)
+  bool det;      // deterministic reduce: result = per-block partials
+  double *result;  // device ptr (2 doubles, or [grid][2] partials if det)
 };
 
 void launch_blas(const BlasCall &c, hipStream_t st);

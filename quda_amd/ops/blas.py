@@ -33,11 +33,32 @@ def _cvals(x: SpinorField) -> torch.Tensor:
                                                   x.ncomp // 2)
 
 
+_DETERMINISTIC = None
+
+
+def deterministic_reductions() -> bool:
+    """Fixed-order GPU reductions (ref: QUDA_DETERMINISTIC_REDUCE):
+    per-block partials + ordered tree sum instead of f64 atomics. Env
+    QUDA_AMD_DETERMINISTIC_REDUCE=1 or set_deterministic(True)."""
+    global _DETERMINISTIC
+    if _DETERMINISTIC is None:
+        import os
+        _DETERMINISTIC = os.environ.get("QUDA_AMD_DETERMINISTIC_REDUCE",
+                                        "0") == "1"
+    return _DETERMINISTIC
+
+
+def set_deterministic(v: bool) -> None:
+    global _DETERMINISTIC
+    _DETERMINISTIC = bool(v)
+
+
 def _gpu_blas(op: int, a, b, x: SpinorField, y: SpinorField):
     ext = hip_ext()
     return ext.blas_op(op, float(a), float(b), x.data, norm_or_empty(x),
                        y.data, norm_or_empty(y), x.volume_cb,
-                       x.n_parity * x.volume_cb, ncomp=x.ncomp)
+                       x.n_parity * x.volume_cb, ncomp=x.ncomp,
+                       deterministic=deterministic_reductions())
 
 
 # -- reductions -------------------------------------------------------------

@@ -208,3 +208,25 @@ def test_cg_clover_gpu(setup):
     d.MdagM(out, x, tmp)
     r = blas.xmy_norm2(b, out)
     assert r < 1e-14 * blas.norm2(b) * 1e6  # rel residual < 1e-4 in norm2
+
+
+@pytest.mark.gpu
+def test_deterministic_reduce_gpu():
+    """QUDA_DETERMINISTIC_REDUCE analogue: det path bit-stable and
+    consistent with the atomic path."""
+    from quda_amd import LatticeGeometry, SpinorField
+    from quda_amd.ops import blas
+    geo = LatticeGeometry((8, 8, 8, 16))
+    x = SpinorField(geo, "double", "cuda").gaussian_(seed=191)
+    y = SpinorField(geo, "double", "cuda").gaussian_(seed=192)
+    blas.set_deterministic(True)
+    try:
+        r1 = blas.norm2(x)
+        r2 = blas.norm2(x)
+        c1 = blas.c_dot(x, y)
+        c2 = blas.c_dot(x, y)
+        assert r1 == r2 and c1 == c2  # bitwise stable
+    finally:
+        blas.set_deterministic(False)
+    r3 = blas.norm2(x)
+    assert abs(r3 - r1) < 1e-10 * abs(r1)
