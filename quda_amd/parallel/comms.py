@@ -116,7 +116,7 @@ def parity_offset_of_rank(local_dims) -> int:
 def allreduce_sum(x):
     """Global sum of a python scalar (float or complex-as-2-floats caller
     side). The latency-critical per-iteration collective."""
-    if not is_distributed():
+    if not is_distributed() or _solo():
         return x
     dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
     t = torch.tensor([x], dtype=torch.float64, device=dev)
@@ -125,11 +125,39 @@ def allreduce_sum(x):
 
 
 def allreduce_tensor(t: torch.Tensor) -> torch.Tensor:
-    if is_distributed():
+    if is_distributed() and not _solo():
         dist.all_reduce(t)
     return t
 
 
 def barrier():
-    if is_distributed():
+    if is_distributed() and not _solo():
         dist.barrier()
+
+
+# -- communicator stack (ref: lib/communicator_stack.cpp push_communicator:
+# split-grid repartitions ranks into independent sub-grids) ----------------
+
+from contextlib import contextmanager
+
+
+@contextmanager
+def solo_mode():
+    """Push a single-rank communicator: collectives and halo exchange
+    become local no-ops while each rank works on an independent problem
+    (the sub-grid-size-1 split-grid case; general sub-grids are a tracked
+    gap). Restores the outer topology on exit."""
+    saved = dict(_STATE)
+    saved_solo = _STATE.get("solo", False)
+    _STATE.update(grid=(1, 1, 1, 1), coords=(0, 0, 0, 0), forced_mask=0,
+                  solo=True)
+    try:
+        yield
+    finally:
+        _STATE.clear()
+        _STATE.update(saved)
+        _STATE["solo"] = saved_solo
+
+
+def _solo() -> bool:
+    return bool(_STATE.get("solo", False))
