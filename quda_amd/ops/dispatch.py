@@ -40,6 +40,42 @@ def _load_ext():
     return _EXT
 
 
+_TRACE = None
+_TRACE_LOG = []
+
+
+def trace_enabled() -> bool:
+    """Per-launch trace (ref: QUDA_ENABLE_TRACE -> trace_N.tsv): records
+    (op, key) tuples in dispatch order; QUDA_AMD_TRACE=1 enables,
+    trace_log()/trace_dump() read/persist."""
+    global _TRACE
+    if _TRACE is None:
+        import os
+        _TRACE = os.environ.get("QUDA_AMD_TRACE", "0") == "1"
+    return _TRACE
+
+
+def set_trace(v: bool) -> None:
+    global _TRACE
+    _TRACE = bool(v)
+
+
+def trace_log():
+    return _TRACE_LOG
+
+
+def trace_record(op: str, key: str) -> None:
+    if trace_enabled():
+        _TRACE_LOG.append((op, key))
+
+
+def trace_dump(path: str) -> None:
+    with open(path, "w") as f:
+        f.write("# op\tkey\n")
+        for op, key in _TRACE_LOG:
+            f.write(f"{op}\t{key}\n")
+
+
 _POLICY = None
 
 
@@ -100,6 +136,8 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     geo = out.geo
     xpay = x is not None
     mask = comms.comm_mask()
+    trace_record("dslash_wilson",
+                 f"{geo.dims}/p{parity}/m{mode}/dag{int(dagger)}/{inp.precision}")
     if on_gpu(out, inp):
         ext = hip_ext()
         cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)

@@ -61,3 +61,23 @@ def test_gauge_io(tmp_path):
     p = str(tmp_path / "gauge.pt")
     save_gauge(p, u)
     assert (load_gauge(p) - u).abs().max().item() == 0
+
+
+def test_launch_trace(tmp_path):
+    from quda_amd.ops import dispatch
+    from quda_amd import GaugeField, LatticeGeometry, SpinorField
+    from quda_amd.ops.dispatch import dslash_wilson
+    dispatch.set_trace(True)
+    try:
+        geo = LatticeGeometry((4, 4, 4, 4))
+        g = GaugeField(geo, "double").random_su3_(seed=1)
+        s = SpinorField(geo, "double", n_parity=1).gaussian_(seed=2)
+        o = SpinorField(geo, "double", n_parity=1)
+        n0 = len(dispatch.trace_log())
+        dslash_wilson(o, s, g, 0)
+        assert len(dispatch.trace_log()) == n0 + 1
+        p = str(tmp_path / "trace.tsv")
+        dispatch.trace_dump(p)
+        assert "dslash_wilson" in open(p).read()
+    finally:
+        dispatch.set_trace(False)
