@@ -252,13 +252,16 @@ static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
 
 static void twist_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
                         at::Tensor in_n, double b_re, double b_im,
-                        int64_t Vcb, int64_t sites) {
+                        int64_t Vcb, int64_t sites, int64_t tau3_vcb = 0,
+                        bool acc = false) {
   TwistApplyCall c{};
   c.out = field_of(out, out_n, Vcb);
   c.in = field_of(in, in_n, Vcb);
   c.b_re = b_re;
   c.b_im = b_im;
   c.sites = sites;
+  c.tau3_vcb = tau3_vcb;
+  c.acc = acc;
   c.prec = prec_of(out);
   launch_twist_apply(c, stream());
   check_launch("twist_apply");
@@ -319,7 +322,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dst_n"), py::arg("src"), py::arg("src_n"), py::arg("Vcb"),
         py::arg("sites"), py::arg("ncomp") = 24);
   m.def("clover_apply", &clover_apply, "clover site-matrix apply");
-  m.def("twist_apply", &twist_apply, "twisted-mass T(b) apply");
+  m.def("twist_apply", &twist_apply, "twisted-mass T(b) apply",
+        py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
+        py::arg("b_re"), py::arg("b_im"), py::arg("Vcb"), py::arg("sites"),
+        py::arg("tau3_vcb") = 0, py::arg("acc") = false);
   m.attr("BLAS_AXPY") = (int)BLAS_AXPY;
   m.attr("BLAS_AXPY_NORM2") = (int)BLAS_AXPY_NORM2;
   m.attr("BLAS_XPAY") = (int)BLAS_XPAY;

@@ -21,17 +21,31 @@ void launch_clover_apply(const CloverApplyCall &c, hipStream_t st) {
   }
 }
 
-// standalone twist apply: out = T(b) in = b_re in + i b_im g5 in
-template <typename Prec>
+// standalone twist apply: out = [out +] T(b) in = b_re in + i b_im g5 in.
+// tau3_vcb > 0: flavor-doublet mode (ls=2 field, flavor = site/tau3_vcb):
+// the g5 coefficient flips sign on the second flavor (g5 tau3).
+template <typename Prec, bool ACC, bool TAU3>
 __global__ __launch_bounds__(256) void k_twist_apply(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, typename Prec::Real br,
-    typename Prec::Real bi, long sites) {
+    typename Prec::Real bi, long sites, long tau3_vcb) {
   using R = typename Prec::Real;
   long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (g >= sites) return;
   cplx<R> v[4][3];
   in.load_g(v, g);
-  twist_mul(v, br, bi);
+  R b2 = bi;
+  if constexpr (TAU3) {
+    if ((g / tau3_vcb) & 1) b2 = -bi;
+  }
+  twist_mul(v, br, b2);
+  if constexpr (ACC) {
+    cplx<R> o[4][3];
+    out.load_g(o, g);
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) v[s][c] += o[s][c];
+  }
   out.store_g(v, g);
 }
 
@@ -42,8 +56,13 @@ static void twist_apply_t(const TwistApplyCall &c, hipStream_t st) {
   int blk = 256;
   int grid = (int)((c.sites + blk - 1) / blk);
   using R = typename Prec::Real;
-  hipLaunchKernelGGL((k_twist_apply<Prec>), dim3(grid), dim3(blk), 0, st, out,
-                     in, (R)c.b_re, (R)c.b_im, c.sites);
+#define QA_TW(ACC, TAU3)                                                      \
+  hipLaunchKernelGGL((k_twist_apply<Prec, ACC, TAU3>), dim3(grid), dim3(blk), \
+                     0, st, out, in, (R)c.b_re, (R)c.b_im, c.sites,           \
+                     c.tau3_vcb)
+  if (c.acc) { if (c.tau3_vcb) QA_TW(true, true); else QA_TW(true, false); }
+  else       { if (c.tau3_vcb) QA_TW(false, true); else QA_TW(false, false); }
+#undef QA_TW
 }
 
 void launch_twist_apply(const TwistApplyCall &c, hipStream_t st) {

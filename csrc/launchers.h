@@ -114,6 +114,8 @@ struct TwistApplyCall {
   BlasField out, in;
   double b_re, b_im;
   long sites;
+  long tau3_vcb;  // >0: flavor-doublet g5*tau3 mode (flavor = site/tau3_vcb)
+  bool acc;       // accumulate into out instead of overwrite
   int prec;
 };
 void launch_twist_apply(const TwistApplyCall &c, hipStream_t st);

@@ -1,10 +1,12 @@
 from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
                         DiracStaggered, DiracStaggeredPC)
-from .dirac import (Dirac, DiracClover, DiracCloverPC, DiracTwistedClover,
-                    DiracTwistedMass, DiracTwistedMassPC, DiracWilson,
-                    DiracWilsonPC)
+from .dirac import (Dirac, DiracClover, DiracCloverPC,
+                    DiracNdegTwistedMass, DiracNdegTwistedMassPC,
+                    DiracTwistedClover, DiracTwistedMass, DiracTwistedMassPC,
+                    DiracWilson, DiracWilsonPC)
 
 __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracCloverPC", "DiracTwistedMass", "DiracTwistedMassPC",
            "DiracTwistedClover", "DiracStaggered", "DiracStaggeredPC",
-           "DiracImprovedStaggered", "DiracImprovedStaggeredPC"]
+           "DiracImprovedStaggered", "DiracImprovedStaggeredPC",
+           "DiracNdegTwistedMass", "DiracNdegTwistedMassPC"]

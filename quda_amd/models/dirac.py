@@ -316,3 +316,123 @@ class DiracTwistedClover(Dirac, _CloverMixin):
 
     def flops_per_site(self) -> int:
         return 1320 + 504 + 48
+
+
+class DiracNdegTwistedMass(Dirac):
+    """Non-degenerate twisted-mass doublet (ref: lib/dirac_twisted_mass.cpp
+    two-flavor branch + kernels/dslash_ndeg_twisted_mass.cuh):
+    M = (1 + i 2 kappa mu g5 tau3 - 2 kappa eps tau1) - kappa D on a
+    flavor doublet, represented as an ls=2 field (slice = flavor; tau1 is
+    the Ls=2 s-hop with mf=-1, g5 tau3 the per-flavor-signed twist)."""
+
+    def __init__(self, gauge: GaugeField, kappa: float, mu: float,
+                 epsilon: float):
+        super().__init__(gauge, kappa)
+        self.mu = float(mu)
+        self.epsilon = float(epsilon)
+
+    @property
+    def a_t(self) -> float:
+        return 2.0 * self.kappa * self.mu
+
+    @property
+    def b_t(self) -> float:
+        return -2.0 * self.kappa * self.epsilon
+
+    def new_spinor(self, precision=None, n_parity=1) -> SpinorField:
+        return SpinorField(self.geo, precision or self.gauge.precision,
+                           self.gauge.device, n_parity, ls=2)
+
+    def _apply_A(self, out, inp, dagger=False):
+        """out = (1 + i a g5 tau3 + b tau1) in  (per parity view)."""
+        from ..ops.dispatch import apply_twist_field, dwf5_op
+        sgn = -1.0 if dagger else 1.0
+        # out = in + b tau1 in  (Ls=2 s-hop with mf=-1 gives psi(1-s))
+        dwf5_op(out, inp, 1.0, self.b_t, -1.0, kind=0)
+        # out += i a g5 tau3 in
+        apply_twist_field(out, inp, 0.0, sgn * self.a_t, tau3=True, acc=True)
+        return out
+
+    def _apply_Ainv(self, out, inp, dagger=False):
+        """A^-1 = (1 - i a g5 tau3 - b tau1)/(1 + a^2 - b^2)."""
+        from ..ops import blas
+        from ..ops.dispatch import apply_twist_field, dwf5_op
+        sgn = -1.0 if dagger else 1.0
+        den = 1.0 + self.a_t ** 2 - self.b_t ** 2
+        dwf5_op(out, inp, 1.0, -self.b_t, -1.0, kind=0)
+        apply_twist_field(out, inp, 0.0, -sgn * self.a_t, tau3=True, acc=True)
+        blas.scal(1.0 / den, out)
+        return out
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import dslash_wilson_slice, dwf_halo_exchange
+        assert inp.n_parity == 2
+        for p in (0, 1):
+            op = out.parity_view(p)
+            self._apply_A(op, inp.parity_view(p), dagger)
+            io = inp.parity_view(1 - p)
+            h = dwf_halo_exchange(io, 1 - p, dagger)
+            for s in (0, 1):
+                dslash_wilson_slice(op, io, self.gauge, p, s, dagger,
+                                    a=-self.kappa, x=op, halo=h)
+        return out
+
+    def MdagM(self, out, inp, tmp):
+        self.M(tmp, inp, dagger=False)
+        self.M(out, tmp, dagger=True)
+        return out
+
+    def flops_per_site(self) -> int:
+        return 2 * 1320 + 2 * 96  # per 4-d site (two flavors)
+
+
+class DiracNdegTwistedMassPC(DiracNdegTwistedMass):
+    """Symmetric even-odd PC doublet:
+    M_pc = 1 - kappa^2 Ainv D Ainv D (A flavor-structured, x-local;
+    A and the 4-d hops commute since A is spin... A contains g5: it does
+    NOT commute with D — apply in operator order like the Moebius PC)."""
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import dslash_wilson_slice, dwf_halo_exchange
+        assert inp.n_parity == 1
+        t = self.tmp("ndeg_t", inp)
+        u = self.tmp("ndeg_u", inp)
+        k2 = -self.kappa ** 2
+
+        def dhat(dst, src, parity, acc_into=None, a=1.0):
+            h = dwf_halo_exchange(src, 1 - parity, dagger)
+            if acc_into is None:
+                dst.zero_()
+            for s in (0, 1):
+                dslash_wilson_slice(dst, src, self.gauge, parity, s, dagger,
+                                    a=a, x=acc_into if acc_into is not None
+                                    else dst, halo=h)
+            return dst
+
+        if not dagger:
+            dhat(t, inp, 1)                    # t_o = D_oe in
+            self._apply_Ainv(u, t)             # u = Ainv t
+            dhat(t, u, 0)                      # t_e = D_eo u
+            self._apply_Ainv(u, t)             # u = Ainv t
+            from ..ops import blas
+            blas.copy(out, inp)
+            blas.axpy(k2, u, out)
+        else:
+            # M^dag = 1 - k^2 D^d Ainv^d D^d Ainv^d
+            self._apply_Ainv(u, inp, dagger=True)
+            dhat(t, u, 1)                      # (D_eo)^d -> odd
+            self._apply_Ainv(u, t, dagger=True)
+            dhat(t, u, 0)                      # (D_oe)^d -> even
+            from ..ops import blas
+            blas.copy(out, inp)
+            blas.axpy(k2, t, out)
+        return out
+
+    def tmp(self, name, like):
+        key = (name, like.precision, str(like.device), like.n_parity, 2)
+        t = self._tmps.get(key)
+        if t is None:
+            t = SpinorField(self.geo, like.precision, like.device,
+                            like.n_parity, ls=2)
+            self._tmps[key] = t
+        return t

@@ -211,15 +211,30 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     return out
 
 
-def apply_twist_field(out: SpinorField, inp: SpinorField, br: float, bi: float):
-    """out = T(b) in = br*in + i*bi*g5*in (site-local)."""
+def apply_twist_field(out: SpinorField, inp: SpinorField, br: float,
+                      bi: float, tau3: bool = False, acc: bool = False):
+    """out = [out +] br*in + i*bi*g5*in. tau3=True: flavor-doublet mode
+    (ls=2 field): the g5 coefficient flips sign for flavor 1 (g5 tau3)."""
     if on_gpu(out, inp):
         ext = hip_ext()
         ext.twist_apply(out.data, norm_or_empty(out), inp.data,
                         norm_or_empty(inp), float(br), float(bi),
-                        out.geo.volume_cb, out.n_parity * out.geo.volume_cb)
+                        out.volume_cb, out.n_parity * out.volume_cb,
+                        tau3_vcb=(out.geo.volume_cb if tau3 else 0),
+                        acc=acc)
         return out
-    out.from_complex(ref.apply_twist(inp.to_complex(), br, bi))
+    c = inp.to_complex()
+    if tau3:
+        assert inp.ls == 2
+        V = inp.geo.volume_cb
+        r = torch.empty_like(c)
+        r[:, :V] = ref.apply_twist(c[:, :V], br, bi)
+        r[:, V:] = ref.apply_twist(c[:, V:], br, -bi)
+    else:
+        r = ref.apply_twist(c, br, bi)
+    if acc:
+        r = r + out.to_complex()
+    out.from_complex(r)
     return out
 
 
