@@ -13,8 +13,13 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 def test_bench_two_ranks_gloo():
     env = dict(os.environ)
+    import socket
+    sock = socket.socket()
+    sock.bind(("127.0.0.1", 0))
+    port = sock.getsockname()[1]
+    sock.close()
     env["MASTER_ADDR"] = "127.0.0.1"
-    env["MASTER_PORT"] = "29511"
+    env["MASTER_PORT"] = str(port)
     procs = []
     for r in range(2):
         e = dict(env, RANK=str(r), LOCAL_RANK=str(r), WORLD_SIZE="2")
