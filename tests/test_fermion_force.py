@@ -59,3 +59,67 @@ def test_hmc_2f_energy_conservation(setup):
         dHs.append(abs(dH))
     assert dHs[1] < dHs[0]  # second-order integrator
     assert dHs[1] < 1.0
+
+
+def test_autograd_force_matches_analytic_wilson(setup):
+    """Autograd force == the hand-derived Wilson force."""
+    from quda_amd.gauge.fermion_force import (autograd_fermion_force,
+                                              fermion_action_and_force,
+                                              wilson_fermion_force)
+    from quda_amd.ops.reference import mat_wilson
+    from quda_amd.solvers.variants import _NormalOp
+    from quda_amd.solvers import cg_solve
+    from quda_amd import GaugeField, SpinorField
+    geo, u, phi = setup
+    g = GaugeField(geo, "double").from_complex(u)
+    d = DiracWilson(g, KAPPA)
+    X = d.new_spinor(n_parity=2)
+    cg_solve(_NormalOp(d, mmdag=True), X, phi, tol=1e-12, maxiter=2000)
+    Y = d.new_spinor(n_parity=2)
+    d.M(Y, X, dagger=True)
+    F_analytic = wilson_fermion_force(u, geo, KAPPA, X.to_complex(),
+                                      Y.to_complex())
+
+    def apply_M(u_t, psi):
+        return mat_wilson(u_t, psi, geo, KAPPA)
+
+    F_auto = autograd_fermion_force(u, geo, apply_M, X.to_complex(),
+                                    Y.to_complex())
+    err = (F_auto - F_analytic).abs().max().item()
+    scale = F_analytic.abs().max().item()
+    assert err < 1e-10 * max(scale, 1.0), (err, scale)
+
+
+def test_clover_force_finite_difference(setup):
+    """Clover-term force (autograd through the field-strength chain) vs
+    finite differences — the computeCloverForceQuda check."""
+    from quda_amd.gauge.fermion_force import clover_fermion_force
+    geo, u, _ = setup
+    from quda_amd.fields.clover import CloverField
+    from quda_amd.models import DiracClover
+    from quda_amd.ops.reference import clover_matrix
+    from quda_amd import GaugeField, SpinorField
+    csw = 1.0
+    # fresh pseudofermion under the clover operator
+    g = GaugeField(geo, "double").from_complex(u)
+    A = clover_matrix(u, geo, KAPPA, csw)
+    cl = CloverField(geo, "double").from_matrices(A)
+    d = DiracClover(g, cl, KAPPA)
+    eta = d.new_spinor(n_parity=2)
+    eta.gaussian_(seed=171)
+    phi = d.new_spinor(n_parity=2)
+    d.M(phi, eta)
+
+    P = random_momentum(geo, seed=172)
+    eps = 1e-6
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    Sp, _ = clover_fermion_force(up, geo, KAPPA, csw, phi, cg_tol=1e-12)
+    Sm, _ = clover_fermion_force(um, geo, KAPPA, csw, phi, cg_tol=1e-12)
+    dSdt = (Sp - Sm) / (2 * eps)
+    _, F = clover_fermion_force(u, geo, KAPPA, csw, phi, cg_tol=1e-12)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
