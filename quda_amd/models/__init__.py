@@ -1,6 +1,7 @@
 from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
                         DiracStaggered, DiracStaggeredPC)
-from .dirac import (Dirac, DiracClover, DiracCloverPC,
+from .dirac import (Dirac, DiracClover, DiracCloverHasenbuschTwist,
+                    DiracCloverHasenbuschTwistPC, DiracCloverPC,
                     DiracNdegTwistedMass, DiracNdegTwistedMassPC,
                     DiracTwistedClover, DiracTwistedMass, DiracTwistedMassPC,
                     DiracWilson, DiracWilsonPC)
@@ -9,4 +10,5 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracCloverPC", "DiracTwistedMass", "DiracTwistedMassPC",
            "DiracTwistedClover", "DiracStaggered", "DiracStaggeredPC",
            "DiracImprovedStaggered", "DiracImprovedStaggeredPC",
-           "DiracNdegTwistedMass", "DiracNdegTwistedMassPC"]
+           "DiracNdegTwistedMass", "DiracNdegTwistedMassPC",
+           "DiracCloverHasenbuschTwist", "DiracCloverHasenbuschTwistPC"]

@@ -436,3 +436,38 @@ class DiracNdegTwistedMassPC(DiracNdegTwistedMass):
                             like.n_parity, ls=2)
             self._tmps[key] = t
         return t
+
+
+class DiracCloverHasenbuschTwist(DiracClover):
+    """Clover + Hasenbusch twist (ref: lib/dirac_clover_hasenbusch_twist.cpp):
+    M' = M_clover + i mu_h g5 — used for Hasenbusch mass splitting in HMC.
+    Composed as the fused clover op plus one accumulate-twist launch."""
+
+    def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float,
+                 mu_h: float):
+        super().__init__(gauge, clover, kappa)
+        self.mu_h = float(mu_h)
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import apply_twist_field
+        super().M(out, inp, dagger)
+        sgn = -1.0 if dagger else 1.0
+        apply_twist_field(out, inp, 0.0, sgn * self.mu_h, acc=True)
+        return out
+
+
+class DiracCloverHasenbuschTwistPC(DiracCloverPC):
+    """PC version: M'_pc = M_pc + i mu_h g5
+    (ref: dirac_quda.h DiracCloverHasenbuschTwistPC)."""
+
+    def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float,
+                 mu_h: float):
+        super().__init__(gauge, clover, kappa)
+        self.mu_h = float(mu_h)
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import apply_twist_field
+        super().M(out, inp, dagger)
+        sgn = -1.0 if dagger else 1.0
+        apply_twist_field(out, inp, 0.0, sgn * self.mu_h, acc=True)
+        return out

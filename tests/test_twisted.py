@@ -159,3 +159,36 @@ def test_tm_pc_cg_gpu(setup):
     x = SpinorField(geo, "double", "cuda", n_parity=1)
     st = cg_solve(pc, x, b, tol=1e-9, maxiter=400)
     assert st.converged
+
+
+def test_hasenbusch_twist(setup):
+    """M' = M_clover + i mu g5; adjointness + CGNR solve."""
+    from quda_amd.models import (DiracCloverHasenbuschTwist,
+                                 DiracCloverHasenbuschTwistPC)
+    geo, g = setup
+    u = g.to_complex()
+    A = ref.clover_matrix(u, geo, KAPPA, 1.0)
+    cl = CloverField(geo, "double").from_matrices(A)
+    d = DiracCloverHasenbuschTwist(g, cl, KAPPA, 0.05)
+    a = SpinorField(geo, "double").gaussian_(seed=251)
+    b = SpinorField(geo, "double").gaussian_(seed=252)
+    Ma = SpinorField(geo, "double")
+    Mdb = SpinorField(geo, "double")
+    d.M(Ma, a)
+    d.M(Mdb, b, dagger=True)
+    lhs = (b.to_complex().conj() * Ma.to_complex()).sum()
+    rhs = (Mdb.to_complex().conj() * a.to_complex()).sum()
+    assert abs(lhs - rhs) < 1e-10 * abs(lhs)
+    # explicit check: M' psi = M psi + i mu g5 psi
+    from quda_amd.models import DiracClover
+    M0 = SpinorField(geo, "double")
+    DiracClover(g, cl, KAPPA).M(M0, a)
+    expect = M0.to_complex() + ref.apply_twist(a.to_complex(), 0.0, 0.05)
+    assert (Ma.to_complex() - expect).abs().max().item() < 1e-12
+    # PC solve
+    pc = DiracCloverHasenbuschTwistPC(g, cl, KAPPA, 0.05)
+    be = SpinorField(geo, "double", n_parity=1).gaussian_(seed=253)
+    x = SpinorField(geo, "double", n_parity=1)
+    from quda_amd.solvers import cgnr_solve
+    st = cgnr_solve(pc, x, be, tol=1e-10, maxiter=600)
+    assert st.converged
