@@ -79,3 +79,26 @@ def test_parity_view_shares_storage(small_geo):
     e = s.parity_view(0)
     e.data.mul_(2.0)
     assert torch.allclose(s.data[0], e.data[0])
+
+
+def test_interop_roundtrips(small_geo):
+    import torch
+    from quda_amd import GaugeField, SpinorField
+    from quda_amd.fields import interop
+    geo = small_geo
+    u = GaugeField(geo, "double").random_su3_(seed=261).to_complex()
+    s = SpinorField(geo, "double").gaussian_(seed=262).to_complex()
+    for to, frm in ((interop.gauge_to_qdp, interop.gauge_from_qdp),
+                    (interop.gauge_to_milc, interop.gauge_from_milc),
+                    (interop.gauge_to_cps, interop.gauge_from_cps)):
+        back = frm(to(u, geo), geo)
+        assert (back - u).abs().max().item() == 0
+    for to, frm in ((interop.spinor_to_qdp, interop.spinor_from_qdp),
+                    (interop.spinor_to_milc, interop.spinor_from_milc),
+                    (interop.spinor_to_cps, interop.spinor_from_cps)):
+        back = frm(to(s, geo), geo)
+        assert (back - s).abs().max().item() == 0
+    # MILC gauge layout structure: site-major with direction index
+    m = interop.gauge_to_milc(u, geo)
+    assert m.shape == (geo.volume, 4, 3, 3)
+    assert (m[0, 2] - u[2, 0, 0]).abs().max().item() == 0
