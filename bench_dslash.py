@@ -57,25 +57,106 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--lattice", default="32,32,32,64")
     ap.add_argument("--reps", type=int, default=100)
+    ap.add_argument("--families", default="wilson",
+                    help="comma list: wilson,staggered,mobius,all")
     args = ap.parse_args()
+    fams = set(args.families.split(","))
+    if "all" in fams:
+        fams = {"wilson", "staggered", "mobius"}
     dims = tuple(int(x) for x in args.lattice.split(","))
     geo = LatticeGeometry(dims)
     gen = torch.Generator().manual_seed(9)
     m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen, dtype=torch.float32)
     from quda_amd.fields.gauge import project_su3
     u = project_su3(torch.view_as_complex(m.to(torch.float64)).cuda())
-    A = ref.clover_matrix(u, geo, 0.135, 1.0)
     results = {}
-    for prec, recon in [("double", "none"), ("single", "none"),
-                        ("single", "twelve"), ("half", "twelve")]:
-        for mode, mname in [(PLAIN, "wilson"), (CLOV_POST, "wilson_clover")]:
-            r = run(prec, recon, mode, geo, u, A, args.reps)
-            key = f"{mname}/{prec}/r{18 if recon == 'none' else 12}"
-            results[key] = r
-            print(f"{key:32s} {r['us']:8.1f} us  {r['gflops']:8.0f} GFLOPS  "
-                  f"{r['gbs']:7.0f} GB/s(model)", flush=True)
-    print(json.dumps({k: round(v["gflops"]) for k, v in results.items()}))
+    if "wilson" in fams:
+        A = ref.clover_matrix(u, geo, 0.135, 1.0)
+        for prec, recon in [("double", "none"), ("single", "none"),
+                            ("single", "twelve"), ("half", "twelve")]:
+            for mode, mname in [(PLAIN, "wilson"), (CLOV_POST, "wilson_clover")]:
+                r = run(prec, recon, mode, geo, u, A, args.reps)
+                key = f"{mname}/{prec}/r{18 if recon == 'none' else 12}"
+                results[key] = r["gflops"]
+                print(f"{key:32s} {r['us']:8.1f} us  {r['gflops']:8.0f} GFLOPS  "
+                      f"{r['gbs']:7.0f} GB/s(model)", flush=True)
+    if "staggered" in fams:
+        results.update(bench_staggered(geo, u, args.reps))
+    if "mobius" in fams:
+        results.update(bench_mobius(geo, u, max(args.reps // 5, 3)))
+    print(json.dumps({k: round(v) for k, v in results.items()}))
 
+
+def bench_staggered(geo, u, reps):
+    """Naive + improved staggered rates (ref staggered flop model 570/site
+    naive, 1146 improved)."""
+    from quda_amd.gauge.hisq import asqtad_coefficients, fat_links, naik_links
+    from quda_amd.ops.dispatch import dslash_staggered
+    dev = "cuda"
+    out = {}
+    for prec in ("single", "half"):
+        g = GaugeField(geo, prec, dev).from_complex(u)
+        s_in = SpinorField(geo, prec, dev, n_parity=1, nspin=1).gaussian_(seed=7)
+        s_out = SpinorField(geo, prec, dev, n_parity=1, nspin=1)
+        for _ in range(5):
+            dslash_staggered(s_out, s_in, g, 0)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            dslash_staggered(s_out, s_in, g, 0)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / reps
+        gf = geo.volume_cb * 570 / dt / 1e9
+        key = f"staggered/{prec}/r18"
+        out[key] = gf
+        print(f"{key:32s} {dt*1e6:8.1f} us  {gf:8.0f} GFLOPS", flush=True)
+    # improved (fat+long)
+    fat = fat_links(u, geo, asqtad_coefficients())
+    lng = naik_links(u, geo)
+    for prec in ("single",):
+        gf_ = GaugeField(geo, prec, dev).from_complex(fat)
+        gl = GaugeField(geo, prec, dev, shift=3).from_complex(lng)
+        s_in = SpinorField(geo, prec, dev, n_parity=1, nspin=1).gaussian_(seed=8)
+        s_out = SpinorField(geo, prec, dev, n_parity=1, nspin=1)
+        for _ in range(5):
+            dslash_staggered(s_out, s_in, gf_, 0, long_gauge=gl)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            dslash_staggered(s_out, s_in, gf_, 0, long_gauge=gl)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / reps
+        gf2 = geo.volume_cb * 1146 / dt / 1e9
+        key = f"hisq/{prec}/r18"
+        out[key] = gf2
+        print(f"{key:32s} {dt*1e6:8.1f} us  {gf2:8.0f} GFLOPS", flush=True)
+    return out
+
+
+def bench_mobius(geo, u, reps, Ls=12):
+    """Moebius full-operator application rate (4-d hops x Ls + s-structure;
+    per-slice Wilson flop model + M5 ops)."""
+    from quda_amd.models.dwf import DiracMobius
+    dev = "cuda"
+    out = {}
+    for prec in ("single",):
+        g = GaugeField(geo, prec, dev).from_complex(u)
+        d = DiracMobius(g, 0.04, 1.8, Ls)
+        psi = SpinorField(geo, prec, dev, ls=Ls).gaussian_(seed=9)
+        res = SpinorField(geo, prec, dev, ls=Ls)
+        for _ in range(3):
+            d.M(res, psi)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            d.M(res, psi)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / reps
+        gf = geo.volume * Ls * (1320 + 96) / dt / 1e9
+        key = f"mobius_M_Ls{Ls}/{prec}"
+        out[key] = gf
+        print(f"{key:32s} {dt*1e6:8.1f} us  {gf:8.0f} GFLOPS", flush=True)
+    return out
 
 if __name__ == "__main__":
     main()
