@@ -25,7 +25,10 @@ from quda_amd.solvers import gcr_solve  # noqa: E402
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--lattice", default="16,16,16,32")
-    ap.add_argument("--kappa", type=float, default=0.142)
+    ap.add_argument("--kappa", type=float, default=0.1245)
+    ap.add_argument("--smear", type=int, default=6,
+                    help="stout steps on the random field (smooth fields "
+                         "have near-critical physics; hot ones are gapped)")
     ap.add_argument("--block", default="4,4,4,4")
     ap.add_argument("--nvec", type=int, default=8)
     ap.add_argument("--tol", type=float, default=1e-8)
@@ -39,6 +42,9 @@ def main():
     m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
                     dtype=torch.float32)
     u = project_su3(torch.view_as_complex(m.to(torch.float64)).to(dev))
+    if args.smear:
+        from quda_amd.gauge import stout_smear
+        u = stout_smear(u, geo, 0.12, args.smear)
     g = GaugeField(geo, "double", dev).from_complex(u)
     A = ref.clover_matrix(u, geo, args.kappa, 1.0)
     cl = CloverField(geo, "double", dev).from_matrices(A)
