@@ -2,6 +2,7 @@
 inv_*_quda.cpp implementations)."""
 
 from .bicgstab import bicgstab_solve, bicgstabl_solve
+from .block_cg import block_cg_solve
 from .ca import ca_cg_solve, ca_gcr_solve
 from .cg import SolverStats, cg_solve
 from .gcr import gcr_solve, mr_solve
@@ -24,6 +25,7 @@ SOLVERS = {
     "ca-cg": ca_cg_solve,
     "ca-gcr": ca_gcr_solve,
     "multishift-cg": multishift_cg_solve,
+    "block-cg": block_cg_solve,
 }
 
 
@@ -39,4 +41,4 @@ __all__ = ["cg_solve", "SolverStats", "bicgstab_solve", "bicgstabl_solve",
            "gcr_solve", "mr_solve", "ca_cg_solve", "ca_gcr_solve",
            "multishift_cg_solve", "cgne_solve", "cgnr_solve", "cg3_solve",
            "sd_solve", "pcg_solve", "ChronoForecaster", "create_solver",
-           "SOLVERS"]
+           "SOLVERS", "block_cg_solve"]
