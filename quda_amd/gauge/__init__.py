@@ -7,7 +7,7 @@ setup/analysis utilities, not the solver hot path."""
 
 from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
                   plaquette, polyakov_loop, project_ta, staple_sum,
-                  stout_smear, topological_charge, wilson_flow)
+                  stout_smear, topological_charge, wilson_flow, wilson_loop)
 from .hmc import hmc_trajectory, leapfrog, mom_action, random_momentum
 from .fix import gauge_fix_ovr, gauge_fix_quality
 from .heatbath import heatbath_sweep, overrelax_sweep
@@ -20,4 +20,4 @@ __all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
            "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",
            "heatbath_sweep", "overrelax_sweep", "wilson_fermion_force",
            "fermion_action_and_force", "hmc_trajectory_2f",
-           "pseudofermion_refresh", "gauge_fix_ovr", "gauge_fix_quality"]
+           "pseudofermion_refresh", "gauge_fix_ovr", "gauge_fix_quality", "wilson_loop"]

@@ -202,3 +202,42 @@ def topological_charge(u: torch.Tensor, geo: LatticeGeometry) -> float:
 
     q = trprod((0, 1), (2, 3)) - trprod((0, 2), (1, 3)) + trprod((0, 3), (1, 2))
     return q / (4.0 * math.pi ** 2)
+
+
+def wilson_loop(u: torch.Tensor, geo: LatticeGeometry, R: int, T: int,
+                mu: int = 0, nu: int = 3) -> complex:
+    """Mean R x T rectangular Wilson loop in the (mu,nu) plane
+    (ref: lib/gauge_loop_trace.cu batched loop traces)."""
+    U = _to_lex(u, geo)
+
+    # path product: R steps of +mu, T of +nu, R of -mu, T of -nu
+    V = geo.volume
+    P = torch.eye(3, dtype=u.dtype, device=u.device).expand(V, 3, 3).clone()
+    shift_total = [0, 0, 0, 0]
+
+    def shifted(f):
+        out = f
+        for d in range(4):
+            s = shift_total[d]
+            while s > 0:
+                out = _shift(out, geo, d, +1)
+                s -= 1
+            while s < 0:
+                out = _shift(out, geo, d, -1)
+                s += 1
+        return out
+
+    for _ in range(R):
+        P = P @ shifted(U[mu])
+        shift_total[mu] += 1
+    for _ in range(T):
+        P = P @ shifted(U[nu])
+        shift_total[nu] += 1
+    for _ in range(R):
+        shift_total[mu] -= 1
+        P = P @ shifted(U[mu]).conj().mT
+    for _ in range(T):
+        shift_total[nu] -= 1
+        P = P @ shifted(U[nu]).conj().mT
+    tr = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).mean() / 3.0
+    return complex(tr.real.item(), tr.imag.item())
