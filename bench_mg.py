@@ -26,9 +26,12 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--lattice", default="16,16,16,32")
     ap.add_argument("--kappa", type=float, default=0.1245)
-    ap.add_argument("--smear", type=int, default=6,
-                    help="stout steps on the random field (smooth fields "
-                         "have near-critical physics; hot ones are gapped)")
+    ap.add_argument("--smear", type=int, default=0,
+                    help="stout steps on the random field")
+    ap.add_argument("--therm", type=int, default=120,
+                    help="heatbath(+3 OR) thermalization iterations at "
+                         "--beta from a cold start (0 = random field)")
+    ap.add_argument("--beta", type=float, default=6.0)
     ap.add_argument("--block", default="4,4,4,4")
     ap.add_argument("--nvec", type=int, default=8)
     ap.add_argument("--tol", type=float, default=1e-8)
@@ -41,10 +44,22 @@ def main():
     from quda_amd.fields.gauge import project_su3
     m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
                     dtype=torch.float32)
-    u = project_su3(torch.view_as_complex(m.to(torch.float64)).to(dev))
-    if args.smear:
-        from quda_amd.gauge import stout_smear
-        u = stout_smear(u, geo, 0.12, args.smear)
+    if args.therm:
+        from quda_amd.gauge import heatbath_sweep, overrelax_sweep, plaquette
+        u = GaugeField(geo, "double", dev).unit_().to_complex()
+        t0 = time.perf_counter()
+        for it in range(args.therm):
+            u = heatbath_sweep(u, geo, args.beta, seed=1000 + 7 * it)
+            for j in range(3):
+                u = overrelax_sweep(u, geo, args.beta, seed=5000 + 13 * it + j)
+        p, _, _ = plaquette(u, geo)
+        print(f"# thermalized {args.therm} it at beta={args.beta}: "
+              f"plaq={p:.4f} ({time.perf_counter()-t0:.0f}s)", flush=True)
+    else:
+        u = project_su3(torch.view_as_complex(m.to(torch.float64)).to(dev))
+        if args.smear:
+            from quda_amd.gauge import stout_smear
+            u = stout_smear(u, geo, 0.12, args.smear)
     g = GaugeField(geo, "double", dev).from_complex(u)
     A = ref.clover_matrix(u, geo, args.kappa, 1.0)
     cl = CloverField(geo, "double", dev).from_matrices(A)

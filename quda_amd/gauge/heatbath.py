@@ -33,27 +33,33 @@ def _su2_embed_mul(U: torch.Tensor, r0, r1, r2, r3, i: int, j: int):
     return out
 
 
+def _rand(n, gen, device, dtype=torch.float64):
+    return torch.rand(n, generator=gen, dtype=dtype).to(device)
+
+
 def _kp_sample(k: torch.Tensor, beta_eff: float, gen) -> torch.Tensor:
     """Kennedy-Pendleton sampling of a0 with density
-    ~ sqrt(1-a0^2) exp(beta_eff k a0) (vectorized rejection)."""
+    ~ sqrt(1-a0^2) exp(beta_eff k a0) (vectorized rejection; RNG on host,
+    sampling math on the field's device)."""
     n = k.shape
+    dev = k.device
     a0 = torch.empty_like(k)
     todo = torch.ones_like(k, dtype=torch.bool)
     alpha = beta_eff * k
     it = 0
+    import math
     while todo.any() and it < 100:
-        r1 = torch.rand(n, generator=gen).clamp_min(1e-12)
-        r2 = torch.rand(n, generator=gen)
-        r3 = torch.rand(n, generator=gen).clamp_min(1e-12)
-        import math
+        r1 = _rand(n, gen, dev).clamp_min(1e-12)
+        r2 = _rand(n, gen, dev)
+        r3 = _rand(n, gen, dev).clamp_min(1e-12)
         x = -(torch.log(r1) + (torch.cos(2 * math.pi * r2)) ** 2 * torch.log(r3)) / alpha
-        acc = (torch.rand(n, generator=gen) ** 2 <= 1 - 0.5 * x) & (x <= 2.0)
+        acc = (_rand(n, gen, dev) ** 2 <= 1 - 0.5 * x) & (x <= 2.0)
         upd = todo & acc
         a0[upd] = (1 - x)[upd]
         todo = todo & ~acc
         it += 1
-    a0[todo] = 2 * torch.rand(int(todo.sum().item()), generator=gen,
-                              dtype=k.dtype) - 1  # fallback
+    if todo.any():
+        a0[todo] = 2 * _rand((int(todo.sum().item()),), gen, dev) - 1  # fallback
     return a0
 
 
@@ -74,9 +80,9 @@ def _sweep(u, geo, beta, gen, mode: str):
                                     gen).to(a0.dtype)
                     rho = torch.sqrt((1 - b0 ** 2).clamp_min(0))
                     import math
-                    ct = 2 * torch.rand(a0.shape, generator=gen, dtype=a0.dtype) - 1
+                    ct = 2 * _rand(a0.shape, gen, a0.device, a0.dtype) - 1
                     st = torch.sqrt((1 - ct ** 2).clamp_min(0))
-                    ph = 2 * math.pi * torch.rand(a0.shape, generator=gen, dtype=a0.dtype)
+                    ph = 2 * math.pi * _rand(a0.shape, gen, a0.device, a0.dtype)
                     b1 = rho * st * torch.cos(ph)
                     b2 = rho * st * torch.sin(ph)
                     b3 = rho * ct
