@@ -28,7 +28,7 @@ def main():
     ap.add_argument("--kappa", type=float, default=0.1245)
     ap.add_argument("--smear", type=int, default=0,
                     help="stout steps on the random field")
-    ap.add_argument("--therm", type=int, default=120,
+    ap.add_argument("--therm", type=int, default=0,
                     help="heatbath(+3 OR) thermalization iterations at "
                          "--beta from a cold start (0 = random field)")
     ap.add_argument("--beta", type=float, default=6.0)

@@ -396,3 +396,43 @@ def perform_gauge_smear_quda(kind: str, n_steps: int, coeff: float) -> None:
     else:
         raise ValueError(kind)
     load_gauge_quda(u, _R.gauge_param)
+
+
+# -- HMC entry points (ref: computeGaugeForceQuda, updateGaugeFieldQuda,
+# momActionQuda, momResidentQuda) ------------------------------------------
+
+def compute_gauge_force_quda(beta: float) -> torch.Tensor:
+    from .gauge import gauge_force
+    return gauge_force(_R.u_complex, _R.geo, beta)
+
+
+def update_gauge_field_quda(mom: torch.Tensor, dt: float) -> None:
+    """U <- exp(dt P) U on the resident field."""
+    from .gauge.ops import _from_lex, _to_lex, exp_su3
+    U = _to_lex(_R.u_complex, _R.geo)
+    P = _to_lex(mom, _R.geo)
+    load_gauge_quda(_from_lex(exp_su3(P, dt) @ U, _R.geo), _R.gauge_param)
+
+
+def mom_action_quda(mom: torch.Tensor) -> float:
+    from .gauge import mom_action
+    return mom_action(mom)
+
+
+def gauss_mom_quda(seed: int) -> torch.Tensor:
+    from .gauge import random_momentum
+    return random_momentum(_R.geo, _R.gauge.device, seed)
+
+
+def compute_gauge_fixing_ovr_quda(gauge: str = "landau", **kw) -> None:
+    """Fix the resident field (ref: computeGaugeFixingOVRQuda)."""
+    from .gauge.fix import gauge_fix_ovr
+    load_gauge_quda(gauge_fix_ovr(_R.u_complex, _R.geo, gauge=gauge, **kw),
+                    _R.gauge_param)
+
+
+def compute_ks_link_quda(coeffs=None):
+    """Fat + long links of the resident field (ref: computeKSLinkQuda)."""
+    from .gauge.hisq import asqtad_coefficients, fat_links, naik_links
+    c = coeffs or asqtad_coefficients()
+    return fat_links(_R.u_complex, _R.geo, c), naik_links(_R.u_complex, _R.geo)

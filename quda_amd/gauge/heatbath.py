@@ -33,7 +33,20 @@ def _su2_embed_mul(U: torch.Tensor, r0, r1, r2, r3, i: int, j: int):
     return out
 
 
+_DEVICE_RNG = False
+
+
+def set_device_rng(v: bool) -> None:
+    """Draw Monte-Carlo randoms directly on the field's device (fast path
+    for GPU thermalization; host RNG with a seeded generator is the
+    deterministic default — the host path proved PCIe-bound at scale)."""
+    global _DEVICE_RNG
+    _DEVICE_RNG = bool(v)
+
+
 def _rand(n, gen, device, dtype=torch.float64):
+    if _DEVICE_RNG and str(device) != "cpu":
+        return torch.rand(n, dtype=dtype, device=device)
     return torch.rand(n, generator=gen, dtype=dtype).to(device)
 
 

@@ -113,3 +113,28 @@ def test_twisted_and_staggered_paths(resident):
     bs = _rand_spinor((2, 128, 3), 138)
     api.invert_quda(bs, ps)
     assert ps.true_res < 1e-8
+
+
+def test_hmc_entry_points(resident):
+    import torch
+    F = api.compute_gauge_force_quda(5.5)
+    assert F.shape == (4, 2, 128, 3, 3)
+    P = api.gauss_mom_quda(seed=301)
+    k0 = api.mom_action_quda(P)
+    assert k0 > 0
+    p_before = api.plaq_quda()[0]
+    api.update_gauge_field_quda(P, 0.01)
+    assert abs(api.plaq_quda()[0] - p_before) > 1e-8  # moved
+    fat, lng = api.compute_ks_link_quda()
+    assert fat.shape == lng.shape == (4, 2, 128, 3, 3)
+
+
+def test_gauge_fixing_entry(resident):
+    from quda_amd.gauge.fix import gauge_fix_quality
+    # reload a fresh field (earlier tests mutate the resident gauge)
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    api.compute_gauge_fixing_ovr_quda("landau", max_iter=400, tol=1e-8)
+    _, th = gauge_fix_quality(api._R.u_complex, api._R.geo, 4)
+    assert th < 1e-6
