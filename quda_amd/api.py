@@ -25,7 +25,8 @@ from .fields.gauge import GaugeField
 from .fields.geometry import LatticeGeometry
 from .fields.spinor import SpinorField
 from .models import (DiracClover, DiracCloverPC, DiracStaggered,
-                     DiracStaggeredPC, DiracTwistedClover, DiracTwistedMass,
+                     DiracStaggeredPC, DiracTwistedClover,
+                     DiracTwistedCloverPC, DiracTwistedMass,
                      DiracTwistedMassPC, DiracWilson, DiracWilsonPC)
 from .models.dwf import (DiracDomainWall, DiracDomainWallPC, DiracMobius,
                          DiracMobiusPC)
@@ -202,8 +203,9 @@ def _make_dirac(p: InvertParam, sloppy: bool = False):
         return (DiracTwistedMassPC(g, p.kappa, p.mu) if pc
                 else DiracTwistedMass(g, p.kappa, p.mu))
     if t == DslashType.TWISTED_CLOVER:
-        assert cl is not None and not pc
-        return DiracTwistedClover(g, cl, p.kappa, p.mu)
+        assert cl is not None, "load_clover_quda first"
+        return (DiracTwistedCloverPC(g, cl, p.kappa, p.mu) if pc
+                else DiracTwistedClover(g, cl, p.kappa, p.mu))
     if t == DslashType.STAGGERED:
         return (DiracStaggeredPC(g, p.mass) if pc
                 else DiracStaggered(g, p.mass))

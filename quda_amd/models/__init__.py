@@ -3,7 +3,8 @@ from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
 from .dirac import (Dirac, DiracClover, DiracCloverHasenbuschTwist,
                     DiracCloverHasenbuschTwistPC, DiracCloverPC,
                     DiracNdegTwistedMass, DiracNdegTwistedMassPC,
-                    DiracTwistedClover, DiracTwistedMass, DiracTwistedMassPC,
+                    DiracTwistedClover, DiracTwistedCloverPC,
+                    DiracTwistedMass, DiracTwistedMassPC,
                     DiracWilson, DiracWilsonPC)
 
 __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
@@ -11,4 +12,5 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracTwistedClover", "DiracStaggered", "DiracStaggeredPC",
            "DiracImprovedStaggered", "DiracImprovedStaggeredPC",
            "DiracNdegTwistedMass", "DiracNdegTwistedMassPC",
-           "DiracCloverHasenbuschTwist", "DiracCloverHasenbuschTwistPC"]
+           "DiracCloverHasenbuschTwist", "DiracCloverHasenbuschTwistPC",
+           "DiracTwistedCloverPC"]

@@ -471,3 +471,84 @@ class DiracCloverHasenbuschTwistPC(DiracCloverPC):
         sgn = -1.0 if dagger else 1.0
         apply_twist_field(out, inp, 0.0, sgn * self.mu_h, acc=True)
         return out
+
+
+class DiracTwistedCloverPC(Dirac, _CloverMixin):
+    """Symmetric even-odd preconditioned twisted clover
+    (ref: lib/dirac_twisted_clover.cpp DiracTwistedCloverPC):
+    M_pc = 1 - kappa^2 Atc^-1 D Atc^-1 D with Atc = A + i eps g5.
+    Atc^-1 = (A - i eps g5)(A^2 + eps^2)^-1 — both factors hermitian and
+    chirality-block-diagonal (they all commute), so the inverse is two
+    packed-clover applies + one twist accumulate per application.
+    `b_inv` holds the precomputed (A^2 + eps^2)^-1 as a CloverField."""
+
+    def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float,
+                 mu: float):
+        super().__init__(gauge, kappa)
+        self.clover = clover
+        self.mu = float(mu)
+        self.eps = 2.0 * kappa * mu
+        # build (A^2 + eps^2)^{-1} once (setup; ref "dynamic clover" role)
+        import torch
+        A = clover.to_complex()
+        eye = torch.eye(12, dtype=A.dtype, device=A.device)
+        B = torch.linalg.inv(A @ A + self.eps ** 2 * eye)
+        self.b_inv = CloverField(gauge.geo, clover.precision,
+                                 gauge.device).from_matrices(B)
+
+    def _apply_Atc_inv(self, out, inp, parity, dagger=False):
+        from ..ops.dispatch import apply_clover, apply_twist_field
+        t = self.tmp("tc_t", inp)
+        apply_clover(t, inp, self.b_inv, parity)        # t = B in
+        apply_clover(out, t, self.clover, parity)       # out = A t
+        sgn = 1.0 if dagger else -1.0
+        apply_twist_field(out, t, 0.0, sgn * self.eps, acc=True)
+        return out
+
+    def M(self, out, inp, dagger: bool = False):
+        k2 = -self.kappa ** 2
+        t = self.tmp("tc_o", inp)
+        u = self.tmp("tc_e", inp)
+        if not dagger:
+            self.dslash(t, inp, 1)
+            self._apply_Atc_inv(u, t, 1)
+            self.dslash(t, u, 0)
+            self._apply_Atc_inv(u, t, 0)
+            blas.copy(out, inp)
+            blas.axpy(k2, u, out)
+        else:
+            # M^dag = 1 - k^2 D^d Atc^-dag D^d Atc^-dag
+            self._apply_Atc_inv(u, inp, 0, dagger=True)
+            self.dslash(t, u, 1, dagger=True)
+            self._apply_Atc_inv(u, t, 1, dagger=True)
+            self.dslash(t, u, 0, dagger=True)
+            blas.copy(out, inp)
+            blas.axpy(k2, t, out)
+        return out
+
+    def MdagM(self, out, inp, tmp):
+        self.M(tmp, inp, dagger=False)
+        self.M(out, tmp, dagger=True)
+        return out
+
+    def prepare(self, b_full: SpinorField) -> SpinorField:
+        """b' = Atc_ee^-1 (b_e + kappa D_eo Atc_oo^-1 b_o)."""
+        t = self.new_spinor(b_full.precision)
+        be = self.new_spinor(b_full.precision)
+        self._apply_Atc_inv(t, b_full.parity_view(1), 1)
+        self.dslash(be, t, 0, a=self.kappa, x=b_full.parity_view(0))
+        self._apply_Atc_inv(t, be, 0)
+        blas.copy(be, t)
+        return be
+
+    def reconstruct(self, x_full: SpinorField, x_e: SpinorField,
+                    b_full: SpinorField):
+        """x_o = Atc_oo^-1 (b_o + kappa D_oe x_e)."""
+        blas.copy(x_full.parity_view(0), x_e)
+        t = self.tmp("tc_o", x_full)
+        self.dslash(t, x_e, 1, a=self.kappa, x=b_full.parity_view(1))
+        self._apply_Atc_inv(x_full.parity_view(1), t, 1)
+        return x_full
+
+    def flops_per_site(self) -> int:
+        return 1320 + 2 * 504 + 48
