@@ -20,6 +20,9 @@ from .transfer import Transfer, generate_null_vectors
 class MGParam:
     block: tuple = (2, 2, 2, 2)
     n_vec: int = 4
+    levels: int = 2              # 3 = recurse once more on the coarse op
+    block2: tuple = (2, 2, 2, 2)
+    n_vec2: int = 4
     nu_pre: int = 0
     nu_post: int = 4
     smoother_omega: float = 0.85
@@ -44,6 +47,11 @@ class MG:
                 maxiter=param.null_maxiter, seed=param.seed)
         self.transfer = Transfer(op.geo, param.block, vectors)
         self.coarse = build_coarse_op(op, self.transfer)
+        self.coarse_mg = None
+        if param.levels >= 3:
+            from .coarse_level import CoarseMG
+            self.coarse_mg = CoarseMG(self.coarse, block2=param.block2,
+                                      n_vec2=param.n_vec2)
 
     # -- verification (ref: multigrid.cpp MG::verify) -----------------------
     def verify(self) -> dict:
@@ -88,8 +96,12 @@ class MG:
         else:
             rr = r
         rc = self.transfer.restrict(rr).reshape(self.transfer.n_agg, -1)
-        ec = coarse_bicgstab(self.coarse, rc, tol=p.coarse_tol,
-                             maxiter=p.coarse_maxiter)
+        if self.coarse_mg is not None:
+            ec = self.coarse_mg.solve(rc, tol=p.coarse_tol,
+                                      maxiter=p.coarse_maxiter)
+        else:
+            ec = coarse_bicgstab(self.coarse, rc, tol=p.coarse_tol,
+                                 maxiter=p.coarse_maxiter)
         e = op.new_spinor(n_parity=2)
         self.transfer.prolong(ec.reshape(self.transfer.n_agg, 2, -1), e)
         blas.axpy(1.0, e, z)
