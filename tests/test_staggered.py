@@ -205,3 +205,40 @@ def test_staggered_multishift_gpu(setup):
         blas.axpy(s, xs[i], r)
         tr = math.sqrt(blas.xmy_norm2(b, r) / blas.norm2(b))
         assert tr < 1e-7, f"shift {s}: {tr}"
+
+
+def test_spin_taste_phases(setup):
+    from quda_amd.models.spin_taste import apply_spin_taste, spin_taste_phase
+    geo, g = setup
+    psi = stag(geo, 281)
+    out = apply_spin_taste(psi, "g5-g5")
+    # epsilon(x)^2 = 1
+    back = apply_spin_taste(out, "g5-g5")
+    assert (back.to_complex() - psi.to_complex()).abs().max().item() < 1e-12
+    # epsilon anticommutes with D: D eps psi = -eps D psi
+    u = g.to_complex()
+    eps_psi = apply_spin_taste(psi, "g5-g5")
+    lhs = ref.dslash_staggered_full(u, eps_psi.to_complex(), geo)
+    rhs = ref.dslash_staggered_full(u, psi.to_complex(), geo)
+    ph = spin_taste_phase(geo, "g5-g5")
+    from quda_amd.fields.geometry import checkerboard_split
+    eps_cb = checkerboard_split(ph.unsqueeze(-1), geo).squeeze(-1)
+    assert (lhs + eps_cb.unsqueeze(-1) * rhs).abs().max().item() < 1e-10
+
+
+def test_two_link_gaussian_smear(setup):
+    from quda_amd.models.spin_taste import gaussian_smear_two_link, two_links
+    geo, g = setup
+    u = g.to_complex()
+    # two-links are products of unitaries
+    N = two_links(u, geo)
+    import torch
+    det = torch.linalg.det(N.reshape(-1, 3, 3))
+    assert (det.abs() - 1).abs().max().item() < 1e-10
+    # point-source smearing spreads support on even 2-hop sublattice
+    psi = SpinorField(geo, "double", nspin=1)
+    c = torch.zeros((2, geo.volume_cb, 3), dtype=torch.complex128)
+    c[0, 0, 0] = 1.0
+    psi.from_complex(c)
+    sm = gaussian_smear_two_link(u, geo, psi, width=2.0, n_steps=4)
+    assert (sm.to_complex().abs() > 1e-10).sum().item() > 20
