@@ -35,3 +35,18 @@ def test_gauge_fixing_converges(setup, gauge, dirs):
     # links still SU(3)
     det = torch.linalg.det(uf.reshape(-1, 3, 3))
     assert (det - 1).abs().max().item() < 1e-8
+
+
+def test_gauge_fix_fft(setup):
+    from quda_amd.gauge.fix import gauge_fix_fft
+    geo, u = setup
+    f0, th0 = gauge_fix_quality(u, geo, 4)
+    uf = gauge_fix_fft(u, geo, max_iter=600, tol=1e-9, alpha=0.08)
+    f1, th1 = gauge_fix_quality(uf, geo, 4)
+    assert th1 < 1e-8, (th0, th1)
+    assert f1 > f0
+    p0, _, _ = plaquette(u, geo)
+    p1, _, _ = plaquette(uf, geo)
+    assert abs(p0 - p1) < 1e-9
+    det = torch.linalg.det(uf.reshape(-1, 3, 3))
+    assert (det - 1).abs().max().item() < 1e-7
