@@ -133,6 +133,9 @@ def autograd_fermion_force(u: torch.Tensor, geo: LatticeGeometry,
     dS/dt = 2 Re tr(P U g^dag) (g = torch grad) and the K = -tr P^2
     convention give F = (1/2) TA[U g^dag] (verified exactly against the
     hand-derived Wilson force and by finite differences)."""
+    from ..parallel import comms
+    assert comms.comm_mask() == 0, \
+        "fermion forces are single-rank this round (multi-rank HMC: round 2)"
     u_req = u.detach().clone().requires_grad_(True)
     MY = apply_M(u_req, Y)
     s = -2.0 * (X.conj() * MY).sum().real

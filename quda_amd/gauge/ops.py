@@ -26,14 +26,17 @@ def _from_lex(U: torch.Tensor, geo: LatticeGeometry) -> torch.Tensor:
 
 
 def _shift(f: torch.Tensor, geo: LatticeGeometry, mu: int, disp: int):
-    """f: [Vlex,...] -> f(x + disp*mu)."""
-    idx = geo.neighbor_lex(mu, disp).to(f.device)
-    return f[idx]
+    """f: [Vlex,...] -> f(x + disp*mu) (neighbor-rank slab on partitioned
+    dims — staples/forces/smearing are multi-rank correct)."""
+    from ..parallel.halo import shift_lex
+    return shift_lex(f, geo, mu, disp)
 
 
 def plaquette(u: torch.Tensor, geo: LatticeGeometry):
     """(total, spatial, temporal) mean plaquette Re tr P / 3
     (ref: lib/gauge_plaq.cu)."""
+    from ..parallel import comms
+    n_ranks = comms.comm_size() if comms.comm_mask() else 1
     U = _to_lex(u, geo)
     tot_s = tot_t = 0.0
     n_s = n_t = 0
@@ -42,7 +45,9 @@ def plaquette(u: torch.Tensor, geo: LatticeGeometry):
             Unu_xmu = _shift(U[nu], geo, mu, +1)
             Umu_xnu = _shift(U[mu], geo, nu, +1)
             P = U[mu] @ Unu_xmu @ Umu_xnu.conj().mT @ U[nu].conj().mT
-            val = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).real.mean().item() / 3.0
+            val = comms.allreduce_sum(
+                torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).real.sum().item()
+            ) / (3.0 * geo.volume * n_ranks)
             if nu == 3:
                 tot_t += val
                 n_t += 1
@@ -77,9 +82,11 @@ def staple_sum(U: torch.Tensor, geo: LatticeGeometry, mu: int) -> torch.Tensor:
 
 
 def gauge_action(u: torch.Tensor, geo: LatticeGeometry, beta: float) -> float:
-    """Wilson gauge action S = beta * sum_P (1 - Re tr P / 3)."""
+    """Wilson gauge action S = beta * sum_P (1 - Re tr P / 3) (global)."""
+    from ..parallel import comms
+    n_ranks = comms.comm_size() if comms.comm_mask() else 1
     p_tot, _, _ = plaquette(u, geo)
-    n_plaq = 6 * geo.volume
+    n_plaq = 6 * geo.volume * n_ranks
     return beta * n_plaq * (1.0 - p_tot)
 
 
@@ -200,8 +207,9 @@ def topological_charge(u: torch.Tensor, geo: LatticeGeometry) -> float:
     def trprod(a, b):
         return torch.einsum("pvij,pvji->", F[a], F[b]).real.item()
 
+    from ..parallel import comms
     q = trprod((0, 1), (2, 3)) - trprod((0, 2), (1, 3)) + trprod((0, 3), (1, 2))
-    return q / (4.0 * math.pi ** 2)
+    return comms.allreduce_sum(q) / (4.0 * math.pi ** 2)
 
 
 def wilson_loop(u: torch.Tensor, geo: LatticeGeometry, R: int, T: int,

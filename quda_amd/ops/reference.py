@@ -125,9 +125,11 @@ def field_strength(u: torch.Tensor, geo: LatticeGeometry):
         U[mu, lo[1]] = u[mu, 1]
 
     def shift(f, mu, disp):
-        """f: [V,3,3] field; returns f(x + disp*mu_hat)."""
-        idx = geo.neighbor_lex(mu, disp).to(dev)
-        return f[idx]
+        """f: [V,3,3] field; returns f(x + disp*mu_hat) — neighbor-rank
+        slabs on partitioned dims (the clover term is then consistent at
+        rank boundaries)."""
+        from ..parallel.halo import shift_lex
+        return shift_lex(f, geo, mu, disp)
 
     out = {}
     for mu in range(4):

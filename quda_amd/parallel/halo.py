@@ -334,3 +334,45 @@ def exchange_psi5_oracle(psi5: torch.Tensor, geo: LatticeGeometry,
         recvs[(mu, 1)] = torch.empty_like(sends[(mu, 0)])
     exchange_tensors(sends, recvs)
     return recvs
+
+
+# ---------------------------------------------------------------------------
+# distributed lexicographic-field shift (gauge-sector halo: staples, field
+# strength, smearing, forces become multi-rank correct through this)
+# ---------------------------------------------------------------------------
+
+def shift_lex(f: torch.Tensor, geo: LatticeGeometry, mu: int, disp: int
+              ) -> torch.Tensor:
+    """f: [Vlex, ...] -> f(x + disp*mu). |disp| must be 1 (compose for
+    more). On partitioned dims the wrapped face comes from the neighbor
+    rank (ref: the gauge exchangeGhost/exchangeExtendedGhost role)."""
+    assert disp in (1, -1)
+    from . import comms
+    idx = geo.neighbor_lex(mu, disp).to(f.device)
+    out = f[idx]
+    if not ((comms.comm_mask() >> mu) & 1):
+        return out
+    hi = geo.dims[mu] - 1
+    if disp == 1:
+        # my x_mu = hi sites need the +mu neighbor's x_mu = 0 slab
+        send = {(mu, 0): f[_face_lex(geo, mu, 0).to(f.device)].contiguous()}
+        recv = {(mu, 1): torch.empty_like(send[(mu, 0)])}
+        exchange_tensors(send, recv)
+        out[_face_lex(geo, mu, hi).to(f.device)] = recv[(mu, 1)]
+    else:
+        send = {(mu, 1): f[_face_lex(geo, mu, hi).to(f.device)].contiguous()}
+        recv = {(mu, 0): torch.empty_like(send[(mu, 1)])}
+        exchange_tensors(send, recv)
+        out[_face_lex(geo, mu, 0).to(f.device)] = recv[(mu, 0)]
+    return out
+
+
+def _face_lex(geo: LatticeGeometry, mu: int, edge: int) -> torch.Tensor:
+    """Lex indices of the face x_mu == edge, in natural lex order (both
+    ends agree: transverse coords are identical across the boundary)."""
+    key = ("face_lex", mu, edge)
+    cache = geo.__dict__.setdefault("_nbr_cache", {})
+    if key not in cache:
+        c = geo.coords[:, mu]
+        cache[key] = (c == edge).nonzero(as_tuple=True)[0].contiguous()
+    return cache[key]
