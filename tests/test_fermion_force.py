@@ -123,3 +123,44 @@ def test_clover_force_finite_difference(setup):
     trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
     assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
         (dSdt, -2 * trPF)
+
+
+def test_hisq_force_finite_difference():
+    """The hisq_paths_force check: backprop through fat7+Lepage+Naik
+    matches finite differences of the improved-staggered action."""
+    from quda_amd.gauge.fermion_force import hisq_fermion_force
+    from quda_amd.gauge.hisq import asqtad_coefficients, fat_links, naik_links
+    from quda_amd.fields.gauge import GaugeField, project_su3
+    from quda_amd.models import DiracImprovedStaggered
+    from quda_amd import SpinorField
+    geo = LatticeGeometry((4, 4, 4, 4))
+    gen = torch.Generator().manual_seed(176)
+    eye = torch.eye(3, dtype=torch.complex128)
+    m = eye + 0.3 * torch.view_as_complex(
+        torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64))
+    u = project_su3(m)
+    mass = 0.2
+    # pseudofermion under the improved operator
+    c = asqtad_coefficients()
+    gf = GaugeField(geo, "double").from_complex(fat_links(u, geo, c))
+    gl = GaugeField(geo, "double", shift=3).from_complex(naik_links(u, geo))
+    d = DiracImprovedStaggered(gf, gl, mass)
+    eta = d.new_spinor(n_parity=2)
+    eta.gaussian_(seed=177)
+    phi = d.new_spinor(n_parity=2)
+    d.M(phi, eta)
+
+    P = random_momentum(geo, seed=178)
+    eps = 1e-6
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    Sp, _ = hisq_fermion_force(up, geo, mass, phi, cg_tol=1e-12)
+    Sm, _ = hisq_fermion_force(um, geo, mass, phi, cg_tol=1e-12)
+    dSdt = (Sp - Sm) / (2 * eps)
+    _, F = hisq_fermion_force(u, geo, mass, phi, cg_tol=1e-12)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
