@@ -7,6 +7,8 @@ from .ca import ca_cg_solve, ca_gcr_solve
 from .cg import SolverStats, cg_solve
 from .gcr import gcr_solve, mr_solve
 from .mre import ChronoForecaster
+from .rational import (RationalApprox, rational_approx, rational_apply,
+                       rhmc_pseudofermion_action)
 from .multishift import multishift_cg_solve
 from .variants import (cg3_solve, cgne_solve, cgnr_solve, pcg_solve,
                        sd_solve)
@@ -41,4 +43,5 @@ __all__ = ["cg_solve", "SolverStats", "bicgstab_solve", "bicgstabl_solve",
            "gcr_solve", "mr_solve", "ca_cg_solve", "ca_gcr_solve",
            "multishift_cg_solve", "cgne_solve", "cgnr_solve", "cg3_solve",
            "sd_solve", "pcg_solve", "ChronoForecaster", "create_solver",
-           "SOLVERS", "block_cg_solve"]
+           "SOLVERS", "block_cg_solve", "RationalApprox", "rational_approx",
+           "rational_apply", "rhmc_pseudofermion_action"]

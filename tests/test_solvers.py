@@ -251,3 +251,66 @@ def test_block_cg(system):
     x1 = SpinorField(geo, "double", n_parity=1)
     st1 = cg_solve(d, x1, bs[0], tol=1e-9, maxiter=300)
     assert st.iters <= st1.iters + 1
+
+
+def test_rational_approx_accuracy():
+    import numpy as np
+    from quda_amd.solvers.rational import rational_approx
+    for alpha in (-0.5, 0.5, -0.25, 0.25):
+        r = rational_approx(alpha, 1e-3, 10.0, n=14)
+        assert r.max_rel_err < 5e-6, (alpha, r.max_rel_err)
+    # accuracy improves with n (exponentially for log-spaced poles)
+    e8 = rational_approx(-0.5, 1e-3, 10.0, n=8).max_rel_err
+    e16 = rational_approx(-0.5, 1e-3, 10.0, n=16).max_rel_err
+    assert e16 < e8 / 100
+
+
+def test_rational_apply_matches_dense(system):
+    """A^{-1/2} phi via multishift == dense eigendecomposition."""
+    import numpy as np
+    import torch
+    from quda_amd.solvers.rational import rational_approx, rational_apply
+    geo = LatticeGeometry((2, 2, 2, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=330)
+    d = DiracWilsonPC(g, 0.1)
+    # dense A for the truth
+    dim = geo.volume_cb * 12
+    A = np.zeros((dim, dim), dtype=complex)
+    xb = SpinorField(geo, "double", n_parity=1)
+    yb = SpinorField(geo, "double", n_parity=1)
+    tb = SpinorField(geo, "double", n_parity=1)
+    for j in range(dim):
+        c = torch.zeros(dim, dtype=torch.complex128)
+        c[j] = 1.0
+        xb.from_complex(c.reshape(1, geo.volume_cb, 4, 3))
+        d.MdagM(yb, xb, tb)
+        A[:, j] = yb.to_complex().reshape(-1).numpy()
+    w, V = np.linalg.eigh(A)
+    phi = SpinorField(geo, "double", n_parity=1).gaussian_(seed=331)
+    pv = phi.to_complex().reshape(-1).numpy()
+    truth = V @ (np.diag(w ** -0.5) @ (V.conj().T @ pv))
+    ap = rational_approx(-0.5, w.min() * 0.8, w.max() * 1.2, n=14)
+    out = SpinorField(geo, "double", n_parity=1)
+    rational_apply(d, out, phi, ap, tol=1e-12, maxiter=3000)
+    got = out.to_complex().reshape(-1).numpy()
+    rel = np.abs(got - truth).max() / np.abs(truth).max()
+    assert rel < 1e-6, rel
+
+
+def test_rhmc_action_consistency(system):
+    """phi^dag A^{-1/2} phi > 0 and matches <phi, A^{-1/2} phi> dense-free
+    sanity via A^{-1/4}(A^{-1/4}) composition."""
+    from quda_amd.solvers.rational import (rational_approx, rational_apply,
+                                           rhmc_pseudofermion_action)
+    geo, g, cl, _, b_e = system
+    d = DiracCloverPC(g, cl, KAPPA)
+    ap_half = rational_approx(-0.5, 1e-2, 20.0, n=14)
+    ap_quarter = rational_approx(-0.25, 1e-2, 20.0, n=14)
+    s = rhmc_pseudofermion_action(d, b_e, ap_half, tol=1e-11)
+    assert s > 0
+    t1 = b_e.clone_empty()
+    rational_apply(d, t1, b_e, ap_quarter, tol=1e-11)
+    t2 = b_e.clone_empty()
+    rational_apply(d, t2, t1, ap_quarter, tol=1e-11)
+    s2 = blas.re_dot(b_e, t2)
+    assert abs(s - s2) < 1e-5 * abs(s)
