@@ -34,6 +34,7 @@ def wilson_fermion_force(u: torch.Tensor, geo: LatticeGeometry, kappa: float,
                          X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
     """F_mu(x) = kappa TA[U_mu(x) G1 - G2 U_mu(x)^dag]; X/Y are full-parity
     oracle fields [2,Vcb,4,3]."""
+    from ..parallel.halo import shift_lex
     dev, dt = u.device, u.dtype
     P = _gamma_tensors(dev, dt)  # [mu, 0:minus/1:plus, 4, 4]
     U = _to_lex(u, geo)
@@ -41,11 +42,12 @@ def wilson_fermion_force(u: torch.Tensor, geo: LatticeGeometry, kappa: float,
     Yl = checkerboard_join(Y, geo)
     F = torch.empty_like(U)
     for mu in range(4):
-        idx = geo.neighbor_lex(mu, +1).to(dev)
-        PmY_xmu = torch.einsum("st,vtc->vsc", P[mu, 0], Yl[idx])
+        Y_xmu = shift_lex(Yl, geo, mu, +1)   # neighbor-rank slab when split
+        X_xmu = shift_lex(Xl, geo, mu, +1)
+        PmY_xmu = torch.einsum("st,vtc->vsc", P[mu, 0], Y_xmu)
         PpY = torch.einsum("st,vtc->vsc", P[mu, 1], Yl)
         G1 = torch.einsum("vsc,vsa->vca", PmY_xmu, Xl.conj())
-        G2 = torch.einsum("vsc,vsb->vcb", PpY, Xl[idx].conj())
+        G2 = torch.einsum("vsc,vsb->vcb", PpY, X_xmu.conj())
         F[mu] = kappa * project_ta(U[mu] @ G1 - G2 @ U[mu].conj().mT)
     return _from_lex(F, geo)
 
@@ -63,7 +65,8 @@ def fermion_action_and_force(u: torch.Tensor, geo: LatticeGeometry,
                              kappa: float, phi: SpinorField, *,
                              cg_tol: float = 1e-10, cg_maxiter: int = 2000
                              ) -> Tuple[float, torch.Tensor]:
-    """S_f = phi^dag (M Mdag)^{-1} phi and its MD force."""
+    """S_f = phi^dag (M Mdag)^{-1} phi (global) and its MD force
+    (multi-rank: the analytic Wilson force uses distributed shifts)."""
     g = GaugeField(geo, "double", phi.device).from_complex(u)
     d = DiracWilson(g, kappa)
     X = d.new_spinor(n_parity=2)

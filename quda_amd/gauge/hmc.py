@@ -48,9 +48,11 @@ def random_momentum(geo: LatticeGeometry, device="cpu", seed: Optional[int] = No
 
 
 def mom_action(P: torch.Tensor) -> float:
-    """S_mom = -sum tr(P^2) (P antihermitian => -tr P^2 >= 0;
-    ref: lib/momentum.cu momAction)."""
-    return -torch.einsum("dpvij,dpvji->", P, P).real.item()
+    """S_mom = -sum tr(P^2), globally summed (P antihermitian => -tr P^2
+    >= 0; ref: lib/momentum.cu momAction)."""
+    from ..parallel import comms
+    return comms.allreduce_sum(
+        -torch.einsum("dpvij,dpvji->", P, P).real.item())
 
 
 def leapfrog(u: torch.Tensor, P: torch.Tensor, geo: LatticeGeometry,

@@ -341,3 +341,70 @@ def test_gauge_sector_multiproc_gloo():
         init_file = f.name
     os.unlink(init_file)
     mp.spawn(_worker_gauge, args=(2, init_file), nprocs=2, join=True)
+
+
+# ---------------------------------------------------------------------------
+# multi-rank HMC (gauge + 2-flavor Wilson pseudofermions)
+# ---------------------------------------------------------------------------
+
+def _worker_hmc(rank, world, init_file):
+    import torch.distributed as dist
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=(1, 1, 1, world))
+        from quda_amd.fields.gauge import project_su3
+        from quda_amd.gauge import (gauge_action, mom_action, random_momentum)
+        from quda_amd.gauge.fermion_force import (fermion_action_and_force,
+                                                  pseudofermion_refresh)
+        from quda_amd.gauge.ops import _from_lex, _to_lex, exp_su3
+        from quda_amd.models import DiracWilson
+        geo = LatticeGeometry((4, 4, 4, 4))  # local; global T = 4*world
+        gen = torch.Generator().manual_seed(400 + rank)
+        eye = torch.eye(3, dtype=torch.complex128)
+        m = eye + 0.25 * torch.view_as_complex(
+            torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                        dtype=torch.float64))
+        u = project_su3(m)
+        g = GaugeField(geo, "double").from_complex(u)  # exchanges boundaries
+        u = g.to_complex()
+        beta, kappa = 5.5, 0.11
+        phi = pseudofermion_refresh(DiracWilson(g, kappa), seed=500 + rank)
+        P = random_momentum(geo, seed=600 + rank)
+
+        def total_force(uc):
+            from quda_amd.gauge import gauge_force
+            Sf, Ff = fermion_action_and_force(uc, geo, kappa, phi,
+                                              cg_tol=1e-11)
+            return gauge_force(uc, geo, beta) + Ff
+
+        def hamiltonian(uc, Pc):
+            Sf, _ = fermion_action_and_force(uc, geo, kappa, phi,
+                                             cg_tol=1e-11)
+            return mom_action(Pc) + gauge_action(uc, geo, beta) + Sf
+
+        dHs = []
+        for n in (6, 12):
+            dt = 0.2 / n
+            uc, Pc = u.clone(), P.clone()
+            H0 = hamiltonian(uc, Pc)
+            Pc = Pc + 0.5 * dt * total_force(uc)
+            for k in range(n):
+                U = _to_lex(uc, geo)
+                U = exp_su3(_to_lex(Pc, geo), dt) @ U
+                uc = _from_lex(U, geo)
+                Pc = Pc + (0.5 if k == n - 1 else 1.0) * dt * total_force(uc)
+            dHs.append(abs(hamiltonian(uc, Pc) - H0))
+        # O(dt^2) integrator on the DISTRIBUTED lattice: halving dt must
+        # shrink dH ~4x; loose factor for the small trajectory
+        assert dHs[1] < dHs[0] / 2.5, (rank, dHs)
+        assert dHs[1] < 1.0, (rank, dHs)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_hmc_2f_multirank_energy_conservation():
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        init_file = f.name
+    os.unlink(init_file)
+    mp.spawn(_worker_hmc, args=(2, init_file), nprocs=2, join=True)
