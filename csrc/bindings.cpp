@@ -290,6 +290,53 @@ static void dwf5(at::Tensor out, at::Tensor out_n, at::Tensor in,
   check_launch("dwf5");
 }
 
+static void zdwf5(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                  at::Tensor in_n, at::Tensor x, at::Tensor x_n,
+                  int64_t Vcb4, int64_t Ls, bool xpay, double a_re,
+                  double a_im, int64_t kind,
+                  std::vector<double> au, std::vector<double> al,
+                  std::vector<double> wu, std::vector<double> wl,
+                  std::vector<int64_t> su, std::vector<int64_t> sl,
+                  std::vector<int64_t> ord_u, std::vector<int64_t> ord_l,
+                  std::vector<double> diu, std::vector<double> eu,
+                  std::vector<double> dil, std::vector<double> el,
+                  double cwu_re, double cwu_im, double cwl_re, double cwl_im) {
+  TORCH_CHECK(Ls <= QA_ZMAX, "zMobius Ls > ", QA_ZMAX);
+  ZCoef zc{};
+  auto put2 = [&](double dst[][2], const std::vector<double> &v) {
+    for (int s = 0; s < (int)Ls && 2 * s + 1 < (int)v.size(); ++s) {
+      dst[s][0] = v[2 * s];
+      dst[s][1] = v[2 * s + 1];
+    }
+  };
+  put2(zc.au, au); put2(zc.al, al); put2(zc.wu, wu); put2(zc.wl, wl);
+  put2(zc.diu, diu); put2(zc.eu, eu); put2(zc.dil, dil); put2(zc.el, el);
+  for (int s = 0; s < (int)Ls; ++s) {
+    if (s < (int)su.size()) zc.su[s] = (int)su[s];
+    if (s < (int)sl.size()) zc.sl[s] = (int)sl[s];
+    if (s < (int)ord_u.size()) zc.ord_u[s] = (int)ord_u[s];
+    if (s < (int)ord_l.size()) zc.ord_l[s] = (int)ord_l[s];
+  }
+  zc.cwu[0] = cwu_re; zc.cwu[1] = cwu_im;
+  zc.cwl[0] = cwl_re; zc.cwl[1] = cwl_im;
+  ZDwf5Call c{};
+  long stride = Vcb4 * Ls;
+  c.out = field_of(out, out_n, stride);
+  c.in = field_of(in, in_n, stride);
+  c.x = field_of(x, x_n, stride);
+  c.Vcb4 = Vcb4;
+  c.Ls = (int)Ls;
+  c.xpay = xpay;
+  c.a_re = a_re;
+  c.a_im = a_im;
+  c.prec = prec_of(out);
+  c.kind = (int)kind;
+  TORCH_CHECK(c.prec != 2, "zdwf5: half precision unsupported");
+  c.zc = &zc;
+  launch_zdwf5(c, stream());
+  check_launch("zdwf5");
+}
+
 static void set_dslash_block(int64_t b) {
   if (b == 64 || b == 128 || b == 256) qa_dslash_block_ref() = (int)b;
 }
@@ -313,6 +360,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("dwf5", &dwf5, "DWF/Moebius 5th-dim ops (Ds apply / M5 inverse)");
+  m.def("zdwf5", &zdwf5, "zMobius per-slice-complex 5th-dim ops");
   m.def("blas_op", &blas_op, "fused blas/reduction",
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
         py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),

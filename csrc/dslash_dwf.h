@@ -19,6 +19,7 @@
 #pragma once
 
 #include "common.h"
+#include "launchers.h"
 
 template <typename Prec, bool XPAY, bool DAG>
 __global__ __launch_bounds__(256) void k_dslash5(
@@ -145,5 +146,137 @@ __global__ __launch_bounds__(256) void k_m5inv(
     }
     out.store_v(vin, (long)s * Vcb4 + g);
     kp = kp / kap;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// zMobius: complex per-slice coefficients b5[s], c5[s] (ref: the zMobius
+// branch of dslash_domain_wall_m5.cuh / dslash5_domain_wall.cu — redesigned:
+// the HOST assembles, per chirality block, the hop source/weight tables for
+// the 5th-dim hop operator and the sequence-ordered bidiagonal (1/diag,
+// off-coef, corner) for its inverse; the kernels are generic array-driven
+// O(Ls) solvers with no in-kernel index logic. Ls <= 32.)
+// ---------------------------------------------------------------------------
+template <typename R>
+__device__ __forceinline__ cplx<R> zcc(const double c[2]) {
+  return {(R)c[0], (R)c[1]};
+}
+
+template <typename Prec, bool XPAY>
+__global__ __launch_bounds__(256) void k_zdslash5(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, SpinorAcc<Prec> x, long Vcb4,
+    int Ls, typename Prec::Real ar, typename Prec::Real ai,
+    const ZCoef *__restrict__ zcp) {
+  using R = typename Prec::Real;
+  const ZCoef &zc = *zcp;
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long n = Vcb4 * Ls;
+  if (tid >= n) return;
+  long g = tid % Vcb4;
+  int s = (int)(tid / Vcb4);
+  cplx<R> vin[12], vu[12], vl[12], res[12];
+  in.load_v(vin, (long)s * Vcb4 + g);
+  in.load_v(vu, (long)zc.su[s] * Vcb4 + g);
+  in.load_v(vl, (long)zc.sl[s] * Vcb4 + g);
+  cplx<R> au = zcc<R>(zc.au[s]), al = zcc<R>(zc.al[s]);
+  cplx<R> wu = zcc<R>(zc.wu[s]), wl = zcc<R>(zc.wl[s]);
+#pragma unroll
+  for (int k = 0; k < 6; ++k) res[k] = cfma(au, vin[k], wu * vu[k]);
+#pragma unroll
+  for (int k = 6; k < 12; ++k) res[k] = cfma(al, vin[k], wl * vl[k]);
+  if constexpr (XPAY) {
+    cplx<R> a{ar, ai}, xv[12];
+    x.load_v(xv, (long)s * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 12; ++k) res[k] = cfma(a, xv[k], res[k]);
+  }
+  out.store_v(res, (long)s * Vcb4 + g);
+}
+
+// Array-driven bidiagonal+corner solve, one thread per 4-d site. Each pass-1
+// step does a read-modify-write of ONE chirality half of a slice, so `out`
+// must be distinct from `in` (launcher allocates); half precision is NOT
+// supported here (the block-float RMW would round the other half) — the
+// launcher dispatches double/single only.
+template <typename Prec, bool XPAY>
+__global__ __launch_bounds__(256) void k_zm5inv(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, SpinorAcc<Prec> x, long Vcb4,
+    int Ls, typename Prec::Real ar, typename Prec::Real ai,
+    const ZCoef *__restrict__ zcp) {
+  using R = typename Prec::Real;
+  const ZCoef &zc = *zcp;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= Vcb4) return;
+  cplx<R> vin[12], tmp[12];
+  cplx<R> prev_u[6], prev_l[6], zu{(R)0, (R)0}, zl{(R)0, (R)0};
+  // pass 1: forward substitution in sequence order for both chirality
+  // blocks; track z = B^{-1}(corner column) by the same recursion
+  for (int i = 0; i < Ls; ++i) {
+    int su = zc.ord_u[i], sl = zc.ord_l[i];
+    cplx<R> diu = zcc<R>(zc.diu[i]), eu = zcc<R>(zc.eu[i]);
+    cplx<R> dil = zcc<R>(zc.dil[i]), el = zcc<R>(zc.el[i]);
+    in.load_v(vin, (long)su * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 6; ++k)
+      prev_u[k] = (i == 0) ? diu * vin[k]
+                           : diu * (vin[k] - eu * prev_u[k]);
+    zu = (i == 0) ? diu * zcc<R>(zc.cwu) : neg(diu * (eu * zu));
+    in.load_v(vin, (long)sl * Vcb4 + g);
+#pragma unroll
+    for (int k = 6; k < 12; ++k)
+      prev_l[k - 6] = (i == 0) ? dil * vin[k]
+                               : dil * (vin[k] - el * prev_l[k - 6]);
+    zl = (i == 0) ? dil * zcc<R>(zc.cwl) : neg(dil * (el * zl));
+    // stash y' halves (RMW preserves the other chirality half)
+    out.load_v(tmp, (long)su * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 6; ++k) tmp[k] = prev_u[k];
+    out.store_v(tmp, (long)su * Vcb4 + g);
+    out.load_v(tmp, (long)sl * Vcb4 + g);
+#pragma unroll
+    for (int k = 6; k < 12; ++k) tmp[k] = prev_l[k - 6];
+    out.store_v(tmp, (long)sl * Vcb4 + g);
+  }
+  // Sherman-Morrison: y[i] -= z[i] * y'_last / (1 + z_last)
+  cplx<R> one{(R)1, (R)0};
+  cplx<R> den_u = one + zu, den_l = one + zl;
+  cplx<R> fu = ((R)1 / (den_u.re * den_u.re + den_u.im * den_u.im)) * conj(den_u);
+  cplx<R> fl = ((R)1 / (den_l.re * den_l.re + den_l.im * den_l.im)) * conj(den_l);
+  cplx<R> yu_last[6], yl_last[6];
+#pragma unroll
+  for (int k = 0; k < 6; ++k) { yu_last[k] = fu * prev_u[k]; yl_last[k] = fl * prev_l[k]; }
+  // pass 2: apply corner correction (z recomputed) + optional xpay/scale
+  cplx<R> a{ar, ai};
+  zu = {(R)0, (R)0};
+  zl = {(R)0, (R)0};
+  for (int i = 0; i < Ls; ++i) {
+    int su = zc.ord_u[i], sl = zc.ord_l[i];
+    cplx<R> diu = zcc<R>(zc.diu[i]), dil = zcc<R>(zc.dil[i]);
+    zu = (i == 0) ? diu * zcc<R>(zc.cwu) : neg(diu * (zcc<R>(zc.eu[i]) * zu));
+    zl = (i == 0) ? dil * zcc<R>(zc.cwl) : neg(dil * (zcc<R>(zc.el[i]) * zl));
+    out.load_v(tmp, (long)su * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 6; ++k) tmp[k] = tmp[k] - zu * yu_last[k];
+    out.store_v(tmp, (long)su * Vcb4 + g);
+    out.load_v(tmp, (long)sl * Vcb4 + g);
+#pragma unroll
+    for (int k = 6; k < 12; ++k) tmp[k] = tmp[k] - zl * yl_last[k - 6];
+    out.store_v(tmp, (long)sl * Vcb4 + g);
+  }
+  const bool scale = (ar != (R)1 || ai != (R)0);
+  if (XPAY || scale) {
+    for (int s = 0; s < Ls; ++s) {
+      out.load_v(tmp, (long)s * Vcb4 + g);
+      if constexpr (XPAY) {
+        cplx<R> xv[12];
+        x.load_v(xv, (long)s * Vcb4 + g);
+#pragma unroll
+        for (int k = 0; k < 12; ++k) tmp[k] = cfma(a, tmp[k], xv[k]);
+      } else {
+#pragma unroll
+        for (int k = 0; k < 12; ++k) tmp[k] = a * tmp[k];
+      }
+      out.store_v(tmp, (long)s * Vcb4 + g);
+    }
   }
 }

@@ -35,3 +35,38 @@ void launch_dwf5(const Dwf5Call &c, hipStream_t st) {
     case 2: dwf5_t<PrecHalf>(c, st); break;
   }
 }
+
+template <typename Prec>
+static void zdwf5_t(const ZDwf5Call &c, const ZCoef *dzc, hipStream_t st) {
+  using R = typename Prec::Real;
+  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.out.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
+  SpinorAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.x.Vcb};
+  int blk = 256;
+  R ar = (R)c.a_re, ai = (R)c.a_im;
+#define QA_Z5(XPAY)                                                           \
+  if (c.kind == 0) {                                                          \
+    long n = c.Vcb4 * c.Ls;                                                   \
+    hipLaunchKernelGGL((k_zdslash5<Prec, XPAY>),                              \
+                       dim3((int)((n + blk - 1) / blk)), dim3(blk), 0, st,    \
+                       out, in, x, c.Vcb4, c.Ls, ar, ai, dzc);                \
+  } else {                                                                    \
+    hipLaunchKernelGGL((k_zm5inv<Prec, XPAY>),                                \
+                       dim3((int)((c.Vcb4 + blk - 1) / blk)), dim3(blk), 0,   \
+                       st, out, in, x, c.Vcb4, c.Ls, ar, ai, dzc);            \
+  }
+  if (c.xpay) QA_Z5(true) else QA_Z5(false)
+#undef QA_Z5
+}
+
+void launch_zdwf5(const ZDwf5Call &c, hipStream_t st) {
+  // stage the host-assembled table into a persistent device buffer
+  // (stream-ordered: the async copy precedes the kernel on the same stream)
+  static ZCoef *dzc = nullptr;
+  if (!dzc) (void)hipMalloc(&dzc, sizeof(ZCoef));
+  (void)hipMemcpyAsync(dzc, c.zc, sizeof(ZCoef), hipMemcpyHostToDevice, st);
+  switch (c.prec) {
+    case 0: zdwf5_t<PrecDouble>(c, dzc, st); break;
+    case 1: zdwf5_t<PrecSingle>(c, dzc, st); break;
+  }
+}

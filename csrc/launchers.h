@@ -154,3 +154,30 @@ struct Dwf5Call {
   int kind;  // 0 = dslash5 (Ds apply), 1 = m5inv
 };
 void launch_dwf5(const Dwf5Call &c, hipStream_t st);
+
+#define QA_ZMAX 32
+
+struct ZCoef {  // host-filled complex coefficient tables (double re/im)
+  // zdslash5: out(s)_upper = au[s]*in(s) + wu[s]*in(su[s]); lower likewise
+  double au[QA_ZMAX][2], al[QA_ZMAX][2];
+  double wu[QA_ZMAX][2], wl[QA_ZMAX][2];
+  int su[QA_ZMAX], sl[QA_ZMAX];
+  // zm5inv: per chirality a sequence order ord[i] = physical slice of step
+  // i and the recursion y[i] = di[i] * (r[i] - e[i] y[i-1]) with di = 1/d;
+  // Sherman-Morrison corner weight cw couples step 0 to the LAST step.
+  int ord_u[QA_ZMAX], ord_l[QA_ZMAX];
+  double diu[QA_ZMAX][2], eu[QA_ZMAX][2], cwu[2];
+  double dil[QA_ZMAX][2], el[QA_ZMAX][2], cwl[2];
+};
+
+struct ZDwf5Call {
+  BlasField out, in, x;
+  long Vcb4;
+  int Ls;
+  bool xpay;
+  double a_re, a_im;
+  int prec;
+  int kind;  // 0 = zdslash5, 1 = zm5inv (double/single only)
+  const ZCoef *zc;  // HOST pointer; launcher stages to device
+};
+void launch_zdwf5(const ZDwf5Call &c, hipStream_t st);

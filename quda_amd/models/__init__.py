@@ -1,5 +1,7 @@
 from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
                         DiracStaggered, DiracStaggeredPC)
+from .dwf import (DiracDomainWall, DiracDomainWallPC, DiracMobius,
+                  DiracMobiusPC, DiracZMobius, DiracZMobiusPC)
 from .dirac import (Dirac, DiracClover, DiracCloverHasenbuschTwist,
                     DiracCloverHasenbuschTwistPC, DiracCloverPC,
                     DiracNdegTwistedMass, DiracNdegTwistedMassPC,
@@ -13,4 +15,5 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracImprovedStaggered", "DiracImprovedStaggeredPC",
            "DiracNdegTwistedMass", "DiracNdegTwistedMassPC",
            "DiracCloverHasenbuschTwist", "DiracCloverHasenbuschTwistPC",
-           "DiracTwistedCloverPC"]
+           "DiracTwistedCloverPC", "DiracDomainWall", "DiracDomainWallPC",
+           "DiracMobius", "DiracMobiusPC", "DiracZMobius", "DiracZMobiusPC"]

@@ -192,3 +192,49 @@ class DiracDomainWall(DiracMobius):
 class DiracDomainWallPC(DiracMobiusPC):
     def __init__(self, gauge: GaugeField, mf: float, m5: float, Ls: int):
         super().__init__(gauge, mf, m5, Ls, b5=1.0, c5=0.0)
+
+
+class DiracZMobius(DiracMobius):
+    """zMobius: complex per-slice b5[s], c5[s] (ref: lib/dirac_mobius.cpp
+    zMobius branch + dslash5_domain_wall.cu M5_ZMOBIUS — re-derived; the
+    per-slice tables are assembled host-side, see ops.dispatch._ztables).
+
+      A_s = 1 + b5[s](4 - M5),  B hop = c5[s];  M = A - (1/2) Dhat B
+    """
+
+    def __init__(self, gauge: GaugeField, mf: float, m5: float, Ls: int,
+                 b5, c5):
+        b5 = [complex(v) for v in b5]
+        c5 = [complex(v) for v in c5]
+        assert len(b5) == Ls and len(c5) == Ls
+        self.gauge = gauge
+        self.geo = gauge.geo
+        self.mf = float(mf)
+        self.m5 = float(m5)
+        self.Ls = int(Ls)
+        self.b5v = b5
+        self.c5v = c5
+        d4 = 4.0 - self.m5
+        self.alpha5 = [1.0 + b * d4 for b in b5]
+        self.beta5 = [c * d4 - 1.0 for c in c5]
+
+    def apply_A(self, out, inp, dagger=False):
+        from ..ops.dispatch import zdwf5_op
+        return zdwf5_op(out, inp, self.alpha5, self.beta5, self.mf, 0,
+                        dagger=dagger)
+
+    def apply_Ainv(self, out, inp, dagger=False, a=1.0, x=None):
+        from ..ops.dispatch import zdwf5_op
+        return zdwf5_op(out, inp, self.alpha5, self.beta5, self.mf, 1,
+                        dagger=dagger, a=a, x=x)
+
+    def apply_B(self, out, inp, dagger=False):
+        from ..ops.dispatch import zdwf5_op
+        return zdwf5_op(out, inp, self.b5v, self.c5v, self.mf, 0,
+                        dagger=dagger)
+
+
+class DiracZMobiusPC(DiracZMobius, DiracMobiusPC):
+    """Symmetric even-even PC zMobius (M/prepare/reconstruct inherited from
+    DiracMobiusPC; the s-operators from DiracZMobius)."""
+    pass

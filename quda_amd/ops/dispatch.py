@@ -351,6 +351,101 @@ def dwf5_op(out: SpinorField, inp: SpinorField, alpha: float, beta: float,
     return out
 
 
+def _ztables(Ls, diag, hop, mf, dagger):
+    """Host-side zMobius table assembly (csrc ZCoef): hop source/weight per
+    chirality, and the sequence-ordered bidiagonal (1/diag, off, corner)
+    for the inverse. diag[s] + hop[s]*Ds rows; dagger = conjugate transpose
+    (coefficients then attach to the SOURCE slice)."""
+    diag = [complex(d) for d in diag]
+    hop = [complex(h) for h in hop]
+    if not dagger:
+        dg = diag
+        su = [(s - 1) % Ls for s in range(Ls)]
+        wu = [hop[s] * (-mf if s == 0 else 1.0) for s in range(Ls)]
+        sl = [(s + 1) % Ls for s in range(Ls)]
+        wl = [hop[s] * (-mf if s == Ls - 1 else 1.0) for s in range(Ls)]
+    else:
+        dg = [d.conjugate() for d in diag]
+        su = [(s + 1) % Ls for s in range(Ls)]
+        wu = [hop[(s + 1) % Ls].conjugate() * (-mf if s == Ls - 1 else 1.0)
+              for s in range(Ls)]
+        sl = [(s - 1) % Ls for s in range(Ls)]
+        wl = [hop[(s - 1) % Ls].conjugate() * (-mf if s == 0 else 1.0)
+              for s in range(Ls)]
+
+    def seq(src, w):
+        # order the rows so each row's off-diagonal hits the previous
+        # step's variable; the remaining corner gets Sherman-Morrison
+        inv = {src[r]: r for r in range(Ls)}
+        o = [0]
+        for _ in range(1, Ls):
+            o.append(inv[o[-1]])
+        di = [1.0 / dg[r] for r in o]
+        e = [0j] + [w[o[i]] for i in range(1, Ls)]
+        cw = w[o[0]]
+        return o, di, e, cw
+
+    ou, diu, eu, cwu = seq(su, wu)
+    ol, dil, el, cwl = seq(sl, wl)
+
+    def flat(v):
+        out = []
+        for z in v:
+            out.extend((z.real, z.imag))
+        return out
+
+    return dict(au=flat(dg), al=flat(dg), wu=flat(wu), wl=flat(wl),
+                su=su, sl=sl, ord_u=ou, ord_l=ol,
+                diu=flat(diu), eu=flat(eu), dil=flat(dil), el=flat(el),
+                cwu_re=cwu.real, cwu_im=cwu.imag,
+                cwl_re=cwl.real, cwl_im=cwl.imag)
+
+
+_ztable_cache = {}
+
+
+def zdwf5_op(out: SpinorField, inp: SpinorField, diag, hop, mf: float,
+             kind: int, dagger: bool = False, a=1.0,
+             x: Optional[SpinorField] = None):
+    """zMobius per-slice-complex 5th-dimension ops (csrc k_zdslash5 /
+    k_zm5inv; ref: the zMobius branch of lib/dslash5_domain_wall.cu):
+    kind=0: out = [a*x +] (diag + hop*Ds) in
+    kind=1: out = [x +] a * (diag + hop*Ds)^{-1} in"""
+    Ls = inp.ls
+    a = complex(a)
+    xpay = x is not None
+    if on_gpu(out, inp):
+        key = (tuple(complex(d) for d in diag), tuple(complex(h) for h in hop),
+               float(mf), bool(dagger), Ls)
+        t = _ztable_cache.get(key)
+        if t is None:
+            t = _ztables(Ls, diag, hop, float(mf), bool(dagger))
+            if len(_ztable_cache) > 64:
+                _ztable_cache.clear()
+            _ztable_cache[key] = t
+        ext = hip_ext()
+        xf = x if x is not None else out
+        ext.zdwf5(out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+                  xf.data, norm_or_empty(xf), inp.geo.volume_cb, Ls,
+                  xpay, a.real, a.imag, kind,
+                  t["au"], t["al"], t["wu"], t["wl"], t["su"], t["sl"],
+                  t["ord_u"], t["ord_l"], t["diu"], t["eu"], t["dil"],
+                  t["el"], t["cwu_re"], t["cwu_im"], t["cwl_re"],
+                  t["cwl_im"])
+        return out
+    psi = inp.to_complex()[0]
+    if kind == 0:
+        res = ref.zdslash5(psi, Ls, diag, hop, mf, dagger)
+        if xpay:
+            res = a * x.to_complex()[0] + res
+    else:
+        res = a * ref.zm5inv(psi, Ls, diag, hop, mf, dagger)
+        if xpay:
+            res = x.to_complex()[0] + res
+    out.from_complex(res.unsqueeze(0))
+    return out
+
+
 def dwf_halo_exchange(inp: SpinorField, parity_in: int, dagger: bool):
     """Pack + exchange all s-slice faces of a 5-d input (blocking; returns
     the halo object to hand to dslash_wilson_slice, or None when no dim is

@@ -369,3 +369,50 @@ def dslash_staggered_naik_parity(n: torch.Tensor, psi: torch.Tensor,
         bwd = nbr3(parity, mu, -3)
         out -= e * torch.einsum("vji,vj->vi", n[mu, other][bwd].conj(), psi[bwd])
     return out
+
+
+def _zm5_matrix(Ls, diag, hop, mf, upper: bool, dagger: bool):
+    """Dense [Ls,Ls] complex chirality-block matrix of the zMobius
+    s-operator diag[s] + hop[s]*Ds (per-slice complex coefficients;
+    dagger = conjugate transpose of the non-dagger matrix)."""
+    import numpy as np
+    A = np.diag(np.asarray(diag, dtype=complex))
+    for s in range(Ls):
+        src = s - 1 if upper else s + 1
+        w = 1.0
+        if src < 0:
+            src += Ls
+            w = -mf
+        if src >= Ls:
+            src -= Ls
+            w = -mf
+        A[s, src] += complex(hop[s]) * w
+    return A.conj().T if dagger else A
+
+
+def zdslash5(psi5: torch.Tensor, Ls: int, diag, hop, mf: float,
+             dagger: bool = False) -> torch.Tensor:
+    """zMobius s-operator apply via dense chirality-block matrices."""
+    V = psi5.shape[0] // Ls
+    v = psi5.reshape(Ls, V, 4, 3)
+    out = torch.empty_like(v)
+    for upper, sl in ((True, slice(0, 2)), (False, slice(2, 4))):
+        A = torch.tensor(_zm5_matrix(Ls, diag, hop, mf, upper, dagger),
+                         dtype=psi5.dtype, device=psi5.device)
+        out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", A, v[:, :, sl, :])
+    return out.reshape(Ls * V, 4, 3)
+
+
+def zm5inv(psi5: torch.Tensor, Ls: int, diag, hop, mf: float,
+           dagger: bool = False) -> torch.Tensor:
+    """Inverse of the zMobius s-operator via dense solves."""
+    import numpy as np
+    V = psi5.shape[0] // Ls
+    v = psi5.reshape(Ls, V, 4, 3)
+    out = torch.empty_like(v)
+    for upper, sl in ((True, slice(0, 2)), (False, slice(2, 4))):
+        A = _zm5_matrix(Ls, diag, hop, mf, upper, dagger)
+        Ainv = torch.tensor(np.linalg.inv(A), dtype=psi5.dtype,
+                            device=psi5.device)
+        out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", Ainv, v[:, :, sl, :])
+    return out.reshape(Ls * V, 4, 3)
