@@ -133,7 +133,8 @@ static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
 static void pack_face_stag(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
                            at::Tensor in_n, std::vector<int64_t> dims,
                            int64_t parity_offset, int64_t Vcb, int64_t parity,
-                           int64_t mu, int64_t edge, int64_t Fcb) {
+                           int64_t mu, int64_t edge, int64_t Fcb,
+                           int64_t depth) {
   PackCall c{};
   c.in = field_of(in, in_n, Vcb);
   c.dst = dst.data_ptr();
@@ -146,6 +147,7 @@ static void pack_face_stag(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
   c.edge = (int)edge;
   c.Fcb = Fcb;
   c.prec = prec_of(in);
+  c.depth = (int)depth;
   launch_pack_face_stag(c, stream());
   check_launch("pack_face_stag");
 }
@@ -159,7 +161,7 @@ static void dslash_staggered(at::Tensor out, at::Tensor out_n, at::Tensor in,
                              int64_t recon, std::vector<at::Tensor> ghost,
                              std::vector<at::Tensor> ghost_nrm,
                              std::vector<int64_t> face_cb, int64_t comm_mask,
-                             int64_t kt) {
+                             int64_t kt, int64_t ghost_depth) {
   TORCH_CHECK(out.is_contiguous() && in.is_contiguous() && gauge.is_contiguous());
   StagDslashCall c{};
   c.out = field_of(out, out_n, Vcb);
@@ -178,6 +180,7 @@ static void dslash_staggered(at::Tensor out, at::Tensor out_n, at::Tensor in,
   c.comm_mask = (int)comm_mask;
   c.kt = (int)kt;
   c.prec = prec_of(out);
+  c.ghost_depth = (int)ghost_depth;
   if (comm_mask) {
     TORCH_CHECK(ghost.size() == 8 && ghost_nrm.size() == 8 && face_cb.size() == 4);
     for (int k = 0; k < 8; ++k) {

@@ -101,18 +101,31 @@ __global__ __launch_bounds__(256) void k_dslash_staggered(
   QA_SDIR(3)
 #undef QA_SDIR
 
-  // Naik 3-hop long-link term (improved staggered; single-rank local wrap,
-  // nFace=3 halos land with the multi-rank HISQ support)
+  // Naik 3-hop long-link term (improved staggered). Multi-rank runs the
+  // FUSED policy with depth-3 ghosts (gh.depth == 3): a 3-hop from layer
+  // k of the boundary lands in ghost layer l (flat index l*Fcb + ghost_idx);
+  // interior/exterior overlap for the Naik term is deferred (the dispatch
+  // forces the fused policy when long links are present and dims are cut).
   if constexpr (IMP) {
     int esum2 = 0;
 #define QA_LDIR(MU)                                                       \
     {                                                                     \
       R eta = (esum2 & 1) ? (R)-1 : (R)1;                                 \
-      in.load_v(p, neighbor_cb3(xc, MU, +1, d));                          \
+      int lp = xc[MU] + 3 - d.X[MU];  /* fwd ghost layer if >= 0 */       \
+      if (KT == KT_FUSED && gh.active(MU) && lp >= 0) {                   \
+        gh.load_v(p, MU, 1, (long)lp * gh.Fcb[MU] + ghost_idx(xc, MU, d)); \
+      } else {                                                            \
+        in.load_v(p, neighbor_cb3(xc, MU, +1, d));                        \
+      }                                                                   \
       lng.template load<MU>(U, i);                                        \
       su3_mul_vec(up, U, p);                                              \
       for (int c = 0; c < 3; ++c) acc[c] += eta * up[c];                  \
-      in.load_v(p, neighbor_cb3(xc, MU, -1, d));                          \
+      int lm = 2 - xc[MU];            /* bwd ghost layer if >= 0 */       \
+      if (KT == KT_FUSED && gh.active(MU) && lm >= 0) {                   \
+        gh.load_v(p, MU, 0, (long)lm * gh.Fcb[MU] + ghost_idx(xc, MU, d)); \
+      } else {                                                            \
+        in.load_v(p, neighbor_cb3(xc, MU, -1, d));                        \
+      }                                                                   \
       lng.template load<4 + MU>(U, i);                                    \
       su3_dagmul_vec(up, U, p);                                           \
       for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \
@@ -214,18 +227,22 @@ __global__ __launch_bounds__(256) void k_dslash_staggered_exterior(
   out.store_v(prev, i);
 }
 
-// staggered face pack: full site (6 reals), no projection
+// staggered face pack: full site (6 reals), no projection. depth layers:
+// layer l packs coord l (low edge) / X-1-l (high edge) into flat slot
+// l*Fcb + f with chunk stride depth*Fcb (matches GhostAcc.depth loads).
 template <typename Prec, bool EDGE>
 __global__ __launch_bounds__(256) void k_pack_face_stag(
     typename Prec::Store *dst, float *dst_nrm, StagAcc<Prec> in, LatDims d,
-    int parity, int mu, long Fcb) {
+    int parity, int mu, long Fcb, int depth) {
   using R = typename Prec::Real;
-  long f = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (f >= Fcb) return;
+  long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= Fcb * depth) return;
+  int l = (int)(t / Fcb);
+  long f = t % Fcb;
   int xc[4];
-  face_coords(xc, f, mu, EDGE ? d.X[mu] - 1 : 0, d, parity);
+  face_coords(xc, f, mu, EDGE ? d.X[mu] - 1 - l : l, d, parity);
   long i = cb_from_coords(xc, d);
   cplx<R> v[3];
   in.load_v(v, i);
-  ghost_store_v<Prec, 6>(dst, dst_nrm, Fcb, f, v);
+  ghost_store_v<Prec, 6>(dst, dst_nrm, (long)depth * Fcb, t, v);
 }

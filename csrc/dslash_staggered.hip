@@ -24,6 +24,7 @@ static void stag_launch(const StagDslashCall &c, hipStream_t st) {
     gh.nrm[k] = c.ghost_nrm[k];
   }
   for (int k = 0; k < 4; ++k) gh.Fcb[k] = c.face_cb[k];
+  gh.depth = c.ghost_depth > 0 ? c.ghost_depth : 1;
   int blk = 256;
   int grid = (int)((c.Vcb + blk - 1) / blk);
   R a = (R)c.a, b = (R)c.b;
@@ -74,14 +75,18 @@ static void stag_pack(const PackCall &c, hipStream_t st) {
   StagAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.Vcb};
   LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
   int blk = 256;
-  int grid = (int)((c.Fcb + blk - 1) / blk);
+  int depth = c.depth > 0 ? c.depth : 1;
+  long n = c.Fcb * depth;
+  int grid = (int)((n + blk - 1) / blk);
   auto *dst = (typename Prec::Store *)c.dst;
   if (c.edge)
     hipLaunchKernelGGL((k_pack_face_stag<Prec, true>), dim3(grid), dim3(blk),
-                       0, st, dst, c.dst_nrm, in, d, c.parity, c.mu, c.Fcb);
+                       0, st, dst, c.dst_nrm, in, d, c.parity, c.mu, c.Fcb,
+                       depth);
   else
     hipLaunchKernelGGL((k_pack_face_stag<Prec, false>), dim3(grid), dim3(blk),
-                       0, st, dst, c.dst_nrm, in, d, c.parity, c.mu, c.Fcb);
+                       0, st, dst, c.dst_nrm, in, d, c.parity, c.mu, c.Fcb,
+                       depth);
 }
 
 void launch_pack_face_stag(const PackCall &c, hipStream_t st) {

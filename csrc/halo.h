@@ -57,6 +57,7 @@ struct GhostAcc {
   const float *nrm[8];  // half only
   long Fcb[4];
   int mask;             // bit mu set => dim mu partitioned
+  int depth = 1;        // ghost layers per (mu,dir) (3 for Naik long links)
 
   __device__ __forceinline__ bool active(int mu) const { return (mask >> mu) & 1; }
 
@@ -64,9 +65,10 @@ struct GhostAcc {
                                          long f) const {
     const S *b = buf[2 * mu + dir];
     S tmp[NCOMP];
+    long stride = (long)depth * Fcb[mu];
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch)
-      load_chunk<S, GW>(b + ((long)ch * Fcb[mu] + f) * GW, tmp + ch * GW);
+      load_chunk<S, GW>(b + ((long)ch * stride + f) * GW, tmp + ch * GW);
     R scale = (R)1;
     if constexpr (Prec::has_norm) scale = nrm[2 * mu + dir][f];
 #pragma unroll

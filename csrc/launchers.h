@@ -95,6 +95,7 @@ struct PackCall {
   int edge;           // 0: x_mu = 0 face, 1: x_mu = X-1 face
   long Fcb;
   int prec;
+  int depth = 1;      // ghost layers (staggered Naik = 3)
 };
 void launch_pack_face_double(const PackCall &c, hipStream_t st);
 void launch_pack_face_single(const PackCall &c, hipStream_t st);
@@ -138,6 +139,7 @@ struct StagDslashCall {
   int comm_mask;
   int kt;  // 0 local, 1 fused, 2 interior, 3 exterior
   int prec;
+  int ghost_depth;  // 1, or 3 when long links cross rank boundaries
 };
 void launch_dslash_staggered(const StagDslashCall &c, hipStream_t st);
 void launch_pack_face_stag(const PackCall &c, hipStream_t st);

@@ -120,7 +120,10 @@ class GaugeField:
                 bwd = u[mu, (1 - p) if self.shift % 2 else p][bwd_idx][:, 0:nrows, :]
                 flat[p, :, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
                     bwd).reshape(V, L)
-        # fix up bwd slots on partitioned-dim boundary faces
+        # fix up bwd slots on partitioned-dim boundary faces: the x_mu = l
+        # (l < shift) sites need U_mu(x - shift*mu) owned by the -mu
+        # neighbor at its coord X-shift+l (one exchange per layer; shift=3
+        # long links exchange three layers).
         from ..parallel import comms
         from ..parallel.halo import active_dims, exchange_tensors
         mask = comms.comm_mask()
@@ -128,34 +131,38 @@ class GaugeField:
             geo = self.geo
             hi = geo.dims[mu] - 1
             fcb = geo.face_volume_cb(mu)
-            # send my far-face U_mu rows (both parities) toward +mu
-            send = torch.empty((2, fcb, nrows, 3), dtype=u.dtype, device=dev)
-            for q in (0, 1):
-                send[q] = u[mu, q][geo.face_index_cb(q, mu, hi).to(dev)][:, 0:nrows, :]
-            recv = torch.empty_like(send)
-            exchange_tensors({(mu, 1): send}, {(mu, 0): recv})
-            for p in (0, 1):
-                fidx0 = geo.face_index_cb(p, mu, 0).to(dev)
-                flat[p, fidx0, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
-                    recv[1 - p]).reshape(fcb, L)
+            for l in range(self.shift):
+                c_src = hi - (self.shift - 1) + l
+                send = torch.empty((2, fcb, nrows, 3), dtype=u.dtype,
+                                   device=dev)
+                for q in (0, 1):
+                    send[q] = u[mu, q][geo.face_index_cb(q, mu, c_src)
+                                       .to(dev)][:, 0:nrows, :]
+                recv = torch.empty_like(send)
+                exchange_tensors({(mu, 1): send}, {(mu, 0): recv})
+                for p in (0, 1):
+                    fidx = geo.face_index_cb(p, mu, l).to(dev)
+                    q = (1 - p) if self.shift % 2 else p
+                    flat[p, fidx, (4 + mu) * L:(5 + mu) * L] = \
+                        torch.view_as_real(recv[q]).reshape(fcb, L)
         w = WIDTH_OF[self.precision]
         native = flat.reshape(2, V, (8 * L) // w, w).movedim(2, 1).contiguous()
         self.data.copy_(native.to(self.data.dtype))
         self._bwd_ghost_cache = {}
         return self
 
-    def bwd_ghost(self, mu: int, parity: int) -> torch.Tensor:
-        """[Fcb, 3, 3] complex U_mu(x-mu) for the x_mu=0 face sites of
-        `parity`, in ghost order — decoded from the stored bwd slots (used
-        by the CPU oracle on partitioned dims; exact same values the HIP
-        kernel reads)."""
+    def bwd_ghost(self, mu: int, parity: int, coord: int = 0) -> torch.Tensor:
+        """[Fcb, 3, 3] complex U_mu(x-shift*mu) for the x_mu=coord face
+        sites of `parity`, in ghost order — decoded from the stored bwd
+        slots (used by the CPU oracle on partitioned dims; exact same
+        values the HIP kernel reads)."""
         cache = self.__dict__.setdefault("_bwd_ghost_cache", {})
-        key = (mu, parity)
+        key = (mu, parity, coord)
         if key not in cache:
             L = self.L
             geo = self.geo
             V = geo.volume_cb
-            fidx0 = geo.face_index_cb(parity, mu, 0)
+            fidx0 = geo.face_index_cb(parity, mu, coord)
             d = self.data[parity].to(torch.float64)          # [NCH, V, w]
             flat = d.movedim(0, 1).reshape(V, 8 * L)
             rows = torch.view_as_complex(

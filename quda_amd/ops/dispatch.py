@@ -261,13 +261,17 @@ def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     """Staggered stencil: out = [a*x +] b*(D in); out at `parity`, in at
     the opposite parity (nspin=1 fields). D^dag = -D: pass b=-b for the
     dagger. `long_gauge` (shift=3 stencil field) adds the Naik 3-hop term
-    (improved staggered; single-rank only until nFace=3 halos land)."""
+    (improved staggered). Multi-rank Naik uses depth-3 ghosts and always
+    runs the fused policy (no interior/exterior overlap for 3-hop halos
+    yet); the long-link gauge needs no ghost at apply time — the stencil
+    layout pre-shifts and exchanges bwd links at load time."""
     from ..parallel import comms
     geo = out.geo
     xpay = x is not None
     mask = comms.comm_mask()
-    if long_gauge is not None and mask:
-        raise NotImplementedError("improved staggered needs nFace=3 halos")
+    depth = 3 if (long_gauge is not None and mask) else 1
+    if depth == 3:
+        assert min(geo.dims) >= 4, "Naik halos need local extents >= 4"
     if on_gpu(out, inp):
         ext = hip_ext()
         xf = x if x is not None else out
@@ -280,16 +284,17 @@ def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                 gauge.data, lng, xf.data, norm_or_empty(xf), list(geo.dims),
                 geo.parity_offset, geo.volume_cb, parity, xpay, float(a),
                 float(b), RECON_COMPS[gauge.reconstruct], ghosts, nrms,
-                face_cb, mask if kt else 0, kt)
+                face_cb, mask if kt else 0, kt, depth)
 
         if not mask:
             launch(0)
             return out
         from ..parallel.halo import get_spinor_halo
-        h = get_spinor_halo(geo, inp.precision, inp.device, mask, ncomp=6)
+        h = get_spinor_halo(geo, inp.precision, inp.device, mask, ncomp=6,
+                            depth=depth)
         h.pack(ext, inp, 1 - parity, False)
         ghosts, nrms, face_cb = h.ghost_args()
-        if dslash_policy() == "fused":
+        if depth == 3 or dslash_policy() == "fused":
             h.exchange()
             launch(1, ghosts, nrms, face_cb)
         else:
@@ -313,8 +318,18 @@ def dslash_staggered(out: SpinorField, inp: SpinorField, gauge: GaugeField,
         }
     res = ref.dslash_staggered_parity(u, psi, geo, parity, halo=halo)
     if long_gauge is not None:
+        halo3 = None
+        if mask:
+            from ..parallel.halo import active_dims, exchange_psi_oracle
+            halo3 = {
+                "mask": mask,
+                "psi3": exchange_psi_oracle(psi, geo, 1 - parity, mask,
+                                            depth=3),
+                "n_bwd": {(mu, c): long_gauge.bwd_ghost(mu, parity, c)
+                          for mu in active_dims(mask) for c in range(3)},
+            }
         res = res + ref.dslash_staggered_naik_parity(
-            long_gauge.to_complex(), psi, geo, parity)
+            long_gauge.to_complex(), psi, geo, parity, halo=halo3)
     res = b * res
     if xpay:
         res = a * x.to_complex()[0] + res

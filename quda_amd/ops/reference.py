@@ -347,13 +347,20 @@ def m5inv(psi5: torch.Tensor, Ls: int, alpha: float, beta: float, mf: float,
 
 
 def dslash_staggered_naik_parity(n: torch.Tensor, psi: torch.Tensor,
-                                 geo: LatticeGeometry, parity: int) -> torch.Tensor:
+                                 geo: LatticeGeometry, parity: int,
+                                 halo=None) -> torch.Tensor:
     """Naik 3-hop term: sum_mu eta_mu(x)[N_mu(x) psi(x+3mu)
-    - N_mu(x-3mu)^dag psi(x-3mu)]. n: long links [4,2,V,3,3]."""
+    - N_mu(x-3mu)^dag psi(x-3mu)]. n: long links [4,2,V,3,3].
+
+    halo (partitioned dims): {"mask": int, "psi3": depth-3 ghosts from
+    exchange_psi_oracle(..., depth=3), "n_bwd": {(mu, l): [Fcb,3,3]}
+    long links N_mu(x-3mu) for the x_mu = l face (from
+    GaugeField.bwd_ghost of the shift-3 stencil field)}."""
     dev, dt = psi.device, psi.dtype
     other = 1 - parity
     out = torch.zeros_like(psi)
     eta = staggered_phases(geo, parity).to(dev)
+    mask = halo["mask"] if halo else 0
 
     def nbr3(p, mu, disp):
         c = geo.coords_of_cb(p).to(torch.int64).clone()
@@ -363,11 +370,26 @@ def dslash_staggered_naik_parity(n: torch.Tensor, psi: torch.Tensor,
         return geo.cb_of_lex[lex].to(dev)
 
     for mu in range(4):
+        part = (mask >> mu) & 1
+        Xm = geo.dims[mu]
         e = eta[:, mu].to(dt).unsqueeze(-1)
         fwd = nbr3(parity, mu, +3)
-        out += e * torch.einsum("vij,vj->vi", n[mu, parity], psi[fwd])
+        psi_f = psi[fwd]
+        if part:
+            for l in range(3):  # sites x_mu = Xm-3+l read fwd ghost layer l
+                fidx = geo.face_index_cb(parity, mu, Xm - 3 + l).to(dev)
+                psi_f[fidx] = halo["psi3"][(mu, 1)][l].to(dt)
+        out += e * torch.einsum("vij,vj->vi", n[mu, parity], psi_f)
         bwd = nbr3(parity, mu, -3)
-        out -= e * torch.einsum("vji,vj->vi", n[mu, other][bwd].conj(), psi[bwd])
+        psi_b = psi[bwd]
+        n_b = n[mu, other][bwd]
+        if part:
+            n_b = n_b.clone()
+            for l in range(3):  # sites x_mu = 2-l read bwd ghost layer l
+                fidx = geo.face_index_cb(parity, mu, 2 - l).to(dev)
+                psi_b[fidx] = halo["psi3"][(mu, 0)][l].to(dt)
+                n_b[fidx] = halo["n_bwd"][(mu, 2 - l)].to(dt)
+        out -= e * torch.einsum("vji,vj->vi", n_b.conj(), psi_b)
     return out
 
 

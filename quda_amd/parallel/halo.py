@@ -114,12 +114,13 @@ class SpinorHalo:
     6 (staggered full site)."""
 
     def __init__(self, geo: LatticeGeometry, precision: str, device,
-                 mask: int, ncomp: int = 12):
+                 mask: int, ncomp: int = 12, depth: int = 1):
         from ..fields.layout import DTYPE_OF
         self.geo = geo
         self.precision = precision
         self.mask = mask
         self.ncomp = ncomp
+        self.depth = depth  # ghost layers (3 for the staggered Naik term)
         self.device = torch.device(device)
         self.send: Dict[Key, torch.Tensor] = {}
         self.recv: Dict[Key, torch.Tensor] = {}
@@ -128,7 +129,7 @@ class SpinorHalo:
         gw = ghost_width(ncomp, precision)
         dt = DTYPE_OF[precision]
         for mu in active_dims(mask):
-            fcb = geo.face_volume_cb(mu)
+            fcb = geo.face_volume_cb(mu) * depth
             for d in (0, 1):
                 shape = (ncomp // gw, fcb, gw)
                 self.send[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
@@ -155,7 +156,8 @@ class SpinorHalo:
                                        self.send_nrm.get((mu, d), empty),
                                        inp.data, norm_or_empty(inp),
                                        list(geo.dims), geo.parity_offset,
-                                       geo.volume_cb, parity, mu, edge, fcb)
+                                       geo.volume_cb, parity, mu, edge, fcb,
+                                       self.depth)
                 else:
                     s01 = d ^ (1 if dagger else 0)
                     ext.pack_face(self.send[(mu, d)],
@@ -194,11 +196,12 @@ _HALO_CACHE: Dict[tuple, SpinorHalo] = {}
 
 
 def get_spinor_halo(geo: LatticeGeometry, precision: str, device,
-                    mask: int, ncomp: int = 12) -> SpinorHalo:
-    key = (geo.dims, geo.parity_offset, precision, str(device), mask, ncomp)
+                    mask: int, ncomp: int = 12, depth: int = 1) -> SpinorHalo:
+    key = (geo.dims, geo.parity_offset, precision, str(device), mask, ncomp,
+           depth)
     h = _HALO_CACHE.get(key)
     if h is None:
-        h = SpinorHalo(geo, precision, device, mask, ncomp)
+        h = SpinorHalo(geo, precision, device, mask, ncomp, depth)
         _HALO_CACHE[key] = h
     return h
 
@@ -208,20 +211,27 @@ def get_spinor_halo(geo: LatticeGeometry, precision: str, device,
 # ---------------------------------------------------------------------------
 
 def exchange_psi_oracle(psi: torch.Tensor, geo: LatticeGeometry,
-                        parity_in: int, mask: int) -> Dict[Key, torch.Tensor]:
+                        parity_in: int, mask: int,
+                        depth: int = 1) -> Dict[Key, torch.Tensor]:
     """Exchange full (unprojected) spinor faces of `psi` ([V_cb,4,3] or
     [V_cb,3] complex at parity_in). Returns {(mu,dir): faces} in
-    ghost-index order."""
+    ghost-index order; with depth > 1 each face is [depth, Fcb, ...] where
+    layer l of (mu,1) holds the +mu neighbor's x_mu = l sites and layer l
+    of (mu,0) the -mu neighbor's x_mu = X-1-l sites (Naik nFace=3)."""
     sends, recvs = {}, {}
     for mu in active_dims(mask):
         hi = geo.dims[mu] - 1
-        idx0 = geo.face_index_cb(parity_in, mu, 0)
-        idx1 = geo.face_index_cb(parity_in, mu, hi)
-        sends[(mu, 0)] = psi[idx0].contiguous()
-        sends[(mu, 1)] = psi[idx1].contiguous()
+        lo_layers = [psi[geo.face_index_cb(parity_in, mu, l)]
+                     for l in range(depth)]
+        hi_layers = [psi[geo.face_index_cb(parity_in, mu, hi - l)]
+                     for l in range(depth)]
+        sends[(mu, 0)] = torch.stack(lo_layers).contiguous()
+        sends[(mu, 1)] = torch.stack(hi_layers).contiguous()
         recvs[(mu, 0)] = torch.empty_like(sends[(mu, 1)])
         recvs[(mu, 1)] = torch.empty_like(sends[(mu, 0)])
     exchange_tensors(sends, recvs)
+    if depth == 1:
+        recvs = {k: v[0] for k, v in recvs.items()}
     return recvs
 
 
