@@ -367,3 +367,79 @@ def test_zmobius_pc_cg_gpu(setup):
     import math as _m
     tr = _m.sqrt(blas.xmy_norm2(b, r) / blas.norm2(b))
     assert tr < 1e-8, tr
+
+
+# ---------------------------------------------------------------------------
+# MADWF (Ls -> Ls' accelerated solver)
+# ---------------------------------------------------------------------------
+
+def test_madwf_transfer_adjoint(setup):
+    from quda_amd.solvers.madwf import TransferLs
+    geo, _ = setup
+    T = TransferLs(LS, LS // 2)
+    # randomize so the adjoint check is non-trivial
+    gen = torch.Generator().manual_seed(320)
+    T.Tp = torch.randn((LS // 2, LS, 2), generator=gen,
+                       dtype=torch.float64)
+    T.Tp = torch.view_as_complex(T.Tp)
+    T.Tm = torch.view_as_complex(
+        torch.randn((LS // 2, LS, 2), generator=gen, dtype=torch.float64))
+    big = spin5(geo, 321, n_parity=1)
+    small = SpinorField(geo, "double", n_parity=1, ls=LS // 2).gaussian_(seed=322)
+    tb = SpinorField(geo, "double", n_parity=1, ls=LS // 2)
+    T.apply(tb, big)
+    td = SpinorField(geo, "double", n_parity=1, ls=LS)
+    T.apply_dag(td, small)
+    lhs = (small.to_complex().conj() * tb.to_complex()).sum()
+    rhs = (td.to_complex().conj() * big.to_complex()).sum()
+    assert abs(lhs - rhs) < 1e-12 * abs(lhs)
+
+
+def test_madwf_identity_transfer_exact(setup):
+    """Ls' = Ls with the identity transfer makes the preconditioner an
+    exact solve: outer GCR must converge in one iteration."""
+    from quda_amd.models import DiracMobiusPC
+    from quda_amd.solvers.madwf import TransferLs, madwf_solve
+    geo, g = setup
+    big = DiracMobiusPC(g, MF, M5, LS, b5=1.5, c5=0.5)
+    b = spin5(geo, 323, n_parity=1)
+    x = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st = madwf_solve(big, big, x, b, T=TransferLs(LS, LS), tol=1e-8,
+                     inner_tol=1e-10, inner_maxiter=500)
+    assert st.converged and st.iters <= 2, st.iters
+    r = SpinorField(geo, "double", n_parity=1, ls=LS)
+    big.M(r, x)
+    import math as _m
+    assert _m.sqrt(blas.xmy_norm2(b, r) / blas.norm2(b)) < 1e-7
+
+
+def test_madwf_truncated_solver(setup):
+    """Ls'=Ls/2 truncation: preconditioned solve converges to the same
+    solution, in no more outer iterations than plain GCR."""
+    from quda_amd.models import DiracMobiusPC
+    from quda_amd.solvers import gcr_solve
+    from quda_amd.solvers.madwf import madwf_solve
+    geo, g = setup
+    big = DiracMobiusPC(g, MF, M5, LS, b5=1.5, c5=0.5)
+    small = DiracMobiusPC(g, MF, M5, LS // 2, b5=1.5, c5=0.5)
+    b = spin5(geo, 324, n_parity=1)
+    x0 = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st0 = gcr_solve(big, x0, b, tol=1e-8, maxiter=300)
+    x1 = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st1 = madwf_solve(big, small, x1, b, tol=1e-8, inner_tol=1e-4)
+    assert st0.converged and st1.converged
+    assert st1.iters <= st0.iters + 1, (st1.iters, st0.iters)
+    err = (x1.to_complex() - x0.to_complex()).abs().max().item()
+    assert err < 1e-6, err
+
+
+def test_madwf_training_reduces_chi2(setup):
+    from quda_amd.models import DiracMobiusPC
+    from quda_amd.solvers.madwf import TransferLs, train_transfer
+    geo, g = setup
+    big = DiracMobiusPC(g, MF, M5, LS, b5=1.5, c5=0.5)
+    small = DiracMobiusPC(g, MF, M5, LS // 2, b5=1.5, c5=0.5)
+    T = TransferLs(LS, LS // 2)
+    hist = train_transfer(big, small, T, n_samples=2, iters=10, lr=0.05,
+                          inner_tol=1e-6)
+    assert hist[-1] < hist[0], hist
