@@ -340,6 +340,38 @@ static void zdwf5(at::Tensor out, at::Tensor out_n, at::Tensor in,
   check_launch("zdwf5");
 }
 
+static void eofa5(at::Tensor out, at::Tensor out_n, at::Tensor in,
+                  at::Tensor in_n, at::Tensor x, at::Tensor x_n,
+                  int64_t Vcb4, int64_t Ls, bool xpay, bool dagger, double a,
+                  double alpha, double beta, double mf,
+                  std::vector<double> u, std::vector<double> w, double sh,
+                  int64_t pm, int64_t kind) {
+  TORCH_CHECK(Ls <= 32, "eofa5: Ls > 32");
+  Eofa5Call c{};
+  long stride = Vcb4 * Ls;
+  c.out = field_of(out, out_n, stride);
+  c.in = field_of(in, in_n, stride);
+  c.x = field_of(x, x_n, stride);
+  c.Vcb4 = Vcb4;
+  c.Ls = (int)Ls;
+  c.xpay = xpay;
+  c.dagger = dagger;
+  c.a = a;
+  c.alpha = alpha;
+  c.beta = beta;
+  c.mf = mf;
+  for (int s = 0; s < (int)Ls; ++s) {
+    if (s < (int)u.size()) c.u[s] = u[s];
+    if (s < (int)w.size()) c.w[s] = w[s];
+  }
+  c.sh = sh;
+  c.pm = (int)pm;
+  c.prec = prec_of(out);
+  c.kind = (int)kind;
+  launch_eofa5(c, stream());
+  check_launch("eofa5");
+}
+
 static void set_dslash_block(int64_t b) {
   if (b == 64 || b == 128 || b == 256) qa_dslash_block_ref() = (int)b;
 }
@@ -364,6 +396,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("dwf5", &dwf5, "DWF/Moebius 5th-dim ops (Ds apply / M5 inverse)");
   m.def("zdwf5", &zdwf5, "zMobius per-slice-complex 5th-dim ops");
+  m.def("eofa5", &eofa5, "EOFA rank-1 extended M5 ops");
   m.def("blas_op", &blas_op, "fused blas/reduction",
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
         py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),

@@ -280,3 +280,164 @@ __global__ __launch_bounds__(256) void k_zm5inv(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// EOFA rank-1 extended M5 ops (ref: kernels/dslash_mobius_eofa.cuh +
+// lib/dslash5_mobius_eofa.cu — re-derived: the one-flavor-algorithm 5th-dim
+// blocks are the Moebius M5 plus a rank-1 term sh * P_pm |u><w| in s-space;
+// the inverse is the plain M5 inverse followed by one Sherman-Morrison
+// correction y -= sh * (B^-1 u) (w^dag y) / (1 + sh w^dag B^-1 u), with
+// B^-1 u and the denominator precomputed on the HOST. The dagger swaps
+// u <-> w (host-side) and daggers the base op.)
+// ---------------------------------------------------------------------------
+
+struct EofaVec {  // host-filled (Ls <= 32); real Moebius coefficients
+  double u[QA_ZMAX];    // rank-1 left vector (m5_eofa) / B^-1 u (m5inv_eofa)
+  double w[QA_ZMAX];    // rank-1 right vector
+  double sh;            // shift strength (m5inv: sh/denom folded in)
+  int pm;               // +1: rank-1 acts on upper spins (0,1); -1: lower
+};
+
+// out(s) = [a*x(s) +] alpha in(s) + beta (Ds in)(s) + sh u_s sum_s' w_s' in_pm(s')
+template <typename Prec, bool XPAY, bool DAG>
+__global__ __launch_bounds__(256) void k_m5_eofa(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, SpinorAcc<Prec> x, long Vcb4,
+    int Ls, typename Prec::Real a, typename Prec::Real alpha,
+    typename Prec::Real beta, typename Prec::Real mf, EofaVec ev) {
+  using R = typename Prec::Real;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= Vcb4) return;
+  cplx<R> vin[12], vu[12], vl[12], res[12];
+  // pass A: dot = sum_s w_s in_pm(s)  (one chirality, 6 components)
+  cplx<R> dot[6];
+#pragma unroll
+  for (int k = 0; k < 6; ++k) dot[k] = {(R)0, (R)0};
+  const int off = (ev.pm > 0) ? 0 : 6;
+  for (int s = 0; s < Ls; ++s) {
+    in.load_v(vin, (long)s * Vcb4 + g);
+    R ws = (R)ev.w[s];
+#pragma unroll
+    for (int k = 0; k < 6; ++k) dot[k] = dot[k] + ws * vin[k + off];
+  }
+  // pass B: base Moebius M5 + rank-1 add
+  for (int s = 0; s < Ls; ++s) {
+    int su = DAG ? s + 1 : s - 1;
+    int sl = DAG ? s - 1 : s + 1;
+    R wu = (R)1, wl = (R)1;
+    if (su < 0) { su += Ls; wu = -mf; }
+    if (su >= Ls) { su -= Ls; wu = -mf; }
+    if (sl < 0) { sl += Ls; wl = -mf; }
+    if (sl >= Ls) { sl -= Ls; wl = -mf; }
+    in.load_v(vin, (long)s * Vcb4 + g);
+    in.load_v(vu, (long)su * Vcb4 + g);
+    in.load_v(vl, (long)sl * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 6; ++k) res[k] = alpha * vin[k] + (beta * wu) * vu[k];
+#pragma unroll
+    for (int k = 6; k < 12; ++k) res[k] = alpha * vin[k] + (beta * wl) * vl[k];
+    R c = (R)(ev.sh * ev.u[s]);
+#pragma unroll
+    for (int k = 0; k < 6; ++k) res[k + off] = res[k + off] + c * dot[k];
+    if constexpr (XPAY) {
+      cplx<R> xv[12];
+      x.load_v(xv, (long)s * Vcb4 + g);
+#pragma unroll
+      for (int k = 0; k < 12; ++k) res[k] = a * xv[k] + res[k];
+    }
+    out.store_v(res, (long)s * Vcb4 + g);
+  }
+}
+
+// out(s) = [x(s) +] a * [(M5 + sh P_pm u w^dag)^{-1} in](s); ev.u holds
+// B^-1 u and ev.sh holds sh/denom (host-folded Sherman-Morrison).
+template <typename Prec, bool XPAY, bool DAG>
+__global__ __launch_bounds__(256) void k_m5inv_eofa(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, SpinorAcc<Prec> x, long Vcb4,
+    int Ls, typename Prec::Real a, typename Prec::Real alpha,
+    typename Prec::Real beta, typename Prec::Real mf, EofaVec ev) {
+  using R = typename Prec::Real;
+  long g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= Vcb4) return;
+  const R kap = -beta / alpha;
+  R kpow = (R)1;
+  for (int s = 0; s < Ls; ++s) kpow *= kap;
+  const R denom = (R)1 + mf * kpow;
+  const R inv_alpha = (R)1 / alpha;
+  cplx<R> prev_u[6], prev_l[6], vin[12], vout[12];
+  // passes 1-3: plain M5 inverse into out (no scale/xpay yet)
+  for (int si = 0; si < Ls; ++si) {
+    int s = DAG ? (Ls - 1 - si) : si;
+    in.load_v(vin, (long)s * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 6; ++k) {
+      cplx<R> yu = (si == 0) ? inv_alpha * vin[k]
+                             : inv_alpha * (vin[k] - beta * prev_u[k]);
+      prev_u[k] = yu;
+      vout[k] = yu;
+    }
+#pragma unroll
+    for (int k = 6; k < 12; ++k) vout[k] = vin[k];
+    out.store_v(vout, (long)s * Vcb4 + g);
+  }
+  cplx<R> yu_last[6];
+#pragma unroll
+  for (int k = 0; k < 6; ++k) yu_last[k] = prev_u[k];
+  R kp = kpow / kap;
+  const R cu = mf * beta * inv_alpha / denom;
+  for (int si = 0; si < Ls; ++si) {
+    int s = DAG ? si : (Ls - 1 - si);
+    out.load_v(vin, (long)s * Vcb4 + g);
+#pragma unroll
+    for (int k = 0; k < 6; ++k) vin[k] = vin[k] + (cu * kp) * yu_last[k];
+#pragma unroll
+    for (int k = 6; k < 12; ++k) {
+      cplx<R> yl = (si == 0) ? inv_alpha * vin[k]
+                             : inv_alpha * (vin[k] - beta * prev_l[k - 6]);
+      prev_l[k - 6] = yl;
+      vin[k] = yl;
+    }
+    out.store_v(vin, (long)s * Vcb4 + g);
+    kp = kp / kap;
+  }
+  cplx<R> yl_last[6];
+#pragma unroll
+  for (int k = 0; k < 6; ++k) yl_last[k] = prev_l[k];
+  kp = kpow / kap;
+  for (int si = 0; si < Ls; ++si) {
+    int s = DAG ? (Ls - 1 - si) : si;
+    out.load_v(vin, (long)s * Vcb4 + g);
+#pragma unroll
+    for (int k = 6; k < 12; ++k) vin[k] = vin[k] + (cu * kp) * yl_last[k - 6];
+    out.store_v(vin, (long)s * Vcb4 + g);
+    kp = kp / kap;
+  }
+  // pass 4: Sherman-Morrison on chirality pm, then scale/xpay epilogue
+  const int off = (ev.pm > 0) ? 0 : 6;
+  cplx<R> dot[6];
+#pragma unroll
+  for (int k = 0; k < 6; ++k) dot[k] = {(R)0, (R)0};
+  for (int s = 0; s < Ls; ++s) {
+    out.load_v(vin, (long)s * Vcb4 + g);
+    R ws = (R)ev.w[s];
+#pragma unroll
+    for (int k = 0; k < 6; ++k) dot[k] = dot[k] + ws * vin[k + off];
+  }
+  for (int s = 0; s < Ls; ++s) {
+    out.load_v(vin, (long)s * Vcb4 + g);
+    R c = (R)(ev.sh * ev.u[s]);  // sh/denom * (B^-1 u)_s
+#pragma unroll
+    for (int k = 0; k < 6; ++k) vin[k + off] = vin[k + off] - c * dot[k];
+    if constexpr (XPAY) {
+      cplx<R> xv[12];
+      x.load_v(xv, (long)s * Vcb4 + g);
+#pragma unroll
+      for (int k = 0; k < 12; ++k) vin[k] = xv[k] + a * vin[k];
+    } else {
+      if (a != (R)1) {
+#pragma unroll
+        for (int k = 0; k < 12; ++k) vin[k] = a * vin[k];
+      }
+    }
+    out.store_v(vin, (long)s * Vcb4 + g);
+  }
+}

@@ -70,3 +70,40 @@ void launch_zdwf5(const ZDwf5Call &c, hipStream_t st) {
     case 1: zdwf5_t<PrecSingle>(c, dzc, st); break;
   }
 }
+
+template <typename Prec>
+static void eofa5_t(const Eofa5Call &c, hipStream_t st) {
+  using R = typename Prec::Real;
+  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.out.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
+  SpinorAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.x.Vcb};
+  EofaVec ev{};
+  for (int s_ = 0; s_ < c.Ls && s_ < QA_ZMAX; ++s_) {
+    ev.u[s_] = c.u[s_];
+    ev.w[s_] = c.w[s_];
+  }
+  ev.sh = c.sh;
+  ev.pm = c.pm;
+  int blk = 256;
+  int grid = (int)((c.Vcb4 + blk - 1) / blk);
+  R a = (R)c.a, al = (R)c.alpha, be = (R)c.beta, mf = (R)c.mf;
+#define QA_E5(XPAY, DAG)                                                      \
+  if (c.kind == 0)                                                            \
+    hipLaunchKernelGGL((k_m5_eofa<Prec, XPAY, DAG>), dim3(grid), dim3(blk),   \
+                       0, st, out, in, x, c.Vcb4, c.Ls, a, al, be, mf, ev);   \
+  else                                                                        \
+    hipLaunchKernelGGL((k_m5inv_eofa<Prec, XPAY, DAG>), dim3(grid),           \
+                       dim3(blk), 0, st, out, in, x, c.Vcb4, c.Ls, a, al,     \
+                       be, mf, ev);
+  if (c.xpay) { if (c.dagger) QA_E5(true, true) else QA_E5(true, false) }
+  else        { if (c.dagger) QA_E5(false, true) else QA_E5(false, false) }
+#undef QA_E5
+}
+
+void launch_eofa5(const Eofa5Call &c, hipStream_t st) {
+  switch (c.prec) {
+    case 0: eofa5_t<PrecDouble>(c, st); break;
+    case 1: eofa5_t<PrecSingle>(c, st); break;
+    case 2: eofa5_t<PrecHalf>(c, st); break;
+  }
+}

@@ -183,3 +183,19 @@ struct ZDwf5Call {
   const ZCoef *zc;  // HOST pointer; launcher stages to device
 };
 void launch_zdwf5(const ZDwf5Call &c, hipStream_t st);
+
+struct EofaVec;  // defined in dslash_dwf.h
+struct Eofa5Call {
+  BlasField out, in, x;
+  long Vcb4;
+  int Ls;
+  bool xpay;
+  bool dagger;
+  double a, alpha, beta, mf;
+  double u[32], w[32];  // see EofaVec semantics (m5inv: u = B^-1 u, sh/denom)
+  double sh;
+  int pm;
+  int prec;
+  int kind;  // 0 = m5_eofa, 1 = m5inv_eofa
+};
+void launch_eofa5(const Eofa5Call &c, hipStream_t st);

@@ -1,7 +1,8 @@
 from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
                         DiracStaggered, DiracStaggeredPC)
 from .dwf import (DiracDomainWall, DiracDomainWallPC, DiracMobius,
-                  DiracMobiusPC, DiracZMobius, DiracZMobiusPC)
+                  DiracMobiusPC, DiracZMobius, DiracZMobiusPC,
+                  DiracMobiusEofa, DiracMobiusEofaPC)
 from .dirac import (Dirac, DiracClover, DiracCloverHasenbuschTwist,
                     DiracCloverHasenbuschTwistPC, DiracCloverPC,
                     DiracNdegTwistedMass, DiracNdegTwistedMassPC,
@@ -16,4 +17,5 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracNdegTwistedMass", "DiracNdegTwistedMassPC",
            "DiracCloverHasenbuschTwist", "DiracCloverHasenbuschTwistPC",
            "DiracTwistedCloverPC", "DiracDomainWall", "DiracDomainWallPC",
-           "DiracMobius", "DiracMobiusPC", "DiracZMobius", "DiracZMobiusPC"]
+           "DiracMobius", "DiracMobiusPC", "DiracZMobius", "DiracZMobiusPC",
+           "DiracMobiusEofa", "DiracMobiusEofaPC"]

@@ -238,3 +238,53 @@ class DiracZMobiusPC(DiracZMobius, DiracMobiusPC):
     """Symmetric even-even PC zMobius (M/prepare/reconstruct inherited from
     DiracMobiusPC; the s-operators from DiracZMobius)."""
     pass
+
+
+class DiracMobiusEofa(DiracMobius):
+    """Moebius EOFA operator (ref: lib/dirac_mobius_eofa.cpp
+    DiracMobiusEofa + dslash5_mobius_eofa.cu — re-derived): the 5th-dim
+    diagonal block A gains a rank-1 chiral term
+
+        A_eofa = A + eofa_shift * P_pm |u><u|,   u_s ~ kappa^s
+
+    where kappa = -beta/alpha is the bulk decay rate of the M5 recursion
+    (|u> is the exponential surface-mode profile at the pm wall; pm=+1
+    profiles decay from s=0, pm=-1 from s=Ls-1). mq1 is the base quark
+    mass; mq2/mq3 are stored for the one-flavor action bookkeeping (the
+    det-ratio pseudofermion wiring is layered above these operators).
+    The inverse applies one host-folded Sherman-Morrison correction on
+    top of the O(Ls) M5 inverse."""
+
+    def __init__(self, gauge: GaugeField, m5: float, Ls: int,
+                 b5: float = 1.5, c5: float = 0.5, *, mq1: float = 0.01,
+                 mq2: float = 0.08, mq3: float = 0.08, eofa_pm: int = 1,
+                 eofa_shift: float = -0.1):
+        super().__init__(gauge, mq1, m5, Ls, b5, c5)
+        self.mq1, self.mq2, self.mq3 = float(mq1), float(mq2), float(mq3)
+        self.eofa_pm = int(eofa_pm)
+        self.eofa_shift = float(eofa_shift)
+        kap = -self.beta / self.alpha
+        prof = [kap ** s for s in range(Ls)]
+        if self.eofa_pm < 0:
+            prof = prof[::-1]
+        import math
+        nrm = math.sqrt(sum(p * p for p in prof))
+        self.eofa_u = [p / nrm for p in prof]
+
+    def apply_A(self, out, inp, dagger=False):
+        from ..ops.dispatch import eofa5_op
+        return eofa5_op(out, inp, self.alpha, self.beta, self.mf, 0,
+                        self.eofa_shift, self.eofa_pm, self.eofa_u,
+                        self.eofa_u, dagger=dagger)
+
+    def apply_Ainv(self, out, inp, dagger=False, a=1.0, x=None):
+        from ..ops.dispatch import eofa5_op
+        return eofa5_op(out, inp, self.alpha, self.beta, self.mf, 1,
+                        self.eofa_shift, self.eofa_pm, self.eofa_u,
+                        self.eofa_u, dagger=dagger, a=a, x=x)
+
+
+class DiracMobiusEofaPC(DiracMobiusEofa, DiracMobiusPC):
+    """Symmetric even-even PC EOFA (structure from DiracMobiusPC, the
+    rank-1-extended s-ops from DiracMobiusEofa)."""
+    pass

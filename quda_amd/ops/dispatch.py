@@ -461,6 +461,56 @@ def zdwf5_op(out: SpinorField, inp: SpinorField, diag, hop, mf: float,
     return out
 
 
+def eofa5_op(out: SpinorField, inp: SpinorField, alpha: float, beta: float,
+             mf: float, kind: int, sh: float, pm: int, u, w,
+             dagger: bool = False, a: float = 1.0,
+             x: Optional[SpinorField] = None):
+    """EOFA rank-1-extended 5th-dimension ops (csrc k_m5_eofa /
+    k_m5inv_eofa; ref lib/dslash5_mobius_eofa.cu):
+    kind=0: out = [a*x +] (alpha + beta Ds + sh P_pm |u><w|) in
+    kind=1: out = [x +] a * (alpha + beta Ds + sh P_pm |u><w|)^{-1} in
+    The dagger swaps u <-> w and daggers the base (host-side); for kind=1
+    the kernel receives B^-1 u and sh/denom (Sherman-Morrison folded)."""
+    import numpy as np
+    Ls = inp.ls
+    xpay = x is not None
+    if on_gpu(out, inp):
+        ext = hip_ext()
+        xf = x if x is not None else out
+        uu = [float(v) for v in u]
+        ww = [float(v) for v in w]
+        if dagger:
+            uu, ww = ww, uu
+        if kind == 1:
+            # host Sherman-Morrison setup on the base bidiagonal+corner
+            # (the rank-1 lives on chirality pm; the base block there is
+            # the upper block for pm=+1)
+            A = ref._m5_matrix(Ls, alpha, beta, mf, pm > 0, dagger)
+            bu = np.linalg.solve(A, np.asarray(uu, dtype=float))
+            den = 1.0 + sh * float(np.dot(ww, bu))
+            uu = [float(v) for v in bu]
+            sh_k = sh / den
+        else:
+            sh_k = sh
+        ext.eofa5(out.data, norm_or_empty(out), inp.data, norm_or_empty(inp),
+                  xf.data, norm_or_empty(xf), inp.geo.volume_cb, Ls, xpay,
+                  bool(dagger), float(a), float(alpha), float(beta),
+                  float(mf), uu, ww, float(sh_k), int(pm), kind)
+        return out
+    psi = inp.to_complex()[0]
+    if kind == 0:
+        res = ref.m5_eofa(psi, Ls, alpha, beta, mf, sh, pm, u, w, dagger)
+        if xpay:
+            res = a * x.to_complex()[0] + res
+    else:
+        res = a * ref.m5inv_eofa(psi, Ls, alpha, beta, mf, sh, pm, u, w,
+                                 dagger)
+        if xpay:
+            res = x.to_complex()[0] + res
+    out.from_complex(res.unsqueeze(0))
+    return out
+
+
 def dwf_halo_exchange(inp: SpinorField, parity_in: int, dagger: bool):
     """Pack + exchange all s-slice faces of a 5-d input (blocking; returns
     the halo object to hand to dslash_wilson_slice, or None when no dim is

@@ -438,3 +438,43 @@ def zm5inv(psi5: torch.Tensor, Ls: int, diag, hop, mf: float,
                             device=psi5.device)
         out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", Ainv, v[:, :, sl, :])
     return out.reshape(Ls * V, 4, 3)
+
+
+def _m5_eofa_matrix(Ls, alpha, beta, mf, sh, pm, u, w, upper: bool,
+                    dagger: bool):
+    """Dense chirality block of the EOFA M5: base Moebius + rank-1
+    sh |u><w| on the pm chirality (pm=+1 -> upper block)."""
+    import numpy as np
+    A = _m5_matrix(Ls, alpha, beta, mf, upper, dagger)
+    on = (pm > 0) == upper
+    if on:
+        uu = np.asarray(u, dtype=float)
+        ww = np.asarray(w, dtype=float)
+        R1 = sh * np.outer(uu, ww)
+        A = A + (R1.conj().T if dagger else R1)
+    return A
+
+
+def m5_eofa(psi5, Ls, alpha, beta, mf, sh, pm, u, w, dagger=False):
+    V = psi5.shape[0] // Ls
+    v = psi5.reshape(Ls, V, 4, 3)
+    out = torch.empty_like(v)
+    for upper, sl in ((True, slice(0, 2)), (False, slice(2, 4))):
+        A = torch.tensor(_m5_eofa_matrix(Ls, alpha, beta, mf, sh, pm, u, w,
+                                         upper, dagger),
+                         dtype=psi5.dtype, device=psi5.device)
+        out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", A, v[:, :, sl, :])
+    return out.reshape(Ls * V, 4, 3)
+
+
+def m5inv_eofa(psi5, Ls, alpha, beta, mf, sh, pm, u, w, dagger=False):
+    import numpy as np
+    V = psi5.shape[0] // Ls
+    v = psi5.reshape(Ls, V, 4, 3)
+    out = torch.empty_like(v)
+    for upper, sl in ((True, slice(0, 2)), (False, slice(2, 4))):
+        A = _m5_eofa_matrix(Ls, alpha, beta, mf, sh, pm, u, w, upper, dagger)
+        Ainv = torch.tensor(np.linalg.inv(A), dtype=psi5.dtype,
+                            device=psi5.device)
+        out[:, :, sl, :] = torch.einsum("st,tvxc->svxc", Ainv, v[:, :, sl, :])
+    return out.reshape(Ls * V, 4, 3)

@@ -443,3 +443,110 @@ def test_madwf_training_reduces_chi2(setup):
     hist = train_transfer(big, small, T, n_samples=2, iters=10, lr=0.05,
                           inner_tol=1e-6)
     assert hist[-1] < hist[0], hist
+
+
+# ---------------------------------------------------------------------------
+# EOFA (rank-1 chiral extension of M5)
+# ---------------------------------------------------------------------------
+
+EOFA_KW = dict(mq1=MF, eofa_pm=1, eofa_shift=-0.21)
+
+
+def _eofa_vecs(op):
+    return dict(alpha=op.alpha, beta=op.beta, mf=op.mf, sh=op.eofa_shift,
+                pm=op.eofa_pm, u=op.eofa_u, w=op.eofa_u)
+
+
+@pytest.mark.parametrize("pm", [1, -1])
+@pytest.mark.parametrize("dagger", [False, True])
+def test_m5inv_eofa_oracle_inverts(setup, pm, dagger):
+    geo, g = setup
+    from quda_amd.models import DiracMobiusEofa
+    op = DiracMobiusEofa(g, M5, LS, mq1=MF, eofa_pm=pm, eofa_shift=-0.21)
+    kw = _eofa_vecs(op)
+    psi = spin5(geo, 331, n_parity=1).to_complex()[0]
+    y = ref.m5inv_eofa(psi, LS, kw["alpha"], kw["beta"], kw["mf"], kw["sh"],
+                       kw["pm"], kw["u"], kw["w"], dagger)
+    back = ref.m5_eofa(y, LS, kw["alpha"], kw["beta"], kw["mf"], kw["sh"],
+                       kw["pm"], kw["u"], kw["w"], dagger)
+    assert (back - psi).abs().max().item() < 1e-11
+
+
+@pytest.mark.parametrize("dagger", [False, True])
+def test_eofa_dispatch_ainv_inverts_a(setup, dagger):
+    """apply_Ainv(apply_A(psi)) == psi through the dispatch layer (the
+    GPU variant of this exercises the host Sherman-Morrison setup)."""
+    geo, g = setup
+    from quda_amd.models import DiracMobiusEofa
+    op = DiracMobiusEofa(g, M5, LS, **EOFA_KW)
+    psi = spin5(geo, 332, n_parity=1)
+    t = SpinorField(geo, "double", n_parity=1, ls=LS)
+    b = SpinorField(geo, "double", n_parity=1, ls=LS)
+    op.apply_A(t, psi, dagger=dagger)
+    op.apply_Ainv(b, t, dagger=dagger)
+    assert (b.to_complex() - psi.to_complex()).abs().max().item() < 1e-11
+
+
+def test_eofa_M_dagger_adjoint(setup):
+    geo, g = setup
+    from quda_amd.models import DiracMobiusEofa
+    op = DiracMobiusEofa(g, M5, LS, **EOFA_KW)
+    a = spin5(geo, 333)
+    b = spin5(geo, 334)
+    Ma = SpinorField(geo, "double", ls=LS)
+    Mdb = SpinorField(geo, "double", ls=LS)
+    op.M(Ma, a)
+    op.M(Mdb, b, dagger=True)
+    lhs = (b.to_complex().conj() * Ma.to_complex()).sum()
+    rhs = (Mdb.to_complex().conj() * a.to_complex()).sum()
+    assert abs(lhs - rhs) < 1e-10 * abs(lhs)
+
+
+def test_eofa_shift_zero_reduces_to_mobius(setup):
+    geo, g = setup
+    from quda_amd.models import DiracMobiusEofa
+    op0 = DiracMobius(g, MF, M5, LS)
+    ope = DiracMobiusEofa(g, M5, LS, mq1=MF, eofa_shift=0.0)
+    psi = spin5(geo, 335)
+    a = spin5(geo, 0)
+    b = spin5(geo, 0)
+    op0.M(a, psi)
+    ope.M(b, psi)
+    assert (a.to_complex() - b.to_complex()).abs().max().item() < 1e-12
+
+
+def test_eofa_pc_solve(setup):
+    geo, g = setup
+    from quda_amd.models import DiracMobiusEofaPC
+    pc = DiracMobiusEofaPC(g, M5, LS, **EOFA_KW)
+    b = spin5(geo, 336, n_parity=1)
+    x = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st = cgnr_solve(pc, x, b, tol=1e-10, maxiter=4000)
+    assert st.converged
+    r = SpinorField(geo, "double", n_parity=1, ls=LS)
+    pc.M(r, x)
+    tr = math.sqrt(blas.xmy_norm2(b, r) / blas.norm2(b))
+    assert tr < 1e-8, tr
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("prec", ["double", "single"])
+@pytest.mark.parametrize("kind", [0, 1])
+@pytest.mark.parametrize("dagger", [False, True])
+@pytest.mark.parametrize("pm", [1, -1])
+def test_eofa5_gpu_vs_oracle(setup, prec, kind, dagger, pm):
+    from quda_amd.ops.dispatch import eofa5_op
+    geo, g = setup
+    from quda_amd.models import DiracMobiusEofa
+    op = DiracMobiusEofa(g, M5, LS, mq1=MF, eofa_pm=pm, eofa_shift=-0.21)
+    kw = _eofa_vecs(op)
+    inp = SpinorField(geo, prec, "cuda", n_parity=1, ls=LS).gaussian_(seed=337)
+    out = SpinorField(geo, prec, "cuda", n_parity=1, ls=LS)
+    eofa5_op(out, inp, kw["alpha"], kw["beta"], kw["mf"], kind, kw["sh"],
+             kw["pm"], kw["u"], kw["w"], dagger=dagger)
+    psi = inp.to_complex()[0]
+    fn = ref.m5_eofa if kind == 0 else ref.m5inv_eofa
+    expect = fn(psi, LS, kw["alpha"], kw["beta"], kw["mf"], kw["sh"],
+                kw["pm"], kw["u"], kw["w"], dagger)
+    tol = 1e-11 if prec == "double" else 1e-4
+    assert (out.to_complex()[0] - expect).abs().max().item() < tol
