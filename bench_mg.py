@@ -46,6 +46,9 @@ def main():
                     dtype=torch.float32)
     if args.therm:
         from quda_amd.gauge import heatbath_sweep, overrelax_sweep, plaquette
+        if dev != "cpu":
+            from quda_amd.gauge.heatbath import set_device_rng
+            set_device_rng(True)
         u = GaugeField(geo, "double", dev).unit_().to_complex()
         t0 = time.perf_counter()
         for it in range(args.therm):
