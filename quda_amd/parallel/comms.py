@@ -19,7 +19,8 @@ import torch
 import torch.distributed as dist
 
 _STATE = {"grid": (1, 1, 1, 1), "coords": (0, 0, 0, 0), "initialized": False,
-          "forced_mask": 0}
+          "forced_mask": 0, "group": None, "group_ranks": None,
+          "group_rank": 0}
 
 
 def is_distributed() -> bool:
@@ -27,11 +28,22 @@ def is_distributed() -> bool:
 
 
 def comm_rank() -> int:
+    if _STATE.get("group_ranks") is not None:
+        return _STATE["group_rank"]
     return dist.get_rank() if is_distributed() else 0
 
 
 def comm_size() -> int:
+    if _STATE.get("group_ranks") is not None:
+        return len(_STATE["group_ranks"])
     return dist.get_world_size() if is_distributed() else 1
+
+
+def to_global_rank(local: int) -> int:
+    """Translate a topology-local rank to the global torch.distributed
+    rank (identity outside split-grid mode)."""
+    gr = _STATE.get("group_ranks")
+    return gr[local] if gr is not None else local
 
 
 def grid_dims() -> Tuple[int, int, int, int]:
@@ -79,9 +91,11 @@ def rank_of_coords(c: Tuple[int, int, int, int]) -> int:
 
 
 def neighbor_rank(dim: int, displacement: int) -> int:
+    """GLOBAL rank of the grid neighbor (p2p ops take global ranks even
+    inside a split-grid sub-communicator)."""
     c = list(_STATE["coords"])
     c[dim] += displacement
-    return rank_of_coords(tuple(c))
+    return to_global_rank(rank_of_coords(tuple(c)))
 
 
 def is_partitioned(dim: int) -> bool:
@@ -120,19 +134,19 @@ def allreduce_sum(x):
         return x
     dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
     t = torch.tensor([x], dtype=torch.float64, device=dev)
-    dist.all_reduce(t)
+    dist.all_reduce(t, group=_STATE.get("group"))
     return t.item()
 
 
 def allreduce_tensor(t: torch.Tensor) -> torch.Tensor:
     if is_distributed() and not _solo():
-        dist.all_reduce(t)
+        dist.all_reduce(t, group=_STATE.get("group"))
     return t
 
 
 def barrier():
     if is_distributed() and not _solo():
-        dist.barrier()
+        dist.barrier(group=_STATE.get("group"))
 
 
 # -- communicator stack (ref: lib/communicator_stack.cpp push_communicator:
@@ -161,3 +175,70 @@ def solo_mode():
 
 def _solo() -> bool:
     return bool(_STATE.get("solo", False))
+
+
+# -- general split-grid sub-communicators ----------------------------------
+
+_GROUP_CACHE = {}
+
+
+@contextmanager
+def split_grid_mode(splits):
+    """Partition the process grid into prod(splits) independent sub-grids
+    (ref: lib/communicator_stack.cpp push_communicator with a non-trivial
+    split key). Each sub-grid has process grid g_i/splits_i and spans the
+    WHOLE lattice (fields must be redistributed — see
+    split_grid.split_grid_solve). Yields (subgrid_index, n_subgrids).
+    Topology queries, collectives and halo exchange all act within the
+    sub-grid while inside the context."""
+    g = _STATE["grid"]
+    c = _STATE["coords"]
+    for i in range(4):
+        assert g[i] % splits[i] == 0, (g, splits)
+    sub = tuple(g[i] // splits[i] for i in range(4))
+    n_sub = splits[0] * splits[1] * splits[2] * splits[3]
+    block = tuple(c[i] // sub[i] for i in range(4))
+    newc = tuple(c[i] % sub[i] for i in range(4))
+    sub_idx = ((block[3] * splits[2] + block[2]) * splits[1]
+               + block[1]) * splits[0] + block[0]
+    key = (g, tuple(splits))
+    if n_sub == 1 or not is_distributed():
+        groups = {0: (None, list(range(comm_size())))}
+    elif key in _GROUP_CACHE:
+        groups = _GROUP_CACHE[key]
+    else:
+        # build ALL sub-groups in the same (lex) order on every rank
+        groups = {}
+        idx = 0
+        for bt in range(splits[3]):
+            for bz in range(splits[2]):
+                for by in range(splits[1]):
+                    for bx in range(splits[0]):
+                        b = (bx, by, bz, bt)
+                        ranks = []
+                        for t in range(sub[3]):
+                            for z in range(sub[2]):
+                                for y in range(sub[1]):
+                                    for x in range(sub[0]):
+                                        cc = (b[0] * sub[0] + x,
+                                              b[1] * sub[1] + y,
+                                              b[2] * sub[2] + z,
+                                              b[3] * sub[3] + t)
+                                        gx, gy, gz, gt = g
+                                        ranks.append(((cc[3] * gz + cc[2])
+                                                      * gy + cc[1]) * gx
+                                                     + cc[0])
+                        grp = dist.new_group(ranks)
+                        groups[idx] = (grp, ranks)
+                        idx += 1
+        _GROUP_CACHE[key] = groups
+    grp, ranks = groups[sub_idx if n_sub > 1 else 0]
+    saved = dict(_STATE)
+    me = dist.get_rank() if is_distributed() else 0
+    _STATE.update(grid=sub, coords=newc, forced_mask=0, group=grp,
+                  group_ranks=ranks, group_rank=ranks.index(me))
+    try:
+        yield sub_idx, n_sub
+    finally:
+        _STATE.clear()
+        _STATE.update(saved)

@@ -58,7 +58,10 @@ def exchange_tensors_start(sends: Dict[Key, torch.Tensor],
     host does not block, so an interior kernel launched in between overlaps
     with the transfer)."""
     ops = []  # (order_key, is_send, tensor, peer)
-    my_rank = comms.comm_rank()
+    # self-wraparound check must use the GLOBAL rank: neighbor_rank returns
+    # global ranks, and inside a split-grid sub-communicator comm_rank() is
+    # group-local
+    my_rank = dist.get_rank() if comms.is_distributed() else 0
     local_copies = []
     for (mu, d), s in sends.items():
         peer = comms.neighbor_rank(mu, -1 if d == 0 else +1)

@@ -40,3 +40,217 @@ def multi_src_solve(sources: List[SpinorField],
             if xs[i].norm is not None:
                 dist.broadcast(xs[i].norm, src=owner)
     return xs
+
+
+# ---------------------------------------------------------------------------
+# General split-grid: prod(splits) sub-grids of >1 rank each, with field
+# redistribution (ref: lib/split_grid.cpp splitFieldQuda + the
+# communicator_stack split key — re-designed over torch.distributed p2p:
+# each sub-grid spans the WHOLE lattice on fewer ranks, so every new rank
+# aggregates the local slabs of prod(splits) old ranks; messages are issued
+# in deterministic sorted peer order, self-pairs copied directly).
+# ---------------------------------------------------------------------------
+
+
+def _old_rank_of(coords, grid):
+    gx, gy, gz, gt = grid
+    return ((coords[3] * gz + coords[2]) * gy + coords[1]) * gx + coords[0]
+
+
+def _lex_nd(t, local_dims, extra_shape):
+    """[V, *extra] local-lex tensor -> [T,Z,Y,X, *extra] block view."""
+    X, Y, Z, T = local_dims
+    return t.reshape(T, Z, Y, X, *extra_shape)
+
+
+def scatter_to_subgrids(t_lex, local_dims, splits):
+    """Redistribute a local-lex field ([V_loc, *extra]) from the current
+    process grid onto EVERY sub-grid of `splits`: returns this rank's
+    enlarged local-lex tensor ([V_loc * prod(splits), *extra]) for its own
+    sub-grid copy. Call OUTSIDE split_grid_mode."""
+    import torch.distributed as dist
+    from . import comms
+    g = comms.grid_dims()
+    t = comms.grid_coords()
+    f = tuple(splits)
+    sub = tuple(g[i] // f[i] for i in range(4))
+    n_sub = f[0] * f[1] * f[2] * f[3]
+    extra = t_lex.shape[1:]
+    me = comms.comm_rank()
+    # my slab's position within any sub-grid copy: new coord t_i // f_i
+    cdest = tuple(t[i] // f[i] for i in range(4))
+    # my OWN sub-grid coordinate (split_grid_mode convention: t_i % sub_i)
+    myc = tuple(t[i] % sub[i] for i in range(4))
+
+    # destinations: for each sub-grid block b, the rank at old-coords
+    # (b_i*sub_i + cdest_i)
+    sends = []
+    for bt in range(f[3]):
+        for bz in range(f[2]):
+            for by in range(f[1]):
+                for bx in range(f[0]):
+                    b = (bx, by, bz, bt)
+                    dest = tuple(b[i] * sub[i] + cdest[i] for i in range(4))
+                    sends.append(_old_rank_of(dest, g))
+    # sources: old ranks covering MY new region (coord myc)
+    recvs = {}
+    for o3 in range(myc[3] * f[3], (myc[3] + 1) * f[3]):
+        for o2 in range(myc[2] * f[2], (myc[2] + 1) * f[2]):
+            for o1 in range(myc[1] * f[1], (myc[1] + 1) * f[1]):
+                for o0 in range(myc[0] * f[0], (myc[0] + 1) * f[0]):
+                    src = _old_rank_of((o0, o1, o2, o3), g)
+                    off = (o0 - myc[0] * f[0], o1 - myc[1] * f[1],
+                           o2 - myc[2] * f[2], o3 - myc[3] * f[3])
+                    recvs[src] = off
+
+    send_buf = t_lex.contiguous()
+    bufs = {src: (torch.empty_like(send_buf) if src != me else send_buf)
+            for src in recvs}
+    reqs = []
+    for peer in sorted(set(sends)):
+        cnt = sends.count(peer)
+        for k in range(cnt):
+            if peer == me:
+                continue
+            reqs.append(dist.isend(send_buf, peer, tag=k))
+    for i, (peer, _) in enumerate(sorted(recvs.items())):
+        if peer == me:
+            continue
+        reqs.append(dist.irecv(bufs[peer], peer, tag=0))
+    for r in reqs:
+        r.wait()
+
+    X, Y, Z, T = local_dims
+    big = torch.empty((T * f[3], Z * f[2], Y * f[1], X * f[0], *extra),
+                      dtype=t_lex.dtype, device=t_lex.device)
+    for src, off in recvs.items():
+        blk = _lex_nd(bufs[src], local_dims, extra)
+        big[off[3] * T:(off[3] + 1) * T, off[2] * Z:(off[2] + 1) * Z,
+            off[1] * Y:(off[1] + 1) * Y, off[0] * X:(off[0] + 1) * X] = blk
+    return big.reshape(-1, *extra)
+
+
+def gather_from_subgrid(big_lex, local_dims, splits, owner_subgrid):
+    """Inverse of scatter for ONE field solved on sub-grid
+    `owner_subgrid`: every rank (all sub-grids) receives its original
+    local-lex slab ([V_loc, *extra]). big_lex is this rank's enlarged
+    tensor (ignored unless this rank belongs to the owner sub-grid).
+    Call OUTSIDE split_grid_mode."""
+    import torch.distributed as dist
+    from . import comms
+    g = comms.grid_dims()
+    t = comms.grid_coords()
+    f = tuple(splits)
+    sub = tuple(g[i] // f[i] for i in range(4))
+    me = comms.comm_rank()
+    block = tuple(t[i] // sub[i] for i in range(4))
+    my_sub = ((block[3] * f[2] + block[2]) * f[1] + block[1]) * f[0] + block[0]
+    newc = tuple(t[i] % sub[i] for i in range(4))
+    X, Y, Z, T = local_dims
+    extra = big_lex.shape[1:] if big_lex is not None else None
+
+    reqs = []
+    out = None
+    # my slab comes from the owner sub-grid's rank at coord c'_i = t_i//f_i
+    cov = tuple(t[i] // f[i] for i in range(4))
+    ob = (owner_subgrid % f[0], (owner_subgrid // f[0]) % f[1],
+          (owner_subgrid // (f[0] * f[1])) % f[2],
+          owner_subgrid // (f[0] * f[1] * f[2]))
+    sender = _old_rank_of(tuple(ob[i] * sub[i] + cov[i] for i in range(4)), g)
+
+    send_slabs = []
+    if my_sub == owner_subgrid:
+        # I hold a solved block: ship sub-slabs to every old rank I cover
+        bigv = _lex_nd(big_lex, (X * f[0], Y * f[1], Z * f[2], T * f[3]),
+                       big_lex.shape[1:])
+        for o3 in range(newc[3] * f[3], (newc[3] + 1) * f[3]):
+            for o2 in range(newc[2] * f[2], (newc[2] + 1) * f[2]):
+                for o1 in range(newc[1] * f[1], (newc[1] + 1) * f[1]):
+                    for o0 in range(newc[0] * f[0], (newc[0] + 1) * f[0]):
+                        # coverage of new coord cov: old-grid coords
+                        # o_i in [cov_i*f_i, (cov_i+1)*f_i)
+                        dst = _old_rank_of((o0, o1, o2, o3), g)
+                        l3, l2 = o3 - newc[3] * f[3], o2 - newc[2] * f[2]
+                        l1, l0 = o1 - newc[1] * f[1], o0 - newc[0] * f[0]
+                        slab = bigv[l3 * T:(l3 + 1) * T, l2 * Z:(l2 + 1) * Z,
+                                    l1 * Y:(l1 + 1) * Y, l0 * X:(l0 + 1) * X]
+                        slab = slab.reshape(-1, *big_lex.shape[1:]).contiguous()
+                        send_slabs.append((dst, slab))
+    # post in sorted deterministic order
+    my_recv = None
+    for dst, slab in sorted(send_slabs, key=lambda p: p[0]):
+        if dst == me:
+            my_recv = slab
+        else:
+            reqs.append(dist.isend(slab, dst))
+    if my_recv is None:
+        shape0 = (X * Y * Z * T,)
+        # element shape: infer from big_lex if present else from a probe --
+        # caller always passes element shape via big_lex on owner ranks;
+        # non-owner ranks pass a template tensor instead
+        raise_if = big_lex is None
+        assert not raise_if, "pass a template tensor for non-owner ranks"
+        out = torch.empty((X * Y * Z * T, *big_lex.shape[1:]),
+                          dtype=big_lex.dtype, device=big_lex.device)
+        reqs.append(dist.irecv(out, sender))
+    else:
+        out = my_recv
+    for r in reqs:
+        r.wait()
+    return out
+
+
+def split_grid_solve(u_cb, sources_cb, geo, splits, solve_one):
+    """Solve len(sources) systems distributed round-robin over
+    prod(splits) sub-grids, each sub-grid a (grid/splits) process grid
+    spanning the whole lattice (general splitGridQuda). u_cb:
+    [4,2,Vcb,3,3] local complex gauge; sources_cb: local complex cb
+    fields [2,Vcb,...]; solve_one(u_big_cb, b_big_cb, geo_big) -> x_big_cb
+    runs inside the sub-communicator. Returns all solutions in the
+    original decomposition."""
+    from . import comms
+    from ..fields.geometry import (LatticeGeometry, checkerboard_join,
+                                   checkerboard_split)
+    f = tuple(splits)
+    n_sub = f[0] * f[1] * f[2] * f[3]
+    if n_sub == 1 or not comms.is_distributed():
+        with comms.split_grid_mode(f) if comms.is_distributed() else _null():
+            return [solve_one(u_cb, b, geo) for b in sources_cb]
+    ld = geo.dims
+    big_dims = tuple(ld[i] * f[i] for i in range(4))
+    # lexify + scatter
+    u_lex = checkerboard_join(u_cb.permute(1, 2, 0, 3, 4), geo)  # [V,4,3,3]
+    u_big_lex = scatter_to_subgrids(u_lex, ld, f)
+    srcs_big = []
+    for j, b in enumerate(sources_cb):
+        b_lex = checkerboard_join(b, geo)
+        srcs_big.append(scatter_to_subgrids(b_lex, ld, f))
+    results_big = [None] * len(sources_cb)
+    with comms.split_grid_mode(f) as (sub_idx, _):
+        geo_big = LatticeGeometry(big_dims)
+        u_big_cb = checkerboard_split(u_big_lex, geo_big)
+        u_big_cb = u_big_cb.permute(2, 0, 1, 3, 4).contiguous()
+        for j in range(len(sources_cb)):
+            if j % n_sub == sub_idx:
+                b_big = checkerboard_split(srcs_big[j], geo_big)
+                results_big[j] = solve_one(u_big_cb, b_big, geo_big)
+    # gather all solutions back (deterministic per-source order)
+    out = []
+    geo_big = LatticeGeometry(big_dims)
+    for j in range(len(sources_cb)):
+        owner = j % n_sub
+        if results_big[j] is not None:
+            x_lex = checkerboard_join(results_big[j], geo_big)
+        else:
+            x_lex = srcs_big[j]  # template (same shape/dtype)
+        slab = gather_from_subgrid(x_lex, ld, f, owner)
+        out.append(checkerboard_split(slab, geo))
+    return out
+
+
+from contextlib import contextmanager
+
+
+@contextmanager
+def _null():
+    yield 0, 1

@@ -408,3 +408,72 @@ def test_hmc_2f_multirank_energy_conservation():
         init_file = f.name
     os.unlink(init_file)
     mp.spawn(_worker_hmc, args=(2, init_file), nprocs=2, join=True)
+
+
+# ---------------------------------------------------------------------------
+# general split-grid (2 sub-grids x 2 ranks each)
+# ---------------------------------------------------------------------------
+
+def _worker_splitgrid(rank, world, init_file):
+    import torch.distributed as dist
+    from quda_amd.parallel.split_grid import split_grid_solve
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=(1, 1, 1, world))
+        gg, u_lex, src_lex = _global_fields(seed=31)
+        from quda_amd.fields.geometry import checkerboard_join, checkerboard_split
+        lg, u_loc_lex = _local_slice(gg, (1, 1, 1, world), comms.grid_coords(),
+                                     u_lex.movedim(0, 1))
+        u_loc = checkerboard_split(u_loc_lex, lg).permute(2, 0, 1, 3, 4).contiguous()
+        # four sources (deterministic, global)
+        gens = [torch.Generator().manual_seed(600 + j) for j in range(4)]
+        srcs_lex = [torch.view_as_complex(
+            torch.randn((gg.volume, 4, 3, 2), generator=g0,
+                        dtype=torch.float64)) for g0 in gens]
+        srcs_loc = []
+        for sl in srcs_lex:
+            _, s_loc_lex = _local_slice(gg, (1, 1, 1, world),
+                                        comms.grid_coords(), sl)
+            srcs_loc.append(checkerboard_split(s_loc_lex, lg))
+
+        kappa = 0.11
+
+        def solve_one(u_big_cb, b_big_cb, geo_big):
+            g = GaugeField(geo_big, "double").from_complex(u_big_cb)
+            d = DiracWilson(g, kappa)
+            b = SpinorField(geo_big, "double")
+            b.from_complex(b_big_cb)
+            x = SpinorField(geo_big, "double")
+            st = cg_solve(d, x, b, tol=1e-10, maxiter=400)
+            assert st.converged
+            return x.to_complex()
+
+        xs = split_grid_solve(u_loc, srcs_loc, lg, (1, 1, 1, 2), solve_one)
+
+        # truth: global single-process solves (grab MY coords before solo
+        # mode re-roots the topology)
+        mycoords = comms.grid_coords()
+        u_g = checkerboard_split(u_lex.movedim(0, 1), gg).permute(
+            2, 0, 1, 3, 4).contiguous()
+        with comms.solo_mode():
+            dg = DiracWilson(GaugeField(gg, "double").from_complex(u_g), kappa)
+            for j in range(4):
+                sg = SpinorField(gg, "double")
+                sg.from_complex(checkerboard_split(srcs_lex[j], gg))
+                xg = SpinorField(gg, "double")
+                st = cg_solve(dg, xg, sg, tol=1e-10, maxiter=400)
+                assert st.converged
+                xg_lex = checkerboard_join(xg.to_complex(), gg)
+                _, x_loc_lex = _local_slice(gg, (1, 1, 1, world),
+                                            mycoords, xg_lex)
+                x_loc = checkerboard_split(x_loc_lex, lg)
+                err = (xs[j] - x_loc).abs().max().item()
+                assert err < 1e-7, f"rank{rank} src{j} err={err}"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_split_grid_general_gloo(tmp_path):
+    init_file = str(tmp_path / "init_sg")
+    mp.spawn(_worker_splitgrid, args=(4, init_file), nprocs=4, join=True)
