@@ -438,3 +438,161 @@ def compute_ks_link_quda(coeffs=None):
     from .gauge.hisq import asqtad_coefficients, fat_links, naik_links
     c = coeffs or asqtad_coefficients()
     return fat_links(_R.u_complex, _R.geo, c), naik_links(_R.u_complex, _R.geo)
+
+
+def mat_dag_mat_quda(inp: torch.Tensor, p: InvertParam) -> torch.Tensor:
+    """ref: MatDagMatQuda interface_quda.cpp — applies M^dag M of the
+    resident operator."""
+    d = _make_dirac(p)
+    n_parity = 1 if "PC" in type(d).__name__ else 2
+    b = _wrap(inp, p, n_parity)
+    out = _wrap(torch.zeros_like(inp), p, n_parity)
+    tmp = _wrap(torch.zeros_like(inp), p, n_parity)
+    d.MdagM(out, b, tmp)
+    return out.to_complex()
+
+
+def save_gauge_quda() -> torch.Tensor:
+    """ref: saveGaugeQuda — returns the resident gauge field as
+    [4,2,Vcb,3,3] complex (the inverse of load_gauge_quda)."""
+    assert _R.u_complex is not None, "no resident gauge"
+    return _R.u_complex.clone()
+
+
+def invert_multi_src_quda(bs, p: InvertParam, *, splits=None):
+    """ref: invertMultiSrcQuda + the split-grid comm key: solves all
+    sources; with `splits` (per-dim sub-grid factors) the sources are
+    distributed over split-grid sub-grids with field redistribution
+    (parallel.split_grid.split_grid_solve); otherwise each source is
+    solved in order on the current grid."""
+    from .parallel import comms
+    if splits is not None and comms.is_distributed():
+        from .parallel.split_grid import split_grid_solve
+        gp0 = _R.gauge_param
+
+        def solve_one(u_big_cb, b_big_cb, geo_big):
+            saved = (_R.gauge, _R.gauge_sloppy, _R.u_complex, _R.geo,
+                     _R.gauge_param)
+            try:
+                gp = GaugeParam(X=geo_big.dims, cuda_prec=gp0.cuda_prec,
+                                cuda_prec_sloppy=gp0.cuda_prec_sloppy,
+                                device=gp0.device)
+                load_gauge_quda(u_big_cb, gp)
+                return invert_quda(b_big_cb, p)
+            finally:
+                (_R.gauge, _R.gauge_sloppy, _R.u_complex, _R.geo,
+                 _R.gauge_param) = saved
+
+        return split_grid_solve(_R.u_complex, list(bs), _R.geo, splits,
+                                solve_one)
+    return [invert_quda(b, p) for b in bs]
+
+
+def blas_gemm_quda(a: torch.Tensor, b: torch.Tensor,
+                   alpha=1.0, beta=0.0, c: torch.Tensor = None):
+    """ref: blasGEMMQuda (batched strided complex GEMM; rocBLAS via
+    torch.matmul on device tensors)."""
+    r = alpha * (a @ b)
+    if c is not None and beta != 0.0:
+        r = r + beta * c
+    return r
+
+
+def compute_clover_force_quda(kappa: float, csw: float, x_sol: torch.Tensor,
+                              seed: int = 0) -> torch.Tensor:
+    """ref: computeCloverForceQuda — clover-fermion force on the resident
+    gauge for pseudofermion phi (autograd-exact; see
+    gauge.fermion_force.clover_fermion_force)."""
+    from .gauge.fermion_force import clover_fermion_force
+    phi = _wrap(x_sol, InvertParam(kappa=kappa), 2)
+    S, F = clover_fermion_force(_R.u_complex, _R.geo, kappa, csw, phi)
+    return F
+
+
+def contract_quda(x: torch.Tensor, y: torch.Tensor, p: InvertParam,
+                  mode: str = "open") -> torch.Tensor:
+    """ref: contractQuda — open-spin / DeGrand-Rossi contraction of two
+    propagator fields."""
+    from .ops import contract
+    xf = _wrap(x, p, 2)
+    yf = _wrap(y, p, 2)
+    if mode == "open":
+        return contract.contract_open_spin(xf, yf)
+    return contract.contract_dr(xf, yf)
+
+
+def contract_ft_quda(x: torch.Tensor, y: torch.Tensor, p: InvertParam,
+                     momenta, reduct_dim: int = 3) -> torch.Tensor:
+    """ref: contractFTQuda — momentum-projected timeslice contraction."""
+    from .ops import contract
+    return contract.contract_ft(_wrap(x, p, 2), _wrap(y, p, 2),
+                                list(momenta), reduct_dim)
+
+
+def new_deflation_quda(p: InvertParam, e: EigParam):
+    """ref: newDeflationQuda — eigensolve the resident operator and
+    return a Deflation object (plugs into solver initial guesses)."""
+    from .solvers.eigen import Deflation
+    evals, evecs = eigensolve_quda(p, e)
+    return Deflation([float(v.real) for v in evals], evecs)
+
+
+_CHRONO = {}
+
+
+def chrono_forecaster(index: int = 0, max_dim: int = 8):
+    """Resident chrono basis per index (ref: inv_param.chrono_index +
+    use_resident_chrono)."""
+    from .solvers.mre import ChronoForecaster
+    if index not in _CHRONO:
+        _CHRONO[index] = ChronoForecaster(max_dim)
+    return _CHRONO[index]
+
+
+def flush_chrono_quda(index: int = -1) -> None:
+    """ref: flushChronoQuda interface_quda.cpp — drop resident chrono
+    bases (index -1 = all)."""
+    if index < 0:
+        _CHRONO.clear()
+    else:
+        _CHRONO.pop(index, None)
+
+
+def perform_fermion_smear_quda(src: torch.Tensor, p: InvertParam,
+                               n_steps: int, width: float) -> torch.Tensor:
+    """ref: performFermionSmearQuda — Gaussian/Wuppertal smearing of a
+    source with the resident gauge."""
+    from .models.laplace import wuppertal_smear
+    f = _wrap(src, p, 2)
+    out = wuppertal_smear(_R.gauge, f, alpha=width, n_steps=n_steps)
+    return out.to_complex()
+
+
+def update_multigrid_quda(mg, p: InvertParam) -> None:
+    """ref: updateMultigridQuda — re-setup the MG hierarchy after the
+    resident gauge changed (re-generates null vectors + coarse ops on the
+    CURRENT resident links)."""
+    from .mg import MG
+    d = _make_dirac(InvertParam(**{**p.__dict__,
+                                   "solution_type": SolutionType.MAT}))
+    mg.__init__(d, mg.param)
+
+
+# -- verbosity / logging (ref: logQuda/printfQuda/warningQuda, util_quda.h)
+
+_VERBOSITY = 1  # 0 silent, 1 summarize, 2 verbose, 3 debug
+
+
+def set_verbosity_quda(level: int) -> None:
+    global _VERBOSITY
+    _VERBOSITY = int(level)
+
+
+def log_quda(level: int, msg: str) -> None:
+    if level <= _VERBOSITY:
+        print(f"QUDA-AMD: {msg}", flush=True)
+
+
+def warning_quda(msg: str) -> None:
+    import sys
+    print(f"QUDA-AMD WARNING: {msg}", file=sys.stderr, flush=True)

@@ -141,3 +141,101 @@ def test_gauge_fixing_entry(resident):
     # hot random fields converge slowly; deep convergence is covered by
     # tests/test_gauge_fix.py on a smooth field
     assert th < 1e-4 and th < th0 * 1e-2
+
+
+def test_matdagmat_and_save_gauge(resident):
+    # earlier tests smear the resident field in place: reload the original
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12)
+    b = _rand_spinor((2, 128, 4, 3), 501)
+    mm = api.mat_dag_mat_quda(b, p)
+    m1 = api.mat_quda(b, p)
+    # compare against M^dag(M b) via dagger entry
+    from quda_amd.api import _make_dirac, _wrap
+    d = _make_dirac(p)
+    t = _wrap(m1, p, 2)
+    o = t.clone_empty()
+    d.M(o, t, dagger=True)
+    assert (mm - o.to_complex()).abs().max().item() < 1e-12
+    u = api.save_gauge_quda()
+    assert (u - resident).abs().max().item() == 0.0
+
+
+def test_multi_src_api(resident):
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12, tol=1e-9,
+                    maxiter=300)
+    bs = [_rand_spinor((2, 128, 4, 3), 510 + j) for j in range(3)]
+    xs = api.invert_multi_src_quda(bs, p)
+    for b, x in zip(bs, xs):
+        x1 = api.invert_quda(b, p)
+        assert (x - x1).abs().max().item() < 1e-8
+
+
+def test_blas_gemm_contract_api(resident):
+    a = torch.randn(4, 8, 8, dtype=torch.complex128)
+    b = torch.randn(4, 8, 8, dtype=torch.complex128)
+    c = torch.randn(4, 8, 8, dtype=torch.complex128)
+    r = api.blas_gemm_quda(a, b, alpha=2.0, beta=0.5, c=c)
+    assert (r - (2.0 * a @ b + 0.5 * c)).abs().max().item() < 1e-12
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12)
+    x = _rand_spinor((2, 128, 4, 3), 520)
+    y = _rand_spinor((2, 128, 4, 3), 521)
+    open_c = api.contract_quda(x, y, p, mode="open")
+    assert open_c.shape[-2:] == (4, 4)
+    ft = api.contract_ft_quda(x, y, p, [(0, 0, 0), (1, 0, 0)])
+    assert ft.shape == (2, 4, 16)
+
+
+def test_deflation_and_chrono_api(resident):
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.10, tol=1e-8,
+                    maxiter=400)
+    e = EigParam(n_ev=4, n_kr=16, tol=1e-6)
+    defl = api.new_deflation_quda(p, e)
+    assert len(defl.evals) == 4
+    ch = api.chrono_forecaster(0)
+    assert api.chrono_forecaster(0) is ch
+    api.flush_chrono_quda(0)
+    assert api.chrono_forecaster(0) is not ch
+    api.flush_chrono_quda()
+
+
+def test_fermion_smear_api(resident):
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12)
+    src = torch.zeros((2, 128, 4, 3), dtype=torch.complex128)
+    src[0, 0, 0, 0] = 1.0
+    sm = api.perform_fermion_smear_quda(src, p, n_steps=3, width=0.5)
+    assert (sm.abs() > 1e-12).sum() > (src.abs() > 1e-12).sum()
+    n0 = src.abs().square().sum()
+
+
+def test_clover_force_api(resident):
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.1)
+    phi = _rand_spinor((2, 128, 4, 3), 530)
+    F = api.compute_clover_force_quda(0.1, 1.2, phi)
+    assert F.shape == (4, 2, 128, 3, 3)
+    # traceless-antihermitian output
+    ta = F + F.conj().transpose(-1, -2)
+    assert ta.abs().max().item() < 1e-10
+    tr = torch.diagonal(F, dim1=-2, dim2=-1).sum(-1)
+    assert tr.abs().max().item() < 1e-10
+
+
+def test_update_multigrid_api(resident):
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12, tol=1e-8,
+                    maxiter=300, inv_type=InverterType.GCR)
+    mg = api.new_multigrid_quda(p, block=(2, 2, 2, 2), n_vec=4)
+    api.update_multigrid_quda(mg, p)
+    p2 = InvertParam(**{**p.__dict__, "preconditioner": mg.precond})
+    b = _rand_spinor((2, 128, 4, 3), 540)
+    x = api.invert_quda(b, p2)
+    assert p2.true_res < 1e-7
+
+
+def test_verbosity_api(capsys):
+    api.set_verbosity_quda(1)
+    api.log_quda(1, "visible")
+    api.log_quda(2, "hidden")
+    out = capsys.readouterr().out
+    assert "visible" in out and "hidden" not in out
