@@ -1,5 +1,7 @@
 from .staggered import (DiracImprovedStaggered, DiracImprovedStaggeredPC,
                         DiracStaggered, DiracStaggeredPC)
+from .staggered_kd import (DiracImprovedStaggeredKD, DiracStaggeredKD,
+                           KDBlockInverse)
 from .dwf import (DiracDomainWall, DiracDomainWallPC, DiracMobius,
                   DiracMobiusPC, DiracZMobius, DiracZMobiusPC,
                   DiracMobiusEofa, DiracMobiusEofaPC)
@@ -18,4 +20,5 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracCloverHasenbuschTwist", "DiracCloverHasenbuschTwistPC",
            "DiracTwistedCloverPC", "DiracDomainWall", "DiracDomainWallPC",
            "DiracMobius", "DiracMobiusPC", "DiracZMobius", "DiracZMobiusPC",
-           "DiracMobiusEofa", "DiracMobiusEofaPC"]
+           "DiracMobiusEofa", "DiracMobiusEofaPC", "DiracStaggeredKD",
+           "DiracImprovedStaggeredKD", "KDBlockInverse"]
