@@ -7,7 +7,9 @@ setup/analysis utilities, not the solver hot path."""
 
 from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
                   plaquette, polyakov_loop, project_ta, staple_sum,
-                  stout_smear, topological_charge, wilson_flow, wilson_loop)
+                  stout_smear, topological_charge, wilson_flow, wilson_loop,
+                  energy_density, wilson_flow_measure, flow_scale_t0,
+                  flow_scale_w0)
 from .hmc import hmc_trajectory, leapfrog, mom_action, random_momentum
 from .fix import gauge_fix_ovr, gauge_fix_quality
 from .heatbath import heatbath_sweep, overrelax_sweep
@@ -17,6 +19,8 @@ from .fermion_force import (fermion_action_and_force, hmc_trajectory_2f,
 __all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
            "project_ta", "exp_su3", "ape_smear", "stout_smear",
            "wilson_flow", "polyakov_loop", "topological_charge",
+           "energy_density", "wilson_flow_measure", "flow_scale_t0",
+           "flow_scale_w0",
            "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",
            "heatbath_sweep", "overrelax_sweep", "wilson_fermion_force",
            "fermion_action_and_force", "hmc_trajectory_2f",

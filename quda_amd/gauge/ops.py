@@ -249,3 +249,65 @@ def wilson_loop(u: torch.Tensor, geo: LatticeGeometry, R: int, T: int,
         P = P @ shifted(U[nu]).conj().mT
     tr = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).mean() / 3.0
     return complex(tr.real.item(), tr.imag.item())
+
+
+def energy_density(u: torch.Tensor, geo: LatticeGeometry) -> tuple:
+    """(E_plaq, E_clover): gradient-flow energy densities
+    (ref: lib/gauge_wilson_flow.cu energy measurement + gauge_qcharge.cuh
+    E from the clover F_munu). E_plaq = 2 sum_{mu<nu} Re tr(1 - P);
+    E_clover = sum_{mu<nu} -tr(F F) with F the TA clover field strength.
+    Both are global means per site (allreduced)."""
+    from ..parallel import comms
+    from ..ops.reference import field_strength
+    tot, _, _ = plaquette(u, geo)
+    # plaquette() returns mean Re tr P / 3 over 6 planes
+    e_plaq = 2.0 * 6 * 3 * (1.0 - tot)
+    e_clov = 0.0
+    Fd = field_strength(u, geo)
+    for F in Fd.values():
+        # tr[F F^dag] = sum |F|^2 is convention-proof (hermitian or
+        # antihermitian F_munu normalization)
+        e_clov += torch.einsum("pvij,pvij->", F, F.conj()).real.item()
+    n = comms.allreduce_sum(float(geo.volume))
+    e_clov = comms.allreduce_sum(e_clov) / n
+    return e_plaq, e_clov
+
+
+def wilson_flow_measure(u: torch.Tensor, geo: LatticeGeometry, eps: float,
+                        n_steps: int):
+    """Flow + measure: returns (u_flowed, history) with history entries
+    (t, E_plaq, E_clover, t^2 E_clover) per step (ref: the wflow path of
+    performGaugeSmearQuda + gaugeObservables)."""
+    hist = []
+    out = u
+    t = 0.0
+    for k in range(n_steps):
+        out = wilson_flow(out, geo, eps, 1)
+        t += eps
+        ep, ec = energy_density(out, geo)
+        hist.append((t, ep, ec, t * t * ec))
+    return out, hist
+
+
+def flow_scale_t0(hist, target: float = 0.3):
+    """t0 from t^2 E(t) = target by linear interpolation (Luscher scale
+    setting); None if the flow history never crosses the target."""
+    for (t1, _, _, s1), (t2, _, _, s2) in zip(hist, hist[1:]):
+        if s1 < target <= s2:
+            return t1 + (target - s1) * (t2 - t1) / (s2 - s1)
+    return None
+
+
+def flow_scale_w0(hist, target: float = 0.3):
+    """w0^2 from t d/dt[t^2 E] = target (BMW w0 scale), finite-difference
+    derivative on the measured history."""
+    for i in range(1, len(hist) - 1):
+        t = hist[i][0]
+        d = t * (hist[i + 1][3] - hist[i - 1][3]) / (hist[i + 1][0]
+                                                     - hist[i - 1][0])
+        d_next = hist[i + 1][0] * (hist[min(i + 2, len(hist) - 1)][3]
+                                   - hist[i][3]) / (
+            hist[min(i + 2, len(hist) - 1)][0] - hist[i][0])
+        if d < target <= d_next:
+            return t + (target - d) * (hist[i + 1][0] - t) / (d_next - d)
+    return None
