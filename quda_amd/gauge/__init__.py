@@ -10,7 +10,8 @@ from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
                   stout_smear, topological_charge, wilson_flow, wilson_loop,
                   energy_density, wilson_flow_measure, flow_scale_t0,
                   flow_scale_w0)
-from .hmc import hmc_trajectory, leapfrog, mom_action, random_momentum
+from .hmc import (hmc_trajectory, leapfrog, mom_action, nested_leapfrog,
+                  omelyan, random_momentum)
 from .fix import gauge_fix_ovr, gauge_fix_quality
 from .heatbath import heatbath_sweep, overrelax_sweep
 from .fermion_force import (fermion_action_and_force, hmc_trajectory_2f,
@@ -22,6 +23,7 @@ __all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
            "energy_density", "wilson_flow_measure", "flow_scale_t0",
            "flow_scale_w0",
            "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",
+           "omelyan", "nested_leapfrog",
            "heatbath_sweep", "overrelax_sweep", "wilson_fermion_force",
            "fermion_action_and_force", "hmc_trajectory_2f",
            "pseudofermion_refresh", "gauge_fix_ovr", "gauge_fix_quality", "wilson_loop"]

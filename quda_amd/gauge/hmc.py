@@ -86,3 +86,58 @@ def hmc_trajectory(u: torch.Tensor, geo: LatticeGeometry, beta: float,
         g.manual_seed(seed + 1)
     accept = torch.rand(1, generator=g).item() < min(1.0, float(torch.exp(torch.tensor(-dH))))
     return (u1 if accept else u), accept, dH
+
+
+def _evolve_u(u: torch.Tensor, P: torch.Tensor, geo: LatticeGeometry,
+              dt: float) -> torch.Tensor:
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    return _from_lex(exp_su3(Pl, dt) @ U, geo)
+
+
+_OMELYAN_LAMBDA = 0.1931833275037836  # minimal-norm 2nd-order coefficient
+
+
+def omelyan(u: torch.Tensor, P: torch.Tensor, geo: LatticeGeometry,
+            force_fn, n_steps: int, dt: float,
+            lam: float = _OMELYAN_LAMBDA):
+    """Second-order minimal-norm (Omelyan/2MN) integrator: per step
+    P(lam dt) U(dt/2) P((1-2lam)dt) U(dt/2) P(lam dt). Same O(dt^2)
+    per-trajectory energy error order as leapfrog with a ~10x smaller
+    coefficient at 1.5x the force evaluations (ref: the integrator the
+    reference's HMC consumers pair with its force kernels)."""
+    u = u.clone()
+    for k in range(n_steps):
+        P = P + (lam * dt) * force_fn(u)
+        u = _evolve_u(u, P, geo, 0.5 * dt)
+        P = P + ((1.0 - 2.0 * lam) * dt) * force_fn(u)
+        u = _evolve_u(u, P, geo, 0.5 * dt)
+        P = P + (lam * dt) * force_fn(u)
+    return u, P
+
+
+def nested_leapfrog(u: torch.Tensor, P: torch.Tensor, geo: LatticeGeometry,
+                    levels, n_steps: int, dt: float):
+    """Sexton-Weingarten multi-timescale leapfrog. `levels` is a list of
+    (force_fn, n_sub) outermost (expensive, e.g. fermion) first; each
+    inner level subdivides its parent's step by n_sub; the innermost
+    drift is the link update. The expensive force is evaluated n_sub
+    times less often than the cheap one."""
+    state = {"u": u.clone(), "P": P}
+
+    def step(h: float, lev: int):
+        if lev == len(levels):
+            state["u"] = _evolve_u(state["u"], state["P"], geo, h)
+            return
+        force, nsub = levels[lev]
+        hs = h / nsub
+        state["P"] = state["P"] + (0.5 * hs) * force(state["u"])
+        for i in range(nsub):
+            step(hs, lev + 1)
+            if i != nsub - 1:
+                state["P"] = state["P"] + hs * force(state["u"])
+        state["P"] = state["P"] + (0.5 * hs) * force(state["u"])
+
+    for _ in range(n_steps):
+        step(dt, 0)
+    return state["u"], state["P"]

@@ -172,3 +172,70 @@ def test_flow_energy_and_scale(geo, u_rand):
     t0 = flow_scale_t0(hist, target)
     if max(t2e) > target > t2e[0]:
         assert t0 is not None and hist[0][0] <= t0 <= hist[-1][0]
+
+
+def test_omelyan_beats_leapfrog(geo, u_rand):
+    """Omelyan 2MN: same O(dt^2) order, much smaller dH than leapfrog at
+    equal step count (and equal-force-eval comparison still wins)."""
+    from quda_amd.gauge import gauge_force, omelyan
+    beta = 5.5
+    P = random_momentum(geo, seed=93)
+
+    def force(uc):
+        return gauge_force(uc, geo, beta)
+
+    n, tau = 10, 0.5
+    u_lf, P_lf = leapfrog(u_rand, P, geo, beta, n, tau / n)
+    u_om, P_om = omelyan(u_rand, P, geo, force, n, tau / n)
+    H0 = mom_action(P) + gauge_action(u_rand, geo, beta)
+    dH_lf = abs(mom_action(P_lf) + gauge_action(u_lf, geo, beta) - H0)
+    dH_om = abs(mom_action(P_om) + gauge_action(u_om, geo, beta) - H0)
+    assert dH_om < dH_lf / 3, (dH_om, dH_lf)
+    # order check: 2x smaller dt -> ~4x smaller dH
+    u2, P2 = omelyan(u_rand, P, geo, force, 2 * n, tau / (2 * n))
+    dH2 = abs(mom_action(P2) + gauge_action(u2, geo, beta) - H0)
+    assert dH2 < dH_om / 2.5
+
+
+def test_nested_leapfrog_multiscale(geo, u_rand):
+    """Sexton-Weingarten: putting a scaled copy of the gauge force on an
+    inner scale conserves energy while evaluating the outer force 4x
+    less; single-level nesting reproduces plain leapfrog exactly."""
+    from quda_amd.gauge import gauge_force, nested_leapfrog
+    beta = 5.5
+    P = random_momentum(geo, seed=94)
+    calls = {"outer": 0, "inner": 0}
+
+    def f_outer(uc):
+        calls["outer"] += 1
+        return gauge_force(uc, geo, 0.3 * beta)
+
+    def f_inner(uc):
+        calls["inner"] += 1
+        return gauge_force(uc, geo, 0.7 * beta)
+
+    n, tau = 8, 0.4
+    u1, P1 = nested_leapfrog(u_rand, P, geo,
+                             [(f_outer, 1), (f_inner, 4)], n, tau / n)
+
+    def act(uc):
+        return (gauge_action(uc, geo, 0.3 * beta)
+                + gauge_action(uc, geo, 0.7 * beta))
+
+    H0 = mom_action(P) + act(u_rand)
+    dH = abs(mom_action(P1) + act(u1) - H0)
+    # O(dt^2): doubling the step count shrinks dH ~4x (hot-field forces
+    # are large, so check the order rather than an absolute bound)
+    u2, P2 = nested_leapfrog(u_rand, P, geo,
+                             [(f_outer, 1), (f_inner, 4)], 2 * n,
+                             tau / (2 * n))
+    dH2 = abs(mom_action(P2) + act(u2) - H0)
+    assert dH2 < dH / 2.5, (dH, dH2)
+    assert calls["inner"] > 3 * calls["outer"]
+    # single-level nested == plain leapfrog (same splitting)
+    u_lf, P_lf = leapfrog(u_rand, P, geo, beta, n, tau / n)
+    u_n1, P_n1 = nested_leapfrog(
+        u_rand, P, geo, [(lambda uc: gauge_force(uc, geo, beta), 1)],
+        n, tau / n)
+    err = (u_n1 - u_lf).abs().max().item()
+    assert err < 1e-12, err
