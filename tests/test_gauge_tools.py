@@ -214,7 +214,7 @@ def test_nested_leapfrog_multiscale(geo, u_rand):
         calls["inner"] += 1
         return gauge_force(uc, geo, 0.7 * beta)
 
-    n, tau = 8, 0.4
+    n, tau = 16, 0.4
     u1, P1 = nested_leapfrog(u_rand, P, geo,
                              [(f_outer, 1), (f_inner, 4)], n, tau / n)
 
