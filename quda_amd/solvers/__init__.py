@@ -7,6 +7,7 @@ from .ca import ca_cg_solve, ca_gcr_solve
 from .cg import SolverStats, cg_solve
 from .gcr import gcr_solve, mr_solve
 from .madwf import TransferLs, madwf_solve, train_transfer
+from .eigcg import IncrementalDeflation, eigcg_solve, inc_eigcg_solve
 from .mre import ChronoForecaster
 from .rational import (RationalApprox, rational_approx, rational_apply,
                        rhmc_pseudofermion_action)

@@ -314,3 +314,63 @@ def test_rhmc_action_consistency(system):
     rational_apply(d, t2, t1, ap_quarter, tol=1e-11)
     s2 = blas.re_dot(b_e, t2)
     assert abs(s - s2) < 1e-5 * abs(s)
+
+
+def test_eigcg_harvests_accurate_ritz_pairs():
+    """eigCG's side-harvested Ritz values must match the dense lowest
+    eigenvalues of MdagM (free field: lowest = (1-4 kappa)^2)."""
+    from quda_amd.models import DiracWilson
+    from quda_amd.solvers.eigcg import eigcg_solve
+    geo = LatticeGeometry((4, 4, 4, 8))
+    g = GaugeField(geo, "double").unit_()
+    kap = 0.245
+    d = DiracWilson(g, kap)
+    b = SpinorField(geo, "double").gaussian_(seed=701)
+    x = SpinorField(geo, "double")
+    harvest = []
+    st = eigcg_solve(d, x, b, nev=4, m=16, tol=1e-10, maxiter=4000,
+                     harvest=harvest)
+    assert st.converged
+    lam0 = (1.0 - 4.0 * kap) ** 2
+    lams = sorted(l for l, _ in harvest)
+    assert abs(lams[0] - lam0) < 1e-4 * max(lam0, 1e-6), (lams[0], lam0)
+    # residual of the best Ritz pair: ||A u - lam u|| small
+    lam, u = min(harvest, key=lambda p: p[0])
+    t = SpinorField(geo, "double")
+    tmp = SpinorField(geo, "double")
+    d.MdagM(t, u, tmp)
+    blas.axpy(-lam, u, t)
+    assert math.sqrt(blas.norm2(t)) < 1e-3
+
+
+def test_inc_eigcg_deflation_accelerates():
+    """Successive RHS on a weakly-disordered near-critical operator
+    converge faster once the incremental deflation space accumulates
+    (and every solution is a true solve)."""
+    from quda_amd.fields.gauge import project_su3
+    from quda_amd.models import DiracWilson
+    from quda_amd.solvers.eigcg import inc_eigcg_solve
+    geo = LatticeGeometry((4, 4, 4, 8))
+    gen = torch.Generator().manual_seed(41)
+    m_ = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                     dtype=torch.float64)
+    eye = torch.eye(3, dtype=torch.complex128)
+    u = project_su3(eye + 0.25 * torch.view_as_complex(m_))
+    g = GaugeField(geo, "double").from_complex(u)
+    d = DiracWilson(g, 0.24)
+    bs = [SpinorField(geo, "double").gaussian_(seed=900 + j)
+          for j in range(5)]
+    xs = [SpinorField(geo, "double") for _ in bs]
+    stats, defl = inc_eigcg_solve(d, xs, bs, nev=8, m=24, tol=1e-8,
+                                  maxiter=8000)
+    assert all(s.converged for s in stats)
+    assert len(defl.U) > 8
+    later = [s.iters for s in stats[1:]]
+    assert sum(later) / len(later) < stats[0].iters - 4, \
+        [s.iters for s in stats]
+    # solutions are true solutions of MdagM x = b
+    r = SpinorField(geo, "double")
+    t = SpinorField(geo, "double")
+    d.MdagM(r, xs[-1], t)
+    tr = math.sqrt(blas.xmy_norm2(bs[-1], r) / blas.norm2(bs[-1]))
+    assert tr < 1e-7, tr
