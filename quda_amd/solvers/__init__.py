@@ -8,6 +8,7 @@ from .cg import SolverStats, cg_solve
 from .gcr import gcr_solve, mr_solve
 from .madwf import TransferLs, madwf_solve, train_transfer
 from .eigcg import IncrementalDeflation, eigcg_solve, inc_eigcg_solve
+from .gmresdr import gmresdr_solve
 from .mre import ChronoForecaster
 from .rational import (RationalApprox, rational_approx, rational_apply,
                        rhmc_pseudofermion_action)
