@@ -57,6 +57,8 @@ class InverterType(Enum):
     MR = "mr"
     CA_CG = "ca-cg"
     CA_GCR = "ca-gcr"
+    GMRESDR = "gmresdr"
+    EIGCG = "eigcg"
 
 
 class SolutionType(Enum):
@@ -313,6 +315,16 @@ def _run_solver(d, x, b, p: InvertParam, sloppy_pair: bool):
     if inv == InverterType.GCR:
         return gcr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter,
                          precond=p.preconditioner)
+    if inv == InverterType.GMRESDR:
+        from .solvers.gmresdr import gmresdr_solve
+        return gmresdr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
+    if inv == InverterType.EIGCG:
+        from .solvers.eigcg import eigcg_solve
+        rhs = b
+        if not isinstance(d, DiracStaggeredPC):
+            rhs = b.clone_empty()
+            d.M(rhs, b, dagger=True)
+        return eigcg_solve(d, x, rhs, tol=p.tol, maxiter=p.maxiter)
     if inv == InverterType.MR:
         return mr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
     if inv == InverterType.CA_CG:

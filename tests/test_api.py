@@ -239,3 +239,15 @@ def test_verbosity_api(capsys):
     api.log_quda(2, "hidden")
     out = capsys.readouterr().out
     assert "visible" in out and "hidden" not in out
+
+
+def test_gmresdr_and_eigcg_through_api(resident):
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    b = _rand_spinor((2, 128, 4, 3), 551)
+    for inv in (InverterType.GMRESDR, InverterType.EIGCG):
+        p = InvertParam(dslash_type=DslashType.WILSON, inv_type=inv,
+                        kappa=0.12, tol=1e-9, maxiter=800)
+        x = api.invert_quda(b, p)
+        assert p.true_res < 1e-7, (inv, p.true_res)

@@ -11,13 +11,8 @@ import pytest
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def test_bench_two_ranks_gloo():
+def _launch_two_rank(port):
     env = dict(os.environ)
-    import socket
-    sock = socket.socket()
-    sock.bind(("127.0.0.1", 0))
-    port = sock.getsockname()[1]
-    sock.close()
     env["MASTER_ADDR"] = "127.0.0.1"
     env["MASTER_PORT"] = str(port)
     procs = []
@@ -29,6 +24,21 @@ def test_bench_two_ranks_gloo():
              "--lattice", "4,4,4,8", "--sloppy", "double", "--device", "cpu"],
             env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
     outs = [p.communicate(timeout=300) for p in procs]
+    return procs, outs
+
+
+def test_bench_two_ranks_gloo():
+    import socket
+    # the free-port probe is racy against lingering TCPStores from earlier
+    # gloo tests in the same session: retry with a fresh port
+    for attempt in range(3):
+        sock = socket.socket()
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
+        sock.close()
+        procs, outs = _launch_two_rank(port)
+        if all(p.returncode == 0 for p in procs):
+            break
     for p in procs:
         assert p.returncode == 0, outs
     line = [l for l in outs[0][0].splitlines() if l.startswith("{")][-1]
