@@ -31,6 +31,8 @@ SOLVERS = {
     "ca-gcr": ca_gcr_solve,
     "multishift-cg": multishift_cg_solve,
     "block-cg": block_cg_solve,
+    "gmresdr": gmresdr_solve,
+    "eigcg": eigcg_solve,
 }
 
 
@@ -47,4 +49,6 @@ __all__ = ["cg_solve", "SolverStats", "bicgstab_solve", "bicgstabl_solve",
            "multishift_cg_solve", "cgne_solve", "cgnr_solve", "cg3_solve",
            "sd_solve", "pcg_solve", "ChronoForecaster", "create_solver",
            "SOLVERS", "block_cg_solve", "RationalApprox", "rational_approx",
-           "rational_apply", "rhmc_pseudofermion_action"]
+           "rational_apply", "rhmc_pseudofermion_action", "gmresdr_solve",
+           "eigcg_solve", "inc_eigcg_solve", "IncrementalDeflation",
+           "TransferLs", "madwf_solve", "train_transfer"]

@@ -310,3 +310,144 @@ class Deflation:
             c = blas.c_dot(v, b) / lam
             blas.caxpy(c, v, x)
         return x
+
+
+def block_trlm_solve(op, n_ev: int, n_kr: int, x0: SpinorField, *,
+                     block_size: int = 4, tol: float = 1e-8,
+                     max_restarts: int = 100,
+                     poly: Optional[ChebyshevOp] = None,
+                     which: str = "smallest") -> EigResult:
+    """Block thick-restarted Lanczos (ref: lib/eig_block_trlm.cpp —
+    re-designed: the basis grows in `block_size` panels so the operator
+    and BLAS phases batch over the panel (GEMM-shaped on GPU); the
+    projected matrix is rebuilt from the explicit orthogonalization
+    coefficients, and the thick restart keeps the Ritz vectors plus the
+    pending (unprocessed) panel, whose couplings are recomputed when the
+    panel is processed — no closed-form boundary algebra to get wrong.
+    Per-matvec convergence trails scalar TRLM (smaller Krylov depth per
+    restart); the payoff is the batched panel apply on GPU."""
+    res = EigResult()
+    bs = block_size
+    assert n_kr > n_ev + bs
+    apply_op = (poly.apply if poly is not None
+                else (lambda o, i, t: op.MdagM(o, i, t)))
+    tmp = _new_like(x0)
+    w = _new_like(x0)
+
+    def ortho_append(vec, V, H, col):
+        """Orthogonalize `vec` against V, recording coefficients in
+        H[:, col]; append the normalized remainder (returns False on
+        breakdown)."""
+        for i, u in enumerate(V):
+            c = blas.c_dot(u, vec)
+            H[i, col] += c
+            blas.caxpy(-c, u, vec)
+        # second Gram-Schmidt pass for orthogonality at restart scale
+        for i, u in enumerate(V):
+            c = blas.c_dot(u, vec)
+            H[i, col] += c
+            blas.caxpy(-c, u, vec)
+        nb = sqrt(blas.norm2(vec))
+        if nb < 1e-14:
+            return False
+        H[len(V), col] += nb
+        blas.scal(1.0 / nb, vec)
+        V.append(vec)
+        return True
+
+    # initial panel
+    gen_seed = 4321
+    V: List[SpinorField] = []
+    H = np.zeros((n_kr + bs, n_kr + bs), dtype=complex)
+    if blas.norm2(x0) == 0.0:
+        x0.gaussian_(seed=gen_seed)
+    first = _new_like(x0)
+    blas.copy(first, x0)
+    blas.scal(1.0 / sqrt(blas.norm2(first)), first)
+    V.append(first)
+    while len(V) < bs:
+        v = _new_like(x0)
+        v.gaussian_(seed=gen_seed + len(V))
+        scratch = np.zeros((n_kr + bs, 1), dtype=complex)
+        if not ortho_append(v, V, scratch, 0):
+            v.gaussian_(seed=gen_seed + 1000 + len(V))
+            ortho_append(v, V, scratch, 0)
+
+    jp = 0          # processed columns
+    iters = 0
+    n_conv = 0
+    for restart in range(max_restarts):
+        # process panels until the basis would exceed n_kr
+        while jp + bs <= len(V) and len(V) + bs <= n_kr:
+            for c in range(jp, jp + bs):
+                apply_op(w, V[c], tmp)
+                iters += 1
+                vec = _new_like(x0)
+                blas.copy(vec, w)
+                if not ortho_append(vec, V, H, c):
+                    vv = _new_like(x0)
+                    vv.gaussian_(seed=gen_seed + 2000 + iters)
+                    scratch = np.zeros((n_kr + bs, 1), dtype=complex)
+                    ortho_append(vv, V, scratch, 0)
+            jp += bs
+        # Rayleigh-Ritz over the processed prefix. H[i, c] is filled when
+        # column c is processed; the mirror H[c, i] stayed zero whenever
+        # row c did not exist yet -- restore it from hermiticity instead
+        # of averaging (averaging would halve one-sided couplings).
+        Hp = H[:jp, :jp].copy()
+        mask = Hp == 0
+        Hp[mask] = Hp.conj().T[mask]
+        Hp = 0.5 * (Hp + Hp.conj().T)
+        wv, Z = np.linalg.eigh(Hp)
+        order = np.argsort(wv if which == "smallest" else -wv)
+        wv, Z = wv[order], Z[:, order]
+        # residual estimates from the pending-panel coupling rows
+        C = H[jp:len(V), :jp]
+        rnorm = np.linalg.norm(C @ Z, axis=0)
+        n_conv = 0
+        for i in range(n_ev):
+            if rnorm[i] < tol * max(abs(wv[i]), 1e-30):
+                n_conv += 1
+            else:
+                break
+        keep = min(max(n_ev + bs, 2 * n_ev), jp - 1)
+        if n_conv >= n_ev or restart == max_restarts - 1:
+            keep = n_ev
+        # rotate: new basis = Ritz vectors (keep) + pending panel
+        newV = []
+        for jcol in range(keep):
+            vj = _new_like(x0)
+            vj.zero_()
+            for i in range(jp):
+                blas.caxpy(complex(Z[i, jcol]), V[i], vj)
+            newV.append(vj)
+        pend = V[jp:len(V)]
+        # re-orthogonalize pending panel against the rotated Ritz basis
+        # (exact in theory; re-done for float hygiene)
+        Hn = np.zeros((n_kr + bs, n_kr + bs), dtype=complex)
+        Hn[:keep, :keep] = np.diag(wv[:keep])
+        V2 = list(newV)
+        scratch = np.zeros((n_kr + bs, 1), dtype=complex)
+        for pv in pend:
+            ortho_append(pv, V2, scratch, 0)
+        V = V2
+        H = Hn
+        jp = keep
+        if n_conv >= n_ev:
+            break
+    # final: exact eigenpairs + residuals
+    evecs = V[:n_ev]
+    evals, resids = [], []
+    r = _new_like(x0)
+    for i in range(n_ev):
+        op.MdagM(r, evecs[i], tmp)
+        lam = blas.re_dot(evecs[i], r)
+        blas.axpy(-lam, evecs[i], r)
+        evals.append(lam)
+        resids.append(sqrt(blas.norm2(r)))
+    res.evals = evals
+    res.evecs = evecs
+    res.residuals = resids
+    res.iters = iters
+    res.converged = n_conv >= n_ev
+    return res

@@ -139,3 +139,19 @@ def test_deflated_cg_fewer_iters(small):
     st1 = cg_solve(d, x_defl, b, tol=1e-10, maxiter=1000)
     assert st1.converged
     assert st1.iters <= st0.iters
+
+
+def test_block_trlm_matches_trlm():
+    """Block TRLM finds the same lowest spectrum as scalar TRLM."""
+    from quda_amd.models import DiracWilson
+    from quda_amd.solvers.eigen import block_trlm_solve, trlm_solve
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=71)
+    d = DiracWilson(g, 0.12)
+    r1 = trlm_solve(d, 6, 24, SpinorField(geo, "double"), tol=1e-8)
+    r2 = block_trlm_solve(d, 6, 32, SpinorField(geo, "double"),
+                          block_size=4, tol=1e-8, max_restarts=200)
+    assert r1.converged and r2.converged
+    for a, b in zip(r1.evals, r2.evals):
+        assert abs(a - b) < 1e-7 * max(abs(a), 1e-10), (a, b)
+    assert all(r < 1e-6 for r in r2.residuals), r2.residuals
