@@ -9,6 +9,7 @@ from .gcr import gcr_solve, mr_solve
 from .madwf import TransferLs, madwf_solve, train_transfer
 from .eigcg import IncrementalDeflation, eigcg_solve, inc_eigcg_solve
 from .gmresdr import gmresdr_solve
+from .mspcg import mspcg_solve, schwarz_precond
 from .mre import ChronoForecaster
 from .rational import (RationalApprox, rational_approx, rational_apply,
                        rhmc_pseudofermion_action)
@@ -33,6 +34,7 @@ SOLVERS = {
     "block-cg": block_cg_solve,
     "gmresdr": gmresdr_solve,
     "eigcg": eigcg_solve,
+    "mspcg": mspcg_solve,
 }
 
 
@@ -51,4 +53,5 @@ __all__ = ["cg_solve", "SolverStats", "bicgstab_solve", "bicgstabl_solve",
            "SOLVERS", "block_cg_solve", "RationalApprox", "rational_approx",
            "rational_apply", "rhmc_pseudofermion_action", "gmresdr_solve",
            "eigcg_solve", "inc_eigcg_solve", "IncrementalDeflation",
-           "TransferLs", "madwf_solve", "train_transfer"]
+           "TransferLs", "madwf_solve", "train_transfer", "mspcg_solve",
+           "schwarz_precond"]
