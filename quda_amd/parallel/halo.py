@@ -389,3 +389,94 @@ def _face_lex(geo: LatticeGeometry, mu: int, edge: int) -> torch.Tensor:
         c = geo.coords[:, mu]
         cache[key] = (c == edge).nonzero(as_tuple=True)[0].contiguous()
     return cache[key]
+
+
+class BatchSpinorHalo:
+    """Multi-RHS ghost buffers: per (mu,dir) ONE contiguous tensor
+    [n_rhs, ncomp/gw, depth*Fcb, gw] so the whole batch ships as a single
+    message per face (ref: dslash create_comms_batch, the merged-halo
+    multi-RHS path — the per-RHS kernels then consume their slice)."""
+
+    def __init__(self, geo: LatticeGeometry, precision: str, device,
+                 mask: int, n_rhs: int, ncomp: int = 12):
+        from ..fields.layout import DTYPE_OF
+        self.geo = geo
+        self.precision = precision
+        self.mask = mask
+        self.n_rhs = n_rhs
+        self.ncomp = ncomp
+        self.device = torch.device(device)
+        gw = ghost_width(ncomp, precision)
+        dt = DTYPE_OF[precision]
+        self.send, self.recv = {}, {}
+        self.send_nrm, self.recv_nrm = {}, {}
+        for mu in active_dims(mask):
+            fcb = geo.face_volume_cb(mu)
+            for d in (0, 1):
+                shape = (n_rhs, ncomp // gw, fcb, gw)
+                self.send[(mu, d)] = torch.empty(shape, dtype=dt,
+                                                 device=device)
+                self.recv[(mu, d)] = torch.empty(shape, dtype=dt,
+                                                 device=device)
+                if precision == "half":
+                    self.send_nrm[(mu, d)] = torch.empty(
+                        (n_rhs, fcb), dtype=torch.float32, device=device)
+                    self.recv_nrm[(mu, d)] = torch.empty(
+                        (n_rhs, fcb), dtype=torch.float32, device=device)
+
+    def pack_one(self, ext, i: int, inp, parity: int, dagger: bool) -> None:
+        geo = self.geo
+        empty = torch.empty(0, dtype=torch.float32, device=self.device)
+        from ..ops.dispatch import norm_or_empty
+        for mu in active_dims(self.mask):
+            fcb = geo.face_volume_cb(mu)
+            for d in (0, 1):
+                s01 = d ^ (1 if dagger else 0)
+                nrm = self.send_nrm.get((mu, d))
+                ext.pack_face(self.send[(mu, d)][i],
+                              nrm[i] if nrm is not None else empty,
+                              inp.data, norm_or_empty(inp),
+                              list(geo.dims), geo.parity_offset,
+                              geo.volume_cb, parity, mu, s01, d, fcb)
+
+    def exchange(self) -> None:
+        reqs = exchange_tensors_start(self.send, self.recv)
+        if self.precision == "half":
+            reqs += exchange_tensors_start(self.send_nrm, self.recv_nrm)
+        for r in reqs:
+            r.wait()
+
+    def exchange_start(self) -> list:
+        reqs = exchange_tensors_start(self.send, self.recv)
+        if self.precision == "half":
+            reqs += exchange_tensors_start(self.send_nrm, self.recv_nrm)
+        return reqs
+
+    def ghost_args(self, i: int):
+        """Per-RHS (ghost[8], ghost_nrm[8], face_cb[4]) slice views."""
+        empty = torch.empty(0, dtype=next(iter(self.recv.values())).dtype,
+                            device=self.device) if self.recv else None
+        empty_n = torch.empty(0, dtype=torch.float32, device=self.device)
+        ghosts, nrms = [], []
+        for mu in range(4):
+            for d in (0, 1):
+                g = self.recv.get((mu, d))
+                ghosts.append(g[i] if g is not None else empty)
+                n = self.recv_nrm.get((mu, d))
+                nrms.append(n[i] if n is not None else empty_n)
+        face_cb = [self.geo.face_volume_cb(mu) for mu in range(4)]
+        return ghosts, nrms, face_cb
+
+
+_BATCH_HALO_CACHE: Dict[tuple, "BatchSpinorHalo"] = {}
+
+
+def get_batch_halo(geo: LatticeGeometry, precision: str, device, mask: int,
+                   n_rhs: int, ncomp: int = 12) -> BatchSpinorHalo:
+    key = (geo.dims, geo.parity_offset, precision, str(device), mask,
+           n_rhs, ncomp)
+    h = _BATCH_HALO_CACHE.get(key)
+    if h is None:
+        h = BatchSpinorHalo(geo, precision, device, mask, n_rhs, ncomp)
+        _BATCH_HALO_CACHE[key] = h
+    return h
