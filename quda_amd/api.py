@@ -340,7 +340,11 @@ def invert_multishift_quda(b: torch.Tensor, p: InvertParam,
     the even-odd system."""
     assert p.solution_type == SolutionType.MATPC
     d = _make_dirac(p)
-    b_f = _wrap(b.unsqueeze(0) if b.dim() == 3 else b, p, 1)
+    # accept the source with or without the leading parity axis
+    # (Wilson site shape (4,3), staggered (3,))
+    wilson_site = b.shape[-2:] == (4, 3) if b.dim() >= 2 else False
+    has_par = b.dim() == (4 if wilson_site else 3)
+    b_f = _wrap(b if has_par else b.unsqueeze(0), p, 1)
     xs = [b_f.clone_empty() for _ in shifts]
     st = multishift_cg_solve(d, xs, b_f, shifts, tol=p.tol,
                              maxiter=p.maxiter)
