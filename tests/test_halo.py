@@ -557,3 +557,26 @@ def _worker_batch_dslash(rank, world, init_file):
 def test_batch_dslash_merged_halos_gloo(tmp_path):
     init_file = str(tmp_path / "init_batch")
     mp.spawn(_worker_batch_dslash, args=(2, init_file), nprocs=2, join=True)
+
+
+@pytest.mark.parametrize("mask", list(range(16)))
+@pytest.mark.parametrize("dagger", [False, True])
+def test_wilson_all_partition_masks_cpu(mask, dagger):
+    """All 16 comm-partition patterns as self-wraparound (the reference's
+    dslash_ctest partition axis): forced-partition result must equal the
+    unpartitioned one exactly."""
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g0 = GaugeField(geo, "double").random_su3_(seed=57)
+    u = g0.to_complex()
+    src = SpinorField(geo, "double", n_parity=1).gaussian_(seed=58)
+    out_ref = SpinorField(geo, "double", n_parity=1)
+    dslash_wilson(out_ref, src, g0, 0, dagger=dagger)
+    try:
+        comms.set_forced_partition(mask)
+        g = GaugeField(geo, "double").from_complex(u)
+        out = SpinorField(geo, "double", n_parity=1)
+        dslash_wilson(out, src, g, 0, dagger=dagger)
+    finally:
+        comms.set_forced_partition(0)
+    err = (out.to_complex() - out_ref.to_complex()).abs().max().item()
+    assert err < 1e-13, (mask, dagger, err)
