@@ -89,8 +89,20 @@ def naik_links(u: torch.Tensor, geo: LatticeGeometry) -> torch.Tensor:
     return _from_lex(N, geo)
 
 
-def unitarize_links(f: torch.Tensor) -> torch.Tensor:
+def unitarize_links(f: torch.Tensor, *, svd_rel_error: float = 1e-6,
+                    return_failures: bool = False):
     """Project fattened links to U(3) via the symmetric polar factor
-    W = F (F^d F)^{-1/2} (ref: lib/unitarize_links_quda.cu — SVD route)."""
+    W = F (F^d F)^{-1/2} (ref: lib/unitarize_links_quda.cu — SVD route).
+    Counts links whose smallest singular value is below
+    svd_rel_error * largest (the reference's `num_failures` device
+    counter for ill-conditioned fattened links, interface_quda.cpp:119);
+    pass return_failures=True to receive (W, n_failures)."""
     U_, S_, Vh = torch.linalg.svd(f)
-    return U_ @ Vh
+    W = U_ @ Vh
+    fails = int((S_[..., -1] < svd_rel_error * S_[..., 0]).sum().item())
+    if return_failures:
+        return W, fails
+    if fails:
+        import warnings
+        warnings.warn(f"unitarize_links: {fails} ill-conditioned links")
+    return W

@@ -45,6 +45,7 @@ class MG:
             vectors = generate_null_vectors(
                 op, param.n_vec, tol=param.null_tol,
                 maxiter=param.null_maxiter, seed=param.seed)
+        self.null_vectors = vectors
         self.transfer = Transfer(op.geo, param.block, vectors)
         self.coarse = build_coarse_op(op, self.transfer)
         self.coarse_mg = None
@@ -109,3 +110,17 @@ class MG:
         if p.nu_post > 0:
             mr_solve(op, z, r, tol=1e-10, maxiter=p.nu_post,
                      omega=p.smoother_omega, zero_init=False)
+
+
+    # -- null-space persistence (ref: multigrid.cpp:1249 vec_load /
+    # vec_outfile over VectorIO; here utils.io checksummed storage) -------
+    def save_vectors(self, path: str) -> None:
+        from ..utils.io import save_field
+        save_field(path, self.null_vectors,
+                   meta={"block": self.param.block,
+                         "n_vec": self.param.n_vec})
+
+    @staticmethod
+    def load_vectors(path: str, device="cpu", precision="double"):
+        from ..utils.io import load_field
+        return load_field(path, device=device, precision=precision)

@@ -16,6 +16,7 @@ class Tuner:
         rp = os.environ.get("QUDA_AMD_RESOURCE_PATH")
         self.path = path or (os.path.join(rp, "tunecache.tsv") if rp else None)
         self.cache: Dict[str, Tuple[str, float]] = {}
+        self.counts: Dict[str, int] = {}  # per-key consults (profile role)
         if self.path and os.path.exists(self.path):
             self._load()
 
@@ -51,6 +52,7 @@ class Tuner:
              warmup: int = 2, iters: int = 5) -> str:
         """Measure `run` under each candidate config (applied by `setup`),
         cache + persist the winner, and leave it applied."""
+        self.counts[key] = self.counts.get(key, 0) + 1
         if key in self.cache:
             cfg = self.cache[key][0]
             setup(cfg)
@@ -76,6 +78,16 @@ class Tuner:
         self._save()
         setup(best)
         return best
+
+    def profile_dump(self, path: str) -> None:
+        """Per-key consult-count x tuned-time table (role of the
+        reference's profile_N.tsv, tune.cpp:566)."""
+        with open(path, "w") as f:
+            f.write("# key\tcount\ttuned_us\ttotal_us\n")
+            for k in sorted(self.counts):
+                cfg_t = self.cache.get(k, ("?", 0.0))
+                n = self.counts[k]
+                f.write(f"{k}\t{n}\t{cfg_t[1]:.3f}\t{n * cfg_t[1]:.3f}\n")
 
 
 _TUNER: Optional[Tuner] = None

@@ -81,3 +81,45 @@ def test_launch_trace(tmp_path):
         assert "dslash_wilson" in open(p).read()
     finally:
         dispatch.set_trace(False)
+
+
+def test_mg_vector_io_roundtrip(tmp_path):
+    """MG null-vector persistence (vec_outfile/vec_load role): saved
+    vectors rebuild an equivalent hierarchy."""
+    from quda_amd import GaugeField, LatticeGeometry, SpinorField
+    from quda_amd.models import DiracWilson
+    from quda_amd.mg.mg import MG, MGParam
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=191)
+    d = DiracWilson(g, 0.12)
+    mg = MG(d, MGParam(block=(2, 2, 2, 2), n_vec=4))
+    p = str(tmp_path / "nullvecs.pt")
+    mg.save_vectors(p)
+    vecs = MG.load_vectors(p)
+    mg2 = MG(d, MGParam(block=(2, 2, 2, 2), n_vec=4), vectors=vecs)
+    import torch
+    assert torch.allclose(mg.transfer.V, mg2.transfer.V)
+
+
+def test_unitarize_failure_counter():
+    import torch
+    from quda_amd.gauge.hisq import unitarize_links
+    good = torch.eye(3, dtype=torch.complex128).expand(10, 3, 3).contiguous()
+    w, f = unitarize_links(good, return_failures=True)
+    assert f == 0
+    bad = good.clone()
+    bad[0, :, 2] = bad[0, :, 0]  # rank-deficient link
+    w, f = unitarize_links(bad, return_failures=True)
+    assert f == 1, f
+
+
+def test_tuner_profile_dump(tmp_path):
+    from quda_amd.utils.tune import Tuner
+    t = Tuner()
+    calls = []
+    t.tune("k1", ["a", "b"], lambda c: calls.append(c), lambda: None)
+    t.tune("k1", ["a", "b"], lambda c: calls.append(c), lambda: None)
+    p = str(tmp_path / "profile.tsv")
+    t.profile_dump(p)
+    lines = open(p).read().splitlines()
+    assert any(l.startswith("k1\t2\t") for l in lines), lines
