@@ -201,6 +201,17 @@ class GaugeField:
                 f"recon={self.reconstruct}, device={self.device})")
 
 
+def project_su3_polar(u: torch.Tensor) -> torch.Tensor:
+    """GAUGE-COVARIANT SU(3) projection: U(3) polar factor (SVD) with the
+    det phase divided out (the reference's smearing-grade projection,
+    su3_project.cuh; Gram-Schmidt below is NOT covariant — fine for
+    random-field generation, wrong inside smearing)."""
+    U_, _, Vh = torch.linalg.svd(u)
+    W = U_ @ Vh
+    ph = torch.linalg.det(W) ** (1.0 / 3.0)
+    return W / ph[..., None, None]
+
+
 def project_su3(u: torch.Tensor) -> torch.Tensor:
     """Project [..., 3, 3] complex onto SU(3): Gram-Schmidt rows + det fix."""
     r0 = u[..., 0, :]

@@ -311,3 +311,213 @@ def flow_scale_w0(hist, target: float = 0.3):
         if d < target <= d_next:
             return t + (target - d) * (hist[i + 1][0] - t) / (d_next - d)
     return None
+
+
+def path_product(u: torch.Tensor, geo: LatticeGeometry,
+                 path) -> torch.Tensor:
+    """[V,3,3] product of links along `path` starting at every site
+    (ref: lib/gauge_loop_trace.cu + gauge_path_helper.cuh — generic
+    signed-direction paths). path entries: +-(mu+1), e.g. the (0,3)
+    plaquette is (1, 4, -1, -4). Comm-aware via shift_lex."""
+    from ..parallel.halo import shift_lex
+    U = _to_lex(u, geo)
+    V = geo.volume
+    P = torch.eye(3, dtype=u.dtype, device=u.device).expand(V, 3, 3).clone()
+    # walk the path: keep the running product in the frame of the START
+    # site by shifting link fields back by the accumulated displacement
+    disp = [0, 0, 0, 0]
+
+    def fetch(field):
+        out = field
+        for d in range(4):
+            s = disp[d]
+            step = 1 if s > 0 else -1
+            for _ in range(abs(s)):
+                out = shift_lex(out, geo, d, step)
+        return out
+
+    for step in path:
+        mu = abs(step) - 1
+        if step > 0:
+            P = P @ fetch(U[mu])
+            disp[mu] += 1
+        else:
+            disp[mu] -= 1
+            P = P @ fetch(U[mu]).conj().mT
+    return P
+
+
+def loop_trace(u: torch.Tensor, geo: LatticeGeometry, paths,
+               coeffs=None) -> complex:
+    """sum_i c_i <tr P_i>/3 over closed paths, globally averaged (the
+    gaugeLoopTraceQuda role). Default coefficients 1."""
+    from ..parallel import comms
+    coeffs = coeffs or [1.0] * len(paths)
+    tot = 0.0 + 0.0j
+    n = comms.allreduce_sum(float(geo.volume))
+    for c, p in zip(coeffs, paths):
+        P = path_product(u, geo, p)
+        tr = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1) / 3.0
+        s = tr.sum()
+        tot += c * complex(comms.allreduce_sum(s.real.item()),
+                           comms.allreduce_sum(s.imag.item())) / n
+    return tot
+
+
+def improved_gauge_action(u: torch.Tensor, geo: LatticeGeometry,
+                          beta: float, *, c1: float = -1.0 / 12.0) -> float:
+    """Luscher-Weisz/Symanzik-improved action
+    S = beta sum_x [ c0 sum_pl (1 - Re tr P/3)
+                   + c1 sum_rect (1 - Re tr R/3) ],  c0 = 1 - 8 c1
+    (c1 = -1/12 tree-level Symanzik, -0.331 Iwasaki; ref: the
+    computeGaugeForceQuda path-coefficient interface)."""
+    from ..parallel import comms
+    c0 = 1.0 - 8.0 * c1
+    n = comms.allreduce_sum(float(geo.volume))
+    s_pl = 0.0
+    s_rt = 0.0
+    for mu in range(4):
+        for nu in range(4):
+            if mu == nu:
+                continue
+            if mu < nu:
+                P = path_product(u, geo, (mu + 1, nu + 1, -(mu + 1),
+                                          -(nu + 1)))
+                tr = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).real / 3.0
+                s_pl += (1.0 - comms.allreduce_sum(tr.sum().item()) / n)
+            # 2x1 rectangles: both orientations (mu-long and nu-long)
+            R = path_product(u, geo, (mu + 1, mu + 1, nu + 1, -(mu + 1),
+                                      -(mu + 1), -(nu + 1)))
+            tr = torch.diagonal(R, dim1=-2, dim2=-1).sum(-1).real / 3.0
+            s_rt += (1.0 - comms.allreduce_sum(tr.sum().item()) / n)
+    return beta * (c0 * s_pl + c1 * s_rt) * n
+
+
+def improved_gauge_force(u: torch.Tensor, geo: LatticeGeometry,
+                         beta: float, *, c1: float = -1.0 / 12.0
+                         ) -> torch.Tensor:
+    """Force of the improved action by reverse-mode differentiation of
+    the loop traces (single-rank; the plaquette-only force keeps the
+    analytic multi-rank path). Same validated convention as
+    autograd_fermion_force: F = (1/2) TA[U g^dag] with g the torch
+    Wirtinger grad of S — c1=0 reproduces gauge_force exactly and the
+    tests check finite differences + dH conservation."""
+    from ..parallel import comms
+    assert comms.comm_size() == 1, "autograd gauge force is single-rank"
+    u_req = u.detach().clone().requires_grad_(True)
+    s = _action_autograd(u_req, geo, beta, c1)
+    s.backward()
+    g = u_req.grad
+    F = torch.empty_like(u)
+    for mu in range(4):
+        for p in (0, 1):
+            F[mu, p] = 0.5 * project_ta(u[mu, p] @ g[mu, p].conj().mT)
+    return F
+
+
+def _action_autograd(z, geo, beta, c1):
+    """improved_gauge_action as a differentiable torch scalar (no
+    .item() calls)."""
+    c0 = 1.0 - 8.0 * c1
+    from ..parallel.halo import shift_lex
+    s_pl = 0.0
+    s_rt = 0.0
+    for mu in range(4):
+        for nu in range(4):
+            if mu == nu:
+                continue
+            if mu < nu:
+                P = path_product(z, geo, (mu + 1, nu + 1, -(mu + 1),
+                                          -(nu + 1)))
+                s_pl = s_pl + (geo.volume
+                               - torch.diagonal(P, dim1=-2, dim2=-1)
+                               .sum(-1).real.sum() / 3.0)
+            R = path_product(z, geo, (mu + 1, mu + 1, nu + 1, -(mu + 1),
+                                      -(mu + 1), -(nu + 1)))
+            s_rt = s_rt + (geo.volume
+                           - torch.diagonal(R, dim1=-2, dim2=-1)
+                           .sum(-1).real.sum() / 3.0)
+    return beta * (c0 * s_pl + c1 * s_rt)
+
+
+def det_trace(u: torch.Tensor, geo: LatticeGeometry) -> tuple:
+    """Global mean link determinant and trace (ref:
+    lib/pgauge_det_trace.cu)."""
+    from ..parallel import comms
+    U = _to_lex(u, geo)
+    det = torch.linalg.det(U.reshape(-1, 3, 3))
+    tr = torch.diagonal(U, dim1=-2, dim2=-1).sum(-1).reshape(-1) / 3.0
+    n = comms.allreduce_sum(float(det.numel()))
+    d = complex(comms.allreduce_sum(det.real.mean().item() * det.numel()) / n,
+                comms.allreduce_sum(det.imag.mean().item() * det.numel()) / n)
+    t = complex(comms.allreduce_sum(tr.real.sum().item()) / n,
+                comms.allreduce_sum(tr.imag.sum().item()) / n)
+    return d, t
+
+
+def hyp_smear(u: torch.Tensor, geo: LatticeGeometry,
+              alpha: tuple = (0.75, 0.6, 0.3), n_iter: int = 1
+              ) -> torch.Tensor:
+    """HYP (hypercubic) smearing, Hasenfratz-Knechtli 3-level restricted
+    staples (ref: lib/gauge_hyp.cu / kernels/gauge_hyp.cuh — re-derived:
+    level-1 decorated links Vbar_{mu;nu rho} use staples only in the
+    remaining direction, level-2 Vtil_{mu;nu} from Vbar staples, level 3
+    assembles the fat link; each level applies the COVARIANT polar SU(3)
+    projection)."""
+    from ..fields.gauge import project_su3_polar as project_su3
+    from ..parallel.halo import shift_lex
+    a1, a2, a3 = alpha
+
+    def staple_dir(Ua, Ub, mu, nu):
+        """staple of link field Ua (direction mu) through links Ub
+        (direction nu): Ub(x) Ua(x+nu) Ub(x+mu)^dag + backward."""
+        fwd = Ub @ shift_lex(Ua, geo, nu, +1) @ \
+            shift_lex(Ub, geo, mu, +1).conj().mT
+        Ub_m = shift_lex(Ub, geo, nu, -1)
+        bwd = Ub_m.conj().mT @ shift_lex(Ua, geo, nu, -1) @ \
+            shift_lex(Ub_m, geo, mu, +1)
+        return fwd + bwd
+
+    out = u
+    for _ in range(n_iter):
+        U = _to_lex(out, geo)
+        # level 1: Vbar[mu][(nu,rho)] decorated in the single remaining dir
+        Vbar = {}
+        for mu in range(4):
+            for nu in range(4):
+                if nu == mu:
+                    continue
+                for rho in range(nu + 1, 4):
+                    if rho == mu:
+                        continue
+                    eta = next(d for d in range(4)
+                               if d not in (mu, nu, rho))
+                    S = staple_dir(U[mu], U[eta], mu, eta)
+                    Vbar[(mu, nu, rho)] = project_su3(
+                        (1 - a3) * U[mu] + (a3 / 2.0) * S)
+        # level 2: Vtil[mu][nu] from Vbar staples in the two remaining dirs
+        Vtil = {}
+        for mu in range(4):
+            for nu in range(4):
+                if nu == mu:
+                    continue
+                S = torch.zeros_like(U[mu])
+                for rho in range(4):
+                    if rho in (mu, nu):
+                        continue
+                    key_a = (mu,) + tuple(sorted((nu, rho)))
+                    key_b = (rho,) + tuple(sorted((nu, mu)))
+                    S = S + staple_dir(Vbar[key_a], Vbar[key_b], mu, rho)
+                Vtil[(mu, nu)] = project_su3(
+                    (1 - a2) * U[mu] + (a2 / 4.0) * S)
+        # level 3: final fat link from Vtil staples in all three dirs
+        Unew = torch.empty_like(U)
+        for mu in range(4):
+            S = torch.zeros_like(U[mu])
+            for nu in range(4):
+                if nu == mu:
+                    continue
+                S = S + staple_dir(Vtil[(mu, nu)], Vtil[(nu, mu)], mu, nu)
+            Unew[mu] = project_su3((1 - a1) * U[mu] + (a1 / 6.0) * S)
+        out = _from_lex(Unew, geo)
+    return out

@@ -239,3 +239,83 @@ def test_nested_leapfrog_multiscale(geo, u_rand):
         n, tau / n)
     err = (u_n1 - u_lf).abs().max().item()
     assert err < 1e-12, err
+
+
+def test_path_product_and_loop_trace(geo, u_rand):
+    """Generic path machinery: the (0,3) plaquette path equals the 1x1
+    Wilson loop; a path and its reverse give conjugate traces."""
+    from quda_amd.gauge.ops import loop_trace, path_product, wilson_loop
+    lt = loop_trace(u_rand, geo, [(1, 4, -1, -4)])
+    wl = wilson_loop(u_rand, geo, 1, 1, mu=0, nu=3)
+    assert abs(lt - wl) < 1e-12
+    fwd = loop_trace(u_rand, geo, [(1, 2, -1, -2)])
+    rev = loop_trace(u_rand, geo, [(2, 1, -2, -1)])
+    assert abs(fwd - rev.conjugate()) < 1e-12
+
+
+def test_improved_gauge_force_matches_plaquette_at_c1_zero(geo, u_rand):
+    from quda_amd.gauge.ops import gauge_force, improved_gauge_force
+    F0 = improved_gauge_force(u_rand, geo, 5.5, c1=0.0)
+    F1 = gauge_force(u_rand, geo, 5.5)
+    assert (F0 - F1).abs().max().item() < 1e-12
+
+
+def test_improved_gauge_force_finite_difference(geo, u_rand):
+    """Symanzik (c1=-1/12) force: dS/dt along Udot = P U equals
+    -2 tr(P F)."""
+    import torch
+    from quda_amd.gauge.ops import (_from_lex, _to_lex,
+                                    improved_gauge_action,
+                                    improved_gauge_force)
+    P = random_momentum(geo, seed=183)
+    eps = 1e-6
+    U = _to_lex(u_rand, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    beta, c1 = 5.5, -1.0 / 12.0
+    dSdt = (improved_gauge_action(up, geo, beta, c1=c1)
+            - improved_gauge_action(um, geo, beta, c1=c1)) / (2 * eps)
+    F = improved_gauge_force(u_rand, geo, beta, c1=c1)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
+
+
+def test_hyp_smear_raises_plaquette(geo, u_rand):
+    from quda_amd.gauge import plaquette
+    from quda_amd.gauge.ops import hyp_smear
+    p0, _, _ = plaquette(u_rand, geo)
+    ph, _, _ = plaquette(hyp_smear(u_rand, geo), geo)
+    pa, _, _ = plaquette(ape_smear(u_rand, geo, 0.5, 1), geo)
+    assert ph > p0
+    assert ph > pa  # HYP fattens harder than one APE step
+
+
+def test_hyp_smear_gauge_covariance(geo, u_rand):
+    """HYP of a gauge-transformed field = transform of the HYP field."""
+    import torch
+    from quda_amd.fields.gauge import project_su3
+    from quda_amd.gauge.ops import _from_lex, _to_lex, hyp_smear
+    gen = torch.Generator().manual_seed(185)
+    m = torch.randn((geo.volume, 3, 3, 2), generator=gen,
+                    dtype=torch.float64)
+    gt = project_su3(torch.view_as_complex(m))
+    U = _to_lex(u_rand, geo)
+    Ut = torch.empty_like(U)
+    for mu in range(4):
+        idx = geo.neighbor_lex(mu, +1)
+        Ut[mu] = gt @ U[mu] @ gt[idx].conj().mT
+    lhs = _to_lex(hyp_smear(_from_lex(Ut, geo), geo), geo)
+    W = _to_lex(hyp_smear(u_rand, geo), geo)
+    for mu in range(4):
+        idx = geo.neighbor_lex(mu, +1)
+        Wt = gt @ W[mu] @ gt[idx].conj().mT
+        assert (lhs[mu] - Wt).abs().max().item() < 1e-10
+
+
+def test_det_trace(geo, u_rand):
+    from quda_amd.gauge.ops import det_trace
+    d, t = det_trace(u_rand, geo)
+    assert abs(d - 1.0) < 1e-10  # SU(3)
+    assert abs(t) < 1.0
