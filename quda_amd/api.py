@@ -411,6 +411,9 @@ def perform_gauge_smear_quda(kind: str, n_steps: int, coeff: float) -> None:
         u = stout_smear(u, _R.geo, coeff, n_steps)
     elif kind == "wilson_flow":
         u = wilson_flow(u, _R.geo, coeff, n_steps)
+    elif kind == "hyp":
+        from .gauge import hyp_smear
+        u = hyp_smear(u, _R.geo, n_iter=n_steps)
     else:
         raise ValueError(kind)
     load_gauge_quda(u, _R.gauge_param)

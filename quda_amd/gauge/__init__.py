@@ -8,6 +8,8 @@ setup/analysis utilities, not the solver hot path."""
 from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
                   plaquette, polyakov_loop, project_ta, staple_sum,
                   stout_smear, topological_charge, wilson_flow, wilson_loop,
+                  hyp_smear, loop_trace, path_product, det_trace,
+                  improved_gauge_action, improved_gauge_force,
                   energy_density, wilson_flow_measure, flow_scale_t0,
                   flow_scale_w0)
 from .hmc import (hmc_trajectory, leapfrog, mom_action, nested_leapfrog,
@@ -21,6 +23,8 @@ __all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
            "project_ta", "exp_su3", "ape_smear", "stout_smear",
            "wilson_flow", "polyakov_loop", "topological_charge",
            "energy_density", "wilson_flow_measure", "flow_scale_t0",
+           "hyp_smear", "loop_trace", "path_product", "det_trace",
+           "improved_gauge_action", "improved_gauge_force",
            "flow_scale_w0",
            "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",
            "omelyan", "nested_leapfrog",
