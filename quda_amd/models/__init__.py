@@ -4,8 +4,10 @@ from .staggered_kd import (DiracImprovedStaggeredKD, DiracStaggeredKD,
                            KDBlockInverse)
 from .dwf import (DiracDomainWall, DiracDomainWallPC, DiracMobius,
                   DiracMobiusPC, DiracZMobius, DiracZMobiusPC,
-                  DiracMobiusEofa, DiracMobiusEofaPC)
-from .dirac import (Dirac, DiracClover, DiracCloverHasenbuschTwist,
+                  DiracMobiusEofa, DiracMobiusEofaPC,
+                  DiracDomainWall4D, DiracDomainWall4DPC)
+from .dirac import (Dirac, DiracClover, DiracG5M, DiracMdagMLocal,
+                    apply_gamma5, DiracCloverHasenbuschTwist,
                     DiracCloverHasenbuschTwistPC, DiracCloverPC,
                     DiracNdegTwistedMass, DiracNdegTwistedMassPC,
                     DiracTwistedClover, DiracTwistedCloverPC,
@@ -21,4 +23,6 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracTwistedCloverPC", "DiracDomainWall", "DiracDomainWallPC",
            "DiracMobius", "DiracMobiusPC", "DiracZMobius", "DiracZMobiusPC",
            "DiracMobiusEofa", "DiracMobiusEofaPC", "DiracStaggeredKD",
-           "DiracImprovedStaggeredKD", "KDBlockInverse"]
+           "DiracImprovedStaggeredKD", "KDBlockInverse",
+           "DiracDomainWall4D", "DiracDomainWall4DPC", "DiracG5M",
+           "DiracMdagMLocal", "apply_gamma5"]

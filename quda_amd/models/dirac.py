@@ -552,3 +552,58 @@ class DiracTwistedCloverPC(Dirac, _CloverMixin):
 
     def flops_per_site(self) -> int:
         return 1320 + 2 * 504 + 48
+
+
+def apply_gamma5(out: "SpinorField", inp: "SpinorField") -> "SpinorField":
+    """out = g5 in (DeGrand-Rossi diag(1,1,-1,-1)); device path rides the
+    twist kernel (i*g5 then a -i rescale), CPU flips the lower spins."""
+    from ..ops import blas
+    from ..ops.dispatch import apply_twist_field, on_gpu
+    if on_gpu(out, inp):
+        out.zero_()
+        apply_twist_field(out, inp, 0.0, 1.0)  # out = i g5 in
+        blas.caxpby(complex(0.0, -1.0), out, 0.0, out)
+        return out
+    v = inp.to_complex().clone()
+    v[..., 2:4, :] = -v[..., 2:4, :]
+    out.from_complex(v)
+    return out
+
+
+class DiracG5M:
+    """gamma5-wrapped operator functor (ref: DiracG5M dirac_quda.h:2441):
+    hermitian (indefinite) for gamma5-hermitian actions, so hermitian
+    eigensolvers and MINRES-type methods apply directly."""
+
+    def __init__(self, op):
+        self.op = op
+        self.geo = op.geo
+
+    def new_spinor(self, precision="double", n_parity=2):
+        return self.op.new_spinor(precision, n_parity)
+
+    def M(self, out, inp, dagger=False):
+        # for g5-hermitian M, (g5 M)^dag = M^dag g5 = g5 M: self-adjoint
+        t = self.op.new_spinor(inp.precision, inp.n_parity)
+        self.op.M(t, inp, dagger=False)
+        return apply_gamma5(out, t)
+
+    def MdagM(self, out, inp, tmp):
+        # (g5 M)^2 = g5 M g5 M = M^dag M for g5-hermitian M
+        self.M(tmp, inp)
+        return self.M(out, tmp)
+
+
+class DiracMdagMLocal:
+    """MdagM with communications disabled (ref: DiracMdagMLocal
+    dirac_quda.h:2510 — the MSPCG inner operator: each rank applies its
+    local operator with frozen boundaries)."""
+
+    def __init__(self, op):
+        self.op = op
+        self.geo = op.geo
+
+    def MdagM(self, out, inp, tmp):
+        from ..parallel import comms
+        with comms.solo_mode():
+            return self.op.MdagM(out, inp, tmp)

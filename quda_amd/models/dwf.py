@@ -288,3 +288,19 @@ class DiracMobiusEofaPC(DiracMobiusEofa, DiracMobiusPC):
     """Symmetric even-even PC EOFA (structure from DiracMobiusPC, the
     rank-1-extended s-ops from DiracMobiusEofa)."""
     pass
+
+
+class DiracDomainWall4D(DiracDomainWall):
+    """4-d-preconditioning flavor of Shamir DWF (ref:
+    lib/dirac_domain_wall_4d.cpp). In this engine's decomposition
+    M = A - (1/2) Dhat B with B = 1 for Shamir, the 4-d even-odd Schur
+    complement IS the preconditioned system DiracDomainWallPC builds
+    (M_pc = 1 - 1/4 Ainv Dhat Ainv Dhat), so the 4-d-PC operator
+    coincides with the symmetric PC class; the alias keeps the
+    reference's operator taxonomy addressable."""
+    pass
+
+
+class DiracDomainWall4DPC(DiracDomainWallPC):
+    """See DiracDomainWall4D."""
+    pass

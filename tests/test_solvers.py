@@ -437,3 +437,42 @@ def test_gmresdr_progresses_on_kd_operator():
     import math as _m
     tr1 = _m.sqrt(blas.xmy_norm2(bp, r1) / blas.norm2(bp))
     assert st2.true_resid < tr1 / 3, (st2.true_resid, tr1)
+
+
+def test_g5m_hermitian_and_squares_to_mdagm():
+    from quda_amd.models import DiracG5M, DiracWilson
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=621)
+    d = DiracWilson(g, 0.12)
+    h = DiracG5M(d)
+    a = SpinorField(geo, "double").gaussian_(seed=622)
+    b = SpinorField(geo, "double").gaussian_(seed=623)
+    Ha = SpinorField(geo, "double")
+    Hb = SpinorField(geo, "double")
+    h.M(Ha, a)
+    h.M(Hb, b)
+    lhs = (b.to_complex().conj() * Ha.to_complex()).sum()
+    rhs = (Hb.to_complex().conj() * a.to_complex()).sum()
+    assert abs(lhs - rhs) < 1e-10 * abs(lhs)  # hermitian: <b,Ha> = <Hb,a>
+    # (g5 M)^2 == MdagM
+    sq = SpinorField(geo, "double")
+    t = SpinorField(geo, "double")
+    h.MdagM(sq, a, t)
+    mm = SpinorField(geo, "double")
+    d.MdagM(mm, a, t)
+    assert (sq.to_complex() - mm.to_complex()).abs().max().item() < 1e-11
+
+
+def test_mdagm_local_equals_global_single_rank():
+    from quda_amd.models import DiracMdagMLocal, DiracWilson
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=624)
+    d = DiracWilson(g, 0.12)
+    loc = DiracMdagMLocal(d)
+    a = SpinorField(geo, "double").gaussian_(seed=625)
+    o1 = SpinorField(geo, "double")
+    o2 = SpinorField(geo, "double")
+    t = SpinorField(geo, "double")
+    d.MdagM(o1, a, t)
+    loc.MdagM(o2, a, t)
+    assert (o1.to_complex() - o2.to_complex()).abs().max().item() < 1e-13
