@@ -98,6 +98,7 @@ class InvertParam:
     cuda_prec: str = "double"
     cuda_prec_sloppy: str = "half"
     preconditioner: Optional[object] = None  # e.g. MG.precond
+    deflation: Optional[object] = None  # Deflation (newDeflationQuda)
     # outputs (ref: out-fields of QudaInvertParam)
     iter: int = 0
     true_res: float = 0.0
@@ -298,6 +299,10 @@ def _run_solver(d, x, b, p: InvertParam, sloppy_pair: bool):
         if not isinstance(d, DiracStaggeredPC):
             rhs = b.clone_empty()
             d.M(rhs, b, dagger=True)
+        if p.deflation is not None:
+            # deflated initial guess (ref: deflated_invert_test /
+            # solver.cpp eig-deflation hookup)
+            p.deflation.guess(x, rhs)
         if sloppy_pair and p.cuda_prec_sloppy != p.cuda_prec:
             ds = _make_dirac(p, sloppy=True)
             return cg_solve(d, x, rhs, op_sloppy=ds,
@@ -553,7 +558,9 @@ def new_deflation_quda(p: InvertParam, e: EigParam):
     return a Deflation object (plugs into solver initial guesses)."""
     from .solvers.eigen import Deflation
     evals, evecs = eigensolve_quda(p, e)
-    return Deflation([float(v.real) for v in evals], evecs)
+    npar = 1 if p.solution_type == SolutionType.MATPC else 2
+    fields = [_wrap(v, p, npar) for v in evecs]
+    return Deflation([float(complex(v).real) for v in evals], fields)
 
 
 _CHRONO = {}

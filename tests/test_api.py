@@ -251,3 +251,23 @@ def test_gmresdr_and_eigcg_through_api(resident):
                         kappa=0.12, tol=1e-9, maxiter=800)
         x = api.invert_quda(b, p)
         assert p.true_res < 1e-7, (inv, p.true_res)
+
+
+def test_deflated_invert_through_api(resident):
+    """newDeflationQuda + deflated invertQuda (deflated_invert_test
+    role): the deflated solve converges in no more iterations and to the
+    same solution."""
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.124, tol=1e-9,
+                    maxiter=600)
+    e = EigParam(n_ev=6, n_kr=24, tol=1e-7)
+    defl = api.new_deflation_quda(p, e)
+    b = _rand_spinor((2, 128, 4, 3), 560)
+    x0 = api.invert_quda(b, p)
+    it0 = p.iter
+    p2 = InvertParam(**{**p.__dict__, "deflation": defl})
+    x1 = api.invert_quda(b, p2)
+    assert p2.iter <= it0, (p2.iter, it0)
+    assert (x1 - x0).abs().max().item() < 1e-6
