@@ -240,10 +240,10 @@ static void convert(at::Tensor dst, at::Tensor dst_n, at::Tensor src,
 
 static void clover_apply(at::Tensor out, at::Tensor out_n, at::Tensor in,
                          at::Tensor in_n, at::Tensor clover, int64_t parity,
-                         int64_t Vcb) {
+                         int64_t Vcb, int64_t v_stride, int64_t s_offset) {
   CloverApplyCall c{};
-  c.out = field_of(out, out_n, Vcb);
-  c.in = field_of(in, in_n, Vcb);
+  c.out = field_of_off(out, out_n, v_stride ? v_stride : Vcb, s_offset);
+  c.in = field_of_off(in, in_n, v_stride ? v_stride : Vcb, s_offset);
   c.clover = clover.data_ptr();
   c.parity = (int)parity;
   c.Vcb = Vcb;
@@ -405,7 +405,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convert", &convert, "precision conversion copy", py::arg("dst"),
         py::arg("dst_n"), py::arg("src"), py::arg("src_n"), py::arg("Vcb"),
         py::arg("sites"), py::arg("ncomp") = 24);
-  m.def("clover_apply", &clover_apply, "clover site-matrix apply");
+  m.def("clover_apply", &clover_apply, "clover term apply",
+        py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
+        py::arg("clover"), py::arg("parity"), py::arg("Vcb"),
+        py::arg("v_stride") = 0, py::arg("s_offset") = 0);
   m.def("twist_apply", &twist_apply, "twisted-mass T(b) apply",
         py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
         py::arg("b_re"), py::arg("b_im"), py::arg("Vcb"), py::arg("sites"),

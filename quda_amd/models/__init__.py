@@ -10,6 +10,7 @@ from .dirac import (Dirac, DiracClover, DiracG5M, DiracMdagMLocal,
                     apply_gamma5, DiracCloverHasenbuschTwist,
                     DiracCloverHasenbuschTwistPC, DiracCloverPC,
                     DiracNdegTwistedMass, DiracNdegTwistedMassPC,
+                    DiracNdegTwistedClover, DiracNdegTwistedCloverPC,
                     DiracTwistedClover, DiracTwistedCloverPC,
                     DiracTwistedMass, DiracTwistedMassPC,
                     DiracWilson, DiracWilsonPC)
@@ -25,4 +26,5 @@ __all__ = ["Dirac", "DiracWilson", "DiracWilsonPC", "DiracClover",
            "DiracMobiusEofa", "DiracMobiusEofaPC", "DiracStaggeredKD",
            "DiracImprovedStaggeredKD", "KDBlockInverse",
            "DiracDomainWall4D", "DiracDomainWall4DPC", "DiracG5M",
-           "DiracMdagMLocal", "apply_gamma5"]
+           "DiracMdagMLocal", "apply_gamma5", "DiracNdegTwistedClover",
+           "DiracNdegTwistedCloverPC"]

@@ -13,6 +13,8 @@ Conventions (kappa normalization, matching the reference semantics):
 
 from __future__ import annotations
 
+import torch
+
 from typing import Optional
 
 from ..fields.clover import CloverField
@@ -607,3 +609,131 @@ class DiracMdagMLocal:
         from ..parallel import comms
         with comms.solo_mode():
             return self.op.MdagM(out, inp, tmp)
+
+
+class DiracNdegTwistedClover(DiracNdegTwistedMass):
+    """Non-degenerate twisted-clover doublet (ref: the
+    QUDA_TWISTED_CLOVER ndeg branch of lib/dirac_twisted_clover.cpp /
+    kernels/dslash_ndeg_twisted_clover.cuh):
+      M = (C + i 2 kappa mu g5 tau3 - 2 kappa eps tau1) - kappa D
+    on an ls=2 flavor doublet. With X = i a g5 tau3 + b tau1 and
+    [C, X] = 0, the flavor-local inverse is
+      A^-1 = (C - X) (C^2 + a^2 - b^2)^-1
+    with the site-local G^-1 = (C^2 + a^2 - b^2)^-1 precomputed as a
+    CloverField (same dynamic-inversion role as the degenerate PC)."""
+
+    def __init__(self, gauge: GaugeField, clover: CloverField, kappa: float,
+                 mu: float, epsilon: float):
+        super().__init__(gauge, kappa, mu, epsilon)
+        self.clover = clover
+        import torch
+        A = clover.to_complex()
+        eye = torch.eye(12, dtype=A.dtype, device=A.device)
+        G = A @ A + (self.a_t ** 2 - self.b_t ** 2) * eye
+        self.g_inv = CloverField(gauge.geo, clover.precision,
+                                 gauge.device).from_matrices(
+                                     torch.linalg.inv(G))
+
+    def tmp(self, name, like):
+        key = (name, like.precision, str(like.device), like.n_parity, 2)
+        t = self._tmps.get(key)
+        if t is None:
+            t = SpinorField(self.geo, like.precision, like.device,
+                            like.n_parity, ls=2)
+            self._tmps[key] = t
+        return t
+
+    def _clov5(self, out: SpinorField, inp: SpinorField, parity: int,
+               field: CloverField) -> SpinorField:
+        """out = field * in on BOTH flavor slices of a single-parity
+        ls=2 field."""
+        from ..ops.dispatch import apply_clover, on_gpu
+        V = self.geo.volume_cb
+        if on_gpu(out, inp):
+            for s in (0, 1):
+                apply_clover(out, inp, field, parity, v_stride=2 * V,
+                             s_offset=s * V)
+            return out
+        from ..ops import reference as ref
+        A = field.to_complex()[parity]
+        psi = inp.to_complex()[0].reshape(2, V, 4, 3)
+        res = torch.stack([ref.apply_clover(A, psi[0]),
+                           ref.apply_clover(A, psi[1])])
+        out.from_complex(res.reshape(2 * V, 4, 3).unsqueeze(0))
+        return out
+
+    # parity-aware A applications (clover needs the parity; the TM base
+    # class applications are x-local and parity-blind)
+    def _apply_A_p(self, out, inp, parity, dagger=False):
+        from ..ops.dispatch import apply_twist_field
+        sgn = -1.0 if dagger else 1.0
+        self._clov5(out, inp, parity, self.clover)       # out = C in
+        from ..ops.dispatch import dwf5_op
+        dwf5_op(out, inp, 0.0, self.b_t, -1.0, kind=0, a=1.0, x=out)
+        apply_twist_field(out, inp, 0.0, sgn * self.a_t, tau3=True,
+                          acc=True)
+        return out
+
+    def _apply_Ainv_p(self, out, inp, parity, dagger=False):
+        """out = (C - X) G^-1 in."""
+        from ..ops import blas
+        from ..ops.dispatch import apply_twist_field, dwf5_op
+        sgn = -1.0 if dagger else 1.0
+        t = self.tmp("ntc_g", inp)
+        self._clov5(t, inp, parity, self.g_inv)          # t = G^-1 in
+        self._clov5(out, t, parity, self.clover)         # out = C t
+        # out -= X t  (X = i a g5 tau3 + b tau1, dagger flips a)
+        dwf5_op(out, t, 0.0, -self.b_t, -1.0, kind=0, a=1.0, x=out)
+        apply_twist_field(out, t, 0.0, -sgn * self.a_t, tau3=True,
+                          acc=True)
+        return out
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops.dispatch import dslash_wilson_slice, dwf_halo_exchange
+        assert inp.n_parity == 2
+        for p in (0, 1):
+            op = out.parity_view(p)
+            self._apply_A_p(op, inp.parity_view(p), p, dagger)
+            io = inp.parity_view(1 - p)
+            h = dwf_halo_exchange(io, 1 - p, dagger)
+            for s in (0, 1):
+                dslash_wilson_slice(op, io, self.gauge, p, s, dagger,
+                                    a=-self.kappa, x=op, halo=h)
+        return out
+
+
+class DiracNdegTwistedCloverPC(DiracNdegTwistedClover):
+    """Symmetric even-odd PC doublet with clover:
+    M_pc = 1 - kappa^2 Ainv D Ainv D (operator order as in the TM PC)."""
+
+    def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
+        from ..ops import blas
+        from ..ops.dispatch import dslash_wilson_slice, dwf_halo_exchange
+        assert inp.n_parity == 1
+        t = self.tmp("ntc_t", inp)
+        u = self.tmp("ntc_u", inp)
+        k2 = -self.kappa ** 2
+
+        def dhat(dst, src, parity):
+            h = dwf_halo_exchange(src, 1 - parity, dagger)
+            dst.zero_()
+            for s in (0, 1):
+                dslash_wilson_slice(dst, src, self.gauge, parity, s, dagger,
+                                    a=1.0, x=dst, halo=h)
+            return dst
+
+        if not dagger:
+            dhat(t, inp, 1)
+            self._apply_Ainv_p(u, t, 1)
+            dhat(t, u, 0)
+            self._apply_Ainv_p(u, t, 0)
+            blas.copy(out, inp)
+            blas.axpy(k2, u, out)
+        else:
+            self._apply_Ainv_p(u, inp, 0, dagger=True)
+            dhat(t, u, 1)
+            self._apply_Ainv_p(u, t, 1, dagger=True)
+            dhat(t, u, 0)
+            blas.copy(out, inp)
+            blas.axpy(k2, t, out)
+        return out

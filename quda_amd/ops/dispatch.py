@@ -317,15 +317,20 @@ def apply_twist_field(out: SpinorField, inp: SpinorField, br: float,
 
 
 def apply_clover(out: SpinorField, inp: SpinorField, clover, parity: int,
-                 inverse: bool = False):
-    """out = A(parity) in (standalone; used by prepare/reconstruct)."""
+                 inverse: bool = False, v_stride: int = 0,
+                 s_offset: int = 0):
+    """out = A(parity) in (standalone; used by prepare/reconstruct).
+    v_stride/s_offset address one 4-d slice of a 5-d (doublet) field, as
+    in dslash_wilson_slice; the CPU path for slices is handled by the
+    callers (they stage the full complex tensor)."""
     if on_gpu(out, inp):
         ext = hip_ext()
         ext.clover_apply(out.data, norm_or_empty(out), inp.data,
                          norm_or_empty(inp),
                          clover.inv_data if inverse else clover.data,
-                         parity, out.geo.volume_cb)
+                         parity, out.geo.volume_cb, v_stride, s_offset)
         return out
+    assert v_stride == 0, "CPU slice path handled by callers"
     A = clover.to_complex(inverse=inverse)[parity]
     psi = inp.to_complex()[0]
     out.from_complex(ref.apply_clover(A, psi).unsqueeze(0))
