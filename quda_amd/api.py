@@ -43,8 +43,14 @@ class DslashType(Enum):
     TWISTED_MASS = "twisted-mass"
     TWISTED_CLOVER = "twisted-clover"
     STAGGERED = "staggered"
+    ASQTAD = "asqtad"               # improved staggered (fat+long links)
     DOMAIN_WALL = "domain-wall"
+    DOMAIN_WALL_4D = "domain-wall-4d"
     MOBIUS = "mobius"
+    ZMOBIUS = "zmobius"
+    MOBIUS_EOFA = "mobius-eofa"
+    NDEG_TWISTED_MASS = "ndeg-twisted-mass"
+    NDEG_TWISTED_CLOVER = "ndeg-twisted-clover"
 
 
 class InverterType(Enum):
@@ -87,6 +93,14 @@ class InvertParam:
     kappa: float = 0.135
     mass: float = 0.05          # staggered
     mu: float = 0.0             # twisted
+    epsilon: float = 0.0        # non-degenerate doublet splitting
+    b5_z: Optional[list] = None  # zMobius complex per-slice b5
+    c5_z: Optional[list] = None  # zMobius complex per-slice c5
+    eofa_pm: int = 1
+    eofa_shift: float = 0.0
+    mq1: float = 0.01
+    mq2: float = 0.08
+    mq3: float = 0.08
     m5: float = 1.8             # DWF height
     Ls: int = 8
     b5: float = 1.5
@@ -126,6 +140,8 @@ class EigParam:
 class _Resident:
     geo: Optional[LatticeGeometry] = None
     gauge: Optional[GaugeField] = None
+    fat_gauge: Optional[GaugeField] = None
+    long_gauge: Optional[GaugeField] = None
     gauge_sloppy: Optional[GaugeField] = None
     clover: Optional[CloverField] = None
     clover_sloppy: Optional[CloverField] = None
@@ -218,14 +234,54 @@ def _make_dirac(p: InvertParam, sloppy: bool = False):
     if t == DslashType.MOBIUS:
         return (DiracMobiusPC(g, p.mass, p.m5, p.Ls, p.b5, p.c5) if pc
                 else DiracMobius(g, p.mass, p.m5, p.Ls, p.b5, p.c5))
+    if t == DslashType.DOMAIN_WALL_4D:
+        from .models import DiracDomainWall4D, DiracDomainWall4DPC
+        return (DiracDomainWall4DPC(g, p.mass, p.m5, p.Ls) if pc
+                else DiracDomainWall4D(g, p.mass, p.m5, p.Ls))
+    if t == DslashType.ZMOBIUS:
+        from .models import DiracZMobius, DiracZMobiusPC
+        b5 = p.b5_z or [p.b5] * p.Ls
+        c5 = p.c5_z or [p.c5] * p.Ls
+        return (DiracZMobiusPC(g, p.mass, p.m5, p.Ls, b5, c5) if pc
+                else DiracZMobius(g, p.mass, p.m5, p.Ls, b5, c5))
+    if t == DslashType.MOBIUS_EOFA:
+        from .models import DiracMobiusEofa, DiracMobiusEofaPC
+        kw = dict(mq1=p.mq1, mq2=p.mq2, mq3=p.mq3, eofa_pm=p.eofa_pm,
+                  eofa_shift=p.eofa_shift)
+        return (DiracMobiusEofaPC(g, p.m5, p.Ls, p.b5, p.c5, **kw) if pc
+                else DiracMobiusEofa(g, p.m5, p.Ls, p.b5, p.c5, **kw))
+    if t == DslashType.NDEG_TWISTED_MASS:
+        from .models import DiracNdegTwistedMass, DiracNdegTwistedMassPC
+        return (DiracNdegTwistedMassPC(g, p.kappa, p.mu, p.epsilon) if pc
+                else DiracNdegTwistedMass(g, p.kappa, p.mu, p.epsilon))
+    if t == DslashType.NDEG_TWISTED_CLOVER:
+        assert cl is not None, "load_clover_quda first"
+        from .models import (DiracNdegTwistedClover,
+                             DiracNdegTwistedCloverPC)
+        return (DiracNdegTwistedCloverPC(g, cl, p.kappa, p.mu, p.epsilon)
+                if pc else
+                DiracNdegTwistedClover(g, cl, p.kappa, p.mu, p.epsilon))
+    if t == DslashType.ASQTAD:
+        from .models import (DiracImprovedStaggered,
+                             DiracImprovedStaggeredPC)
+        assert _R.fat_gauge is not None, "compute/load KS links first"
+        return (DiracImprovedStaggeredPC(_R.fat_gauge, _R.long_gauge,
+                                         p.mass) if pc
+                else DiracImprovedStaggered(_R.fat_gauge, _R.long_gauge,
+                                            p.mass))
     raise ValueError(t)
 
 
 def _wrap(v: torch.Tensor, p: InvertParam, n_parity: int) -> SpinorField:
     geo = _R.geo
-    nspin = 1 if p.dslash_type == DslashType.STAGGERED else 4
-    ls = p.Ls if p.dslash_type in (DslashType.DOMAIN_WALL,
-                                   DslashType.MOBIUS) else 1
+    nspin = 1 if p.dslash_type in (DslashType.STAGGERED,
+                                   DslashType.ASQTAD) else 4
+    ls = p.Ls if p.dslash_type in (
+        DslashType.DOMAIN_WALL, DslashType.DOMAIN_WALL_4D,
+        DslashType.MOBIUS, DslashType.ZMOBIUS,
+        DslashType.MOBIUS_EOFA) else (
+        2 if p.dslash_type in (DslashType.NDEG_TWISTED_MASS,
+                               DslashType.NDEG_TWISTED_CLOVER) else 1)
     f = SpinorField(geo, p.cuda_prec, _R.gauge.device, n_parity,
                     nspin=nspin, ls=ls)
     f.from_complex(v.to(_R.gauge.device))
@@ -461,7 +517,13 @@ def compute_ks_link_quda(coeffs=None):
     """Fat + long links of the resident field (ref: computeKSLinkQuda)."""
     from .gauge.hisq import asqtad_coefficients, fat_links, naik_links
     c = coeffs or asqtad_coefficients()
-    return fat_links(_R.u_complex, _R.geo, c), naik_links(_R.u_complex, _R.geo)
+    fat = fat_links(_R.u_complex, _R.geo, c)
+    lng = naik_links(_R.u_complex, _R.geo)
+    gp = _R.gauge_param
+    _R.fat_gauge = GaugeField(_R.geo, gp.cuda_prec, gp.device).from_complex(fat)
+    _R.long_gauge = GaugeField(_R.geo, gp.cuda_prec, gp.device,
+                               shift=3).from_complex(lng)
+    return fat, lng
 
 
 def mat_dag_mat_quda(inp: torch.Tensor, p: InvertParam) -> torch.Tensor:

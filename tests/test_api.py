@@ -271,3 +271,38 @@ def test_deflated_invert_through_api(resident):
     x1 = api.invert_quda(b, p2)
     assert p2.iter <= it0, (p2.iter, it0)
     assert (x1 - x0).abs().max().item() < 1e-6
+
+
+def test_all_action_types_through_api(resident):
+    """Every DslashType constructs and solves through invertQuda (the
+    invert_test dslash-type axis)."""
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    pcl = InvertParam(dslash_type=DslashType.CLOVER, kappa=0.11,
+                      clover_csw=1.2)
+    api.load_clover_quda(pcl)
+    api.compute_ks_link_quda()
+    cases = [
+        (DslashType.DOMAIN_WALL_4D, dict(mass=0.04, Ls=4), (4, 3), 4),
+        (DslashType.ZMOBIUS,
+         dict(mass=0.04, Ls=4,
+              b5_z=[1.5 + 0.1j, 1.4 - 0.05j, 1.6 + 0.02j, 1.5 - 0.08j],
+              c5_z=[0.5 + 0.1j, 0.4 - 0.05j, 0.6 + 0.02j, 0.5 - 0.08j]),
+         (4, 3), 4),
+        (DslashType.MOBIUS_EOFA,
+         dict(Ls=4, mq1=0.04, eofa_shift=-0.2), (4, 3), 4),
+        (DslashType.NDEG_TWISTED_MASS,
+         dict(kappa=0.11, mu=0.2, epsilon=0.1), (4, 3), 2),
+        (DslashType.NDEG_TWISTED_CLOVER,
+         dict(kappa=0.11, mu=0.2, epsilon=0.1), (4, 3), 2),
+        (DslashType.ASQTAD, dict(mass=0.1), (3,), 1),
+    ]
+    g = torch.Generator().manual_seed(570)
+    for t, kw, site, ls in cases:
+        p = InvertParam(dslash_type=t, inv_type=InverterType.CGNR,
+                        tol=1e-8, maxiter=3000, **kw)
+        b = torch.view_as_complex(torch.randn(
+            (2, 128 * ls, *site, 2), generator=g, dtype=torch.float64))
+        x = api.invert_quda(b, p)
+        assert p.true_res < 1e-6, (t, p.true_res)
