@@ -453,11 +453,17 @@ def plaq_quda() -> tuple:
 
 def gauge_observables_quda() -> dict:
     from .gauge import plaquette, polyakov_loop, topological_charge
+    from .gauge import det_trace, energy_density
     tot, sp, tm = plaquette(_R.u_complex, _R.geo)
+    ep, ec = energy_density(_R.u_complex, _R.geo)
+    d, t = det_trace(_R.u_complex, _R.geo)
     return {
         "plaquette": (tot, sp, tm),
         "polyakov_loop": polyakov_loop(_R.u_complex, _R.geo),
         "qcharge": topological_charge(_R.u_complex, _R.geo),
+        "energy": (ep, ec),
+        "det": d,
+        "trace": t,
     }
 
 

@@ -111,7 +111,8 @@ def gaussian_noise(f: SpinorField, seed: int) -> SpinorField:
 
 def dilute(src: SpinorField, scheme: str = "spin") -> List[SpinorField]:
     """Split a source into orthogonal dilution components
-    (ref: spinor_dilute.cuh; schemes: spin, color, even-odd)."""
+    (ref: spinor_dilute.cuh; schemes: spin, color, even-odd,
+    time-slice)."""
     geo = src.geo
     c = src.to_complex()
     out = []
@@ -138,6 +139,20 @@ def dilute(src: SpinorField, scheme: str = "spin") -> List[SpinorField]:
             f = src.clone_empty()
             cc = torch.zeros_like(c)
             cc[p] = c[p]
+            f.from_complex(cc)
+            out.append(f)
+    elif scheme == "time":
+        tcoord = geo.coords[:, 3].to(torch.int64)
+        from ..fields.geometry import checkerboard_split
+        tc = torch.stack([tcoord[geo.lex_of_cb[0]],
+                          tcoord[geo.lex_of_cb[1]]])  # [2, Vcb]
+        if src.ls > 1:
+            tc = tc.repeat(1, src.ls)
+        for t in range(geo.dims[3]):
+            f = src.clone_empty()
+            cc = torch.zeros_like(c)
+            m = (tc == t)
+            cc[m] = c[m]
             f.from_complex(cc)
             out.append(f)
     else:

@@ -101,3 +101,21 @@ def test_pion_correlator_free_field():
     # monotone decay toward the midpoint
     for t in range(T // 2):
         assert C[t] > C[t + 1] * 0.999, (t, C[t].item(), C[t + 1].item())
+
+
+def test_time_dilution_partition_of_unity():
+    from quda_amd import GaugeField, LatticeGeometry, SpinorField
+    from quda_amd.ops.contract import dilute
+    import torch
+    geo = LatticeGeometry((4, 4, 4, 8))
+    src = SpinorField(geo, "double").gaussian_(seed=281)
+    parts = dilute(src, "time")
+    assert len(parts) == 8
+    tot = torch.zeros_like(src.to_complex())
+    for p in parts:
+        tot += p.to_complex()
+    assert (tot - src.to_complex()).abs().max().item() < 1e-14
+    # orthogonality of distinct slices
+    a = parts[0].to_complex()
+    b = parts[3].to_complex()
+    assert abs((a.conj() * b).sum()) < 1e-14
