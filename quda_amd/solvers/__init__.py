@@ -9,7 +9,7 @@ from .gcr import gcr_solve, mr_solve
 from .madwf import TransferLs, madwf_solve, train_transfer
 from .eigcg import IncrementalDeflation, eigcg_solve, inc_eigcg_solve
 from .gmresdr import gmresdr_solve
-from .mspcg import mspcg_solve, schwarz_precond
+from .mspcg import dd_gcr_solve, mspcg_solve, schwarz_precond
 from .mre import ChronoForecaster
 from .rational import (RationalApprox, rational_approx, rational_apply,
                        rhmc_pseudofermion_action)
@@ -35,6 +35,7 @@ SOLVERS = {
     "gmresdr": gmresdr_solve,
     "eigcg": eigcg_solve,
     "mspcg": mspcg_solve,
+    "dd-gcr": dd_gcr_solve,
 }
 
 
@@ -54,4 +55,4 @@ __all__ = ["cg_solve", "SolverStats", "bicgstab_solve", "bicgstabl_solve",
            "rational_apply", "rhmc_pseudofermion_action", "gmresdr_solve",
            "eigcg_solve", "inc_eigcg_solve", "IncrementalDeflation",
            "TransferLs", "madwf_solve", "train_transfer", "mspcg_solve",
-           "schwarz_precond"]
+           "schwarz_precond", "dd_gcr_solve"]

@@ -39,3 +39,24 @@ def mspcg_solve(op, x: SpinorField, b: SpinorField, *,
     communication/compute ratio of the machine."""
     return pcg_solve(op, x, b, precond=schwarz_precond(
         op, inner_iters=inner_iters), tol=tol, maxiter=maxiter)
+
+
+def dd_gcr_solve(op, x: SpinorField, b: SpinorField, *,
+                 inner_iters: int = 6, tol: float = 1e-8,
+                 maxiter: int = 500, nkrylov: int = 10) -> SolverStats:
+    """Domain-decomposed GCR: flexible GCR preconditioned by the same
+    communication-free local solve (ref: the DD preconditioner option of
+    lib/inv_gcr_quda.cpp — the >100-GPU strong-scaling configuration of
+    the reference, arXiv:1109.2935). Nonsymmetric-friendly flavor of
+    mspcg_solve."""
+    from .gcr import gcr_solve
+
+    def precond(z: SpinorField, r: SpinorField):
+        from ..parallel import comms
+        from .variants import cgnr_solve
+        with comms.solo_mode():
+            z.zero_()
+            cgnr_solve(op, z, r, tol=1e-30, maxiter=inner_iters)
+
+    return gcr_solve(op, x, b, tol=tol, maxiter=maxiter, nkrylov=nkrylov,
+                     precond=precond)

@@ -514,6 +514,17 @@ def _worker_mspcg(rank, world, init_file):
         assert st1.iters < st0.iters, (st1.iters, st0.iters)
         err = (x1.to_complex() - x0.to_complex()).abs().max().item()
         assert err < 1e-6, err
+        # DD-GCR flavor: solves the NONSYMMETRIC system directly
+        from quda_amd.solvers import dd_gcr_solve
+        x2 = SpinorField(lg, "double")
+        st2 = dd_gcr_solve(d, x2, b, inner_iters=6, tol=1e-9, maxiter=500)
+        assert st2.converged, st2
+        r2f = SpinorField(lg, "double")
+        d.M(r2f, x2)
+        import math as _m
+        from quda_amd.ops import blas as _bl
+        tr = _m.sqrt(_bl.xmy_norm2(b, r2f) / _bl.norm2(b))
+        assert tr < 1e-7, tr
     finally:
         dist.destroy_process_group()
 
