@@ -227,3 +227,128 @@ def hisq_fermion_force(u: torch.Tensor, geo: LatticeGeometry, mass: float,
     F = autograd_fermion_force(u, geo, apply_M, X.to_complex(),
                                Y.to_complex())
     return S_f, F
+
+
+def _oracle_cg(apply_A, b: torch.Tensor, tol: float, maxiter: int
+               ) -> torch.Tensor:
+    """Plain CG on a hermitian-PD complex-tensor operator."""
+    x = torch.zeros_like(b)
+    r = b.clone()
+    p = r.clone()
+    r2 = (r.conj() * r).sum().real
+    b2 = r2.clone()
+    for _ in range(maxiter):
+        Ap = apply_A(p)
+        alpha = r2 / (p.conj() * Ap).sum().real
+        x = x + alpha * p
+        r = r - alpha * Ap
+        r2n = (r.conj() * r).sum().real
+        if r2n < tol * tol * b2:
+            break
+        p = r + (r2n / r2) * p
+        r2 = r2n
+    return x
+
+
+def hasenbusch_action_and_force(u: torch.Tensor, geo: LatticeGeometry,
+                                kappa: float, mu_h: float,
+                                phi1: torch.Tensor, phi2: torch.Tensor, *,
+                                cg_tol: float = 1e-10,
+                                cg_maxiter: int = 2000):
+    """Hasenbusch mass-preconditioned two-flavor Wilson action
+    (ref: the DiracCloverHasenbuschTwist consumers; W = M + i mu_h g5):
+
+        S = phi1^d (W W^d)^-1 phi1  +  phi2^d W (M M^d)^-1 W^d phi2
+
+    (det(MM^d) = det(WW^d) * det ratio). Forces by the adjoint trick:
+    solve once with u detached, then differentiate the surrogate
+    2 Re<w(u), y> - <y, A(u) y> whose u-gradient equals the true dS.
+    Returns (S, F) with the validated F = 1/2 TA[U g^d] convention."""
+    from ..ops.reference import dslash_wilson_full
+    from .ops import project_ta
+
+    def apply_M(ut, psi):
+        return psi - kappa * dslash_wilson_full(ut, psi, geo)
+
+    def apply_Mdag(ut, psi):
+        g5psi = psi.clone()
+        g5psi[..., 2:4, :] = -g5psi[..., 2:4, :]
+        t = apply_M(ut, g5psi)
+        t[..., 2:4, :] = -t[..., 2:4, :]
+        return t
+
+    def g5(psi):
+        o = psi.clone()
+        o[..., 2:4, :] = -o[..., 2:4, :]
+        return o
+
+    def apply_W(ut, psi):
+        return apply_M(ut, psi) + 1j * mu_h * g5(psi)
+
+    def apply_Wdag(ut, psi):
+        return apply_Mdag(ut, psi) - 1j * mu_h * g5(psi)
+
+    ud = u.detach()
+    # solves at detached u
+    X1 = _oracle_cg(lambda v: apply_W(ud, apply_Wdag(ud, v)), phi1,
+                    cg_tol, cg_maxiter)
+    w2 = apply_Wdag(ud, phi2)
+    Y2 = _oracle_cg(lambda v: apply_M(ud, apply_Mdag(ud, v)), w2,
+                    cg_tol, cg_maxiter)
+
+    u_req = u.detach().clone().requires_grad_(True)
+    # S1 surrogate: 2Re<phi1, X1> - <X1, W W^d X1>
+    s1 = (2.0 * (phi1.conj() * X1).sum().real
+          - (X1.conj() * apply_W(u_req, apply_Wdag(u_req, X1))).sum().real)
+    # S2 surrogate: 2Re<W^d(u) phi2, Y2> - <Y2, M M^d Y2>
+    s2 = (2.0 * (apply_Wdag(u_req, phi2).conj() * Y2).sum().real
+          - (Y2.conj() * apply_M(u_req, apply_Mdag(u_req, Y2))).sum().real)
+    s = s1 + s2
+    s.backward()
+    g = u_req.grad
+    F = torch.empty_like(u)
+    for mu in range(4):
+        for p in (0, 1):
+            F[mu, p] = 0.5 * project_ta(u[mu, p] @ g[mu, p].conj().mT)
+    S = float((phi1.conj() * X1).sum().real
+              + (w2.conj() * Y2).sum().real)
+    return S, F
+
+
+def hasenbusch_refresh(u: torch.Tensor, geo: LatticeGeometry, kappa: float,
+                       mu_h: float, seed: int, *, cg_tol: float = 1e-10,
+                       cg_maxiter: int = 2000):
+    """(phi1, phi2) heatbath: phi1 = W eta1 and phi2 = W^-dag M eta2 make
+    both action terms eta^d eta distributed."""
+    from ..ops.reference import dslash_wilson_full
+    gen = torch.Generator().manual_seed(seed)
+
+    def gauss():
+        return torch.view_as_complex(
+            torch.randn((2, geo.volume_cb, 4, 3, 2), generator=gen,
+                        dtype=torch.float64)) / (2.0 ** 0.5)
+
+    def apply_M(psi):
+        return psi - kappa * dslash_wilson_full(u, psi, geo)
+
+    def g5(psi):
+        o = psi.clone()
+        o[..., 2:4, :] = -o[..., 2:4, :]
+        return o
+
+    def apply_Mdag(psi):
+        return g5(apply_M(g5(psi)))
+
+    def apply_W(psi):
+        return apply_M(psi) + 1j * mu_h * g5(psi)
+
+    def apply_Wdag(psi):
+        return apply_Mdag(psi) - 1j * mu_h * g5(psi)
+
+    eta1, eta2 = gauss(), gauss()
+    phi1 = apply_W(eta1)
+    b = apply_M(eta2)
+    y = _oracle_cg(lambda v: apply_W(apply_Wdag(v)), apply_W(b),
+                   cg_tol, cg_maxiter)
+    # y = (W W^d)^-1 W b = W^-dag b
+    return phi1, y

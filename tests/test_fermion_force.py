@@ -164,3 +164,63 @@ def test_hisq_force_finite_difference():
     trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
     assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
         (dSdt, -2 * trPF)
+
+
+def test_hasenbusch_force_finite_difference(setup):
+    """Mass-preconditioned (Hasenbusch) two-term action: combined force
+    matches finite differences."""
+    from quda_amd.gauge.fermion_force import (hasenbusch_action_and_force,
+                                              hasenbusch_refresh)
+    geo, u, _ = setup
+    mu_h = 0.15
+    phi1, phi2 = hasenbusch_refresh(u, geo, KAPPA, mu_h, seed=166)
+    P = random_momentum(geo, seed=167)
+    eps = 1e-6
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    Sp, _ = hasenbusch_action_and_force(up, geo, KAPPA, mu_h, phi1, phi2,
+                                        cg_tol=1e-12)
+    Sm, _ = hasenbusch_action_and_force(um, geo, KAPPA, mu_h, phi1, phi2,
+                                        cg_tol=1e-12)
+    dSdt = (Sp - Sm) / (2 * eps)
+    _, F = hasenbusch_action_and_force(u, geo, KAPPA, mu_h, phi1, phi2,
+                                       cg_tol=1e-12)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
+
+
+def test_hasenbusch_hmc_energy_conservation(setup):
+    """Leapfrog with gauge + both Hasenbusch terms conserves H at
+    O(dt^2)."""
+    from quda_amd.gauge import gauge_action, gauge_force, leapfrog
+    from quda_amd.gauge.fermion_force import (hasenbusch_action_and_force,
+                                              hasenbusch_refresh)
+    from quda_amd.gauge.hmc import _evolve_u, mom_action
+    geo, u, _ = setup
+    mu_h = 0.2
+    phi1, phi2 = hasenbusch_refresh(u, geo, KAPPA, mu_h, seed=168)
+
+    def total_force(uc):
+        _, Ff = hasenbusch_action_and_force(uc, geo, KAPPA, mu_h, phi1,
+                                            phi2, cg_tol=1e-11)
+        return gauge_force(uc, geo, BETA) + Ff
+
+    def H(uc, Pc):
+        Sf, _ = hasenbusch_action_and_force(uc, geo, KAPPA, mu_h, phi1,
+                                            phi2, cg_tol=1e-11)
+        return mom_action(Pc) + gauge_action(uc, geo, BETA) + Sf
+
+    P0 = random_momentum(geo, seed=169)
+    dHs = []
+    for n in (8, 16):
+        dt = 0.25 / n
+        uu, PP = u.clone(), P0 + 0.5 * dt * total_force(u)
+        for k in range(n):
+            uu = _evolve_u(uu, PP, geo, dt)
+            PP = PP + (0.5 if k == n - 1 else 1.0) * dt * total_force(uu)
+        dHs.append(abs(H(uu, PP) - H(u, P0)))
+    assert dHs[1] < dHs[0]
+    assert dHs[1] < 1.0, dHs
