@@ -224,3 +224,65 @@ def test_hasenbusch_hmc_energy_conservation(setup):
         dHs.append(abs(H(uu, PP) - H(u, P0)))
     assert dHs[1] < dHs[0]
     assert dHs[1] < 1.0, dHs
+
+
+def test_rhmc_force_finite_difference():
+    """Rational-HMC (rooted staggered) force from one multishift solve +
+    adjoint surrogate matches finite differences."""
+    from quda_amd.gauge.fermion_force import rhmc_force, rhmc_refresh
+    from quda_amd.solvers import rational_approx
+    geo = LatticeGeometry((4, 4, 4, 4))
+    u = GaugeField(geo, "double").random_su3_(seed=81).to_complex()
+    mass = 0.1
+    lo, hi = 4 * mass * mass, 30.0
+    r_inv = rational_approx(-0.25, lo, hi, n=10)
+    r_half = rational_approx(0.125, lo, hi, n=10)
+    phi = rhmc_refresh(u, geo, mass, r_half, seed=33)
+    P = random_momentum(geo, seed=34)
+    eps = 1e-6
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    Sp, _ = rhmc_force(up, geo, mass, phi, r_inv, cg_tol=1e-12)
+    Sm, _ = rhmc_force(um, geo, mass, phi, r_inv, cg_tol=1e-12)
+    dSdt = (Sp - Sm) / (2 * eps)
+    _, F = rhmc_force(u, geo, mass, phi, r_inv, cg_tol=1e-12)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    assert abs(-2 * trPF + dSdt) < 1e-5 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
+
+
+def test_rhmc_energy_conservation():
+    """One-flavor-rooted staggered RHMC leapfrog: dH shrinks O(dt^2)."""
+    from quda_amd.gauge import gauge_action, gauge_force
+    from quda_amd.gauge.fermion_force import rhmc_force, rhmc_refresh
+    from quda_amd.gauge.hmc import _evolve_u, mom_action
+    from quda_amd.solvers import rational_approx
+    geo = LatticeGeometry((4, 4, 4, 4))
+    u = GaugeField(geo, "double").random_su3_(seed=82).to_complex()
+    mass, beta = 0.1, 5.5
+    lo, hi = 4 * mass * mass, 30.0
+    r_inv = rational_approx(-0.25, lo, hi, n=10)
+    r_half = rational_approx(0.125, lo, hi, n=10)
+    phi = rhmc_refresh(u, geo, mass, r_half, seed=35)
+
+    def force(uc):
+        _, Ff = rhmc_force(uc, geo, mass, phi, r_inv, cg_tol=1e-11)
+        return gauge_force(uc, geo, beta) + Ff
+
+    def H(uc, Pc):
+        Sf, _ = rhmc_force(uc, geo, mass, phi, r_inv, cg_tol=1e-11)
+        return mom_action(Pc) + gauge_action(uc, geo, beta) + Sf
+
+    P0 = random_momentum(geo, seed=36)
+    dHs = []
+    for n in (8, 16):
+        dt = 0.2 / n
+        uu, PP = u.clone(), P0 + 0.5 * dt * force(u)
+        for k in range(n):
+            uu = _evolve_u(uu, PP, geo, dt)
+            PP = PP + (0.5 if k == n - 1 else 1.0) * dt * force(uu)
+        dHs.append(abs(H(uu, PP) - H(u, P0)))
+    assert dHs[1] < dHs[0]
+    assert dHs[1] < 1.0, dHs
