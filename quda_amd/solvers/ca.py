@@ -26,12 +26,15 @@ def _gram(vs):
 
 
 def ca_cg_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
-                maxiter: int = 1000, basis_size: int = 4) -> SolverStats:
-    """CA-CG on MdagM: per cycle build the power basis
-    V = [r, A r, ..., A^{s-1} r], Galerkin-project
+                maxiter: int = 1000, basis_size: int = 4,
+                basis: str = "power", lambda_min: float = 0.0,
+                lambda_max: float = 10.0) -> SolverStats:
+    """CA-CG on MdagM: per cycle build an s-step basis, Galerkin-project
     (V^dag A V) c = V^dag r, and update x += V c. One batched reduction
-    set per s matrix applications (power basis; the reference also offers
-    Chebyshev, QudaCABasis enum_quda.h:207)."""
+    set per s matrix applications. basis = "power"
+    (V = [r, Ar, ..., A^{s-1} r]) or "chebyshev" (QudaCABasis
+    enum_quda.h:207 — shifted Chebyshev T_k((A-c)/h) r, numerically
+    stable at larger s; supply spectrum bounds lambda_min/max)."""
     stats = SolverStats()
     b2 = blas.norm2(b)
     if b2 == 0.0:
@@ -40,12 +43,16 @@ def ca_cg_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
         return stats
     stop = tol * tol * b2
     s = basis_size
+    cheb = basis == "chebyshev"
+    cc = 0.5 * (lambda_max + lambda_min)
+    h = 0.5 * (lambda_max - lambda_min)
 
     def new():
         return SpinorField(x.geo, x.precision, x.device, x.n_parity, nspin=x.nspin, ls=x.ls)
 
     r, tmp = new(), new()
-    V = [new() for _ in range(s + 1)]  # V[s] = A V[s-1] completes AV
+    V = [new() for _ in range(s + 1)]
+    W = [new() for _ in range(s)]  # W[j] = A V[j]
     x2 = blas.norm2(x)
     if x2 > 0.0:
         op.MdagM(r, x, tmp)
@@ -57,14 +64,23 @@ def ca_cg_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
     while r2 > stop and k < maxiter:
         blas.copy(V[0], r)
         for j in range(s):
-            op.MdagM(V[j + 1], V[j], tmp)
+            op.MdagM(W[j], V[j], tmp)
             k += 1
-        # G = V^dag (A V) using AV_j = V[j+1]; rhs_j = <V_j, r>
+            if not cheb:
+                blas.copy(V[j + 1], W[j])
+            else:
+                # V[j+1] = (2 - [j==0]) * (W[j] - c V[j])/h - V[j-1]
+                fac = (1.0 if j == 0 else 2.0) / h
+                blas.copy(V[j + 1], W[j])
+                blas.axpy(-cc, V[j], V[j + 1])
+                blas.scal(fac, V[j + 1])
+                if j > 0:
+                    blas.axpy(-1.0, V[j - 1], V[j + 1])
         G = np.empty((s, s), dtype=complex)
         rhs = np.empty(s, dtype=complex)
         for i in range(s):
             for j in range(s):
-                G[i, j] = blas.c_dot(V[i], V[j + 1])
+                G[i, j] = blas.c_dot(V[i], W[j])
             rhs[i] = blas.c_dot(V[i], r)
         try:
             c = np.linalg.solve(G, rhs)
@@ -72,7 +88,7 @@ def ca_cg_solve(op, x: SpinorField, b: SpinorField, *, tol: float = 1e-8,
             break
         for j in range(s):
             blas.caxpy(complex(c[j]), V[j], x)
-            blas.caxpy(-complex(c[j]), V[j + 1], r)
+            blas.caxpy(-complex(c[j]), W[j], r)
         r2 = blas.norm2(r)
 
     stats.iters = k

@@ -110,3 +110,13 @@ def tune_dslash(example_call: Callable[[], None], key: str) -> str:
         ext.set_dslash_block(int(cfg))
 
     return get_tuner().tune(key, ["64", "128", "256"], setup, example_call)
+
+
+def tune_dslash_policy(example_call, key: str) -> str:
+    """Autotune the comm-overlap policy (overlap vs fused) for a
+    representative distributed dslash (ref: the policy-level autotuning
+    of lib/dslash_policy.hpp, profile_async tsv)."""
+    from ..ops.dispatch import set_dslash_policy
+
+    return get_tuner().tune(key, ["overlap", "fused"],
+                            lambda c: set_dslash_policy(c), example_call)

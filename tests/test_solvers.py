@@ -476,3 +476,39 @@ def test_mdagm_local_equals_global_single_rank():
     d.MdagM(o1, a, t)
     loc.MdagM(o2, a, t)
     assert (o1.to_complex() - o2.to_complex()).abs().max().item() < 1e-13
+
+
+def test_ca_cg_chebyshev_basis():
+    """Chebyshev s-step basis converges where the power basis at the
+    same s is numerically fragile, and matches the CG solution."""
+    from quda_amd.models import DiracWilson
+    from quda_amd.solvers.ca import ca_cg_solve
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=631)
+    d = DiracWilson(g, 0.12)
+    b = SpinorField(geo, "double").gaussian_(seed=632)
+    x0 = SpinorField(geo, "double")
+    st0 = cg_solve(d, x0, b, tol=1e-9, maxiter=500)
+    x1 = SpinorField(geo, "double")
+    st1 = ca_cg_solve(d, x1, b, tol=1e-9, maxiter=500, basis_size=8,
+                      basis="chebyshev", lambda_min=0.05, lambda_max=6.0)
+    assert st0.converged and st1.converged
+    err = (x1.to_complex() - x0.to_complex()).abs().max().item()
+    assert err < 1e-6, err
+
+
+def test_tune_dslash_policy_caches():
+    from quda_amd.utils.tune import Tuner
+    import quda_amd.utils.tune as tn
+    t = Tuner()
+    old = tn._TUNER
+    tn._TUNER = t
+    try:
+        from quda_amd.utils.tune import tune_dslash_policy
+        picks = []
+        from quda_amd.ops.dispatch import dslash_policy
+        tune_dslash_policy(lambda: picks.append(dslash_policy()),
+                           "policy-test-key")
+        assert t.cache["policy-test-key"][0] in ("overlap", "fused")
+    finally:
+        tn._TUNER = old
