@@ -415,10 +415,12 @@ def improved_gauge_force(u: torch.Tensor, geo: LatticeGeometry,
     return F
 
 
-def _action_autograd(z, geo, beta, c1):
+def _action_autograd(z, geo, beta, c1, c0=None):
     """improved_gauge_action as a differentiable torch scalar (no
-    .item() calls)."""
-    c0 = 1.0 - 8.0 * c1
+    .item() calls). c0 defaults to the Luscher-Weisz normalization
+    1 - 8 c1; pass it explicitly for over-improved smearing weights."""
+    if c0 is None:
+        c0 = 1.0 - 8.0 * c1
     from ..parallel.halo import shift_lex
     s_pl = 0.0
     s_rt = 0.0
@@ -520,4 +522,37 @@ def hyp_smear(u: torch.Tensor, geo: LatticeGeometry,
                 S = S + staple_dir(Vtil[(mu, nu)], Vtil[(nu, mu)], mu, nu)
             Unew[mu] = project_su3((1 - a1) * U[mu] + (a1 / 6.0) * S)
         out = _from_lex(Unew, geo)
+    return out
+
+
+def over_improved_stout_smear(u: torch.Tensor, geo: LatticeGeometry,
+                              rho: float, n_iter: int = 1,
+                              epsilon: float = -0.25) -> torch.Tensor:
+    """Over-improved stout smearing (Moran-Leinweber; ref: the
+    over-improved branch of lib/gauge_stout.cu): the smearing direction
+    is the traceless-antihermitian derivative of the action with
+    plaquette weight (5 - 2 eps)/3 and rectangle weight -(1 - eps)/12;
+    eps = 1 reduces EXACTLY to plain stout (rectangles drop out), the
+    default eps = -0.25 is tuned to preserve instantons. Computed by
+    reverse-mode differentiation of the loop traces (single-rank)."""
+    from ..parallel import comms
+    assert comms.comm_size() == 1, "autograd smear direction: single-rank"
+    c0 = (5.0 - 2.0 * epsilon) / 3.0
+    c1 = -(1.0 - epsilon) / 12.0
+    out = u
+    for _ in range(n_iter):
+        u_req = out.detach().clone().requires_grad_(True)
+        s = _action_autograd(u_req, geo, 1.0, c1, c0=c0)
+        s.backward()
+        g = u_req.grad
+        Q = torch.empty_like(out)
+        for mu in range(4):
+            for p in (0, 1):
+                # Q_stout = TA[S U^d] = +3 TA[U g^d] (from the validated
+                # force relation F = 1/2 TA[U g^d] = -(1/6) TA[U S^d])
+                Q[mu, p] = 3.0 * project_ta(out[mu, p]
+                                            @ g[mu, p].conj().mT)
+        U = _to_lex(out, geo)
+        Ql = _to_lex(Q, geo)
+        out = _from_lex(exp_su3(Ql, rho) @ U, geo)
     return out

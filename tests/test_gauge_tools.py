@@ -319,3 +319,20 @@ def test_det_trace(geo, u_rand):
     d, t = det_trace(u_rand, geo)
     assert abs(d - 1.0) < 1e-10  # SU(3)
     assert abs(t) < 1.0
+
+
+def test_over_improved_stout(geo, u_rand):
+    """eps=1 reduces EXACTLY to plain stout; the default eps raises the
+    plaquette and keeps links unitary."""
+    import torch
+    from quda_amd.gauge import plaquette
+    from quda_amd.gauge.ops import over_improved_stout_smear, stout_smear
+    a = stout_smear(u_rand, geo, 0.1, 1)
+    b = over_improved_stout_smear(u_rand, geo, 0.1, 1, epsilon=1.0)
+    assert (a - b).abs().max().item() < 1e-12
+    c = over_improved_stout_smear(u_rand, geo, 0.1, 2)
+    p0, _, _ = plaquette(u_rand, geo)
+    pc, _, _ = plaquette(c, geo)
+    assert pc > p0
+    det = torch.linalg.det(c.reshape(-1, 3, 3))
+    assert (det.abs() - 1).abs().max().item() < 1e-10
