@@ -512,3 +512,18 @@ def test_tune_dslash_policy_caches():
         assert t.cache["policy-test-key"][0] in ("overlap", "fused")
     finally:
         tn._TUNER = old
+
+
+def test_heavy_quark_residual_criterion():
+    from quda_amd.models import DiracWilson
+    from quda_amd.solvers.cg import cg_solve, hq_residual
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g = GaugeField(geo, "double").random_su3_(seed=661)
+    d = DiracWilson(g, 0.12)
+    b = SpinorField(geo, "double").gaussian_(seed=662)
+    bp = SpinorField(geo, "double")
+    d.M(bp, b, dagger=True)
+    x = SpinorField(geo, "double")
+    st = cg_solve(d, x, bp, tol=1e-6, maxiter=800, hq_tol=1e-7)
+    assert st.converged
+    assert 0 < st.hq_resid <= 1e-7, st.hq_resid
