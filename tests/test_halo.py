@@ -638,3 +638,73 @@ def test_dslash_odd_shapes_self_wrap(dims):
         comms.set_forced_partition(0)
     err = (out.to_complex() - out_ref.to_complex()).abs().max().item()
     assert err < 1e-13, err
+
+
+def _worker_mobius(rank, world, init_file):
+    import torch.distributed as dist
+    from quda_amd.models.dwf import DiracMobiusPC
+    from quda_amd.solvers import cgnr_solve
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=(1, 1, 1, world))
+        gg, u_lex, _ = _global_fields(seed=67)
+        from quda_amd.fields.geometry import checkerboard_join, checkerboard_split
+        lg, u_loc_lex = _local_slice(gg, (1, 1, 1, world),
+                                     comms.grid_coords(), u_lex.movedim(0, 1))
+        u_loc = checkerboard_split(u_loc_lex, lg).permute(
+            2, 0, 1, 3, 4).contiguous()
+        g = GaugeField(lg, "double").from_complex(u_loc)
+        LS, MF, M5 = 4, 0.04, 1.8
+        pc = DiracMobiusPC(g, MF, M5, LS, b5=1.5, c5=0.5)
+        # deterministic global 5-d source on EVEN sites: build per slice
+        gen = torch.Generator().manual_seed(680)
+        src_slices = [torch.view_as_complex(
+            torch.randn((gg.volume, 4, 3, 2), generator=gen,
+                        dtype=torch.float64)) for _ in range(LS)]
+        mycoords = comms.grid_coords()
+
+        def local_5d(slices, geo_l, coords):
+            loc = []
+            for sl in slices:
+                _, l = _local_slice(gg, (1, 1, 1, world), coords, sl)
+                loc.append(checkerboard_split(l, geo_l)[0])  # even cb
+            return torch.cat(loc, dim=0)  # [Ls*Vcb, 4, 3]
+
+        b = SpinorField(lg, "double", n_parity=1, ls=LS)
+        b.from_complex(local_5d(src_slices, lg, mycoords).unsqueeze(0))
+        x = SpinorField(lg, "double", n_parity=1, ls=LS)
+        st = cgnr_solve(pc, x, b, tol=1e-10, maxiter=3000)
+        assert st.converged, f"rank{rank}: {st}"
+
+        # global truth in one solo solve
+        with comms.solo_mode():
+            gglob = GaugeField(gg, "double").from_complex(
+                checkerboard_split(u_lex.movedim(0, 1), gg).permute(
+                    2, 0, 1, 3, 4).contiguous())
+            pcg = DiracMobiusPC(gglob, MF, M5, LS, b5=1.5, c5=0.5)
+            bg = SpinorField(gg, "double", n_parity=1, ls=LS)
+            bg_c = torch.cat([checkerboard_split(sl, gg)[0]
+                              for sl in src_slices], dim=0)
+            bg.from_complex(bg_c.unsqueeze(0))
+            xg = SpinorField(gg, "double", n_parity=1, ls=LS)
+            stg = cgnr_solve(pcg, xg, bg, tol=1e-10, maxiter=3000)
+            assert stg.converged
+        # compare my slab: global even-cb per slice -> lex -> local slice
+        Vg = gg.volume_cb
+        xg_c = xg.to_complex()[0].reshape(LS, Vg, 4, 3)
+        xloc = x.to_complex()[0].reshape(LS, lg.volume_cb, 4, 3)
+        for s in range(LS):
+            lex = torch.zeros((gg.volume, 4, 3), dtype=torch.complex128)
+            lex[gg.lex_of_cb[0]] = xg_c[s]
+            _, loc_lex = _local_slice(gg, (1, 1, 1, world), mycoords, lex)
+            want = checkerboard_split(loc_lex, lg)[0]
+            err = (xloc[s] - want).abs().max().item()
+            assert err < 1e-7, f"rank{rank} s{s} err={err}"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_mobius_multiproc_gloo(tmp_path):
+    init_file = str(tmp_path / "init_mob")
+    mp.spawn(_worker_mobius, args=(2, init_file), nprocs=2, join=True)
