@@ -708,3 +708,50 @@ def _worker_mobius(rank, world, init_file):
 def test_mobius_multiproc_gloo(tmp_path):
     init_file = str(tmp_path / "init_mob")
     mp.spawn(_worker_mobius, args=(2, init_file), nprocs=2, join=True)
+
+
+def _worker_twisted(rank, world, init_file):
+    import torch.distributed as dist
+    from quda_amd.models import DiracTwistedMass
+    from quda_amd.solvers import cgnr_solve
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=(1, 1, 1, world))
+        gg, u_lex, src_lex = _global_fields(seed=69)
+        from quda_amd.fields.geometry import checkerboard_join, checkerboard_split
+        lg, u_loc_lex = _local_slice(gg, (1, 1, 1, world),
+                                     comms.grid_coords(), u_lex.movedim(0, 1))
+        u_loc = checkerboard_split(u_loc_lex, lg).permute(
+            2, 0, 1, 3, 4).contiguous()
+        g = GaugeField(lg, "double").from_complex(u_loc)
+        mycoords = comms.grid_coords()
+        _, s_loc_lex = _local_slice(gg, (1, 1, 1, world), mycoords, src_lex)
+        b = SpinorField(lg, "double")
+        b.from_complex(checkerboard_split(s_loc_lex, lg))
+        d = DiracTwistedMass(g, 0.12, 0.3)
+        x = SpinorField(lg, "double")
+        st = cgnr_solve(d, x, b, tol=1e-10, maxiter=1000)
+        assert st.converged
+        with comms.solo_mode():
+            gglob = GaugeField(gg, "double").from_complex(
+                checkerboard_split(u_lex.movedim(0, 1), gg).permute(
+                    2, 0, 1, 3, 4).contiguous())
+            dg = DiracTwistedMass(gglob, 0.12, 0.3)
+            bg = SpinorField(gg, "double")
+            bg.from_complex(checkerboard_split(src_lex, gg))
+            xg = SpinorField(gg, "double")
+            stg = cgnr_solve(dg, xg, bg, tol=1e-10, maxiter=1000)
+            assert stg.converged
+        xg_lex = checkerboard_join(xg.to_complex(), gg)
+        _, xl = _local_slice(gg, (1, 1, 1, world), mycoords, xg_lex)
+        want = checkerboard_split(xl, lg)
+        err = (x.to_complex() - want).abs().max().item()
+        assert err < 1e-7, f"rank{rank}: {err}"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_twisted_mass_multiproc_gloo(tmp_path):
+    init_file = str(tmp_path / "init_tm")
+    mp.spawn(_worker_twisted, args=(2, init_file), nprocs=2, join=True)
