@@ -53,3 +53,36 @@ def load_gauge(path: str, device="cpu") -> torch.Tensor:
     if field_checksum(blob["u"]) != blob["checksum"]:
         raise IOError(f"checksum mismatch in {path}")
     return blob["u"].to(device)
+
+
+def save_gauge(path: str, u: torch.Tensor, geo: LatticeGeometry,
+               meta: dict = None) -> None:
+    """Checksummed gauge-configuration storage ([4,2,Vcb,3,3] complex;
+    the gauge-side VectorIO/QIO role). Stores plaquette in the header
+    for load-time integrity beyond the byte checksum."""
+    from ..gauge import plaquette
+    tot, sp, tm = plaquette(u, geo)
+    blob = {
+        "u": u.cpu(),
+        "dims": geo.dims,
+        "checksum": field_checksum(u.cpu()),
+        "plaquette": (tot, sp, tm),
+        "meta": meta or {},
+    }
+    torch.save(blob, path)
+
+
+def load_gauge(path: str, device="cpu", check_plaquette: bool = True):
+    """-> (u [4,2,Vcb,3,3] complex on device, geo, meta). Raises on
+    checksum or plaquette mismatch."""
+    blob = torch.load(path, weights_only=False)
+    u = blob["u"]
+    if field_checksum(u) != blob["checksum"]:
+        raise IOError(f"gauge checksum mismatch in {path}")
+    geo = LatticeGeometry(blob["dims"])
+    if check_plaquette:
+        from ..gauge import plaquette
+        tot, _, _ = plaquette(u, geo)
+        if abs(tot - blob["plaquette"][0]) > 1e-10:
+            raise IOError(f"gauge plaquette mismatch in {path}")
+    return u.to(device), geo, blob["meta"]

@@ -123,3 +123,23 @@ def test_tuner_profile_dump(tmp_path):
     t.profile_dump(p)
     lines = open(p).read().splitlines()
     assert any(l.startswith("k1\t2\t") for l in lines), lines
+
+
+def test_gauge_io_roundtrip(tmp_path):
+    import torch
+    from quda_amd import GaugeField, LatticeGeometry
+    from quda_amd.utils.io import load_gauge, save_gauge
+    geo = LatticeGeometry((4, 4, 4, 4))
+    u = GaugeField(geo, "double").random_su3_(seed=671).to_complex()
+    p = str(tmp_path / "conf.pt")
+    save_gauge(p, u, geo, meta={"beta": 6.0})
+    u2, geo2, meta = load_gauge(p)
+    assert (u2 - u).abs().max().item() == 0.0
+    assert geo2.dims == geo.dims and meta["beta"] == 6.0
+    # corruption detection
+    blob = torch.load(p, weights_only=False)
+    blob["u"][0, 0, 0, 0, 0] += 1.0
+    torch.save(blob, p)
+    import pytest as _pt
+    with _pt.raises(IOError):
+        load_gauge(p)
