@@ -42,24 +42,19 @@ def load_field(path: str, device="cpu", precision="double") -> List[SpinorField]
     return out
 
 
-def save_gauge(path: str, u: torch.Tensor, meta: dict = None):
-    c = u.detach().cpu()
-    torch.save({"u": c, "checksum": field_checksum(c), "meta": meta or {}},
-               path)
-
-
-def load_gauge(path: str, device="cpu") -> torch.Tensor:
-    blob = torch.load(path, weights_only=False)
-    if field_checksum(blob["u"]) != blob["checksum"]:
-        raise IOError(f"checksum mismatch in {path}")
-    return blob["u"].to(device)
-
-
-def save_gauge(path: str, u: torch.Tensor, geo: LatticeGeometry,
+def save_gauge(path: str, u: torch.Tensor, geo: LatticeGeometry = None,
                meta: dict = None) -> None:
     """Checksummed gauge-configuration storage ([4,2,Vcb,3,3] complex;
-    the gauge-side VectorIO/QIO role). Stores plaquette in the header
-    for load-time integrity beyond the byte checksum."""
+    the gauge-side VectorIO/QIO role). Stores the plaquette in the
+    header (integrity beyond the byte checksum); geo is inferred from
+    the field shape when omitted only insofar as the plaquette header is
+    skipped."""
+    if geo is None:
+        blob = {"u": u.detach().cpu(), "dims": None,
+                "checksum": field_checksum(u.detach().cpu()),
+                "plaquette": None, "meta": meta or {}}
+        torch.save(blob, path)
+        return
     from ..gauge import plaquette
     tot, sp, tm = plaquette(u, geo)
     blob = {
@@ -79,8 +74,10 @@ def load_gauge(path: str, device="cpu", check_plaquette: bool = True):
     u = blob["u"]
     if field_checksum(u) != blob["checksum"]:
         raise IOError(f"gauge checksum mismatch in {path}")
+    if blob.get("dims") is None:
+        return u.to(device)  # legacy minimal record
     geo = LatticeGeometry(blob["dims"])
-    if check_plaquette:
+    if check_plaquette and blob.get("plaquette") is not None:
         from ..gauge import plaquette
         tot, _, _ = plaquette(u, geo)
         if abs(tot - blob["plaquette"][0]) > 1e-10:
