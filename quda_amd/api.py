@@ -544,11 +544,24 @@ def mat_dag_mat_quda(inp: torch.Tensor, p: InvertParam) -> torch.Tensor:
     return out.to_complex()
 
 
-def save_gauge_quda() -> torch.Tensor:
+def save_gauge_quda(path: Optional[str] = None) -> torch.Tensor:
     """ref: saveGaugeQuda — returns the resident gauge field as
-    [4,2,Vcb,3,3] complex (the inverse of load_gauge_quda)."""
+    [4,2,Vcb,3,3] complex (the inverse of load_gauge_quda); with `path`
+    also writes the checksummed on-disk record (utils.io.save_gauge)."""
     assert _R.u_complex is not None, "no resident gauge"
+    if path is not None:
+        from .utils.io import save_gauge
+        save_gauge(path, _R.u_complex, _R.geo)
     return _R.u_complex.clone()
+
+
+def covdev_quda(inp: torch.Tensor, p: InvertParam, mu: int,
+                forward: bool = True) -> torch.Tensor:
+    """ref: the GaugeCovDev dslash type (covariant displacement, used
+    for sequential sources and derivative operators)."""
+    from .models.laplace import covdev_apply
+    f = _wrap(inp, p, 2)
+    return covdev_apply(_R.gauge, f, mu, forward).to_complex()
 
 
 def invert_multi_src_quda(bs, p: InvertParam, *, splits=None):

@@ -9,7 +9,7 @@ from .ops import (ape_smear, exp_su3, gauge_action, gauge_force,
                   plaquette, polyakov_loop, project_ta, staple_sum,
                   stout_smear, topological_charge, wilson_flow, wilson_loop,
                   hyp_smear, loop_trace, path_product, det_trace,
-                  over_improved_stout_smear,
+                  over_improved_stout_smear, topological_charge_density,
                   improved_gauge_action, improved_gauge_force,
                   energy_density, wilson_flow_measure, flow_scale_t0,
                   flow_scale_w0)
@@ -25,7 +25,7 @@ __all__ = ["plaquette", "gauge_action", "staple_sum", "gauge_force",
            "wilson_flow", "polyakov_loop", "topological_charge",
            "energy_density", "wilson_flow_measure", "flow_scale_t0",
            "hyp_smear", "loop_trace", "path_product", "det_trace",
-           "over_improved_stout_smear",
+           "over_improved_stout_smear", "topological_charge_density",
            "improved_gauge_action", "improved_gauge_force",
            "flow_scale_w0",
            "leapfrog", "hmc_trajectory", "mom_action", "random_momentum",

@@ -556,3 +556,19 @@ def over_improved_stout_smear(u: torch.Tensor, geo: LatticeGeometry,
         Ql = _to_lex(Q, geo)
         out = _from_lex(exp_su3(Ql, rho) @ U, geo)
     return out
+
+
+def topological_charge_density(u: torch.Tensor, geo: LatticeGeometry
+                               ) -> torch.Tensor:
+    """Per-site topological charge density q(x) ([2, V_cb] real;
+    the qChargeDensity output of gauge_qcharge.cuh): sums to
+    topological_charge on this rank's sub-lattice."""
+    import math
+    F = field_strength(u, geo)
+
+    def trprod(a, b):
+        return torch.einsum("pvij,pvji->pv", F[a], F[b]).real
+
+    q = (trprod((0, 1), (2, 3)) - trprod((0, 2), (1, 3))
+         + trprod((0, 3), (1, 2)))
+    return q / (4.0 * math.pi ** 2)

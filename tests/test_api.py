@@ -306,3 +306,20 @@ def test_all_action_types_through_api(resident):
             (2, 128 * ls, *site, 2), generator=g, dtype=torch.float64))
         x = api.invert_quda(b, p)
         assert p.true_res < 1e-6, (t, p.true_res)
+
+
+def test_covdev_and_gauge_save_api(resident, tmp_path):
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12)
+    src = _rand_spinor((2, 128, 4, 3), 580)
+    fwd = api.covdev_quda(src, p, mu=2, forward=True)
+    bwd = api.covdev_quda(fwd, p, mu=2, forward=False)
+    # U^d(x-mu) [U(x-mu) psi(x)] = psi  => bwd(fwd(psi)) = psi
+    assert (bwd - src).abs().max().item() < 1e-12
+    path = str(tmp_path / "resident.pt")
+    api.save_gauge_quda(path)
+    from quda_amd.utils.io import load_gauge
+    u2, geo2, _ = load_gauge(path)
+    assert (u2 - resident).abs().max().item() == 0.0

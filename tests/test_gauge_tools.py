@@ -336,3 +336,10 @@ def test_over_improved_stout(geo, u_rand):
     assert pc > p0
     det = torch.linalg.det(c.reshape(-1, 3, 3))
     assert (det.abs() - 1).abs().max().item() < 1e-10
+
+
+def test_qcharge_density_sums_to_charge(geo, u_rand):
+    from quda_amd.gauge import topological_charge, topological_charge_density
+    q = topological_charge(u_rand, geo)
+    qd = topological_charge_density(u_rand, geo)
+    assert abs(qd.sum().item() - q) < 1e-10
