@@ -158,3 +158,25 @@ def dilute(src: SpinorField, scheme: str = "spin") -> List[SpinorField]:
     else:
         raise ValueError(scheme)
     return out
+
+
+def sequential_source(prop_col: SpinorField, t_sink: int,
+                      gamma: str = "g5") -> SpinorField:
+    """Sequential source for 3-point functions (ref: the sequential-
+    propagator workflow the reference's covdev/contract kernels serve):
+    restrict one propagator column to the sink timeslice and hit it with
+    the sink gamma (g5 for the pion)."""
+    import torch
+    geo = prop_col.geo
+    c = prop_col.to_complex().clone()
+    tcoord = geo.coords[:, 3].to(torch.int64)
+    tc = torch.stack([tcoord[geo.lex_of_cb[0]], tcoord[geo.lex_of_cb[1]]])
+    mask = (tc == t_sink).unsqueeze(-1).unsqueeze(-1)
+    c = torch.where(mask, c, torch.zeros_like(c))
+    if gamma == "g5":
+        c[..., 2:4, :] = -c[..., 2:4, :]
+    else:
+        raise ValueError(gamma)
+    out = prop_col.clone_empty()
+    out.from_complex(c)
+    return out
