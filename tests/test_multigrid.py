@@ -227,3 +227,39 @@ def test_staggered_null_vectors_are_near_null():
         w = SpinorField(geo, "double", nspin=1)
         d.M(w, v)
         assert math.sqrt(blas.norm2(w)) < 0.1  # ||v|| = 1
+
+
+def test_mg_survives_gauge_evolution():
+    """updateMultigridQuda role across real HMC evolution: evolve the
+    resident gauge a few MD steps, refresh the hierarchy, and the
+    MG-preconditioned solve still converges on the NEW links
+    (multigrid_evolve_test analogue)."""
+    from quda_amd import api
+    from quda_amd.api import (DslashType, GaugeParam, InvertParam,
+                              InverterType)
+    from quda_amd.fields.gauge import GaugeField
+    from quda_amd.fields.geometry import LatticeGeometry
+    import torch
+    geo = LatticeGeometry((4, 4, 4, 4))
+    u = GaugeField(geo, "double").random_su3_(seed=695).to_complex()
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.init_quda()
+    api.load_gauge_quda(u, gp)
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12,
+                    inv_type=InverterType.GCR, tol=1e-8, maxiter=400)
+    mg = api.new_multigrid_quda(p, block=(2, 2, 2, 2), n_vec=4)
+    # evolve: a few leapfrog steps on the resident field
+    P = api.gauss_mom_quda(seed=696)
+    for _ in range(3):
+        F = api.compute_gauge_force_quda(5.5)
+        P = P + 0.05 * F
+        api.update_gauge_field_quda(P, 0.05)
+    api.update_multigrid_quda(mg, p)
+    p2 = InvertParam(**{**p.__dict__, "preconditioner": mg.precond})
+    g = torch.Generator().manual_seed(697)
+    b = torch.view_as_complex(torch.randn((2, 128, 4, 3, 2), generator=g,
+                                          dtype=torch.float64))
+    api.invert_quda(b, p2)
+    assert p2.true_res < 1e-7, p2.true_res
+    api.end_quda()
