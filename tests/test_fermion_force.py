@@ -286,3 +286,36 @@ def test_rhmc_energy_conservation():
         dHs.append(abs(H(uu, PP) - H(u, P0)))
     assert dHs[1] < dHs[0]
     assert dHs[1] < 1.0, dHs
+
+
+def test_hisq_full_force_through_unitarization():
+    """HISQ force with the SVD unitarization inside the differentiated
+    chain (unitarize_force role) matches finite differences."""
+    from quda_amd.gauge.hisq import hisq_full_force, hisq_two_level_links
+    from quda_amd.models import DiracImprovedStaggered
+    from quda_amd.fields.gauge import GaugeField
+    from quda_amd import SpinorField
+    geo = LatticeGeometry((4, 4, 4, 4))
+    u = GaugeField(geo, "double").random_su3_(seed=83).to_complex()
+    mass = 0.2
+    # pseudofermion refresh: phi = M eta on the two-level links
+    fat, lng = hisq_two_level_links(u, geo)
+    gf = GaugeField(geo, "double").from_complex(fat)
+    gl = GaugeField(geo, "double", shift=3).from_complex(lng)
+    d = DiracImprovedStaggered(gf, gl, mass)
+    eta = SpinorField(geo, "double", nspin=1).gaussian_(seed=84)
+    phi = SpinorField(geo, "double", nspin=1)
+    d.M(phi, eta)
+    P = random_momentum(geo, seed=85)
+    eps = 1e-6
+    U = _to_lex(u, geo)
+    Pl = _to_lex(P, geo)
+    up = _from_lex(torch.matrix_exp(eps * Pl) @ U, geo)
+    um = _from_lex(torch.matrix_exp(-eps * Pl) @ U, geo)
+    Sp, _ = hisq_full_force(up, geo, mass, phi, cg_tol=1e-12)
+    Sm, _ = hisq_full_force(um, geo, mass, phi, cg_tol=1e-12)
+    dSdt = (Sp - Sm) / (2 * eps)
+    _, F = hisq_full_force(u, geo, mass, phi, cg_tol=1e-12)
+    trPF = torch.einsum("dpvij,dpvji->", P, F).real.item()
+    assert abs(-2 * trPF + dSdt) < 1e-4 * max(abs(dSdt), 1.0), \
+        (dSdt, -2 * trPF)
