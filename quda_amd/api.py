@@ -81,6 +81,7 @@ class GaugeParam:
     reconstruct: str = "none"
     reconstruct_sloppy: str = "twelve"
     anisotropy: float = 1.0
+    t_boundary: str = "periodic"  # or "anti" (ref: QudaTboundary)
     device: str = "cuda" if torch.cuda.is_available() else "cpu"
 
 
@@ -179,6 +180,21 @@ def load_gauge_quda(u: torch.Tensor, param: GaugeParam) -> None:
     if isinstance(u, GaugeField):
         u = u.to_complex()
     u = u.to(param.device)
+    if param.t_boundary == "anti":
+        # fold the anti-periodic fermion boundary into the links: negate
+        # U_t on the last timeslice (ref: QudaTboundary / the gauge-fix
+        # phase application at load in gauge_field.cpp). Applied on a
+        # copy so the caller's field is untouched; on a partitioned T
+        # grid only the LAST rank's boundary slice carries the phase.
+        from .parallel import comms
+        u = u.clone()
+        gt = comms.grid_dims()[3]
+        ct = comms.grid_coords()[3]
+        if ct == gt - 1:
+            t_hi = geo.dims[3] - 1
+            for p in (0, 1):
+                idx = geo.face_index_cb(p, 3, t_hi).to(u.device)
+                u[3, p, idx] = -u[3, p, idx]
     _R.geo = geo
     _R.gauge_param = param
     _R.u_complex = u
