@@ -180,6 +180,15 @@ def load_gauge_quda(u: torch.Tensor, param: GaugeParam) -> None:
     if isinstance(u, GaugeField):
         u = u.to_complex()
     u = u.to(param.device)
+    if param.anisotropy != 1.0:
+        # fold the bare anisotropy into the SPATIAL links (xi_0
+        # convention: U_i -> U_i / xi; ref QudaGaugeParam.anisotropy).
+        # Scaled links are no longer unitary, so reconstruction must be
+        # off.
+        assert param.reconstruct == "none", \
+            "anisotropy needs reconstruct='none'"
+        u = u.clone()
+        u[0:3] = u[0:3] / param.anisotropy
     if param.t_boundary == "anti":
         # fold the anti-periodic fermion boundary into the links: negate
         # U_t on the last timeslice (ref: QudaTboundary / the gauge-fix

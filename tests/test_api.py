@@ -353,3 +353,19 @@ def test_antiperiodic_t_boundary(resident):
     api.load_gauge_quda(u, gp_p)
     out_per = api.mat_quda(b, p)
     assert (out_a - out_per).abs().max().item() > 1e-3
+
+
+def test_anisotropy_folds_into_spatial_links(resident):
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double", anisotropy=2.5)
+    api.load_gauge_quda(resident, gp)
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12)
+    b = _rand_spinor((2, 128, 4, 3), 595)
+    out_a = api.mat_quda(b, p)
+    u2 = resident.clone()
+    u2[0:3] = u2[0:3] / 2.5
+    gp_p = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                      cuda_prec_sloppy="double")
+    api.load_gauge_quda(u2, gp_p)
+    out_m = api.mat_quda(b, p)
+    assert (out_a - out_m).abs().max().item() < 1e-13

@@ -288,3 +288,63 @@ def test_improved_multiproc_gloo(tmp_path):
     import torch.multiprocessing as mp
     init_file = str(tmp_path / "init_imp")
     mp.spawn(_worker_improved, args=(2, init_file), nprocs=2, join=True)
+
+
+def _worker_improved_2d(rank, world, init_file):
+    import torch.distributed as dist
+    from quda_amd.parallel import comms
+    from quda_amd.fields.geometry import checkerboard_split
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        grid = (1, 1, 2, 2)
+        comms.init_comms(grid=grid)
+        GD = (4, 4, 8, 8)
+        gg = LatticeGeometry(GD)
+        gen = torch.Generator().manual_seed(411)
+        from quda_amd.fields.gauge import project_su3
+        fat_lex = project_su3(torch.view_as_complex(torch.randn(
+            (4, gg.volume, 3, 3, 2), generator=gen, dtype=torch.float64)))
+        lng_lex = project_su3(torch.view_as_complex(torch.randn(
+            (4, gg.volume, 3, 3, 2), generator=gen, dtype=torch.float64)))
+        src_lex = torch.view_as_complex(torch.randn(
+            (gg.volume, 3, 2), generator=gen, dtype=torch.float64))
+        ldims = tuple(GD[i] // grid[i] for i in range(4))
+        lg = LatticeGeometry(ldims)
+        coords = comms.grid_coords()
+        off = torch.tensor([coords[i] * ldims[i] for i in range(4)])
+        c = lg.coords.to(torch.int64) + off
+        X, Y, Z, _ = GD
+        glex = ((c[:, 3] * Z + c[:, 2]) * Y + c[:, 1]) * X + c[:, 0]
+        fat_loc = checkerboard_split(fat_lex.movedim(0, 1)[glex], lg
+                                     ).permute(2, 0, 1, 3, 4).contiguous()
+        lng_loc = checkerboard_split(lng_lex.movedim(0, 1)[glex], lg
+                                     ).permute(2, 0, 1, 3, 4).contiguous()
+        gf = GaugeField(lg, "double").from_complex(fat_loc)
+        gl = GaugeField(lg, "double", shift=3).from_complex(lng_loc)
+        src = SpinorField(lg, "double", nspin=1)
+        src.from_complex(checkerboard_split(src_lex[glex], lg))
+        out = SpinorField(lg, "double", n_parity=1, nspin=1)
+        dslash_staggered(out, src.parity_view(1), gf, 0, long_gauge=gl)
+        fat_g = checkerboard_split(fat_lex.movedim(0, 1), gg).permute(
+            2, 0, 1, 3, 4).contiguous()
+        lng_g = checkerboard_split(lng_lex.movedim(0, 1), gg).permute(
+            2, 0, 1, 3, 4).contiguous()
+        src_g = checkerboard_split(src_lex, gg)
+        truth = (ref.dslash_staggered_parity(fat_g, src_g[1], gg, 0)
+                 + ref.dslash_staggered_naik_parity(lng_g, src_g[1], gg, 0))
+        truth_lex = torch.zeros((gg.volume, 3), dtype=torch.complex128)
+        truth_lex[gg.lex_of_cb[0]] = truth
+        want = checkerboard_split(truth_lex[glex], lg)[0]
+        err = (out.to_complex()[0] - want).abs().max().item()
+        assert err < 1e-12, f"rank{rank} err={err}"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_improved_multiproc_2d_grid_gloo(tmp_path):
+    """4 ranks on a (1,1,2,2) grid: depth-3 Naik halos across TWO
+    partitioned dims simultaneously (corner-adjacent exchange paths)."""
+    import torch.multiprocessing as mp
+    init_file = str(tmp_path / "init_imp2d")
+    mp.spawn(_worker_improved_2d, args=(4, init_file), nprocs=4, join=True)
