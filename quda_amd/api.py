@@ -82,6 +82,7 @@ class GaugeParam:
     reconstruct_sloppy: str = "twelve"
     anisotropy: float = 1.0
     t_boundary: str = "periodic"  # or "anti" (ref: QudaTboundary)
+    staggered_phase_applied: bool = False  # input links carry eta(x)
     device: str = "cuda" if torch.cuda.is_available() else "cpu"
 
 
@@ -114,6 +115,8 @@ class InvertParam:
     cuda_prec_sloppy: str = "half"
     preconditioner: Optional[object] = None  # e.g. MG.precond
     deflation: Optional[object] = None  # Deflation (newDeflationQuda)
+    chrono_index: int = -1      # >=0: use/extend the resident chrono basis
+    chrono_max_dim: int = 8
     # outputs (ref: out-fields of QudaInvertParam)
     iter: int = 0
     true_res: float = 0.0
@@ -180,6 +183,18 @@ def load_gauge_quda(u: torch.Tensor, param: GaugeParam) -> None:
     if isinstance(u, GaugeField):
         u = u.to_complex()
     u = u.to(param.device)
+    if param.staggered_phase_applied:
+        # interop: the caller's links already contain the staggered eta
+        # phases (MILC convention). The stencil kernels generate eta
+        # in-kernel, so STRIP them at load (eta^2 = 1: multiply again)
+        # (ref: QudaStaggeredPhase / applyStaggeredPhase).
+        from .ops.reference import staggered_phases
+        u = u.clone()
+        for par in (0, 1):
+            ph = staggered_phases(geo, par)  # [Vcb, 4]
+            for mu in range(4):
+                u[mu, par] = u[mu, par] * ph[:, mu].to(u.dtype).reshape(
+                    -1, 1, 1)
     if param.anisotropy != 1.0:
         # fold the bare anisotropy into the SPATIAL links (xi_0
         # convention: U_i -> U_i / xi; ref QudaGaugeParam.anisotropy).
@@ -384,12 +399,21 @@ def _run_solver(d, x, b, p: InvertParam, sloppy_pair: bool):
             # deflated initial guess (ref: deflated_invert_test /
             # solver.cpp eig-deflation hookup)
             p.deflation.guess(x, rhs)
+        if p.chrono_index >= 0:
+            # chronological initial-guess projection over past solutions
+            # (ref: inv_param.chrono_make_resident/use_resident_chrono)
+            ch = chrono_forecaster(p.chrono_index, p.chrono_max_dim)
+            ch.forecast(d, x, rhs)
         if sloppy_pair and p.cuda_prec_sloppy != p.cuda_prec:
             ds = _make_dirac(p, sloppy=True)
-            return cg_solve(d, x, rhs, op_sloppy=ds,
-                            sloppy=p.cuda_prec_sloppy, tol=p.tol,
-                            maxiter=p.maxiter, delta=p.reliable_delta)
-        return cg_solve(d, x, rhs, tol=p.tol, maxiter=p.maxiter)
+            st = cg_solve(d, x, rhs, op_sloppy=ds,
+                          sloppy=p.cuda_prec_sloppy, tol=p.tol,
+                          maxiter=p.maxiter, delta=p.reliable_delta)
+        else:
+            st = cg_solve(d, x, rhs, tol=p.tol, maxiter=p.maxiter)
+        if p.chrono_index >= 0:
+            chrono_forecaster(p.chrono_index, p.chrono_max_dim).append(x)
+        return st
     if inv == InverterType.CGNR:
         return cgnr_solve(d, x, b, tol=p.tol, maxiter=p.maxiter)
     if inv == InverterType.CGNE:

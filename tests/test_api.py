@@ -369,3 +369,59 @@ def test_anisotropy_folds_into_spatial_links(resident):
     api.load_gauge_quda(u2, gp_p)
     out_m = api.mat_quda(b, p)
     assert (out_a - out_m).abs().max().item() < 1e-13
+
+
+def test_staggered_phase_applied_interop(resident):
+    """MILC-convention phase-folded links load to the same staggered
+    operator as bare links."""
+    from quda_amd.ops.reference import staggered_phases
+    geo = LatticeGeometry((4, 4, 4, 4))
+    u = resident.clone()
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(u, gp)
+    p = InvertParam(dslash_type=DslashType.STAGGERED, mass=0.1)
+    g = torch.Generator().manual_seed(598)
+    b = torch.view_as_complex(torch.randn((2, 128, 3, 2), generator=g,
+                                          dtype=torch.float64))
+    out_bare = api.mat_quda(b, p)
+    u_ph = u.clone()
+    for par in (0, 1):
+        ph = staggered_phases(geo, par)
+        for mu in range(4):
+            u_ph[mu, par] = u_ph[mu, par] * ph[:, mu].to(u.dtype).reshape(
+                -1, 1, 1)
+    gp2 = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                     cuda_prec_sloppy="double",
+                     staggered_phase_applied=True)
+    api.load_gauge_quda(u_ph, gp2)
+    out_ph = api.mat_quda(b, p)
+    assert (out_bare - out_ph).abs().max().item() < 1e-13
+
+
+def test_chrono_guess_accelerates_sequence(resident):
+    """use_resident_chrono role: successive related solves reuse the
+    chrono basis and converge in fewer iterations."""
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    api.flush_chrono_quda()
+    g = torch.Generator().manual_seed(599)
+    b0 = torch.view_as_complex(torch.randn((2, 128, 4, 3, 2), generator=g,
+                                           dtype=torch.float64))
+    db = torch.view_as_complex(torch.randn((2, 128, 4, 3, 2), generator=g,
+                                           dtype=torch.float64))
+    iters_plain, iters_chrono = [], []
+    for i in range(3):
+        b = b0 + 0.01 * i * db  # slowly-varying source sequence
+        p1 = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12,
+                         tol=1e-9, maxiter=500)
+        api.invert_quda(b, p1)
+        iters_plain.append(p1.iter)
+        p2 = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12,
+                         tol=1e-9, maxiter=500, chrono_index=1)
+        api.invert_quda(b, p2)
+        iters_chrono.append(p2.iter)
+    api.flush_chrono_quda()
+    assert iters_chrono[-1] < iters_plain[-1], (iters_chrono, iters_plain)
+
