@@ -556,6 +556,47 @@ def mom_action_quda(mom: torch.Tensor) -> float:
     return mom_action(mom)
 
 
+_MOM = {"p": None}
+
+
+def mom_resident_quda(mom: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """ref: momResidentQuda — set (mom given) or get the resident
+    momentum field."""
+    if mom is not None:
+        _MOM["p"] = mom
+    return _MOM["p"]
+
+
+def compute_gauge_path_force_quda(paths, coeffs, beta: float = 1.0
+                                  ) -> torch.Tensor:
+    """ref: computeGaugeForceQuda with explicit input paths — the force
+    of S = beta * sum_i c_i sum_x Re tr[1 - P_i(x)/3] for arbitrary
+    signed-direction paths, by reverse-mode differentiation of the loop
+    traces (validated against the analytic plaquette force and FD in
+    the tests). Single-rank (autograd path); the plaquette action has
+    the analytic multi-rank force."""
+    from .gauge.ops import path_product, project_ta
+    from .parallel import comms
+    assert comms.comm_size() == 1, "autograd path force is single-rank"
+    geo = _R.geo
+    u_req = _R.u_complex.detach().clone().requires_grad_(True)
+    s = None
+    for c, p in zip(coeffs, paths):
+        P = path_product(u_req, geo, tuple(p))
+        tr = torch.diagonal(P, dim1=-2, dim2=-1).sum(-1).real.sum() / 3.0
+        term = beta * c * (geo.volume - tr)
+        s = term if s is None else s + term
+    s.backward()
+    g = u_req.grad
+    u = _R.u_complex
+    F = torch.empty_like(u)
+    for mu in range(4):
+        for par in (0, 1):
+            F[mu, par] = 0.5 * project_ta(u[mu, par]
+                                          @ g[mu, par].conj().mT)
+    return F
+
+
 def gauss_mom_quda(seed: int) -> torch.Tensor:
     from .gauge import random_momentum
     return random_momentum(_R.geo, _R.gauge.device, seed)

@@ -425,3 +425,24 @@ def test_chrono_guess_accelerates_sequence(resident):
     api.flush_chrono_quda()
     assert iters_chrono[-1] < iters_plain[-1], (iters_chrono, iters_plain)
 
+
+
+def test_gauge_path_force_matches_plaquette(resident):
+    """computeGaugeForceQuda with explicit plaquette paths reproduces
+    the analytic plaquette force."""
+    from quda_amd.gauge import gauge_force
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    paths, coeffs = [], []
+    for mu in range(4):
+        for nu in range(mu + 1, 4):
+            paths.append((mu + 1, nu + 1, -(mu + 1), -(nu + 1)))
+            coeffs.append(1.0)
+    beta = 5.5
+    F = api.compute_gauge_path_force_quda(paths, coeffs, beta)
+    F_ref = gauge_force(resident, api._R.geo, beta)
+    assert (F - F_ref).abs().max().item() < 1e-12
+    # resident momentum round trip
+    api.mom_resident_quda(F)
+    assert api.mom_resident_quda() is F
