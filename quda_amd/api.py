@@ -166,6 +166,10 @@ def init_quda(device: int = 0) -> None:
 
 def end_quda() -> None:
     free_gauge_quda()
+    _R.fat_gauge = None
+    _R.long_gauge = None
+    _MOM["p"] = None
+    _CHRONO.clear()
 
 
 def free_gauge_quda() -> None:

@@ -343,3 +343,26 @@ def test_qcharge_density_sums_to_charge(geo, u_rand):
     q = topological_charge(u_rand, geo)
     qd = topological_charge_density(u_rand, geo)
     assert abs(qd.sum().item() - q) < 1e-10
+
+
+def test_improved_action_hmc_conserves(geo, u_rand):
+    """Symanzik-improved gauge HMC: Omelyan trajectory with the autograd
+    improved force conserves H at O(dt^2)."""
+    import torch
+    from quda_amd.gauge import omelyan
+    from quda_amd.gauge.ops import improved_gauge_action, improved_gauge_force
+    beta, c1 = 5.0, -1.0 / 12.0
+    P = random_momentum(geo, seed=197)
+
+    def force(uc):
+        return improved_gauge_force(uc, geo, beta, c1=c1)
+
+    def H(uc, Pc):
+        return (mom_action(Pc)
+                + improved_gauge_action(uc, geo, beta, c1=c1))
+
+    dHs = []
+    for n in (8, 16):
+        u1, P1 = omelyan(u_rand, P, geo, force, n, 0.3 / n)
+        dHs.append(abs(H(u1, P1) - H(u_rand, P)))
+    assert dHs[1] < dHs[0] / 2.5, dHs
