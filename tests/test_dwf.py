@@ -550,3 +550,23 @@ def test_eofa5_gpu_vs_oracle(setup, prec, kind, dagger, pm):
                 kw["pm"], kw["u"], kw["w"], dagger)
     tol = 1e-11 if prec == "double" else 1e-4
     assert (out.to_complex()[0] - expect).abs().max().item() < tol
+
+
+def test_madwf_with_zmobius_inner(setup):
+    """The production MADWF configuration: a cheap COMPLEX-coefficient
+    (zMobius) inner operator accelerating a Moebius outer solve."""
+    from quda_amd.models import DiracMobiusPC, DiracZMobiusPC
+    from quda_amd.solvers import gcr_solve
+    from quda_amd.solvers.madwf import madwf_solve
+    geo, g = setup
+    big = DiracMobiusPC(g, MF, M5, LS, b5=1.5, c5=0.5)
+    b5z, c5z = _zcoefs()
+    small = DiracZMobiusPC(g, MF, M5, LS // 2, b5z[:LS // 2], c5z[:LS // 2])
+    b = spin5(geo, 341, n_parity=1)
+    x0 = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st0 = gcr_solve(big, x0, b, tol=1e-8, maxiter=300)
+    x1 = SpinorField(geo, "double", n_parity=1, ls=LS)
+    st1 = madwf_solve(big, small, x1, b, tol=1e-8, inner_tol=1e-4)
+    assert st0.converged and st1.converged
+    err = (x1.to_complex() - x0.to_complex()).abs().max().item()
+    assert err < 1e-6, err
