@@ -446,3 +446,24 @@ def test_gauge_path_force_matches_plaquette(resident):
     # resident momentum round trip
     api.mom_resident_quda(F)
     assert api.mom_resident_quda() is F
+
+
+def test_mixed_precision_and_dwf_matpc_through_api(resident):
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="single")
+    api.load_gauge_quda(resident, gp)
+    # mixed-precision CG (reliable updates) through the API
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12, tol=1e-9,
+                    maxiter=600, cuda_prec="double",
+                    cuda_prec_sloppy="single")
+    b = _rand_spinor((2, 128, 4, 3), 601)
+    api.invert_quda(b, p)
+    assert p.true_res < 1e-8, p.true_res
+    # Mobius MATPC solve + reconstruct through the API
+    p2 = InvertParam(dslash_type=DslashType.MOBIUS, mass=0.04, Ls=4,
+                     solution_type=SolutionType.MATPC,
+                     inv_type=InverterType.CGNR, tol=1e-9, maxiter=2000,
+                     cuda_prec="double", cuda_prec_sloppy="double")
+    b5 = _rand_spinor((2, 128 * 4, 4, 3), 602)
+    api.invert_quda(b5, p2)
+    assert p2.true_res < 1e-7, p2.true_res
