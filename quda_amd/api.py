@@ -352,10 +352,13 @@ def mat_quda(inp: torch.Tensor, p: InvertParam,
     return out.to_complex().cpu()
 
 
-def invert_quda(b: torch.Tensor, p: InvertParam) -> torch.Tensor:
+def invert_quda(b: torch.Tensor, p: InvertParam,
+                x0: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Solve M x = b (ref: invertQuda interface_quda.cpp:2986): builds the
     precise/sloppy operator pair, runs the selected solver (PC solves go
-    through prepare/reconstruct), fills p.iter/true_res/secs/gflops."""
+    through prepare/reconstruct), fills p.iter/true_res/secs/gflops.
+    x0 (optional) is the initial guess (use_init_guess role) for
+    non-PC solves."""
     t0 = time.perf_counter()
     d = _make_dirac(p)
     b_f = _wrap(b, p, 2)
@@ -368,7 +371,8 @@ def invert_quda(b: torch.Tensor, p: InvertParam) -> torch.Tensor:
         x_f = b_f.clone_empty()
         d.reconstruct(x_f, x_e, b_f)
     else:
-        x_f = b_f.clone_empty()
+        x_f = (_wrap(x0, p, 2) if x0 is not None
+               else b_f.clone_empty())
         stats = _run_solver(d, x_f, b_f, p, sloppy_pair=False)
 
     # true residual on the requested system
@@ -520,11 +524,19 @@ def gauge_observables_quda() -> dict:
     }
 
 
-def perform_gauge_smear_quda(kind: str, n_steps: int, coeff: float) -> None:
+def perform_gauge_smear_quda(kind: str, n_steps: int, coeff: float,
+                             measure: bool = False):
     """ref: performGaugeSmearQuda / performWFlowQuda — smears the resident
-    field in place."""
+    field in place; measure=True (wilson_flow only) also returns the
+    per-step (t, E_plaq, E_clover, t^2 E) history (the
+    gaugeSmearParam.meas_interval role)."""
     from .gauge import ape_smear, stout_smear, wilson_flow
     u = _R.u_complex
+    if kind == "wilson_flow" and measure:
+        from .gauge import wilson_flow_measure
+        u, hist = wilson_flow_measure(u, _R.geo, coeff, n_steps)
+        load_gauge_quda(u, _R.gauge_param)
+        return hist
     if kind == "ape":
         u = ape_smear(u, _R.geo, coeff, n_steps)
     elif kind == "stout":

@@ -467,3 +467,19 @@ def test_mixed_precision_and_dwf_matpc_through_api(resident):
     b5 = _rand_spinor((2, 128 * 4, 4, 3), 602)
     api.invert_quda(b5, p2)
     assert p2.true_res < 1e-7, p2.true_res
+
+
+def test_init_guess_and_flow_measure(resident):
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    p = InvertParam(dslash_type=DslashType.WILSON, kappa=0.12,
+                    inv_type=InverterType.CGNR, tol=1e-9, maxiter=500)
+    b = _rand_spinor((2, 128, 4, 3), 611)
+    x1 = api.invert_quda(b, p)
+    it_cold = p.iter
+    x2 = api.invert_quda(b, p, x0=x1)  # warm start at the solution
+    assert p.iter < max(it_cold // 2, 2), (p.iter, it_cold)
+    hist = api.perform_gauge_smear_quda("wilson_flow", 3, 0.02,
+                                        measure=True)
+    assert len(hist) == 3 and hist[0][0] < hist[-1][0]
