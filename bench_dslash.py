@@ -58,7 +58,7 @@ def main():
     ap.add_argument("--lattice", default="32,32,32,64")
     ap.add_argument("--reps", type=int, default=100)
     ap.add_argument("--families", default="wilson",
-                    help="comma list: wilson,staggered,mobius,all")
+                    help="comma list: wilson,staggered,mobius,dwf5,all")
     args = ap.parse_args()
     fams = set(args.families.split(","))
     if "all" in fams:
@@ -84,6 +84,8 @@ def main():
         results.update(bench_staggered(geo, u, args.reps))
     if "mobius" in fams:
         results.update(bench_mobius(geo, u, max(args.reps // 5, 3)))
+    if "dwf5" in fams:
+        results.update(bench_dwf5(geo, max(args.reps, 20)))
     print(json.dumps({k: round(v) for k, v in results.items()}))
 
 
@@ -130,6 +132,51 @@ def bench_staggered(geo, u, reps):
         key = f"hisq/{prec}/r18"
         out[key] = gf2
         print(f"{key:32s} {dt*1e6:8.1f} us  {gf2:8.0f} GFLOPS", flush=True)
+    return out
+
+
+def bench_dwf5(geo, reps, Ls=12):
+    """5th-dimension kernel rates: M5 apply/inverse, zMobius, EOFA
+    (bandwidth-bound s-structure ops; useful to confirm they stay
+    negligible next to the 4-d hops)."""
+    from quda_amd.ops.dispatch import dwf5_op, eofa5_op, zdwf5_op
+    dev = "cuda"
+    out = {}
+    V5 = geo.volume_cb * Ls
+    for prec in ("double", "single"):
+        a = SpinorField(geo, prec, dev, n_parity=1, ls=Ls).gaussian_(seed=7)
+        o = SpinorField(geo, prec, dev, n_parity=1, ls=Ls)
+        zb = [1.5 + 0.05j * s for s in range(Ls)]
+        zc = [0.5 - 0.03j * s for s in range(Ls)]
+        import math
+        kap = 0.5
+        eu = [kap ** s for s in range(Ls)]
+        n = math.sqrt(sum(x * x for x in eu))
+        eu = [x / n for x in eu]
+        cases = {
+            "m5": lambda: dwf5_op(o, a, 1.8, -0.5, 0.04, kind=0),
+            "m5inv": lambda: dwf5_op(o, a, 1.8, -0.5, 0.04, kind=1),
+            "zm5": lambda: zdwf5_op(o, a, zb, zc, 0.04, 0),
+            "zm5inv": lambda: zdwf5_op(o, a, zb, zc, 0.04, 1),
+            "eofa_m5": lambda: eofa5_op(o, a, 1.8, -0.5, 0.04, 0, -0.2, 1,
+                                        eu, eu),
+            "eofa_m5inv": lambda: eofa5_op(o, a, 1.8, -0.5, 0.04, 1, -0.2,
+                                           1, eu, eu),
+        }
+        bytes_per = {"double": 8, "single": 4}[prec] * 24 * 2  # r+w
+        for name, fn in cases.items():
+            for _ in range(3):
+                fn()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(reps):
+                fn()
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / reps
+            out[f"{name}_Ls{Ls}/{prec}"] = {
+                "us": dt * 1e6,
+                "GBps": V5 * bytes_per / dt / 1e9,
+            }
     return out
 
 
