@@ -51,6 +51,8 @@ class DslashType(Enum):
     MOBIUS_EOFA = "mobius-eofa"
     NDEG_TWISTED_MASS = "ndeg-twisted-mass"
     NDEG_TWISTED_CLOVER = "ndeg-twisted-clover"
+    CLOVER_HASENBUSCH_TWIST = "clover-hasenbusch-twist"
+    LAPLACE = "laplace"
 
 
 class InverterType(Enum):
@@ -305,6 +307,15 @@ def _make_dirac(p: InvertParam, sloppy: bool = False):
         return (DiracNdegTwistedCloverPC(g, cl, p.kappa, p.mu, p.epsilon)
                 if pc else
                 DiracNdegTwistedClover(g, cl, p.kappa, p.mu, p.epsilon))
+    if t == DslashType.CLOVER_HASENBUSCH_TWIST:
+        assert cl is not None, "load_clover_quda first"
+        from .models import (DiracCloverHasenbuschTwist,
+                             DiracCloverHasenbuschTwistPC)
+        return (DiracCloverHasenbuschTwistPC(g, cl, p.kappa, p.mu) if pc
+                else DiracCloverHasenbuschTwist(g, cl, p.kappa, p.mu))
+    if t == DslashType.LAPLACE:
+        from .models.laplace import GaugeLaplace
+        return GaugeLaplace(g, m2=p.mass, ndim=3)
     if t == DslashType.ASQTAD:
         from .models import (DiracImprovedStaggered,
                              DiracImprovedStaggeredPC)

@@ -483,3 +483,22 @@ def test_init_guess_and_flow_measure(resident):
     hist = api.perform_gauge_smear_quda("wilson_flow", 3, 0.02,
                                         measure=True)
     assert len(hist) == 3 and hist[0][0] < hist[-1][0]
+
+
+def test_hasenbusch_and_laplace_types(resident):
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double")
+    api.load_gauge_quda(resident, gp)
+    pcl = InvertParam(dslash_type=DslashType.CLOVER, kappa=0.11,
+                      clover_csw=1.2)
+    api.load_clover_quda(pcl)
+    p = InvertParam(dslash_type=DslashType.CLOVER_HASENBUSCH_TWIST,
+                    kappa=0.11, mu=0.2, inv_type=InverterType.CGNR,
+                    tol=1e-8, maxiter=800)
+    b = _rand_spinor((2, 128, 4, 3), 621)
+    api.invert_quda(b, p)
+    assert p.true_res < 1e-6, p.true_res
+    p2 = InvertParam(dslash_type=DslashType.LAPLACE, mass=0.5,
+                     inv_type=InverterType.CG, tol=1e-9, maxiter=400)
+    x = api.invert_quda(b, p2)
+    assert p2.true_res < 1e-7, p2.true_res
