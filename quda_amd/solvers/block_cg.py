@@ -41,9 +41,15 @@ def block_cg_solve(op, xs: List[SpinorField], bs: List[SpinorField], *,
         blas.copy(P[i], bs[i])
     rho = block_cdot(R, R)          # [N,N]
     k = 0
+    batched = hasattr(op, "MdagM_batch")
     while k < maxiter:
-        for i in range(N):
-            op.MdagM(AP[i], P[i], tmp)
+        if batched:
+            # merged-halo multi-RHS path: one message per face for the
+            # whole block (dispatch.dslash_wilson_batch underneath)
+            op.MdagM_batch(AP, P)
+        else:
+            for i in range(N):
+                op.MdagM(AP[i], P[i], tmp)
         gamma = block_cdot(P, AP).cpu().numpy()
         try:
             alpha = np.linalg.solve(gamma, rho.cpu().numpy())
