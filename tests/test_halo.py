@@ -848,3 +848,45 @@ def test_block_cg_batched_halos_gloo(tmp_path):
     init_file = str(tmp_path / "init_bcgb")
     mp.spawn(_worker_block_cg_batched, args=(2, init_file), nprocs=2,
              join=True)
+
+
+def _worker_determinism(rank, world, init_file):
+    import torch.distributed as dist
+    from quda_amd.ops.blas import set_deterministic
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=(1, 1, 1, world))
+        gg, u_lex, src_lex = _global_fields(seed=79)
+        from quda_amd.fields.geometry import checkerboard_split
+        lg, u_loc_lex = _local_slice(gg, (1, 1, 1, world),
+                                     comms.grid_coords(), u_lex.movedim(0, 1))
+        u_loc = checkerboard_split(u_loc_lex, lg).permute(
+            2, 0, 1, 3, 4).contiguous()
+        g = GaugeField(lg, "double").from_complex(u_loc)
+        mycoords = comms.grid_coords()
+        _, s_loc = _local_slice(gg, (1, 1, 1, world), mycoords, src_lex)
+        b = SpinorField(lg, "double")
+        b.from_complex(checkerboard_split(s_loc, lg))
+        d = DiracWilson(g, 0.12)
+        try:
+            set_deterministic(True)
+            sols = []
+            for _ in range(2):
+                x = SpinorField(lg, "double")
+                st = cg_solve(d, x, b, tol=1e-9, maxiter=400)
+                assert st.converged
+                sols.append(x.to_complex().clone())
+        finally:
+            set_deterministic(False)
+        # bitwise identical across repeated distributed solves
+        assert torch.equal(sols[0], sols[1]), f"rank{rank} nondeterministic"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_deterministic_distributed_solve_gloo(tmp_path):
+    """QUDA_DETERMINISTIC_REDUCE role: repeated multi-rank solves are
+    bitwise identical under deterministic reductions."""
+    init_file = str(tmp_path / "init_det")
+    mp.spawn(_worker_determinism, args=(2, init_file), nprocs=2, join=True)

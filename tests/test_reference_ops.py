@@ -97,3 +97,27 @@ def test_clover_unit_gauge_is_identity(small_geo):
     A = ref.clover_matrix(u, geo, kappa=0.1, csw=1.0)
     eye = torch.eye(12, dtype=torch.complex128)
     assert (A - eye).abs().max().item() < 1e-13
+
+
+@pytest.mark.parametrize("dims,kappa,seed", [
+    ((2, 4, 6, 8), 0.09, 11), ((6, 6, 2, 4), 0.21, 12),
+    ((4, 2, 8, 2), 0.13, 13)])
+def test_dslash_adjoint_identity_shape_sweep(dims, kappa, seed):
+    """Property sweep: gamma5-hermiticity <chi, D psi> = <g5 D g5 chi,
+    psi>* holds on arbitrary even anisotropic shapes (layout corner
+    cases)."""
+    import torch
+    from quda_amd import GaugeField, LatticeGeometry, SpinorField
+    from quda_amd.ops import reference as ref
+    geo = LatticeGeometry(dims)
+    u = GaugeField(geo, "double").random_su3_(seed=seed).to_complex()
+    gen = torch.Generator().manual_seed(seed + 1)
+    psi = torch.view_as_complex(torch.randn(
+        (geo.volume_cb, 4, 3, 2), generator=gen, dtype=torch.float64))
+    chi = torch.view_as_complex(torch.randn(
+        (geo.volume_cb, 4, 3, 2), generator=gen, dtype=torch.float64))
+    Dp = ref.dslash_wilson_parity(u, psi, geo, 0, False)
+    Ddc = ref.dslash_wilson_parity(u, chi, geo, 1, True)
+    lhs = (chi.conj() * Dp).sum()
+    rhs = (Ddc.conj() * psi).sum()
+    assert abs(lhs - rhs) < 1e-10 * max(abs(lhs), 1e-30)
