@@ -106,6 +106,18 @@ class GaugeField:
         L = self.L
         V = self.geo.volume_cb
         assert u.shape == (4, 2, V, 3, 3)
+        if self.reconstruct != "none":
+            # recon-12 rebuilds row 2 as conj(row0 x row1): only valid for
+            # unitary links. Fat/smeared links MUST use reconstruct="none"
+            # — catch the misuse at load time on a sample.
+            s = u[:, :, : min(V, 64)]
+            r2 = torch.cross(s[..., 0, :], s[..., 1, :], dim=-1).conj()
+            err = (r2 - s[..., 2, :]).abs().max().item()
+            if err > 1e-3:
+                raise ValueError(
+                    "reconstruct='twelve' on non-unitary links (row-2 "
+                    f"reconstruction error {err:.2e}); use "
+                    "reconstruct='none' for fat/smeared fields")
         dev = u.device
         flat = torch.empty((2, V, 8 * L),
                            dtype=torch.float64 if u.dtype == torch.complex128 else torch.float32,

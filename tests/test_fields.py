@@ -102,3 +102,19 @@ def test_interop_roundtrips(small_geo):
     m = interop.gauge_to_milc(u, geo)
     assert m.shape == (geo.volume, 4, 3, 3)
     assert (m[0, 2] - u[2, 0, 0]).abs().max().item() == 0
+
+
+def test_recon12_rejects_nonunitary_links(small_geo):
+    """Fat links loaded into a recon-12 field would be silently
+    corrupted; the load must refuse."""
+    import pytest
+    import torch
+    from quda_amd.fields.gauge import GaugeField
+    geo = small_geo
+    gen = torch.Generator().manual_seed(17)
+    u = torch.view_as_complex(torch.randn(
+        (4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+        dtype=torch.float64))  # NOT unitary
+    g = GaugeField(geo, "double", reconstruct="twelve")
+    with pytest.raises(ValueError):
+        g.from_complex(u)
