@@ -208,6 +208,7 @@ def load_gauge_quda(u: torch.Tensor, param: GaugeParam) -> None:
         # off.
         assert param.reconstruct == "none", \
             "anisotropy needs reconstruct='none'"
+        param.reconstruct_sloppy = "none"  # scaled links are not unitary
         u = u.clone()
         u[0:3] = u[0:3] / param.anisotropy
     if param.t_boundary == "anti":
