@@ -4,8 +4,11 @@
 
 template <typename Prec>
 static void clover_apply_t(const CloverApplyCall &c, hipStream_t st) {
-  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.Vcb};
-  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.Vcb};
+  // spinor strides come from the BlasField (slice views carry
+  // v_stride/s_offset for 5-d doublet fields); the clover field is
+  // always 4-d with stride c.Vcb
+  SpinorAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.out.Vcb};
+  SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
   CloverAcc<Prec> cl{(const typename Prec::Store *)c.clover, c.Vcb};
   int blk = 256;
   int grid = (int)((c.sites + blk - 1) / blk);
