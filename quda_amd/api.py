@@ -217,6 +217,13 @@ def load_gauge_quda(u: torch.Tensor, param: GaugeParam) -> None:
         # phase application at load in gauge_field.cpp). Applied on a
         # copy so the caller's field is untouched; on a partitioned T
         # grid only the LAST rank's boundary slice carries the phase.
+        # Negated SU(3) links cannot be represented by recon-12 (row2 =
+        # conj(r0 x r1) always rebuilds +u2, never -u2), so reconstruction
+        # must be off for BOTH residencies until the codec applies the
+        # boundary phase at decode time (QUDA does this inside
+        # gauge_field_order.h's Reconstruct functors).
+        param.reconstruct = "none"
+        param.reconstruct_sloppy = "none"
         from .parallel import comms
         u = u.clone()
         gt = comms.grid_dims()[3]

@@ -109,8 +109,11 @@ class GaugeField:
         if self.reconstruct != "none":
             # recon-12 rebuilds row 2 as conj(row0 x row1): only valid for
             # unitary links. Fat/smeared links MUST use reconstruct="none"
-            # — catch the misuse at load time on a sample.
-            s = u[:, :, : min(V, 64)]
+            # — catch the misuse at load time on a sample STRIDED across
+            # the whole volume (a t=0-only sample misses boundary-phased
+            # timeslices and partially-smeared regions).
+            idx = torch.arange(0, V, max(1, V // 64), device=u.device)
+            s = u[:, :, idx]
             r2 = torch.cross(s[..., 0, :], s[..., 1, :], dim=-1).conj()
             err = (r2 - s[..., 2, :]).abs().max().item()
             if err > 1e-3:

@@ -104,6 +104,7 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
 
     maxr = sqrt(r2)    # max residual since last reliable update
     k = 0
+    x_s_dirty = False  # x_s holds progress not yet folded into x
     while r2 > stop and k < maxiter:
         op_sloppy.MdagM(Ap, p, tmp_s)
         pAp = blas.re_dot(p, Ap)
@@ -111,6 +112,7 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
             break  # breakdown (ref: inv_cg_quda.cpp:265 checks)
         alpha = r2 / pAp
         blas.axpy(alpha, p, x_s)
+        x_s_dirty = True
         r2_old = r2
         r2 = blas.axpy_norm2(-alpha, Ap, r_s)
         k += 1
@@ -125,6 +127,7 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
             op.MdagM(r, x, tmp_hi)
             r2 = blas.xmy_norm2(b, r)
             x_s.zero_()
+            x_s_dirty = False
             blas.copy(r_s, r)
             # restart direction with beta continuation:
             # p = r_s + beta p  with beta = r2_new / r2_old
@@ -138,8 +141,11 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
             beta = r2 / r2_old
             blas.xpay(r_s, beta, p)
 
-    if not mixed:
-        # accumulate the (high-precision) iterate
+    if x_s_dirty:
+        # Fold the sloppy accumulator into x unconditionally — a maxiter
+        # or pAp<=0 exit must not discard progress since the last
+        # reliable update (ref: inv_cg_quda.cpp copies xSloppy back after
+        # the loop) — and recompute the true residual.
         blas.copy(tmp_hi, x_s)
         blas.axpy(1.0, tmp_hi, x)
         op.MdagM(r, x, tmp_hi)

@@ -344,8 +344,9 @@ def test_antiperiodic_t_boundary(resident):
     for par in (0, 1):
         idx = g2.face_index_cb(par, 3, 3)
         u2[3, par, idx] = -u2[3, par, idx]
+    # manually-negated boundary links are not SU(3): recon must be off
     gp_p = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
-                      cuda_prec_sloppy="double")
+                      cuda_prec_sloppy="double", reconstruct_sloppy="none")
     api.load_gauge_quda(u2, gp_p)
     out_p = api.mat_quda(b, p)
     assert (out_a - out_p).abs().max().item() < 1e-13
@@ -503,3 +504,18 @@ def test_hasenbusch_and_laplace_types(resident):
                      inv_type=InverterType.CG, tol=1e-9, maxiter=400)
     x = api.invert_quda(b, p2)
     assert p2.true_res < 1e-7, p2.true_res
+
+
+def test_anti_boundary_forces_reconstruct_off(resident):
+    """anti-periodic t-boundary negates U_t on the last
+    timeslice; recon-12 cannot represent negated SU(3) links, so the load
+    must force reconstruction off for both residencies, and the sloppy
+    operator must agree with the precise one."""
+    gp = GaugeParam(X=(4, 4, 4, 4), device="cpu", cuda_prec="double",
+                    cuda_prec_sloppy="double", t_boundary="anti")
+    assert gp.reconstruct_sloppy == "twelve"  # the default, pre-load
+    api.load_gauge_quda(resident.clone(), gp)
+    assert gp.reconstruct_sloppy == "none"
+    # sloppy links bit-match the precise ones (same precision here)
+    d = (api._R.gauge.to_complex() - api._R.gauge_sloppy.to_complex())
+    assert d.abs().max().item() < 1e-14

@@ -161,3 +161,20 @@ def test_clover_full_vs_blocks(setup):
     expect = (ref.apply_clover(A, b.to_complex())
               - KAPPA * ref.dslash_wilson_full(u, b.to_complex(), geo))
     assert (out.to_complex() - expect).abs().max().item() < 1e-12
+
+
+def test_cg_mixed_maxiter_exit_keeps_progress(setup):
+    """exiting the mixed-precision loop via maxiter
+    (no reliable update yet) must still fold the sloppy accumulator into
+    x — a few iterations must beat x=0."""
+    geo, g, b = setup
+    d = DiracWilsonPC(g, KAPPA)
+    rhs = b.parity_view(0)
+    x = SpinorField(geo, "double", n_parity=1)
+    stats = cg_solve(d, x, rhs, sloppy="single", tol=1e-30, maxiter=7,
+                     delta=1e-30)  # delta=0 never triggers a reliable update
+    assert stats.iters == 7 and stats.reliable_updates == 0
+    from quda_amd.ops import blas as _blas
+    assert _blas.norm2(x) > 0.0, "progress discarded on maxiter exit"
+    # true residual must have decreased vs the b (x=0) residual
+    assert stats.resid < 1.0, stats.resid
