@@ -4,17 +4,20 @@
 the BASELINE.json headline config ("Wilson-clover Dslash GFLOPS/GPU +
 CG time-to-solution, 32^3x64 mixed-prec").
 
-One "step" = `--iters` mixed-precision CG iterations of the hot loop
-(fixed work; the production solver path, nothing skipped). The metric is
-aggregate sustained GFLOPS over the whole job computed from the reference
-flop model (ref: include/dslash.h:467 — Wilson 1320 flop/site, clover 504;
-blas/reduction flops counted per op), on synthetic data: random SU(3)
-gauge + gaussian source, random-init = same protocol as the reference's
-tests (tests/utils/host_utils.cpp constructRandomGaugeField).
+One "step" = ONE FULL mixed-precision CG solve from x=0 to --tol
+(relative L2 residual) with PRODUCTION settings: reliable updates ON at
+delta=0.1, nothing skipped. The gauge field is a random SU(3) field and
+the source is gaussian — the reference's own benchmark protocol
+(tests/utils/host_utils.cpp constructRandomGaugeField;
+tests/invert_test.cpp:300-376 verification scheme). The metric is
+aggregate sustained GFLOPS over the whole job from the reference flop
+model (include/dslash.h:467: Wilson 1320 flop/site, clover 504; blas and
+reliable-update work counted per op from the MEASURED iteration and
+update counts). secs-per-solve (time-to-solution) is reported alongside.
 
 Weak scaling: each rank owns a full 32^3x64 local lattice; ranks are laid
-out along T. (Multi-rank path requires the halo exchange engine; N=1 is
-the single-GPU path.)
+out along T (one global 4-d lattice, T-partitioned; the CG reductions and
+halo exchanges are global, so the iteration count is rank-independent).
 """
 
 from __future__ import annotations
@@ -64,7 +67,7 @@ def setup_fields(geo, device, sloppy_prec, seed):
 
 
 def flops_per_iter(Vcb: int) -> float:
-    """Flop count of one CG iteration on the even-odd clover system.
+    """Flop count of one sloppy CG iteration on the even-odd clover system.
 
     MdagM = M (2 fused dslash-clover launches: 2*(1320+504+48 xpay/scale))
           + Mdag (clover-inv 504 + fused 1824 + dslash 1320 + 48)
@@ -75,13 +78,22 @@ def flops_per_iter(Vcb: int) -> float:
     return float(Vcb) * (mdagm + blas_f)
 
 
+def flops_per_reliable(Vcb: int) -> float:
+    """One reliable update: precise MdagM + xmy_norm2 + accumulate/copy blas."""
+    mdagm = (2 * (1320 + 504) + 48) + (504 + 1824 + 1320 + 48)
+    return float(Vcb) * (mdagm + 96 + 48 + 48)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
-    ap.add_argument("--iters", type=int, default=25,
-                    help="CG iterations per step (fixed work)")
+    ap.add_argument("--tol", type=float, default=1e-8,
+                    help="relative residual target of each solve")
+    ap.add_argument("--delta", type=float, default=0.1,
+                    help="reliable-update trigger (production default 0.1)")
+    ap.add_argument("--maxiter", type=int, default=2000)
     ap.add_argument("--lattice", type=str, default="32,32,32,64")
     ap.add_argument("--sloppy", type=str, default="half",
                     choices=["double", "single", "half"])
@@ -108,22 +120,27 @@ def main():
     b = SpinorField(geo, "double", device, n_parity=1).gaussian_(seed=args.seed + 100 + rank)
     x = SpinorField(geo, "double", device, n_parity=1)
 
-    def step():
+    counts = {"iters": 0, "reliable": 0, "resid": 0.0}
+
+    def step(record=False):
         x.zero_()
-        # fixed-iteration mixed-precision CG hot loop (tol=0: never exits
-        # early; delta tiny: reliable updates off at this iteration count,
-        # matching production behaviour for <50 iters between updates)
-        cg_solve(d, x, b, op_sloppy=ds, sloppy=args.sloppy, tol=0.0,
-                 maxiter=args.iters, delta=1e-30)
+        # one production solve: solve-to-tolerance, reliable updates ON
+        st = cg_solve(d, x, b, op_sloppy=ds, sloppy=args.sloppy,
+                      tol=args.tol, maxiter=args.maxiter, delta=args.delta)
+        if record:
+            counts["iters"] += st.iters
+            counts["reliable"] += st.reliable_updates
+            counts["resid"] = st.resid
+        return st
 
     for _ in range(args.warmup):
-        step()
+        st0 = step()
     comms.barrier()
     if torch.cuda.is_available():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        step(record=True)
     comms.barrier()
     if torch.cuda.is_available():
         torch.cuda.synchronize()
@@ -139,7 +156,8 @@ def main():
         elapsed = t.item()
 
     ms_per_step = elapsed / args.steps * 1000.0
-    total_flops = flops_per_iter(geo.volume_cb) * args.iters * args.steps * world
+    total_flops = (flops_per_iter(geo.volume_cb) * counts["iters"]
+                   + flops_per_reliable(geo.volume_cb) * counts["reliable"]) * world
     gflops = total_flops / elapsed / 1e9
 
     if rank == 0:
@@ -159,11 +177,16 @@ def main():
             "config": {
                 "model": "wilson_clover_eo_pc_cg",
                 "lattice_per_gpu": "x".join(str(v) for v in dims),
-                "global_batch": args.iters,
+                "global_batch": 1,
                 "seq_len": geo.volume,
                 "kappa": KAPPA,
                 "csw": CSW,
-                "cg_iters_per_step": args.iters,
+                "solve_tol": args.tol,
+                "reliable_delta": args.delta,
+                "secs_per_solve": round(elapsed / args.steps, 4),
+                "iters_per_solve": counts["iters"] / args.steps,
+                "reliable_updates_per_solve": counts["reliable"] / args.steps,
+                "final_resid": counts["resid"],
                 "parallelism": f"dd_t{world}",
                 "gauge_recon_sloppy": 12 if args.sloppy != "double" else 18,
             },

@@ -53,12 +53,36 @@ def run(prec, recon_name, mode, geo, u, A, reps):
     return dict(us=dt * 1e6, gflops=flops / dt / 1e9, gbs=bytes_ / dt / 1e9)
 
 
+def sweep(geo, u, reps):
+    """(block x waves) launch-config sweep of the hot Wilson kernels."""
+    from quda_amd.ops.dispatch import hip_ext as _ext
+    A = ref.clover_matrix(u, geo, 0.135, 1.0)
+    results = {}
+    for waves in (0, 3):
+        _ext().set_dslash_waves(waves)
+        blocks = [64] if waves == 3 else [64, 128, 256]
+        for blk in blocks:
+            _ext().set_dslash_block(blk)
+            for prec, recon in [("half", "twelve"), ("single", "twelve")]:
+                for mode, mname in [(PLAIN, "wilson"),
+                                    (CLOV_POST, "wilson_clover")]:
+                    r = run(prec, recon, mode, geo, u, A, reps)
+                    key = f"{mname}/{prec}/r12/b{blk}w{waves}"
+                    results[key] = round(r["gflops"])
+                    print(f"{key:40s} {r['us']:8.1f} us  {r['gflops']:8.0f} "
+                          f"GFLOPS", flush=True)
+    _ext().set_dslash_waves(0)
+    _ext().set_dslash_block(64)
+    print(json.dumps(results))
+    return results
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--lattice", default="32,32,32,64")
     ap.add_argument("--reps", type=int, default=100)
     ap.add_argument("--families", default="wilson",
-                    help="comma list: wilson,staggered,mobius,dwf5,all")
+                    help="comma list: wilson,staggered,mobius,dwf5,all,sweep")
     args = ap.parse_args()
     fams = set(args.families.split(","))
     if "all" in fams:
@@ -70,6 +94,11 @@ def main():
     from quda_amd.fields.gauge import project_su3
     u = project_su3(torch.view_as_complex(m.to(torch.float64)).cuda())
     results = {}
+    if "sweep" in fams:
+        results.update(sweep(geo, u, args.reps))
+        fams.discard("sweep")
+        if not fams:
+            return
     if "wilson" in fams:
         A = ref.clover_matrix(u, geo, 0.135, 1.0)
         for prec, recon in [("double", "none"), ("single", "none"),

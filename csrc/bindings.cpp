@@ -376,6 +376,10 @@ static void set_dslash_block(int64_t b) {
   if (b == 64 || b == 128 || b == 256) qa_dslash_block_ref() = (int)b;
 }
 
+static void set_dslash_waves(int64_t w) {
+  if (w == 0 || w == 3) qa_dslash_waves_ref() = (int)w;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover/-twisted) dslash",
         py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
@@ -394,6 +398,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
+  m.def("set_dslash_waves", &set_dslash_waves,
+        "occupancy experiment: 0 default, 3 = 64-thread/3-wave variant");
   m.def("dwf5", &dwf5, "DWF/Moebius 5th-dim ops (Ds apply / M5 inverse)");
   m.def("zdwf5", &zdwf5, "zMobius per-slice-complex 5th-dim ops");
   m.def("eofa5", &eofa5, "EOFA rank-1 extended M5 ops");

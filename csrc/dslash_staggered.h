@@ -85,10 +85,15 @@ __global__ __launch_bounds__(256) void k_dslash_staggered(
     if (!(KT == KT_INTERIOR && cross_m)) {                                \
       if (KT == KT_FUSED && cross_m) {                                    \
         gh.load_v(p, MU, 0, ghost_idx(xc, MU, d));                        \
+        g.template load<4 + MU>(U, i);                                    \
       } else {                                                            \
-        in.load_v(p, neighbor_cb(xc, MU, -1, d));                         \
+        /* bwd link from the -mu neighbor's FWD slot: same duplicated */  \
+        /* value, shared L2 line -> halves unique gauge traffic (see */   \
+        /* dslash_wilson.h). Own bwd slot only for rank-crossing hops. */ \
+        long j = neighbor_cb(xc, MU, -1, d);                              \
+        in.load_v(p, j);                                                  \
+        g.template load<MU>(U, j);                                        \
       }                                                                   \
-      g.template load<4 + MU>(U, i);                                      \
       su3_dagmul_vec(up, U, p);                                           \
       for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \
     }                                                                     \
@@ -123,10 +128,12 @@ __global__ __launch_bounds__(256) void k_dslash_staggered(
       int lm = 2 - xc[MU];            /* bwd ghost layer if >= 0 */       \
       if (KT == KT_FUSED && gh.active(MU) && lm >= 0) {                   \
         gh.load_v(p, MU, 0, (long)lm * gh.Fcb[MU] + ghost_idx(xc, MU, d)); \
+        lng.template load<4 + MU>(U, i);                                  \
       } else {                                                            \
-        in.load_v(p, neighbor_cb3(xc, MU, -1, d));                        \
+        long j3 = neighbor_cb3(xc, MU, -1, d);                            \
+        in.load_v(p, j3);                                                 \
+        lng.template load<MU>(U, j3);                                     \
       }                                                                   \
-      lng.template load<4 + MU>(U, i);                                    \
       su3_dagmul_vec(up, U, p);                                           \
       for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \
       esum2 += xc[MU];                                                    \

@@ -98,8 +98,13 @@ __device__ __forceinline__ void clover_mul(cplx<R> out[4][3], const R diag[2][6]
   }
 }
 
-template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY, int KT = KT_LOCAL>
-__global__ __launch_bounds__(256) void k_dslash_wilson(
+// LBW = 0: 256-thread cap, allocator unconstrained (measured 222 VGPRs ->
+// 2 waves/SIMD). LBW > 0: 64-thread workgroups with a minimum of LBW waves
+// per SIMD — caps the register allocation (512/LBW) to raise occupancy on
+// this latency-bound kernel (profiles/r01_dslash_pmc_analysis.md).
+template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY, int KT = KT_LOCAL,
+          int LBW = 0>
+__global__ __launch_bounds__(LBW ? 64 : 256, LBW ? LBW : 1) void k_dslash_wilson(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
     SpinorAcc<Prec> x, GhostAcc<Prec> gh, typename Prec::Real br,
@@ -129,6 +134,17 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
   bool skip[8];
 #pragma unroll
   for (int k = 0; k < 8; ++k) skip[k] = false;
+  // -mu neighbor index, kept for phase 2: the backward link U_mu(x-mu) is
+  // read from the NEIGHBOR's forward slot instead of this site's bwd slot.
+  // The stencil layout stores every link twice (fwd slot of x == bwd slot
+  // of x+mu); reading only fwd slots halves the dslash's UNIQUE gauge
+  // traffic — neighbor fwd-slot lines are shared through L2 exactly like
+  // the neighbor spinor loads. Boundary-crossing hops (jm < 0) still read
+  // the bwd slot, which holds the -mu RANK's link from the load-time
+  // exchange (fields/gauge.py from_complex).
+  long jm[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k) jm[k] = -1;
 
   // Phase 1: gather + spin-project all 8 neighbor half-spinors first — the
   // loads are independent, so the wave has 8 spinor fetches in flight
@@ -155,19 +171,15 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
       gh.load(h[2 * MU + 1], MU, 0, ghost_idx(xc, MU, d));                \
     } else {                                                              \
       long j = neighbor_cb(xc, MU, -1, d);                                \
+      jm[MU] = j;                                                         \
       in.load(p, j);                                                      \
       if constexpr (!DAG) proj_##MU##_1(h[2 * MU + 1], p);                \
       else proj_##MU##_0(h[2 * MU + 1], p);                               \
     }                                                                     \
   }
 
-  QA_GATHER(0)
-  QA_GATHER(1)
-  QA_GATHER(2)
-  QA_GATHER(3)
-#undef QA_GATHER
-
-  // Phase 2: gauge multiplies + spin reconstruction (site-local gauge)
+  // Phase 2 macro: gauge multiplies + spin reconstruction (fwd link at own
+  // site; bwd link from the -mu neighbor's fwd slot, see jm above)
 #define QA_MUL(MU)                                                        \
   {                                                                       \
     if (!skip[2 * MU]) {                                                  \
@@ -177,17 +189,42 @@ __global__ __launch_bounds__(256) void k_dslash_wilson(
       else recon_##MU##_1(acc, uh, one);                                  \
     }                                                                     \
     if (!skip[2 * MU + 1]) {                                              \
-      g.template load<4 + MU>(U, i);                                      \
+      if (jm[MU] >= 0) g.template load<MU>(U, jm[MU]);                    \
+      else g.template load<4 + MU>(U, i);                                 \
       su3_dagmul_half(uh, U, h[2 * MU + 1]);                              \
       if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                   \
       else recon_##MU##_0(acc, uh, one);                                  \
     }                                                                     \
   }
 
-  QA_MUL(0)
-  QA_MUL(1)
-  QA_MUL(2)
-  QA_MUL(3)
+  if constexpr (LBW == 0) {
+    // all 8 gathers in flight, then all multiplies (max load ILP; ~210
+    // VGPRs -> 2 waves/SIMD)
+    QA_GATHER(0)
+    QA_GATHER(1)
+    QA_GATHER(2)
+    QA_GATHER(3)
+    QA_MUL(0)
+    QA_MUL(1)
+    QA_MUL(2)
+    QA_MUL(3)
+  } else {
+    // split-gather: two 4-neighbor rounds halve the live half-spinor
+    // state so the LBW-wave occupancy cap is met without spilling —
+    // trades per-wave ILP for thread-level parallelism.
+    QA_GATHER(0)
+    QA_GATHER(1)
+    QA_MUL(0)
+    QA_MUL(1)
+    // pin the round split: without this the scheduler hoists round-2 loads
+    // into round 1 and the live state spills right back to the 8-gather shape
+    __builtin_amdgcn_sched_barrier(0);
+    QA_GATHER(2)
+    QA_GATHER(3)
+    QA_MUL(2)
+    QA_MUL(3)
+  }
+#undef QA_GATHER
 #undef QA_MUL
 
   // boundary sites under CLOV_POST defer the whole epilogue: store raw sum

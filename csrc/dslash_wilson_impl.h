@@ -34,11 +34,18 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
     if ((c.comm_mask >> m) & 1) n_ext += 2 * c.face_cb[m];
   int grid_ext = (int)((n_ext + blk - 1) / blk);
 
+  int waves = qa_dslash_waves();
+  int grid64 = (int)((c.Vcb + 63) / 64);
+
 #define QA_LAUNCH(DAG, MODE, XPAY, KT)                                        \
   if (c.kt == 3)                                                              \
     hipLaunchKernelGGL((k_dslash_wilson_exterior<Prec, RECON, DAG, MODE, XPAY>), \
                        dim3(grid_ext), dim3(blk), 0, st, out, in, g, cl, d,   \
                        c.parity, a, x, gh, n_ext, br, bi);                    \
+  else if (waves == 3)                                                        \
+    hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY, KT, 3>),\
+                       dim3(grid64), dim3(64), 0, st, out, in, g, cl, d,      \
+                       c.parity, a, x, gh, br, bi);                           \
   else                                                                        \
     hipLaunchKernelGGL((k_dslash_wilson<Prec, RECON, DAG, MODE, XPAY, KT>),   \
                        dim3(grid), dim3(blk), 0, st, out, in, g, cl, d,       \

@@ -16,6 +16,19 @@ inline int &qa_dslash_block_ref() {
 }
 inline int qa_dslash_block() { return qa_dslash_block_ref(); }
 
+// occupancy experiment: 0 = default (256-thread cap, ~2 waves/SIMD at the
+// measured 222 VGPRs), 3 = 64-thread workgroups with __launch_bounds__
+// forcing >= 3 waves/SIMD (VGPR cap 168). Only {0,3} are compiled.
+inline int &qa_dslash_waves_ref() {
+  static int w = [] {
+    const char *e = getenv("QUDA_AMD_DSLASH_WAVES");
+    int v = e ? atoi(e) : 0;
+    return v == 3 ? 3 : 0;
+  }();
+  return w;
+}
+inline int qa_dslash_waves() { return qa_dslash_waves_ref(); }
+
 struct BlasField {
   void *data;
   void *norm;  // nullptr unless half

@@ -20,7 +20,7 @@ def _launch_two_rank(port):
         e = dict(env, RANK=str(r), LOCAL_RANK=str(r), WORLD_SIZE="2")
         procs.append(subprocess.Popen(
             [sys.executable, os.path.join(ROOT, "bench.py"), "--gpus", "2",
-             "--steps", "1", "--warmup", "0", "--iters", "2",
+             "--steps", "1", "--warmup", "0", "--maxiter", "4",
              "--lattice", "4,4,4,8", "--sloppy", "double", "--device", "cpu"],
             env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
     outs = [p.communicate(timeout=300) for p in procs]
@@ -50,7 +50,7 @@ def test_bench_two_ranks_gloo():
 def test_bench_single_rank_cpu():
     out = subprocess.run(
         [sys.executable, os.path.join(ROOT, "bench.py"), "--steps", "1",
-         "--warmup", "0", "--iters", "2", "--lattice", "4,4,4,8",
+         "--warmup", "0", "--maxiter", "4", "--lattice", "4,4,4,8",
          "--sloppy", "double", "--device", "cpu"],
         capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stderr

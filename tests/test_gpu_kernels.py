@@ -264,3 +264,25 @@ def test_batch_dslash_self_wrap_gpu():
         err = (outs[i].to_complex()
                - outs_ref[i].to_complex()).abs().max().item()
         assert err < 1e-13, (i, err)
+
+
+@pytest.mark.parametrize("prec", ["single", "half"])
+def test_dslash_occupancy_variant(setup, prec):
+    """The LBW=3 (64-thread / 3-wave __launch_bounds__) kernel variant must
+    match the oracle exactly like the default variant."""
+    from quda_amd.ops.dispatch import hip_ext as _ext
+    geo, g, psi, chi, A = setup
+    u = g.to_complex()
+    gd, sd = _gpu_fields(geo, g, psi, prec, 12)
+    out = SpinorField(geo, prec, "cuda", n_parity=1)
+    _ext().set_dslash_waves(3)
+    try:
+        for dagger in (False, True):
+            dslash_wilson(out, sd.parity_view(1), gd, 0, dagger)
+            got = out.to_complex().cpu()[0]
+            want = ref.dslash_wilson_parity(u, psi.to_complex()[1], geo, 0,
+                                            dagger)
+            err = (got - want).abs().max().item() / want.abs().max().item()
+            assert err < TOL[prec], (prec, dagger, err)
+    finally:
+        _ext().set_dslash_waves(0)
