@@ -184,11 +184,14 @@ struct GaugeAcc {
   using R = typename Prec::Real;
   static constexpr int W = Prec::W;
   static constexpr int NCH = (8 * RECON) / W;  // chunks per site
-  const S *data;  // parity-adjusted base: [NCH][V][W]
+  const S *data;   // parity-adjusted base: [NCH][V][W]
+  const S *other;  // OPPOSITE-parity base (bwd links read from the -mu
+                   // neighbor's fwd slot live there: x-mu flips parity)
   long V;
 
   template <int Q>
-  __device__ __forceinline__ void load(cplx<R> u[3][3], long i) const {
+  __device__ __forceinline__ void load_base(const S *base, cplx<R> u[3][3],
+                                            long i) const {
     constexpr int e0 = Q * RECON;
     constexpr int c0 = e0 / W;
     constexpr int c1 = (e0 + RECON - 1) / W;
@@ -197,7 +200,7 @@ struct GaugeAcc {
     S tmp[NC * W];
 #pragma unroll
     for (int c = 0; c < NC; ++c)
-      load_chunk<S, W>(data + ((long)(c0 + c) * V + i) * W, tmp + c * W);
+      load_chunk<S, W>(base + ((long)(c0 + c) * V + i) * W, tmp + c * W);
 #pragma unroll
     for (int k = 0; k < RECON / 2; ++k)
       u[k / 3][k % 3] = {(R)tmp[off + 2 * k], (R)tmp[off + 2 * k + 1]};
@@ -208,6 +211,16 @@ struct GaugeAcc {
         u[2][c] = conj(u[0][a] * u[1][b] - u[0][b] * u[1][a]);
       }
     }
+  }
+
+  template <int Q>
+  __device__ __forceinline__ void load(cplx<R> u[3][3], long i) const {
+    load_base<Q>(data, u, i);
+  }
+  // fwd slot Q at site i of the OTHER parity block
+  template <int Q>
+  __device__ __forceinline__ void load_o(cplx<R> u[3][3], long i) const {
+    load_base<Q>(other, u, i);
   }
 };
 

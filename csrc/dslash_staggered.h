@@ -92,7 +92,7 @@ __global__ __launch_bounds__(256) void k_dslash_staggered(
         /* dslash_wilson.h). Own bwd slot only for rank-crossing hops. */ \
         long j = neighbor_cb(xc, MU, -1, d);                              \
         in.load_v(p, j);                                                  \
-        g.template load<MU>(U, j);                                        \
+        g.template load_o<MU>(U, j);                                      \
       }                                                                   \
       su3_dagmul_vec(up, U, p);                                           \
       for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \
@@ -132,7 +132,7 @@ __global__ __launch_bounds__(256) void k_dslash_staggered(
       } else {                                                            \
         long j3 = neighbor_cb3(xc, MU, -1, d);                            \
         in.load_v(p, j3);                                                 \
-        lng.template load<MU>(U, j3);                                     \
+        lng.template load_o<MU>(U, j3);                                   \
       }                                                                   \
       su3_dagmul_vec(up, U, p);                                           \
       for (int c = 0; c < 3; ++c) acc[c] += (-eta) * up[c];               \

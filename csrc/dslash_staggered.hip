@@ -8,14 +8,16 @@ static void stag_launch(const StagDslashCall &c, hipStream_t st) {
   StagAcc<Prec> out{(typename Prec::Store *)c.out.data, (float *)c.out.norm, c.Vcb};
   StagAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.Vcb};
   StagAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.Vcb};
-  const auto *gbase = (const typename Prec::Store *)c.gauge +
-                      (long)c.parity * GaugeAcc<Prec, RECON>::NCH * c.Vcb * Prec::W;
-  GaugeAcc<Prec, RECON> g{gbase, c.Vcb};
-  const typename Prec::Store *lbase = nullptr;
-  if (c.long_gauge)
-    lbase = (const typename Prec::Store *)c.long_gauge +
-            (long)c.parity * GaugeAcc<Prec, 18>::NCH * c.Vcb * Prec::W;
-  GaugeAcc<Prec, 18> lng{lbase, c.Vcb};
+  const long gpar = (long)GaugeAcc<Prec, RECON>::NCH * c.Vcb * Prec::W;
+  const auto *g0 = (const typename Prec::Store *)c.gauge;
+  GaugeAcc<Prec, RECON> g{g0 + c.parity * gpar, g0 + (1 - c.parity) * gpar,
+                          c.Vcb};
+  const typename Prec::Store *l0 = (const typename Prec::Store *)c.long_gauge;
+  const long lpar = (long)GaugeAcc<Prec, 18>::NCH * c.Vcb * Prec::W;
+  // 3-hop neighbors flip parity too (odd shift): bwd long links also read
+  // the other parity block's fwd slots
+  GaugeAcc<Prec, 18> lng{l0 ? l0 + c.parity * lpar : nullptr,
+                         l0 ? l0 + (1 - c.parity) * lpar : nullptr, c.Vcb};
   LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
   GhostAcc<Prec, 6> gh{};
   gh.mask = c.comm_mask;

@@ -189,7 +189,7 @@ __global__ __launch_bounds__(LBW ? 64 : 256, LBW ? LBW : 1) void k_dslash_wilson
       else recon_##MU##_1(acc, uh, one);                                  \
     }                                                                     \
     if (!skip[2 * MU + 1]) {                                              \
-      if (jm[MU] >= 0) g.template load<MU>(U, jm[MU]);                    \
+      if (jm[MU] >= 0) g.template load_o<MU>(U, jm[MU]);                  \
       else g.template load<4 + MU>(U, i);                                 \
       su3_dagmul_half(uh, U, h[2 * MU + 1]);                              \
       if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                   \

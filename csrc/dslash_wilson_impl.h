@@ -13,9 +13,10 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   SpinorAcc<Prec> in{(typename Prec::Store *)c.in.data, (float *)c.in.norm, c.in.Vcb};
   SpinorAcc<Prec> x{(typename Prec::Store *)c.x.data, (float *)c.x.norm, c.x.Vcb};
   // parity-offset the stencil gauge base: [2][NCH][V][W]
-  const auto *gbase = (const typename Prec::Store *)c.gauge +
-                      (long)c.parity * GaugeAcc<Prec, RECON>::NCH * c.Vcb * Prec::W;
-  GaugeAcc<Prec, RECON> g{gbase, c.Vcb};
+  const long gpar = (long)GaugeAcc<Prec, RECON>::NCH * c.Vcb * Prec::W;
+  const auto *g0 = (const typename Prec::Store *)c.gauge;
+  GaugeAcc<Prec, RECON> g{g0 + c.parity * gpar, g0 + (1 - c.parity) * gpar,
+                          c.Vcb};
   CloverAcc<Prec> cl{(const typename Prec::Store *)c.clover, c.Vcb};
   LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset, c.Vcb};
   GhostAcc<Prec> gh{};

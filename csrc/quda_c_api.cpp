@@ -1,0 +1,355 @@
+// quda_amd C ABI implementation (include/quda_amd.h).
+//
+// The engine is the quda_amd Python package driving the in-tree HIP
+// kernels; this translation unit embeds the interpreter (pybind11::embed)
+// and marshals host arrays <-> torch tensors with zero-copy views
+// (torch.frombuffer on a memoryview of the caller's buffer), so a C or
+// Fortran client gets the full solver stack through plain extern "C"
+// symbols (role of the reference's lib/interface_quda.cpp entry points;
+// the resident-field caching lives in quda_amd/api.py).
+//
+// Build: g++/hipcc -shared -fPIC quda_c_api.cpp -lpython3.10 (build_hip.py
+// target libquda_amd_c.so). No torch C++ linkage — all torch calls go
+// through Python, so the .so only depends on libpython.
+
+#include <pybind11/embed.h>
+
+#include <cstring>
+#include <string>
+
+#include "../include/quda_amd.h"
+
+namespace py = pybind11;
+
+static std::string g_err;
+static bool g_inited = false;
+static py::object g_api;        // quda_amd.api module
+static py::object g_torch;      // torch module
+static std::string g_device;    // "cuda:N" or "cpu"
+
+const char *qudaAmdLastError(void) { return g_err.c_str(); }
+
+#define QA_TRY                                                                \
+  try {                                                                       \
+    py::gil_scoped_acquire gil;
+
+#define QA_END                                                                \
+    g_err.clear();                                                            \
+    return 0;                                                                 \
+  } catch (const std::exception &e) {                                         \
+    g_err = e.what();                                                         \
+    return -1;                                                                \
+  }
+
+static const char *prec_str(QudaAmdPrecision p) {
+  switch (p) {
+    case QUDA_AMD_SINGLE: return "single";
+    case QUDA_AMD_HALF: return "half";
+    default: return "double";
+  }
+}
+
+static const char *recon_str(QudaAmdReconstruct r) {
+  switch (r) {
+    case QUDA_AMD_RECON_12: return "twelve";
+    case QUDA_AMD_RECON_8: return "eight";
+    default: return "none";
+  }
+}
+
+static long vcb_of(const int X[4]) {
+  return (long)X[0] * X[1] * X[2] * X[3] / 2;
+}
+
+// zero-copy torch view of a host buffer (complex128), shaped
+static py::object tensor_view(const void *ptr, size_t n_cplx,
+                              py::tuple shape) {
+  auto mv = py::memoryview::from_memory(const_cast<void *>(ptr),
+                                        n_cplx * 16, false);
+  py::object t = g_torch.attr("frombuffer")(
+      mv, py::arg("dtype") = g_torch.attr("complex128"));
+  return t.attr("reshape")(shape);
+}
+
+// copy a (cpu, contiguous, complex128) tensor's data into a host buffer
+static void tensor_out(py::object t, void *dst) {
+  t = t.attr("contiguous")().attr("cpu")();
+  t = t.attr("to")(g_torch.attr("complex128"));
+  auto nbytes = t.attr("numel")().cast<long>() * 16;
+  auto ptr = t.attr("data_ptr")().cast<uintptr_t>();
+  std::memcpy(dst, reinterpret_cast<void *>(ptr), nbytes);
+}
+
+QudaAmdGaugeParam newQudaAmdGaugeParam(void) {
+  QudaAmdGaugeParam p;
+  std::memset(&p, 0, sizeof(p));
+  p.X[0] = p.X[1] = p.X[2] = p.X[3] = 8;
+  p.cpu_prec = QUDA_AMD_DOUBLE;
+  p.cuda_prec = QUDA_AMD_DOUBLE;
+  p.cuda_prec_sloppy = QUDA_AMD_HALF;
+  p.reconstruct = QUDA_AMD_RECON_NO;
+  p.reconstruct_sloppy = QUDA_AMD_RECON_12;
+  p.anisotropy = 1.0;
+  p.t_boundary = QUDA_AMD_PERIODIC_T;
+  return p;
+}
+
+QudaAmdInvertParam newQudaAmdInvertParam(void) {
+  QudaAmdInvertParam p;
+  std::memset(&p, 0, sizeof(p));
+  p.dslash_type = QUDA_AMD_WILSON_DSLASH;
+  p.inv_type = QUDA_AMD_CG_INVERTER;
+  p.solution_type = QUDA_AMD_MAT_SOLUTION;
+  p.kappa = 0.12;
+  p.mass = 0.05;
+  p.tol = 1e-8;
+  p.maxiter = 1000;
+  p.reliable_delta = 0.1;
+  p.cpu_prec = QUDA_AMD_DOUBLE;
+  p.cuda_prec = QUDA_AMD_DOUBLE;
+  p.cuda_prec_sloppy = QUDA_AMD_HALF;
+  p.Ls = 8;
+  p.m5 = 1.8;
+  p.b5 = 1.5;
+  p.c5 = 0.5;
+  return p;
+}
+
+int initQuda(int device) {
+  try {
+    if (!g_inited) {
+      if (!Py_IsInitialized()) py::initialize_interpreter();
+      py::gil_scoped_acquire gil;
+      py::module_ sys = py::module_::import("sys");
+      const char *root = getenv("QUDA_AMD_ROOT");
+#ifdef QA_REPO_ROOT
+      sys.attr("path").attr("insert")(0, root ? root : QA_REPO_ROOT);
+#else
+      if (root) sys.attr("path").attr("insert")(0, root);
+#endif
+      g_torch = py::module_::import("torch");
+      g_api = py::module_::import("quda_amd.api");
+      const char *dev = getenv("QUDA_AMD_DEVICE");
+      if (dev) {
+        g_device = dev;
+      } else if (g_torch.attr("cuda").attr("is_available")().cast<bool>()) {
+        g_device = "cuda:" + std::to_string(device < 0 ? 0 : device);
+      } else {
+        g_device = "cpu";
+      }
+      g_api.attr("init_quda")(device < 0 ? 0 : device);
+      g_inited = true;
+    }
+    g_err.clear();
+    return 0;
+  } catch (const std::exception &e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
+int endQuda(void) {
+  QA_TRY
+  if (g_inited) g_api.attr("end_quda")();
+  QA_END
+}
+
+int freeGaugeQuda(void) {
+  QA_TRY
+  g_api.attr("free_gauge_quda")();
+  QA_END
+}
+
+static py::object make_gauge_param(const QudaAmdGaugeParam *p) {
+  using namespace py::literals;
+  return g_api.attr("GaugeParam")(
+      "X"_a = py::make_tuple(p->X[0], p->X[1], p->X[2], p->X[3]),
+      "cuda_prec"_a = prec_str(p->cuda_prec),
+      "cuda_prec_sloppy"_a = prec_str(p->cuda_prec_sloppy),
+      "reconstruct"_a = recon_str(p->reconstruct),
+      "reconstruct_sloppy"_a = recon_str(p->reconstruct_sloppy),
+      "anisotropy"_a = p->anisotropy,
+      "t_boundary"_a = (p->t_boundary == QUDA_AMD_ANTI_PERIODIC_T
+                            ? "anti" : "periodic"),
+      "staggered_phase_applied"_a = (bool)p->staggered_phase_applied,
+      "device"_a = g_device);
+}
+
+int loadGaugeQuda(const void *h_gauge, QudaAmdGaugeParam *param) {
+  QA_TRY
+  if (param->cpu_prec != QUDA_AMD_DOUBLE)
+    throw std::runtime_error("cpu_prec: only double host data supported");
+  long Vcb = vcb_of(param->X);
+  py::object u = tensor_view(h_gauge, (size_t)4 * 2 * Vcb * 9,
+                             py::make_tuple(4, 2, Vcb, 3, 3));
+  // clone: the engine keeps the field resident beyond this call
+  u = u.attr("clone")();
+  g_api.attr("load_gauge_quda")(u, make_gauge_param(param));
+  QA_END
+}
+
+int saveGaugeQuda(void *h_gauge, QudaAmdGaugeParam *param) {
+  QA_TRY
+  py::object u = g_api.attr("save_gauge_quda")();
+  tensor_out(u, h_gauge);
+  QA_END
+}
+
+static const char *dslash_name(QudaAmdDslashType t) {
+  switch (t) {
+    case QUDA_AMD_CLOVER_WILSON_DSLASH: return "clover";
+    case QUDA_AMD_TWISTED_MASS_DSLASH: return "twisted_mass";
+    case QUDA_AMD_TWISTED_CLOVER_DSLASH: return "twisted_clover";
+    case QUDA_AMD_STAGGERED_DSLASH: return "staggered";
+    case QUDA_AMD_ASQTAD_DSLASH: return "asqtad";
+    case QUDA_AMD_DOMAIN_WALL_DSLASH: return "domain_wall";
+    case QUDA_AMD_MOBIUS_DWF_DSLASH: return "mobius";
+    case QUDA_AMD_LAPLACE_DSLASH: return "laplace";
+    default: return "wilson";
+  }
+}
+
+static const char *inv_name(QudaAmdInverterType t) {
+  switch (t) {
+    case QUDA_AMD_BICGSTAB_INVERTER: return "bicgstab";
+    case QUDA_AMD_GCR_INVERTER: return "gcr";
+    case QUDA_AMD_MR_INVERTER: return "mr";
+    case QUDA_AMD_BICGSTABL_INVERTER: return "bicgstab-l";
+    case QUDA_AMD_CA_CG_INVERTER: return "ca-cg";
+    case QUDA_AMD_CA_GCR_INVERTER: return "ca-gcr";
+    case QUDA_AMD_CGNE_INVERTER: return "cgne";
+    case QUDA_AMD_CGNR_INVERTER: return "cgnr";
+    default: return "cg";
+  }
+}
+
+static py::object make_invert_param(const QudaAmdInvertParam *p) {
+  using namespace py::literals;
+  py::object enums = g_api;
+  py::object dt = enums.attr("DslashType")(dslash_name(p->dslash_type));
+  py::object it = enums.attr("InverterType")(inv_name(p->inv_type));
+  py::object st = enums.attr("SolutionType")(
+      p->solution_type == QUDA_AMD_MATPC_SOLUTION ? "matpc" : "mat");
+  return g_api.attr("InvertParam")(
+      "dslash_type"_a = dt, "inv_type"_a = it, "solution_type"_a = st,
+      "kappa"_a = p->kappa, "mass"_a = p->mass, "mu"_a = p->mu,
+      "clover_csw"_a = p->clover_csw, "tol"_a = p->tol,
+      "maxiter"_a = p->maxiter, "reliable_delta"_a = p->reliable_delta,
+      "cuda_prec"_a = prec_str(p->cuda_prec),
+      "cuda_prec_sloppy"_a = prec_str(p->cuda_prec_sloppy),
+      "Ls"_a = p->Ls, "m5"_a = p->m5, "b5"_a = p->b5, "c5"_a = p->c5);
+}
+
+// complex components per 4-d site (both parities) of the action's spinor
+static long spinor_cplx(const QudaAmdInvertParam *p, long Vcb) {
+  long per_site =
+      (p->dslash_type == QUDA_AMD_STAGGERED_DSLASH ||
+       p->dslash_type == QUDA_AMD_ASQTAD_DSLASH) ? 3 : 12;
+  long ls = (p->dslash_type == QUDA_AMD_DOMAIN_WALL_DSLASH ||
+             p->dslash_type == QUDA_AMD_MOBIUS_DWF_DSLASH) ? p->Ls : 1;
+  return 2 * Vcb * ls * per_site;
+}
+
+static py::tuple spinor_shape(const QudaAmdInvertParam *p, long Vcb,
+                              int n_parity) {
+  long ls = (p->dslash_type == QUDA_AMD_DOMAIN_WALL_DSLASH ||
+             p->dslash_type == QUDA_AMD_MOBIUS_DWF_DSLASH) ? p->Ls : 1;
+  if (p->dslash_type == QUDA_AMD_STAGGERED_DSLASH ||
+      p->dslash_type == QUDA_AMD_ASQTAD_DSLASH)
+    return py::make_tuple(n_parity, Vcb, 3);
+  return py::make_tuple(n_parity, ls * Vcb, 4, 3);
+}
+
+static long resident_vcb() {
+  return g_api.attr("_R").attr("geo").attr("volume_cb").cast<long>();
+}
+
+int loadCloverQuda(const void *h_clover, const void *h_clovinv,
+                   QudaAmdInvertParam *param) {
+  QA_TRY
+  (void)h_clovinv;  // the engine inverts the resident term itself
+  py::object ip = make_invert_param(param);
+  if (h_clover == nullptr) {
+    g_api.attr("load_clover_quda")(ip);
+  } else {
+    long Vcb = resident_vcb();
+    py::object A = tensor_view(h_clover, (size_t)2 * Vcb * 144,
+                               py::make_tuple(2, Vcb, 12, 12));
+    g_api.attr("load_clover_quda")(ip, A.attr("clone")());
+  }
+  QA_END
+}
+
+int invertQuda(void *h_x, const void *h_b, QudaAmdInvertParam *param) {
+  QA_TRY
+  long Vcb = resident_vcb();
+  py::object b = tensor_view(h_b, spinor_cplx(param, Vcb),
+                             spinor_shape(param, Vcb, 2));
+  py::object ip = make_invert_param(param);
+  py::object x = g_api.attr("invert_quda")(b, ip);
+  tensor_out(x, h_x);
+  param->iter = ip.attr("iter").cast<int>();
+  param->true_res = ip.attr("true_res").cast<double>();
+  param->secs = ip.attr("secs").cast<double>();
+  param->gflops = ip.attr("gflops").cast<double>();
+  QA_END
+}
+
+int invertMultiShiftQuda(void **h_x, const void *h_b,
+                         QudaAmdInvertParam *param, const double *offsets,
+                         int num_offset) {
+  QA_TRY
+  long Vcb = resident_vcb();
+  py::object b = tensor_view(h_b, spinor_cplx(param, Vcb) / 2,
+                             spinor_shape(param, Vcb, 1));
+  py::object ip = make_invert_param(param);
+  py::list sh;
+  for (int i = 0; i < num_offset; ++i) sh.append(offsets[i]);
+  py::object xs = g_api.attr("invert_multishift_quda")(b, ip, sh);
+  for (int i = 0; i < num_offset; ++i)
+    tensor_out(py::reinterpret_borrow<py::list>(xs)[i], h_x[i]);
+  param->iter = ip.attr("iter").cast<int>();
+  QA_END
+}
+
+int dslashQuda(void *h_out, const void *h_in, QudaAmdInvertParam *param,
+               int parity) {
+  QA_TRY
+  long Vcb = resident_vcb();
+  py::object in = tensor_view(h_in, spinor_cplx(param, Vcb) / 2,
+                              spinor_shape(param, Vcb, 1));
+  py::object out =
+      g_api.attr("dslash_quda")(in.attr("__getitem__")(0), make_invert_param(param),
+                                parity);
+  tensor_out(out, h_out);
+  QA_END
+}
+
+int MatQuda(void *h_out, const void *h_in, QudaAmdInvertParam *param) {
+  QA_TRY
+  long Vcb = resident_vcb();
+  py::object in = tensor_view(h_in, spinor_cplx(param, Vcb),
+                              spinor_shape(param, Vcb, 2));
+  py::object out = g_api.attr("mat_quda")(in, make_invert_param(param));
+  tensor_out(out, h_out);
+  QA_END
+}
+
+int MatDagMatQuda(void *h_out, const void *h_in, QudaAmdInvertParam *param) {
+  QA_TRY
+  long Vcb = resident_vcb();
+  py::object in = tensor_view(h_in, spinor_cplx(param, Vcb),
+                              spinor_shape(param, Vcb, 2));
+  py::object out = g_api.attr("mat_dag_mat_quda")(in, make_invert_param(param));
+  tensor_out(out, h_out);
+  QA_END
+}
+
+int plaqQuda(double plaq[3]) {
+  QA_TRY
+  py::tuple t = g_api.attr("plaq_quda")();
+  plaq[0] = t[0].cast<double>();
+  plaq[1] = t[1].cast<double>();
+  plaq[2] = t[2].cast<double>();
+  QA_END
+}
