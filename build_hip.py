@@ -110,7 +110,35 @@ def build(force: bool = False, verbose: bool = True) -> str:
             raise RuntimeError(f"link failed:\n{r.stdout}\n{r.stderr}")
         if verbose:
             print(f"linked {OUT_SO}")
+    build_c_api(force=force, verbose=verbose)
     return OUT_SO
+
+
+C_API_SO = os.path.join(ROOT, "libquda_amd_c.so")
+
+
+def build_c_api(force: bool = False, verbose: bool = True) -> str:
+    """extern \"C\" ABI library (include/quda_amd.h): embeds the engine via
+    libpython; no torch C++ linkage, so it builds with plain g++."""
+    import sysconfig
+
+    import pybind11
+    src = os.path.join(CSRC, "quda_c_api.cpp")
+    hdr = os.path.join(ROOT, "include", "quda_amd.h")
+    if not (force or _newer(src, C_API_SO) or _newer(hdr, C_API_SO)):
+        return C_API_SO
+    py_inc = sysconfig.get_paths()["include"]
+    libdir = sysconfig.get_config_var("LIBDIR")
+    pyver = f"python{sys.version_info.major}.{sys.version_info.minor}"
+    cmd = ["g++", "-O2", "-std=c++17", "-fPIC", "-shared",
+           f"-I{pybind11.get_include()}", f"-I{py_inc}",
+           src, "-o", C_API_SO, f"-L{libdir}", f"-l{pyver}"]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        raise RuntimeError(f"C API build failed:\n{r.stdout}\n{r.stderr}")
+    if verbose:
+        print(f"linked {C_API_SO}")
+    return C_API_SO
 
 
 def _newer_f(mtime: float, path: str) -> bool:

@@ -12,6 +12,8 @@
 // target libquda_amd_c.so). No torch C++ linkage — all torch calls go
 // through Python, so the .so only depends on libpython.
 
+#include <dlfcn.h>
+#include <libgen.h>
 #include <pybind11/embed.h>
 
 #include <cstring>
@@ -121,12 +123,20 @@ int initQuda(int device) {
       if (!Py_IsInitialized()) py::initialize_interpreter();
       py::gil_scoped_acquire gil;
       py::module_ sys = py::module_::import("sys");
+      // package location: QUDA_AMD_ROOT env, else the directory holding
+      // this .so (libquda_amd_c.so lives at the repo root next to the
+      // quda_amd package and quda_amd_hip.so)
       const char *root = getenv("QUDA_AMD_ROOT");
-#ifdef QA_REPO_ROOT
-      sys.attr("path").attr("insert")(0, root ? root : QA_REPO_ROOT);
-#else
+      std::string selfdir;
+      if (!root) {
+        Dl_info info;
+        if (dladdr((void *)&initQuda, &info) && info.dli_fname) {
+          std::string p(info.dli_fname);
+          selfdir = p.substr(0, p.find_last_of('/'));
+          root = selfdir.c_str();
+        }
+      }
       if (root) sys.attr("path").attr("insert")(0, root);
-#endif
       g_torch = py::module_::import("torch");
       g_api = py::module_::import("quda_amd.api");
       const char *dev = getenv("QUDA_AMD_DEVICE");
