@@ -30,6 +30,7 @@ KERNEL_SOURCES = [
     "dslash_wilson_h.hip",
     "dslash_staggered.hip",
     "dslash_dwf.hip",
+    "dslash_wilson_mrhs.hip",
 ]
 BINDING_SOURCES = ["bindings.cpp"]
 

@@ -104,6 +104,52 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
   check_launch("dslash_wilson");
 }
 
+static void dslash_wilson_mrhs(
+    std::vector<at::Tensor> out, std::vector<at::Tensor> out_n,
+    std::vector<at::Tensor> in, std::vector<at::Tensor> in_n,
+    at::Tensor gauge, at::Tensor clover, std::vector<at::Tensor> x,
+    std::vector<at::Tensor> x_n, std::vector<int64_t> dims,
+    int64_t parity_offset, int64_t Vcb, int64_t parity, bool dagger,
+    int64_t mode, bool xpay, double a, int64_t recon,
+    std::vector<at::Tensor> ghost, std::vector<at::Tensor> ghost_nrm,
+    std::vector<int64_t> face_cb, int64_t comm_mask, int64_t kt) {
+  int n = (int)out.size();
+  TORCH_CHECK(n == 2 || n == 4, "mrhs kernel supports 2 or 4 RHS");
+  TORCH_CHECK((int)in.size() == n);
+  DslashMrhsCall c{};
+  c.nrhs = n;
+  for (int r = 0; r < n; ++r) {
+    TORCH_CHECK(out[r].is_contiguous() && in[r].is_contiguous());
+    c.out[r] = field_of(out[r], out_n[r], Vcb);
+    c.in[r] = field_of(in[r], in_n[r], Vcb);
+    if (!x.empty()) c.x[r] = field_of(x[r], x_n[r], Vcb);
+  }
+  c.comm_mask = (int)comm_mask;
+  c.kt = (int)kt;
+  if (comm_mask) {
+    TORCH_CHECK(ghost.size() == 8 && ghost_nrm.size() == 8 && face_cb.size() == 4);
+    for (int k = 0; k < 8; ++k) {
+      c.ghost[k] = ptr_or_null(ghost[k]);
+      c.ghost_nrm[k] = (const float *)ptr_or_null(ghost_nrm[k]);
+    }
+    for (int k = 0; k < 4; ++k) c.face_cb[k] = face_cb[k];
+  }
+  c.gauge = gauge.data_ptr();
+  c.clover = ptr_or_null(clover);
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.dagger = dagger;
+  c.mode = (int)mode;
+  c.xpay = xpay;
+  c.a = a;
+  c.recon = (int)recon;
+  c.prec = prec_of(out[0]);
+  launch_dslash_wilson_mrhs(c, stream());
+  check_launch("dslash_wilson_mrhs");
+}
+
 static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
                       at::Tensor in_n, std::vector<int64_t> dims,
                       int64_t parity_offset, int64_t Vcb, int64_t parity,
@@ -396,6 +442,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("parity"), py::arg("mu"), py::arg("s01"), py::arg("edge"),
         py::arg("Fcb"), py::arg("v_stride") = 0, py::arg("s_offset") = 0);
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
+  m.def("dslash_wilson_mrhs", &dslash_wilson_mrhs,
+        "multi-RHS Wilson(-clover) dslash: NRHS sides per gauge load");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("set_dslash_waves", &set_dslash_waves,

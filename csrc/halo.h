@@ -82,6 +82,26 @@ struct GhostAcc {
     static_assert(NCOMP == 12);
     load_v(reinterpret_cast<cplx<R> *>(h), mu, dir, f);
   }
+
+  // batched-halo view: rhs slice r of a [n_rhs][NCH][depth*Fcb][GW]
+  // buffer (parallel/halo.py BatchSpinorHalo — buf[] points at slice 0)
+  __device__ __forceinline__ void load_r(cplx<R> (&h)[2][3], int mu, int dir,
+                                         long f, int r) const {
+    static_assert(NCOMP == 12);
+    const S *b = buf[2 * mu + dir];
+    long stride = (long)depth * Fcb[mu];
+    b += (long)r * NCH * stride * GW;
+    S tmp[NCOMP];
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      load_chunk<S, GW>(b + ((long)ch * stride + f) * GW, tmp + ch * GW);
+    R scale = (R)1;
+    if constexpr (Prec::has_norm) scale = nrm[2 * mu + dir][(long)r * stride + f];
+    auto *out = reinterpret_cast<cplx<R> *>(h);
+#pragma unroll
+    for (int k = 0; k < NCPLX; ++k)
+      out[k] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+  }
 };
 
 // write NCOMP/2 complex values into a send buffer slot

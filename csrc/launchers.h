@@ -94,6 +94,34 @@ void launch_dslash_wilson_double(const DslashCall &c, hipStream_t st);
 void launch_dslash_wilson_single(const DslashCall &c, hipStream_t st);
 void launch_dslash_wilson_half(const DslashCall &c, hipStream_t st);
 
+// ---------------------------------------------------------------------------
+// multi-RHS Wilson(-clover): NRHS in {2,4} sides share each gauge/clover
+// load (csrc/dslash_wilson_mrhs.h). mode PLAIN or CLOV_POST only.
+#define QA_MRHS_MAX 4
+struct DslashMrhsCall {
+  int nrhs;
+  BlasField out[QA_MRHS_MAX], in[QA_MRHS_MAX], x[QA_MRHS_MAX];
+  const void *gauge;
+  const void *clover;
+  int Xdim[4];
+  int parity_offset;
+  long Vcb;
+  int parity;
+  bool dagger;
+  int mode;
+  bool xpay;
+  double a;
+  int recon;
+  // batched ghosts: slice-0 base per [2*mu+dir]; kernel offsets by rhs
+  const void *ghost[8];
+  const float *ghost_nrm[8];
+  long face_cb[4];
+  int comm_mask;
+  int kt;  // 0 local, 1 fused, 2 interior (exterior runs per-RHS)
+  int prec;
+};
+void launch_dslash_wilson_mrhs(const DslashMrhsCall &c, hipStream_t st);
+
 // pack one (mu, s01, edge) face of `in` into dst (see csrc/halo.h)
 struct PackCall {
   BlasField in;       // dslash input spinor (single parity view)

@@ -211,54 +211,114 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     return out
 
 
+def _mrhs_groups(n: int):
+    """Split n RHS into kernel-supported group sizes (4, then 2; odd
+    remainder runs the per-RHS kernel)."""
+    groups, r0 = [], 0
+    while n - r0 >= 4:
+        groups.append((r0, 4))
+        r0 += 4
+    if n - r0 >= 2:
+        groups.append((r0, 2))
+        r0 += 2
+    return groups, r0
+
+
 def dslash_wilson_batch(outs, inps, gauge: GaugeField, parity: int,
-                        dagger: bool = False, a: float = 1.0, xs=None):
-    """Multi-RHS Wilson dslash with MERGED halos: the whole batch ships
-    as one message per face (ref: the reference's create_comms_batch
-    multi-RHS path, dslash_wilson.hpp:48 — here the merged buffer is a
-    [n_rhs, ...] tensor exchanged once; each RHS's kernel consumes its
-    slice). Falls back to the per-RHS path when nothing is partitioned."""
+                        dagger: bool = False, a: float = 1.0, xs=None,
+                        mode: int = PLAIN, clover=None,
+                        clover_inverse: bool = False):
+    """Multi-RHS Wilson(-clover) dslash: kernel-level NRHS batching (the
+    k_dslash_wilson_mrhs kernel processes 2 or 4 sides per gauge/clover
+    load — ref dslash_wilson.cuh:38-40 MAX_MULTI_RHS arrays) plus MERGED
+    halos (one message per face for the whole batch; ref
+    create_comms_batch, dslash_wilson.hpp:48)."""
     from ..parallel import comms
     geo = outs[0].geo
     n = len(inps)
     mask = comms.comm_mask()
-    if not mask or n == 1:
-        for i in range(n):
-            dslash_wilson(outs[i], inps[i], gauge, parity, dagger=dagger,
-                          a=a, x=xs[i] if xs else None)
-        return outs
     xpay = xs is not None
-    if on_gpu(outs[0], inps[0]):
-        ext = hip_ext()
-        from ..parallel.halo import get_batch_halo
-        h = get_batch_halo(geo, inps[0].precision, inps[0].device, mask, n)
-        for i in range(n):
-            h.pack_one(ext, i, inps[i], 1 - parity, bool(dagger))
-        reqs = h.exchange_start()
 
-        def launch(i, kt, ghosts, nrms, face_cb):
-            xf = xs[i] if xs else outs[i]
-            cl_t = torch.empty(0, dtype=outs[i].data.dtype,
-                               device=outs[i].device)
-            ext.dslash_wilson(
-                outs[i].data, norm_or_empty(outs[i]), inps[i].data,
-                norm_or_empty(inps[i]), gauge.data, cl_t, xf.data,
-                norm_or_empty(xf), list(geo.dims), geo.parity_offset,
-                geo.volume_cb, parity, bool(dagger), PLAIN, xpay, float(a),
-                RECON_COMPS[gauge.reconstruct], ghosts, nrms, face_cb,
-                mask, kt, 0.0, 0.0)
+    def per_rhs(i, kt=None, ghosts=None, nrms=None, face_cb=None, h=None):
+        dslash_wilson(outs[i], inps[i], gauge, parity, dagger=dagger,
+                      a=a, x=xs[i] if xs else None, mode=mode,
+                      clover=clover, clover_inverse=clover_inverse)
 
-        for i in range(n):  # interiors overlap the batched transfer
-            ghosts, nrms, face_cb = h.ghost_args(i)
-            launch(i, 2, ghosts, nrms, face_cb)
-        for r in reqs:
-            r.wait()
-        for i in range(n):
-            ghosts, nrms, face_cb = h.ghost_args(i)
-            launch(i, 3, ghosts, nrms, face_cb)
+    if n == 1 or not on_gpu(outs[0], inps[0]):
+        if not mask or n == 1:
+            for i in range(n):
+                per_rhs(i)
+            return outs
+        return _dslash_batch_oracle(outs, inps, gauge, parity, dagger, a,
+                                    xs, mode, clover, clover_inverse)
+
+    ext = hip_ext()
+    groups, rem0 = _mrhs_groups(n)
+    cl_t = torch.empty(0, dtype=outs[0].data.dtype, device=outs[0].device)
+    if mode == CLOV_POST:
+        cl_t = clover.inv_data if clover_inverse else clover.data
+
+    def launch_group(r0, g, kt, ghosts, nrms, face_cb):
+        sl = slice(r0, r0 + g)
+        xg = xs[sl] if xs else []
+        ext.dslash_wilson_mrhs(
+            [o.data for o in outs[sl]], [norm_or_empty(o) for o in outs[sl]],
+            [i_.data for i_ in inps[sl]],
+            [norm_or_empty(i_) for i_ in inps[sl]],
+            gauge.data, cl_t,
+            [x_.data for x_ in xg], [norm_or_empty(x_) for x_ in xg],
+            list(geo.dims), geo.parity_offset, geo.volume_cb, parity,
+            bool(dagger), mode, xpay, float(a),
+            RECON_COMPS[gauge.reconstruct], ghosts, nrms, face_cb,
+            mask if kt else 0, kt)
+
+    if not mask:
+        for r0, g in groups:
+            launch_group(r0, g, 0, [], [], [])
+        for i in range(rem0, n):
+            per_rhs(i)
         return outs
+
+    from ..parallel.halo import get_batch_halo
+    h = get_batch_halo(geo, inps[0].precision, inps[0].device, mask, n)
+    for i in range(n):
+        h.pack_one(ext, i, inps[i], 1 - parity, bool(dagger))
+    reqs = h.exchange_start()
+
+    def launch_one(i, kt):
+        ghosts, nrms, face_cb = h.ghost_args(i)
+        xf = xs[i] if xs else outs[i]
+        ext.dslash_wilson(
+            outs[i].data, norm_or_empty(outs[i]), inps[i].data,
+            norm_or_empty(inps[i]), gauge.data, cl_t, xf.data,
+            norm_or_empty(xf), list(geo.dims), geo.parity_offset,
+            geo.volume_cb, parity, bool(dagger), mode, xpay, float(a),
+            RECON_COMPS[gauge.reconstruct], ghosts, nrms, face_cb,
+            mask, kt, 0.0, 0.0)
+
+    # interiors overlap the batched transfer: mrhs kernels offset into
+    # the batch ghost slabs from the group's base slice
+    for r0, g in groups:
+        ghosts, nrms, face_cb = h.ghost_args(r0)
+        launch_group(r0, g, 2, ghosts, nrms, face_cb)
+    for i in range(rem0, n):
+        launch_one(i, 2)
+    for r in reqs:
+        r.wait()
+    for i in range(n):  # exteriors are boundary-only: per-RHS kernels
+        launch_one(i, 3)
+    return outs
+
+
+def _dslash_batch_oracle(outs, inps, gauge, parity, dagger, a, xs,
+                         mode=PLAIN, clover=None, clover_inverse=False):
     # ---- CPU oracle path: ONE merged exchange for the batch ----
+    from ..parallel import comms
     from ..parallel.halo import (active_dims, exchange_tensors)
+    geo = outs[0].geo
+    n = len(inps)
+    mask = comms.comm_mask()
+    xpay = xs is not None
     psis = [inp.to_complex()[0] for inp in inps]
     sends, recvs = {}, {}
     pin = 1 - parity
@@ -281,6 +341,9 @@ def dslash_wilson_batch(outs, inps, gauge: GaugeField, parity: int,
         }
         res = ref.dslash_wilson_parity(u, psis[i], geo, parity, dagger,
                                        halo=halo)
+        if mode == CLOV_POST:
+            A = clover.to_complex(inverse=clover_inverse)[parity]
+            res = ref.apply_clover(A, res)
         if xpay:
             res = xs[i].to_complex()[0] + a * res
         else:

@@ -286,3 +286,42 @@ def test_dslash_occupancy_variant(setup, prec):
             assert err < TOL[prec], (prec, dagger, err)
     finally:
         _ext().set_dslash_waves(0)
+
+
+@pytest.mark.parametrize("prec", ["half", "single", "double"])
+@pytest.mark.parametrize("nrhs", [2, 4])
+def test_dslash_mrhs_vs_per_rhs(setup, prec, nrhs):
+    """k_dslash_wilson_mrhs (NRHS sides per gauge load) must match the
+    per-RHS kernel bit-for-bit in PLAIN and CLOV_POST(+xpay) modes."""
+    from quda_amd.fields.clover import pack_clover
+    from quda_amd.ops.dispatch import dslash_wilson_batch
+    geo, g, psi, chi, A = setup
+    recon = 12 if prec != "double" else 18
+    gd, _ = _gpu_fields(geo, g, psi, prec, recon)
+    cl = CloverField(geo, prec, "cuda")
+    cl.data.copy_(cl._to_native(pack_clover(A.cuda())))
+    cl.inv_data.copy_(cl.data)  # any hermitian packed field works here
+    gen = torch.Generator().manual_seed(77)
+    inps, outs_m, outs_1, xs = [], [], [], []
+    for r in range(nrhs):
+        v = torch.view_as_complex(
+            torch.randn(1, geo.volume_cb, 4, 3, 2, generator=gen,
+                        dtype=torch.float64)).cuda()
+        inps.append(SpinorField(geo, prec, "cuda", n_parity=1).from_complex(v))
+        xs.append(SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+            0.5 * v))
+        outs_m.append(SpinorField(geo, prec, "cuda", n_parity=1))
+        outs_1.append(SpinorField(geo, prec, "cuda", n_parity=1))
+    for mode, use_x, use_cl in [(PLAIN, False, None), (PLAIN, True, None),
+                                (CLOV_POST, True, cl)]:
+        for dag in (False, True):
+            dslash_wilson_batch(outs_m, inps, gd, 0, dagger=dag, a=-0.3,
+                                xs=xs if use_x else None, mode=mode,
+                                clover=use_cl)
+            for r in range(nrhs):
+                dslash_wilson(outs_1[r], inps[r], gd, 0, dagger=dag,
+                              a=-0.3, x=xs[r] if use_x else None,
+                              mode=mode, clover=use_cl)
+                dm = (outs_m[r].to_complex() - outs_1[r].to_complex())
+                err = dm.abs().max().item()
+                assert err == 0.0, (prec, nrhs, mode, use_x, dag, r, err)
