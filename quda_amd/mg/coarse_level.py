@@ -114,6 +114,9 @@ def build_coarse2_op(co: CoarseOp, t2: CoarseTransfer) -> CoarseOp:
         col = torch.zeros((Na, Nc), dtype=co.X.dtype, device=dev)
         col[t2.sites_by_agg.reshape(-1)] = \
             t2.V[:, :, :, v].reshape(Na, Nc)
+        # multi-rank: hops crossing a rank boundary read the neighbor
+        # rank's column family through the CoarseOp's own face machinery
+        ghosts = co._exchange_c(col)
         # self/X coupling
         WX = torch.einsum("aij,aj->ai", co.X, col)
         X2[:, :, v] += t2.restrict(WX)
@@ -121,14 +124,14 @@ def build_coarse2_op(co: CoarseOp, t2: CoarseTransfer) -> CoarseOp:
         # nbr's; crossing iff a is on the d-boundary of its aggregate
         for d in range(8):
             mu, fwd = d // 2, d % 2
-            src = co.nbr[:, d]
-            W = torch.einsum("aij,aj->ai", co.Y[d], col[src])
+            W = torch.einsum("aij,aj->ai", co.Y[d],
+                             co._nbr_c(col, d, ghosts))
             m = bnd[(mu, fwd)].unsqueeze(-1)
             W_int = torch.where(m, torch.zeros_like(W), W)
             W_bnd = torch.where(m, W, torch.zeros_like(W))
             X2[:, :, v] += t2.restrict(W_int)
             Y2[d][:, :, v] += t2.restrict(W_bnd)
-    return CoarseOp(X2, Y2, t2.cd2)
+    return CoarseOp(X2, Y2, t2.cd2, mask=co.mask)
 
 
 class CoarseMG:
@@ -138,10 +141,10 @@ class CoarseMG:
 
     def __init__(self, co: CoarseOp, block2=(2, 2, 2, 2), n_vec2: int = 4,
                  null_tol: float = 5e-2, null_maxiter: int = 100,
-                 seed: int = 700):
+                 seed: int = 700, vectors=None):
         self.co = co
-        vecs = generate_coarse_null_vectors(co, n_vec2, tol=null_tol,
-                                            maxiter=null_maxiter, seed=seed)
+        vecs = vectors if vectors is not None else generate_coarse_null_vectors(
+            co, n_vec2, tol=null_tol, maxiter=null_maxiter, seed=seed)
         self.t2 = CoarseTransfer(co.cd, block2, vecs)
         self.co2 = build_coarse2_op(co, self.t2)
 
@@ -149,13 +152,14 @@ class CoarseMG:
               maxiter: int = 50, nu: int = 2,
               bottom_tol: float = 1e-2, bottom_maxiter: int = 200
               ) -> torch.Tensor:
+        from .coarse import gnorm2
         x = torch.zeros_like(b)
-        b2 = (b.conj() * b).sum().real.item()
+        b2 = gnorm2(b)
         if b2 == 0:
             return x
         for _ in range(maxiter):
             r = b - self.co.apply(x)
-            r2 = (r.conj() * r).sum().real.item()
+            r2 = gnorm2(r)
             if r2 <= tol * tol * b2:
                 break
             # pre-smooth: a couple of bicgstab steps on the residual eq

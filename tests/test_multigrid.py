@@ -263,3 +263,169 @@ def test_mg_survives_gauge_evolution():
     api.invert_quda(b, p2)
     assert p2.true_res < 1e-7, p2.true_res
     api.end_quda()
+
+
+# ---------------------------------------------------------------------------
+# Multi-rank multigrid: the distributed coarse op, V-cycle
+# and 3-level recursion must match the single-rank build of the same
+# GLOBAL system (ref: coarse ghost machinery, lib/dslash_coarse.hpp:30).
+# ---------------------------------------------------------------------------
+
+def _mg_worker(rank, world, grid, init_file):
+    import torch.distributed as dist
+    from quda_amd.fields.geometry import (checkerboard_join,
+                                          checkerboard_split)
+    from quda_amd.mg.coarse_level import CoarseMG
+    from quda_amd.parallel import comms
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        comms.init_comms(grid=grid)
+        GD = (4, 4, 4, 8)
+        gg = LatticeGeometry(GD)
+        gen = torch.Generator().manual_seed(411)
+        m = torch.randn((4, 2, gg.volume_cb, 3, 3, 2), generator=gen,
+                        dtype=torch.float64)
+        from quda_amd.fields.gauge import project_su3
+        u_g = project_su3(torch.view_as_complex(m))
+        kappa = 0.14
+
+        ldims = tuple(GD[i] // grid[i] for i in range(4))
+        lg = LatticeGeometry(
+            ldims, parity_offset=comms.parity_offset_of_rank(ldims))
+        coords = comms.grid_coords()
+        off = torch.tensor([coords[i] * ldims[i] for i in range(4)])
+
+        def local_lex(lex_field):
+            c = lg.coords.to(torch.int64) + off
+            X, Y, Z, _ = GD
+            glex = ((c[:, 3] * Z + c[:, 2]) * Y + c[:, 1]) * X + c[:, 0]
+            return lex_field[glex]
+
+        # per-direction lex join of the global links, sliced to this rank
+        u_loc = torch.empty((4, 2, lg.volume_cb, 3, 3),
+                            dtype=torch.complex128)
+        for mu in range(4):
+            lexd = checkerboard_join(u_g[mu], gg)  # [V,3,3]
+            loc = local_lex(lexd)
+            u_loc[mu] = checkerboard_split(loc, lg)
+
+        g_loc = GaugeField(lg, "double").from_complex(u_loc)
+        d_loc = DiracWilson(g_loc, kappa)
+
+        # single-rank global truth (each rank computes it identically)
+        with comms.solo_mode():
+            g_glob = GaugeField(gg, "double").from_complex(u_g)
+            d_glob = DiracWilson(g_glob, kappa)
+            from quda_amd.mg.transfer import generate_null_vectors
+            vecs_g = generate_null_vectors(d_glob, 3, tol=1e-3,
+                                           maxiter=150, seed=500)
+            mg_g = MG(d_glob, MGParam(block=(2, 2, 2, 2), n_vec=3,
+                                      nu_post=2, coarse_tol=1e-2,
+                                      coarse_maxiter=100),
+                      vectors=vecs_g)
+
+        # distributed build from the SLICED global null vectors
+        vecs_l = []
+        for v in vecs_g:
+            lexv = checkerboard_join(v.to_complex(), gg)
+            vecs_l.append(SpinorField(lg, "double").from_complex(
+                checkerboard_split(local_lex(lexv), lg)))
+        mg_l = MG(d_loc, MGParam(block=(2, 2, 2, 2), n_vec=3, nu_post=2,
+                                 coarse_tol=1e-2, coarse_maxiter=100),
+                  vectors=vecs_l)
+
+        # (a) coarse tensors: local X/Y equal the global slice
+        cd_g = mg_g.transfer.coarse_dims
+        cd_l = mg_l.transfer.coarse_dims
+        cidx = torch.arange(cd_l[0] * cd_l[1] * cd_l[2] * cd_l[3])
+        cc = torch.stack([cidx % cd_l[0],
+                          (cidx // cd_l[0]) % cd_l[1],
+                          (cidx // (cd_l[0] * cd_l[1])) % cd_l[2],
+                          cidx // (cd_l[0] * cd_l[1] * cd_l[2])], dim=1)
+        coff = torch.tensor([coords[i] * cd_l[i] for i in range(4)])
+        gc = cc + coff
+        c_glex = (((gc[:, 3] * cd_g[2] + gc[:, 2]) * cd_g[1] + gc[:, 1])
+                  * cd_g[0] + gc[:, 0])
+        errX = (mg_l.coarse.X - mg_g.coarse.X[c_glex]).abs().max().item()
+        assert errX < 1e-11, f"rank{rank} X err {errX}"
+        for dd in range(8):
+            errY = (mg_l.coarse.Y[dd]
+                    - mg_g.coarse.Y[dd][c_glex]).abs().max().item()
+            assert errY < 1e-11, f"rank{rank} Y[{dd}] err {errY}"
+
+        # (b) coarse apply + dagger match the global apply's local slice
+        genc = torch.Generator().manual_seed(97)
+        cg_vec = torch.view_as_complex(
+            torch.randn((mg_g.coarse.Na, mg_g.coarse.Nc, 2), generator=genc,
+                        dtype=torch.float64))
+        cl_vec = cg_vec[c_glex].clone()
+        for dag in (False, True):
+            with comms.solo_mode():
+                want = mg_g.coarse.apply(cg_vec, dagger=dag)
+            got = mg_l.coarse.apply(cl_vec, dagger=dag)
+            err = (got - want[c_glex]).abs().max().item()
+            assert err < 1e-11, f"rank{rank} apply dag={dag} err {err}"
+
+        # (c) full V-cycle matches the single-rank V-cycle
+        genr = torch.Generator().manual_seed(98)
+        r_lex = torch.view_as_complex(
+            torch.randn((gg.volume, 4, 3, 2), generator=genr,
+                        dtype=torch.float64))
+        r_g = SpinorField(gg, "double").from_complex(
+            checkerboard_split(r_lex, gg))
+        r_l = SpinorField(lg, "double").from_complex(
+            checkerboard_split(local_lex(r_lex), lg))
+        z_g = SpinorField(gg, "double")
+        z_l = SpinorField(lg, "double")
+        with comms.solo_mode():
+            mg_g.precond(z_g, r_g)
+        mg_l.precond(z_l, r_l)
+        zg_lex = checkerboard_join(z_g.to_complex(), gg)
+        err = (z_l.to_complex()
+               - checkerboard_split(local_lex(zg_lex), lg)).abs().max().item()
+        assert err < 1e-8, f"rank{rank} V-cycle err {err}"
+
+        # (d) 3-level: level-2 Galerkin tensors match the global build
+        from quda_amd.mg.coarse_level import (CoarseTransfer,
+                                              build_coarse2_op,
+                                              generate_coarse_null_vectors)
+        with comms.solo_mode():
+            v2_g = generate_coarse_null_vectors(mg_g.coarse, 2, tol=1e-2,
+                                                maxiter=60, seed=700)
+            cmg_g = CoarseMG(mg_g.coarse, block2=(2, 2, 2, 2), n_vec2=2,
+                             vectors=v2_g)
+        v2_l = [v[c_glex].clone() for v in v2_g]
+        cmg_l = CoarseMG(mg_l.coarse, block2=(2, 2, 2, 2), n_vec2=2,
+                         vectors=v2_l)
+        cd2_g, cd2_l = cmg_g.t2.cd2, cmg_l.t2.cd2
+        c2idx = torch.arange(cd2_l[0] * cd2_l[1] * cd2_l[2] * cd2_l[3])
+        c2c = torch.stack([c2idx % cd2_l[0],
+                           (c2idx // cd2_l[0]) % cd2_l[1],
+                           (c2idx // (cd2_l[0] * cd2_l[1])) % cd2_l[2],
+                           c2idx // (cd2_l[0] * cd2_l[1] * cd2_l[2])], dim=1)
+        c2off = torch.tensor([coords[i] * cd2_l[i] for i in range(4)])
+        g2 = c2c + c2off
+        c2_glex = (((g2[:, 3] * cd2_g[2] + g2[:, 2]) * cd2_g[1] + g2[:, 1])
+                   * cd2_g[0] + g2[:, 0])
+        errX2 = (cmg_l.co2.X - cmg_g.co2.X[c2_glex]).abs().max().item()
+        assert errX2 < 1e-10, f"rank{rank} X2 err {errX2}"
+        for dd in range(8):
+            errY2 = (cmg_l.co2.Y[dd]
+                     - cmg_g.co2.Y[dd][c2_glex]).abs().max().item()
+            assert errY2 < 1e-10, f"rank{rank} Y2[{dd}] err {errY2}"
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("grid,world", [((1, 1, 1, 2), 2)])
+def test_multirank_mg_matches_global(grid, world):
+    import os
+    import tempfile
+
+    import torch.multiprocessing as mp
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        init_file = f.name
+    os.unlink(init_file)
+    mp.spawn(_mg_worker, args=(world, grid, init_file), nprocs=world,
+             join=True)
