@@ -31,6 +31,7 @@ KERNEL_SOURCES = [
     "dslash_staggered.hip",
     "dslash_dwf.hip",
     "dslash_wilson_mrhs.hip",
+    "coarse.hip",
 ]
 BINDING_SOURCES = ["bindings.cpp"]
 

@@ -426,6 +426,22 @@ static void set_dslash_waves(int64_t w) {
   if (w == 0 || w == 3) qa_dslash_waves_ref() = (int)w;
 }
 
+static void coarse_dslash_mfma(at::Tensor mats, at::Tensor nbr9,
+                               at::Tensor c, at::Tensor out, int64_t Na,
+                               int64_t Nc, int64_t NR) {
+  TORCH_CHECK(mats.scalar_type() == at::kComplexFloat &&
+              c.scalar_type() == at::kComplexFloat &&
+              out.scalar_type() == at::kComplexFloat);
+  TORCH_CHECK(nbr9.scalar_type() == at::kLong);
+  TORCH_CHECK(mats.is_contiguous() && nbr9.is_contiguous() &&
+              c.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(Nc % 16 == 0 && NR >= 1 && NR <= 16);
+  CoarseMfmaCall cc{mats.data_ptr(), nbr9.data_ptr(), c.data_ptr(),
+                    out.data_ptr(), Na, (int)Nc, (int)NR};
+  launch_coarse_dslash_mfma(cc, stream());
+  check_launch("coarse_dslash_mfma");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dslash_wilson", &dslash_wilson, "Wilson(-clover/-twisted) dslash",
         py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
@@ -444,6 +460,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
   m.def("dslash_wilson_mrhs", &dslash_wilson_mrhs,
         "multi-RHS Wilson(-clover) dslash: NRHS sides per gauge load");
+  m.def("coarse_dslash_mfma", &coarse_dslash_mfma,
+        "coarse-grid 9-matrix dslash on f32 MFMA tiles");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("set_dslash_waves", &set_dslash_waves,

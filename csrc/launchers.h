@@ -198,6 +198,19 @@ struct Dwf5Call {
 };
 void launch_dwf5(const Dwf5Call &c, hipStream_t st);
 
+// ---------------------------------------------------------------------------
+// coarse-grid dslash on MFMA (csrc/coarse.hip)
+struct CoarseMfmaCall {
+  const void *mats;  // [9][Na][Nc][Nc] complex64
+  const void *nbr9;  // [Na][9] int64 source sites (>=Na -> ghost rows)
+  const void *c;     // [Na + n_ghost][Nc][NR] complex64
+  void *out;         // [Na][Nc][NR] complex64
+  long Na;
+  int Nc;            // multiple of 16
+  int NR;            // <= 16
+};
+void launch_coarse_dslash_mfma(const CoarseMfmaCall &c, hipStream_t st);
+
 #define QA_ZMAX 32
 
 struct ZCoef {  // host-filled complex coefficient tables (double re/im)

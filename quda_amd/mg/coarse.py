@@ -157,8 +157,67 @@ class CoarseOp:
             cn[self.face[(mu, fwd)]] = ghosts[(mu, fwd)]
         return cn
 
+    # -- MFMA fast path (GPU): one fused 9-matrix kernel ------------------
+    def _hip_setup(self):
+        """Lazy: stacked [9,Na,Nc,Nc] c64 matrices + ghost-extended nbr
+        table for k_coarse_dslash_mfma (csrc/coarse.hip)."""
+        if getattr(self, "_m9", None) is not None:
+            return
+        dev = self.X.device
+        self._m9 = torch.stack([self.X] + list(self.Y)).to(
+            torch.complex64).contiguous()
+        nbr9 = torch.empty((self.Na, 9), dtype=torch.int64, device=dev)
+        nbr9[:, 0] = torch.arange(self.Na, device=dev)
+        nbr9[:, 1:] = self.nbr
+        self._ghost_order = []  # [(mu, fwd, face_len)] in c_ext append order
+        off = self.Na
+        for d in range(8):
+            mu, fwd = d // 2, d % 2
+            if self.mask and (self.mask >> mu) & 1:
+                f = self.face[(mu, fwd)]
+                nbr9[f, 1 + d] = off + torch.arange(len(f), device=dev)
+                self._ghost_order.append((mu, fwd, len(f)))
+                off += len(f)
+        self._n_ghost = off - self.Na
+        self._nbr9 = nbr9.contiguous()
+
+    def _apply_hip(self, c: torch.Tensor, ghosts) -> torch.Tensor:
+        from ..ops.dispatch import hip_ext
+        self._hip_setup()
+        nr = 1 if c.dim() == 2 else c.shape[2]
+        cb = c.reshape(self.Na, self.Nc, nr).to(torch.complex64)
+        if self._n_ghost:
+            gs = [ghosts[(mu, fwd)].reshape(n, self.Nc, nr)
+                  for mu, fwd, n in self._ghost_order]
+            cb = torch.cat([cb] + [g.to(torch.complex64) for g in gs])
+        cb = cb.contiguous()
+        out = torch.empty((self.Na, self.Nc, nr), dtype=torch.complex64,
+                          device=c.device)
+        hip_ext().coarse_dslash_mfma(self._m9, self._nbr9, cb, out,
+                                     self.Na, self.Nc, nr)
+        out = out.to(c.dtype)
+        return out.reshape(self.Na, self.Nc) if c.dim() == 2 else out
+
+    def apply_block(self, C: torch.Tensor) -> torch.Tensor:
+        """Forward apply on an RHS block [Na, Nc, NR] (multi-RHS coarse
+        dslash — the reference's dslash_coarse_mma shape)."""
+        ghosts = self._exchange_c(C)
+        if (C.device.type == "cuda" and self.Nc % 16 == 0
+                and C.shape[2] <= 16 and getattr(self, "use_hip", True)):
+            return self._apply_hip(C, ghosts)
+        out = torch.einsum("aij,ajn->ain", self.X, C)
+        for d in range(8):
+            out += torch.einsum("aij,ajn->ain", self.Y[d],
+                                self._nbr_c(C, d, ghosts))
+        return out
+
     def apply(self, c: torch.Tensor, dagger: bool = False) -> torch.Tensor:
+        if c.dim() == 3 and not dagger:
+            return self.apply_block(c)
         ghosts = self._exchange_c(c)
+        if (not dagger and c.device.type == "cuda" and self.Nc % 16 == 0
+                and getattr(self, "use_hip", True)):
+            return self._apply_hip(c, ghosts)
         if not dagger:
             out = torch.einsum("aij,aj->ai", self.X, c)
             for d in range(8):
