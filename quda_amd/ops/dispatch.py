@@ -122,6 +122,18 @@ def norm_or_empty(f: Optional[SpinorField]):
 
 from ..fields.gauge import RECON_COMPS
 
+_AUTOTUNE = None
+
+
+def _autotune_on() -> bool:
+    """Dispatch-level autotuning (QUDA_AMD_AUTOTUNE=0 disables; on by
+    default like the reference's tuneLaunch)."""
+    global _AUTOTUNE
+    if _AUTOTUNE is None:
+        import os
+        _AUTOTUNE = os.environ.get("QUDA_AMD_AUTOTUNE", "1") != "0"
+    return _AUTOTUNE
+
 
 def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                   parity: int, dagger: bool = False, mode: int = PLAIN,
@@ -155,6 +167,16 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                 mask if kt else 0, kt, float(twist[0]), float(twist[1]))
 
         if not mask:
+            # autotuned workgroup size (ref: lib/tune.cpp tuneLaunch —
+            # first call per (kernel, dims, precision, mode) key measures
+            # the candidates, later calls hit the cache; the dslash
+            # overwrites `out`, so re-running it during tuning is safe)
+            if _autotune_on():
+                from ..utils.tune import tune_dslash
+                key = (f"dslash_wilson/{'x'.join(map(str, geo.dims))}/"
+                       f"{inp.precision}/r{RECON_COMPS[gauge.reconstruct]}"
+                       f"/m{mode}/x{int(xpay)}")
+                tune_dslash(lambda: launch(0), key)
             launch(0)
             return out
         # comm-overlap policy (role of ref lib/dslash_policy.hpp): pack ->
@@ -163,17 +185,29 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
         # the transfers -> EXTERIOR adds ghost hops. "fused" = blocking.
         from ..parallel.halo import get_spinor_halo
         h = get_spinor_halo(geo, inp.precision, inp.device, mask)
-        h.pack(ext, inp, 1 - parity, bool(dagger))
-        ghosts, nrms, face_cb = h.ghost_args()
-        if dslash_policy() == "fused":
-            h.exchange()
-            launch(1, ghosts, nrms, face_cb)
-        else:
-            reqs = h.exchange_start()
-            launch(2, ghosts, nrms, face_cb)   # interior
-            for r in reqs:
-                r.wait()
-            launch(3, ghosts, nrms, face_cb)   # exterior
+
+        def run_halo():
+            h.pack(ext, inp, 1 - parity, bool(dagger))
+            ghosts, nrms, face_cb = h.ghost_args()
+            if dslash_policy() == "fused":
+                h.exchange()
+                launch(1, ghosts, nrms, face_cb)
+            else:
+                reqs = h.exchange_start()
+                launch(2, ghosts, nrms, face_cb)   # interior
+                for r in reqs:
+                    r.wait()
+                launch(3, ghosts, nrms, face_cb)   # exterior
+
+        if _autotune_on():
+            # policy-level autotune (ref DslashPolicyTune): every rank
+            # reaches this collectively, so candidate runs stay in
+            # lockstep; rank-0's winner is broadcast by the tuner
+            from ..utils.tune import tune_dslash_policy
+            key = (f"dslash_policy/{'x'.join(map(str, geo.dims))}/"
+                   f"{inp.precision}/mask{mask}")
+            tune_dslash_policy(run_halo, key)
+        run_halo()
         return out
     # ---- oracle path ----
     u = gauge.to_complex()

@@ -74,7 +74,11 @@ class Tuner:
             if dt < best_t:
                 best, best_t = cfg, dt
         self.cache[key] = (best, best_t)
+        # rank-0's winner becomes THE winner everywhere (launch configs
+        # must agree grid-wide, ref broadcastTuneCache): re-read after
+        # the broadcast before applying
         self._broadcast()
+        best = self.cache[key][0]
         self._save()
         setup(best)
         return best
