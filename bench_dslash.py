@@ -115,6 +115,8 @@ def main():
         results.update(bench_mobius(geo, u, max(args.reps // 5, 3)))
     if "dwf5" in fams:
         results.update(bench_dwf5(geo, max(args.reps, 20)))
+    if "mrhs" in fams:
+        results.update(bench_mrhs(geo, u, max(args.reps // 2, 10)))
     print(json.dumps({k: round(v) for k, v in results.items()}))
 
 
@@ -236,3 +238,36 @@ def bench_mobius(geo, u, reps, Ls=12):
 
 if __name__ == "__main__":
     main()
+
+
+def bench_mrhs(geo, u, reps):
+    """Kernel-level multi-RHS amortization: 8 RHS as groups of 4/2 vs
+    per-RHS launches (gauge+clover loads shared across the group)."""
+    import quda_amd.ops.dispatch as dsp
+    from quda_amd.ops.dispatch import dslash_wilson_batch
+    dev = "cuda"
+    out = {}
+    n = 8
+    for prec in ("half", "single"):
+        g = GaugeField(geo, prec, dev, reconstruct="twelve").from_complex(u)
+        inps = [SpinorField(geo, prec, dev, n_parity=1).gaussian_(seed=40 + r)
+                for r in range(n)]
+        outs = [SpinorField(geo, prec, dev, n_parity=1) for _ in range(n)]
+        for gsz in (1, 2, 4):
+            dsp._MRHS_GROUP = gsz
+            for _ in range(3):
+                dslash_wilson_batch(outs, inps, g, 0)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(reps):
+                dslash_wilson_batch(outs, inps, g, 0)
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / reps
+            gf = n * geo.volume_cb * 1320 / dt / 1e9
+            key = f"mrhs{gsz}/{prec}"
+            out[key] = round(gf)
+            print(f"{key:32s} {dt*1e6:8.1f} us ({n} rhs) {gf:8.0f} GFLOPS",
+                  flush=True)
+        dsp._MRHS_GROUP = None
+    print(json.dumps(out))
+    return out

@@ -245,16 +245,34 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
     return out
 
 
+_MRHS_GROUP = None
+
+
+def mrhs_group_size() -> int:
+    """Preferred kernel NRHS (QUDA_AMD_MRHS_GROUP: 4, 2, or 1=disable).
+    Default 2: the NRHS=4 instantiation spills (1-5 KB/lane scratch at
+    256-VGPR cap) and measures slower than 2x NRHS=2."""
+    global _MRHS_GROUP
+    if _MRHS_GROUP is None:
+        import os
+        v = int(os.environ.get("QUDA_AMD_MRHS_GROUP", "2"))
+        _MRHS_GROUP = v if v in (1, 2, 4) else 2
+    return _MRHS_GROUP
+
+
 def _mrhs_groups(n: int):
-    """Split n RHS into kernel-supported group sizes (4, then 2; odd
-    remainder runs the per-RHS kernel)."""
+    """Split n RHS into kernel-supported group sizes; the remainder runs
+    the per-RHS kernel."""
+    g = mrhs_group_size()
     groups, r0 = [], 0
-    while n - r0 >= 4:
-        groups.append((r0, 4))
-        r0 += 4
-    if n - r0 >= 2:
-        groups.append((r0, 2))
-        r0 += 2
+    if g >= 4:
+        while n - r0 >= 4:
+            groups.append((r0, 4))
+            r0 += 4
+    if g >= 2:
+        while n - r0 >= 2:
+            groups.append((r0, 2))
+            r0 += 2
     return groups, r0
 
 

@@ -312,6 +312,11 @@ def test_dslash_mrhs_vs_per_rhs(setup, prec, nrhs):
             0.5 * v))
         outs_m.append(SpinorField(geo, prec, "cuda", n_parity=1))
         outs_1.append(SpinorField(geo, prec, "cuda", n_parity=1))
+    # LSB-level fp-contraction differences between the two template
+    # instantiations are expected (the compiler fuses mul+add chains
+    # differently per kernel); require agreement at the precision's
+    # roundoff, not bitwise.
+    rtol = {"half": 2e-3, "single": 1e-5, "double": 1e-13}[prec]
     for mode, use_x, use_cl in [(PLAIN, False, None), (PLAIN, True, None),
                                 (CLOV_POST, True, cl)]:
         for dag in (False, True):
@@ -322,6 +327,7 @@ def test_dslash_mrhs_vs_per_rhs(setup, prec, nrhs):
                 dslash_wilson(outs_1[r], inps[r], gd, 0, dagger=dag,
                               a=-0.3, x=xs[r] if use_x else None,
                               mode=mode, clover=use_cl)
-                dm = (outs_m[r].to_complex() - outs_1[r].to_complex())
-                err = dm.abs().max().item()
-                assert err == 0.0, (prec, nrhs, mode, use_x, dag, r, err)
+                ref_c = outs_1[r].to_complex()
+                dm = (outs_m[r].to_complex() - ref_c)
+                err = dm.abs().max().item() / ref_c.abs().max().item()
+                assert err < rtol, (prec, nrhs, mode, use_x, dag, r, err)
