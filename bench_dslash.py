@@ -236,8 +236,7 @@ def bench_mobius(geo, u, reps, Ls=12):
         print(f"{key:32s} {dt*1e6:8.1f} us  {gf:8.0f} GFLOPS", flush=True)
     return out
 
-if __name__ == "__main__":
-    main()
+
 
 
 def bench_mrhs(geo, u, reps):
@@ -271,3 +270,7 @@ def bench_mrhs(geo, u, reps):
         dsp._MRHS_GROUP = None
     print(json.dumps(out))
     return out
+
+
+if __name__ == "__main__":
+    main()
