@@ -110,6 +110,8 @@ class InvertParam:
     b5: float = 1.5
     c5: float = 0.5
     clover_csw: float = 0.0
+    distance_pc_alpha0: float = 0.0  # ref quda.h:464 (arXiv:1006.4028)
+    distance_pc_t0: int = 0
     tol: float = 1e-8
     maxiter: int = 1000
     reliable_delta: float = 0.1
@@ -266,6 +268,20 @@ def _make_dirac(p: InvertParam, sloppy: bool = False):
     cl = _R.clover_sloppy if sloppy else _R.clover
     pc = p.solution_type == SolutionType.MATPC
     t = p.dslash_type
+    if p.distance_pc_alpha0 != 0.0 and t in (DslashType.WILSON,
+                                             DslashType.CLOVER):
+        # distance preconditioning (ref: lib/dslash_wilson_distance.cu)
+        from .models.distance import (DiracCloverDistance,
+                                      DiracCloverDistancePC,
+                                      DiracWilsonDistance,
+                                      DiracWilsonDistancePC)
+        kw = dict(alpha0=p.distance_pc_alpha0, t0=p.distance_pc_t0)
+        if t == DslashType.WILSON:
+            return (DiracWilsonDistancePC(g, p.kappa, **kw) if pc
+                    else DiracWilsonDistance(g, p.kappa, **kw))
+        assert cl is not None, "load_clover_quda first"
+        return (DiracCloverDistancePC(g, cl, p.kappa, **kw) if pc
+                else DiracCloverDistance(g, cl, p.kappa, **kw))
     if t == DslashType.WILSON:
         return DiracWilsonPC(g, p.kappa) if pc else DiracWilson(g, p.kappa)
     if t == DslashType.CLOVER:
@@ -382,8 +398,10 @@ def invert_quda(b: torch.Tensor, p: InvertParam,
     d = _make_dirac(p)
     b_f = _wrap(b, p, 2)
     pc_capable = hasattr(d, "prepare")
+    distw = getattr(d, "distance", None)
 
     if p.solution_type == SolutionType.MATPC and pc_capable:
+        # distance ops fold the P/P^-1 weights into prepare/reconstruct
         src = d.prepare(b_f)
         x_e = src.clone_empty()
         stats = _run_solver(d, x_e, src, p, sloppy_pair=True)
@@ -392,15 +410,22 @@ def invert_quda(b: torch.Tensor, p: InvertParam,
     else:
         x_f = (_wrap(x0, p, 2) if x0 is not None
                else b_f.clone_empty())
+        if distw is not None:
+            distw.apply(b_f, inverse=True)   # solve M_dist x' = P^-1 b
         stats = _run_solver(d, x_f, b_f, p, sloppy_pair=False)
+        if distw is not None:
+            distw.apply(x_f)                 # x = P x'
+            distw.apply(b_f)                 # restore
 
-    # true residual on the requested system
+    # true residual on the requested (PHYSICAL) system: the plain operator
     r = b_f.clone_empty()
-    full = _make_dirac(InvertParam(**{**p.__dict__,
+    plain = {**p.__dict__, "distance_pc_alpha0": 0.0}
+    full = _make_dirac(InvertParam(**{**plain,
                                       "solution_type": SolutionType.MAT})) \
-        if p.solution_type == SolutionType.MATPC else d
+        if p.solution_type == SolutionType.MATPC else (
+            _make_dirac(InvertParam(**plain)) if distw is not None else d)
     full_b = b_f
-    full.M(r, x_f) if p.solution_type == SolutionType.MATPC else d.M(r, x_f)
+    full.M(r, x_f)
     import math
     p.true_res = math.sqrt(blas.xmy_norm2(full_b, r) / blas.norm2(full_b))
     p.iter = stats.iters
