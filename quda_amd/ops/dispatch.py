@@ -124,6 +124,10 @@ from ..fields.gauge import RECON_COMPS
 
 _AUTOTUNE = None
 
+# hook: zero-arg callable executed inside the dslash comms window (ref:
+# dslash::aux_worker, lib/dslash_quda.cu:73 / inv_multi_cg_quda.cpp:115)
+aux_worker = None
+
 
 def _autotune_on() -> bool:
     """Dispatch-level autotuning (QUDA_AMD_AUTOTUNE=0 disables; on by
@@ -195,6 +199,11 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
             else:
                 reqs = h.exchange_start()
                 launch(2, ghosts, nrms, face_cb)   # interior
+                if aux_worker is not None:
+                    # independent work queued into the comms window (ref:
+                    # dslash::aux_worker lib/dslash_quda.cu:73 — the
+                    # multishift p-updates ride here)
+                    aux_worker()
                 for r in reqs:
                     r.wait()
                 launch(3, ghosts, nrms, face_cb)   # exterior

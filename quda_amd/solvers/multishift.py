@@ -53,8 +53,26 @@ def multishift_cg_solve(op, xs: List[SpinorField], b: SpinorField,
     alpha_prev = 1.0       # base alpha_{k-1} (init per Jegerlehner)
     beta_prev = 0.0        # base beta_{k-1}
     k = 0
+    # Shifted p-updates are DEFERRED into the next MdagM's comms window
+    # (ref: inv_multi_cg_quda.cpp:115 ShiftUpdate via dslash::aux_worker):
+    # they touch only ps[i>=1], which nothing reads until the following
+    # x-update, so they legally overlap the halo wait.
+    pending = [None]
+
+    def _aux():
+        if pending[0] is not None:
+            fn, pending[0] = pending[0], None
+            fn()
+
+    from ..ops import dispatch as _dsp
+
     while r2 > stop and k < maxiter:
-        op.MdagM(Ap, ps[0], tmp)
+        _dsp.aux_worker = _aux
+        try:
+            op.MdagM(Ap, ps[0], tmp)
+        finally:
+            _dsp.aux_worker = None
+        _aux()  # no comms window ran it (local / fused policy)
         if shifts[0] != 0.0:
             blas.axpy(shifts[0], ps[0], Ap)
         pAp = blas.re_dot(ps[0], Ap)
@@ -82,20 +100,29 @@ def multishift_cg_solve(op, xs: List[SpinorField], b: SpinorField,
             if not converged[i]:
                 blas.axpy(alpha_s[i], ps[i], xs[i])
         beta0 = r2 / r2_old
-        # p updates
+        # p updates: base now, shifted ones deferred to the next comms
+        # window (captured state: this iteration's zeta/beta coefficients)
         blas.xpay(r, beta0, ps[0])
+        updates = []
         for i in range(1, n):
             if converged[i]:
                 continue
             ratio = zeta_next[i] / zeta[i]
             beta_s = beta0 * ratio * ratio
-            blas.caxpby(zeta_next[i], r, beta_s, ps[i])
+            updates.append((i, zeta_next[i], beta_s))
             if zeta_next[i] * zeta_next[i] * r2 < stop:
                 converged[i] = True
             zeta_old[i], zeta[i] = zeta[i], zeta_next[i]
+
+        def _shift_p_updates(us=updates):
+            for i, zn, bs in us:
+                blas.caxpby(zn, r, bs, ps[i])
+
+        pending[0] = _shift_p_updates
         alpha_prev, beta_prev = alpha0, beta0
         k += 1
 
+    pending[0] = None  # trailing p-update is never consumed
     stats.iters = k
     stats.resid = sqrt(r2 / b2)
     stats.converged = r2 <= stop
