@@ -132,9 +132,10 @@ def build_c_api(force: bool = False, verbose: bool = True) -> str:
     py_inc = sysconfig.get_paths()["include"]
     libdir = sysconfig.get_config_var("LIBDIR")
     pyver = f"python{sys.version_info.major}.{sys.version_info.minor}"
+    fsrc = os.path.join(CSRC, "quda_fortran_api.cpp")
     cmd = ["g++", "-O2", "-std=c++17", "-fPIC", "-shared",
            f"-I{pybind11.get_include()}", f"-I{py_inc}",
-           src, "-o", C_API_SO, f"-L{libdir}", f"-l{pyver}"]
+           src, fsrc, "-o", C_API_SO, f"-L{libdir}", f"-l{pyver}"]
     r = subprocess.run(cmd, capture_output=True, text=True)
     if r.returncode != 0:
         raise RuntimeError(f"C API build failed:\n{r.stdout}\n{r.stderr}")

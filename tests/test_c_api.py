@@ -46,3 +46,19 @@ def test_c_interface_gpu(c_test_bin):
     r = _run(c_test_bin, "cuda:0")
     assert r.returncode == 0, (r.stdout, r.stderr)
     assert "ALL PASSED" in r.stdout
+
+
+def test_fortran_interface_symbols(c_test_bin, tmp_path):
+    """The trailing-underscore Fortran entry points (csrc/
+    quda_fortran_api.cpp, role of lib/quda_fortran.F90) drive a full
+    solve through pass-by-reference calls."""
+    exe = str(tmp_path / "fortran_interface_test")
+    src = os.path.join(ROOT, "tests", "c_api", "fortran_interface_test.c")
+    r = subprocess.run(
+        ["gcc", "-std=c99", "-O2", src, "-o", exe, f"-L{ROOT}",
+         "-lquda_amd_c", f"-Wl,-rpath,{ROOT}", "-lm"],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = _run(exe, "cpu")
+    assert out.returncode == 0, (out.returncode, out.stdout, out.stderr)
+    assert "ALL PASSED" in out.stdout
