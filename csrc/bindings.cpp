@@ -112,17 +112,22 @@ static void dslash_wilson_mrhs(
     int64_t parity_offset, int64_t Vcb, int64_t parity, bool dagger,
     int64_t mode, bool xpay, double a, int64_t recon,
     std::vector<at::Tensor> ghost, std::vector<at::Tensor> ghost_nrm,
-    std::vector<int64_t> face_cb, int64_t comm_mask, int64_t kt) {
+    std::vector<int64_t> face_cb, int64_t comm_mask, int64_t kt,
+    int64_t v_stride, std::vector<int64_t> s_offsets) {
+  // v_stride/s_offsets: address 4-d slices of 5-d fields (domain-wall
+  // s-batching — the same tensor appears n times with different offsets)
   int n = (int)out.size();
   TORCH_CHECK(n == 2 || n == 4, "mrhs kernel supports 2 or 4 RHS");
   TORCH_CHECK((int)in.size() == n);
+  if (v_stride == 0) v_stride = Vcb;
   DslashMrhsCall c{};
   c.nrhs = n;
   for (int r = 0; r < n; ++r) {
     TORCH_CHECK(out[r].is_contiguous() && in[r].is_contiguous());
-    c.out[r] = field_of(out[r], out_n[r], Vcb);
-    c.in[r] = field_of(in[r], in_n[r], Vcb);
-    if (!x.empty()) c.x[r] = field_of(x[r], x_n[r], Vcb);
+    long so = s_offsets.empty() ? 0 : s_offsets[r];
+    c.out[r] = field_of_off(out[r], out_n[r], v_stride, so);
+    c.in[r] = field_of_off(in[r], in_n[r], v_stride, so);
+    if (!x.empty()) c.x[r] = field_of_off(x[r], x_n[r], v_stride, so);
   }
   c.comm_mask = (int)comm_mask;
   c.kt = (int)kt;
@@ -459,7 +464,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("Fcb"), py::arg("v_stride") = 0, py::arg("s_offset") = 0);
   m.def("pack_face_stag", &pack_face_stag, "staggered halo face pack");
   m.def("dslash_wilson_mrhs", &dslash_wilson_mrhs,
-        "multi-RHS Wilson(-clover) dslash: NRHS sides per gauge load");
+        "multi-RHS Wilson(-clover) dslash: NRHS sides per gauge load",
+        py::arg("out"), py::arg("out_n"), py::arg("in"), py::arg("in_n"),
+        py::arg("gauge"), py::arg("clover"), py::arg("x"), py::arg("x_n"),
+        py::arg("dims"), py::arg("parity_offset"), py::arg("Vcb"),
+        py::arg("parity"), py::arg("dagger"), py::arg("mode"),
+        py::arg("xpay"), py::arg("a"), py::arg("recon"), py::arg("ghost"),
+        py::arg("ghost_nrm"), py::arg("face_cb"), py::arg("comm_mask"),
+        py::arg("kt"), py::arg("v_stride") = 0,
+        py::arg("s_offsets") = std::vector<int64_t>{});
   m.def("coarse_dslash_mfma", &coarse_dslash_mfma,
         "coarse-grid 9-matrix dslash on f32 MFMA tiles");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");

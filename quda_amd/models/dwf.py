@@ -73,13 +73,13 @@ class DiracMobius:
                        dagger=dagger)
 
     def _dhat(self, out, inp, parity, dagger=False, a=1.0, xpay=True):
-        """out[s] = [out[s] +] a * Dhat in[s] for every slice (one halo
-        exchange for all slices when dims are partitioned)."""
-        from ..ops.dispatch import dwf_halo_exchange
+        """out[s] = [out[s] +] a * Dhat in[s] for every slice: one halo
+        exchange for all slices + s-batched multi-RHS kernels sharing the
+        gauge loads (dispatch.dslash_wilson_slices)."""
+        from ..ops.dispatch import dslash_wilson_slices, dwf_halo_exchange
         h = dwf_halo_exchange(inp, 1 - parity, dagger)
-        for s in range(self.Ls):
-            dslash_wilson_slice(out, inp, self.gauge, parity, s, dagger,
-                                a=a, x=out if xpay else None, halo=h)
+        dslash_wilson_slices(out, inp, self.gauge, parity, dagger, a=a,
+                             x=out if xpay else None, halo=h)
         return out
 
     # -- full operator ------------------------------------------------------

@@ -737,6 +737,50 @@ def dwf_halo_exchange(inp: SpinorField, parity_in: int, dagger: bool):
     return ("oracle", mask, ghosts)
 
 
+def dslash_wilson_slices(out: SpinorField, inp: SpinorField,
+                         gauge: GaugeField, parity: int, dagger: bool = False,
+                         a: float = 1.0, x: Optional[SpinorField] = None,
+                         halo=None):
+    """4-d Wilson hop on ALL Ls slices of 5-d fields, s-batched through
+    the multi-RHS kernel: groups of slices share each gauge load (the
+    domain-wall traffic fusion the reference gets from
+    dslash_domain_wall_4d_fused_m5 — on MI355X the gauge stream, not the
+    M5, is the fusible term; the s-batch amortizes it NRHS-fold)."""
+    from ..parallel import comms
+    geo = out.geo
+    mask = comms.comm_mask()
+    Ls = out.ls
+    Vcb = geo.volume_cb
+    gpu = on_gpu(out, inp)
+    groups, rem0 = _mrhs_groups(Ls) if gpu else ([], 0)
+    if gpu and groups:
+        ext = hip_ext()
+        xpay = x is not None
+        xf = x if x is not None else out
+        cl_t = torch.empty(0, dtype=out.data.dtype, device=out.device)
+        for s0, gsz in groups:
+            ghosts, nrms, face_cb, kt = [], [], [], 0
+            if mask:
+                kind, hmask, h = halo
+                assert kind == "native" and hmask == mask
+                ghosts, nrms, face_cb = h.ghost_args(s0)
+                kt = 1
+            ext.dslash_wilson_mrhs(
+                [out.data] * gsz, [norm_or_empty(out)] * gsz,
+                [inp.data] * gsz, [norm_or_empty(inp)] * gsz,
+                gauge.data, cl_t, [xf.data] * gsz if xpay else [],
+                [norm_or_empty(xf)] * gsz if xpay else [],
+                list(geo.dims), geo.parity_offset, Vcb, parity,
+                bool(dagger), PLAIN, xpay, float(a),
+                RECON_COMPS[gauge.reconstruct], ghosts, nrms, face_cb,
+                mask, kt, inp.volume_cb,
+                [(s0 + r) * Vcb for r in range(gsz)])
+    for s in range(rem0, Ls):
+        dslash_wilson_slice(out, inp, gauge, parity, s, dagger, a=a, x=x,
+                            halo=halo)
+    return out
+
+
 def dslash_wilson_slice(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                         parity: int, s: int, dagger: bool = False,
                         a: float = 1.0, x: Optional[SpinorField] = None,
