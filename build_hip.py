@@ -28,6 +28,7 @@ KERNEL_SOURCES = [
     "dslash_wilson_d.hip",
     "dslash_wilson_s.hip",
     "dslash_wilson_h.hip",
+    "dslash_wilson_q.hip",
     "dslash_staggered.hip",
     "dslash_dwf.hip",
     "dslash_wilson_mrhs.hip",

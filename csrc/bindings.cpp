@@ -12,6 +12,7 @@ int prec_of(const at::Tensor &t) {
     case at::kDouble: return 0;
     case at::kFloat: return 1;
     case at::kHalf: return 2;
+    case at::kFloat8_e4m3fn: return 3;
     default: TORCH_CHECK(false, "unsupported dtype");
   }
 }
@@ -100,6 +101,7 @@ static void dslash_wilson(at::Tensor out, at::Tensor out_n, at::Tensor in,
     case 0: launch_dslash_wilson_double(c, stream()); break;
     case 1: launch_dslash_wilson_single(c, stream()); break;
     case 2: launch_dslash_wilson_half(c, stream()); break;
+    case 3: launch_dslash_wilson_quarter(c, stream()); break;
   }
   check_launch("dslash_wilson");
 }
@@ -177,6 +179,7 @@ static void pack_face(at::Tensor dst, at::Tensor dst_nrm, at::Tensor in,
     case 0: launch_pack_face_double(c, stream()); break;
     case 1: launch_pack_face_single(c, stream()); break;
     case 2: launch_pack_face_half(c, stream()); break;
+    case 3: launch_pack_face_quarter(c, stream()); break;
   }
   check_launch("pack_face");
 }

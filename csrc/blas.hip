@@ -279,12 +279,14 @@ void launch_blas(const BlasCall &c, hipStream_t st) {
       case 0: blas_dispatch<StagAcc<PrecDouble>>(c, st); break;
       case 1: blas_dispatch<StagAcc<PrecSingle>>(c, st); break;
       case 2: blas_dispatch<StagAcc<PrecHalf>>(c, st); break;
+      case 3: blas_dispatch<StagAcc<PrecQuarter>>(c, st); break;
     }
   } else {
     switch (c.prec) {
       case 0: blas_dispatch<SpinorAcc<PrecDouble>>(c, st); break;
       case 1: blas_dispatch<SpinorAcc<PrecSingle>>(c, st); break;
       case 2: blas_dispatch<SpinorAcc<PrecHalf>>(c, st); break;
+      case 3: blas_dispatch<SpinorAcc<PrecQuarter>>(c, st); break;
     }
   }
 }
@@ -300,16 +302,23 @@ static void conv(const BlasField &d, const BlasField &s, long sites, hipStream_t
 template <template <typename> class AT>
 static void conv_dispatch(const BlasField &dst, int pdst, const BlasField &src,
                           int psrc, long sites, hipStream_t st) {
-  switch (pdst * 3 + psrc) {
-    case 0 * 3 + 1: conv<AT, PrecDouble, PrecSingle>(dst, src, sites, st); break;
-    case 0 * 3 + 2: conv<AT, PrecDouble, PrecHalf>(dst, src, sites, st); break;
-    case 1 * 3 + 0: conv<AT, PrecSingle, PrecDouble>(dst, src, sites, st); break;
-    case 1 * 3 + 2: conv<AT, PrecSingle, PrecHalf>(dst, src, sites, st); break;
-    case 2 * 3 + 0: conv<AT, PrecHalf, PrecDouble>(dst, src, sites, st); break;
-    case 2 * 3 + 1: conv<AT, PrecHalf, PrecSingle>(dst, src, sites, st); break;
-    case 0 * 3 + 0: conv<AT, PrecDouble, PrecDouble>(dst, src, sites, st); break;
-    case 1 * 3 + 1: conv<AT, PrecSingle, PrecSingle>(dst, src, sites, st); break;
-    case 2 * 3 + 2: conv<AT, PrecHalf, PrecHalf>(dst, src, sites, st); break;
+  switch (pdst * 4 + psrc) {
+    case 0 * 4 + 1: conv<AT, PrecDouble, PrecSingle>(dst, src, sites, st); break;
+    case 0 * 4 + 2: conv<AT, PrecDouble, PrecHalf>(dst, src, sites, st); break;
+    case 1 * 4 + 0: conv<AT, PrecSingle, PrecDouble>(dst, src, sites, st); break;
+    case 1 * 4 + 2: conv<AT, PrecSingle, PrecHalf>(dst, src, sites, st); break;
+    case 2 * 4 + 0: conv<AT, PrecHalf, PrecDouble>(dst, src, sites, st); break;
+    case 2 * 4 + 1: conv<AT, PrecHalf, PrecSingle>(dst, src, sites, st); break;
+    case 0 * 4 + 0: conv<AT, PrecDouble, PrecDouble>(dst, src, sites, st); break;
+    case 1 * 4 + 1: conv<AT, PrecSingle, PrecSingle>(dst, src, sites, st); break;
+    case 2 * 4 + 2: conv<AT, PrecHalf, PrecHalf>(dst, src, sites, st); break;
+    case 0 * 4 + 3: conv<AT, PrecDouble, PrecQuarter>(dst, src, sites, st); break;
+    case 1 * 4 + 3: conv<AT, PrecSingle, PrecQuarter>(dst, src, sites, st); break;
+    case 2 * 4 + 3: conv<AT, PrecHalf, PrecQuarter>(dst, src, sites, st); break;
+    case 3 * 4 + 0: conv<AT, PrecQuarter, PrecDouble>(dst, src, sites, st); break;
+    case 3 * 4 + 1: conv<AT, PrecQuarter, PrecSingle>(dst, src, sites, st); break;
+    case 3 * 4 + 2: conv<AT, PrecQuarter, PrecHalf>(dst, src, sites, st); break;
+    case 3 * 4 + 3: conv<AT, PrecQuarter, PrecQuarter>(dst, src, sites, st); break;
   }
 }
 

@@ -10,6 +10,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
+#include <hip/hip_fp8.h>
 
 // ---------------------------------------------------------------------------
 // complex
@@ -51,9 +52,33 @@ __device__ __forceinline__ cplx<T> cfma_conj(const cplx<T> &a, const cplx<T> &b,
 // ---------------------------------------------------------------------------
 // storage traits: Store = memory type, Real = compute type
 // ---------------------------------------------------------------------------
-struct PrecDouble { using Store = double; using Real = double; static constexpr int W = 2;  static constexpr bool has_norm = false; };
-struct PrecSingle { using Store = float;  using Real = float;  static constexpr int W = 4;  static constexpr bool has_norm = false; };
-struct PrecHalf   { using Store = __half; using Real = float;  static constexpr int W = 8;  static constexpr bool has_norm = true;  };
+// distinct 1-byte wrapper so overloads can route fp8-e4m3 (OCP, gfx950
+// hardware-converted) storage through the block-float decode/encode —
+// the "quarter" precision (ref: color_spinor_field_order.h:1426 FLOAT8 /
+// quarter fixed-point, re-based on true fp8 since CDNA4 has native
+// e4m3 converts)
+struct fp8s {
+  unsigned char b;
+};
+
+struct PrecDouble  { using Store = double; using Real = double; static constexpr int W = 2;  static constexpr bool has_norm = false; };
+struct PrecSingle  { using Store = float;  using Real = float;  static constexpr int W = 4;  static constexpr bool has_norm = false; };
+struct PrecHalf    { using Store = __half; using Real = float;  static constexpr int W = 8;  static constexpr bool has_norm = true;  };
+struct PrecQuarter { using Store = fp8s;   using Real = float;  static constexpr int W = 16; static constexpr bool has_norm = true;  };
+
+template <typename R, typename S>
+__device__ __forceinline__ R qa_tor(const S &v) { return (R)v; }
+template <typename R>
+__device__ __forceinline__ R qa_tor(const fp8s &v) {
+  return (R)__half(__hip_cvt_fp8_to_halfraw(v.b, __HIP_E4M3));
+}
+
+template <typename S, typename R>
+__device__ __forceinline__ S qa_tos(const R &v) { return (S)v; }
+template <>
+__device__ __forceinline__ fp8s qa_tos<fp8s, float>(const float &v) {
+  return {__hip_cvt_float_to_fp8(v, __HIP_SATFINITE, __HIP_E4M3)};
+}
 
 // 16-byte opaque chunk for vector loads
 struct alignas(16) chunk16 { unsigned int u[4]; };
@@ -121,7 +146,7 @@ struct SpinorAcc {
     if constexpr (Prec::has_norm) scale = norm[g];
 #pragma unroll
     for (int k = 0; k < NCPLX; ++k)
-      out[k] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+      out[k] = {scale * qa_tor<R>(tmp[2 * k]), scale * qa_tor<R>(tmp[2 * k + 1])};
   }
 
   __device__ __forceinline__ void store_v(const cplx<R> in[NCPLX], long g) const {
@@ -136,8 +161,8 @@ struct SpinorAcc {
       R inv = m > (R)0 ? (R)1 / m : (R)0;
 #pragma unroll
       for (int k = 0; k < NCPLX; ++k) {
-        tmp[2 * k] = (S)(in[k].re * inv);
-        tmp[2 * k + 1] = (S)(in[k].im * inv);
+        tmp[2 * k] = qa_tos<S>(in[k].re * inv);
+        tmp[2 * k + 1] = qa_tos<S>(in[k].im * inv);
       }
     } else {
 #pragma unroll
@@ -203,7 +228,7 @@ struct GaugeAcc {
       load_chunk<S, W>(base + ((long)(c0 + c) * V + i) * W, tmp + c * W);
 #pragma unroll
     for (int k = 0; k < RECON / 2; ++k)
-      u[k / 3][k % 3] = {(R)tmp[off + 2 * k], (R)tmp[off + 2 * k + 1]};
+      u[k / 3][k % 3] = {qa_tor<R>(tmp[off + 2 * k]), qa_tor<R>(tmp[off + 2 * k + 1])};
     if constexpr (RECON == 12) {
 #pragma unroll
       for (int c = 0; c < 3; ++c) {

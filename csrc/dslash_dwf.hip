@@ -33,6 +33,7 @@ void launch_dwf5(const Dwf5Call &c, hipStream_t st) {
     case 0: dwf5_t<PrecDouble>(c, st); break;
     case 1: dwf5_t<PrecSingle>(c, st); break;
     case 2: dwf5_t<PrecHalf>(c, st); break;
+    case 3: dwf5_t<PrecQuarter>(c, st); break;
   }
 }
 

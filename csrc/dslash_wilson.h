@@ -62,10 +62,10 @@ struct CloverAcc {
 #pragma unroll
     for (int b = 0; b < 2; ++b) {
 #pragma unroll
-      for (int k = 0; k < 6; ++k) diag[b][k] = (R)tmp[36 * b + k];
+      for (int k = 0; k < 6; ++k) diag[b][k] = qa_tor<R>(tmp[36 * b + k]);
 #pragma unroll
       for (int k = 0; k < 15; ++k)
-        tri[b][k] = {(R)tmp[36 * b + 6 + 2 * k], (R)tmp[36 * b + 6 + 2 * k + 1]};
+        tri[b][k] = {qa_tor<R>(tmp[36 * b + 6 + 2 * k]), qa_tor<R>(tmp[36 * b + 6 + 2 * k + 1])};
     }
   }
 };

@@ -73,7 +73,7 @@ struct GhostAcc {
     if constexpr (Prec::has_norm) scale = nrm[2 * mu + dir][f];
 #pragma unroll
     for (int k = 0; k < NCPLX; ++k)
-      out[k] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+      out[k] = {scale * qa_tor<R>(tmp[2 * k]), scale * qa_tor<R>(tmp[2 * k + 1])};
   }
 
   // Wilson half-spinor view (NCOMP == 12)
@@ -100,7 +100,7 @@ struct GhostAcc {
     auto *out = reinterpret_cast<cplx<R> *>(h);
 #pragma unroll
     for (int k = 0; k < NCPLX; ++k)
-      out[k] = {scale * (R)tmp[2 * k], scale * (R)tmp[2 * k + 1]};
+      out[k] = {scale * qa_tor<R>(tmp[2 * k]), scale * qa_tor<R>(tmp[2 * k + 1])};
   }
 };
 
@@ -124,8 +124,8 @@ __device__ __forceinline__ void ghost_store_v(
     R inv = m > (R)0 ? (R)1 / m : (R)0;
 #pragma unroll
     for (int k = 0; k < NCPLX; ++k) {
-      tmp[2 * k] = (S)(h[k].re * inv);
-      tmp[2 * k + 1] = (S)(h[k].im * inv);
+      tmp[2 * k] = qa_tos<S>(h[k].re * inv);
+      tmp[2 * k + 1] = qa_tos<S>(h[k].im * inv);
     }
   } else {
 #pragma unroll

@@ -93,6 +93,7 @@ struct DslashCall {
 void launch_dslash_wilson_double(const DslashCall &c, hipStream_t st);
 void launch_dslash_wilson_single(const DslashCall &c, hipStream_t st);
 void launch_dslash_wilson_half(const DslashCall &c, hipStream_t st);
+void launch_dslash_wilson_quarter(const DslashCall &c, hipStream_t st);
 
 // ---------------------------------------------------------------------------
 // multi-RHS Wilson(-clover): NRHS in {2,4} sides share each gauge/clover
@@ -141,6 +142,7 @@ struct PackCall {
 void launch_pack_face_double(const PackCall &c, hipStream_t st);
 void launch_pack_face_single(const PackCall &c, hipStream_t st);
 void launch_pack_face_half(const PackCall &c, hipStream_t st);
+void launch_pack_face_quarter(const PackCall &c, hipStream_t st);
 
 struct CloverApplyCall {
   BlasField out, in;

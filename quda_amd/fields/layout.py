@@ -28,12 +28,14 @@ DTYPE_OF = {
     "double": torch.float64,
     "single": torch.float32,
     "half": torch.float16,
+    "quarter": torch.float8_e4m3fn,
 }
 
 WIDTH_OF = {  # reals per 16-byte chunk
     "double": 2,
     "single": 4,
     "half": 8,
+    "quarter": 16,
 }
 
 

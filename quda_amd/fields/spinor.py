@@ -47,7 +47,7 @@ class SpinorField:
             self.data = data
         else:
             self.data = torch.zeros(shape, dtype=DTYPE_OF[precision], device=device)
-        if precision == "half":
+        if precision in ("half", "quarter"):
             nshape = (n_parity, ls * geo.volume_cb)
             if norm is not None:
                 assert tuple(norm.shape) == nshape
@@ -97,7 +97,7 @@ class SpinorField:
         assert c.shape == (self.n_parity, self.volume_cb, *self.site_shape), \
             (c.shape, self.site_shape)
         flat = c.reshape(self.n_parity, self.volume_cb, self.ncomp // 2)
-        if self.precision == "half":
+        if self.precision in ("half", "quarter"):
             mags = torch.view_as_real(flat).abs().amax(dim=(-1, -2))  # [P,V]
             self.norm.copy_(mags.to(torch.float32))
             scale = torch.where(mags > 0, 1.0 / mags, torch.zeros_like(mags))

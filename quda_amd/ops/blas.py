@@ -17,7 +17,7 @@ from .dispatch import hip_ext, norm_or_empty, on_gpu
 
 
 def _is_half(x: SpinorField) -> bool:
-    return x.precision == "half"
+    return x.precision in ("half", "quarter")
 
 
 def _pairs(t: torch.Tensor) -> torch.Tensor:

@@ -99,7 +99,7 @@ def exchange_tensors_start(sends: Dict[Key, torch.Tensor],
 # native (device-layout) spinor halo: persistent buffers + HIP pack kernels
 # ---------------------------------------------------------------------------
 
-GHOST_W0 = {"double": 2, "single": 4, "half": 4}  # max reals per ghost chunk
+GHOST_W0 = {"double": 2, "single": 4, "half": 4, "quarter": 4}  # max reals per ghost chunk
 
 
 def ghost_width(ncomp: int, precision: str) -> int:
@@ -137,7 +137,7 @@ class SpinorHalo:
                 shape = (ncomp // gw, fcb, gw)
                 self.send[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
                 self.recv[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
-                if precision == "half":
+                if precision in ("half", "quarter"):
                     self.send_nrm[(mu, d)] = torch.empty(fcb, dtype=torch.float32,
                                                          device=device)
                     self.recv_nrm[(mu, d)] = torch.empty(fcb, dtype=torch.float32,
@@ -175,7 +175,7 @@ class SpinorHalo:
 
     def exchange_start(self) -> list:
         reqs = exchange_tensors_start(self.send, self.recv)
-        if self.precision == "half":
+        if self.precision in ("half", "quarter"):
             reqs += exchange_tensors_start(self.send_nrm, self.recv_nrm)
         return reqs
 
@@ -265,7 +265,7 @@ class DwfHalo:
                 shape = (ls, 12 // gw, fcb, gw)
                 self.send[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
                 self.recv[(mu, d)] = torch.empty(shape, dtype=dt, device=device)
-                if precision == "half":
+                if precision in ("half", "quarter"):
                     self.send_nrm[(mu, d)] = torch.empty((ls, fcb),
                                                          dtype=torch.float32,
                                                          device=device)
@@ -284,14 +284,14 @@ class DwfHalo:
                 edge = 0 if d == 0 else 1
                 for s in range(self.ls):
                     nrm = (self.send_nrm[(mu, d)][s]
-                           if self.precision == "half" else empty)
+                           if self.precision in ("half", "quarter") else empty)
                     ext.pack_face(self.send[(mu, d)][s], nrm,
                                   inp.data, _norm_or_empty(inp),
                                   list(geo.dims), geo.parity_offset, Vcb,
                                   parity, mu, s01, edge, fcb,
                                   v_stride=self.ls * Vcb, s_offset=s * Vcb)
         exchange_tensors(self.send, self.recv)
-        if self.precision == "half":
+        if self.precision in ("half", "quarter"):
             exchange_tensors(self.send_nrm, self.recv_nrm)
 
     def ghost_args(self, s: int):
@@ -307,7 +307,7 @@ class DwfHalo:
                 else:
                     ghosts.append(g[s])
                     nrms.append(self.recv_nrm[(mu, d)][s]
-                                if self.precision == "half" else empty_n)
+                                if self.precision in ("half", "quarter") else empty_n)
         face_cb = [self.geo.face_volume_cb(mu) for mu in range(4)]
         return ghosts, nrms, face_cb
 
@@ -418,7 +418,7 @@ class BatchSpinorHalo:
                                                  device=device)
                 self.recv[(mu, d)] = torch.empty(shape, dtype=dt,
                                                  device=device)
-                if precision == "half":
+                if precision in ("half", "quarter"):
                     self.send_nrm[(mu, d)] = torch.empty(
                         (n_rhs, fcb), dtype=torch.float32, device=device)
                     self.recv_nrm[(mu, d)] = torch.empty(
@@ -441,14 +441,14 @@ class BatchSpinorHalo:
 
     def exchange(self) -> None:
         reqs = exchange_tensors_start(self.send, self.recv)
-        if self.precision == "half":
+        if self.precision in ("half", "quarter"):
             reqs += exchange_tensors_start(self.send_nrm, self.recv_nrm)
         for r in reqs:
             r.wait()
 
     def exchange_start(self) -> list:
         reqs = exchange_tensors_start(self.send, self.recv)
-        if self.precision == "half":
+        if self.precision in ("half", "quarter"):
             reqs += exchange_tensors_start(self.send_nrm, self.recv_nrm)
         return reqs
 

@@ -1,0 +1,121 @@
+"""Quarter precision (fp8-e4m3 block-float) fields and kernels
+(ref: the reference's quarter fixed-point storage,
+color_spinor_field_order.h:1426 / gauge quarter — re-based on CDNA4's
+native OCP e4m3 converts; per-site fp32 norms carry the dynamic range)."""
+import math
+
+import pytest
+import torch
+
+from quda_amd import GaugeField, LatticeGeometry, SpinorField
+from quda_amd.ops import blas
+from quda_amd.ops import reference as ref
+from quda_amd.ops.dispatch import PLAIN, dslash_wilson
+from quda_amd.solvers import cg_solve
+from quda_amd.models import DiracWilsonPC
+
+
+def test_quarter_spinor_roundtrip():
+    geo = LatticeGeometry((4, 4, 4, 4))
+    s = SpinorField(geo, "quarter")
+    gen = torch.Generator().manual_seed(881)
+    v = torch.view_as_complex(torch.randn(2, geo.volume_cb, 4, 3, 2,
+                                          generator=gen, dtype=torch.float64))
+    s.from_complex(v)
+    back = s.to_complex()
+    # e4m3 has a 3-bit mantissa: |err| <= 2^-4 * per-site max (block float)
+    site_max = v.abs().amax(dim=(-2, -1), keepdim=True)
+    rel = ((back - v).abs() / site_max).max().item()
+    assert rel < 0.07, rel
+    assert s.data.dtype == torch.float8_e4m3fn
+    assert s.norm is not None
+
+
+def test_quarter_gauge_roundtrip():
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g64 = GaugeField(geo, "double").random_su3_(seed=882)
+    u = g64.to_complex()
+    gq = GaugeField(geo, "quarter", reconstruct="none").from_complex(u)
+    back = gq.to_complex()
+    assert (back - u).abs().max().item() < 0.07  # |U| <= 1 entries
+
+
+def test_quarter_cpu_dslash_oracle():
+    """The CPU fallback path runs the oracle on dequantized fields —
+    checks the conversions compose through the dispatch layer."""
+    geo = LatticeGeometry((4, 4, 4, 4))
+    g64 = GaugeField(geo, "double").random_su3_(seed=883)
+    u = g64.to_complex()
+    gq = GaugeField(geo, "quarter", reconstruct="none").from_complex(u)
+    psi = SpinorField(geo, "double").gaussian_(seed=884)
+    sq = SpinorField(geo, "quarter").from_complex(psi.to_complex())
+    out = SpinorField(geo, "quarter", n_parity=1)
+    dslash_wilson(out, sq.parity_view(1), gq, 0)
+    want = ref.dslash_wilson_parity(u, psi.to_complex()[1], geo, 0)
+    rel = ((out.to_complex()[0] - want).abs().max()
+           / want.abs().max()).item()
+    assert rel < 0.25, rel  # two fp8 quantizations + 8-hop sum
+
+
+@pytest.mark.gpu
+def test_quarter_dslash_gpu_vs_oracle():
+    geo = LatticeGeometry((8, 8, 8, 8))
+    g64 = GaugeField(geo, "double").random_su3_(seed=885)
+    u = g64.to_complex()
+    psi = SpinorField(geo, "double").gaussian_(seed=886)
+    want = ref.dslash_wilson_parity(u, psi.to_complex()[1], geo, 0)
+    for recon in ("none", "twelve"):
+        gq = GaugeField(geo, "quarter", "cuda", reconstruct=recon)
+        gq.from_complex(u.cuda())
+        sq = SpinorField(geo, "quarter", "cuda").from_complex(
+            psi.to_complex().cuda())
+        out = SpinorField(geo, "quarter", "cuda", n_parity=1)
+        dslash_wilson(out, sq.parity_view(1), gq, 0)
+        rel = ((out.to_complex().cpu()[0] - want).abs().max()
+               / want.abs().max()).item()
+        assert rel < 0.25, (recon, rel)
+
+
+@pytest.mark.gpu
+def test_quarter_m5_gpu():
+    """fp8 block-float M5 family (the BASELINE config-5 precision): the
+    dwf5 kernels at quarter vs the fp64 oracle."""
+    from quda_amd.ops.dispatch import dwf5_op
+    geo = LatticeGeometry((4, 4, 4, 8))
+    LS, MF = 12, 0.05
+    inp = SpinorField(geo, "quarter", "cuda", n_parity=1, ls=LS)
+    gen = torch.Generator().manual_seed(887)
+    v = torch.view_as_complex(torch.randn(1, LS * geo.volume_cb, 4, 3, 2,
+                                          generator=gen, dtype=torch.float64))
+    inp.from_complex(v.cuda())
+    out = SpinorField(geo, "quarter", "cuda", n_parity=1, ls=LS)
+    for kind in (0, 1):
+        dwf5_op(out, inp, 1.9, -0.55, MF, kind=kind)
+        psi = inp.to_complex().cpu()[0]
+        expect = (ref.dslash5(psi, LS, 1.9, -0.55, MF, False) if kind == 0
+                  else ref.m5inv(psi, LS, 1.9, -0.55, MF, False))
+        rel = ((out.to_complex().cpu()[0] - expect).abs().max()
+               / expect.abs().max()).item()
+        assert rel < 0.15, (kind, rel)
+
+
+@pytest.mark.gpu
+def test_quarter_sloppy_cg_gpu():
+    """double-quarter mixed CG with reliable updates still converges to
+    1e-8 (the quarter operator is only the sloppy inner operator)."""
+    geo = LatticeGeometry((8, 8, 8, 8))
+    gen = torch.Generator().manual_seed(888)
+    from quda_amd.fields.gauge import project_su3
+    m = torch.randn((4, 2, geo.volume_cb, 3, 3, 2), generator=gen,
+                    dtype=torch.float64)
+    u = project_su3(torch.view_as_complex(m)).cuda()
+    g = GaugeField(geo, "double", "cuda").from_complex(u)
+    gq = GaugeField(geo, "quarter", "cuda", reconstruct="none").from_complex(u)
+    d = DiracWilsonPC(g, 0.12)
+    dq = DiracWilsonPC(gq, 0.12)
+    b = SpinorField(geo, "double", "cuda", n_parity=1).gaussian_(seed=889)
+    x = SpinorField(geo, "double", "cuda", n_parity=1)
+    st = cg_solve(d, x, b, op_sloppy=dq, sloppy="quarter", tol=1e-8,
+                  maxiter=2000, delta=0.03)
+    assert st.converged, (st.iters, st.resid)
+    assert st.reliable_updates > 0
