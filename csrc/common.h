@@ -80,6 +80,13 @@ __device__ __forceinline__ fp8s qa_tos<fp8s, float>(const float &v) {
   return {__hip_cvt_float_to_fp8(v, __HIP_SATFINITE, __HIP_E4M3)};
 }
 
+__device__ __forceinline__ void qa_sincospi(float x, float *s, float *c) {
+  sincospif(x, s, c);  // sin(pi x), cos(pi x)
+}
+__device__ __forceinline__ void qa_sincospi(double x, double *s, double *c) {
+  sincospi(x, s, c);
+}
+
 // 16-byte opaque chunk for vector loads
 struct alignas(16) chunk16 { unsigned int u[4]; };
 
@@ -226,6 +233,47 @@ struct GaugeAcc {
 #pragma unroll
     for (int c = 0; c < NC; ++c)
       load_chunk<S, W>(base + ((long)(c0 + c) * V + i) * W, tmp + c * W);
+    if constexpr (RECON == 8) {
+      // arXiv:0911.3191 codec in the row-permuted {{b},{a},{-c}} form
+      // (ref gauge_field_order.h Reconstruct<8>; u0 = 1 — anisotropy and
+      // boundary phases force reconstruct 'none' at load time)
+      R st[8];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) st[k] = qa_tor<R>(tmp[off + k]);
+      cplx<R> b2{st[2], st[3]}, b3{st[4], st[5]}, a1{st[6], st[7]};
+      R row_sum = b2.re * b2.re + b2.im * b2.im + b3.re * b3.re + b3.im * b3.im;
+      R rinv = (R)1 / row_sum;
+      R d1 = (R)1 - row_sum;
+      R b1m = d1 > (R)0 ? sqrt(d1) : (R)0;
+      R sn, cs;
+      qa_sincospi(st[0], &sn, &cs);
+      cplx<R> b1{cs * b1m, sn * b1m};
+      R col_sum = b1.re * b1.re + b1.im * b1.im + a1.re * a1.re + a1.im * a1.im;
+      R d2 = (R)1 - col_sum;
+      R cm = d2 > (R)0 ? sqrt(d2) : (R)0;
+      qa_sincospi(st[1], &sn, &cs);
+      cplx<R> mc1{cs * cm, sn * cm};  // -c1
+      cplx<R> A = conj(b1) * a1;
+      cplx<R> a2 = neg(conj(mc1) * conj(b3) + A * b2);
+      a2 = {a2.re * rinv, a2.im * rinv};
+      cplx<R> a3 = conj(mc1) * conj(b2) - A * b3;
+      a3 = {a3.re * rinv, a3.im * rinv};
+      cplx<R> B = conj(b1) * mc1;
+      cplx<R> c2 = neg(conj(a1) * conj(b3) - B * b2);  // -(mc2)
+      c2 = {c2.re * rinv, c2.im * rinv};
+      cplx<R> c3 = conj(a1) * conj(b2) + B * b3;       // -(mc3)
+      c3 = {c3.re * rinv, c3.im * rinv};
+      u[0][0] = a1;
+      u[0][1] = a2;
+      u[0][2] = a3;
+      u[1][0] = b1;
+      u[1][1] = b2;
+      u[1][2] = b3;
+      u[2][0] = neg(mc1);
+      u[2][1] = c2;
+      u[2][2] = c3;
+      return;
+    }
 #pragma unroll
     for (int k = 0; k < RECON / 2; ++k)
       u[k / 3][k % 3] = {qa_tor<R>(tmp[off + 2 * k]), qa_tor<R>(tmp[off + 2 * k + 1])};

@@ -3,6 +3,7 @@
 
 void launch_dslash_wilson_half(const DslashCall &c, hipStream_t st) {
   if (c.recon == 12) dslash_launch_all<PrecHalf, 12>(c, st);
+  else if (c.recon == 8) dslash_launch_all<PrecHalf, 8>(c, st);
   else dslash_launch_all<PrecHalf, 18>(c, st);
 }
 

@@ -48,7 +48,13 @@ __global__ __launch_bounds__(256) void k_dslash_wilson_mrhs(
 #pragma unroll
       for (int c = 0; c < 3; ++c) acc[r][s][c] = {(R)0, (R)0};
 
-  cplx<R> p[4][3], h[2][3], uh[2][3], U[3][3];
+  cplx<R> uh[2][3], U[3][3];
+  // all NRHS projected half-spinors of a hop live simultaneously so their
+  // NRHS x 4 chunk loads issue back-to-back — the gather-first discipline
+  // of the single-RHS kernel (profiles/r01_dslash_pmc_analysis.md); a
+  // shared scratch buffer here serializes the loads and measures 2-4x
+  // slower (r02 finding).
+  cplx<R> hh[NRHS][2][3];
   const R one = (R)0.5;
 
   bool bnd = false;
@@ -70,15 +76,18 @@ __global__ __launch_bounds__(256) void k_dslash_wilson_mrhs(
       long j = neighbor_cb(xc, MU, +1, d);                                 \
       _Pragma("unroll") for (int r = 0; r < NRHS; ++r) {                   \
         if (KT == KT_FUSED && cross_p) {                                   \
-          gh.load_r(h, MU, 1, ghost_idx(xc, MU, d), r);                    \
+          gh.load_r(hh[r], MU, 1, ghost_idx(xc, MU, d), r);                \
         } else {                                                           \
+          cplx<R> p[4][3];                                                 \
           in_r.data = const_cast<typename Prec::Store *>(ptr.in[r]);       \
           in_r.norm = const_cast<float *>(ptr.in_n[r]);                    \
           in_r.load(p, j);                                                 \
-          if constexpr (!DAG) proj_##MU##_0(h, p);                         \
-          else proj_##MU##_1(h, p);                                        \
+          if constexpr (!DAG) proj_##MU##_0(hh[r], p);                     \
+          else proj_##MU##_1(hh[r], p);                                    \
         }                                                                  \
-        su3_mul_half(uh, U, h);                                            \
+      }                                                                    \
+      _Pragma("unroll") for (int r = 0; r < NRHS; ++r) {                   \
+        su3_mul_half(uh, U, hh[r]);                                        \
         if constexpr (!DAG) recon_##MU##_0(acc[r], uh, one);               \
         else recon_##MU##_1(acc[r], uh, one);                              \
       }                                                                    \
@@ -94,15 +103,18 @@ __global__ __launch_bounds__(256) void k_dslash_wilson_mrhs(
       }                                                                    \
       _Pragma("unroll") for (int r = 0; r < NRHS; ++r) {                   \
         if (KT == KT_FUSED && cross_m) {                                   \
-          gh.load_r(h, MU, 0, ghost_idx(xc, MU, d), r);                    \
+          gh.load_r(hh[r], MU, 0, ghost_idx(xc, MU, d), r);                \
         } else {                                                           \
+          cplx<R> p[4][3];                                                 \
           in_r.data = const_cast<typename Prec::Store *>(ptr.in[r]);       \
           in_r.norm = const_cast<float *>(ptr.in_n[r]);                    \
           in_r.load(p, j);                                                 \
-          if constexpr (!DAG) proj_##MU##_1(h, p);                         \
-          else proj_##MU##_0(h, p);                                        \
+          if constexpr (!DAG) proj_##MU##_1(hh[r], p);                     \
+          else proj_##MU##_0(hh[r], p);                                    \
         }                                                                  \
-        su3_dagmul_half(uh, U, h);                                         \
+      }                                                                    \
+      _Pragma("unroll") for (int r = 0; r < NRHS; ++r) {                   \
+        su3_dagmul_half(uh, U, hh[r]);                                     \
         if constexpr (!DAG) recon_##MU##_1(acc[r], uh, one);               \
         else recon_##MU##_0(acc[r], uh, one);                              \
       }                                                                    \

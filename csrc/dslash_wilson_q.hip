@@ -5,6 +5,7 @@
 
 void launch_dslash_wilson_quarter(const DslashCall &c, hipStream_t st) {
   if (c.recon == 12) dslash_launch_all<PrecQuarter, 12>(c, st);
+  else if (c.recon == 8) dslash_launch_all<PrecQuarter, 8>(c, st);
   else dslash_launch_all<PrecQuarter, 18>(c, st);
 }
 

@@ -3,6 +3,7 @@
 
 void launch_dslash_wilson_single(const DslashCall &c, hipStream_t st) {
   if (c.recon == 12) dslash_launch_all<PrecSingle, 12>(c, st);
+  else if (c.recon == 8) dslash_launch_all<PrecSingle, 8>(c, st);
   else dslash_launch_all<PrecSingle, 18>(c, st);
 }
 

@@ -30,7 +30,59 @@ import torch
 from .geometry import LatticeGeometry
 from .layout import DTYPE_OF, WIDTH_OF
 
-RECON_COMPS = {"none": 18, "twelve": 12}
+RECON_COMPS = {"none": 18, "twelve": 12, "eight": 8}
+
+
+def pack_recon8(u: torch.Tensor) -> torch.Tensor:
+    """[..., 3, 3] complex SU(3) -> [..., 8] real (the arXiv:0911.3191
+    codec as the reference packs it, gauge_field_order.h Reconstruct<8>:
+    phases of b1 and -c1 (units of pi) + b2, b3, a1 re/im; the matrix is
+    compressed in the row-permuted form {{b},{a},{-c}} to avoid the
+    unit-gauge singularity)."""
+    import math
+    sh = u.shape[:-2]
+    out = torch.empty((*sh, 8), dtype=torch.float64)
+    a1, b = u[..., 0, 0], u[..., 1, :]
+    c1 = u[..., 2, 0]
+    out[..., 0] = torch.atan2(b[..., 0].imag, b[..., 0].real) / math.pi
+    out[..., 1] = torch.atan2(-c1.imag, -c1.real) / math.pi
+    out[..., 2] = b[..., 1].real
+    out[..., 3] = b[..., 1].imag
+    out[..., 4] = b[..., 2].real
+    out[..., 5] = b[..., 2].imag
+    out[..., 6] = a1.real
+    out[..., 7] = a1.imag
+    return out.to(u.real.dtype)
+
+
+def unpack_recon8(p: torch.Tensor, dtype=torch.complex128) -> torch.Tensor:
+    """[..., 8] real -> [..., 3, 3] complex (mirrors the kernel decode in
+    csrc/common.h GaugeAcc::load_base<RECON=8>; u0 = 1)."""
+    import math
+    rd = torch.float64 if dtype == torch.complex128 else torch.float32
+    p = p.to(rd)
+    M = torch.empty((*p.shape[:-1], 3, 3), dtype=dtype)
+    b2 = torch.complex(p[..., 2], p[..., 3])
+    b3 = torch.complex(p[..., 4], p[..., 5])
+    a1 = torch.complex(p[..., 6], p[..., 7])
+    # permuted M rows: row0 = b, row1 = a, row2 = -c
+    row_sum = (b2.abs() ** 2 + b3.abs() ** 2)
+    b1 = torch.polar((1.0 - row_sum).clamp_min(0).sqrt(),
+                     p[..., 0] * math.pi)
+    col_sum = b1.abs() ** 2 + a1.abs() ** 2
+    mc1 = torch.polar((1.0 - col_sum).clamp_min(0).sqrt(),
+                      p[..., 1] * math.pi)
+    r_inv2 = 1.0 / row_sum
+    A = b1.conj() * a1
+    a2 = -(mc1.conj() * b3.conj() + A * b2) * r_inv2
+    a3 = (mc1.conj() * b2.conj() - A * b3) * r_inv2
+    B = b1.conj() * mc1
+    mc2 = (a1.conj() * b3.conj() - B * b2) * r_inv2
+    mc3 = -(a1.conj() * b2.conj() + B * b3) * r_inv2
+    M[..., 0, 0], M[..., 0, 1], M[..., 0, 2] = a1, a2, a3
+    M[..., 1, 0], M[..., 1, 1], M[..., 1, 2] = b1, b2, b3
+    M[..., 2, 0], M[..., 2, 1], M[..., 2, 2] = -mc1, -mc2, -mc3
+    return M
 
 
 class GaugeField:
@@ -84,8 +136,12 @@ class GaugeField:
         flat = d.movedim(1, 2).reshape(2, V, 8 * L)  # [p, V, comps]
         out = torch.empty((4, 2, V, 3, 3), dtype=dtype, device=d.device)
         for mu in range(4):
+            comps = flat[:, :, mu * L:(mu + 1) * L]
+            if self.reconstruct == "eight":
+                out[mu] = unpack_recon8(comps, dtype).to(d.device)
+                continue
             rows = torch.view_as_complex(
-                flat[:, :, mu * L:(mu + 1) * L].reshape(2, V, L // 2, 2).contiguous())
+                comps.reshape(2, V, L // 2, 2).contiguous())
             if self.reconstruct == "none":
                 out[mu] = rows.reshape(2, V, 3, 3)
             else:
@@ -93,6 +149,14 @@ class GaugeField:
                 r2 = torch.cross(r01[..., 0, :], r01[..., 1, :], dim=-1).conj()
                 out[mu] = torch.cat([r01, r2.unsqueeze(-2)], dim=-2)
         return out
+
+    def _encode(self, links: torch.Tensor) -> torch.Tensor:
+        """[..., 3, 3] complex -> [..., L] real per the reconstruct codec."""
+        if self.reconstruct == "eight":
+            return pack_recon8(links).to(links.device)
+        nrows = self.L // 6
+        return torch.view_as_real(links[..., 0:nrows, :].contiguous()).reshape(
+            *links.shape[:-2], self.L)
 
     def from_complex(self, u: torch.Tensor) -> "GaugeField":
         """u: [4, 2, V_cb, 3, 3] complex -> fill fwd + shifted bwd slots.
@@ -125,16 +189,12 @@ class GaugeField:
         flat = torch.empty((2, V, 8 * L),
                            dtype=torch.float64 if u.dtype == torch.complex128 else torch.float32,
                            device=dev)
-        nrows = L // 6  # 3 or 2 link rows stored
         for p in (0, 1):
             for mu in range(4):
-                fwd = u[mu, p][:, 0:nrows, :]                      # [V,nrows,3]
-                flat[p, :, mu * L:(mu + 1) * L] = torch.view_as_real(
-                    fwd).reshape(V, L)
+                flat[p, :, mu * L:(mu + 1) * L] = self._encode(u[mu, p])
                 bwd_idx = self._nbr_shift(p, mu).to(dev)
-                bwd = u[mu, (1 - p) if self.shift % 2 else p][bwd_idx][:, 0:nrows, :]
-                flat[p, :, (4 + mu) * L:(5 + mu) * L] = torch.view_as_real(
-                    bwd).reshape(V, L)
+                bwd = u[mu, (1 - p) if self.shift % 2 else p][bwd_idx]
+                flat[p, :, (4 + mu) * L:(5 + mu) * L] = self._encode(bwd)
         # fix up bwd slots on partitioned-dim boundary faces: the x_mu = l
         # (l < shift) sites need U_mu(x - shift*mu) owned by the -mu
         # neighbor at its coord X-shift+l (one exchange per layer; shift=3
@@ -148,18 +208,18 @@ class GaugeField:
             fcb = geo.face_volume_cb(mu)
             for l in range(self.shift):
                 c_src = hi - (self.shift - 1) + l
-                send = torch.empty((2, fcb, nrows, 3), dtype=u.dtype,
+                send = torch.empty((2, fcb, 3, 3), dtype=u.dtype,
                                    device=dev)
                 for q in (0, 1):
                     send[q] = u[mu, q][geo.face_index_cb(q, mu, c_src)
-                                       .to(dev)][:, 0:nrows, :]
+                                       .to(dev)]
                 recv = torch.empty_like(send)
                 exchange_tensors({(mu, 1): send}, {(mu, 0): recv})
                 for p in (0, 1):
                     fidx = geo.face_index_cb(p, mu, l).to(dev)
                     q = (1 - p) if self.shift % 2 else p
                     flat[p, fidx, (4 + mu) * L:(5 + mu) * L] = \
-                        torch.view_as_real(recv[q]).reshape(fcb, L)
+                        self._encode(recv[q])
         w = WIDTH_OF[self.precision]
         native = flat.reshape(2, V, (8 * L) // w, w).movedim(2, 1).contiguous()
         self.data.copy_(native.to(self.data.dtype))

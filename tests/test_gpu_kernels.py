@@ -331,3 +331,22 @@ def test_dslash_mrhs_vs_per_rhs(setup, prec, nrhs):
                 dm = (outs_m[r].to_complex() - ref_c)
                 err = dm.abs().max().item() / ref_c.abs().max().item()
                 assert err < rtol, (prec, nrhs, mode, use_x, dag, r, err)
+
+
+@pytest.mark.parametrize("prec", ["half", "single"])
+def test_dslash_recon8_gpu(setup, prec):
+    """recon-8 kernel decode vs the oracle and vs the recon-12 kernel."""
+    geo, g, psi, chi, A = setup
+    u = g.to_complex()
+    g8 = GaugeField(geo, prec, "cuda", reconstruct="eight")
+    g8.from_complex(u.cuda())
+    sd = SpinorField(geo, prec, "cuda", n_parity=2).from_complex(
+        psi.to_complex().cuda())
+    out = SpinorField(geo, prec, "cuda", n_parity=1)
+    for dagger in (False, True):
+        dslash_wilson(out, sd.parity_view(1), g8, 0, dagger)
+        want = ref.dslash_wilson_parity(u, psi.to_complex()[1], geo, 0,
+                                        dagger)
+        err = ((out.to_complex().cpu()[0] - want).abs().max()
+               / want.abs().max()).item()
+        assert err < TOL[prec], (prec, dagger, err)
