@@ -259,13 +259,21 @@ _MRHS_GROUP = None
 
 def mrhs_group_size() -> int:
     """Preferred kernel NRHS (QUDA_AMD_MRHS_GROUP: 4, 2, or 1=disable).
-    Default 2: the NRHS=4 instantiation spills (1-5 KB/lane scratch at
-    256-VGPR cap) and measures slower than 2x NRHS=2."""
+
+    Default 1 (per-RHS launches) — measured on MI355X: the 256 MB
+    Infinity Cache keeps the ~100 MB gauge+clover streams resident
+    ACROSS per-RHS launches at 32^3x64, so kernel-level RHS batching
+    saves no HBM traffic there, while the direction-major NRHS loop caps
+    per-wave load ILP at NRHS x 4 chunks (vs 32 for the 8-direction
+    gather-first single-RHS kernel) and measures ~2x slower
+    (profiles/r02_dslash_sweep.md). The batched-HALO merging (one
+    message per face for the whole batch) is independent of this knob
+    and always on. Set 2/4 for volumes whose gauge stream exceeds L3."""
     global _MRHS_GROUP
     if _MRHS_GROUP is None:
         import os
-        v = int(os.environ.get("QUDA_AMD_MRHS_GROUP", "2"))
-        _MRHS_GROUP = v if v in (1, 2, 4) else 2
+        v = int(os.environ.get("QUDA_AMD_MRHS_GROUP", "1"))
+        _MRHS_GROUP = v if v in (1, 2, 4) else 1
     return _MRHS_GROUP
 
 
