@@ -467,3 +467,76 @@ def test_coarse_mfma_vs_torch_gpu(nr):
             assert rel1 < 5e-6, (nr, mask, rel1)
         finally:
             comms.set_forced_partition(0)
+
+
+def test_staggered_kd_free_spectrum_circle():
+    """ROUND2_PLAN §3 derivation check: on the free field the
+    Kahler-Dirac-preconditioned staggered operator X^-1 M has its whole
+    spectrum on the Wilson-like circle |lambda - 1| = sqrt(1 - (2m)^2/..)
+    ~ 1, tangent to zero from the right — the property that makes
+    staggered MG coarsening contract (ref dirac_staggered_kd.cpp role)."""
+    import numpy as np
+    from quda_amd.models import DiracStaggered
+    from quda_amd.models.staggered_kd import KDBlockInverse
+    geo = LatticeGeometry((4, 4, 2, 2))
+    u = torch.zeros((4, 2, geo.volume_cb, 3, 3), dtype=torch.complex128)
+    for c in range(3):
+        u[..., c, c] = 1.0
+    m = 0.1
+    d = DiracStaggered(GaugeField(geo, "double").from_complex(u), m)
+    kd = KDBlockInverse(u, geo, m)
+
+    def apply_op(f):
+        s = SpinorField(geo, "double", n_parity=2, nspin=1)
+        s.from_complex(f)
+        w = SpinorField(geo, "double", n_parity=2, nspin=1)
+        d.M(w, s)
+        t = SpinorField(geo, "double", n_parity=2, nspin=1)
+        kd.apply(t, w)
+        return t.to_complex()
+
+    n = geo.volume * 3
+    cols = []
+    E = torch.zeros((2, geo.volume_cb, 3), dtype=torch.complex128)
+    for p in range(2):
+        for i in range(geo.volume_cb):
+            for c in range(3):
+                E.zero_()
+                E[p, i, c] = 1.0
+                cols.append(apply_op(E).reshape(-1).numpy())
+    A = np.array(cols).T
+    ev = np.linalg.eigvals(A)
+    assert ev.real.min() > 0, ev.real.min()           # right half plane
+    r = np.abs(ev - 1.0)
+    assert r.max() - r.min() < 1e-6, (r.min(), r.max())  # one circle
+    assert abs(r.max() - np.sqrt(1 - ev.real.min() * (2 - ev.real.min()))) \
+        < 1.0  # radius ~ <=1
+
+
+def test_staggered_kd_mg_contracts():
+    """VERDICT r1 #10 'Done' criterion: measured iteration reduction on
+    the staggered operator — the V-cycle over the KD-preconditioned op
+    (spectrum on the Wilson circle) contracts where coarsening plain
+    staggered could not (round-1 finding, mg/staggered.py)."""
+    from quda_amd.models.staggered_kd import DiracStaggeredKD
+    from quda_amd.mg.staggered import StaggeredMG
+    geo = LatticeGeometry((4, 4, 4, 8))
+    g = GaugeField(geo, "double").random_su3_(seed=17)
+    dkd = DiracStaggeredKD(g, 0.3)
+    b = SpinorField(geo, "double", n_parity=2, nspin=1).gaussian_(seed=5)
+    bp = dkd.prepare(b)
+    x0 = SpinorField(geo, "double", n_parity=2, nspin=1)
+    st_p = gcr_solve(dkd, x0, bp, tol=1e-8, maxiter=300, nkrylov=16)
+    mg = StaggeredMG(dkd, geo, block=(2, 2, 2, 4), n_vec=8, n_smooth=4)
+    x1 = SpinorField(geo, "double", n_parity=2, nspin=1)
+    st_m = gcr_solve(dkd, x1, bp, tol=1e-8, maxiter=300, nkrylov=16,
+                     precond=mg.precond)
+    assert st_p.converged and st_m.converged
+    assert st_m.iters * 2 < st_p.iters, (st_m.iters, st_p.iters)
+    # solution solves the PLAIN staggered system
+    from quda_amd.models import DiracStaggered
+    dpl = DiracStaggered(g, 0.3)
+    r = SpinorField(geo, "double", n_parity=2, nspin=1)
+    dpl.M(r, x1)
+    tr = math.sqrt(blas.xmy_norm2(b, r) / blas.norm2(b))
+    assert tr < 1e-6, tr
