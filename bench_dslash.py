@@ -63,11 +63,16 @@ def sweep(geo, u, reps):
         blocks = [64] if waves == 3 else [64, 128, 256]
         for blk in blocks:
             _ext().set_dslash_block(blk)
-            for prec, recon in [("half", "twelve"), ("single", "twelve")]:
+            recons = ([("half", "twelve"), ("half", "eight"),
+                       ("single", "twelve"), ("single", "eight"),
+                       ("quarter", "eight")] if (waves == 0 and blk == 64)
+                      else [("half", "twelve"), ("single", "twelve")])
+            for prec, recon in recons:
                 for mode, mname in [(PLAIN, "wilson"),
                                     (CLOV_POST, "wilson_clover")]:
                     r = run(prec, recon, mode, geo, u, A, reps)
-                    key = f"{mname}/{prec}/r12/b{blk}w{waves}"
+                    rc = {"twelve": 12, "eight": 8, "none": 18}[recon]
+                    key = f"{mname}/{prec}/r{rc}/b{blk}w{waves}"
                     results[key] = round(r["gflops"])
                     print(f"{key:40s} {r['us']:8.1f} us  {r['gflops']:8.0f} "
                           f"GFLOPS", flush=True)

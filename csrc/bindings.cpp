@@ -434,6 +434,62 @@ static void set_dslash_waves(int64_t w) {
   if (w == 0 || w == 3) qa_dslash_waves_ref() = (int)w;
 }
 
+// ---------------------------------------------------------------------------
+// HIP-IPC remote-write halos (role of comm_target.cpp:41-134 P2P
+// remote-write): export/import device-buffer handles so the pack kernel
+// writes directly into the PEER rank's recv buffer over xGMI.
+static py::bytes ipc_get_handle(at::Tensor t) {
+  hipIpcMemHandle_t h;
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous());
+  hipError_t e = hipIpcGetMemHandle(&h, t.data_ptr());
+  TORCH_CHECK(e == hipSuccess, "hipIpcGetMemHandle: ", hipGetErrorString(e));
+  return py::bytes(reinterpret_cast<const char *>(&h), sizeof(h));
+}
+
+static int64_t ipc_open_handle(py::bytes handle) {
+  std::string s = handle;
+  TORCH_CHECK(s.size() == sizeof(hipIpcMemHandle_t));
+  hipIpcMemHandle_t h;
+  memcpy(&h, s.data(), sizeof(h));
+  void *ptr = nullptr;
+  hipError_t e =
+      hipIpcOpenMemHandle(&ptr, h, hipIpcMemLazyEnablePeerAccess);
+  TORCH_CHECK(e == hipSuccess, "hipIpcOpenMemHandle: ", hipGetErrorString(e));
+  return (int64_t)(uintptr_t)ptr;
+}
+
+static void ipc_close_handle(int64_t ptr) {
+  hipIpcCloseMemHandle((void *)(uintptr_t)ptr);
+}
+
+// pack a face directly into a raw (peer) destination pointer
+static void pack_face_ptr(int64_t dst, int64_t dst_nrm, at::Tensor in,
+                          at::Tensor in_n, std::vector<int64_t> dims,
+                          int64_t parity_offset, int64_t Vcb, int64_t parity,
+                          int64_t mu, int64_t s01, int64_t edge, int64_t Fcb,
+                          int64_t v_stride, int64_t s_offset) {
+  PackCall c{};
+  c.in = field_of_off(in, in_n, v_stride ? v_stride : Vcb, s_offset);
+  c.dst = (void *)(uintptr_t)dst;
+  c.dst_nrm = (float *)(uintptr_t)dst_nrm;
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.mu = (int)mu;
+  c.s01 = (int)s01;
+  c.edge = (int)edge;
+  c.Fcb = Fcb;
+  c.prec = prec_of(in);
+  switch (c.prec) {
+    case 0: launch_pack_face_double(c, stream()); break;
+    case 1: launch_pack_face_single(c, stream()); break;
+    case 2: launch_pack_face_half(c, stream()); break;
+    case 3: launch_pack_face_quarter(c, stream()); break;
+  }
+  check_launch("pack_face_ptr");
+}
+
 static void coarse_dslash_mfma(at::Tensor mats, at::Tensor nbr9,
                                at::Tensor c, at::Tensor out, int64_t Na,
                                int64_t Nc, int64_t NR) {
@@ -482,6 +538,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("set_dslash_waves", &set_dslash_waves,
         "occupancy experiment: 0 default, 3 = 64-thread/3-wave variant");
+  m.def("ipc_get_handle", &ipc_get_handle, "hipIpcGetMemHandle of a tensor");
+  m.def("ipc_open_handle", &ipc_open_handle, "open a peer IPC handle");
+  m.def("ipc_close_handle", &ipc_close_handle, "close a peer IPC mapping");
+  m.def("pack_face_ptr", &pack_face_ptr,
+        "pack a halo face into a raw (IPC peer) pointer");
   m.def("dwf5", &dwf5, "DWF/Moebius 5th-dim ops (Ds apply / M5 inverse)");
   m.def("zdwf5", &zdwf5, "zMobius per-slice-complex 5th-dim ops");
   m.def("eofa5", &eofa5, "EOFA rank-1 extended M5 ops");

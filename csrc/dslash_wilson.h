@@ -47,7 +47,9 @@ template <typename Prec>
 struct CloverAcc {
   using S = typename Prec::Store;
   using R = typename Prec::Real;
-  static constexpr int W = Prec::W;
+  // widest chunk dividing the 72-real site (quarter: 16 does not divide
+  // 72 -> 8-byte chunks, mirroring fields/layout.py chunk_width)
+  static constexpr int W = chunk_w<72, Prec::W>::value;
   static constexpr int NCH = 72 / W;
   const S *data;  // [parity][NCH][V][W]
   long V;

@@ -82,17 +82,22 @@ _POLICY = None
 def dslash_policy() -> str:
     """'overlap' (interior/exterior split around async RCCL transfers,
     default) or 'fused' (blocking exchange, single ghost-aware kernel).
-    Env QUDA_AMD_DSLASH_POLICY overrides."""
+    Env QUDA_AMD_DSLASH_POLICY overrides; QUDA_AMD_BLOCKING_COMMS=1 is
+    the safe-mode switch (forces 'fused' AND disables the policy
+    autotuner's overlap candidates — the multi-GPU first-run fallback)."""
     global _POLICY
     if _POLICY is None:
         import os
-        _POLICY = os.environ.get("QUDA_AMD_DSLASH_POLICY", "overlap")
+        if os.environ.get("QUDA_AMD_BLOCKING_COMMS", "0") == "1":
+            _POLICY = "fused"
+        else:
+            _POLICY = os.environ.get("QUDA_AMD_DSLASH_POLICY", "overlap")
     return _POLICY
 
 
 def set_dslash_policy(p: str) -> None:
     global _POLICY
-    assert p in ("overlap", "fused")
+    assert p in ("overlap", "fused", "ipc")
     _POLICY = p
 
 
@@ -190,6 +195,19 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
         from ..parallel.halo import get_spinor_halo
         h = get_spinor_halo(geo, inp.precision, inp.device, mask)
 
+        if dslash_policy() == "ipc":
+            # remote-write transport (comm_target.cpp:41-134 role): pack
+            # kernels scatter each face straight into the peer rank's
+            # recv buffer over xGMI; interior overlaps the peers' packs
+            from ..parallel.ipc_halo import get_ipc_halo
+            hi = get_ipc_halo(geo, inp.precision, inp.device, mask)
+            hi.pack_remote(ext, inp, 1 - parity, bool(dagger))
+            ghosts, nrms, face_cb = hi.ghost_args()
+            launch(2, ghosts, nrms, face_cb)   # interior
+            hi.complete()
+            launch(3, ghosts, nrms, face_cb)   # exterior
+            return out
+
         def run_halo():
             h.pack(ext, inp, 1 - parity, bool(dagger))
             ghosts, nrms, face_cb = h.ghost_args()
@@ -208,7 +226,9 @@ def dslash_wilson(out: SpinorField, inp: SpinorField, gauge: GaugeField,
                     r.wait()
                 launch(3, ghosts, nrms, face_cb)   # exterior
 
-        if _autotune_on():
+        import os
+        if _autotune_on() and os.environ.get("QUDA_AMD_BLOCKING_COMMS",
+                                             "0") != "1":
             # policy-level autotune (ref DslashPolicyTune): every rank
             # reaches this collectively, so candidate runs stay in
             # lockstep; rank-0's winner is broadcast by the tuner
