@@ -40,11 +40,10 @@ from quda_amd.ops import reference as ref  # noqa: E402
 from quda_amd.parallel import comms  # noqa: E402
 from quda_amd.solvers import cg_solve  # noqa: E402
 
-KAPPA = 0.135
 CSW = 1.0
 
 
-def setup_fields(geo, device, sloppy_prec, seed):
+def setup_fields(geo, device, sloppy_prec, seed, kappa):
     """Random SU(3) gauge + clover at double and sloppy precision."""
     g = GaugeField(geo, "double", device)
     # generate directly on device (host gen at 32^3x64 is minutes of QR)
@@ -55,7 +54,7 @@ def setup_fields(geo, device, sloppy_prec, seed):
     from quda_amd.fields.gauge import project_su3
     u = project_su3(torch.view_as_complex(m.to(torch.float64)).to(device))
     g.from_complex(u)
-    A = ref.clover_matrix(u, geo, KAPPA, CSW)
+    A = ref.clover_matrix(u, geo, kappa, CSW)
     cl = CloverField(geo, "double", device).from_matrices(A)
     # sloppy copies
     recon = "twelve" if sloppy_prec != "double" else "none"
@@ -95,6 +94,11 @@ def main():
                     help="reliable-update trigger (production default 0.1)")
     ap.add_argument("--maxiter", type=int, default=2000)
     ap.add_argument("--lattice", type=str, default="32,32,32,64")
+    ap.add_argument("--kappa", type=float, default=0.2,
+                    help="hopping parameter; the default targets a HARD "
+                         "near-critical solve on the random field "
+                         "(hundreds of iterations), the honest "
+                         "time-to-solution regime")
     ap.add_argument("--sloppy", type=str, default="half",
                     choices=["double", "single", "half"])
     ap.add_argument("--seed", type=int, default=777)
@@ -113,9 +117,10 @@ def main():
         torch.cuda.set_device(device)
 
     geo = LatticeGeometry(dims, parity_offset=comms.parity_offset_of_rank(dims))
-    g, cl, gs, cls = setup_fields(geo, device, args.sloppy, args.seed + rank)
-    d = DiracCloverPC(g, cl, KAPPA)
-    ds = DiracCloverPC(gs, cls, KAPPA)
+    g, cl, gs, cls = setup_fields(geo, device, args.sloppy, args.seed + rank,
+                                  args.kappa)
+    d = DiracCloverPC(g, cl, args.kappa)
+    ds = DiracCloverPC(gs, cls, args.kappa)
 
     b = SpinorField(geo, "double", device, n_parity=1).gaussian_(seed=args.seed + 100 + rank)
     x = SpinorField(geo, "double", device, n_parity=1)
@@ -179,7 +184,7 @@ def main():
                 "lattice_per_gpu": "x".join(str(v) for v in dims),
                 "global_batch": 1,
                 "seq_len": geo.volume,
-                "kappa": KAPPA,
+                "kappa": args.kappa,
                 "csw": CSW,
                 "solve_tol": args.tol,
                 "reliable_delta": args.delta,
