@@ -94,7 +94,7 @@ def main():
                     help="reliable-update trigger (production default 0.1)")
     ap.add_argument("--maxiter", type=int, default=2000)
     ap.add_argument("--lattice", type=str, default="32,32,32,64")
-    ap.add_argument("--kappa", type=float, default=0.2,
+    ap.add_argument("--kappa", type=float, default=0.34,
                     help="hopping parameter; the default targets a HARD "
                          "near-critical solve on the random field "
                          "(hundreds of iterations), the honest "

@@ -439,23 +439,39 @@ static void set_dslash_waves(int64_t w) {
 // remote-write): export/import device-buffer handles so the pack kernel
 // writes directly into the PEER rank's recv buffer over xGMI.
 static py::bytes ipc_get_handle(at::Tensor t) {
-  hipIpcMemHandle_t h;
+  // torch's caching allocator suballocates: the IPC handle refers to the
+  // BASE allocation, so export (handle, offset-of-tensor-within-it)
   TORCH_CHECK(t.is_cuda() && t.is_contiguous());
-  hipError_t e = hipIpcGetMemHandle(&h, t.data_ptr());
+  void *base = nullptr;
+  size_t sz = 0;
+  hipError_t e = hipMemGetAddressRange((hipDeviceptr_t *)&base, &sz,
+                                       (hipDeviceptr_t)t.data_ptr());
+  TORCH_CHECK(e == hipSuccess, "hipMemGetAddressRange: ",
+              hipGetErrorString(e));
+  hipIpcMemHandle_t h;
+  e = hipIpcGetMemHandle(&h, base);
   TORCH_CHECK(e == hipSuccess, "hipIpcGetMemHandle: ", hipGetErrorString(e));
-  return py::bytes(reinterpret_cast<const char *>(&h), sizeof(h));
+  uint64_t off = (uint64_t)((char *)t.data_ptr() - (char *)base);
+  std::string out(sizeof(h) + sizeof(off), '\0');
+  memcpy(&out[0], &h, sizeof(h));
+  memcpy(&out[sizeof(h)], &off, sizeof(off));
+  return py::bytes(out);
 }
 
-static int64_t ipc_open_handle(py::bytes handle) {
+static py::tuple ipc_open_handle(py::bytes handle) {
+  // returns (tensor_ptr, mapped_base) — close with the BASE
   std::string s = handle;
-  TORCH_CHECK(s.size() == sizeof(hipIpcMemHandle_t));
+  TORCH_CHECK(s.size() == sizeof(hipIpcMemHandle_t) + sizeof(uint64_t));
   hipIpcMemHandle_t h;
+  uint64_t off;
   memcpy(&h, s.data(), sizeof(h));
+  memcpy(&off, s.data() + sizeof(h), sizeof(off));
   void *ptr = nullptr;
   hipError_t e =
       hipIpcOpenMemHandle(&ptr, h, hipIpcMemLazyEnablePeerAccess);
   TORCH_CHECK(e == hipSuccess, "hipIpcOpenMemHandle: ", hipGetErrorString(e));
-  return (int64_t)(uintptr_t)ptr;
+  return py::make_tuple((int64_t)((uintptr_t)ptr + off),
+                        (int64_t)(uintptr_t)ptr);
 }
 
 static void ipc_close_handle(int64_t ptr) {

@@ -66,14 +66,14 @@ class IpcSpinorHalo(SpinorHalo):
                     continue
                 self.self_wrap[(mu, d)] = False
                 h = gathered[peer][dst_key]
-                p = ext.ipc_open_handle(h)
+                p, pbase = ext.ipc_open_handle(h)
                 self.peer_ptr[(mu, d)] = p
-                self._opened.append(p)
+                self._opened.append(pbase)
                 hn = gathered[peer].get(("n", *dst_key))
                 if hn is not None:
-                    pn = ext.ipc_open_handle(hn)
+                    pn, pnb = ext.ipc_open_handle(hn)
                     self.peer_nrm_ptr[(mu, d)] = pn
-                    self._opened.append(pn)
+                    self._opened.append(pnb)
         if comms.is_distributed():
             dist.barrier()  # all mappings open before anyone packs
 
