@@ -369,9 +369,14 @@ def _wrap(v: torch.Tensor, p: InvertParam, n_parity: int) -> SpinorField:
 
 def dslash_quda(inp: torch.Tensor, p: InvertParam, parity: int) -> torch.Tensor:
     """Apply the parity-hopping dslash (ref: dslashQuda
-    interface_quda.cpp:1709). inp at parity 1-parity."""
+    interface_quda.cpp:1709). inp at parity 1-parity ([Vcb, 4, 3] for
+    Wilson-family, [Vcb, 3] for staggered; a leading parity dim of 1 is
+    also accepted)."""
     d = _make_dirac(p)
-    src = _wrap(inp.unsqueeze(0) if inp.dim() == 3 else inp, p, 1)
+    nsite_dims = 1 if p.dslash_type in (DslashType.STAGGERED,
+                                        DslashType.ASQTAD) else 2
+    src = _wrap(inp.unsqueeze(0) if inp.dim() == 1 + nsite_dims else inp,
+                p, 1)
     out = src.clone_empty()
     d.dslash(out, src, parity)
     return out.to_complex().cpu()[0]

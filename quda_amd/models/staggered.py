@@ -23,6 +23,11 @@ class DiracStaggered(Dirac):
         return SpinorField(self.geo, precision or self.gauge.precision,
                            self.gauge.device, n_parity, nspin=1)
 
+    def dslash(self, out, inp, parity, dagger=False, **kw):
+        """Parity-hopping KS term (the dslashQuda entry; D^dag = -D)."""
+        return dslash_staggered(out, inp, self.gauge, parity,
+                                b=-1.0 if dagger else 1.0)
+
     def M(self, out: SpinorField, inp: SpinorField, dagger: bool = False):
         b = -1.0 if dagger else 1.0
         for p in (0, 1):

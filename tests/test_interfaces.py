@@ -73,3 +73,49 @@ def test_milc_multishift(milc_setup):
         bf.from_complex(be.unsqueeze(0))
         tr = math.sqrt(blas.xmy_norm2(bf, r) / blas.norm2(bf))
         assert tr < 1e-7, (s, tr)
+
+
+def test_milc_extended_surface():
+    """The round-2 MILC entries (lifecycle, dslash, clover solves,
+    multi-src, rephase, unitarized links, observables, momenta) run
+    end-to-end on a small lattice."""
+    import torch
+    from quda_amd import api
+    from quda_amd.interfaces import milc
+    dims = (4, 4, 4, 4)
+    milc.qudaInit()
+    milc.qudaSetLayout(dims)
+    geo_v = 4 ** 4
+    gen = torch.Generator().manual_seed(909)
+    from quda_amd.fields.gauge import GaugeField, project_su3
+    from quda_amd.fields.geometry import LatticeGeometry
+    from quda_amd.fields.interop import gauge_to_milc
+    geo = LatticeGeometry(dims)
+    u = GaugeField(geo, "double").random_su3_(seed=909).to_complex()
+    milc.qudaLoadGauge(dims, gauge_to_milc(u, geo))
+    # dslash + staggered solve
+    src = torch.view_as_complex(torch.randn(geo_v, 3, 2, generator=gen,
+                                            dtype=torch.float64))
+    out = milc.qudaDslash(src, parity=0)
+    assert out.shape == src.shape and out.abs().max() > 0
+    x = milc.qudaEigCGInvert(0.3, src, tol=1e-7, maxiter=500)
+    assert x.abs().max() > 0
+    xs = milc.qudaInvertMsrc(0.3, [src, src], tol=1e-7, maxiter=500)
+    assert len(xs) == 2
+    # clover solve through the MILC entry
+    wsrc = torch.view_as_complex(torch.randn(geo_v, 4, 3, 2, generator=gen,
+                                             dtype=torch.float64))
+    xc = milc.qudaCloverInvert(0.12, 1.0, wsrc, tol=1e-8, maxiter=500)
+    assert xc.shape == wsrc.shape
+    # links/observables/momenta
+    w = milc.qudaLoadUnitarizedLink()
+    assert w.shape == u.shape
+    pl = milc.qudaPolyakovLoop()
+    tr = milc.qudaGaugeLoopTrace([[0, 1, 8 + 0, 8 + 1]])
+    mom = torch.zeros_like(u)
+    milc.qudaMomLoad(mom)
+    assert milc.qudaMomSave() is mom
+    milc.qudaRephase(True)
+    milc.qudaFreeCloverField()
+    milc.qudaFreeGaugeField()
+    milc.qudaFinalize()
