@@ -134,11 +134,11 @@ def qudaCloverInvert(kappa: float, csw: float, source_milc: torch.Tensor, *,
                         solution_type=api.SolutionType.MATPC, tol=tol,
                         maxiter=maxiter)
     api.load_clover_quda(p)
-    b4 = source_milc.reshape(geo.volume, 4, 3)
-    from ..fields.interop import spinor_from_qdp, spinor_to_qdp
-    b = spinor_from_qdp(b4, geo)
+    # MILC wilson-spinor site order == lex; convert to the cb oracle order
+    from ..fields.geometry import checkerboard_join, checkerboard_split
+    b = checkerboard_split(source_milc.reshape(geo.volume, 4, 3), geo)
     x = api.invert_quda(b, p)
-    return spinor_to_qdp(x, geo)
+    return checkerboard_join(x, geo)
 
 
 def qudaCloverMultishiftInvert(kappa: float, csw: float, offsets,

@@ -111,7 +111,7 @@ def test_milc_extended_surface():
     w = milc.qudaLoadUnitarizedLink()
     assert w.shape == u.shape
     pl = milc.qudaPolyakovLoop()
-    tr = milc.qudaGaugeLoopTrace([[0, 1, 8 + 0, 8 + 1]])
+    tr = milc.qudaGaugeLoopTrace([[1, 2, -1, -2]])  # (0,1) plaquette
     mom = torch.zeros_like(u)
     milc.qudaMomLoad(mom)
     assert milc.qudaMomSave() is mom
