@@ -241,16 +241,21 @@ def qudaMomSave() -> torch.Tensor:
 
 
 def qudaRephase(flag: bool) -> None:
-    """Apply/remove staggered phases on the resident field (ref
+    """MILC keeps eta-phased links resident; this engine generates the
+    phases in-kernel, so rephasing toggles the staggered_phase_applied
+    load convention (phased input links are stripped at load — ref
     qudaRephase -> gauge_phase.cuh role)."""
+    from dataclasses import replace
     from ..ops.reference import staggered_phases
     geo = _geo()
     u = api._R.u_complex.clone()
-    for par in (0, 1):
-        ph = staggered_phases(geo, par).to(u.device)
-        for mu in range(4):
-            u[mu, par] = u[mu, par] * ph[:, mu].to(u.dtype).reshape(-1, 1, 1)
-    gp = api._R.gauge_param
+    if flag:
+        for par in (0, 1):
+            ph = staggered_phases(geo, par).to(u.device)
+            for mu in range(4):
+                u[mu, par] = u[mu, par] * ph[:, mu].to(u.dtype).reshape(
+                    -1, 1, 1)
+    gp = replace(api._R.gauge_param, staggered_phase_applied=bool(flag))
     api.load_gauge_quda(u, gp)
 
 
