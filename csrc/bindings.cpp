@@ -251,7 +251,11 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
                           at::Tensor x_n, at::Tensor y, at::Tensor y_n,
                           int64_t Vcb, int64_t sites, double c2 = 0.0,
                           double d2 = 0.0, int64_t ncomp = 24,
-                          bool deterministic = false) {
+                          bool deterministic = false,
+                          c10::optional<at::Tensor> z = c10::nullopt,
+                          c10::optional<at::Tensor> z_n = c10::nullopt,
+                          c10::optional<at::Tensor> w = c10::nullopt,
+                          c10::optional<at::Tensor> w_n = c10::nullopt) {
   BlasCall c{};
   c.op = (int)op;
   c.prec = prec_of(x);
@@ -263,10 +267,13 @@ static at::Tensor blas_op(int64_t op, double a, double b, at::Tensor x,
   c.det = deterministic;
   c.x = field_of(x, x_n, Vcb);
   c.y = field_of(y, y_n, Vcb);
+  if (z) c.z = field_of(*z, *z_n, Vcb);
+  if (w) c.w = field_of(*w, *w_n, Vcb);
   c.sites = sites;
   at::Tensor result;
   bool reduction = (op == BLAS_AXPY_NORM2 || op == BLAS_XMY_NORM2 ||
-                    op == BLAS_NORM2 || op == BLAS_REDOT || op == BLAS_CDOT);
+                    op == BLAS_NORM2 || op == BLAS_REDOT || op == BLAS_CDOT ||
+                    op == BLAS_TRIPLE_CG);
   if (reduction) {
     if (deterministic) {
       long g = (sites + 255) / 256;
@@ -566,7 +573,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("op"), py::arg("a"), py::arg("b"), py::arg("x"),
         py::arg("x_n"), py::arg("y"), py::arg("y_n"), py::arg("Vcb"),
         py::arg("sites"), py::arg("c2") = 0.0, py::arg("d2") = 0.0,
-        py::arg("ncomp") = 24, py::arg("deterministic") = false);
+        py::arg("ncomp") = 24, py::arg("deterministic") = false,
+        py::arg("z") = py::none(), py::arg("z_n") = py::none(),
+        py::arg("w") = py::none(), py::arg("w_n") = py::none());
   m.def("convert", &convert, "precision conversion copy", py::arg("dst"),
         py::arg("dst_n"), py::arg("src"), py::arg("src_n"), py::arg("Vcb"),
         py::arg("sites"), py::arg("ncomp") = 24);
@@ -586,6 +595,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.attr("BLAS_XMY_NORM2") = (int)BLAS_XMY_NORM2;
   m.attr("BLAS_SCAL") = (int)BLAS_SCAL;
   m.attr("BLAS_NORM2") = (int)BLAS_NORM2;
+  m.attr("BLAS_TRIPLE_CG") = (int)BLAS_TRIPLE_CG;
   m.attr("BLAS_REDOT") = (int)BLAS_REDOT;
   m.attr("BLAS_CDOT") = (int)BLAS_CDOT;
   m.attr("BLAS_CAXPBY") = (int)BLAS_CAXPBY;

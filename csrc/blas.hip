@@ -219,6 +219,33 @@ int grid_for(long sites) {
 }
 }  // namespace
 
+// fused CG inner update (ref reduce_core.cuh tripleCGUpdate): one pass
+// does the solution accumulation AND the residual update + its norm —
+// two launches and one full residual re-read fewer per iteration
+template <typename A>
+__global__ __launch_bounds__(256) void k_triple_cg(
+    typename A::R a, A p, A ap, A x, A r, long sites, double *result,
+    bool det) {
+  using R = typename A::R;
+  double acc = 0.0;
+  GRID_STRIDE(g, sites) {
+    cplx<R> pv[A::NCPLX], av[A::NCPLX], xv[A::NCPLX], rv[A::NCPLX];
+    p.load_v(pv, g);
+    ap.load_v(av, g);
+    x.load_v(xv, g);
+    r.load_v(rv, g);
+#pragma unroll
+    for (int k = 0; k < A::NCPLX; ++k) {
+      xv[k] = xv[k] + a * pv[k];
+      rv[k] = rv[k] - a * av[k];
+      acc += (double)rv[k].re * rv[k].re + (double)rv[k].im * rv[k].im;
+    }
+    x.store_v(xv, g);
+    r.store_v(rv, g);
+  }
+  block_atomic_add(acc, result, det, 2, 0);
+}
+
 template <typename A>
 static void blas_dispatch(const BlasCall &c, hipStream_t st) {
   using S = typename A::S;
@@ -236,6 +263,13 @@ static void blas_dispatch(const BlasCall &c, hipStream_t st) {
       hipLaunchKernelGGL((k_axpy<A, true>), dim3(gr), dim3(BLK), 0, st,
                          (R)c.a, x, y, n, c.result, c.det);
       break;
+    case BLAS_TRIPLE_CG: {
+      A z{(S *)c.z.data, (float *)c.z.norm, c.z.Vcb};
+      A w{(S *)c.w.data, (float *)c.w.norm, c.w.Vcb};
+      hipLaunchKernelGGL((k_triple_cg<A>), dim3(gr), dim3(BLK), 0, st,
+                         (R)c.a, x, y, z, w, n, c.result, c.det);
+      break;
+    }
     case BLAS_XPAY:
       hipLaunchKernelGGL((k_xpay<A>), dim3(gr), dim3(BLK), 0, st, x, (R)c.a, y, n);
       break;

@@ -47,6 +47,7 @@ enum BlasOp {
   BLAS_REDOT,
   BLAS_CDOT,
   BLAS_CAXPBY,
+  BLAS_TRIPLE_CG,  // x += a p; r -= a Ap; -> ||r||^2 (ref tripleCGUpdate)
 };
 
 struct BlasCall {
@@ -55,6 +56,7 @@ struct BlasCall {
   double a, b;   // scalars (caxpy: a=re, b=im)
   double c, d;   // second complex scalar (caxpby: b=(c,d))
   BlasField x, y;
+  BlasField z, w;  // extra operands (TRIPLE_CG: x=p, y=Ap, z=x, w=r)
   long sites;    // npar * Vcb
   int ncomp;     // reals per site: 24 (Wilson) or 6 (staggered)
   bool det;      // deterministic reduce: result = per-block partials

@@ -112,6 +112,25 @@ def axpy_norm2(a: float, x: SpinorField, y: SpinorField) -> float:
     return norm2(y)
 
 
+def triple_cg_update(a: float, p: SpinorField, Ap: SpinorField,
+                     x: SpinorField, r: SpinorField) -> float:
+    """Fused CG inner update (ref reduce_core.cuh tripleCGUpdate):
+    x += a*p; r -= a*Ap; returns ||r||^2 — one kernel instead of two,
+    one residual pass instead of two."""
+    if on_gpu(p, Ap, x, r):
+        ext = hip_ext()
+        res = ext.blas_op(ext.BLAS_TRIPLE_CG, float(a), 0.0, p.data,
+                          norm_or_empty(p), Ap.data, norm_or_empty(Ap),
+                          p.volume_cb, p.n_parity * p.volume_cb,
+                          ncomp=p.ncomp,
+                          deterministic=deterministic_reductions(),
+                          z=x.data, z_n=norm_or_empty(x),
+                          w=r.data, w_n=norm_or_empty(r))
+        return allreduce_sum(res[0].item())
+    axpy(a, p, x)
+    return axpy_norm2(-a, Ap, r)
+
+
 def xmy_norm2(x: SpinorField, y: SpinorField) -> float:
     """y = x - y; returns ||y||^2."""
     if on_gpu(x, y):

@@ -111,10 +111,10 @@ def cg_solve(op, x: SpinorField, b: SpinorField, *,
         if pAp <= 0.0:
             break  # breakdown (ref: inv_cg_quda.cpp:265 checks)
         alpha = r2 / pAp
-        blas.axpy(alpha, p, x_s)
         x_s_dirty = True
         r2_old = r2
-        r2 = blas.axpy_norm2(-alpha, Ap, r_s)
+        # fused: x_s += alpha p; r_s -= alpha Ap; r2 = ||r_s||^2
+        r2 = blas.triple_cg_update(alpha, p, Ap, x_s, r_s)
         k += 1
         rnorm = sqrt(r2)
         maxr = max(maxr, rnorm)
