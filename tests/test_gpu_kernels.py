@@ -350,3 +350,33 @@ def test_dslash_recon8_gpu(setup, prec):
         err = ((out.to_complex().cpu()[0] - want).abs().max()
                / want.abs().max()).item()
         assert err < TOL[prec], (prec, dagger, err)
+
+
+@pytest.mark.parametrize("prec", ["double", "single", "half"])
+def test_triple_cg_update_gpu(setup, prec):
+    """Fused x += a p; r -= a Ap; ||r||^2 vs the two-op composition."""
+    geo, g, psi, chi, A = setup
+    gen = torch.Generator().manual_seed(515)
+
+    def mk(seed):
+        v = torch.view_as_complex(torch.randn(1, geo.volume_cb, 4, 3, 2,
+                                              generator=gen,
+                                              dtype=torch.float64))
+        return SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+            v.cuda())
+
+    p, ap = mk(1), mk(2)
+    x1, r1 = mk(3), mk(4)
+    x2 = SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+        x1.to_complex())
+    r2 = SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+        r1.to_complex())
+    a = 0.37
+    fused = blas.triple_cg_update(a, p, ap, x1, r1)
+    blas.axpy(a, p, x2)
+    ref_n = blas.axpy_norm2(-a, ap, r2)
+    tol = {"double": 1e-12, "single": 1e-5, "half": 5e-2}[prec]
+    assert abs(fused - ref_n) < tol * (abs(ref_n) + 1), (prec, fused, ref_n)
+    dx = (x1.to_complex() - x2.to_complex()).abs().max().item()
+    dr = (r1.to_complex() - r2.to_complex()).abs().max().item()
+    assert dx < tol * 10 and dr < tol * 10, (prec, dx, dr)
