@@ -33,6 +33,7 @@ KERNEL_SOURCES = [
     "dslash_dwf.hip",
     "dslash_wilson_mrhs.hip",
     "coarse.hip",
+    "heatbath.hip",
 ]
 BINDING_SOURCES = ["bindings.cpp"]
 

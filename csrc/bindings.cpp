@@ -513,6 +513,25 @@ static void pack_face_ptr(int64_t dst, int64_t dst_nrm, at::Tensor in,
   check_launch("pack_face_ptr");
 }
 
+static void heatbath_sweep_dir(at::Tensor u, std::vector<int64_t> dims,
+                               int64_t parity_offset, int64_t Vcb,
+                               int64_t parity, int64_t mu, double beta_eff,
+                               int64_t seed, int64_t mode) {
+  TORCH_CHECK(u.is_contiguous() && u.scalar_type() == at::kComplexDouble);
+  HeatbathCall c{};
+  c.u = u.data_ptr();
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.parity = (int)parity;
+  c.mu = (int)mu;
+  c.beta_eff = beta_eff;
+  c.seed = (unsigned long long)seed;
+  c.mode = (int)mode;
+  launch_heatbath(c, stream());
+  check_launch("heatbath");
+}
+
 static void coarse_dslash_mfma(at::Tensor mats, at::Tensor nbr9,
                                at::Tensor c, at::Tensor out, int64_t Na,
                                int64_t Nc, int64_t NR) {
@@ -555,6 +574,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ghost_nrm"), py::arg("face_cb"), py::arg("comm_mask"),
         py::arg("kt"), py::arg("v_stride") = 0,
         py::arg("s_offsets") = std::vector<int64_t>{});
+  m.def("heatbath_sweep_dir", &heatbath_sweep_dir,
+        "SU(2)-subgroup heatbath/overrelax update of one (parity, mu)");
   m.def("coarse_dslash_mfma", &coarse_dslash_mfma,
         "coarse-grid 9-matrix dslash on f32 MFMA tiles");
   m.def("dslash_staggered", &dslash_staggered, "naive staggered dslash");

@@ -215,6 +215,21 @@ struct CoarseMfmaCall {
 };
 void launch_coarse_dslash_mfma(const CoarseMfmaCall &c, hipStream_t st);
 
+// ---------------------------------------------------------------------------
+// native heatbath / overrelaxation sweep (csrc/heatbath.hip)
+struct HeatbathCall {
+  void *u;  // [4][2][Vcb][3][3] complex double (oracle layout)
+  int Xdim[4];
+  int parity_offset;
+  long Vcb;
+  int parity;
+  int mu;
+  double beta_eff;
+  unsigned long long seed;
+  int mode;  // 0 heatbath, 1 overrelax
+};
+void launch_heatbath(const HeatbathCall &c, hipStream_t st);
+
 #define QA_ZMAX 32
 
 struct ZCoef {  // host-filled complex coefficient tables (double re/im)
