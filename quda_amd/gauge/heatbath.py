@@ -114,9 +114,38 @@ def _sweep(u, geo, beta, gen, mode: str):
     return _from_lex(U, geo)
 
 
+def _native_sweep(u, geo, beta, seed, mode: int):
+    """One sweep through the hand-written HIP kernel (csrc/heatbath.hip):
+    8 launches (mu x parity), staples + all 3 SU(2) subgroup hits fused
+    per site, counter-based in-kernel RNG."""
+    from ..ops.dispatch import hip_ext
+    ext = hip_ext()
+    uc = u if u.is_contiguous() else u.contiguous()
+    for mu in range(4):
+        for p in (0, 1):
+            ext.heatbath_sweep_dir(uc, list(geo.dims), geo.parity_offset,
+                                   geo.volume_cb, p, mu, 2.0 * beta / 3.0,
+                                   int(seed or 0), mode)
+    return uc
+
+
+def _use_native(u, geo) -> bool:
+    from ..parallel import comms
+    import os
+    if os.environ.get("QUDA_AMD_NATIVE_HEATBATH", "1") == "0":
+        return False
+    return (u.device.type == "cuda" and u.dtype == torch.complex128
+            and not comms.comm_mask())
+
+
 def heatbath_sweep(u, geo: LatticeGeometry, beta: float,
                    seed: Optional[int] = None):
-    """One Cabibbo-Marinari heatbath sweep (all links, both parities)."""
+    """One Cabibbo-Marinari heatbath sweep (all links, both parities).
+    On a single GPU the native HIP kernel runs (8 launches/sweep); the
+    torch path is the multi-rank / CPU / deterministic-host-RNG route."""
+    if _use_native(u, geo):
+        return _native_sweep(u, geo, beta, seed if seed is not None else 1,
+                             0)
     gen = torch.Generator()
     if seed is not None:
         gen.manual_seed(seed)
@@ -126,6 +155,8 @@ def heatbath_sweep(u, geo: LatticeGeometry, beta: float,
 def overrelax_sweep(u, geo: LatticeGeometry, beta: float,
                     seed: Optional[int] = None):
     """One SU(2)-subgroup overrelaxation sweep (action-preserving)."""
+    if _use_native(u, geo):
+        return _native_sweep(u, geo, beta, 0, 1)
     gen = torch.Generator()
     if seed is not None:
         gen.manual_seed(seed)
