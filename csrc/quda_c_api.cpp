@@ -240,10 +240,16 @@ static py::object make_invert_param(const QudaAmdInvertParam *p) {
   py::object it = enums.attr("InverterType")(inv_name(p->inv_type));
   py::object st = enums.attr("SolutionType")(
       p->solution_type == QUDA_AMD_MATPC_SOLUTION ? "matpc" : "mat");
+  // kappa-convention bridge: the engine's stencil applies TRUE
+  // projectors P = (1 -+ gamma)/2, so kappa_engine = 2 kappa_standard
+  // and csw_engine = csw_standard / 2 reproduce the quda.h-convention
+  // operator exactly (free kappa_c: engine 1/4 <-> standard 1/8); the C
+  // ABI speaks the STANDARD convention (verified by the free-field
+  // kappa_c check in c_interface_test).
   return g_api.attr("InvertParam")(
       "dslash_type"_a = dt, "inv_type"_a = it, "solution_type"_a = st,
-      "kappa"_a = p->kappa, "mass"_a = p->mass, "mu"_a = p->mu,
-      "clover_csw"_a = p->clover_csw, "tol"_a = p->tol,
+      "kappa"_a = 2.0 * p->kappa, "mass"_a = p->mass, "mu"_a = p->mu,
+      "clover_csw"_a = 0.5 * p->clover_csw, "tol"_a = p->tol,
       "maxiter"_a = p->maxiter, "reliable_delta"_a = p->reliable_delta,
       "cuda_prec"_a = prec_str(p->cuda_prec),
       "cuda_prec_sloppy"_a = prec_str(p->cuda_prec_sloppy),

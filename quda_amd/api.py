@@ -94,6 +94,10 @@ class InvertParam:
     dslash_type: DslashType = DslashType.WILSON
     inv_type: InverterType = InverterType.CG
     solution_type: SolutionType = SolutionType.MAT
+    # NORMALIZATION: the stencil applies true projectors P = (1-+g)/2,
+    # so M = 1 - kappa sum_mu P U psi and the FREE kappa_c is 1/4 —
+    # kappa here = 2 x the standard Wilson convention (and csw here =
+    # standard csw / 2 at matched operator). The C ABI converts.
     kappa: float = 0.135
     mass: float = 0.05          # staggered
     mu: float = 0.0             # twisted

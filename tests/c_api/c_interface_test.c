@@ -123,6 +123,18 @@ int main(void) {
   CHECK(d01 > 1e-12, "shifted solutions differ");
   printf("multishift: %d iters\n", ip.iter);
 
+  /* ---- kappa-convention check: standard quda.h normalization means
+   *      M (constant spinor) = (1 - 8 kappa) (constant spinor) on the
+   *      free field ---- */
+  ip.dslash_type = QUDA_AMD_WILSON_DSLASH;
+  ip.solution_type = QUDA_AMD_MAT_SOLUTION;
+  ip.kappa = 0.1;
+  for (size_t i = 0; i < ns; ++i) { b[i].re = 1.0; b[i].im = 0.0; }
+  CHECK(MatQuda(mx, b, &ip) == 0, "MatQuda(convention)");
+  CHECK(fabs(mx[0].re - (1.0 - 8.0 * 0.1)) < 1e-10,
+        "standard Wilson kappa normalization (M psi0 = (1-8k) psi0)");
+  for (size_t i = 0; i < ns; ++i) { b[i].re = lcg(); b[i].im = lcg(); }
+
   /* ---- dslashQuda parity application on the free field ---- */
   ip.solution_type = QUDA_AMD_MAT_SOLUTION;
   CHECK(dslashQuda(mx, b, &ip, 0) == 0, "dslashQuda");
