@@ -185,6 +185,7 @@ def main():
                 "global_batch": 1,
                 "seq_len": geo.volume,
                 "kappa": args.kappa,
+                "kappa_standard_convention": args.kappa / 2,
                 "csw": CSW,
                 "solve_tol": args.tol,
                 "reliable_delta": args.delta,
