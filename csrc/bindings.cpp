@@ -532,6 +532,24 @@ static void heatbath_sweep_dir(at::Tensor u, std::vector<int64_t> dims,
   check_launch("heatbath");
 }
 
+static void stout_smear_dir(at::Tensor out, at::Tensor in,
+                            std::vector<int64_t> dims,
+                            int64_t parity_offset, int64_t Vcb, int64_t mu,
+                            double rho) {
+  TORCH_CHECK(out.is_contiguous() && in.is_contiguous() &&
+              in.scalar_type() == at::kComplexDouble);
+  StoutCall c{};
+  c.out = out.data_ptr();
+  c.in = in.data_ptr();
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.mu = (int)mu;
+  c.rho = rho;
+  launch_stout(c, stream());
+  check_launch("stout");
+}
+
 static void coarse_dslash_mfma(at::Tensor mats, at::Tensor nbr9,
                                at::Tensor c, at::Tensor out, int64_t Na,
                                int64_t Nc, int64_t NR) {
@@ -574,6 +592,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ghost_nrm"), py::arg("face_cb"), py::arg("comm_mask"),
         py::arg("kt"), py::arg("v_stride") = 0,
         py::arg("s_offsets") = std::vector<int64_t>{});
+  m.def("stout_smear_dir", &stout_smear_dir,
+        "native stout smear of one direction (exp via scale-and-square)");
   m.def("heatbath_sweep_dir", &heatbath_sweep_dir,
         "SU(2)-subgroup heatbath/overrelax update of one (parity, mu)");
   m.def("coarse_dslash_mfma", &coarse_dslash_mfma,

@@ -244,3 +244,159 @@ void launch_heatbath(const HeatbathCall &c, hipStream_t st) {
   hipLaunchKernelGGL(k_heatbath, dim3(grid), dim3(128), 0, st, (cd *)c.u, d,
                      c.parity, c.mu, c.beta_eff, c.seed, c.mode);
 }
+
+// ---------------------------------------------------------------------------
+// Native stout smearing (role of kernels/gauge_stout.cuh):
+//   U'_mu = exp(rho * TA[S U_mu^dag]) U_mu
+// exp of the traceless-antihermitian Q by scaling-and-squaring with an
+// 8-term Taylor series (||rho Q|| <~ 1 for physical rho: scale to < 1/4,
+// exact to ~1e-14 — branch-free, unlike the Morningstar-Peardon
+// Cayley-Hamilton form with its small-w singularities).
+// Reads `gin`, writes `gout` (all links of one mu per launch; every
+// staple comes from the OLD field, so the update is order-free).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+__device__ __forceinline__ void mat_scale_add_eye(M3 &o, const M3 &a,
+                                                  double s) {
+#pragma unroll
+  for (int i = 0; i < 3; ++i)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+      o[i][j] = {a[i][j].re * s + (i == j ? 1.0 : 0.0), a[i][j].im * s};
+    }
+}
+
+__device__ __forceinline__ void mat_exp_ta(M3 &E, const M3 &Q) {
+  // scale Q by 2^-s so its 1-norm < 0.25
+  double n = 0.0;
+#pragma unroll
+  for (int i = 0; i < 3; ++i)
+#pragma unroll
+    for (int j = 0; j < 3; ++j)
+      n += fabs(Q[i][j].re) + fabs(Q[i][j].im);
+  int s = 0;
+  double sc = 1.0;
+  while (n * sc > 0.25 && s < 40) {
+    sc *= 0.5;
+    ++s;
+  }
+  M3 A;
+#pragma unroll
+  for (int i = 0; i < 3; ++i)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) A[i][j] = {Q[i][j].re * sc, Q[i][j].im * sc};
+  // 8-term Horner Taylor: E = I + A(I + A/2 (I + A/3 (...)))
+  M3 T, P;
+  mat_scale_add_eye(T, A, 1.0 / 8.0);  // placeholder start: I + A/8
+  for (int k = 7; k >= 1; --k) {
+    mat_mul(P, A, T);
+#pragma unroll
+    for (int i = 0; i < 3; ++i)
+#pragma unroll
+      for (int j = 0; j < 3; ++j) {
+        double f = 1.0 / k;
+        P[i][j] = {P[i][j].re * f + (i == j ? 1.0 : 0.0), P[i][j].im * f};
+      }
+#pragma unroll
+    for (int i = 0; i < 3; ++i)
+#pragma unroll
+      for (int j = 0; j < 3; ++j) T[i][j] = P[i][j];
+  }
+  // square s times
+  for (int k = 0; k < s; ++k) {
+    mat_mul(P, T, T);
+#pragma unroll
+    for (int i = 0; i < 3; ++i)
+#pragma unroll
+      for (int j = 0; j < 3; ++j) T[i][j] = P[i][j];
+  }
+#pragma unroll
+  for (int i = 0; i < 3; ++i)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) E[i][j] = T[i][j];
+}
+
+}  // namespace
+
+__global__ __launch_bounds__(128) void k_stout(
+    cd *__restrict__ gout, const cd *__restrict__ gin, LatDims d, int mu,
+    double rho) {
+  long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long V = d.Vcb;
+  if (t >= 2 * V) return;
+  int parity = (int)(t / V);
+  long i = t - (long)parity * V;
+  int xc[4];
+  coords_from_cb(xc, i, d, parity);
+
+  M3 S;
+#pragma unroll
+  for (int k = 0; k < 9; ++k) S[k / 3][k % 3] = {0.0, 0.0};
+  const int op = 1 - parity;
+  long x_pmu = neighbor_cb(xc, mu, +1, d);
+#pragma unroll
+  for (int nu = 0; nu < 4; ++nu) {
+    if (nu == mu) continue;
+    M3 a, b, c, tm, st;
+    long x_pnu = neighbor_cb(xc, nu, +1, d);
+    load_link(a, gin, V, nu, parity, i);
+    load_link(b, gin, V, mu, op, x_pnu);
+    load_link(c, gin, V, nu, op, x_pmu);
+    mat_mul(tm, a, b);
+    mat_mul_dag(st, tm, c);
+    mat_acc(S, st);
+    long x_mnu = neighbor_cb(xc, nu, -1, d);
+    int y[4] = {xc[0], xc[1], xc[2], xc[3]};
+    y[nu] = y[nu] - 1;
+    if (y[nu] < 0) y[nu] += d.X[nu];
+    y[mu] = y[mu] + 1;
+    if (y[mu] >= d.X[mu]) y[mu] -= d.X[mu];
+    long x_pmu_mnu = cb_from_coords(y, d);
+    load_link(a, gin, V, nu, op, x_mnu);
+    load_link(b, gin, V, mu, op, x_mnu);
+    load_link(c, gin, V, nu, parity, x_pmu_mnu);
+    mat_dag_mul(tm, a, b);
+    mat_mul(st, tm, c);
+    mat_acc(S, st);
+  }
+
+  M3 U, W, Q, E, Un;
+  load_link(U, gin, V, mu, parity, i);
+  mat_mul_dag(W, S, U);  // S U^dag
+  // Q = rho * TA[W] = rho * ((W - W^dag)/2 - tr/3)
+  cd tr = {0.0, 0.0};
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+      Q[i2][j] = {0.5 * (W[i2][j].re - W[j][i2].re),
+                  0.5 * (W[i2][j].im + W[j][i2].im)};
+    }
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2) tr = cadd(tr, Q[i2][i2]);
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2) {
+    Q[i2][i2].re -= tr.re / 3.0;
+    Q[i2][i2].im -= tr.im / 3.0;
+  }
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+      Q[i2][j].re *= rho;
+      Q[i2][j].im *= rho;
+    }
+  mat_exp_ta(E, Q);
+  mat_mul(Un, E, U);
+  store_link(gout, V, mu, parity, i, Un);
+}
+
+void launch_stout(const StoutCall &c, hipStream_t st) {
+  LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset,
+            c.Vcb};
+  int grid = (int)((2 * c.Vcb + 127) / 128);
+  hipLaunchKernelGGL(k_stout, dim3(grid), dim3(128), 0, st, (cd *)c.out,
+                     (const cd *)c.in, d, c.mu, c.rho);
+}

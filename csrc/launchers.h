@@ -230,6 +230,17 @@ struct HeatbathCall {
 };
 void launch_heatbath(const HeatbathCall &c, hipStream_t st);
 
+struct StoutCall {
+  void *out;       // [4][2][Vcb][3][3] complex double
+  const void *in;
+  int Xdim[4];
+  int parity_offset;
+  long Vcb;
+  int mu;
+  double rho;
+};
+void launch_stout(const StoutCall &c, hipStream_t st);
+
 #define QA_ZMAX 32
 
 struct ZCoef {  // host-filled complex coefficient tables (double re/im)

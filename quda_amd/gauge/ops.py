@@ -143,6 +143,24 @@ def stout_smear(u: torch.Tensor, geo: LatticeGeometry, rho: float,
     Morningstar-Peardon: U' = exp(i rho Q_herm) U; here with antihermitian
     Q: U' = exp(rho * TA[staple U^d... ]) — we use
     U'_mu = exp(rho * TA[S U_mu^d]) U_mu (ref: lib/gauge_stout.cu)."""
+    from ..parallel import comms
+    import os
+    if (u.device.type == "cuda" and u.dtype == torch.complex128
+            and not comms.comm_mask()
+            and os.environ.get("QUDA_AMD_NATIVE_SMEAR", "1") != "0"):
+        # native HIP kernel (csrc/heatbath.hip k_stout): staple + TA +
+        # scale-and-square exp fused per site, 4 launches per iteration
+        from ..ops.dispatch import hip_ext
+        ext = hip_ext()
+        cur = u.contiguous().clone()
+        nxt = torch.empty_like(cur)
+        for _ in range(n_iter):
+            for mu in range(4):
+                ext.stout_smear_dir(nxt, cur, list(geo.dims),
+                                    geo.parity_offset, geo.volume_cb, mu,
+                                    float(rho))
+            cur, nxt = nxt, cur
+        return cur
     out = u
     for _ in range(n_iter):
         U = _to_lex(out, geo)

@@ -402,3 +402,28 @@ def test_native_heatbath_kernel():
     eye = torch.eye(3, dtype=u.dtype, device=u.device)
     assert (uu - eye).abs().max().item() < 1e-10
     print(f"native heatbath: {dt/12*1000:.1f} ms/sweep at 8^4")
+
+
+@pytest.mark.gpu
+def test_native_stout_kernel():
+    """csrc/heatbath.hip k_stout vs the torch stout path: deterministic
+    math (staple + TA + matrix exponential), must agree to fp64
+    roundoff."""
+    import os
+    import torch
+    from quda_amd.fields.gauge import GaugeField
+    from quda_amd.fields.geometry import LatticeGeometry
+    from quda_amd.gauge import ops as gops
+    geo = LatticeGeometry((8, 8, 8, 8))
+    u = GaugeField(geo, "double").random_su3_(seed=641).to_complex().cuda()
+    un = gops.stout_smear(u, geo, 0.12, 2)           # native path
+    os.environ["QUDA_AMD_NATIVE_SMEAR"] = "0"
+    try:
+        ut = gops.stout_smear(u, geo, 0.12, 2)       # torch path
+    finally:
+        os.environ["QUDA_AMD_NATIVE_SMEAR"] = "1"
+    err = (un - ut).abs().max().item()
+    assert err < 1e-11, err
+    # unitarity preserved
+    eye = torch.eye(3, dtype=u.dtype, device=u.device)
+    assert ((un @ un.conj().mT) - eye).abs().max().item() < 1e-11
