@@ -37,6 +37,8 @@ def main():
     ap.add_argument("--beta", type=float, default=6.0)
     ap.add_argument("--block", default="4,4,4,4")
     ap.add_argument("--nvec", type=int, default=8)
+    ap.add_argument("--levels", type=int, default=2)
+    ap.add_argument("--nvec2", type=int, default=8)
     ap.add_argument("--tol", type=float, default=1e-8)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
@@ -84,6 +86,8 @@ def main():
         t0 = time.perf_counter()
         mg = MG(d, MGParam(block=tuple(int(x) for x in args.block.split(",")),
                            n_vec=args.nvec, nu_post=4, coarse_tol=5e-2,
+                           levels=args.levels, n_vec2=args.nvec2,
+                           block2=(2, 2, 2, 2),
                            null_tol=1e-4, null_maxiter=300))
         sync()
         t_setup = time.perf_counter() - t0
