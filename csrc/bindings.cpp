@@ -550,6 +550,39 @@ static void stout_smear_dir(at::Tensor out, at::Tensor in,
   check_launch("stout");
 }
 
+static void flow_zmat_dir(at::Tensor Z, at::Tensor in,
+                          std::vector<int64_t> dims, int64_t parity_offset,
+                          int64_t Vcb, int64_t mu, double eps) {
+  TORCH_CHECK(Z.is_contiguous() && in.is_contiguous());
+  StoutCall c{};
+  c.out = Z.data_ptr();
+  c.in = in.data_ptr();
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.mu = (int)mu;
+  c.rho = eps;
+  launch_zmat(c, stream());
+  check_launch("zmat");
+}
+
+static void flow_expmul_dir(at::Tensor out, at::Tensor in, at::Tensor Zc,
+                            std::vector<int64_t> dims,
+                            int64_t parity_offset, int64_t Vcb, int64_t mu) {
+  TORCH_CHECK(out.is_contiguous() && in.is_contiguous() &&
+              Zc.is_contiguous());
+  StoutCall c{};
+  c.out = out.data_ptr();
+  c.in = in.data_ptr();
+  c.aux = Zc.data_ptr();
+  for (int i = 0; i < 4; ++i) c.Xdim[i] = (int)dims[i];
+  c.parity_offset = (int)parity_offset;
+  c.Vcb = Vcb;
+  c.mu = (int)mu;
+  launch_expmul(c, stream());
+  check_launch("expmul");
+}
+
 static void coarse_dslash_mfma(at::Tensor mats, at::Tensor nbr9,
                                at::Tensor c, at::Tensor out, int64_t Na,
                                int64_t Nc, int64_t NR) {
@@ -592,6 +625,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ghost_nrm"), py::arg("face_cb"), py::arg("comm_mask"),
         py::arg("kt"), py::arg("v_stride") = 0,
         py::arg("s_offsets") = std::vector<int64_t>{});
+  m.def("flow_zmat_dir", &flow_zmat_dir, "wilson-flow Z = eps TA[S U^d]");
+  m.def("flow_expmul_dir", &flow_expmul_dir, "U' = exp(Zc) U");
   m.def("stout_smear_dir", &stout_smear_dir,
         "native stout smear of one direction (exp via scale-and-square)");
   m.def("heatbath_sweep_dir", &heatbath_sweep_dir,

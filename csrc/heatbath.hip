@@ -400,3 +400,110 @@ void launch_stout(const StoutCall &c, hipStream_t st) {
   hipLaunchKernelGGL(k_stout, dim3(grid), dim3(128), 0, st, (cd *)c.out,
                      (const cd *)c.in, d, c.mu, c.rho);
 }
+
+// ---------------------------------------------------------------------------
+// Wilson-flow building blocks (role of kernels/gauge_wilson_flow.cuh):
+// k_zmat computes Z_mu = eps * TA[S U^dag] into a gauge-shaped tensor;
+// k_expmul applies U' = exp(Zc) U for a host-combined Zc (the RK3 stage
+// combinations are cheap elementwise torch ops on the Z tensors).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(128) void k_zmat(
+    cd *__restrict__ Z, const cd *__restrict__ gin, LatDims d, int mu,
+    double eps) {
+  long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long V = d.Vcb;
+  if (t >= 2 * V) return;
+  int parity = (int)(t / V);
+  long i = t - (long)parity * V;
+  int xc[4];
+  coords_from_cb(xc, i, d, parity);
+
+  M3 S;
+#pragma unroll
+  for (int k = 0; k < 9; ++k) S[k / 3][k % 3] = {0.0, 0.0};
+  const int op = 1 - parity;
+  long x_pmu = neighbor_cb(xc, mu, +1, d);
+#pragma unroll
+  for (int nu = 0; nu < 4; ++nu) {
+    if (nu == mu) continue;
+    M3 a, b, c, tm, st;
+    long x_pnu = neighbor_cb(xc, nu, +1, d);
+    load_link(a, gin, V, nu, parity, i);
+    load_link(b, gin, V, mu, op, x_pnu);
+    load_link(c, gin, V, nu, op, x_pmu);
+    mat_mul(tm, a, b);
+    mat_mul_dag(st, tm, c);
+    mat_acc(S, st);
+    long x_mnu = neighbor_cb(xc, nu, -1, d);
+    int y[4] = {xc[0], xc[1], xc[2], xc[3]};
+    y[nu] = y[nu] - 1;
+    if (y[nu] < 0) y[nu] += d.X[nu];
+    y[mu] = y[mu] + 1;
+    if (y[mu] >= d.X[mu]) y[mu] -= d.X[mu];
+    long x_pmu_mnu = cb_from_coords(y, d);
+    load_link(a, gin, V, nu, op, x_mnu);
+    load_link(b, gin, V, mu, op, x_mnu);
+    load_link(c, gin, V, nu, parity, x_pmu_mnu);
+    mat_dag_mul(tm, a, b);
+    mat_mul(st, tm, c);
+    mat_acc(S, st);
+  }
+  M3 U, W, Q;
+  load_link(U, gin, V, mu, parity, i);
+  mat_mul_dag(W, S, U);
+  cd tr = {0.0, 0.0};
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2)
+#pragma unroll
+    for (int j = 0; j < 3; ++j)
+      Q[i2][j] = {0.5 * (W[i2][j].re - W[j][i2].re),
+                  0.5 * (W[i2][j].im + W[j][i2].im)};
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2) tr = cadd(tr, Q[i2][i2]);
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2) {
+    Q[i2][i2].re -= tr.re / 3.0;
+    Q[i2][i2].im -= tr.im / 3.0;
+  }
+#pragma unroll
+  for (int i2 = 0; i2 < 3; ++i2)
+#pragma unroll
+    for (int j = 0; j < 3; ++j) {
+      Q[i2][j].re *= eps;
+      Q[i2][j].im *= eps;
+    }
+  store_link(Z, V, mu, parity, i, Q);
+}
+
+__global__ __launch_bounds__(128) void k_expmul(
+    cd *__restrict__ gout, const cd *__restrict__ gin,
+    const cd *__restrict__ Zc, LatDims d, int mu) {
+  long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long V = d.Vcb;
+  if (t >= 2 * V) return;
+  int parity = (int)(t / V);
+  long i = t - (long)parity * V;
+  M3 Q, E, U, Un;
+  load_link(Q, Zc, V, mu, parity, i);
+  load_link(U, gin, V, mu, parity, i);
+  mat_exp_ta(E, Q);
+  mat_mul(Un, E, U);
+  store_link(gout, V, mu, parity, i, Un);
+}
+
+void launch_zmat(const StoutCall &c, hipStream_t st) {
+  LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset,
+            c.Vcb};
+  int grid = (int)((2 * c.Vcb + 127) / 128);
+  hipLaunchKernelGGL(k_zmat, dim3(grid), dim3(128), 0, st, (cd *)c.out,
+                     (const cd *)c.in, d, c.mu, c.rho);
+}
+
+void launch_expmul(const StoutCall &c, hipStream_t st) {
+  LatDims d{{c.Xdim[0], c.Xdim[1], c.Xdim[2], c.Xdim[3]}, c.parity_offset,
+            c.Vcb};
+  int grid = (int)((2 * c.Vcb + 127) / 128);
+  hipLaunchKernelGGL(k_expmul, dim3(grid), dim3(128), 0, st, (cd *)c.out,
+                     (const cd *)c.in, (const cd *)c.aux, d, c.mu);
+}

