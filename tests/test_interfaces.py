@@ -119,3 +119,114 @@ def test_milc_extended_surface():
     milc.qudaFreeCloverField()
     milc.qudaFreeGaugeField()
     milc.qudaFinalize()
+
+
+def test_milc_round2_surface():
+    """Round-2 additions: memory helpers, gauge-field handles, phased
+    wrappers, shift/spin-taste/two-link smear, gauge fixing, DD invert,
+    MG create, force-chain pieces (oprod, clover derivative/trace)."""
+    import torch
+    from quda_amd import api
+    from quda_amd.interfaces import milc
+    from quda_amd.fields.gauge import GaugeField
+    from quda_amd.fields.geometry import LatticeGeometry
+    from quda_amd.fields.interop import gauge_to_milc
+    dims = (4, 4, 4, 4)
+    milc.qudaInit()
+    geo = LatticeGeometry(dims)
+    V = geo.volume
+    gen = torch.Generator().manual_seed(321)
+    u = GaugeField(geo, "double").random_su3_(seed=321).to_complex()
+    milc.qudaLoadGaugeField(dims, gauge_to_milc(u, geo))
+
+    # memory + comm helpers
+    buf = milc.qudaAllocatePinned(256)
+    assert buf.numel() == 256
+    milc.qudaFreePinned(buf)
+    milc.qudaFreeManaged(milc.qudaAllocateManaged(64))
+    milc.qudaSetMPICommHandle(0)
+
+    # handles
+    h = milc.qudaCreateGaugeField(dims)
+    um = milc.qudaSaveGaugeField(h)
+    assert torch.allclose(um[0, 0], torch.eye(3, dtype=um.dtype))
+    milc.qudaDestroyGaugeField(h)
+    he = milc.qudaCreateExtendedGaugeField(gauge_to_milc(u, geo), dims, 2)
+    milc.qudaResidentExtendedGaugeField(he)
+
+    # phased observables: plaquette is a closed loop -> phases cancel in
+    # pairs only for 2x identical eta factors; compare against explicit
+    # double-rephase instead of the unphased value
+    p_phased = milc.qudaPlaquettePhased()
+    assert all(abs(x) <= 1.0 + 1e-12 for x in p_phased)
+    meas = milc.qudaGaugeMeasurementsPhased()
+    assert set(meas) == {"plaquette", "polyakov_loop", "qcharge"}
+    # after every phased entry the resident field must be back unphased
+    p0 = milc.qudaPlaquette()
+    tot2, _, _ = __import__("quda_amd.gauge", fromlist=["plaquette"]
+                            ).plaquette(api._R.u_complex, geo)
+    assert abs(p0[0] - tot2) < 1e-12
+
+    # covariant shift: free-field shift of a constant vector is itself
+    milc.qudaLoadGaugeField(dims, gauge_to_milc(
+        torch.eye(3, dtype=torch.complex128).expand(4, 2, geo.volume_cb,
+                                                    3, 3).contiguous(), geo))
+    cvec = torch.ones((V, 3), dtype=torch.complex128)
+    sh = milc.qudaShift(cvec, 0, True)
+    assert (sh - cvec).abs().max() < 1e-14
+    milc.qudaLoadGaugeField(dims, gauge_to_milc(u, geo))
+
+    # spin-taste + two-link smear
+    src = torch.view_as_complex(torch.randn(V, 3, 2, generator=gen,
+                                            dtype=torch.float64))
+    st = milc.qudaSpinTaste(src, "g5-g5")
+    assert st.abs().max() > 0 and st.shape == src.shape
+    sm = milc.qudaTwoLinkGaussianSmear(src, width=1.0, n_steps=2)
+    assert sm.shape == src.shape
+    milc.qudaFreeTwoLink()
+
+    # gauge fixing refreshes the resident field and improves quality
+    from quda_amd.gauge.fix import gauge_fix_quality
+    q0 = gauge_fix_quality(api._R.u_complex, geo)
+    milc.qudaGaugeFixingOVR(4, max_iter=12, tol=1e-30)
+    q1 = gauge_fix_quality(api._R.u_complex, geo)
+    assert q1[0] > q0[0]
+    milc.qudaLoadGaugeField(dims, gauge_to_milc(u, geo))
+
+    # DD invert solves (2m + D) x = b
+    from quda_amd.ops import reference as refops
+    from quda_amd.fields.interop import spinor_from_milc
+    xm = milc.qudaDDInvert(0.3, src, tol=1e-8, maxiter=300)
+    x = spinor_from_milc(xm, geo)
+    b = spinor_from_milc(src, geo)
+    r = refops.mat_staggered(u, x, geo, 0.3)
+    assert (r - b).abs().max() < 1e-6
+
+    # MG create + destroy
+    mg = milc.qudaMultigridCreate(0.12, block=(2, 2, 2, 2), n_vec=2)
+    milc.qudaMultigridDestroy(mg)
+
+    # oprod: free-field check O_mu = psi(x+mu) psi(x)^dag
+    o = milc.qudaComputeOprod([1.0], [src])
+    from quda_amd.parallel.halo import shift_lex
+    psi = src.reshape(V, 3)
+    want = torch.einsum(
+        "xa,xb->xab",
+        shift_lex(psi.unsqueeze(-1), geo, 1, +1).squeeze(-1), psi.conj())
+    assert (o[1] - want).abs().max() < 1e-14
+
+    # clover derivative returns a traceless antihermitian field
+    F = milc.qudaCloverDerivative(torch.randn(2, geo.volume_cb, 3, 3,
+                                              dtype=torch.complex128,
+                                              generator=gen), 0, 1)
+    assert (F + F.conj().mT).abs().max() < 1e-10
+    assert F.diagonal(dim1=-2, dim2=-1).sum(-1).abs().max() < 1e-10
+
+    # clover sigma-trace: antihermitian color matrices per mu<nu pair
+    tr = milc.qudaCloverTrace(0.12, 1.0)
+    assert len(tr) == 6
+    t01 = tr[(0, 1)]
+    assert t01.shape == (2, geo.volume_cb, 3, 3)
+
+    milc.qudaFreeGaugeField()
+    milc.qudaFinalize()
