@@ -16,6 +16,7 @@
 #include <libgen.h>
 #include <pybind11/embed.h>
 
+#include <complex>
 #include <cstring>
 #include <string>
 
@@ -24,6 +25,8 @@
 namespace py = pybind11;
 
 static std::string g_err;
+struct QudaAmdMgHandle;  // fwd (round-2 MG lifecycle, defined below)
+static void attach_precond(pybind11::object &ip, void *handle);
 static bool g_inited = false;
 static py::object g_api;        // quda_amd.api module
 static py::object g_torch;      // torch module
@@ -302,6 +305,7 @@ int invertQuda(void *h_x, const void *h_b, QudaAmdInvertParam *param) {
   py::object b = tensor_view(h_b, spinor_cplx(param, Vcb),
                              spinor_shape(param, Vcb, 2));
   py::object ip = make_invert_param(param);
+  if (param->preconditioner) attach_precond(ip, param->preconditioner);
   py::object x = g_api.attr("invert_quda")(b, ip);
   tensor_out(x, h_x);
   param->iter = ip.attr("iter").cast<int>();
@@ -367,5 +371,224 @@ int plaqQuda(double plaq[3]) {
   plaq[0] = t[0].cast<double>();
   plaq[1] = t[1].cast<double>();
   plaq[2] = t[2].cast<double>();
+  QA_END
+}
+
+// ---------------------------------------------------------------------------
+// round-2 additions: eigensolve, multigrid lifecycle, HMC surface,
+// smearing, observables, gauge fixing, contractions
+// ---------------------------------------------------------------------------
+
+QudaAmdEigParam newQudaAmdEigParam(void) {
+  QudaAmdEigParam p;
+  std::memset(&p, 0, sizeof(p));
+  p.n_ev = 8;
+  p.n_kr = 32;
+  p.tol = 1e-8;
+  p.max_restarts = 100;
+  p.use_norm_op = 1;
+  p.poly_deg = 8;
+  p.a_min = 0.1;
+  p.a_max = 10.0;
+  return p;
+}
+
+QudaAmdMultigridParam newQudaAmdMultigridParam(void) {
+  QudaAmdMultigridParam p;
+  std::memset(&p, 0, sizeof(p));
+  p.geo_block_size[0] = p.geo_block_size[1] = p.geo_block_size[2] =
+      p.geo_block_size[3] = 2;
+  p.n_vec = 4;
+  p.n_level = 2;
+  return p;
+}
+
+int eigensolveQuda(double *evals_re, double *evals_im, void **h_evecs,
+                   QudaAmdInvertParam *ip, QudaAmdEigParam *ep) {
+  QA_TRY
+  using namespace py::literals;
+  py::object e = g_api.attr("EigParam")(
+      "n_ev"_a = ep->n_ev, "n_kr"_a = ep->n_kr, "tol"_a = ep->tol,
+      "max_restarts"_a = ep->max_restarts,
+      "use_norm_op"_a = (bool)ep->use_norm_op,
+      "use_poly_acc"_a = (bool)ep->use_poly_acc, "poly_deg"_a = ep->poly_deg,
+      "a_min"_a = ep->a_min, "a_max"_a = ep->a_max,
+      "spectrum"_a = (ep->spectrum_largest ? "largest" : "smallest"));
+  py::tuple res = g_api.attr("eigensolve_quda")(make_invert_param(ip), e);
+  py::list evals = res[0].cast<py::list>();
+  for (int i = 0; i < ep->n_ev && i < (int)evals.size(); ++i) {
+    std::complex<double> v;  // TRLM yields real floats, IRAM complex
+    try {
+      v = evals[i].cast<std::complex<double>>();
+    } catch (const py::cast_error &) {
+      v = {evals[i].cast<double>(), 0.0};
+    }
+    evals_re[i] = v.real();
+    evals_im[i] = v.imag();
+  }
+  if (h_evecs) {
+    py::list evecs = res[1].cast<py::list>();
+    for (int i = 0; i < ep->n_ev && i < (int)evecs.size(); ++i)
+      tensor_out(evecs[i], h_evecs[i]);
+  }
+  QA_END
+}
+
+// heap wrapper keeping the MG pack alive across the C boundary
+struct QudaAmdMgHandle { py::object mg; };
+
+static void attach_precond(py::object &ip, void *handle) {
+  auto *h = static_cast<QudaAmdMgHandle *>(handle);
+  ip.attr("preconditioner") = h->mg.attr("precond");
+}
+
+void *newMultigridQuda(QudaAmdInvertParam *ip, QudaAmdMultigridParam *mp) {
+  try {
+    py::gil_scoped_acquire gil;
+    using namespace py::literals;
+    py::object mg = g_api.attr("new_multigrid_quda")(
+        make_invert_param(ip),
+        "block"_a = py::make_tuple(mp->geo_block_size[0],
+                                   mp->geo_block_size[1],
+                                   mp->geo_block_size[2],
+                                   mp->geo_block_size[3]),
+        "n_vec"_a = mp->n_vec);
+    g_err.clear();
+    return new QudaAmdMgHandle{std::move(mg)};
+  } catch (const std::exception &e) {
+    g_err = e.what();
+    return nullptr;
+  }
+}
+
+int updateMultigridQuda(void *mg, QudaAmdInvertParam *ip) {
+  QA_TRY
+  auto *h = static_cast<QudaAmdMgHandle *>(mg);
+  g_api.attr("update_multigrid_quda")(h->mg, make_invert_param(ip));
+  QA_END
+}
+
+int destroyMultigridQuda(void *mg) {
+  QA_TRY
+  delete static_cast<QudaAmdMgHandle *>(mg);
+  QA_END
+}
+
+static long gauge_cplx() { return 4 * 2 * resident_vcb() * 9; }
+
+static py::object mom_view(const void *h_mom) {
+  long Vcb = resident_vcb();
+  return tensor_view(h_mom, gauge_cplx(), py::make_tuple(4, 2, Vcb, 3, 3));
+}
+
+int computeGaugeForceQuda(void *h_mom, double beta) {
+  QA_TRY
+  py::object F = g_api.attr("compute_gauge_force_quda")(beta);
+  tensor_out(F, h_mom);
+  QA_END
+}
+
+int updateGaugeFieldQuda(const void *h_mom, double dt) {
+  QA_TRY
+  g_api.attr("update_gauge_field_quda")(mom_view(h_mom).attr("clone")(), dt);
+  QA_END
+}
+
+int momActionQuda(double *action, const void *h_mom) {
+  QA_TRY
+  *action =
+      g_api.attr("mom_action_quda")(mom_view(h_mom)).cast<double>();
+  QA_END
+}
+
+int momResidentQuda(const void *h_mom) {
+  QA_TRY
+  if (h_mom)
+    g_api.attr("mom_resident_quda")(mom_view(h_mom).attr("clone")());
+  else
+    g_api.attr("_MOM").attr("__setitem__")("p", py::none());
+  QA_END
+}
+
+int gaussMomQuda(void *h_mom, long seed) {
+  QA_TRY
+  py::object P = g_api.attr("gauss_mom_quda")(seed);
+  tensor_out(P, h_mom);
+  QA_END
+}
+
+int performGaugeSmearQuda(QudaAmdGaugeSmearType type, int n_steps,
+                          double coeff) {
+  QA_TRY
+  const char *kind = "stout";
+  switch (type) {
+    case QUDA_AMD_SMEAR_APE: kind = "ape"; break;
+    case QUDA_AMD_SMEAR_STOUT: kind = "stout"; break;
+    case QUDA_AMD_SMEAR_WILSON_FLOW: kind = "wilson_flow"; break;
+    case QUDA_AMD_SMEAR_HYP: kind = "hyp"; break;
+  }
+  g_api.attr("perform_gauge_smear_quda")(kind, n_steps, coeff);
+  QA_END
+}
+
+int gaugeObservablesQuda(double plaq[3], double *qcharge, double energy[2]) {
+  QA_TRY
+  py::dict obs = g_api.attr("gauge_observables_quda")();
+  py::tuple p = obs["plaquette"].cast<py::tuple>();
+  plaq[0] = p[0].cast<double>();
+  plaq[1] = p[1].cast<double>();
+  plaq[2] = p[2].cast<double>();
+  if (qcharge) *qcharge = obs["qcharge"].cast<double>();
+  if (energy) {
+    py::tuple e = obs["energy"].cast<py::tuple>();
+    energy[0] = e[0].cast<double>();
+    energy[1] = e[1].cast<double>();
+  }
+  QA_END
+}
+
+int projectSU3Quda(void) {
+  QA_TRY
+  py::object fields = py::module_::import("quda_amd.fields.gauge");
+  py::object u = g_api.attr("_R").attr("u_complex");
+  py::object w = fields.attr("project_su3")(u);
+  g_api.attr("load_gauge_quda")(w, g_api.attr("_R").attr("gauge_param"));
+  QA_END
+}
+
+int computeGaugeFixingOVRQuda(int gauge_dir, int max_iter, double tol) {
+  QA_TRY
+  using namespace py::literals;
+  g_api.attr("compute_gauge_fixing_ovr_quda")(
+      gauge_dir == 4 ? "landau" : "coulomb", "max_iter"_a = max_iter,
+      "tol"_a = tol);
+  QA_END
+}
+
+int computeGaugeFixingFFTQuda(int gauge_dir, int max_iter, double alpha,
+                              double tol) {
+  QA_TRY
+  using namespace py::literals;
+  py::object fix = py::module_::import("quda_amd.gauge.fix");
+  py::object R = g_api.attr("_R");
+  py::object w = fix.attr("gauge_fix_fft")(
+      R.attr("u_complex"), R.attr("geo"),
+      "gauge"_a = (gauge_dir == 4 ? "landau" : "coulomb"),
+      "alpha"_a = alpha, "max_iter"_a = max_iter, "tol"_a = tol);
+  g_api.attr("load_gauge_quda")(w, R.attr("gauge_param"));
+  QA_END
+}
+
+int contractQuda(void *h_out, const void *h_x, const void *h_y,
+                 QudaAmdInvertParam *param, int mode) {
+  QA_TRY
+  long Vcb = resident_vcb();
+  py::object x = tensor_view(h_x, spinor_cplx(param, Vcb),
+                             spinor_shape(param, Vcb, 2));
+  py::object y = tensor_view(h_y, spinor_cplx(param, Vcb),
+                             spinor_shape(param, Vcb, 2));
+  py::object out = g_api.attr("contract_quda")(
+      x, y, make_invert_param(param), mode == 0 ? "open" : "dr");
+  tensor_out(out, h_out);
   QA_END
 }

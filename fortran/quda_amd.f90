@@ -40,6 +40,7 @@ module quda_amd
     real(c_double) :: m5, b5, c5
     integer(c_int) :: iter
     real(c_double) :: true_res, secs, gflops
+    type(c_ptr) :: preconditioner
   end type
 
   interface

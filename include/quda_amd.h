@@ -84,11 +84,45 @@ typedef struct QudaAmdInvertParam_s {
   double true_res;
   double secs;
   double gflops;
+  /* opaque MG handle from newMultigridQuda (NULL = unpreconditioned);
+   * consumed by invertQuda with inv_type GCR (ref: the
+   * QudaInvertParam::preconditioner field of quda.h) */
+  void *preconditioner;
 } QudaAmdInvertParam;
+
+/* eigensolver request (ref: QudaEigParam quda.h:471, consumed subset) */
+typedef struct QudaAmdEigParam_s {
+  int n_ev;                      /* eigenpairs wanted                   */
+  int n_kr;                      /* Krylov/Lanczos subspace size        */
+  double tol;
+  int max_restarts;
+  int use_norm_op;               /* 1: TRLM on MdagM; 0: IRAM on M      */
+  int use_poly_acc;              /* Chebyshev acceleration (TRLM)       */
+  int poly_deg;
+  double a_min, a_max;           /* Chebyshev window                    */
+  int spectrum_largest;          /* 0 = smallest end, 1 = largest       */
+} QudaAmdEigParam;
+
+/* multigrid setup request (ref: QudaMultigridParam quda.h:570 subset) */
+typedef struct QudaAmdMultigridParam_s {
+  int geo_block_size[4];         /* aggregation block                   */
+  int n_vec;                     /* near-null vectors                   */
+  int n_level;                   /* 2 or 3                              */
+} QudaAmdMultigridParam;
+
+/* smearing kinds for performGaugeSmearQuda (ref: QudaGaugeSmearType) */
+typedef enum {
+  QUDA_AMD_SMEAR_APE = 0,
+  QUDA_AMD_SMEAR_STOUT = 1,
+  QUDA_AMD_SMEAR_WILSON_FLOW = 2,
+  QUDA_AMD_SMEAR_HYP = 3
+} QudaAmdGaugeSmearType;
 
 /* default-initialized params (role of newQudaGaugeParam/newQudaInvertParam) */
 QudaAmdGaugeParam newQudaAmdGaugeParam(void);
 QudaAmdInvertParam newQudaAmdInvertParam(void);
+QudaAmdEigParam newQudaAmdEigParam(void);
+QudaAmdMultigridParam newQudaAmdMultigridParam(void);
 
 /* lifecycle (ref: initQuda interface_quda.cpp:522 / endQuda) */
 int initQuda(int device);
@@ -118,6 +152,52 @@ int MatDagMatQuda(void *h_out, const void *h_in, QudaAmdInvertParam *param);
 /* observables (ref: plaqQuda quda.h:1462; plaq[0]=total, [1]=spatial,
  * [2]=temporal) */
 int plaqQuda(double plaq[3]);
+
+/* eigensolve the resident operator described by ip (ref: eigensolveQuda
+ * interface_quda.cpp:2524): fills evals_re/evals_im[n_ev]; h_evecs may be
+ * NULL or an array of ep->n_ev host spinor buffers */
+int eigensolveQuda(double *evals_re, double *evals_im, void **h_evecs,
+                   QudaAmdInvertParam *ip, QudaAmdEigParam *ep);
+
+/* multigrid lifecycle (ref: newMultigridQuda interface_quda.cpp:2772,
+ * updateMultigridQuda, destroyMultigridQuda). The returned handle plugs
+ * into QudaAmdInvertParam.preconditioner for GCR solves. */
+void *newMultigridQuda(QudaAmdInvertParam *ip, QudaAmdMultigridParam *mp);
+int updateMultigridQuda(void *mg, QudaAmdInvertParam *ip);
+int destroyMultigridQuda(void *mg);
+
+/* HMC surface (ref: computeGaugeForceQuda, updateGaugeFieldQuda,
+ * momActionQuda, momResidentQuda, gaussMomQuda). Momentum layout matches
+ * the gauge layout: [4][2][Vcb][3][3] complex double (antihermitian). */
+int computeGaugeForceQuda(void *h_mom, double beta);
+int updateGaugeFieldQuda(const void *h_mom, double dt);
+int momActionQuda(double *action, const void *h_mom);
+int momResidentQuda(const void *h_mom);      /* NULL clears              */
+int gaussMomQuda(void *h_mom, long seed);
+
+/* gauge smearing of the resident field in place (ref:
+ * performGaugeSmearQuda / performWFlowQuda; coeff = alpha/rho/epsilon) */
+int performGaugeSmearQuda(QudaAmdGaugeSmearType type, int n_steps,
+                          double coeff);
+
+/* combined gauge observables on the resident field (ref:
+ * gaugeObservablesQuda; energy[0]=plaq-based E, energy[1]=clover E) */
+int gaugeObservablesQuda(double plaq[3], double *qcharge, double energy[2]);
+
+/* project the resident links back onto SU(3) (ref: projectSU3Quda) */
+int projectSU3Quda(void);
+
+/* gauge fixing of the resident field (ref: computeGaugeFixingOVRQuda /
+ * computeGaugeFixingFFTQuda; gauge_dir 4 = Landau, 3 = Coulomb) */
+int computeGaugeFixingOVRQuda(int gauge_dir, int max_iter, double tol);
+int computeGaugeFixingFFTQuda(int gauge_dir, int max_iter, double alpha,
+                              double tol);
+
+/* open-spin (mode 0) / DeGrand-Rossi gamma-insertion (mode 1) contraction
+ * of two host propagator fields; out: [V][4][4] (mode 0) or [V][16]
+ * (mode 1) complex double in lexicographic site order (ref: contractQuda) */
+int contractQuda(void *h_out, const void *h_x, const void *h_y,
+                 QudaAmdInvertParam *param, int mode);
 
 /* last error message ("" when the previous call succeeded) */
 const char *qudaAmdLastError(void);

@@ -143,6 +143,89 @@ int main(void) {
   for (int c = 0; c < 3; ++c)
     CHECK(fabs(gauge[c * 3 + c].re - 1.0) < 1e-12, "saved gauge round-trip");
 
+  /* ---- round-2 surface: observables, smearing, HMC, eigensolve, MG,
+   *      gauge fixing, contraction ---- */
+  double qcharge, energy[2];
+  CHECK(gaugeObservablesQuda(plaq, &qcharge, energy) == 0,
+        "gaugeObservablesQuda");
+  CHECK(fabs(plaq[0] - 1.0) < 1e-12 && fabs(qcharge) < 1e-8,
+        "unit-field observables");
+
+  CHECK(performGaugeSmearQuda(QUDA_AMD_SMEAR_STOUT, 2, 0.1) == 0,
+        "stout smear");
+  CHECK(performGaugeSmearQuda(QUDA_AMD_SMEAR_WILSON_FLOW, 2, 0.02) == 0,
+        "wilson flow");
+  CHECK(projectSU3Quda() == 0, "projectSU3Quda");
+  CHECK(plaqQuda(plaq) == 0 && fabs(plaq[0] - 1.0) < 1e-10,
+        "unit field invariant under smearing");
+
+  cplx *mom = (cplx *)malloc(glinks * 9 * sizeof(cplx));
+  CHECK(gaussMomQuda(mom, 77) == 0, "gaussMomQuda");
+  double mact = 0;
+  CHECK(momActionQuda(&mact, mom) == 0 && mact > 0, "momActionQuda");
+  CHECK(momResidentQuda(mom) == 0, "momResidentQuda(set)");
+  CHECK(momResidentQuda(NULL) == 0, "momResidentQuda(clear)");
+  cplx *force = (cplx *)malloc(glinks * 9 * sizeof(cplx));
+  CHECK(computeGaugeForceQuda(force, 5.5) == 0, "computeGaugeForceQuda");
+  double fmax = 0;
+  for (size_t i = 0; i < glinks * 9; ++i) {
+    if (fabs(force[i].re) > fmax) fmax = fabs(force[i].re);
+    if (fabs(force[i].im) > fmax) fmax = fabs(force[i].im);
+  }
+  CHECK(fmax < 1e-10, "unit-field gauge force vanishes");
+  CHECK(updateGaugeFieldQuda(mom, 0.0) == 0, "updateGaugeFieldQuda(dt=0)");
+  CHECK(plaqQuda(plaq) == 0 && fabs(plaq[0] - 1.0) < 1e-12,
+        "dt=0 update is the identity");
+
+  QudaAmdEigParam ep = newQudaAmdEigParam();
+  ep.n_ev = 4;
+  ep.n_kr = 16;
+  ep.tol = 1e-6;
+  ip.dslash_type = QUDA_AMD_WILSON_DSLASH;
+  ip.solution_type = QUDA_AMD_MATPC_SOLUTION;
+  ip.kappa = 0.10;
+  double ev_re[4], ev_im[4];
+  CHECK(eigensolveQuda(ev_re, ev_im, NULL, &ip, &ep) == 0,
+        "eigensolveQuda");
+  CHECK(ev_re[0] > 0 && ev_re[0] <= ev_re[1] + 1e-12,
+        "MdagM spectrum positive and sorted");
+  printf("eigensolve: lambda_min = %.6f\n", ev_re[0]);
+
+  QudaAmdMultigridParam mp = newQudaAmdMultigridParam();
+  mp.n_vec = 2;
+  ip.solution_type = QUDA_AMD_MAT_SOLUTION;
+  ip.inv_type = QUDA_AMD_GCR_INVERTER;
+  ip.kappa = 0.11;
+  ip.tol = 1e-8;
+  ip.maxiter = 200;
+  void *mg = newMultigridQuda(&ip, &mp);
+  CHECK(mg != NULL, "newMultigridQuda");
+  ip.preconditioner = mg;
+  CHECK(invertQuda(x, b, &ip) == 0, "invertQuda(MG-preconditioned)");
+  CHECK(ip.iter > 0, "MG solve iterated");
+  ip.preconditioner = NULL;
+  CHECK(MatQuda(mx, x, &ip) == 0, "MatQuda(mg check)");
+  r2 = 0;
+  for (size_t i = 0; i < ns; ++i) {
+    double dr = mx[i].re - b[i].re, di = mx[i].im - b[i].im;
+    r2 += dr * dr + di * di;
+  }
+  CHECK(sqrt(r2 / b2) < 1e-6, "MG ||Mx-b||/||b|| < 1e-6");
+  printf("mg-gcr solve: %d iters, check %.2e\n", ip.iter, sqrt(r2 / b2));
+  CHECK(destroyMultigridQuda(mg) == 0, "destroyMultigridQuda");
+
+  CHECK(computeGaugeFixingOVRQuda(4, 3, 1e-10) == 0,
+        "computeGaugeFixingOVRQuda");
+
+  /* contraction of two constant propagator fields: C[x,s,s'] = 3 */
+  for (size_t i = 0; i < ns; ++i) { b[i].re = 1.0; b[i].im = 0.0; }
+  cplx *corr = (cplx *)malloc((size_t)L * L * L * L * 16 * sizeof(cplx));
+  CHECK(contractQuda(corr, b, b, &ip, 0) == 0, "contractQuda");
+  CHECK(fabs(corr[0].re - 3.0) < 1e-12, "open-spin contraction value");
+  free(corr);
+  free(mom);
+  free(force);
+
   CHECK(freeGaugeQuda() == 0, "freeGaugeQuda");
   CHECK(endQuda() == 0, "endQuda");
   printf("c_interface_test: ALL PASSED\n");
