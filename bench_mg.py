@@ -41,6 +41,9 @@ def main():
     ap.add_argument("--nvec2", type=int, default=8)
     ap.add_argument("--tol", type=float, default=1e-8)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--smooth-sweep", action="store_true",
+                    help="after one MG setup, sweep (nu_post, coarse_tol) "
+                         "V-cycle configs reusing the same null vectors")
     args = ap.parse_args()
     dev = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     dims = tuple(int(x) for x in args.lattice.split(","))
@@ -98,6 +101,28 @@ def main():
                              nkrylov=24)
         sync()
         t_plain = time.perf_counter() - t0
+
+        if args.smooth_sweep:
+            import dataclasses
+            for nu_post, ctol, cmax in [(4, 5e-2, 200), (2, 5e-2, 200),
+                                        (4, 1e-1, 100), (2, 1e-1, 100),
+                                        (1, 1e-1, 100), (2, 2e-1, 60),
+                                        (1, 2e-1, 60)]:
+                mg.param = dataclasses.replace(mg.param, nu_post=nu_post,
+                                               coarse_tol=ctol,
+                                               coarse_maxiter=cmax)
+                xs = SpinorField(geo, "double", dev)
+                t0 = time.perf_counter()
+                st = gcr_solve(d, xs, b, tol=args.tol, maxiter=2000,
+                               nkrylov=24, precond=mg.precond)
+                sync()
+                print(json.dumps({
+                    "sweep": {"nu_post": nu_post, "coarse_tol": ctol,
+                              "coarse_maxiter": cmax},
+                    "kappa": kappa, "iters": st.iters,
+                    "secs": round(time.perf_counter() - t0, 2),
+                    "converged": st.converged}), flush=True)
+            continue
 
         x1 = SpinorField(geo, "double", dev)
         t0 = time.perf_counter()
