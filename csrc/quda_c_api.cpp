@@ -478,7 +478,11 @@ static long gauge_cplx() { return 4 * 2 * resident_vcb() * 9; }
 
 static py::object mom_view(const void *h_mom) {
   long Vcb = resident_vcb();
-  return tensor_view(h_mom, gauge_cplx(), py::make_tuple(4, 2, Vcb, 3, 3));
+  // host buffer -> device-resident tensor (the engine's fields live on
+  // g_device; a cpu view would fail device checks in the update ops)
+  py::object t =
+      tensor_view(h_mom, gauge_cplx(), py::make_tuple(4, 2, Vcb, 3, 3));
+  return t.attr("clone")().attr("to")(g_device);
 }
 
 int computeGaugeForceQuda(void *h_mom, double beta) {
@@ -490,7 +494,7 @@ int computeGaugeForceQuda(void *h_mom, double beta) {
 
 int updateGaugeFieldQuda(const void *h_mom, double dt) {
   QA_TRY
-  g_api.attr("update_gauge_field_quda")(mom_view(h_mom).attr("clone")(), dt);
+  g_api.attr("update_gauge_field_quda")(mom_view(h_mom), dt);
   QA_END
 }
 
@@ -504,7 +508,7 @@ int momActionQuda(double *action, const void *h_mom) {
 int momResidentQuda(const void *h_mom) {
   QA_TRY
   if (h_mom)
-    g_api.attr("mom_resident_quda")(mom_view(h_mom).attr("clone")());
+    g_api.attr("mom_resident_quda")(mom_view(h_mom));
   else
     g_api.attr("_MOM").attr("__setitem__")("p", py::none());
   QA_END
