@@ -20,7 +20,7 @@ from quda_amd.fields.clover import CloverField, pack_clover  # noqa: E402
 from quda_amd.ops import reference as ref  # noqa: E402
 from quda_amd.ops.dispatch import CLOV_POST, PLAIN, dslash_wilson  # noqa: E402
 
-SIZEOF = {"double": 8, "single": 4, "half": 2}
+SIZEOF = {"double": 8, "single": 4, "half": 2, "quarter": 1}
 
 
 def run(prec, recon_name, mode, geo, u, A, reps):
@@ -82,6 +82,27 @@ def sweep(geo, u, reps):
     return results
 
 
+def lds_family(geo, u, reps):
+    """LDS-tiled kernel (k_dslash_wilson_lds) vs the gather-first default."""
+    from quda_amd.ops.dispatch import hip_ext as _ext
+    A = ref.clover_matrix(u, geo, 0.135, 1.0)
+    results = {}
+    for lds in (0, 1):
+        _ext().set_dslash_lds(lds)
+        for prec, recon in [("half", "twelve"), ("half", "none"),
+                            ("quarter", "eight"), ("quarter", "twelve")]:
+            for mode, mname in [(PLAIN, "wilson"), (CLOV_POST, "wilson_clover")]:
+                r = run(prec, recon, mode, geo, u, A, reps)
+                rc = {"twelve": 12, "eight": 8, "none": 18}[recon]
+                key = f"{mname}/{prec}/r{rc}/lds{lds}"
+                results[key] = round(r["gflops"])
+                print(f"{key:40s} {r['us']:8.1f} us  {r['gflops']:8.0f} "
+                      f"GFLOPS", flush=True)
+    _ext().set_dslash_lds(0)
+    print(json.dumps(results))
+    return results
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--lattice", default="32,32,32,64")
@@ -102,6 +123,9 @@ def main():
     if "sweep" in fams:
         results.update(sweep(geo, u, args.reps))
         fams.discard("sweep")
+    if "lds" in fams:
+        results.update(lds_family(geo, u, args.reps))
+        fams.discard("lds")
         if not fams:
             return
     if "wilson" in fams:

@@ -441,6 +441,8 @@ static void set_dslash_waves(int64_t w) {
   if (w == 0 || w == 3) qa_dslash_waves_ref() = (int)w;
 }
 
+static void set_dslash_lds(int64_t v) { qa_dslash_lds_ref() = (int)v; }
+
 // ---------------------------------------------------------------------------
 // HIP-IPC remote-write halos (role of comm_target.cpp:41-134 P2P
 // remote-write): export/import device-buffer handles so the pack kernel
@@ -637,6 +639,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_dslash_block", &set_dslash_block, "autotuner: dslash workgroup size");
   m.def("set_dslash_waves", &set_dslash_waves,
         "occupancy experiment: 0 default, 3 = 64-thread/3-wave variant");
+  m.def("set_dslash_lds", &set_dslash_lds,
+        "LDS-tiled dslash policy: 1 = stage the in-spinor halo tile in LDS "
+        "(half/quarter, local, tile-divisible dims)");
   m.def("ipc_get_handle", &ipc_get_handle, "hipIpcGetMemHandle of a tensor");
   m.def("ipc_open_handle", &ipc_open_handle, "open a peer IPC handle");
   m.def("ipc_close_handle", &ipc_close_handle, "close a peer IPC mapping");

@@ -422,3 +422,169 @@ __global__ __launch_bounds__(256) void k_clover_apply(
   clover_mul(r, diag, tri, v);
   out.store(r, i);
 }
+
+// ---------------------------------------------------------------------------
+// LDS-tiled Wilson dslash (half/quarter, local-only).
+//
+// The gather-first kernel above is LOAD-ISSUE bound on MI355X (~13 B/cyc/
+// CU at the mixed-load ceiling, profiles/r02_dslash_final.md): every
+// neighbor spinor body is fetched from global memory TWICE per
+// checkerboard apply (once by each adjacent output site), and all fetches
+// compete for the same vector-load issue port. This variant stages the
+// whole in-spinor halo tile through LDS instead: one workgroup owns a
+// BX x BY x BZ x BT lexicographic tile (256 output sites of one parity),
+// cooperatively loads the (B+2)^-extended in-parity halo (1080 sites,
+// RAW stored chunks + per-site norm = 56 KB at half -> 2 workgroups/CU),
+// then serves all 8 neighbor reads per site from the LDS pipe, which
+// issues independently of the global-load port. Unique global spinor
+// traffic drops ~1.9x and the spinor share of the vector-load issue
+// stream moves off the critical port entirely; gauge links still stream
+// from global (fwd-slot reads as above).
+// (role of reference kernels/dslash_wilson.cuh; the reference deliberately
+// has no shared-memory dslash — on NVIDIA the texture path suffices. On
+// CDNA4 the LDS pipe is the only way past the load-issue ceiling.)
+// ---------------------------------------------------------------------------
+struct LdsTile {
+  static constexpr int BX = 4, BY = 4, BZ = 4, BT = 8;   // 512 lex sites
+  static constexpr int EX = BX + 2, EY = BY + 2, EZ = BZ + 2, ET = BT + 2;
+  static constexpr int NROW = EY * EZ * ET;              // rows of EX sites
+  static constexpr int NSLOT = NROW * (EX / 2);          // in-parity sites
+};
+
+template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
+__global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
+    SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
+    CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
+    SpinorAcc<Prec> x, typename Prec::Real br, typename Prec::Real bi) {
+  using R = typename Prec::Real;
+  using S = typename Prec::Store;
+  static_assert(Prec::has_norm, "LDS dslash: half/quarter only (raw-chunk staging)");
+  constexpr int W = SpinorAcc<Prec>::W;     // 8 for half & quarter
+  constexpr int NCH = SpinorAcc<Prec>::NCH; // 3
+  using T = LdsTile;
+
+  __shared__ S lds[NCH][T::NSLOT][W];
+  __shared__ float lnrm[T::NSLOT];
+
+  // tile origin from block id (x-fastest tile raster)
+  int ntx = d.X[0] / T::BX, nty = d.X[1] / T::BY, ntz = d.X[2] / T::BZ;
+  long bid = blockIdx.x;
+  int ox = (int)(bid % ntx) * T::BX; bid /= ntx;
+  int oy = (int)(bid % nty) * T::BY; bid /= nty;
+  int oz = (int)(bid % ntz) * T::BZ;
+  int ot = (int)(bid / ntz) * T::BT;
+  const int ipar = 1 - parity;
+  const int S0 = ox + oy + oz + ot + d.parity_offset;
+
+  // ---- cooperative halo load: slot s covers ext row r = s / (EX/2),
+  //      within-row k = s % (EX/2); in-parity x positions are r0 + 2k ----
+  for (int s = threadIdx.x; s < T::NSLOT; s += blockDim.x) {
+    int k = s % (T::EX / 2), r = s / (T::EX / 2);
+    int ey = r % T::EY, rr = r / T::EY;
+    int ez = rr % T::EZ, et = rr / T::EZ;
+    int r0 = (ipar + S0 + ey + ez + et) & 1;
+    int gx = ox - 1 + r0 + 2 * k, gy = oy - 1 + ey, gz = oz - 1 + ez,
+        gt = ot - 1 + et;
+    if (gx < 0) gx += d.X[0]; else if (gx >= d.X[0]) gx -= d.X[0];
+    if (gy < 0) gy += d.X[1]; else if (gy >= d.X[1]) gy -= d.X[1];
+    if (gz < 0) gz += d.X[2]; else if (gz >= d.X[2]) gz -= d.X[2];
+    if (gt < 0) gt += d.X[3]; else if (gt >= d.X[3]) gt -= d.X[3];
+    int xc2[4] = {gx, gy, gz, gt};
+    long j = cb_from_coords(xc2, d);
+    long base = j * W;  // parity-sliced accessor: chunk_base(j<V) = j*W
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      load_chunk<S, W>(in.data + base + (long)ch * in.V * W, lds[ch][s]);
+    lnrm[s] = in.norm[j];
+  }
+  __syncthreads();
+
+  // ---- output site of this thread ----
+  int k = threadIdx.x % (T::BX / 2), r = threadIdx.x / (T::BX / 2);
+  int ly = r % T::BY; int rr = r / T::BY;
+  int lz = rr % T::BZ, lt = rr / T::BZ;
+  int r0 = (parity + S0 + ly + lz + lt) & 1;
+  int lx = r0 + 2 * k;
+  int xc[4] = {ox + lx, oy + ly, oz + lz, ot + lt};
+  long i = cb_from_coords(xc, d);
+
+  cplx<R> acc[4][3];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+#pragma unroll
+    for (int c = 0; c < 3; ++c) acc[s][c] = {(R)0, (R)0};
+  cplx<R> p[4][3], h[2][3], uh[2][3], U[3][3];
+  const R one = (R)0.5;
+
+  // LDS slot of the neighbor at ext coords (ex,ey,ez,et); caller
+  // guarantees it holds an in-parity site
+  auto lds_read = [&](int ex, int ey, int ez, int et) {
+    int row = (et * T::EZ + ez) * T::EY + ey;
+    int slot = row * (T::EX / 2) + (ex >> 1);
+    S tmp[24];
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      load_chunk<S, W>(lds[ch][slot], tmp + ch * W);
+    R sc = lnrm[slot];
+#pragma unroll
+    for (int kk = 0; kk < 12; ++kk)
+      p[kk / 3][kk % 3] = {sc * qa_tor<R>(tmp[2 * kk]),
+                           sc * qa_tor<R>(tmp[2 * kk + 1])};
+  };
+
+  const int e0 = lx + 1, e1 = ly + 1, e2 = lz + 1, e3 = lt + 1;
+
+#define QA_LDS_DIR(MU, EXP, EXM)                                          \
+  {                                                                       \
+    EXP;                                                                  \
+    if constexpr (!DAG) proj_##MU##_0(h, p);                              \
+    else proj_##MU##_1(h, p);                                             \
+    g.template load<MU>(U, i);                                            \
+    su3_mul_half(uh, U, h);                                               \
+    if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
+    else recon_##MU##_1(acc, uh, one);                                    \
+    EXM;                                                                  \
+    if constexpr (!DAG) proj_##MU##_1(h, p);                              \
+    else proj_##MU##_0(h, p);                                             \
+    long jm = neighbor_cb(xc, MU, -1, d);                                 \
+    g.template load_o<MU>(U, jm);                                         \
+    su3_dagmul_half(uh, U, h);                                            \
+    if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \
+    else recon_##MU##_0(acc, uh, one);                                    \
+  }
+
+  QA_LDS_DIR(0, lds_read(e0 + 1, e1, e2, e3), lds_read(e0 - 1, e1, e2, e3))
+  QA_LDS_DIR(1, lds_read(e0, e1 + 1, e2, e3), lds_read(e0, e1 - 1, e2, e3))
+  QA_LDS_DIR(2, lds_read(e0, e1, e2 + 1, e3), lds_read(e0, e1, e2 - 1, e3))
+  QA_LDS_DIR(3, lds_read(e0, e1, e2, e3 + 1), lds_read(e0, e1, e2, e3 - 1))
+#undef QA_LDS_DIR
+
+  if constexpr (MODE == CLOV_POST) {
+    R diag[2][6];
+    cplx<R> tri[2][15];
+    clov.load(diag, tri, parity, i);
+    cplx<R> tmp[4][3];
+    clover_mul(tmp, diag, tri, acc);
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) acc[s][c] = tmp[s][c];
+  }
+  if constexpr (MODE == TWIST_POST) twist_mul(acc, br, bi);
+  if constexpr (XPAY) {
+    cplx<R> xv[4][3];
+    x.load(xv, i);
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+#pragma unroll
+      for (int c = 0; c < 3; ++c) acc[s][c] = xv[s][c] + a * acc[s][c];
+  } else {
+    if (a != (R)1) {
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+#pragma unroll
+        for (int c = 0; c < 3; ++c) acc[s][c] = a * acc[s][c];
+    }
+  }
+  out.store(acc, i);
+}

@@ -38,6 +38,28 @@ static void dslash_launch_all(const DslashCall &c, hipStream_t st) {
   int waves = qa_dslash_waves();
   int grid64 = (int)((c.Vcb + 63) / 64);
 
+  // LDS-tiled path (half/quarter, local, 4-d, tile-divisible dims,
+  // PLAIN/CLOV_POST/TWIST_POST): see k_dslash_wilson_lds
+  if constexpr (Prec::has_norm) {
+    bool lds_ok = qa_dslash_lds() && c.kt == 0 && c.comm_mask == 0 &&
+                  c.in.Vcb == c.Vcb && c.out.Vcb == c.Vcb &&
+                  (c.mode == PLAIN || c.mode == CLOV_POST ||
+                   c.mode == TWIST_POST) &&
+                  c.Xdim[0] % LdsTile::BX == 0 &&
+                  c.Xdim[1] % LdsTile::BY == 0 &&
+                  c.Xdim[2] % LdsTile::BZ == 0 && c.Xdim[3] % LdsTile::BT == 0;
+    if (lds_ok) {
+      long ntile = ((long)c.Xdim[0] / LdsTile::BX) * (c.Xdim[1] / LdsTile::BY) *
+                   (c.Xdim[2] / LdsTile::BZ) * (c.Xdim[3] / LdsTile::BT);
+#define QA_LDS_LAUNCH(DAG, MODE, XPAY)                                              hipLaunchKernelGGL((k_dslash_wilson_lds<Prec, RECON, DAG, MODE, XPAY>),                          dim3((unsigned)ntile), dim3(256), 0, st, out, in, g,                          cl, d, c.parity, a, x, br, bi)
+#define QA_LDS_MODES(DAG)                                                           switch (c.mode) {                                                               case PLAIN:                                                                     if (c.xpay) QA_LDS_LAUNCH(DAG, PLAIN, true);                                  else QA_LDS_LAUNCH(DAG, PLAIN, false);                                        break;                                                                      case CLOV_POST:                                                                 if (c.xpay) QA_LDS_LAUNCH(DAG, CLOV_POST, true);                              else QA_LDS_LAUNCH(DAG, CLOV_POST, false);                                    break;                                                                      case TWIST_POST:                                                                if (c.xpay) QA_LDS_LAUNCH(DAG, TWIST_POST, true);                             else QA_LDS_LAUNCH(DAG, TWIST_POST, false);                                   break;                                                                    }
+      if (!c.dagger) { QA_LDS_MODES(false) } else { QA_LDS_MODES(true) }
+#undef QA_LDS_MODES
+#undef QA_LDS_LAUNCH
+      return;
+    }
+  }
+
 #define QA_LAUNCH(DAG, MODE, XPAY, KT)                                        \
   if (c.kt == 3)                                                              \
     hipLaunchKernelGGL((k_dslash_wilson_exterior<Prec, RECON, DAG, MODE, XPAY>), \

@@ -29,6 +29,17 @@ inline int &qa_dslash_waves_ref() {
 }
 inline int qa_dslash_waves() { return qa_dslash_waves_ref(); }
 
+// LDS-tiled dslash policy (0 = off, 1 = on where eligible: half/quarter,
+// local, 4-d, tile-divisible dims; set_dslash_lds binding / autotuner)
+inline int &qa_dslash_lds_ref() {
+  static int v = []() {
+    const char *e = std::getenv("QUDA_AMD_DSLASH_LDS");
+    return e ? std::atoi(e) : 0;
+  }();
+  return v;
+}
+inline int qa_dslash_lds() { return qa_dslash_lds_ref(); }
+
 struct BlasField {
   void *data;
   void *norm;  // nullptr unless half

@@ -380,3 +380,38 @@ def test_triple_cg_update_gpu(setup, prec):
     dx = (x1.to_complex() - x2.to_complex()).abs().max().item()
     dr = (r1.to_complex() - r2.to_complex()).abs().max().item()
     assert dx < tol * 10 and dr < tol * 10, (prec, dx, dr)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("prec", ["half", "quarter"])
+def test_dslash_lds_variant(setup, prec):
+    """k_dslash_wilson_lds (in-spinor halo tile staged through LDS) must
+    reproduce the gather-first kernel on PLAIN (+dagger), CLOV_POST and
+    xpay — same decoded inputs, same accumulation order, so the results
+    agree to same-precision roundoff."""
+    from quda_amd.fields.clover import CloverField
+    from quda_amd.ops.dispatch import CLOV_POST
+    from quda_amd.ops.dispatch import hip_ext as _ext
+    geo, g, psi, chi, A = setup
+    gd, sd = _gpu_fields(geo, g, psi, prec, 12)
+    cl = CloverField(geo, prec, "cuda").from_matrices(A.cuda())
+    xd = SpinorField(geo, prec, "cuda", n_parity=1).from_complex(
+        chi.to_complex()[0:1].cuda())
+    out0 = SpinorField(geo, prec, "cuda", n_parity=1)
+    out1 = SpinorField(geo, prec, "cuda", n_parity=1)
+    cases = [dict(mode=PLAIN, dagger=False), dict(mode=PLAIN, dagger=True),
+             dict(mode=CLOV_POST, clover=cl, clover_inverse=False),
+             dict(mode=PLAIN, a=-0.4, x=xd)]
+    for kw in cases:
+        kw2 = dict(kw)
+        _ext().set_dslash_lds(0)
+        dslash_wilson(out0, sd.parity_view(1), gd, 0, **kw2)
+        _ext().set_dslash_lds(1)
+        try:
+            dslash_wilson(out1, sd.parity_view(1), gd, 0, **kw2)
+        finally:
+            _ext().set_dslash_lds(0)
+        a = out0.to_complex().cpu()
+        b = out1.to_complex().cpu()
+        err = (a - b).abs().max().item() / a.abs().max().item()
+        assert err < 1e-6, (prec, kw.get("mode"), err)
