@@ -90,7 +90,7 @@ def lds_family(geo, u, reps):
     for lds in (0, 1):
         _ext().set_dslash_lds(lds)
         for prec, recon in [("half", "twelve"), ("half", "none"),
-                            ("quarter", "eight"), ("quarter", "twelve")]:
+                            ("quarter", "twelve")]:
             for mode, mname in [(PLAIN, "wilson"), (CLOV_POST, "wilson_clover")]:
                 r = run(prec, recon, mode, geo, u, A, reps)
                 rc = {"twelve": 12, "eight": 8, "none": 18}[recon]

@@ -463,7 +463,8 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
   constexpr int NCH = SpinorAcc<Prec>::NCH; // 3
   using T = LdsTile;
 
-  __shared__ S lds[NCH][T::NSLOT][W];
+  // 16-byte alignment is REQUIRED: chunks move via ds_read/write_b128
+  __shared__ alignas(16) S lds[NCH][T::NSLOT][W];
   __shared__ float lnrm[T::NSLOT];
 
   // tile origin from block id (x-fastest tile raster)

@@ -21,7 +21,7 @@ from typing import Optional
 import torch
 
 from .geometry import LatticeGeometry
-from .layout import DTYPE_OF, WIDTH_OF, n_chunks
+from .layout import DTYPE_OF, chunk_width, n_chunks
 
 N_REALS = 72
 
@@ -65,7 +65,7 @@ class CloverField:
                  device="cpu"):
         self.geo = geo
         self.precision = precision
-        w = WIDTH_OF[precision]
+        w = chunk_width(N_REALS, precision)  # quarter: 16 !| 72 -> 8
         nch = n_chunks(N_REALS, precision)
         shape = (2, nch, geo.volume_cb, w)
         dt = DTYPE_OF[precision]
@@ -78,7 +78,7 @@ class CloverField:
 
     def _to_native(self, packed: torch.Tensor) -> torch.Tensor:
         """[2, V, 72] real -> chunked [2, nch, V, w]."""
-        w = WIDTH_OF[self.precision]
+        w = chunk_width(N_REALS, self.precision)
         V = self.geo.volume_cb
         return (packed.reshape(2, V, -1, w).movedim(2, 1).contiguous()
                 .to(DTYPE_OF[self.precision]))
