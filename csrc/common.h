@@ -221,6 +221,39 @@ struct GaugeAcc {
                    // neighbor's fwd slot live there: x-mu flips parity)
   long V;
 
+  // split raw-load / decode pair (RECON 12/18): lets a kernel issue the
+  // chunk loads of many links early (e.g. behind an LDS fill) and decode
+  // at use. Chunk count: recon-12 at W=8 always spans 2 chunks.
+  template <int Q>
+  static constexpr int raw_nc() {
+    return (Q * RECON + RECON - 1) / W - (Q * RECON) / W + 1;
+  }
+  template <int Q>
+  __device__ __forceinline__ void load_raw(const S *base, S *out,
+                                           long i) const {
+    constexpr int c0 = (Q * RECON) / W;
+#pragma unroll
+    for (int c = 0; c < raw_nc<Q>(); ++c)
+      load_chunk<S, W>(base + ((long)(c0 + c) * V + i) * W, out + c * W);
+  }
+  template <int Q>
+  __device__ __forceinline__ void decode_raw(const S *in,
+                                             cplx<R> u[3][3]) const {
+    static_assert(RECON == 12 || RECON == 18, "raw split: plain codecs only");
+    constexpr int off = Q * RECON - ((Q * RECON) / W) * W;
+#pragma unroll
+    for (int k = 0; k < RECON / 2; ++k)
+      u[k / 3][k % 3] = {qa_tor<R>(in[off + 2 * k]),
+                        qa_tor<R>(in[off + 2 * k + 1])};
+    if constexpr (RECON == 12) {
+#pragma unroll
+      for (int c = 0; c < 3; ++c) {
+        int a = (c + 1) % 3, b = (c + 2) % 3;
+        u[2][c] = conj(u[0][a] * u[1][b] - u[0][b] * u[1][a]);
+      }
+    }
+  }
+
   template <int Q>
   __device__ __forceinline__ void load_base(const S *base, cplx<R> u[3][3],
                                             long i) const {

@@ -466,6 +466,7 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
   // 16-byte alignment is REQUIRED: chunks move via ds_read/write_b128
   __shared__ alignas(16) S lds[NCH][T::NSLOT][W];
   __shared__ float lnrm[T::NSLOT];
+  int qa_xc[4];  // this thread's output-site coords
 
   // tile origin from block id (x-fastest tile raster)
   int ntx = d.X[0] / T::BX, nty = d.X[1] / T::BY, ntz = d.X[2] / T::BZ;
@@ -476,6 +477,36 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
   int ot = (int)(bid / ntz) * T::BT;
   const int ipar = 1 - parity;
   const int S0 = ox + oy + oz + ot + d.parity_offset;
+
+  // ---- output site of this thread (needed first: the raw gauge
+  //      preloads below overlap the cooperative LDS fill) ----
+  {
+    int k0 = threadIdx.x % (T::BX / 2), r1 = threadIdx.x / (T::BX / 2);
+    int ly = r1 % T::BY; int rr1 = r1 / T::BY;
+    int lz = rr1 % T::BZ, lt = rr1 / T::BZ;
+    int rp = (parity + S0 + ly + lz + lt) & 1;
+    qa_xc[0] = ox + rp + 2 * k0; qa_xc[1] = oy + ly;
+    qa_xc[2] = oz + lz; qa_xc[3] = ot + lt;
+  }
+  const long i = cb_from_coords(qa_xc, d);
+  long jm[4];
+#pragma unroll
+  for (int m = 0; m < 4; ++m) jm[m] = neighbor_cb(qa_xc, m, -1, d);
+
+  // raw gauge preload (recon 12: every link spans exactly 2 chunks) —
+  // the 16 chunk loads are in flight for the whole fill + barrier
+  constexpr bool GPRE = (RECON == 12);
+  S graw[GPRE ? 8 : 1][2 * GaugeAcc<Prec, RECON>::W];
+  if constexpr (GPRE) {
+    g.template load_raw<0>(g.data, graw[0], i);
+    g.template load_raw<1>(g.data, graw[2], i);
+    g.template load_raw<2>(g.data, graw[4], i);
+    g.template load_raw<3>(g.data, graw[6], i);
+    g.template load_raw<0>(g.other, graw[1], jm[0]);
+    g.template load_raw<1>(g.other, graw[3], jm[1]);
+    g.template load_raw<2>(g.other, graw[5], jm[2]);
+    g.template load_raw<3>(g.other, graw[7], jm[3]);
+  }
 
   // ---- cooperative halo load: slot s covers ext row r = s / (EX/2),
   //      within-row k = s % (EX/2); in-parity x positions are r0 + 2k ----
@@ -499,15 +530,6 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
     lnrm[s] = in.norm[j];
   }
   __syncthreads();
-
-  // ---- output site of this thread ----
-  int k = threadIdx.x % (T::BX / 2), r = threadIdx.x / (T::BX / 2);
-  int ly = r % T::BY; int rr = r / T::BY;
-  int lz = rr % T::BZ, lt = rr / T::BZ;
-  int r0 = (parity + S0 + ly + lz + lt) & 1;
-  int lx = r0 + 2 * k;
-  int xc[4] = {ox + lx, oy + ly, oz + lz, ot + lt};
-  long i = cb_from_coords(xc, d);
 
   cplx<R> acc[4][3];
 #pragma unroll
@@ -533,22 +555,24 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
                            sc * qa_tor<R>(tmp[2 * kk + 1])};
   };
 
-  const int e0 = lx + 1, e1 = ly + 1, e2 = lz + 1, e3 = lt + 1;
+  const int e0 = qa_xc[0] - ox + 1, e1 = qa_xc[1] - oy + 1,
+            e2 = qa_xc[2] - oz + 1, e3 = qa_xc[3] - ot + 1;
 
 #define QA_LDS_DIR(MU, EXP, EXM)                                          \
   {                                                                       \
     EXP;                                                                  \
     if constexpr (!DAG) proj_##MU##_0(h, p);                              \
     else proj_##MU##_1(h, p);                                             \
-    g.template load<MU>(U, i);                                            \
+    if constexpr (GPRE) g.template decode_raw<MU>(graw[2 * MU], U);       \
+    else g.template load<MU>(U, i);                                       \
     su3_mul_half(uh, U, h);                                               \
     if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
     else recon_##MU##_1(acc, uh, one);                                    \
     EXM;                                                                  \
     if constexpr (!DAG) proj_##MU##_1(h, p);                              \
     else proj_##MU##_0(h, p);                                             \
-    long jm = neighbor_cb(xc, MU, -1, d);                                 \
-    g.template load_o<MU>(U, jm);                                         \
+    if constexpr (GPRE) g.template decode_raw<MU>(graw[2 * MU + 1], U);   \
+    else g.template load_o<MU>(U, jm[MU]);                                \
     su3_dagmul_half(uh, U, h);                                            \
     if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \
     else recon_##MU##_0(acc, uh, one);                                    \

@@ -414,4 +414,7 @@ def test_dslash_lds_variant(setup, prec):
         a = out0.to_complex().cpu()
         b = out1.to_complex().cpu()
         err = (a - b).abs().max().item() / a.abs().max().item()
-        assert err < 1e-6, (prec, kw.get("mode"), err)
+        # same decoded inputs and op order, but the per-site block-float
+        # norm can flip one stored ULP between variants
+        tol = {"half": 5e-4, "quarter": 5e-3}[prec]
+        assert err < tol, (prec, kw.get("mode"), err)
