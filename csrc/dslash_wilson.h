@@ -452,7 +452,7 @@ struct LdsTile {
 };
 
 template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
-__global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
+__global__ __launch_bounds__(256, 1) void k_dslash_wilson_lds(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
     SpinorAcc<Prec> x, typename Prec::Real br, typename Prec::Real bi) {
@@ -489,27 +489,10 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
     qa_xc[2] = oz + lz; qa_xc[3] = ot + lt;
   }
   const long i = cb_from_coords(qa_xc, d);
-  long jm[4];
-#pragma unroll
-  for (int m = 0; m < 4; ++m) jm[m] = neighbor_cb(qa_xc, m, -1, d);
-
-  // raw gauge preload (recon 12: every link spans exactly 2 chunks) —
-  // the 16 chunk loads are in flight for the whole fill + barrier
-  constexpr bool GPRE = (RECON == 12);
-  S graw[GPRE ? 8 : 1][2 * GaugeAcc<Prec, RECON>::W];
-  if constexpr (GPRE) {
-    g.template load_raw<0>(g.data, graw[0], i);
-    g.template load_raw<1>(g.data, graw[2], i);
-    g.template load_raw<2>(g.data, graw[4], i);
-    g.template load_raw<3>(g.data, graw[6], i);
-    g.template load_raw<0>(g.other, graw[1], jm[0]);
-    g.template load_raw<1>(g.other, graw[3], jm[1]);
-    g.template load_raw<2>(g.other, graw[5], jm[2]);
-    g.template load_raw<3>(g.other, graw[7], jm[3]);
-  }
 
   // ---- cooperative halo load: slot s covers ext row r = s / (EX/2),
   //      within-row k = s % (EX/2); in-parity x positions are r0 + 2k ----
+#pragma unroll 1
   for (int s = threadIdx.x; s < T::NSLOT; s += blockDim.x) {
     int k = s % (T::EX / 2), r = s / (T::EX / 2);
     int ey = r % T::EY, rr = r / T::EY;
@@ -539,21 +522,22 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
   cplx<R> p[4][3], h[2][3], uh[2][3], U[3][3];
   const R one = (R)0.5;
 
-  // LDS slot of the neighbor at ext coords (ex,ey,ez,et); caller
-  // guarantees it holds an in-parity site
-  auto lds_read = [&](int ex, int ey, int ez, int et) {
-    int row = (et * T::EZ + ez) * T::EY + ey;
-    int slot = row * (T::EX / 2) + (ex >> 1);
-    S tmp[24];
-#pragma unroll
-    for (int ch = 0; ch < NCH; ++ch)
-      load_chunk<S, W>(lds[ch][slot], tmp + ch * W);
-    R sc = lnrm[slot];
-#pragma unroll
-    for (int kk = 0; kk < 12; ++kk)
-      p[kk / 3][kk % 3] = {sc * qa_tor<R>(tmp[2 * kk]),
-                           sc * qa_tor<R>(tmp[2 * kk + 1])};
-  };
+  // LDS slot of the neighbor at ext coords; the coords always hold an
+  // in-parity site (macro, not lambda: capture state stays dead)
+#define QA_LDS_READ(EXC, EYC, EZC, ETC)                                   \
+  {                                                                       \
+    int slot_ = (((ETC) * T::EZ + (EZC)) * T::EY + (EYC)) * (T::EX / 2) + \
+                ((EXC) >> 1);                                             \
+    S tmp_[24];                                                           \
+    _Pragma("unroll")                                                     \
+    for (int ch_ = 0; ch_ < NCH; ++ch_)                                   \
+      load_chunk<S, W>(lds[ch_][slot_], tmp_ + ch_ * W);                  \
+    R sc_ = lnrm[slot_];                                                  \
+    _Pragma("unroll")                                                     \
+    for (int kk_ = 0; kk_ < 12; ++kk_)                                    \
+      p[kk_ / 3][kk_ % 3] = {sc_ * qa_tor<R>(tmp_[2 * kk_]),              \
+                             sc_ * qa_tor<R>(tmp_[2 * kk_ + 1])};         \
+  }
 
   const int e0 = qa_xc[0] - ox + 1, e1 = qa_xc[1] - oy + 1,
             e2 = qa_xc[2] - oz + 1, e3 = qa_xc[3] - ot + 1;
@@ -563,26 +547,25 @@ __global__ __launch_bounds__(256, 2) void k_dslash_wilson_lds(
     EXP;                                                                  \
     if constexpr (!DAG) proj_##MU##_0(h, p);                              \
     else proj_##MU##_1(h, p);                                             \
-    if constexpr (GPRE) g.template decode_raw<MU>(graw[2 * MU], U);       \
-    else g.template load<MU>(U, i);                                       \
+    g.template load<MU>(U, i);                                            \
     su3_mul_half(uh, U, h);                                               \
     if constexpr (!DAG) recon_##MU##_0(acc, uh, one);                     \
     else recon_##MU##_1(acc, uh, one);                                    \
     EXM;                                                                  \
     if constexpr (!DAG) proj_##MU##_1(h, p);                              \
     else proj_##MU##_0(h, p);                                             \
-    if constexpr (GPRE) g.template decode_raw<MU>(graw[2 * MU + 1], U);   \
-    else g.template load_o<MU>(U, jm[MU]);                                \
+    g.template load_o<MU>(U, neighbor_cb(qa_xc, MU, -1, d));             \
     su3_dagmul_half(uh, U, h);                                            \
     if constexpr (!DAG) recon_##MU##_1(acc, uh, one);                     \
     else recon_##MU##_0(acc, uh, one);                                    \
   }
 
-  QA_LDS_DIR(0, lds_read(e0 + 1, e1, e2, e3), lds_read(e0 - 1, e1, e2, e3))
-  QA_LDS_DIR(1, lds_read(e0, e1 + 1, e2, e3), lds_read(e0, e1 - 1, e2, e3))
-  QA_LDS_DIR(2, lds_read(e0, e1, e2 + 1, e3), lds_read(e0, e1, e2 - 1, e3))
-  QA_LDS_DIR(3, lds_read(e0, e1, e2, e3 + 1), lds_read(e0, e1, e2, e3 - 1))
+  QA_LDS_DIR(0, QA_LDS_READ(e0 + 1, e1, e2, e3), QA_LDS_READ(e0 - 1, e1, e2, e3))
+  QA_LDS_DIR(1, QA_LDS_READ(e0, e1 + 1, e2, e3), QA_LDS_READ(e0, e1 - 1, e2, e3))
+  QA_LDS_DIR(2, QA_LDS_READ(e0, e1, e2 + 1, e3), QA_LDS_READ(e0, e1, e2 - 1, e3))
+  QA_LDS_DIR(3, QA_LDS_READ(e0, e1, e2, e3 + 1), QA_LDS_READ(e0, e1, e2, e3 - 1))
 #undef QA_LDS_DIR
+#undef QA_LDS_READ
 
   if constexpr (MODE == CLOV_POST) {
     R diag[2][6];
