@@ -445,14 +445,20 @@ __global__ __launch_bounds__(256) void k_clover_apply(
 // CDNA4 the LDS pipe is the only way past the load-issue ceiling.)
 // ---------------------------------------------------------------------------
 struct LdsTile {
-  static constexpr int BX = 4, BY = 4, BZ = 4, BT = 8;   // 512 lex sites
+  // 4^4 tile: 128 output sites/WG, ~34 KB LDS at half -> 4 workgroups/CU.
+  // The compute phase's gauge loads serialize within one wave (the
+  // allocation sits exactly at the 2-wave 256-VGPR line, so the
+  // scheduler cannot hoist loads across directions); four independent
+  // workgroups per CU restore the memory-level parallelism instead.
+  static constexpr int BX = 4, BY = 4, BZ = 4, BT = 8;
   static constexpr int EX = BX + 2, EY = BY + 2, EZ = BZ + 2, ET = BT + 2;
   static constexpr int NROW = EY * EZ * ET;              // rows of EX sites
   static constexpr int NSLOT = NROW * (EX / 2);          // in-parity sites
+  static constexpr int NOUT = BX * BY * BZ * BT / 2;     // threads/WG
 };
 
 template <typename Prec, int RECON, bool DAG, int MODE, bool XPAY>
-__global__ __launch_bounds__(256, 1) void k_dslash_wilson_lds(
+__global__ __launch_bounds__(LdsTile::NOUT, 1) void k_dslash_wilson_lds(
     SpinorAcc<Prec> out, SpinorAcc<Prec> in, GaugeAcc<Prec, RECON> g,
     CloverAcc<Prec> clov, LatDims d, int parity, typename Prec::Real a,
     SpinorAcc<Prec> x, typename Prec::Real br, typename Prec::Real bi) {
