@@ -73,6 +73,24 @@ __device__ __forceinline__ R qa_tor(const fp8s &v) {
   return (R)__half(__hip_cvt_fp8_to_halfraw(v.b, __HIP_E4M3));
 }
 
+// decode two adjacent stored values (one complex) at once; the fp8
+// specialization uses the CDNA4 packed converter v_cvt_pk_f32_fp8
+// (gfx950 native OCP e4m3) — one VALU op for the pair instead of two
+// byte-wise __hip_cvt chains (the quarter dslash is decode-VALU bound)
+template <typename R, typename S>
+__device__ __forceinline__ void qa_tor_pair(const S *p, R &a, R &b) {
+  a = qa_tor<R>(p[0]);
+  b = qa_tor<R>(p[1]);
+}
+template <>
+__device__ __forceinline__ void qa_tor_pair<float, fp8s>(const fp8s *p,
+                                                         float &a, float &b) {
+  unsigned short u = (unsigned short)((p[1].b << 8) | p[0].b);
+  auto v = __builtin_amdgcn_cvt_pk_f32_fp8(u, false);
+  a = v[0];
+  b = v[1];
+}
+
 template <typename S, typename R>
 __device__ __forceinline__ S qa_tos(const R &v) { return (S)v; }
 template <>
@@ -152,8 +170,11 @@ struct SpinorAcc {
     R scale = (R)1;
     if constexpr (Prec::has_norm) scale = norm[g];
 #pragma unroll
-    for (int k = 0; k < NCPLX; ++k)
-      out[k] = {scale * qa_tor<R>(tmp[2 * k]), scale * qa_tor<R>(tmp[2 * k + 1])};
+    for (int k = 0; k < NCPLX; ++k) {
+      R re, im;
+      qa_tor_pair<R, S>(tmp + 2 * k, re, im);
+      out[k] = {scale * re, scale * im};
+    }
   }
 
   __device__ __forceinline__ void store_v(const cplx<R> in[NCPLX], long g) const {
@@ -308,8 +329,11 @@ struct GaugeAcc {
       return;
     }
 #pragma unroll
-    for (int k = 0; k < RECON / 2; ++k)
-      u[k / 3][k % 3] = {qa_tor<R>(tmp[off + 2 * k]), qa_tor<R>(tmp[off + 2 * k + 1])};
+    for (int k = 0; k < RECON / 2; ++k) {
+      R re, im;
+      qa_tor_pair<R, S>(tmp + off + 2 * k, re, im);
+      u[k / 3][k % 3] = {re, im};
+    }
     if constexpr (RECON == 12) {
 #pragma unroll
       for (int c = 0; c < 3; ++c) {

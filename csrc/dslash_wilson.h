@@ -66,8 +66,11 @@ struct CloverAcc {
 #pragma unroll
       for (int k = 0; k < 6; ++k) diag[b][k] = qa_tor<R>(tmp[36 * b + k]);
 #pragma unroll
-      for (int k = 0; k < 15; ++k)
-        tri[b][k] = {qa_tor<R>(tmp[36 * b + 6 + 2 * k]), qa_tor<R>(tmp[36 * b + 6 + 2 * k + 1])};
+      for (int k = 0; k < 15; ++k) {
+        R re, im;
+        qa_tor_pair<R, S>(tmp + 36 * b + 6 + 2 * k, re, im);
+        tri[b][k] = {re, im};
+      }
     }
   }
 };
@@ -540,9 +543,11 @@ __global__ __launch_bounds__(LdsTile::NOUT, 1) void k_dslash_wilson_lds(
       load_chunk<S, W>(lds[ch_][slot_], tmp_ + ch_ * W);                  \
     R sc_ = lnrm[slot_];                                                  \
     _Pragma("unroll")                                                     \
-    for (int kk_ = 0; kk_ < 12; ++kk_)                                    \
-      p[kk_ / 3][kk_ % 3] = {sc_ * qa_tor<R>(tmp_[2 * kk_]),              \
-                             sc_ * qa_tor<R>(tmp_[2 * kk_ + 1])};         \
+    for (int kk_ = 0; kk_ < 12; ++kk_) {                                  \
+      R re_, im_;                                                         \
+      qa_tor_pair<R, S>(tmp_ + 2 * kk_, re_, im_);                        \
+      p[kk_ / 3][kk_ % 3] = {sc_ * re_, sc_ * im_};                       \
+    }                                                                     \
   }
 
   const int e0 = qa_xc[0] - ox + 1, e1 = qa_xc[1] - oy + 1,

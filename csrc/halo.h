@@ -73,7 +73,11 @@ struct GhostAcc {
     if constexpr (Prec::has_norm) scale = nrm[2 * mu + dir][f];
 #pragma unroll
     for (int k = 0; k < NCPLX; ++k)
-      out[k] = {scale * qa_tor<R>(tmp[2 * k]), scale * qa_tor<R>(tmp[2 * k + 1])};
+      {
+        R re, im;
+        qa_tor_pair<R, S>(tmp + 2 * k, re, im);
+        out[k] = {scale * re, scale * im};
+      }
   }
 
   // Wilson half-spinor view (NCOMP == 12)
@@ -100,7 +104,11 @@ struct GhostAcc {
     auto *out = reinterpret_cast<cplx<R> *>(h);
 #pragma unroll
     for (int k = 0; k < NCPLX; ++k)
-      out[k] = {scale * qa_tor<R>(tmp[2 * k]), scale * qa_tor<R>(tmp[2 * k + 1])};
+      {
+        R re, im;
+        qa_tor_pair<R, S>(tmp + 2 * k, re, im);
+        out[k] = {scale * re, scale * im};
+      }
   }
 };
 
