@@ -100,7 +100,7 @@ def main():
                          "(hundreds of iterations), the honest "
                          "time-to-solution regime")
     ap.add_argument("--sloppy", type=str, default="half",
-                    choices=["double", "single", "half"])
+                    choices=["double", "single", "half", "quarter"])
     ap.add_argument("--seed", type=int, default=777)
     ap.add_argument("--device", default=None,
                     help="override (cpu for harness tests; default cuda:LOCAL_RANK)")
