@@ -92,5 +92,52 @@ module quda_amd
       real(c_double) :: plaq(3)
       integer(c_int) :: ierr
     end function
+
+    function gaugeObservablesQuda(plaq, qcharge, energy) &
+        bind(c, name="gaugeObservablesQuda") result(ierr)
+      import :: c_double, c_int
+      real(c_double) :: plaq(3), qcharge, energy(2)
+      integer(c_int) :: ierr
+    end function
+
+    function performGaugeSmearQuda(stype, n_steps, coeff) &
+        bind(c, name="performGaugeSmearQuda") result(ierr)
+      import :: c_double, c_int
+      integer(c_int), value :: stype, n_steps
+      real(c_double), value :: coeff
+      integer(c_int) :: ierr
+    end function
+
+    function computeGaugeForceQuda(h_mom, beta) &
+        bind(c, name="computeGaugeForceQuda") result(ierr)
+      import :: c_ptr, c_double, c_int
+      type(c_ptr), value :: h_mom
+      real(c_double), value :: beta
+      integer(c_int) :: ierr
+    end function
+
+    function updateGaugeFieldQuda(h_mom, dt) &
+        bind(c, name="updateGaugeFieldQuda") result(ierr)
+      import :: c_ptr, c_double, c_int
+      type(c_ptr), value :: h_mom
+      real(c_double), value :: dt
+      integer(c_int) :: ierr
+    end function
+
+    function momActionQuda(action, h_mom) &
+        bind(c, name="momActionQuda") result(ierr)
+      import :: c_ptr, c_double, c_int
+      real(c_double) :: action
+      type(c_ptr), value :: h_mom
+      integer(c_int) :: ierr
+    end function
+
+    function computeGaugeFixingOVRQuda(gauge_dir, max_iter, tol) &
+        bind(c, name="computeGaugeFixingOVRQuda") result(ierr)
+      import :: c_double, c_int
+      integer(c_int), value :: gauge_dir, max_iter
+      real(c_double), value :: tol
+      integer(c_int) :: ierr
+    end function
   end interface
 end module quda_amd

@@ -53,6 +53,35 @@ int main(void) {
   for (size_t i = 0; i < ns; ++i) b[i].re = (double)((i * 2654435761u) % 97) / 97.0 - 0.5;
   invert_quda_(x, b, &ip, &ierr);
   if (ierr || ip.true_res > 1e-6 || ip.iter <= 0) return 3;
+
+  /* round-2 Fortran surface: observables + smearing + HMC + eigensolve */
+  double qtop, energy[2];
+  gauge_observables_quda_(plaq, &qtop, energy, &ierr);
+  if (ierr || fabs(plaq[0] - 1.0) > 1e-12) return 4;
+  int stype = 1 /* stout */, nstep = 1;
+  double coeff = 0.05;
+  perform_gauge_smear_quda_(&stype, &nstep, &coeff, &ierr);
+  if (ierr) return 5;
+  project_su3_quda_(&ierr);
+  if (ierr) return 6;
+  cplx *mom = (cplx *)calloc((size_t)4 * 2 * VCB * 9, sizeof(cplx));
+  long seed = 12;
+  gauss_mom_quda_(mom, &seed, &ierr);
+  if (ierr) return 7;
+  double mact = 0;
+  mom_action_quda_(&mact, mom, &ierr);
+  if (ierr || mact <= 0) return 8;
+  QudaAmdEigParam ep;
+  new_quda_eig_param_(&ep);
+  ep.n_ev = 2;
+  ep.n_kr = 8;
+  ep.tol = 1e-4;
+  ip.solution_type = QUDA_AMD_MATPC_SOLUTION;
+  double er[2], ei[2];
+  eigensolve_quda_(er, ei, &ip, &ep, &ierr);
+  if (ierr || er[0] <= 0) return 9;
+  free(mom);
+
   printf("fortran_interface_test: solve %d iters res %.2e — ALL PASSED\n",
          ip.iter, ip.true_res);
   end_quda_();
