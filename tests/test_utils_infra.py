@@ -143,3 +143,38 @@ def test_gauge_io_roundtrip(tmp_path):
     import pytest as _pt
     with _pt.raises(IOError):
         load_gauge(p)
+
+
+def test_param_check_and_print():
+    """check_params.h role: validation lists EVERY violation; print dumps
+    all fields."""
+    import pytest
+    from quda_amd import api
+    from quda_amd.utils.params import (check_eig_param, check_gauge_param,
+                                       check_invert_param, print_gauge_param,
+                                       print_invert_param)
+    gp = api.GaugeParam()
+    check_gauge_param(gp)
+    s = print_gauge_param(gp)
+    assert "anisotropy" in s and "reconstruct_sloppy" in s
+    bad = api.GaugeParam(X=(7, 8, 8, 8), reconstruct="nine",
+                         t_boundary="open", anisotropy=-1.0)
+    with pytest.raises(ValueError) as ei:
+        check_gauge_param(bad)
+    msg = str(ei.value)
+    assert "even" in msg and "nine" in msg and "open" in msg and "> 0" in msg
+
+    ip = api.InvertParam()
+    check_invert_param(ip)
+    assert "reliable_delta" in print_invert_param(ip)
+    with pytest.raises(ValueError) as ei:
+        check_invert_param(api.InvertParam(tol=-1, maxiter=0,
+                                           cuda_prec="half"))
+    assert "tol" in str(ei.value) and "maxiter" in str(ei.value)
+    with pytest.raises(ValueError):
+        check_invert_param(api.InvertParam(
+            dslash_type=api.DslashType.CLOVER, clover_csw=0.0))
+
+    check_eig_param(api.EigParam())
+    with pytest.raises(ValueError):
+        check_eig_param(api.EigParam(n_ev=8, n_kr=4))
